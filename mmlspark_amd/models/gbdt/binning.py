@@ -1,0 +1,66 @@
+"""Quantile binning — builds the per-feature bin mapper shared by all ranks.
+
+Equivalent of the dataset construction the reference delegates to
+``LGBM_DatasetCreateFromMat`` (dataset/DatasetAggregator.scala:335): quantile
+boundaries from a row sample, then the full matrix binned to uint8 in the
+feature-interleaved (ngroups, n_rows, 4) device layout consumed by the HIP
+histogram kernel.  Distributed: every rank contributes a fixed-size sample
+(all_gather), then computes identical boundaries — no broadcast needed.
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from ...ops import backend
+from ...parallel.comm import Comm
+
+
+class BinMapper:
+    """Per-feature quantile bin boundaries. upper_bounds[f, b] = largest value in bin b."""
+
+    def __init__(self, upper_bounds: torch.Tensor, n_bins: int):
+        self.upper_bounds = upper_bounds  # (nf, n_bins-1) float32, +inf padded
+        self.n_bins = n_bins
+        self.n_features = upper_bounds.shape[0]
+
+    @staticmethod
+    def fit(X: torch.Tensor, n_bins: int = 255, sample_size: int = 200_000,
+            comm: Optional[Comm] = None, seed: int = 0) -> "BinMapper":
+        n, nf = X.shape
+        gen = torch.Generator(device="cpu").manual_seed(seed)
+        k = min(n, max(1, sample_size // max(1, comm.world_size if comm else 1)))
+        idx = torch.randperm(n, generator=gen)[:k].to(X.device)
+        sample = X[idx]
+        if comm is not None and comm.is_distributed:
+            # equal-size shards so all_gather works; ranks with fewer rows pad
+            # by repeating (slight weight skew only in tiny-data tests)
+            if sample.shape[0] < k:
+                reps = (k + sample.shape[0] - 1) // max(1, sample.shape[0])
+                sample = sample.repeat(reps, 1)[:k]
+            sample = torch.cat(comm.all_gather(sample.contiguous()), dim=0)
+        qs = torch.linspace(0, 1, n_bins, device=sample.device)[1:]  # n_bins-1 cuts
+        sample = torch.nan_to_num(sample, nan=float("inf"))
+        ub = torch.quantile(sample.double(), qs.double(), dim=0).t().float()  # (nf, n_bins-1)
+        # strictly increasing boundaries; collapse duplicated quantiles
+        ub = torch.cummax(ub, dim=1).values
+        ub[:, -1] = float("inf")
+        return BinMapper(ub.contiguous(), n_bins)
+
+    def transform(self, X: torch.Tensor) -> torch.Tensor:
+        """(n, nf) float -> (ngroups, n, 4) uint8 feature-interleaved bins."""
+        return backend.bin_matrix(X, self.upper_bounds.to(X.device), self.n_bins)
+
+    def bin_upper_value(self, feature: int, b: int) -> float:
+        """Raw-value threshold for 'bin <= b' splits (used at predict time)."""
+        if b >= self.n_bins - 1:
+            return float("inf")
+        return float(self.upper_bounds[feature, b])
+
+    def state_dict(self):
+        return {"upper_bounds": self.upper_bounds.cpu(), "n_bins": self.n_bins}
+
+    @staticmethod
+    def from_state(state):
+        return BinMapper(state["upper_bounds"], int(state["n_bins"]))

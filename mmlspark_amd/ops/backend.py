@@ -1,0 +1,110 @@
+"""Device-op dispatch: HIP/CDNA4 extension on GPU tensors, torch reference on CPU.
+
+The HIP extension is built IN-TREE (mmlspark_amd/ops/_hip_ops*.so) by
+``python setup.py build_ext --inplace`` / ``__graft_entry__.build()`` so the
+.so travels to the GPU box with the repo snapshot.  On a CUDA/ROCm tensor the
+extension is REQUIRED — a missing .so raises instead of silently falling back
+to eager torch (the round-end check records which native libraries the GPU
+processes actually loaded).
+"""
+from __future__ import annotations
+
+import torch
+
+from . import cpu_ref
+
+_EXT = None
+_EXT_ERR: str = ""
+
+
+def _load_ext():
+    global _EXT, _EXT_ERR
+    if _EXT is not None:
+        return _EXT
+    try:
+        from . import _hip_ops  # built in-tree
+        _EXT = _hip_ops
+    except ImportError as e:
+        _EXT_ERR = str(e)
+        _EXT = False
+    return _EXT
+
+
+def hip_available() -> bool:
+    return bool(_load_ext()) and torch.cuda.is_available()
+
+
+def _require_ext():
+    ext = _load_ext()
+    if not ext:
+        raise RuntimeError(
+            "mmlspark_amd HIP extension is required for GPU tensors but is not "
+            "built (import error: %s). Run `python setup.py build_ext --inplace` "
+            "(PYTORCH_ROCM_ARCH=gfx950)." % _EXT_ERR)
+    return ext
+
+
+# --------------------------------------------------------------------------- GBDT
+def hist_build(binned_i4, rows, grad, hess, n_bins):
+    if binned_i4.is_cuda:
+        return _require_ext().hist_build(binned_i4, rows, grad, hess, n_bins)
+    return cpu_ref.hist_build(binned_i4, rows, grad, hess, n_bins)
+
+
+def partition_rows(binned_i4, rows, feature, threshold_bin):
+    if binned_i4.is_cuda:
+        return _require_ext().partition_rows(binned_i4, rows, feature, threshold_bin)
+    return cpu_ref.partition_rows(binned_i4, rows, feature, threshold_bin)
+
+
+def predict_forest(node_feature, node_threshold, node_left, node_right,
+                   node_value, tree_offsets, X, n_outputs,
+                   tree_weights=None, start_tree=0, num_iteration=-1):
+    if X.is_cuda:
+        n_trees = tree_offsets.numel() - 1
+        end_tree = n_trees if num_iteration < 0 else min(
+            n_trees, start_tree + num_iteration * n_outputs)
+        if tree_weights is None:
+            tree_weights = torch.ones(n_trees, dtype=torch.float32, device=X.device)
+        return _require_ext().predict_forest(
+            node_feature, node_threshold, node_left, node_right, node_value,
+            tree_offsets, tree_weights, X, n_outputs, start_tree, end_tree)
+    return cpu_ref.predict_forest(node_feature, node_threshold, node_left,
+                                  node_right, node_value, tree_offsets, X,
+                                  n_outputs, tree_weights, start_tree, num_iteration)
+
+
+def predict_leaf(node_feature, node_threshold, node_left, node_right,
+                 node_leaf_index, tree_offsets, X):
+    if X.is_cuda:
+        return _require_ext().predict_leaf(
+            node_feature, node_threshold, node_left, node_right,
+            node_leaf_index, tree_offsets, X)
+    return cpu_ref.predict_leaf(node_feature, node_threshold, node_left,
+                                node_right, node_leaf_index, tree_offsets, X)
+
+
+def bin_matrix(X, upper_bounds, n_bins):
+    if X.is_cuda:
+        return _require_ext().bin_matrix(X, upper_bounds, n_bins)
+    return cpu_ref.bin_matrix(X, upper_bounds, n_bins)
+
+
+# --------------------------------------------------------------------------- VW
+def vw_sgd_minibatch(indices, values, offsets, labels, weights_tbl, adaptive_tbl,
+                     lr, l2, power_t, loss: str, weight_decay_rounds: int = 1):
+    if weights_tbl.is_cuda:
+        return _require_ext().vw_sgd_minibatch(
+            indices, values, offsets, labels, weights_tbl, adaptive_tbl,
+            lr, l2, power_t, {"squared": 0, "logistic": 1, "hinge": 2}[loss])
+    from ..models.vw import sgd_ref
+    return sgd_ref.vw_sgd_minibatch(indices, values, offsets, labels,
+                                    weights_tbl, adaptive_tbl, lr, l2,
+                                    power_t, loss)
+
+
+def vw_predict(indices, values, offsets, weights_tbl):
+    if weights_tbl.is_cuda:
+        return _require_ext().vw_predict(indices, values, offsets, weights_tbl)
+    from ..models.vw import sgd_ref
+    return sgd_ref.vw_predict(indices, values, offsets, weights_tbl)

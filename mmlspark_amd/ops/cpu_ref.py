@@ -1,0 +1,117 @@
+"""Pure-torch CPU reference implementations of every device op.
+
+These are (a) the CPU execution path for GPU-less environments and
+(b) the numerics reference the HIP kernels are tested against
+(tests compare HIP output to these in fp32).
+"""
+from __future__ import annotations
+
+import torch
+
+
+def hist_build(binned_i4: torch.Tensor, rows: torch.Tensor, grad: torch.Tensor,
+               hess: torch.Tensor, n_bins: int) -> torch.Tensor:
+    """Per-(feature, bin) gradient/hessian/count sums over the given rows.
+
+    binned_i4: (ngroups, n_rows, 4) uint8 — feature-interleaved binned data,
+               feature f lives at [f // 4, :, f % 4].
+    rows:      (m,) int32/int64 row indices of the leaf.
+    grad/hess: (n_rows,) float32.
+    returns:   (ngroups*4, n_bins, 3) float32 [sum_grad, sum_hess, count].
+    """
+    ng = binned_i4.shape[0]
+    nf = ng * 4
+    r = rows.long()
+    b = binned_i4[:, r, :].permute(0, 2, 1).reshape(nf, -1).long()  # (nf, m)
+    g = grad[r].unsqueeze(0).expand(nf, -1)
+    h = hess[r].unsqueeze(0).expand(nf, -1)
+    hist = torch.zeros(nf, n_bins, 3, dtype=torch.float32, device=binned_i4.device)
+    hist[:, :, 0].scatter_add_(1, b, g)
+    hist[:, :, 1].scatter_add_(1, b, h)
+    hist[:, :, 2].scatter_add_(1, b, torch.ones_like(g))
+    return hist
+
+
+def partition_rows(binned_i4: torch.Tensor, rows: torch.Tensor, feature: int,
+                   threshold_bin: int):
+    """Split rows into (left, right): left ⇔ bin[feature] <= threshold_bin."""
+    g, j = feature // 4, feature % 4
+    bins = binned_i4[g, rows.long(), j]
+    mask = bins <= threshold_bin
+    return rows[mask], rows[~mask]
+
+
+def predict_forest(node_feature: torch.Tensor, node_threshold: torch.Tensor,
+                   node_left: torch.Tensor, node_right: torch.Tensor,
+                   node_value: torch.Tensor, tree_offsets: torch.Tensor,
+                   X: torch.Tensor, n_outputs: int,
+                   tree_weights: torch.Tensor = None,
+                   start_tree: int = 0, num_iteration: int = -1) -> torch.Tensor:
+    """Sum of per-tree leaf values over the ensemble. Vectorized traversal.
+
+    Flattened node arrays (concatenated trees); interior node: feature >= 0,
+    go left iff X[:, feature] <= threshold or X is NaN (missing → left);
+    leaf: feature == -1, value = node_value.
+    Trees are laid out round-robin over outputs: tree t contributes to
+    output t % n_outputs (LightGBM multiclass convention).
+    returns (n, n_outputs) float32 raw scores (no base score added).
+    """
+    n = X.shape[0]
+    n_trees = tree_offsets.numel() - 1
+    end_tree = n_trees if num_iteration < 0 else min(n_trees, start_tree + num_iteration * n_outputs)
+    out = torch.zeros(n, n_outputs, dtype=torch.float32, device=X.device)
+    for t in range(start_tree, end_tree):
+        base = int(tree_offsets[t])
+        idx = torch.full((n,), base, dtype=torch.long, device=X.device)
+        active = node_feature[idx] >= 0
+        while bool(active.any()):
+            f = node_feature[idx].clamp(min=0).long()
+            xv = X[torch.arange(n, device=X.device), f]
+            go_left = (xv <= node_threshold[idx]) | torch.isnan(xv)
+            nxt = torch.where(go_left, node_left[idx], node_right[idx]).long()
+            idx = torch.where(active, nxt + base, idx)
+            active = node_feature[idx] >= 0
+        w = 1.0 if tree_weights is None else float(tree_weights[t])
+        out[:, t % n_outputs] += w * node_value[idx]
+    return out
+
+
+def predict_leaf(node_feature, node_threshold, node_left, node_right,
+                 node_leaf_index, tree_offsets, X) -> torch.Tensor:
+    """Per-tree leaf index for each row: (n, n_trees) int32."""
+    n = X.shape[0]
+    n_trees = tree_offsets.numel() - 1
+    out = torch.zeros(n, n_trees, dtype=torch.int32, device=X.device)
+    for t in range(n_trees):
+        base = int(tree_offsets[t])
+        idx = torch.full((n,), base, dtype=torch.long, device=X.device)
+        active = node_feature[idx] >= 0
+        while bool(active.any()):
+            f = node_feature[idx].clamp(min=0).long()
+            xv = X[torch.arange(n, device=X.device), f]
+            go_left = (xv <= node_threshold[idx]) | torch.isnan(xv)
+            nxt = torch.where(go_left, node_left[idx], node_right[idx]).long()
+            idx = torch.where(active, nxt + base, idx)
+            active = node_feature[idx] >= 0
+        out[:, t] = node_leaf_index[idx]
+    return out
+
+
+def bin_matrix(X: torch.Tensor, upper_bounds: torch.Tensor,
+               n_bins: int) -> torch.Tensor:
+    """Quantile-bin a dense (n, nf) matrix into the (ngroups, n, 4) i4 layout.
+
+    upper_bounds: (nf, n_bins-1) float32 ascending per-feature boundaries
+    (+inf padded). bin = searchsorted(bounds, x) so x <= bounds[b] → bin<=b.
+    NaN maps to bin 0.
+    """
+    n, nf = X.shape
+    ng = (nf + 3) // 4
+    Xc = torch.nan_to_num(X, nan=-float("inf"))
+    bins = torch.searchsorted(upper_bounds.contiguous(),
+                              Xc.t().contiguous(), right=False)
+    bins = bins.clamp(max=n_bins - 1).to(torch.uint8)  # (nf, n)
+    out = torch.zeros(ng, n, 4, dtype=torch.uint8, device=X.device)
+    for f in range(nf):
+        out[f // 4, :, f % 4] = bins[f]
+    return out
