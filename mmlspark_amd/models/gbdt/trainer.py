@@ -133,10 +133,12 @@ class TreeGrower:
         self.bin_mapper = bin_mapper
         self.device = binned_i4.device
 
+    _sync_timers = bool(__import__("os").environ.get("MMLSPARK_AMD_SYNC_TIMERS"))
+
     def _hist(self, rows, grad, hess, reduce=True):
         t0 = time.perf_counter()
         h = backend.hist_build(self.binned, rows, grad, hess, self.cfg.max_bin)
-        if self.device.type == "cuda":
+        if self._sync_timers and self.device.type == "cuda":
             torch.cuda.synchronize()
         t1 = time.perf_counter()
         self.stats.hist_s += t1 - t0
@@ -178,10 +180,14 @@ class TreeGrower:
             gain[~feat_mask] = NEG_INF
         gain[self.nf:] = NEG_INF  # padding features
         per_f, per_bin = gain.max(dim=1)
-        bf = int(per_f.argmax())
-        bb = int(per_bin[bf])
-        bg = float(per_f[bf])
-        out = (bg, bf, bb, float(GL[bf, bb]), float(HL[bf, bb]), float(CL[bf, bb]))
+        bf_t = per_f.argmax()
+        bb_t = per_bin[bf_t]
+        # single device→host sync for all six scalars
+        packed = torch.stack([per_f[bf_t], bf_t.float(), bb_t.float(),
+                              GL[bf_t, bb_t], HL[bf_t, bb_t],
+                              CL[bf_t, bb_t]]).cpu()
+        bg, bf, bb, gl, hl, cl = packed.tolist()
+        out = (bg, int(bf), int(bb), gl, hl, cl)
         self.stats.split_s += time.perf_counter() - t0
         return out
 
@@ -319,6 +325,177 @@ def _goss_sample(grad, hess, cfg: TrainConfig, gen) -> torch.Tensor:
     return rows.to(torch.int32).sort().values
 
 
+class TrainingSession:
+    """Step-level training state: one ``step()`` = one boosting iteration
+    (the bench.py "step"; also the iteration-level checkpoint boundary)."""
+
+    def __init__(self, X: torch.Tensor, y: torch.Tensor, cfg: TrainConfig,
+                 objective: Objective, comm: Comm,
+                 weight: Optional[torch.Tensor] = None,
+                 group_sizes: Optional[torch.Tensor] = None,
+                 feature_names: Optional[List[str]] = None,
+                 init_booster: Optional[Booster] = None,
+                 binned_cache=None):
+        self.X, self.y, self.cfg, self.objective = X, y, cfg, objective
+        self.comm, self.weight = comm, weight
+        self.feature_names = feature_names
+        self.stats = TrainingStats()
+        self.device = X.device
+        self.n, self.nf = X.shape
+        self.K = objective.n_outputs
+        if hasattr(objective, "group_sizes"):
+            objective.group_sizes = group_sizes
+
+        if binned_cache is not None:
+            self.bin_mapper, self.binned = binned_cache
+        else:
+            self.bin_mapper = BinMapper.fit(X, n_bins=cfg.max_bin, comm=comm,
+                                            seed=cfg.seed)
+            self.binned = self.bin_mapper.transform(X)
+
+        device, n, K = self.device, self.n, self.K
+        if init_booster is not None and init_booster.trees:
+            base = torch.from_numpy(init_booster.base_score).to(device)
+            self.preds = init_booster.predict_raw(X)
+            self.trees: List[Tree] = list(init_booster.trees)
+            self.tree_w: List[float] = list(init_booster.tree_weights)
+        else:
+            base = self._global_init_score()
+            if base.numel() < K:
+                base = base.expand(K).contiguous()
+            self.preds = base.unsqueeze(0).expand(n, K).clone()
+            self.trees = []
+            self.tree_w = []
+        self.base = base
+        self.grower = TreeGrower(self.binned, self.nf, cfg, comm, self.stats,
+                                 self.bin_mapper)
+        self.gen = torch.Generator(device="cpu")
+        self.all_rows = torch.arange(n, dtype=torch.int32, device=device)
+        self.n_start_trees = len(self.trees)
+        self.it = 0
+
+    def _global_init_score(self):
+        y, weight, device = self.y, self.weight, self.device
+        obj, comm = self.objective, self.comm
+        if comm.is_distributed:
+            w = weight if weight is not None else torch.ones_like(
+                y, dtype=torch.float32)
+            sums = torch.stack([(y.float() * w).sum(), w.sum()]).to(device)
+            comm.all_reduce(sums)
+            if obj.name == "binary":
+                mean = (sums[0] / sums[1]).clamp(1e-6, 1 - 1e-6)
+                return torch.log(mean / (1 - mean)).reshape(1)
+            if obj.name in ("regression", "regression_l2"):
+                return (sums[0] / sums[1]).reshape(1)
+        return obj.init_score(y, weight).to(device)
+
+    # ------------------------------------------------------------------ step
+    def step(self):
+        cfg, gen, K, device = self.cfg, self.gen, self.K, self.device
+        trees, tree_w, preds = self.trees, self.tree_w, self.preds
+        it = self.it
+        rf_mode = cfg.boosting == "rf"
+        dart_mode = cfg.boosting == "dart"
+        goss_mode = cfg.boosting == "goss"
+
+        # ---- gradients ----------------------------------------------------
+        t0 = time.perf_counter()
+        dropped = []
+        if dart_mode and trees[self.n_start_trees:]:
+            gen.manual_seed(cfg.seed * 7919 + it)
+            if float(torch.rand(1, generator=gen)) >= cfg.skip_drop:
+                cand = list(range(self.n_start_trees, len(trees)))
+                mask = torch.rand(len(cand), generator=gen) < cfg.drop_rate
+                dropped = [cand[i] for i in range(len(cand)) if bool(mask[i])]
+                dropped = dropped[: cfg.max_drop]
+        if dropped:
+            drop_contrib = torch.zeros_like(preds)
+            for t in dropped:
+                k = (t - self.n_start_trees) % K
+                drop_contrib[:, k] += tree_w[t] * predict_tree_binned(
+                    trees[t], self.binned, device)
+            preds_used = preds - drop_contrib
+        else:
+            preds_used = preds
+        if rf_mode:
+            preds_used = self.base.unsqueeze(0).expand(self.n, K)
+        grad, hess = self.objective.grad_hess(preds_used, self.y, self.weight)
+        self.stats.grad_s += time.perf_counter() - t0
+
+        # ---- row sampling ---------------------------------------------------
+        gen.manual_seed(cfg.seed * 104729 + it * 31 + self.comm.rank)
+        if goss_mode and it >= 1:
+            rows_root = _goss_sample(grad, hess, cfg, gen)
+        elif (cfg.bagging_freq > 0 and cfg.bagging_fraction < 1.0
+              and it % cfg.bagging_freq == 0) or rf_mode:
+            frac = cfg.bagging_fraction if cfg.bagging_fraction < 1.0 else 0.632
+            m = max(1, int(frac * self.n))
+            perm = torch.randperm(self.n, generator=gen)[:m].to(device)
+            rows_root = perm.to(torch.int32).sort().values
+        else:
+            rows_root = self.all_rows
+
+        # ---- feature sampling (same seed on every rank) ---------------------
+        feat_mask = None
+        if cfg.feature_fraction < 1.0:
+            gen_f = torch.Generator(device="cpu")
+            gen_f.manual_seed(cfg.seed * 524287 + it)
+            kf = max(1, int(cfg.feature_fraction * self.nf))
+            sel = torch.randperm(self.nf, generator=gen_f)[:kf]
+            feat_mask = torch.zeros(self.grower.nf_pad, dtype=torch.bool,
+                                    device=device)
+            feat_mask[sel.to(device)] = True
+
+        # ---- one tree per output class --------------------------------------
+        new_trees = []
+        for k in range(K):
+            tree, leaves = self.grower.grow(rows_root, grad[:, k].contiguous(),
+                                            hess[:, k].contiguous(), feat_mask)
+            tree.shrinkage = 1.0 if rf_mode else cfg.learning_rate
+            if not rf_mode:
+                t0 = time.perf_counter()
+                for lf in leaves:
+                    if lf.rows.numel():
+                        w_leaf = float(tree.value[lf.node_id]) * tree.shrinkage
+                        preds[:, k].index_add_(
+                            0, lf.rows.long(),
+                            torch.full((lf.rows.numel(),), w_leaf, device=device))
+                self.stats.partition_s += time.perf_counter() - t0
+            new_trees.append(tree)
+
+        if dropped:
+            # DART normalization: dropped ×k/(k+1), new tree ×1/(k+1)
+            kdrop = len(dropped)
+            factor = kdrop / (kdrop + 1.0)
+            for t in dropped:
+                k = (t - self.n_start_trees) % K
+                delta = (factor - 1.0) * tree_w[t]
+                preds[:, k] += delta * predict_tree_binned(trees[t], self.binned,
+                                                           device)
+                tree_w[t] *= factor
+            for k, tr in enumerate(new_trees):
+                excess = (1.0 - 1.0 / (kdrop + 1.0)) * tr.shrinkage
+                preds[:, k] -= excess * predict_tree_binned(tr, self.binned, device)
+                tr.shrinkage *= 1.0 / (kdrop + 1.0)
+
+        for tr in new_trees:
+            trees.append(tr)
+            tree_w.append(1.0)
+        self.stats.iterations += 1
+        self.it += 1
+
+    # --------------------------------------------------------------- snapshot
+    def booster(self) -> Booster:
+        tw = list(self.tree_w)
+        if self.cfg.boosting == "rf" and self.stats.iterations:
+            tw = [1.0 / self.stats.iterations] * len(tw)
+        return Booster(list(self.trees), self.objective.name, self.K,
+                       self.base.cpu().numpy(), self.nf, self.feature_names,
+                       self.bin_mapper,
+                       sigmoid=getattr(self.objective, "sigmoid", 1.0),
+                       tree_weights=np.array(tw, dtype=np.float32))
+
+
 def train_booster(X: torch.Tensor, y: torch.Tensor, cfg: TrainConfig,
                   objective: Objective, comm: Comm,
                   weight: Optional[torch.Tensor] = None,
@@ -328,158 +505,23 @@ def train_booster(X: torch.Tensor, y: torch.Tensor, cfg: TrainConfig,
                   init_booster: Optional[Booster] = None,
                   metrics_fn=None,
                   binned_cache=None) -> (Booster, TrainingStats):
-    """Full training loop. X local shard (n, nf) float32; y (n,)."""
+    """Full training loop with eval + early stopping over TrainingSession."""
     t_start = time.perf_counter()
-    stats = TrainingStats()
-    device = X.device
-    n, nf = X.shape
-    K = objective.n_outputs
-
-    if hasattr(objective, "group_sizes"):
-        objective.group_sizes = group_sizes
-
-    if binned_cache is not None:
-        bin_mapper, binned = binned_cache
-    else:
-        bin_mapper = BinMapper.fit(X, n_bins=cfg.max_bin, comm=comm, seed=cfg.seed)
-        binned = bin_mapper.transform(X)
-
-    # init score: global (weighted) statistics via all_reduce of sums
-    if init_booster is not None and init_booster.trees:
-        base = torch.from_numpy(init_booster.base_score).to(device)
-        preds = init_booster.predict_raw(X)
-        trees: List[Tree] = list(init_booster.trees)
-        tree_w = list(init_booster.tree_weights)
-    else:
-        if comm.is_distributed:
-            w = weight if weight is not None else torch.ones_like(y, dtype=torch.float32)
-            sums = torch.stack([(y.float() * w).sum(), w.sum()]).to(device)
-            comm.all_reduce(sums)
-            mean = (sums[0] / sums[1]).clamp(1e-6, 1 - 1e-6) \
-                if objective.name == "binary" else sums[0] / sums[1]
-            if objective.name == "binary":
-                base = torch.log(mean / (1 - mean)).reshape(1)
-            elif objective.name in ("regression", "regression_l2"):
-                base = mean.reshape(1)
-            else:
-                base = objective.init_score(y, weight).to(device)
-        else:
-            base = objective.init_score(y, weight).to(device)
-        if base.numel() < K:
-            base = base.expand(K).contiguous()
-        preds = base.unsqueeze(0).expand(n, K).clone()
-        trees = []
-        tree_w = []
-
-    grower = TreeGrower(binned, nf, cfg, comm, stats, bin_mapper)
-    gen = torch.Generator(device="cpu")
-    all_rows = torch.arange(n, dtype=torch.int32, device=device)
-    rf_mode = cfg.boosting == "rf"
-    dart_mode = cfg.boosting == "dart"
-    goss_mode = cfg.boosting == "goss"
-    n_start_trees = len(trees)
-
+    session = TrainingSession(X, y, cfg, objective, comm, weight=weight,
+                              group_sizes=group_sizes,
+                              feature_names=feature_names,
+                              init_booster=init_booster,
+                              binned_cache=binned_cache)
+    stats = session.stats
     best_score = None
     best_iter = -1
     rounds_no_improve = 0
 
     for it in range(cfg.num_iterations):
-        # ---- gradients -----------------------------------------------------
-        t0 = time.perf_counter()
-        dropped = []
-        if dart_mode and trees[n_start_trees:]:
-            gen.manual_seed(cfg.seed * 7919 + it)
-            if float(torch.rand(1, generator=gen)) >= cfg.skip_drop:
-                cand = list(range(n_start_trees, len(trees)))
-                mask = torch.rand(len(cand), generator=gen) < cfg.drop_rate
-                dropped = [cand[i] for i in range(len(cand)) if bool(mask[i])]
-                dropped = dropped[: cfg.max_drop]
-        if dropped:
-            drop_contrib = torch.zeros_like(preds)
-            for t in dropped:
-                k = (t - n_start_trees) % K
-                drop_contrib[:, k] += tree_w[t] * predict_tree_binned(
-                    trees[t], binned, device)
-            preds_used = preds - drop_contrib
-        else:
-            preds_used = preds
-        grad, hess = objective.grad_hess(preds_used if not rf_mode
-                                         else preds.detach() * 0 + base,
-                                         y, weight)
-        stats.grad_s += time.perf_counter() - t0
-
-        # ---- row sampling --------------------------------------------------
-        gen.manual_seed(cfg.seed * 104729 + it * 31 + comm.rank)
-        if goss_mode and it >= 1:
-            rows_root = _goss_sample(grad, hess, cfg, gen)
-        elif (cfg.bagging_freq > 0 and cfg.bagging_fraction < 1.0
-              and it % cfg.bagging_freq == 0) or rf_mode:
-            frac = cfg.bagging_fraction if cfg.bagging_fraction < 1.0 else 0.632
-            m = max(1, int(frac * n))
-            perm = torch.randperm(n, generator=gen)[:m].to(device)
-            rows_root = perm.to(torch.int32).sort().values
-        else:
-            rows_root = all_rows
-
-        # ---- feature sampling (same on every rank) -------------------------
-        feat_mask = None
-        if cfg.feature_fraction < 1.0:
-            gen_f = torch.Generator(device="cpu")
-            gen_f.manual_seed(cfg.seed * 524287 + it)
-            kf = max(1, int(cfg.feature_fraction * nf))
-            sel = torch.randperm(nf, generator=gen_f)[:kf]
-            feat_mask = torch.zeros(grower.nf_pad, dtype=torch.bool, device=device)
-            feat_mask[sel.to(device)] = True
-
-        # ---- one tree per output class -------------------------------------
-        new_trees = []
-        for k in range(K):
-            tree, leaves = grower.grow(rows_root, grad[:, k].contiguous(),
-                                       hess[:, k].contiguous(), feat_mask)
-            shrink = 1.0 if rf_mode else cfg.learning_rate
-            tree.shrinkage = shrink
-            # update predictions leaf-by-leaf with local row lists
-            if not rf_mode:
-                t0 = time.perf_counter()
-                for lf in leaves:
-                    if lf.rows.numel():
-                        w_leaf = float(tree.value[lf.node_id]) * shrink
-                        preds[:, k].index_add_(
-                            0, lf.rows.long(),
-                            torch.full((lf.rows.numel(),), w_leaf,
-                                       device=device))
-                stats.partition_s += time.perf_counter() - t0
-            new_trees.append(tree)
-
-        if dropped:
-            # DART normalization: dropped scaled by k/(k+1), new by 1/(k+1)
-            kdrop = len(dropped)
-            factor = kdrop / (kdrop + 1.0)
-            for t in dropped:
-                k = (t - n_start_trees) % K
-                delta = (factor - 1.0) * tree_w[t]
-                preds[:, k] += delta * predict_tree_binned(trees[t], binned, device)
-                tree_w[t] *= factor
-            for tr in new_trees:
-                tr.shrinkage *= 1.0 / (kdrop + 1.0)
-            # rebuild preds contribution of the new trees is already applied at
-            # full shrinkage; remove the excess
-            for k, tr in enumerate(new_trees):
-                excess = 1.0 - 1.0 / (kdrop + 1.0)
-                preds[:, k] -= excess * cfg.learning_rate * predict_tree_binned(
-                    tr, binned, device) / max(tr.shrinkage, 1e-12) * tr.shrinkage
-
-        for tr in new_trees:
-            trees.append(tr)
-            tree_w.append(1.0)
-        stats.iterations += 1
-
-        # ---- eval & early stopping ----------------------------------------
+        session.step()
         if metrics_fn is not None and (valid_sets or cfg.is_provide_training_metric):
             t0 = time.perf_counter()
-            booster_now = Booster(trees, objective.name, K,
-                                  base.cpu().numpy(), nf, feature_names,
-                                  bin_mapper, tree_weights=np.array(tree_w))
+            booster_now = session.booster()
             entry = {"iteration": it}
             score = None
             for vi, (Xv, yv, wv) in enumerate(valid_sets or []):
@@ -501,14 +543,7 @@ def train_booster(X: torch.Tensor, y: torch.Tensor, cfg: TrainConfig,
                     if rounds_no_improve >= cfg.early_stopping_round:
                         break
 
-    if rf_mode:
-        w = 1.0 / max(1, stats.iterations)
-        tree_w = [w] * len(tree_w)
-
     stats.total_s = time.perf_counter() - t_start
-    booster = Booster(trees, objective.name, K, base.cpu().numpy(), nf,
-                      feature_names, bin_mapper,
-                      sigmoid=getattr(objective, "sigmoid", 1.0),
-                      tree_weights=np.array(tree_w, dtype=np.float32))
+    booster = session.booster()
     booster.best_iteration = best_iter
     return booster, stats

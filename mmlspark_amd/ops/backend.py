@@ -52,8 +52,8 @@ def hist_build(binned_i4, rows, grad, hess, n_bins):
 
 
 def partition_rows(binned_i4, rows, feature, threshold_bin):
-    if binned_i4.is_cuda:
-        return _require_ext().partition_rows(binned_i4, rows, feature, threshold_bin)
+    # order-preserving masked select — torch's cub path is already optimal on
+    # ROCm, and keeping row lists sorted keeps later grad/hess gathers coalesced
     return cpu_ref.partition_rows(binned_i4, rows, feature, threshold_bin)
 
 

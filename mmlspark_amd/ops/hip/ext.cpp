@@ -1,0 +1,135 @@
+// Python bindings for the MI355X HIP kernels (pure-ROCm torch extension —
+// no hipify, no CUDA shims: this file uses torch's native c10::hip API and
+// links the hipcc-compiled kernel objects).
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#define CHECK_DEV(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
+#define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
+
+extern "C" {
+void launch_hist_build(const void*, long, const int*, long, const float*,
+                       const float*, float*, int, int, hipStream_t);
+void launch_predict_forest(const int*, const float*, const int*, const int*,
+                           const float*, const long*, const float*,
+                           const float*, long, int, float*, int, int, int,
+                           hipStream_t);
+void launch_predict_leaf(const int*, const float*, const int*, const int*,
+                         const int*, const long*, const float*, long, int,
+                         int*, int, hipStream_t);
+void launch_bin_matrix(const float*, const float*, long, int, int, int, void*,
+                       hipStream_t);
+void launch_vw_sgd(const int*, const float*, const long*, const float*,
+                   float*, float*, float, float, float, int, long, float*,
+                   hipStream_t);
+void launch_vw_predict(const int*, const float*, const long*, const float*,
+                       long, float*, hipStream_t);
+}
+
+static hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+torch::Tensor hist_build(torch::Tensor binned_i4, torch::Tensor rows,
+                         torch::Tensor grad, torch::Tensor hess, long n_bins) {
+  CHECK_DEV(binned_i4); CHECK_CONTIG(binned_i4);
+  CHECK_DEV(rows); CHECK_CONTIG(rows);
+  CHECK_DEV(grad); CHECK_CONTIG(grad);
+  CHECK_DEV(hess); CHECK_CONTIG(hess);
+  TORCH_CHECK(binned_i4.dtype() == torch::kUInt8, "binned must be uint8");
+  TORCH_CHECK(rows.dtype() == torch::kInt32, "rows must be int32");
+  const long ngroups = binned_i4.size(0);
+  const long n_rows = binned_i4.size(1);
+  auto hist = torch::zeros({ngroups * 4, n_bins, 3},
+                           grad.options().dtype(torch::kFloat32));
+  launch_hist_build(binned_i4.data_ptr(), n_rows, rows.data_ptr<int>(),
+                    rows.numel(), grad.data_ptr<float>(),
+                    hess.data_ptr<float>(), hist.data_ptr<float>(),
+                    (int)n_bins, (int)ngroups, cur_stream());
+  return hist;
+}
+
+torch::Tensor predict_forest(torch::Tensor feat, torch::Tensor thr,
+                             torch::Tensor left, torch::Tensor right,
+                             torch::Tensor val, torch::Tensor offsets,
+                             torch::Tensor tw, torch::Tensor X,
+                             long n_outputs, long t0, long t1) {
+  CHECK_DEV(X); CHECK_CONTIG(X);
+  const long n = X.size(0);
+  const long nf = X.size(1);
+  auto out = torch::zeros({n, n_outputs}, X.options().dtype(torch::kFloat32));
+  TORCH_CHECK(offsets.dtype() == torch::kInt64, "offsets must be int64");
+  launch_predict_forest(feat.data_ptr<int>(), thr.data_ptr<float>(),
+                        left.data_ptr<int>(), right.data_ptr<int>(),
+                        val.data_ptr<float>(), offsets.data_ptr<long>(),
+                        tw.data_ptr<float>(), X.data_ptr<float>(), n, (int)nf,
+                        out.data_ptr<float>(), (int)n_outputs, (int)t0,
+                        (int)t1, cur_stream());
+  return out;
+}
+
+torch::Tensor predict_leaf(torch::Tensor feat, torch::Tensor thr,
+                           torch::Tensor left, torch::Tensor right,
+                           torch::Tensor leaf_index, torch::Tensor offsets,
+                           torch::Tensor X) {
+  CHECK_DEV(X); CHECK_CONTIG(X);
+  const long n = X.size(0);
+  const long nf = X.size(1);
+  const long n_trees = offsets.numel() - 1;
+  auto out = torch::zeros({n, n_trees}, X.options().dtype(torch::kInt32));
+  launch_predict_leaf(feat.data_ptr<int>(), thr.data_ptr<float>(),
+                      left.data_ptr<int>(), right.data_ptr<int>(),
+                      leaf_index.data_ptr<int>(), offsets.data_ptr<long>(),
+                      X.data_ptr<float>(), n, (int)nf, out.data_ptr<int>(),
+                      (int)n_trees, cur_stream());
+  return out;
+}
+
+torch::Tensor bin_matrix(torch::Tensor X, torch::Tensor ub, long n_bins) {
+  CHECK_DEV(X); CHECK_CONTIG(X);
+  CHECK_DEV(ub); CHECK_CONTIG(ub);
+  const long n = X.size(0);
+  const long nf = X.size(1);
+  const long ngroups = (nf + 3) / 4;
+  auto out = torch::zeros({ngroups, n, 4}, X.options().dtype(torch::kUInt8));
+  launch_bin_matrix(X.data_ptr<float>(), ub.data_ptr<float>(), n, (int)nf,
+                    (int)n_bins, (int)ngroups, out.data_ptr(), cur_stream());
+  return out;
+}
+
+torch::Tensor vw_sgd_minibatch(torch::Tensor idx, torch::Tensor val,
+                               torch::Tensor off, torch::Tensor label,
+                               torch::Tensor w_tbl, torch::Tensor g_tbl,
+                               double lr, double l2, double power_t,
+                               long loss) {
+  CHECK_DEV(w_tbl); CHECK_CONTIG(w_tbl);
+  const long n_ex = off.numel() - 1;
+  auto preds = torch::zeros({n_ex}, w_tbl.options());
+  launch_vw_sgd(idx.data_ptr<int>(), val.data_ptr<float>(),
+                off.data_ptr<long>(), label.data_ptr<float>(),
+                w_tbl.data_ptr<float>(), g_tbl.data_ptr<float>(), (float)lr,
+                (float)l2, (float)power_t, (int)loss, n_ex,
+                preds.data_ptr<float>(), cur_stream());
+  return preds;
+}
+
+torch::Tensor vw_predict(torch::Tensor idx, torch::Tensor val,
+                         torch::Tensor off, torch::Tensor w_tbl) {
+  CHECK_DEV(w_tbl); CHECK_CONTIG(w_tbl);
+  const long n_ex = off.numel() - 1;
+  auto out = torch::zeros({n_ex}, w_tbl.options());
+  launch_vw_predict(idx.data_ptr<int>(), val.data_ptr<float>(),
+                    off.data_ptr<long>(), w_tbl.data_ptr<float>(), n_ex,
+                    out.data_ptr<float>(), cur_stream());
+  return out;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("hist_build", &hist_build, "per-leaf (feature,bin) grad/hess/count histogram");
+  m.def("predict_forest", &predict_forest, "GBDT ensemble raw scores");
+  m.def("predict_leaf", &predict_leaf, "GBDT per-tree leaf indices");
+  m.def("bin_matrix", &bin_matrix, "quantile binning to interleaved uint8");
+  m.def("vw_sgd_minibatch", &vw_sgd_minibatch, "adaptive sparse SGD minibatch");
+  m.def("vw_predict", &vw_predict, "sparse linear predict");
+}

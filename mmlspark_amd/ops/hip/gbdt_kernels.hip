@@ -1,0 +1,234 @@
+// CDNA4 (gfx950 / MI355X) kernels for the GBDT path — written HIP-first.
+//
+// These implement, natively on MI355X, the compute the reference delegates to
+// LightGBM's CPU library behind LGBM_BoosterUpdateOneIter /
+// LGBM_BoosterPredictForMat: per-leaf feature-histogram build,
+// ensemble traversal scoring, and quantile binning.
+//
+// Design notes (cf. /opt/skills/guides/cdna_hip_programming.md):
+//  * wave = 64; blocks are 256 threads (4 waves).
+//  * hist_build: LDS-staged per-(feature,bin) accumulation — each workgroup
+//    owns GPB*4 features × n_bins bins in LDS (~49 KB at GPB=4, 3 blocks/CU)
+//    and a contiguous row chunk; flush with device atomics. Binned data is
+//    feature-interleaved uchar4 so one 4-byte load yields 4 features' bins.
+//  * grid is sized ≫256 workgroups (row-chunk × feature-block 2-D grid) to
+//    fill 8 XCDs; row lists are kept sorted so grad/hess gathers are
+//    near-coalesced.
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#define WAVE 64
+#define DEV_INLINE __device__ __forceinline__
+
+// ---------------------------------------------------------------- histogram
+// binned: (ngroups, n_rows) uchar4; rows: (m,) i32 sorted; grad/hess: (n_rows,)
+// hist:   (ngroups*4, n_bins, 3) f32, pre-zeroed.
+template <int GPB>
+__global__ void hist_build_k(const uchar4* __restrict__ binned, long n_rows,
+                             const int* __restrict__ rows, long m,
+                             const float* __restrict__ grad,
+                             const float* __restrict__ hess,
+                             float* __restrict__ hist, int n_bins,
+                             int ngroups, long chunk) {
+  extern __shared__ float lds[];  // [GPB*4][n_bins][3]
+  const int tid = threadIdx.x;
+  const int nfb = GPB * 4;
+  const int lds_elems = nfb * n_bins * 3;
+  for (int i = tid; i < lds_elems; i += blockDim.x) lds[i] = 0.0f;
+  __syncthreads();
+
+  const int gq0 = blockIdx.y * GPB;  // first feature-group of this block
+  const long start = (long)blockIdx.x * chunk;
+  const long end = min(start + chunk, m);
+
+  for (long i = start + tid; i < end; i += blockDim.x) {
+    const int r = rows[i];
+    const float g = grad[r];
+    const float h = hess[r];
+#pragma unroll
+    for (int q = 0; q < GPB; ++q) {
+      const int grp = gq0 + q;
+      if (grp >= ngroups) break;
+      const uchar4 b4 = binned[(size_t)grp * n_rows + r];
+      const unsigned char bs[4] = {b4.x, b4.y, b4.z, b4.w};
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float* cell = &lds[((q * 4 + j) * n_bins + bs[j]) * 3];
+        atomicAdd(cell + 0, g);
+        atomicAdd(cell + 1, h);
+        atomicAdd(cell + 2, 1.0f);
+      }
+    }
+  }
+  __syncthreads();
+
+  // flush LDS histograms to global with device atomics (row chunks overlap)
+  const size_t base = (size_t)gq0 * 4 * n_bins * 3;
+  const int valid_f = min(nfb, (ngroups - gq0) * 4);
+  const int flush_elems = valid_f * n_bins * 3;
+  for (int i = tid; i < flush_elems; i += blockDim.x) {
+    const float v = lds[i];
+    if (v != 0.0f) atomicAdd(&hist[base + i], v);
+  }
+}
+
+extern "C" void launch_hist_build(const void* binned, long n_rows,
+                                  const int* rows, long m, const float* grad,
+                                  const float* hess, float* hist, int n_bins,
+                                  int ngroups, hipStream_t stream) {
+  constexpr int GPB = 4;
+  if (m == 0) return;
+  const int n_fblocks = (ngroups + GPB - 1) / GPB;
+  // target ≥ 2048 workgroups total to fill 256 CUs × 8 XCDs
+  long chunks = (2048 + n_fblocks - 1) / n_fblocks;
+  long min_chunk = 1024;  // enough rows per block to amortize the LDS flush
+  long chunk = (m + chunks - 1) / chunks;
+  if (chunk < min_chunk) chunk = min_chunk;
+  chunks = (m + chunk - 1) / chunk;
+  dim3 grid((unsigned)chunks, (unsigned)n_fblocks);
+  const size_t lds_bytes = (size_t)GPB * 4 * n_bins * 3 * sizeof(float);
+  hipLaunchKernelGGL((hist_build_k<GPB>), grid, dim3(256), lds_bytes, stream,
+                     (const uchar4*)binned, n_rows, rows, m, grad, hess, hist,
+                     n_bins, ngroups, chunk);
+}
+
+// ------------------------------------------------------------ forest predict
+// Flat node arrays across trees; per-thread row traversal.
+__global__ void predict_forest_k(const int* __restrict__ feat,
+                                 const float* __restrict__ thr,
+                                 const int* __restrict__ left,
+                                 const int* __restrict__ right,
+                                 const float* __restrict__ val,
+                                 const long* __restrict__ offsets,
+                                 const float* __restrict__ tw,
+                                 const float* __restrict__ X, long n, int nf,
+                                 float* __restrict__ out, int n_outputs,
+                                 int t0, int t1) {
+  const long row0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long row = row0; row < n; row += stride) {
+    const float* x = X + row * nf;
+    if (n_outputs == 1) {
+      float acc = 0.0f;
+      for (int t = t0; t < t1; ++t) {
+        long idx = offsets[t];
+        const long base = idx;
+        int f = feat[idx];
+        while (f >= 0) {
+          const float xv = x[f];
+          idx = base + ((xv <= thr[idx] || isnan(xv)) ? left[idx] : right[idx]);
+          f = feat[idx];
+        }
+        acc += tw[t] * val[idx];
+      }
+      out[row] += acc;
+    } else {
+      for (int t = t0; t < t1; ++t) {
+        long idx = offsets[t];
+        const long base = idx;
+        int f = feat[idx];
+        while (f >= 0) {
+          const float xv = x[f];
+          idx = base + ((xv <= thr[idx] || isnan(xv)) ? left[idx] : right[idx]);
+          f = feat[idx];
+        }
+        out[row * n_outputs + (t % n_outputs)] += tw[t] * val[idx];
+      }
+    }
+  }
+}
+
+extern "C" void launch_predict_forest(const int* feat, const float* thr,
+                                      const int* left, const int* right,
+                                      const float* val, const long* offsets,
+                                      const float* tw, const float* X, long n,
+                                      int nf, float* out, int n_outputs,
+                                      int t0, int t1, hipStream_t stream) {
+  if (n == 0 || t1 <= t0) return;
+  long blocks = (n + 255) / 256;
+  if (blocks > 8192) blocks = 8192;
+  hipLaunchKernelGGL(predict_forest_k, dim3((unsigned)blocks), dim3(256), 0,
+                     stream, feat, thr, left, right, val, offsets, tw, X, n,
+                     nf, out, n_outputs, t0, t1);
+}
+
+// ------------------------------------------------------------- leaf indices
+__global__ void predict_leaf_k(const int* __restrict__ feat,
+                               const float* __restrict__ thr,
+                               const int* __restrict__ left,
+                               const int* __restrict__ right,
+                               const int* __restrict__ leaf_index,
+                               const long* __restrict__ offsets,
+                               const float* __restrict__ X, long n, int nf,
+                               int* __restrict__ out, int n_trees) {
+  const long row0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long row = row0; row < n; row += stride) {
+    const float* x = X + row * nf;
+    for (int t = 0; t < n_trees; ++t) {
+      long idx = offsets[t];
+      const long base = idx;
+      int f = feat[idx];
+      while (f >= 0) {
+        const float xv = x[f];
+        idx = base + ((xv <= thr[idx] || isnan(xv)) ? left[idx] : right[idx]);
+        f = feat[idx];
+      }
+      out[row * n_trees + t] = leaf_index[idx];
+    }
+  }
+}
+
+extern "C" void launch_predict_leaf(const int* feat, const float* thr,
+                                    const int* left, const int* right,
+                                    const int* leaf_index, const long* offsets,
+                                    const float* X, long n, int nf, int* out,
+                                    int n_trees, hipStream_t stream) {
+  if (n == 0) return;
+  long blocks = (n + 255) / 256;
+  if (blocks > 8192) blocks = 8192;
+  hipLaunchKernelGGL(predict_leaf_k, dim3((unsigned)blocks), dim3(256), 0,
+                     stream, feat, thr, left, right, leaf_index, offsets, X, n,
+                     nf, out, n_trees);
+}
+
+// ----------------------------------------------------------------- binning
+// X: (n, nf) f32 row-major; ub: (nf, n_bins-1) ascending; out: (ngroups, n) uchar4
+__global__ void bin_matrix_k(const float* __restrict__ X,
+                             const float* __restrict__ ub, long n, int nf,
+                             int n_bins, int ngroups, uchar4* __restrict__ out) {
+  const long r0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  const int nb = n_bins - 1;
+  for (long r = r0; r < n; r += stride) {
+    for (int g = 0; g < ngroups; ++g) {
+      unsigned char b[4] = {0, 0, 0, 0};
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int f = g * 4 + j;
+        if (f >= nf) break;
+        const float x = X[r * nf + f];
+        if (isnan(x)) { b[j] = 0; continue; }
+        const float* bounds = ub + (size_t)f * nb;
+        // lower_bound: first idx with bounds[idx] >= x
+        int lo = 0, hi = nb;
+        while (lo < hi) {
+          const int mid = (lo + hi) >> 1;
+          if (bounds[mid] < x) lo = mid + 1; else hi = mid;
+        }
+        b[j] = (unsigned char)min(lo, n_bins - 1);
+      }
+      out[(size_t)g * n + r] = make_uchar4(b[0], b[1], b[2], b[3]);
+    }
+  }
+}
+
+extern "C" void launch_bin_matrix(const float* X, const float* ub, long n,
+                                  int nf, int n_bins, int ngroups, void* out,
+                                  hipStream_t stream) {
+  if (n == 0) return;
+  long blocks = (n + 255) / 256;
+  if (blocks > 8192) blocks = 8192;
+  hipLaunchKernelGGL(bin_matrix_k, dim3((unsigned)blocks), dim3(256), 0,
+                     stream, X, ub, n, nf, n_bins, ngroups, (uchar4*)out);
+}

@@ -1,0 +1,106 @@
+// CDNA4 kernels for the VW-equivalent path: hashed sparse SGD over a 2^b
+// weight table in HBM.  Replaces the compute inside the reference's
+// VowpalWabbitNative example.learn loop (VowpalWabbitBase.scala:261-292):
+// per-example sparse dot product + adaptive (AdaGrad-style) per-weight update.
+// One 64-lane wave per example; lanes stride the example's features; the dot
+// product reduces with __shfl_xor over the full wave.  Minibatch updates are
+// hogwild (atomics) — end-of-pass weight sync is an RCCL all_reduce
+// (spanning-tree AllReduce → RCCL over xGMI).
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#define WAVE 64
+
+__device__ __forceinline__ float wave_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, WAVE);
+  return v;
+}
+
+// loss: 0=squared 1=logistic(label ±1) 2=hinge(label ±1)
+__device__ __forceinline__ float dloss(int loss, float pred, float y) {
+  switch (loss) {
+    case 0: return pred - y;
+    case 1: { // d/dp log(1+exp(-y p)) = -y * sigmoid(-y p)
+      const float z = -y * pred;
+      const float s = 1.0f / (1.0f + __expf(-z));
+      return -y * s;
+    }
+    case 2: return (y * pred < 1.0f) ? -y : 0.0f;
+  }
+  return 0.0f;
+}
+
+__global__ void vw_sgd_k(const int* __restrict__ idx,
+                         const float* __restrict__ val,
+                         const long* __restrict__ off,
+                         const float* __restrict__ label,
+                         float* __restrict__ w_tbl, float* __restrict__ g_tbl,
+                         float lr, float l2, float power_t, int loss,
+                         long n_ex, float* __restrict__ preds_out) {
+  const long wid0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const long n_waves = ((long)gridDim.x * blockDim.x) / WAVE;
+  for (long ex = wid0; ex < n_ex; ex += n_waves) {
+    const long s = off[ex], e = off[ex + 1];
+    float dot = 0.0f;
+    for (long k = s + lane; k < e; k += WAVE) dot += w_tbl[idx[k]] * val[k];
+    const float pred = wave_sum(dot);
+    if (preds_out && lane == 0) preds_out[ex] = pred;
+    const float gl = dloss(loss, pred, label[ex]);
+    if (gl == 0.0f) continue;
+    for (long k = s + lane; k < e; k += WAVE) {
+      const int i = idx[k];
+      const float x = val[k];
+      float g = gl * x + l2 * w_tbl[i];
+      const float Gold = atomicAdd(&g_tbl[i], g * g);
+      const float G = Gold + g * g;
+      // adaptive per-weight rate: lr * G^(-power_t); power_t=0.5 → rsqrt
+      float scale;
+      if (power_t == 0.5f) scale = __frsqrt_rn(G + 1e-10f);
+      else scale = __powf(G + 1e-10f, -power_t);
+      atomicAdd(&w_tbl[i], -lr * g * scale);
+    }
+  }
+}
+
+extern "C" void launch_vw_sgd(const int* idx, const float* val,
+                              const long* off, const float* label,
+                              float* w_tbl, float* g_tbl, float lr, float l2,
+                              float power_t, int loss, long n_ex,
+                              float* preds_out, hipStream_t stream) {
+  if (n_ex == 0) return;
+  long waves = n_ex;
+  long blocks = (waves * WAVE + 255) / 256;
+  if (blocks > 4096) blocks = 4096;
+  hipLaunchKernelGGL(vw_sgd_k, dim3((unsigned)blocks), dim3(256), 0, stream,
+                     idx, val, off, label, w_tbl, g_tbl, lr, l2, power_t,
+                     loss, n_ex, preds_out);
+}
+
+__global__ void vw_predict_k(const int* __restrict__ idx,
+                             const float* __restrict__ val,
+                             const long* __restrict__ off,
+                             const float* __restrict__ w_tbl, long n_ex,
+                             float* __restrict__ out) {
+  const long wid0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const long n_waves = ((long)gridDim.x * blockDim.x) / WAVE;
+  for (long ex = wid0; ex < n_ex; ex += n_waves) {
+    const long s = off[ex], e = off[ex + 1];
+    float dot = 0.0f;
+    for (long k = s + lane; k < e; k += WAVE) dot += w_tbl[idx[k]] * val[k];
+    const float pred = wave_sum(dot);
+    if (lane == 0) out[ex] = pred;
+  }
+}
+
+extern "C" void launch_vw_predict(const int* idx, const float* val,
+                                  const long* off, const float* w_tbl,
+                                  long n_ex, float* out, hipStream_t stream) {
+  if (n_ex == 0) return;
+  long blocks = (n_ex * WAVE + 255) / 256;
+  if (blocks > 4096) blocks = 4096;
+  hipLaunchKernelGGL(vw_predict_k, dim3((unsigned)blocks), dim3(256), 0,
+                     stream, idx, val, off, w_tbl, n_ex, out);
+}

@@ -1,0 +1,57 @@
+"""In-tree build of the MI355X HIP extension.
+
+`python setup.py build_ext --inplace` produces
+mmlspark_amd/ops/_hip_ops*.so next to the package sources so the .so travels
+with the repo snapshot to GPU boxes.  Kernels (*.hip) are compiled directly
+with hipcc --offload-arch=gfx950 (no hipify, no CUDA path); the torch
+binding (ext.cpp) is built as a normal torch CppExtension and linked against
+the kernel objects + libamdhip64.
+"""
+import os
+import subprocess
+import sys
+
+from setuptools import setup
+from torch.utils import cpp_extension
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+HIP_DIR = os.path.join(ROOT, "mmlspark_amd", "ops", "hip")
+ROCM = os.environ.get("ROCM_PATH", "/opt/rocm")
+ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950").split(";")[0]
+
+KERNEL_SOURCES = ["gbdt_kernels.hip", "vw_kernels.hip"]
+
+
+def compile_hip_kernels():
+    objs = []
+    for src in KERNEL_SOURCES:
+        src_path = os.path.join(HIP_DIR, src)
+        obj_path = src_path.replace(".hip", ".o")
+        if (not os.path.exists(obj_path)
+                or os.path.getmtime(obj_path) < os.path.getmtime(src_path)):
+            cmd = [os.path.join(ROCM, "bin", "hipcc"),
+                   f"--offload-arch={ARCH}", "-O3", "-std=c++17", "-fPIC",
+                   "-c", src_path, "-o", obj_path]
+            print("[hipcc]", " ".join(cmd), flush=True)
+            subprocess.check_call(cmd)
+        objs.append(obj_path)
+    return objs
+
+
+extra_objects = compile_hip_kernels()
+
+setup(
+    name="mmlspark_amd_hip_ops",
+    ext_modules=[
+        cpp_extension.CppExtension(
+            name="mmlspark_amd.ops._hip_ops",
+            sources=[os.path.join(HIP_DIR, "ext.cpp")],
+            extra_objects=extra_objects,
+            extra_compile_args=["-O2", "-D__HIP_PLATFORM_AMD__=1",
+                                f"-I{ROCM}/include"],
+            extra_link_args=[f"-L{ROCM}/lib", "-lamdhip64"],
+        )
+    ],
+    cmdclass={"build_ext": cpp_extension.BuildExtension.with_options(
+        use_ninja=False)},
+)
