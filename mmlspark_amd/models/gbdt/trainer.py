@@ -147,49 +147,24 @@ class TreeGrower:
             self.stats.comm_s += time.perf_counter() - t1
         return h
 
-    def _best_split(self, hist, feat_mask):
-        """Identical on all ranks (input is the reduced histogram)."""
+    def _scan(self, hists, feat_mask):
+        """Fused split scan on stacked histograms; identical on all ranks
+        (input is the reduced histogram). One kernel pair + one readback."""
         t0 = time.perf_counter()
         cfg = self.cfg
-        g = hist[:, :, 0]
-        h = hist[:, :, 1]
-        c = hist[:, :, 2]
-        GL = g.cumsum(1)
-        HL = h.cumsum(1)
-        CL = c.cumsum(1)
-        G = GL[:, -1:]
-        H = HL[:, -1:]
-        C = CL[:, -1:]
-        GR, HR, CR = G - GL, H - HL, C - CL
-
-        if cfg.lambda_l1 > 0:
-            def sc(Gs, Hs):
-                Ga = (Gs.abs() - cfg.lambda_l1).clamp_min(0)
-                return Ga * Ga / (Hs + cfg.lambda_l2 + 1e-32)
-        else:
-            def sc(Gs, Hs):
-                return Gs * Gs / (Hs + cfg.lambda_l2 + 1e-32)
-
-        gain = sc(GL, HL) + sc(GR, HR) - sc(G, H)
-        valid = ((CL >= cfg.min_data_in_leaf) & (CR >= cfg.min_data_in_leaf)
-                 & (HL >= cfg.min_sum_hessian_in_leaf)
-                 & (HR >= cfg.min_sum_hessian_in_leaf))
-        gain = torch.where(valid, gain, torch.full_like(gain, NEG_INF))
-        gain[:, -1] = NEG_INF  # right side empty
-        if feat_mask is not None:
-            gain[~feat_mask] = NEG_INF
-        gain[self.nf:] = NEG_INF  # padding features
-        per_f, per_bin = gain.max(dim=1)
-        bf_t = per_f.argmax()
-        bb_t = per_bin[bf_t]
-        # single device→host sync for all six scalars
-        packed = torch.stack([per_f[bf_t], bf_t.float(), bb_t.float(),
-                              GL[bf_t, bb_t], HL[bf_t, bb_t],
-                              CL[bf_t, bb_t]]).cpu()
-        bg, bf, bb, gl, hl, cl = packed.tolist()
-        out = (bg, int(bf), int(bb), gl, hl, cl)
+        out = backend.split_scan(
+            hists, cfg.max_bin, cfg.lambda_l1, cfg.lambda_l2,
+            float(cfg.min_data_in_leaf), cfg.min_sum_hessian_in_leaf,
+            cfg.min_gain_to_split, self.nf, feat_mask).cpu()
+        res = []
+        for row in out.tolist():
+            bg, bf, bb, gl, hl, cl = row
+            res.append((bg, int(bf), int(bb), gl, hl, cl))
         self.stats.split_s += time.perf_counter() - t0
-        return out
+        return res
+
+    def _best_split(self, hist, feat_mask):
+        return self._scan(hist.unsqueeze(0), feat_mask)[0]
 
     def grow(self, rows_root: torch.Tensor, grad: torch.Tensor,
              hess: torch.Tensor, feat_mask) -> (Tree, List):
@@ -262,9 +237,10 @@ class TreeGrower:
             lc = _Leaf(lid, rows_l, hist_l, GL, HL, CL, leaf.depth + 1)
             rc = _Leaf(rid, rows_r, hist_r, GR, HR, CR, leaf.depth + 1)
             leaf.hist = None  # free parent histogram
-            for ch in (lc, rc):
-                (ch.gain, ch.feat, ch.bin, ch.GL, ch.HL, ch.CL) = \
-                    self._best_split(ch.hist, feat_mask)
+            # both children in ONE fused scan + readback
+            pair = self._scan(torch.stack([hist_l, hist_r]), feat_mask)
+            for ch, res in zip((lc, rc), pair):
+                (ch.gain, ch.feat, ch.bin, ch.GL, ch.HL, ch.CL) = res
                 heapq.heappush(heap, (-ch.gain, seq, ch))
                 seq += 1
                 final_leaves.append(ch)

@@ -84,6 +84,16 @@ def predict_leaf(node_feature, node_threshold, node_left, node_right,
                                 node_right, node_leaf_index, tree_offsets, X)
 
 
+def split_scan(hists, n_bins, l1, l2, min_data, min_hess, min_gain, nf_real,
+               feat_mask=None):
+    if hists.is_cuda:
+        return _require_ext().split_scan(hists.contiguous(), n_bins, l1, l2,
+                                         min_data, min_hess, min_gain,
+                                         nf_real, feat_mask)
+    return cpu_ref.split_scan(hists, n_bins, l1, l2, min_data, min_hess,
+                              min_gain, nf_real, feat_mask)
+
+
 def bin_matrix(X, upper_bounds, n_bins):
     if X.is_cuda:
         return _require_ext().bin_matrix(X, upper_bounds, n_bins)

@@ -115,3 +115,43 @@ def bin_matrix(X: torch.Tensor, upper_bounds: torch.Tensor,
     for f in range(nf):
         out[f // 4, :, f % 4] = bins[f]
     return out
+
+
+def split_scan(hists: torch.Tensor, n_bins: int, l1: float, l2: float,
+               min_data: float, min_hess: float, min_gain: float,
+               nf_real: int, feat_mask=None) -> torch.Tensor:
+    """Best split per histogram. hists (n_hists, nf_pad, n_bins, 3) ->
+    (n_hists, 6) {gain, feature, bin, GL, HL, CL}. Matches split_scan_k."""
+    g = hists[..., 0]
+    h = hists[..., 1]
+    c = hists[..., 2]
+    GL = g.cumsum(-1)
+    HL = h.cumsum(-1)
+    CL = c.cumsum(-1)
+    G = GL[..., -1:]
+    H = HL[..., -1:]
+    C = CL[..., -1:]
+    GR, HR, CR = G - GL, H - HL, C - CL
+
+    def sc(Gs, Hs):
+        Ga = (Gs.abs() - l1).clamp_min(0)
+        return Ga * Ga / (Hs + l2 + 1e-32)
+
+    gain = sc(GL, HL) + sc(GR, HR) - sc(G, H)
+    valid = ((CL >= min_data) & (CR >= min_data)
+             & (HL >= min_hess) & (HR >= min_hess))
+    neg = torch.full_like(gain, float("-inf"))
+    gain = torch.where(valid, gain, neg)
+    gain[..., -1] = float("-inf")
+    if feat_mask is not None:
+        gain[:, ~feat_mask, :] = float("-inf")
+    gain[:, nf_real:, :] = float("-inf")
+    per_f, per_bin = gain.max(dim=-1)           # (nh, nf)
+    bf = per_f.argmax(dim=-1)                   # (nh,)
+    nh = hists.shape[0]
+    ar = torch.arange(nh, device=hists.device)
+    bb = per_bin[ar, bf]
+    out = torch.stack([
+        per_f[ar, bf], bf.float(), bb.float(),
+        GL[ar, bf, bb], HL[ar, bf, bb], CL[ar, bf, bb]], dim=-1)
+    return out

@@ -20,6 +20,9 @@ void launch_predict_leaf(const int*, const float*, const int*, const int*,
                          int*, int, hipStream_t);
 void launch_bin_matrix(const float*, const float*, long, int, int, int, void*,
                        hipStream_t);
+void launch_split_scan(const float*, int, long, int, float, float, float,
+                       float, float, long, const bool*, float*, float*,
+                       hipStream_t);
 void launch_vw_sgd(const int*, const float*, const long*, const float*,
                    float*, float*, float, float, float, int, long, float*,
                    hipStream_t);
@@ -98,6 +101,30 @@ torch::Tensor bin_matrix(torch::Tensor X, torch::Tensor ub, long n_bins) {
   return out;
 }
 
+torch::Tensor split_scan(torch::Tensor hists, long n_bins, double l1,
+                         double l2, double min_data, double min_hess,
+                         double min_gain, long nf_real,
+                         c10::optional<torch::Tensor> feat_mask) {
+  // hists: (n_hists, nf_pad, n_bins, 3) — returns (n_hists, 6) on device:
+  // {gain, feature, bin, GL, HL, CL}
+  CHECK_DEV(hists); CHECK_CONTIG(hists);
+  const long n_hists = hists.size(0);
+  const long nf_pad = hists.size(1);
+  auto scratch = torch::empty({n_hists, nf_pad, 6}, hists.options());
+  auto out = torch::empty({n_hists, 6}, hists.options());
+  const bool* mask_ptr = nullptr;
+  if (feat_mask.has_value()) {
+    TORCH_CHECK(feat_mask->dtype() == torch::kBool, "feat_mask must be bool");
+    mask_ptr = feat_mask->data_ptr<bool>();
+  }
+  launch_split_scan(hists.data_ptr<float>(), (int)n_hists, nf_pad,
+                    (int)n_bins, (float)l1, (float)l2, (float)min_data,
+                    (float)min_hess, (float)min_gain, nf_real, mask_ptr,
+                    scratch.data_ptr<float>(), out.data_ptr<float>(),
+                    cur_stream());
+  return out;
+}
+
 torch::Tensor vw_sgd_minibatch(torch::Tensor idx, torch::Tensor val,
                                torch::Tensor off, torch::Tensor label,
                                torch::Tensor w_tbl, torch::Tensor g_tbl,
@@ -130,6 +157,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("predict_forest", &predict_forest, "GBDT ensemble raw scores");
   m.def("predict_leaf", &predict_leaf, "GBDT per-tree leaf indices");
   m.def("bin_matrix", &bin_matrix, "quantile binning to interleaved uint8");
+  m.def("split_scan", &split_scan, "fused best-split over sibling histograms");
   m.def("vw_sgd_minibatch", &vw_sgd_minibatch, "adaptive sparse SGD minibatch");
   m.def("vw_predict", &vw_predict, "sparse linear predict");
 }

@@ -121,3 +121,24 @@ def test_vw_kernels():
     preds = backend.vw_predict(idx_d, val_d, off_d, w).cpu()
     acc = ((preds.sign() == labels).float().mean())
     assert float(acc) > 0.8, float(acc)
+
+
+@requires_gpu
+def test_split_scan_matches_cpu():
+    from mmlspark_amd.ops import backend, cpu_ref
+    g = torch.Generator().manual_seed(3)
+    nh, nf_pad, nb = 2, 28, 255
+    hists = torch.zeros(nh, nf_pad, nb, 3)
+    hists[:, :, :, 0] = torch.randn(nh, nf_pad, nb, generator=g)
+    hists[:, :, :, 1] = torch.rand(nh, nf_pad, nb, generator=g) + 0.01
+    hists[:, :, :, 2] = torch.randint(0, 50, (nh, nf_pad, nb), generator=g).float()
+    mask = torch.ones(nf_pad, dtype=torch.bool)
+    mask[5] = False
+    for fm in (None, mask):
+        ref = cpu_ref.split_scan(hists, nb, 0.1, 0.5, 3.0, 1e-3, 0.0, 26, fm)
+        out = backend.split_scan(hists.cuda(), nb, 0.1, 0.5, 3.0, 1e-3, 0.0,
+                                 26, fm.cuda() if fm is not None else None).cpu()
+        # gains equal within fp tolerance; chosen split identical
+        assert torch.allclose(ref[:, 0], out[:, 0], rtol=1e-3, atol=1e-3), (ref, out)
+        assert torch.equal(ref[:, 1:3], out[:, 1:3]), (ref, out)
+        assert torch.allclose(ref[:, 3:], out[:, 3:], rtol=1e-3, atol=1e-2)
