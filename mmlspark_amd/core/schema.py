@@ -15,6 +15,34 @@ import numpy as np
 import pandas as pd
 
 
+class SparseVector:
+    """Hashed sparse feature vector (the analog of SparkML SparseVector
+    produced by the reference's VowpalWabbitFeaturizer)."""
+
+    __slots__ = ("size", "indices", "values")
+
+    def __init__(self, size: int, indices, values):
+        self.size = int(size)
+        self.indices = np.asarray(indices, dtype=np.int32)
+        self.values = np.asarray(values, dtype=np.float32)
+
+    def to_dense(self) -> np.ndarray:
+        out = np.zeros(self.size, dtype=np.float32)
+        out[self.indices] = self.values
+        return out
+
+    def __len__(self):
+        return self.size
+
+    def __eq__(self, other):
+        return (isinstance(other, SparseVector) and self.size == other.size
+                and np.array_equal(self.indices, other.indices)
+                and np.array_equal(self.values, other.values))
+
+    def __repr__(self):
+        return f"SparseVector(size={self.size}, nnz={len(self.indices)})"
+
+
 def find_unused_column(df: pd.DataFrame, base: str) -> str:
     name = base
     i = 0

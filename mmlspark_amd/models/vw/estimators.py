@@ -1,0 +1,406 @@
+"""VW-equivalent estimators: hashed sparse online learning on MI355X.
+
+Parity targets: VowpalWabbitClassifier:21 / VowpalWabbitRegressor
+/ VowpalWabbitContextualBandit:106 (vw/src/main/scala/.../VowpalWabbitClassifier.scala
+etc.).  The native VW example.learn loop (VowpalWabbitBase.trainRow:261-292)
+becomes minibatched adaptive sparse-SGD HIP kernels over a 2^b weight table in
+HBM; the end-of-pass spanning-tree AllReduce (VowpalWabbitBase.scala:363-368)
+becomes ONE RCCL all_reduce of the dense weight table over xGMI (2^18 floats =
+1 MB — latency-bound, single launch).  Per-partition perf timers surface as a
+performance-statistics DataFrame (TrainingStats parity,
+VowpalWabbitBase.scala:27-46,464-490).
+"""
+from __future__ import annotations
+
+import time
+from typing import Optional
+
+import numpy as np
+import pandas as pd
+import torch
+
+from ...core.param import Param, Params, toBool, toFloat, toInt, toString
+from ...core.pipeline import Estimator, Model
+from ...core.registry import register
+from ...core.schema import SparseVector, matrix_to_vector_column
+from ...ops import backend
+from ...parallel.comm import get_comm
+from ...utils.devices import default_device
+
+_LOSS_BY_NAME = {"squared": "squared", "logistic": "logistic", "hinge": "hinge"}
+
+
+def _extract_csr(df: pd.DataFrame, col: str, extra_cols=None, device="cpu"):
+    """SparseVector / dense-vector column(s) -> CSR tensors on device."""
+    cols = [col] + list(extra_cols or [])
+    idx_parts, val_parts, counts = [], [], np.zeros(len(df), dtype=np.int64)
+    offset_base = 0
+    sizes = []
+    for c in cols:
+        vals = df[c].to_numpy()
+        if len(vals) and isinstance(vals[0], SparseVector):
+            sizes.append(vals[0].size)
+        else:
+            sizes.append(len(np.asarray(vals[0])) if len(vals) else 0)
+    for ci, c in enumerate(cols):
+        vals = df[c].to_numpy()
+        for i, v in enumerate(vals):
+            if isinstance(v, SparseVector):
+                idx_parts.append((i, v.indices + offset_base, v.values))
+                counts[i] += len(v.indices)
+            else:
+                dense = np.asarray(v, dtype=np.float32)
+                nz = np.nonzero(dense)[0]
+                idx_parts.append((i, nz.astype(np.int32) + offset_base,
+                                  dense[nz]))
+                counts[i] += len(nz)
+        offset_base += sizes[ci]
+    # assemble in row order
+    idx_parts.sort(key=lambda t: t[0])
+    if idx_parts:
+        indices = np.concatenate([p[1] for p in idx_parts])
+        values = np.concatenate([p[2] for p in idx_parts])
+    else:
+        indices = np.zeros(0, dtype=np.int32)
+        values = np.zeros(0, dtype=np.float32)
+    offsets = np.zeros(len(df) + 1, dtype=np.int64)
+    np.cumsum(counts, out=offsets[1:])
+    return (torch.from_numpy(indices.astype(np.int32)).to(device),
+            torch.from_numpy(values.astype(np.float32)).to(device),
+            torch.from_numpy(offsets).to(device),
+            offset_base)
+
+
+class _VWParams(Params):
+    labelCol = Param("labelCol", "label column", "label")
+    featuresCol = Param("featuresCol", "hashed features column", "features")
+    additionalFeatures = Param("additionalFeatures",
+                               "extra feature columns (namespaces)", None)
+    weightCol = Param("weightCol", "importance weight column", None)
+    predictionCol = Param("predictionCol", "prediction column", "prediction")
+    numPasses = Param("numPasses", "passes over the data", 1, toInt)
+    learningRate = Param("learningRate", "initial learning rate", 0.5, toFloat)
+    powerT = Param("powerT", "lr decay exponent", 0.5, toFloat)
+    l1 = Param("l1", "L1 regularization (end-of-pass truncation)", 0.0, toFloat)
+    l2 = Param("l2", "L2 regularization", 0.0, toFloat)
+    numBits = Param("numBits", "log2 weight-table size", 18, toInt)
+    lossFunction = Param("lossFunction", "squared|logistic|hinge", None)
+    batchSize = Param("batchSize", "SGD minibatch size (GPU hogwild window)",
+                      4096, toInt)
+    hashSeed = Param("hashSeed", "murmur seed", 0, toInt)
+    initialModel = Param("initialModel", "warm-start weight table", None,
+                         is_complex=True)
+    passThroughArgs = Param("passThroughArgs",
+                            "VW-style arg string (subset parsed: --l1 --l2 "
+                            "--learning_rate --power_t -b --passes "
+                            "--loss_function)", "", toString)
+    device = Param("device", "cpu|cuda|auto", "auto", toString)
+
+    def _parse_args(self):
+        """Apply passThroughArgs (analog of the reference building the VW arg
+        string, VowpalWabbitBase.scala:531-543 — here parsed back to Params)."""
+        s = (self.get("passThroughArgs") or "").split()
+        i = 0
+        mapping = {"--l1": "l1", "--l2": "l2", "--learning_rate": "learningRate",
+                   "--power_t": "powerT", "-b": "numBits", "--bit_precision":
+                   "numBits", "--passes": "numPasses",
+                   "--loss_function": "lossFunction"}
+        while i < len(s):
+            if s[i] in mapping and i + 1 < len(s):
+                self.set(mapping[s[i]], s[i + 1])
+                i += 2
+            else:
+                i += 1
+
+
+class _VWBase(_VWParams, Estimator):
+    _default_loss = "squared"
+    _binary_labels = False
+
+    def _fit(self, df: pd.DataFrame):
+        self._parse_args()
+        comm = get_comm()
+        device = default_device(self.get("device"))
+        t_ingest = time.perf_counter()
+        idx, val, off, _ = _extract_csr(df, self.get("featuresCol"),
+                                        self.get("additionalFeatures"), device)
+        y = df[self.get("labelCol")].to_numpy(dtype=np.float32)
+        if self._binary_labels:
+            y = np.where(y > 0, 1.0, -1.0).astype(np.float32)
+        labels = torch.from_numpy(y).to(device)
+        ex_w = None
+        if self.get("weightCol"):
+            ex_w = torch.from_numpy(
+                df[self.get("weightCol")].to_numpy(dtype=np.float32)).to(device)
+        ingest_s = time.perf_counter() - t_ingest
+
+        bits = self.get("numBits")
+        tbl = 1 << bits
+        init = self.get("initialModel")
+        if init is not None:
+            w = torch.from_numpy(np.asarray(init["weights"],
+                                            dtype=np.float32)).to(device).clone()
+            g = torch.from_numpy(np.asarray(init["adaptive"],
+                                            dtype=np.float32)).to(device).clone()
+        else:
+            w = torch.zeros(tbl, dtype=torch.float32, device=device)
+            g = torch.zeros(tbl, dtype=torch.float32, device=device)
+
+        loss = self.get("lossFunction") or self._default_loss
+        lr = self.get("learningRate")
+        l1, l2 = self.get("l1"), self.get("l2")
+        bs = self.get("batchSize")
+        n = len(df)
+        learn_s = 0.0
+        multipass_s = 0.0
+        for p in range(self.get("numPasses")):
+            t0 = time.perf_counter()
+            for s in range(0, n, bs):
+                e = min(s + bs, n)
+                o = off[s:e + 1] - off[s]
+                sl = slice(int(off[s]), int(off[e]))
+                backend.vw_sgd_minibatch(
+                    idx[sl], val[sl], o, labels[s:e], w, g, lr, l2,
+                    self.get("powerT"), loss,
+                    ex_w[s:e] if ex_w is not None else None)
+            learn_s += time.perf_counter() - t0
+            # end-of-pass sync: RCCL all_reduce of weights + accumulators
+            t0 = time.perf_counter()
+            if comm.is_distributed:
+                comm.all_reduce(w)
+                w /= comm.world_size
+                comm.all_reduce(g)
+                g /= comm.world_size
+            if l1 > 0:  # proximal truncation
+                w.copy_(torch.sign(w) * (w.abs() - lr * l1).clamp_min(0))
+            multipass_s += time.perf_counter() - t0
+
+        model = self._model_class()(weights=w.cpu().numpy(),
+                                    adaptive=g.cpu().numpy())
+        for pname in ("labelCol", "featuresCol", "additionalFeatures",
+                      "predictionCol", "numBits"):
+            model.set(pname, self.get(pname))
+        model.set("lossFunction", loss)
+        model._stats = pd.DataFrame([{
+            "partitionId": comm.rank, "ipassCurrent": self.get("numPasses"),
+            "numberOfExamplesPerPass": n,
+            "totalNumberOfFeatures": int(off[-1]),
+            "nativeIngestTimeNs": int(ingest_s * 1e9),
+            "learnTimeNs": int(learn_s * 1e9),
+            "multipassTimeNs": int(multipass_s * 1e9),
+        }])
+        return model
+
+    def _model_class(self):
+        raise NotImplementedError
+
+
+class _VWModelBase(_VWParams, Model):
+    weightsArrays = Param("weightsArrays", "weight + adaptive tables", None,
+                          is_complex=True)
+
+    def __init__(self, weights: Optional[np.ndarray] = None,
+                 adaptive: Optional[np.ndarray] = None, **kwargs):
+        super().__init__(**kwargs)
+        if weights is not None:
+            self.set("weightsArrays", {"weights": weights, "adaptive": adaptive})
+        self._stats = None
+
+    @property
+    def weights(self) -> np.ndarray:
+        return self.get("weightsArrays")["weights"]
+
+    def getPerformanceStatistics(self) -> pd.DataFrame:
+        """Per-rank perf stats DF (VowpalWabbitBase.scala:464-490)."""
+        return self._stats if self._stats is not None else pd.DataFrame()
+
+    def saveNativeModel(self, path: str):
+        np.savez(path, **self.get("weightsArrays"))
+
+    def getReadableModel(self) -> pd.DataFrame:
+        w = self.weights
+        nz = np.nonzero(w)[0]
+        return pd.DataFrame({"index": nz, "weight": w[nz]})
+
+    def _raw(self, df: pd.DataFrame) -> np.ndarray:
+        device = default_device(self.get("device"))
+        idx, val, off, _ = _extract_csr(df, self.get("featuresCol"),
+                                        self.get("additionalFeatures"), device)
+        w = torch.from_numpy(self.weights).to(device)
+        return backend.vw_predict(idx, val, off, w).cpu().numpy()
+
+
+@register
+class VowpalWabbitRegressor(_VWBase):
+    _default_loss = "squared"
+
+    def _model_class(self):
+        return VowpalWabbitRegressorModel
+
+
+@register
+class VowpalWabbitRegressorModel(_VWModelBase):
+    def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
+        out = df.copy()
+        out[self.get("predictionCol")] = self._raw(df).astype(np.float64)
+        return out
+
+
+@register
+class VowpalWabbitClassifier(_VWBase):
+    _default_loss = "logistic"
+    _binary_labels = True
+    rawPredictionCol = Param("rawPredictionCol", "margin column", "rawPrediction")
+    probabilityCol = Param("probabilityCol", "probability column", "probability")
+
+    def _fit(self, df):
+        model = super()._fit(df)
+        for p in ("rawPredictionCol", "probabilityCol"):
+            model.set(p, self.get(p))
+        return model
+
+    def _model_class(self):
+        return VowpalWabbitClassificationModel
+
+
+@register
+class VowpalWabbitClassificationModel(_VWModelBase):
+    rawPredictionCol = Param("rawPredictionCol", "margin column", "rawPrediction")
+    probabilityCol = Param("probabilityCol", "probability column", "probability")
+
+    def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
+        raw = self._raw(df)
+        p1 = 1.0 / (1.0 + np.exp(-raw))
+        out = df.copy()
+        out[self.get("rawPredictionCol")] = matrix_to_vector_column(
+            np.stack([-raw, raw], axis=1))
+        out[self.get("probabilityCol")] = matrix_to_vector_column(
+            np.stack([1 - p1, p1], axis=1))
+        out[self.get("predictionCol")] = (raw > 0).astype(np.float64)
+        return out
+
+
+# --------------------------------------------------------------- contextual bandit
+class ContextualBanditMetrics:
+    """IPS / SNIPS estimators (vw/.../VowpalWabbitContextualBandit.scala:53)."""
+
+    def __init__(self):
+        self.total = 0
+        self.ips_num = 0.0
+        self.snips_den = 0.0
+
+    def add(self, prob_logged: float, cost: float, prob_pred_matches: float):
+        self.total += 1
+        w = prob_pred_matches / max(prob_logged, 1e-12)
+        self.ips_num += w * cost
+        self.snips_den += w
+
+    @property
+    def ips_estimate(self):
+        return self.ips_num / max(self.total, 1)
+
+    @property
+    def snips_estimate(self):
+        return self.ips_num / max(self.snips_den, 1e-12)
+
+
+@register
+class VowpalWabbitContextualBandit(_VWBase):
+    """CB cost regression with IPS weighting: the chosen action's
+    (shared ⊕ action ⊕ shared×action) features regress the observed cost with
+    importance weight 1/p_logged (reference --cb_type ips semantics)."""
+    _default_loss = "squared"
+    sharedCol = Param("sharedCol", "shared-context sparse column", "shared")
+    featuresCol = Param("featuresCol", "per-action features (list of "
+                        "SparseVector per row)", "features")
+    chosenActionCol = Param("chosenActionCol", "1-based chosen action index",
+                            "chosenAction")
+    probabilityCol = Param("probabilityCol", "logged action probability",
+                           "probability")
+    labelCol = Param("labelCol", "observed cost", "cost")
+    epsilon = Param("epsilon", "exploration for predicted policy", 0.05, toFloat)
+
+    def _combine(self, shared: SparseVector, action: SparseVector, mask: int):
+        from .featurizer import FNV_PRIME
+        si, sv = shared.indices.astype(np.int64), shared.values
+        ai, av = action.indices.astype(np.int64), action.values
+        cross_i = ((si[:, None] * FNV_PRIME) ^ ai[None, :]).reshape(-1)
+        cross_v = (sv[:, None] * av[None, :]).reshape(-1)
+        idx = np.concatenate([si, ai, cross_i & mask])
+        val = np.concatenate([sv, av, cross_v])
+        return idx.astype(np.int32), val.astype(np.float32)
+
+    def _fit(self, df: pd.DataFrame):
+        self._parse_args()
+        device = default_device(self.get("device"))
+        bits = self.get("numBits")
+        tbl = 1 << bits
+        mask = tbl - 1
+        w = torch.zeros(tbl, dtype=torch.float32, device=device)
+        g = torch.zeros(tbl, dtype=torch.float32, device=device)
+        lr, l2, pt = self.get("learningRate"), self.get("l2"), self.get("powerT")
+
+        idx_parts, val_parts, counts, labels, ws = [], [], [], [], []
+        for _, row in df.iterrows():
+            shared = row[self.get("sharedCol")]
+            actions = row[self.get("featuresCol")]
+            chosen = int(row[self.get("chosenActionCol")]) - 1
+            cost = float(row[self.get("labelCol")])
+            prob = float(row[self.get("probabilityCol")])
+            ia, va = self._combine(shared, actions[chosen], mask)
+            idx_parts.append(ia)
+            val_parts.append(va)
+            counts.append(len(ia))
+            labels.append(cost)
+            ws.append(1.0 / max(prob, 1e-6))
+        off = np.zeros(len(df) + 1, dtype=np.int64)
+        np.cumsum(counts, out=off[1:])
+        idx_t = torch.from_numpy(np.concatenate(idx_parts)).to(device)
+        val_t = torch.from_numpy(np.concatenate(val_parts)).to(device)
+        off_t = torch.from_numpy(off).to(device)
+        y_t = torch.tensor(labels, dtype=torch.float32, device=device)
+        w_t = torch.tensor(ws, dtype=torch.float32, device=device)
+
+        bs = self.get("batchSize")
+        n = len(df)
+        for _ in range(self.get("numPasses")):
+            for s in range(0, n, bs):
+                e = min(s + bs, n)
+                o = off_t[s:e + 1] - off_t[s]
+                sl = slice(int(off_t[s]), int(off_t[e]))
+                backend.vw_sgd_minibatch(idx_t[sl], val_t[sl], o, y_t[s:e],
+                                         w, g, lr, l2, pt, "squared",
+                                         w_t[s:e])
+        model = VowpalWabbitContextualBanditModel(weights=w.cpu().numpy(),
+                                                  adaptive=g.cpu().numpy())
+        for p in ("sharedCol", "featuresCol", "predictionCol", "numBits",
+                  "epsilon"):
+            model.set(p, self.get(p))
+        return model
+
+
+@register
+class VowpalWabbitContextualBanditModel(_VWModelBase):
+    sharedCol = Param("sharedCol", "shared-context sparse column", "shared")
+    epsilon = Param("epsilon", "exploration rate", 0.05, toFloat)
+
+    def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
+        est = VowpalWabbitContextualBandit()
+        est.set("numBits", self.get("numBits"))
+        mask = (1 << self.get("numBits")) - 1
+        w = self.weights
+        scores_col = []
+        chosen_col = []
+        for _, row in df.iterrows():
+            shared = row[self.get("sharedCol")]
+            actions = row[self.get("featuresCol")]
+            scores = []
+            for a in actions:
+                ia, va = est._combine(shared, a, mask)
+                scores.append(float((w[ia] * va).sum()))
+            scores_col.append(np.asarray(scores, dtype=np.float32))
+            chosen_col.append(int(np.argmin(scores)) + 1)
+        out = df.copy()
+        out["predictedCosts"] = scores_col
+        out[self.get("predictionCol")] = chosen_col
+        return out

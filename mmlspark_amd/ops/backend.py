@@ -102,15 +102,16 @@ def bin_matrix(X, upper_bounds, n_bins):
 
 # --------------------------------------------------------------------------- VW
 def vw_sgd_minibatch(indices, values, offsets, labels, weights_tbl, adaptive_tbl,
-                     lr, l2, power_t, loss: str, weight_decay_rounds: int = 1):
+                     lr, l2, power_t, loss: str, ex_weight=None):
     if weights_tbl.is_cuda:
         return _require_ext().vw_sgd_minibatch(
             indices, values, offsets, labels, weights_tbl, adaptive_tbl,
-            lr, l2, power_t, {"squared": 0, "logistic": 1, "hinge": 2}[loss])
+            lr, l2, power_t, {"squared": 0, "logistic": 1, "hinge": 2}[loss],
+            ex_weight)
     from ..models.vw import sgd_ref
     return sgd_ref.vw_sgd_minibatch(indices, values, offsets, labels,
                                     weights_tbl, adaptive_tbl, lr, l2,
-                                    power_t, loss)
+                                    power_t, loss, ex_weight)
 
 
 def vw_predict(indices, values, offsets, weights_tbl):

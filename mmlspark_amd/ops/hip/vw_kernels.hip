@@ -35,6 +35,7 @@ __global__ void vw_sgd_k(const int* __restrict__ idx,
                          const float* __restrict__ val,
                          const long* __restrict__ off,
                          const float* __restrict__ label,
+                         const float* __restrict__ ex_weight,
                          float* __restrict__ w_tbl, float* __restrict__ g_tbl,
                          float lr, float l2, float power_t, int loss,
                          long n_ex, float* __restrict__ preds_out) {
@@ -47,7 +48,8 @@ __global__ void vw_sgd_k(const int* __restrict__ idx,
     for (long k = s + lane; k < e; k += WAVE) dot += w_tbl[idx[k]] * val[k];
     const float pred = wave_sum(dot);
     if (preds_out && lane == 0) preds_out[ex] = pred;
-    const float gl = dloss(loss, pred, label[ex]);
+    float gl = dloss(loss, pred, label[ex]);
+    if (ex_weight) gl *= ex_weight[ex];
     if (gl == 0.0f) continue;
     for (long k = s + lane; k < e; k += WAVE) {
       const int i = idx[k];
@@ -66,6 +68,7 @@ __global__ void vw_sgd_k(const int* __restrict__ idx,
 
 extern "C" void launch_vw_sgd(const int* idx, const float* val,
                               const long* off, const float* label,
+                              const float* ex_weight,
                               float* w_tbl, float* g_tbl, float lr, float l2,
                               float power_t, int loss, long n_ex,
                               float* preds_out, hipStream_t stream) {
@@ -74,8 +77,8 @@ extern "C" void launch_vw_sgd(const int* idx, const float* val,
   long blocks = (waves * WAVE + 255) / 256;
   if (blocks > 4096) blocks = 4096;
   hipLaunchKernelGGL(vw_sgd_k, dim3((unsigned)blocks), dim3(256), 0, stream,
-                     idx, val, off, label, w_tbl, g_tbl, lr, l2, power_t,
-                     loss, n_ex, preds_out);
+                     idx, val, off, label, ex_weight, w_tbl, g_tbl, lr, l2,
+                     power_t, loss, n_ex, preds_out);
 }
 
 __global__ void vw_predict_k(const int* __restrict__ idx,

@@ -1,0 +1,157 @@
+"""VW-equivalent module: murmur parity, featurizer, SGD learners, CB."""
+import numpy as np
+import pandas as pd
+import pytest
+
+from mmlspark_amd.core.schema import SparseVector
+from mmlspark_amd.models.vw.estimators import (
+    VowpalWabbitClassifier, VowpalWabbitContextualBandit,
+    VowpalWabbitRegressor)
+from mmlspark_amd.models.vw.featurizer import (VowpalWabbitFeaturizer,
+                                               VowpalWabbitInteractions)
+from mmlspark_amd.models.vw.murmur import hash_string, murmur3_32
+
+
+def test_murmur3_known_vectors():
+    # canonical MurmurHash3_x86_32 test vectors
+    assert murmur3_32(b"", 0) == 0
+    assert murmur3_32(b"", 1) == 0x514E28B7
+    assert murmur3_32(b"hello", 0) == 0x248BFA47
+    assert murmur3_32(b"hello, world", 0) == 0x149BBB7F
+    assert murmur3_32(b"The quick brown fox jumps over the lazy dog", 0) == 0x2E4FF723
+
+
+def test_featurizer_basic():
+    df = pd.DataFrame({
+        "num": [1.5, 0.0, 2.0],
+        "cat": ["a", "b", "a"],
+        "txt": ["hello world", "foo", ""],
+    })
+    f = VowpalWabbitFeaturizer(inputCols=["num", "cat"],
+                               stringSplitInputCols=["txt"], numBits=15)
+    out = f.transform(df)
+    vecs = out["features"].to_numpy()
+    assert all(isinstance(v, SparseVector) for v in vecs)
+    assert vecs[0].size == 1 << 15
+    # row0: num + cat + 2 tokens = 4 features (assuming no collisions)
+    assert len(vecs[0].indices) == 4
+    # zero numeric dropped: row1 has cat + 1 token = 2
+    assert len(vecs[1].indices) == 2
+    # same string → same hash
+    same_cat = set(vecs[0].indices) & set(vecs[2].indices)
+    assert len(same_cat) >= 1
+
+
+def test_featurizer_deterministic_and_sorted():
+    df = pd.DataFrame({"a": ["x"], "b": [2.0]})
+    f = VowpalWabbitFeaturizer(inputCols=["a", "b"], numBits=10)
+    v1 = f.transform(df)["features"].iloc[0]
+    v2 = f.transform(df)["features"].iloc[0]
+    assert v1 == v2
+    assert (np.diff(v1.indices) > 0).all()
+
+
+def test_interactions():
+    a = SparseVector(1 << 10, [1, 5], [1.0, 2.0])
+    b = SparseVector(1 << 10, [3], [4.0])
+    df = pd.DataFrame({"c1": [a], "c2": [b]})
+    t = VowpalWabbitInteractions(inputCols=["c1", "c2"], numBits=10)
+    v = t.transform(df)["interactions"].iloc[0]
+    assert len(v.indices) == 2
+    assert sorted(v.values.tolist()) == [4.0, 8.0]
+
+
+def _hashed_binary_data(n=5000, nf=30, bits=16, seed=0):
+    rng = np.random.default_rng(seed)
+    size = 1 << bits
+    feat_ids = rng.integers(0, size, size=200)
+    w_true = rng.normal(size=size) * 0.0
+    w_true[feat_ids] = rng.normal(size=len(feat_ids))
+    rows = []
+    labels = []
+    for _ in range(n):
+        k = rng.integers(5, nf)
+        idx = np.unique(rng.choice(feat_ids, size=k))
+        val = rng.normal(size=len(idx)).astype(np.float32)
+        margin = float((w_true[idx] * val).sum())
+        rows.append(SparseVector(size, idx.astype(np.int32), val))
+        labels.append(1.0 if margin + rng.normal() * 0.3 > 0 else 0.0)
+    return pd.DataFrame({"features": rows, "label": labels})
+
+
+def test_vw_classifier_learns():
+    from sklearn.metrics import roc_auc_score
+    df = _hashed_binary_data()
+    m = VowpalWabbitClassifier(numPasses=5, numBits=16, learningRate=0.5).fit(df)
+    out = m.transform(df)
+    prob = np.stack(out["probability"].to_numpy())[:, 1]
+    assert roc_auc_score(df["label"], prob) > 0.85
+    stats = m.getPerformanceStatistics()
+    assert stats["numberOfExamplesPerPass"].iloc[0] == len(df)
+
+
+def test_vw_regressor_learns():
+    rng = np.random.default_rng(1)
+    size = 1 << 14
+    n = 4000
+    w_true = rng.normal(size=size) * 0.05
+    rows, ys = [], []
+    for _ in range(n):
+        idx = np.unique(rng.integers(0, size, size=20))
+        val = np.ones(len(idx), dtype=np.float32)
+        rows.append(SparseVector(size, idx.astype(np.int32), val))
+        ys.append(float((w_true[idx] * val).sum()))
+    df = pd.DataFrame({"features": rows, "label": ys})
+    m = VowpalWabbitRegressor(numPasses=10, numBits=14, learningRate=0.3).fit(df)
+    pred = m.transform(df)["prediction"].to_numpy()
+    y = np.asarray(ys)
+    ss_res = ((pred - y) ** 2).sum()
+    ss_tot = ((y - y.mean()) ** 2).sum()
+    assert 1 - ss_res / ss_tot > 0.5
+
+
+def test_vw_args_string():
+    est = VowpalWabbitRegressor(passThroughArgs="--l2 0.01 -b 20 --passes 3")
+    est._parse_args()
+    assert est.get("l2") == 0.01
+    assert est.get("numBits") == 20
+    assert est.get("numPasses") == 3
+
+
+def test_vw_save_load(tmp_path):
+    import os
+    df = _hashed_binary_data(n=500)
+    m = VowpalWabbitClassifier(numPasses=2, numBits=16).fit(df)
+    p1 = np.stack(m.transform(df)["probability"].to_numpy())
+    path = os.path.join(tmp_path, "vw")
+    m.save(path)
+    from mmlspark_amd.models.vw.estimators import VowpalWabbitClassificationModel
+    m2 = VowpalWabbitClassificationModel.load(path)
+    p2 = np.stack(m2.transform(df)["probability"].to_numpy())
+    np.testing.assert_allclose(p1, p2)
+
+
+def test_contextual_bandit():
+    rng = np.random.default_rng(2)
+    size = 1 << 14
+    n = 2000
+    n_actions = 3
+    # context bit decides which action is cheapest
+    rows = []
+    for _ in range(n):
+        ctx = int(rng.integers(0, n_actions))
+        shared = SparseVector(size, [100 + ctx], [1.0])
+        actions = [SparseVector(size, [2000 + a], [1.0])
+                   for a in range(n_actions)]
+        logged = int(rng.integers(0, n_actions))
+        cost = 0.0 if logged == ctx else 1.0
+        rows.append({"shared": shared, "features": actions,
+                     "chosenAction": logged + 1, "cost": cost,
+                     "probability": 1.0 / n_actions, "ctx": ctx})
+    df = pd.DataFrame(rows)
+    cb = VowpalWabbitContextualBandit(numPasses=5, numBits=14,
+                                      learningRate=0.5).fit(df)
+    out = cb.transform(df)
+    picked = out["prediction"].to_numpy() - 1
+    ctx = df["ctx"].to_numpy()
+    assert (picked == ctx).mean() > 0.9
