@@ -1,0 +1,116 @@
+"""IsolationForest, KNN/ConditionalKNN, SAR."""
+import numpy as np
+import pandas as pd
+import pytest
+
+from mmlspark_amd.models.iforest import IsolationForest, IsolationForestModel
+from mmlspark_amd.models.knn import KNN, ConditionalKNN
+from mmlspark_amd.models.sar import (SAR, RankingEvaluator,
+                                     RecommendationIndexer)
+
+
+def test_isolation_forest_finds_outliers():
+    rng = np.random.default_rng(0)
+    inliers = rng.normal(size=(500, 4)).astype(np.float32)
+    outliers = rng.normal(size=(20, 4)).astype(np.float32) * 6 + 10
+    X = np.concatenate([inliers, outliers])
+    df = pd.DataFrame({"features": list(X)})
+    m = IsolationForest(numEstimators=50, contamination=0.04,
+                        randomSeed=3).fit(df)
+    out = m.transform(df)
+    scores = out["outlierScore"].to_numpy()
+    assert scores[500:].mean() > scores[:500].mean() + 0.1
+    pred = out["predictedLabel"].to_numpy()
+    # most flagged points are true outliers
+    assert pred[500:].mean() > 0.7
+    assert pred[:500].mean() < 0.05
+
+
+def test_isolation_forest_save_load(tmp_path):
+    import os
+    rng = np.random.default_rng(1)
+    X = rng.normal(size=(200, 3)).astype(np.float32)
+    df = pd.DataFrame({"features": list(X)})
+    m = IsolationForest(numEstimators=10).fit(df)
+    s1 = m.transform(df)["outlierScore"].to_numpy()
+    m.save(os.path.join(tmp_path, "if"))
+    m2 = IsolationForestModel.load(os.path.join(tmp_path, "if"))
+    s2 = m2.transform(df)["outlierScore"].to_numpy()
+    np.testing.assert_allclose(s1, s2)
+
+
+def test_knn_exact():
+    rng = np.random.default_rng(2)
+    X = rng.normal(size=(300, 8)).astype(np.float32)
+    df = pd.DataFrame({"features": list(X), "values": [f"v{i}" for i in range(300)]})
+    m = KNN(k=3).fit(df)
+    q = pd.DataFrame({"features": [X[7], X[42]]})
+    out = m.transform(q)
+    matches = out["output"].to_numpy()
+    assert matches[0][0]["value"] == "v7"      # self is nearest
+    assert matches[0][0]["distance"] < 1e-4
+    assert matches[1][0]["value"] == "v42"
+    # verify against brute force
+    d = ((X - X[7]) ** 2).sum(axis=1)
+    expect = set(np.argsort(d)[:3])
+    got = {int(v["value"][1:]) for v in matches[0]}
+    assert got == expect
+
+
+def test_conditional_knn_label_filter():
+    rng = np.random.default_rng(3)
+    X = rng.normal(size=(200, 5)).astype(np.float32)
+    labels = np.array([i % 4 for i in range(200)])
+    df = pd.DataFrame({"features": list(X), "values": list(range(200)),
+                       "labels": labels})
+    m = ConditionalKNN(k=4).fit(df)
+    q = pd.DataFrame({"features": [X[0], X[1]],
+                      "conditioner": [[1, 2], [0]]})
+    out = m.transform(q)
+    for i, allowed in enumerate([[1, 2], [0]]):
+        for match in out["output"].iloc[i]:
+            assert match["label"] in allowed
+
+
+def test_sar_recommendations():
+    # two user groups with disjoint item tastes
+    rows = []
+    for u in range(20):
+        items = [0, 1, 2] if u < 10 else [5, 6, 7]
+        for it in items:
+            rows.append({"user": f"u{u}", "item": f"i{it}", "rating": 1.0})
+    # one crossover item to create co-occurrence signal
+    rows.append({"user": "u0", "item": "i3", "rating": 1.0})
+    rows.append({"user": "u1", "item": "i3", "rating": 1.0})
+    rows.append({"user": "u2", "item": "i3", "rating": 1.0})
+    rows.append({"user": "u3", "item": "i3", "rating": 1.0})
+    df = pd.DataFrame(rows)
+    idx = RecommendationIndexer().fit(df)
+    dfi = idx.transform(df)
+    sar = SAR(supportThreshold=1, similarityFunction="jaccard").fit(dfi)
+    recs = sar.recommendForAllUsers(2)
+    item_map = idx.get("itemMap")
+    i3 = item_map["i3"]
+    # users 4..9 (group A, never saw i3) should get i3 recommended
+    hit = 0
+    for u in range(4, 10):
+        uidx = idx.get("userMap")[f"u{u}"]
+        rec_items = [r["itemIdx"] for r in
+                     recs[recs["userIdx"] == uidx]["recommendations"].iloc[0]]
+        hit += int(i3 in rec_items)
+    assert hit >= 5
+
+    scored = sar.transform(dfi.head(5))
+    assert "prediction" in scored.columns
+
+
+def test_ranking_evaluator():
+    df = pd.DataFrame({
+        "prediction": [[1, 2, 3], [4, 5, 6]],
+        "label": [[1, 3], [9]],
+    })
+    ev = RankingEvaluator(k=3, metricName="precisionAtk")
+    assert abs(ev.evaluate(df) - (2 / 3 + 0) / 2) < 1e-9
+    ev2 = RankingEvaluator(k=3, metricName="ndcgAt")
+    v = ev2.evaluate(df)
+    assert 0 < v < 1
