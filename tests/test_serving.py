@@ -1,0 +1,153 @@
+"""Serving server (continuous + micro-batch) and HTTP client stack."""
+import json
+import threading
+import time
+
+import numpy as np
+import pandas as pd
+import pytest
+import requests
+
+from mmlspark_amd.serving.server import (LowLatencyGBDTScorer, ServingServer,
+                                         TransformerHandler)
+
+
+def _echo_handler(payloads):
+    return [{"echo": p.get("x", None)} for p in payloads]
+
+
+def test_continuous_serving_roundtrip():
+    srv = ServingServer(_echo_handler, port=0, mode="continuous").start()
+    try:
+        url = f"http://127.0.0.1:{srv.port}/"
+        r = requests.post(url, json={"x": 42}, timeout=5)
+        assert r.status_code == 200
+        assert r.json() == {"echo": 42}
+        info = requests.get(url + "__service_info", timeout=5).json()
+        assert info["mode"] == "continuous" and info["served"] >= 1
+    finally:
+        srv.stop()
+
+
+def test_micro_batch_serving_batches_requests():
+    seen_batches = []
+
+    def handler(payloads):
+        seen_batches.append(len(payloads))
+        return [{"y": p["x"] * 2} for p in payloads]
+
+    srv = ServingServer(handler, port=0, mode="micro-batch",
+                        batch_wait_ms=50).start()
+    try:
+        url = f"http://127.0.0.1:{srv.port}/"
+        results = {}
+
+        def post(i):
+            results[i] = requests.post(url, json={"x": i}, timeout=10).json()
+
+        threads = [threading.Thread(target=post, args=(i,)) for i in range(8)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        assert all(results[i] == {"y": 2 * i} for i in range(8))
+        assert max(seen_batches) > 1  # actually batched
+    finally:
+        srv.stop()
+
+
+def test_micro_batch_replay_on_failure():
+    calls = {"n": 0}
+
+    def flaky(payloads):
+        calls["n"] += 1
+        if calls["n"] == 1:
+            raise RuntimeError("transient")
+        return [{"ok": True} for _ in payloads]
+
+    srv = ServingServer(flaky, port=0, mode="micro-batch",
+                        reply_timeout=10).start()
+    try:
+        url = f"http://127.0.0.1:{srv.port}/"
+        r = requests.post(url, json={"x": 1}, timeout=10)
+        assert r.json() == {"ok": True}  # replayed after first failure
+        assert calls["n"] >= 2
+    finally:
+        srv.stop()
+
+
+def test_serving_model_handler(binary_df):
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    model = LightGBMClassifier(numIterations=5, numLeaves=7).fit(binary_df)
+    handler = TransformerHandler(model, ["prediction"])
+    srv = ServingServer(handler, port=0, mode="continuous").start()
+    try:
+        url = f"http://127.0.0.1:{srv.port}/"
+        x = binary_df["features"].iloc[0].tolist()
+        r = requests.post(url, json={"features": x}, timeout=10)
+        assert r.status_code == 200
+        assert r.json()["prediction"] in (0.0, 1.0)
+    finally:
+        srv.stop()
+
+
+def test_low_latency_scorer_cpu(binary_df):
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    model = LightGBMClassifier(numIterations=5, numLeaves=7).fit(binary_df)
+    scorer = LowLatencyGBDTScorer(model.booster, max_batch=4)
+    X = np.stack(binary_df["features"].to_numpy()[:3])
+    p = scorer.score(X)
+    ref = np.stack(model.transform(binary_df.head(3))["probability"]
+                   .to_numpy())[:, 1]
+    np.testing.assert_allclose(p[:, 0], ref, atol=1e-5)
+
+
+def test_http_transformer_against_local_server():
+    from mmlspark_amd.io_http.client import (HTTPTransformer, JSONInputParser,
+                                             JSONOutputParser,
+                                             SimpleHTTPTransformer)
+    from mmlspark_amd.io_http.http_schema import HTTPRequestData
+
+    srv = ServingServer(_echo_handler, port=0, mode="continuous").start()
+    try:
+        url = f"http://127.0.0.1:{srv.port}/"
+        df = pd.DataFrame({"payload": [{"x": 1}, {"x": 2}]})
+        parsed = JSONInputParser(inputCol="payload", outputCol="request",
+                                 url=url).transform(df)
+        assert isinstance(parsed["request"].iloc[0], HTTPRequestData)
+        resp = HTTPTransformer(inputCol="request", outputCol="response",
+                               concurrency=2).transform(parsed)
+        out = JSONOutputParser(inputCol="response",
+                               outputCol="parsed").transform(resp)
+        assert out["parsed"].tolist() == [{"echo": 1}, {"echo": 2}]
+
+        simple = SimpleHTTPTransformer(inputCol="payload", outputCol="out",
+                                       url=url, concurrency=2)
+        out2 = simple.transform(df)
+        assert out2["out"].tolist() == [{"echo": 1}, {"echo": 2}]
+        assert out2["errors"].isna().all()
+    finally:
+        srv.stop()
+
+
+def test_serving_p50_latency_cpu(binary_df):
+    """Latency smoke on CPU (the real p50 target is measured on the GPU box
+    via bench_serving.py; reference bar: 'sub-millisecond')."""
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    model = LightGBMClassifier(numIterations=5, numLeaves=7).fit(binary_df)
+    scorer = LowLatencyGBDTScorer(model.booster, max_batch=1, use_graph=False)
+    srv = ServingServer(scorer, port=0, mode="continuous").start()
+    try:
+        url = f"http://127.0.0.1:{srv.port}/"
+        x = binary_df["features"].iloc[0].tolist()
+        s = requests.Session()
+        lat = []
+        for _ in range(50):
+            t0 = time.perf_counter()
+            r = s.post(url, json={"features": x}, timeout=5)
+            lat.append(time.perf_counter() - t0)
+            assert r.status_code == 200
+        p50 = sorted(lat)[len(lat) // 2]
+        assert p50 < 0.05, p50  # generous CPU bound; GPU bench asserts harder
+    finally:
+        srv.stop()
