@@ -1,0 +1,118 @@
+"""ImageTransformer / ImageSetAugmenter + cognitive HTTP transformers
+(against local mock services)."""
+import json
+
+import numpy as np
+import pandas as pd
+import pytest
+import requests
+
+from mmlspark_amd.models.images import ImageSetAugmenter, ImageTransformer
+from mmlspark_amd.serving.server import ServingServer
+
+
+def _img(h=20, w=30):
+    rng = np.random.default_rng(0)
+    return rng.integers(0, 255, size=(h, w, 3)).astype(np.uint8)
+
+
+def test_image_transformer_pipeline():
+    df = pd.DataFrame({"image": [_img()]})
+    t = (ImageTransformer()
+         .resize(10, 12)
+         .crop(2, 2, 6, 8)
+         .flip(1)
+         .blur(3, 3))
+    out = t.transform(df)
+    res = out["out_image"].iloc[0]
+    assert res.shape == (6, 8, 3)
+
+
+def test_image_transformer_ops():
+    img = _img()
+    df = pd.DataFrame({"image": [img]})
+    gray = ImageTransformer().colorFormat("gray").transform(df)["out_image"].iloc[0]
+    assert gray.shape == (20, 30, 1)
+    th = ImageTransformer().threshold(128).transform(df)["out_image"].iloc[0]
+    assert set(np.unique(th)).issubset({0, 255})
+    g = ImageTransformer().gaussianKernel(5, 1.5).transform(df)["out_image"].iloc[0]
+    assert g.shape == img.shape
+    flipped = ImageTransformer().flip(1).transform(df)["out_image"].iloc[0]
+    np.testing.assert_array_equal(flipped, img[:, ::-1])
+
+
+def test_image_set_augmenter():
+    df = pd.DataFrame({"image": [_img(), _img()]})
+    out = ImageSetAugmenter(flipLeftRight=True, flipUpDown=True).transform(df)
+    assert len(out) == 6
+
+
+def test_image_transformer_save_load(tmp_path):
+    import os
+    t = ImageTransformer().resize(8, 8).threshold(100)
+    t.save(os.path.join(tmp_path, "it"))
+    t2 = ImageTransformer.load(os.path.join(tmp_path, "it"))
+    df = pd.DataFrame({"image": [_img()]})
+    np.testing.assert_array_equal(t.transform(df)["out_image"].iloc[0],
+                                  t2.transform(df)["out_image"].iloc[0])
+
+
+# ----------------------------------------------------------------- cognitive
+@pytest.fixture(scope="module")
+def mock_service():
+    """Local mock that echoes a sentiment-like response."""
+    def handler(payloads):
+        out = []
+        for p in payloads:
+            docs = p.get("documents", [])
+            out.append({"documents": [
+                {"id": d["id"], "sentiment":
+                 "positive" if "good" in d.get("text", "") else "negative"}
+                for d in docs], "errors": []})
+        return out
+    srv = ServingServer(handler, port=0, mode="continuous").start()
+    yield f"http://127.0.0.1:{srv.port}/"
+    srv.stop()
+
+
+def test_text_sentiment_against_mock(mock_service):
+    from mmlspark_amd.io_http.cognitive import TextSentiment
+    df = pd.DataFrame({"text": ["good day", "awful day"],
+                       "key": ["k1", "k2"]})
+    ts = TextSentiment(url=mock_service, subscriptionKeyCol="key",
+                       textCol="text", concurrency=2)
+    out = ts.transform(df)
+    r0 = out["response"].iloc[0]
+    assert r0["documents"][0]["sentiment"] == "positive"
+    r1 = out["response"].iloc[1]
+    assert r1["documents"][0]["sentiment"] == "negative"
+    assert out["errors"].isna().all()
+
+
+def test_cognitive_error_column():
+    from mmlspark_amd.io_http.cognitive import TextSentiment
+    # unreachable endpoint -> error column populated, no exception
+    df = pd.DataFrame({"text": ["x"]})
+    ts = TextSentiment(url="http://127.0.0.1:1/", textCol="text", timeout=0.5)
+    out = ts.transform(df)
+    assert out["errors"].iloc[0] is not None
+
+
+def test_anomaly_detector_shape(mock_service):
+    from mmlspark_amd.io_http.cognitive import DetectLastAnomaly
+    series = [{"timestamp": f"2020-01-{i+1:02d}T00:00:00Z", "value": float(i)}
+              for i in range(12)]
+    df = pd.DataFrame({"series": [series]})
+    det = DetectLastAnomaly(url=mock_service, seriesCol="series",
+                            granularity="daily")
+    out = det.transform(df)
+    assert out["response"].iloc[0] is not None
+
+
+def test_bing_image_search_get(mock_service):
+    from mmlspark_amd.io_http.cognitive import BingImageSearch
+    df = pd.DataFrame({"q": ["cats"]})
+    b = BingImageSearch(url=mock_service, qCol="q", count=3)
+    out = b.transform(df)
+    # mock returns a JSON body for GET-with-params too (echo handler tolerates)
+    assert "response" in out.columns
