@@ -1,0 +1,85 @@
+#!/usr/bin/env python3
+"""Serving p50 latency benchmark (BASELINE config #6: reference bar is
+'sub-millisecond' continuous-mode latency, docs/mmlspark-serving.md:10).
+
+Trains a small GBDT, serves it in continuous mode through the
+hipGraph-captured low-latency scorer, and measures end-to-end HTTP p50/p90/p99
+over loopback plus the raw scorer latency (no HTTP)."""
+import argparse
+import json
+import time
+
+import numpy as np
+import pandas as pd
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--requests", type=int, default=2000)
+    ap.add_argument("--trees", type=int, default=100)
+    ap.add_argument("--features", type=int, default=28)
+    args = ap.parse_args()
+
+    import requests as rq
+    import torch
+
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    from mmlspark_amd.serving.server import LowLatencyGBDTScorer, ServingServer
+
+    rng = np.random.default_rng(0)
+    n, nf = 20000, args.features
+    X = rng.normal(size=(n, nf)).astype(np.float32)
+    y = (X[:, 0] + X[:, 1] > 0).astype(np.float32)
+    df = pd.DataFrame({"features": list(X), "label": y})
+    model = LightGBMClassifier(numIterations=args.trees, numLeaves=31).fit(df)
+
+    scorer = LowLatencyGBDTScorer(model.booster, max_batch=1)
+    x0 = X[0]
+
+    # raw scorer latency (no HTTP)
+    for _ in range(50):
+        scorer.score(x0[None, :])
+    lat_raw = []
+    for _ in range(args.requests):
+        t0 = time.perf_counter()
+        scorer.score(x0[None, :])
+        lat_raw.append((time.perf_counter() - t0) * 1e3)
+
+    srv = ServingServer(scorer, port=0, mode="continuous").start()
+    try:
+        url = f"http://127.0.0.1:{srv.port}/"
+        s = rq.Session()
+        body = {"features": x0.tolist()}
+        for _ in range(100):
+            s.post(url, json=body, timeout=5)
+        lat = []
+        for _ in range(args.requests):
+            t0 = time.perf_counter()
+            r = s.post(url, json=body, timeout=5)
+            lat.append((time.perf_counter() - t0) * 1e3)
+            assert r.status_code == 200
+    finally:
+        srv.stop()
+
+    lat.sort()
+    lat_raw.sort()
+    q = lambda a, p: a[int(p * len(a))]
+    print(json.dumps({
+        "metric": "serving_p50_latency_ms",
+        "value": q(lat, 0.5),
+        "unit": "ms",
+        "higher_is_better": False,
+        "http_p50_ms": q(lat, 0.5),
+        "http_p90_ms": q(lat, 0.9),
+        "http_p99_ms": q(lat, 0.99),
+        "scorer_p50_ms": q(lat_raw, 0.5),
+        "scorer_p99_ms": q(lat_raw, 0.99),
+        "hipgraph": scorer.graph is not None,
+        "config": {"trees": args.trees, "features": args.features,
+                   "mode": "continuous", "transport": "loopback HTTP"},
+        "reference_bar": "sub-millisecond (docs/mmlspark-serving.md:10)",
+    }), flush=True)
+
+
+if __name__ == "__main__":
+    main()
