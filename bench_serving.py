@@ -47,17 +47,28 @@ def main():
 
     srv = ServingServer(scorer, port=0, mode="continuous").start()
     try:
-        url = f"http://127.0.0.1:{srv.port}/"
-        s = rq.Session()
-        body = {"features": x0.tolist()}
+        import http.client
+        import socket
+        conn = http.client.HTTPConnection("127.0.0.1", srv.port)
+        conn.connect()
+        conn.sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+        body = json.dumps({"features": x0.tolist()})
+        hdrs = {"Content-Type": "application/json"}
+
+        def one():
+            conn.request("POST", "/", body=body, headers=hdrs)
+            r = conn.getresponse()
+            r.read()
+            return r.status
+
         for _ in range(100):
-            s.post(url, json=body, timeout=5)
+            assert one() == 200
         lat = []
         for _ in range(args.requests):
             t0 = time.perf_counter()
-            r = s.post(url, json=body, timeout=5)
+            code = one()
             lat.append((time.perf_counter() - t0) * 1e3)
-            assert r.status_code == 200
+            assert code == 200
     finally:
         srv.stop()
 

@@ -72,6 +72,11 @@ class ServingServer:
 
         class Handler(BaseHTTPRequestHandler):
             protocol_version = "HTTP/1.1"
+            # latency: no Nagle on the accepted socket + fully-buffered writes
+            # so status+headers+body leave as ONE segment (else delayed-ACK
+            # interplay costs ~40 ms per response on loopback)
+            disable_nagle_algorithm = True
+            wbufsize = -1
 
             def log_message(self, *a):  # quiet
                 pass
@@ -162,8 +167,14 @@ class ServingServer:
 
     # ------------------------------------------------------------- lifecycle
     def start(self):
-        self._httpd = ThreadingHTTPServer((self.host, self.port),
-                                          self._make_handler())
+        class _Server(ThreadingHTTPServer):
+            # loopback latency: disable Nagle (otherwise delayed-ACK + Nagle
+            # adds ~40 ms per response on keep-alive connections)
+            disable_nagle_algorithm = True
+            daemon_threads = True
+
+        self._httpd = _Server((self.host, self.port),
+                              self._make_handler())
         self.port = self._httpd.server_port
         t = threading.Thread(target=self._httpd.serve_forever, daemon=True)
         t.start()
@@ -239,7 +250,10 @@ class LowLatencyGBDTScorer:
         self.graph = None
         self.out = None
         if use_graph and self.device.type == "cuda":
-            self._capture()
+            try:
+                self._capture()
+            except Exception:
+                self.graph = None  # fall back to plain launches
 
     def _raw(self):
         from ..ops import backend
