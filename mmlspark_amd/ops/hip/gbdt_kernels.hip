@@ -72,12 +72,21 @@ __global__ void hist_build_k(const uchar4* __restrict__ binned, long n_rows,
   }
 }
 
+static int hist_gpb_env() {
+  const char* e = getenv("MMLSPARK_HIST_GPB");
+  if (e) {
+    int v = atoi(e);
+    if (v == 1 || v == 2 || v == 4 || v == 8) return v;
+  }
+  return 4;
+}
+
 extern "C" void launch_hist_build(const void* binned, long n_rows,
                                   const int* rows, long m, const float* grad,
                                   const float* hess, float* hist, int n_bins,
                                   int ngroups, hipStream_t stream) {
-  constexpr int GPB = 4;
   if (m == 0) return;
+  const int GPB = hist_gpb_env();
   const int n_fblocks = (ngroups + GPB - 1) / GPB;
   // target ≥ 2048 workgroups total to fill 256 CUs × 8 XCDs
   long chunks = (2048 + n_fblocks - 1) / n_fblocks;
@@ -87,9 +96,17 @@ extern "C" void launch_hist_build(const void* binned, long n_rows,
   chunks = (m + chunk - 1) / chunk;
   dim3 grid((unsigned)chunks, (unsigned)n_fblocks);
   const size_t lds_bytes = (size_t)GPB * 4 * n_bins * 3 * sizeof(float);
-  hipLaunchKernelGGL((hist_build_k<GPB>), grid, dim3(256), lds_bytes, stream,
-                     (const uchar4*)binned, n_rows, rows, m, grad, hess, hist,
-                     n_bins, ngroups, chunk);
+#define LAUNCH_HB(G) \
+    hipLaunchKernelGGL((hist_build_k<G>), grid, dim3(256), lds_bytes, stream, \
+                       (const uchar4*)binned, n_rows, rows, m, grad, hess,    \
+                       hist, n_bins, ngroups, chunk)
+  switch (GPB) {
+    case 1: LAUNCH_HB(1); break;
+    case 2: LAUNCH_HB(2); break;
+    case 8: LAUNCH_HB(8); break;
+    default: LAUNCH_HB(4); break;
+  }
+#undef LAUNCH_HB
 }
 
 // ------------------------------------------------------------ forest predict
