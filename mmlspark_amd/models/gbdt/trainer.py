@@ -123,7 +123,7 @@ class TreeGrower:
 
     def __init__(self, binned_i4: torch.Tensor, n_features: int,
                  cfg: TrainConfig, comm: Comm, stats: TrainingStats,
-                 bin_mapper: BinMapper):
+                 bin_mapper: BinMapper, n_global: Optional[int] = None):
         self.binned = binned_i4
         self.nf = n_features
         self.nf_pad = binned_i4.shape[0] * 4
@@ -132,12 +132,36 @@ class TreeGrower:
         self.stats = stats
         self.bin_mapper = bin_mapper
         self.device = binned_i4.device
+        # fixed-point integer histograms on GPU (9x faster LDS atomics;
+        # bit-exact all_reduce + sibling subtraction)
+        self.fixed = self.device.type == "cuda"
+        self.n_global = n_global or binned_i4.shape[1]
+        self.scale_g = 1.0
+        self.scale_h = 1.0
 
     _sync_timers = bool(__import__("os").environ.get("MMLSPARK_AMD_SYNC_TIMERS"))
 
+    def set_scales(self, grad: torch.Tensor, hess: torch.Tensor):
+        """Per-tree fixed-point scales from GLOBAL max|g|, max h (identical on
+        every rank: one tiny all_reduce)."""
+        if not self.fixed:
+            return
+        mx = torch.stack([grad.abs().max(), hess.max()])
+        self.comm.all_reduce(mx, op="max")
+        gmax = max(float(mx[0]), 1e-12)
+        hmax = max(float(mx[1]), 1e-12)
+        self.scale_g = (2.0 ** 61) / (max(self.n_global, 1) * gmax)
+        self.scale_h = (2.0 ** 24) / hmax  # chunk ≤ 2^19 rows → fits 44 bits
+
     def _hist(self, rows, grad, hess, reduce=True):
         t0 = time.perf_counter()
-        h = backend.hist_build(self.binned, rows, grad, hess, self.cfg.max_bin)
+        if self.fixed:
+            h = backend.hist_build_fixed(self.binned, rows, grad, hess,
+                                         self.cfg.max_bin, self.scale_g,
+                                         self.scale_h)
+        else:
+            h = backend.hist_build(self.binned, rows, grad, hess,
+                                   self.cfg.max_bin)
         if self._sync_timers and self.device.type == "cuda":
             torch.cuda.synchronize()
         t1 = time.perf_counter()
@@ -147,15 +171,32 @@ class TreeGrower:
             self.stats.comm_s += time.perf_counter() - t1
         return h
 
+    def _to_float_hist(self, hists: torch.Tensor) -> torch.Tensor:
+        """Stacked histograms → float32 real units for the split scan."""
+        if not self.fixed:
+            return hists
+        inv = torch.tensor([1.0 / self.scale_g, 1.0 / self.scale_h, 1.0],
+                           device=hists.device)
+        return (hists.double() * inv.double()).float()
+
+    def _sums(self, hist: torch.Tensor):
+        """(G, H, C) totals of one histogram (feature 0 owns every row)."""
+        s = hist[0].sum(dim=0)
+        if self.fixed:
+            return (float(s[0]) / self.scale_g, float(s[1]) / self.scale_h,
+                    float(s[2]))
+        return float(s[0]), float(s[1]), float(s[2])
+
     def _scan(self, hists, feat_mask):
         """Fused split scan on stacked histograms; identical on all ranks
         (input is the reduced histogram). One kernel pair + one readback."""
         t0 = time.perf_counter()
         cfg = self.cfg
         out = backend.split_scan(
-            hists, cfg.max_bin, cfg.lambda_l1, cfg.lambda_l2,
-            float(cfg.min_data_in_leaf), cfg.min_sum_hessian_in_leaf,
-            cfg.min_gain_to_split, self.nf, feat_mask).cpu()
+            self._to_float_hist(hists), cfg.max_bin, cfg.lambda_l1,
+            cfg.lambda_l2, float(cfg.min_data_in_leaf),
+            cfg.min_sum_hessian_in_leaf, cfg.min_gain_to_split, self.nf,
+            feat_mask).cpu()
         res = []
         for row in out.tolist():
             bg, bf, bb, gl, hl, cl = row
@@ -185,10 +226,9 @@ class TreeGrower:
             leaf_idx.append(-1)
             return len(feature) - 1
 
+        self.set_scales(grad, hess)
         root_hist = self._hist(rows_root, grad, hess)
-        G = float(root_hist[0, :, 0].sum())
-        H = float(root_hist[0, :, 1].sum())
-        C = float(root_hist[0, :, 2].sum())
+        G, H, C = self._sums(root_hist)
         root = _Leaf(new_node(), rows_root, root_hist, G, H, C, 0)
         (root.gain, root.feat, root.bin, root.GL, root.HL, root.CL) = \
             self._best_split(root_hist, feat_mask)
@@ -343,8 +383,13 @@ class TrainingSession:
             self.trees = []
             self.tree_w = []
         self.base = base
+        n_global = n
+        if comm.is_distributed:
+            t = torch.tensor([float(n)], device=device)
+            comm.all_reduce(t)
+            n_global = int(t[0])
         self.grower = TreeGrower(self.binned, self.nf, cfg, comm, self.stats,
-                                 self.bin_mapper)
+                                 self.bin_mapper, n_global=n_global)
         self.gen = torch.Generator(device="cpu")
         self.all_rows = torch.arange(n, dtype=torch.int32, device=device)
         self.n_start_trees = len(self.trees)

@@ -20,6 +20,9 @@ void launch_predict_leaf(const int*, const float*, const int*, const int*,
                          int*, int, hipStream_t);
 void launch_bin_matrix(const float*, const float*, long, int, int, int, void*,
                        hipStream_t);
+void launch_hist_build_fixed(const void*, long, const int*, long, const float*,
+                             const float*, long long*, int, int, double,
+                             double, hipStream_t);
 void launch_split_scan(const float*, int, long, int, float, float, float,
                        float, float, long, const bool*, float*, float*,
                        hipStream_t);
@@ -50,6 +53,24 @@ torch::Tensor hist_build(torch::Tensor binned_i4, torch::Tensor rows,
                     rows.numel(), grad.data_ptr<float>(),
                     hess.data_ptr<float>(), hist.data_ptr<float>(),
                     (int)n_bins, (int)ngroups, cur_stream());
+  return hist;
+}
+
+torch::Tensor hist_build_fixed(torch::Tensor binned_i4, torch::Tensor rows,
+                               torch::Tensor grad, torch::Tensor hess,
+                               long n_bins, double scale_g, double scale_h) {
+  CHECK_DEV(binned_i4); CHECK_CONTIG(binned_i4);
+  CHECK_DEV(rows); CHECK_CONTIG(rows);
+  TORCH_CHECK(rows.dtype() == torch::kInt32, "rows must be int32");
+  const long ngroups = binned_i4.size(0);
+  const long n_rows = binned_i4.size(1);
+  auto hist = torch::zeros({ngroups * 4, n_bins, 3},
+                           grad.options().dtype(torch::kInt64));
+  launch_hist_build_fixed(binned_i4.data_ptr(), n_rows, rows.data_ptr<int>(),
+                          rows.numel(), grad.data_ptr<float>(),
+                          hess.data_ptr<float>(),
+                          (long long*)hist.data_ptr<int64_t>(), (int)n_bins,
+                          (int)ngroups, scale_g, scale_h, cur_stream());
   return hist;
 }
 
@@ -157,6 +178,8 @@ torch::Tensor vw_predict(torch::Tensor idx, torch::Tensor val,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hist_build", &hist_build, "per-leaf (feature,bin) grad/hess/count histogram");
+  m.def("hist_build_fixed", &hist_build_fixed,
+        "fixed-point u64 histogram (fast LDS integer atomics)");
   m.def("predict_forest", &predict_forest, "GBDT ensemble raw scores");
   m.def("predict_leaf", &predict_leaf, "GBDT per-tree leaf indices");
   m.def("bin_matrix", &bin_matrix, "quantile binning to interleaved uint8");

@@ -142,3 +142,29 @@ def test_split_scan_matches_cpu():
         assert torch.allclose(ref[:, 0], out[:, 0], rtol=1e-3, atol=1e-3), (ref, out)
         assert torch.equal(ref[:, 1:3], out[:, 1:3]), (ref, out)
         assert torch.allclose(ref[:, 3:], out[:, 3:], rtol=1e-3, atol=1e-2)
+
+
+@requires_gpu
+def test_hist_build_fixed_matches_cpu():
+    from mmlspark_amd.ops import backend, cpu_ref
+    g = torch.Generator().manual_seed(5)
+    n, nf, nb = 200_000, 26, 255
+    ngroups = (nf + 3) // 4
+    binned = torch.randint(0, nb, (ngroups, n, 4), generator=g,
+                           dtype=torch.uint8)
+    rows = torch.randperm(n, generator=g)[: n // 2].to(torch.int32).sort().values
+    grad = torch.randn(n, generator=g)
+    hess = torch.rand(n, generator=g) + 0.01
+    ref = cpu_ref.hist_build(binned, rows, grad, hess, nb)
+    gmax = float(grad.abs().max())
+    hmax = float(hess.max())
+    sg = (2.0 ** 61) / (n * gmax)
+    sh = (2.0 ** 24) / hmax
+    out = backend.hist_build_fixed(binned.cuda(), rows.cuda(), grad.cuda(),
+                                   hess.cuda(), nb, sg, sh).cpu()
+    outf = out.double() * torch.tensor([1.0 / sg, 1.0 / sh, 1.0]).double()
+    assert torch.equal(outf[:, :, 2].float(), ref[:, :, 2])  # counts exact
+    assert torch.allclose(outf[:, :, 0].float(), ref[:, :, 0], atol=2e-3,
+                          rtol=1e-4)
+    assert torch.allclose(outf[:, :, 1].float(), ref[:, :, 1], atol=2e-3,
+                          rtol=1e-4)
