@@ -474,13 +474,18 @@ class TrainingSession:
                                             hess[:, k].contiguous(), feat_mask)
             tree.shrinkage = 1.0 if rf_mode else cfg.learning_rate
             if not rf_mode:
+                # one fused update: concat leaf row lists + repeat leaf values
                 t0 = time.perf_counter()
-                for lf in leaves:
-                    if lf.rows.numel():
-                        w_leaf = float(tree.value[lf.node_id]) * tree.shrinkage
-                        preds[:, k].index_add_(
-                            0, lf.rows.long(),
-                            torch.full((lf.rows.numel(),), w_leaf, device=device))
+                live = [lf for lf in leaves if lf.rows.numel()]
+                if live:
+                    all_rows = torch.cat([lf.rows for lf in live]).long()
+                    sizes = [lf.rows.numel() for lf in live]
+                    vals = torch.tensor(
+                        [float(tree.value[lf.node_id]) * tree.shrinkage
+                         for lf in live], device=device)
+                    expanded = torch.repeat_interleave(
+                        vals, torch.tensor(sizes, device=device))
+                    preds[:, k].index_add_(0, all_rows, expanded)
                 self.stats.partition_s += time.perf_counter() - t0
             new_trees.append(tree)
 

@@ -63,8 +63,11 @@ def hist_build_fixed(binned_i4, rows, grad, hess, n_bins, scale_g, scale_h):
 
 
 def partition_rows(binned_i4, rows, feature, threshold_bin):
-    # order-preserving masked select — torch's cub path is already optimal on
-    # ROCm, and keeping row lists sorted keeps later grad/hess gathers coalesced
+    if binned_i4.is_cuda:
+        # ordered 3-kernel partition, ONE device→host sync (torch's two
+        # masked_selects cost two nonzero() syncs per split)
+        return _require_ext().partition_rows(binned_i4, rows.contiguous(),
+                                             feature, threshold_bin)
     return cpu_ref.partition_rows(binned_i4, rows, feature, threshold_bin)
 
 

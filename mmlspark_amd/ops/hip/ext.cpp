@@ -23,6 +23,8 @@ void launch_bin_matrix(const float*, const float*, long, int, int, int, void*,
 void launch_hist_build_fixed(const void*, long, const int*, long, const float*,
                              const float*, long long*, int, int, double,
                              double, hipStream_t);
+void launch_partition(const void*, long, const int*, long, int, int, int*,
+                      int*, int*, hipStream_t);
 void launch_split_scan(const float*, int, long, int, float, float, float,
                        float, float, long, const bool*, float*, float*,
                        hipStream_t);
@@ -110,6 +112,25 @@ torch::Tensor predict_leaf(torch::Tensor feat, torch::Tensor thr,
   return out;
 }
 
+std::tuple<torch::Tensor, torch::Tensor> partition_rows(
+    torch::Tensor binned_i4, torch::Tensor rows, long feature, long thr) {
+  CHECK_DEV(binned_i4); CHECK_CONTIG(binned_i4);
+  CHECK_DEV(rows); CHECK_CONTIG(rows);
+  TORCH_CHECK(rows.dtype() == torch::kInt32, "rows must be int32");
+  const long m = rows.numel();
+  auto out = torch::empty({m}, rows.options());
+  auto scratch = torch::empty({4096}, rows.options());
+  auto total = torch::zeros({1}, rows.options());
+  if (m > 0) {
+    launch_partition(binned_i4.data_ptr(), binned_i4.size(1),
+                     rows.data_ptr<int>(), m, (int)feature, (int)thr,
+                     out.data_ptr<int>(), scratch.data_ptr<int>(),
+                     total.data_ptr<int>(), cur_stream());
+  }
+  const long nl = total.item<int>();  // the ONE sync per split
+  return {out.slice(0, 0, nl), out.slice(0, nl, m)};
+}
+
 torch::Tensor bin_matrix(torch::Tensor X, torch::Tensor ub, long n_bins) {
   CHECK_DEV(X); CHECK_CONTIG(X);
   CHECK_DEV(ub); CHECK_CONTIG(ub);
@@ -184,6 +205,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("predict_leaf", &predict_leaf, "GBDT per-tree leaf indices");
   m.def("bin_matrix", &bin_matrix, "quantile binning to interleaved uint8");
   m.def("split_scan", &split_scan, "fused best-split over sibling histograms");
+  m.def("partition_rows", &partition_rows,
+        "stable ordered row partition, single sync");
   m.def("vw_sgd_minibatch", &vw_sgd_minibatch, "adaptive sparse SGD minibatch");
   m.def("vw_predict", &vw_predict, "sparse linear predict");
 }

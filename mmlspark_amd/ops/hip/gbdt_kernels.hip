@@ -197,6 +197,136 @@ extern "C" void launch_hist_build(const void* binned, long n_rows,
 #undef LAUNCH_HB
 }
 
+// -------------------------------------------------------- ordered partition
+// Stable row partition (left = bin[feature] <= thr) in 3 async launches +
+// one count readback — replaces torch's two masked_selects whose internal
+// nonzero() forced TWO device→host syncs per split (profiled at ~13 ms/iter
+// host time).  Output preserves ascending row order on both sides, keeping
+// later grad/hess gathers coalesced.
+__global__ void part_count_k(const uchar4* __restrict__ binned, long n_rows,
+                             const int* __restrict__ rows, long m, int grp,
+                             int j, int thr, long chunk,
+                             int* __restrict__ block_counts) {
+  const long start = (long)blockIdx.x * chunk;
+  const long end = min(start + chunk, m);
+  int cnt = 0;
+  for (long i = start + threadIdx.x; i < end; i += blockDim.x) {
+    const uchar4 b4 = binned[(size_t)grp * n_rows + rows[i]];
+    const unsigned char b = j == 0 ? b4.x : j == 1 ? b4.y : j == 2 ? b4.z : b4.w;
+    cnt += (b <= thr);
+  }
+  __shared__ int sh[256];
+  sh[threadIdx.x] = cnt;
+  __syncthreads();
+  for (int d = 128; d > 0; d >>= 1) {
+    if (threadIdx.x < d) sh[threadIdx.x] += sh[threadIdx.x + d];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) block_counts[blockIdx.x] = sh[0];
+}
+
+__global__ void part_scan_k(int* __restrict__ block_counts, int n_blocks,
+                            int* __restrict__ total_left) {
+  // single block: exclusive scan of block_counts in place
+  __shared__ int carry;
+  __shared__ int sh[256];
+  if (threadIdx.x == 0) carry = 0;
+  __syncthreads();
+  for (int base = 0; base < n_blocks; base += blockDim.x) {
+    const int i = base + threadIdx.x;
+    const int v = (i < n_blocks) ? block_counts[i] : 0;
+    sh[threadIdx.x] = v;
+    __syncthreads();
+    for (int d = 1; d < 256; d <<= 1) {  // Hillis-Steele inclusive scan
+      const int add = threadIdx.x >= d ? sh[threadIdx.x - d] : 0;
+      __syncthreads();
+      sh[threadIdx.x] += add;
+      __syncthreads();
+    }
+    if (i < n_blocks) block_counts[i] = carry + sh[threadIdx.x] - v;
+    __syncthreads();
+    if (threadIdx.x == 0) carry += sh[255];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) *total_left = carry;
+}
+
+__global__ void part_scatter_k(const uchar4* __restrict__ binned, long n_rows,
+                               const int* __restrict__ rows, long m, int grp,
+                               int j, int thr, long chunk,
+                               const int* __restrict__ block_offsets,
+                               const int* __restrict__ total_left,
+                               int* __restrict__ out) {
+  const long start = (long)blockIdx.x * chunk;
+  const long end = min(start + chunk, m);
+  const long nl_total = *total_left;
+  __shared__ long base_l, base_r;
+  __shared__ int wave_l[4], wave_r[4];
+  if (threadIdx.x == 0) {
+    base_l = block_offsets[blockIdx.x];
+    base_r = nl_total + (start - block_offsets[blockIdx.x]);
+  }
+  __syncthreads();
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  for (long i0 = start; i0 < end; i0 += blockDim.x) {
+    const long i = i0 + threadIdx.x;
+    int r = 0;
+    bool valid = i < end, left = false;
+    if (valid) {
+      r = rows[i];
+      const uchar4 b4 = binned[(size_t)grp * n_rows + r];
+      const unsigned char b = j == 0 ? b4.x : j == 1 ? b4.y : j == 2 ? b4.z : b4.w;
+      left = (b <= thr);
+    }
+    const unsigned long long mask_l = __ballot(valid && left);
+    const unsigned long long mask_r = __ballot(valid && !left);
+    const unsigned long long lt = (1ull << lane) - 1ull;
+    if (threadIdx.x == (unsigned)(wid << 6)) {  // lane 0 of each wave
+      wave_l[wid] = __popcll(mask_l);
+      wave_r[wid] = __popcll(mask_r);
+    }
+    __syncthreads();
+    long wl = base_l, wr = base_r;
+    for (int w = 0; w < wid; ++w) {
+      wl += wave_l[w];
+      wr += wave_r[w];
+    }
+    if (valid) {
+      if (left) out[wl + __popcll(mask_l & lt)] = r;
+      else out[wr + __popcll(mask_r & lt)] = r;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      base_l += wave_l[0] + wave_l[1] + wave_l[2] + wave_l[3];
+      base_r += wave_r[0] + wave_r[1] + wave_r[2] + wave_r[3];
+    }
+    __syncthreads();
+  }
+}
+
+extern "C" void launch_partition(const void* binned, long n_rows,
+                                 const int* rows, long m, int feature,
+                                 int thr, int* out, int* scratch,
+                                 int* total_left, hipStream_t stream) {
+  if (m == 0) return;
+  long chunk = 4096;
+  long blocks = (m + chunk - 1) / chunk;
+  if (blocks > 4096) {
+    chunk = (m + 4095) / 4096;
+    blocks = (m + chunk - 1) / chunk;
+  }
+  const int grp = feature / 4, j = feature % 4;
+  hipLaunchKernelGGL(part_count_k, dim3((unsigned)blocks), dim3(256), 0,
+                     stream, (const uchar4*)binned, n_rows, rows, m, grp, j,
+                     thr, chunk, scratch);
+  hipLaunchKernelGGL(part_scan_k, dim3(1), dim3(256), 0, stream, scratch,
+                     (int)blocks, total_left);
+  hipLaunchKernelGGL(part_scatter_k, dim3((unsigned)blocks), dim3(256), 0,
+                     stream, (const uchar4*)binned, n_rows, rows, m, grp, j,
+                     thr, chunk, scratch, total_left, out);
+}
+
 // ------------------------------------------------------------ forest predict
 // Flat node arrays across trees; per-thread row traversal.
 __global__ void predict_forest_k(const int* __restrict__ feat,

@@ -168,3 +168,20 @@ def test_hist_build_fixed_matches_cpu():
                           rtol=1e-4)
     assert torch.allclose(outf[:, :, 1].float(), ref[:, :, 1], atol=2e-3,
                           rtol=1e-4)
+
+
+@requires_gpu
+def test_partition_rows_kernel_matches_cpu():
+    from mmlspark_amd.ops import backend, cpu_ref
+    g = torch.Generator().manual_seed(6)
+    n, nf, nb = 300_000, 12, 255
+    ngroups = (nf + 3) // 4
+    binned = torch.randint(0, nb, (ngroups, n, 4), generator=g,
+                           dtype=torch.uint8)
+    rows = torch.randperm(n, generator=g)[: n // 2].to(torch.int32).sort().values
+    for feature, thr in ((0, 100), (5, 3), (11, 254)):
+        l_ref, r_ref = cpu_ref.partition_rows(binned, rows, feature, thr)
+        l_gpu, r_gpu = backend.partition_rows(binned.cuda(), rows.cuda(),
+                                              feature, thr)
+        assert torch.equal(l_ref, l_gpu.cpu()), feature
+        assert torch.equal(r_ref, r_gpu.cpu()), feature
