@@ -1,0 +1,89 @@
+"""File IO codecs, PowerBI writer, codegen stub/doc generation."""
+import os
+
+import numpy as np
+import pandas as pd
+import pytest
+
+from mmlspark_amd.io_http.files import (BinaryFileReader, ImageReader,
+                                        PowerBIWriter, decode_image,
+                                        encode_image, read_binary_files,
+                                        write_binary_files)
+
+
+def test_binary_file_roundtrip(tmp_path):
+    for i in range(3):
+        with open(tmp_path / f"f{i}.bin", "wb") as f:
+            f.write(bytes([i] * 10))
+    df = read_binary_files(str(tmp_path / "*.bin"))
+    assert len(df) == 3
+    assert df["bytes"].iloc[1] == bytes([1] * 10)
+    out_dir = tmp_path / "out"
+    write_binary_files(df, str(out_dir))
+    assert sorted(os.listdir(out_dir)) == ["f0.bin", "f1.bin", "f2.bin"]
+    rdr = BinaryFileReader()
+    df2 = rdr.transform(pd.DataFrame({"path": [str(tmp_path / "f0.bin")]}))
+    assert df2["bytes"].iloc[0] == bytes([0] * 10)
+
+
+def test_image_codecs_roundtrip():
+    rng = np.random.default_rng(0)
+    img = rng.integers(0, 255, size=(9, 7, 3)).astype(np.uint8)
+    data = encode_image(img, "ppm")
+    back = decode_image(data)
+    np.testing.assert_array_equal(img, back)
+    gray = img[:, :, 0]
+    back_g = decode_image(encode_image(gray, "ppm"))
+    np.testing.assert_array_equal(gray, back_g[:, :, 0])
+    back_npy = decode_image(encode_image(img, "npy"))
+    np.testing.assert_array_equal(img, back_npy)
+    with pytest.raises(ValueError):
+        decode_image(b"\xff\xd8 fake jpeg")
+
+
+def test_image_reader_transformer():
+    img = np.zeros((4, 4, 3), dtype=np.uint8)
+    df = pd.DataFrame({"bytes": [encode_image(img, "ppm"), b"garbage!"]})
+    out = ImageReader(dropInvalid=True).transform(df)
+    assert len(out) == 1
+    assert out["image"].iloc[0].shape == (4, 4, 3)
+
+
+def test_powerbi_writer_against_mock():
+    from mmlspark_amd.serving.server import ServingServer
+    received = []
+
+    def handler(payloads):
+        received.extend(payloads)
+        return [{"ok": True} for _ in payloads]
+
+    srv = ServingServer(handler, port=0, mode="continuous").start()
+    try:
+        df = pd.DataFrame({"a": range(7), "b": ["x"] * 7})
+        w = PowerBIWriter(url=f"http://127.0.0.1:{srv.port}/", batchSize=3)
+        w.transform(df)
+        assert sum(len(p["rows"]) for p in received) == 7
+    finally:
+        srv.stop()
+
+
+def test_codegen_outputs(tmp_path):
+    from mmlspark_amd.core.codegen import (generate_docs, generate_r_wrappers,
+                                           generate_stubs)
+    n = generate_stubs(str(tmp_path / "stubs"))
+    assert n > 40
+    stub_files = os.listdir(tmp_path / "stubs")
+    assert any("gbdt" in f for f in stub_files)
+    content = open(tmp_path / "stubs" /
+                   [f for f in stub_files if "gbdt" in f][0]).read()
+    assert "class LightGBMClassifier" in content
+    assert "def setNumLeaves(self, value: int)" in content
+
+    nd = generate_docs(str(tmp_path / "docs"))
+    assert nd == n
+    assert os.path.exists(tmp_path / "docs" / "LightGBMClassifier.md")
+
+    nr = generate_r_wrappers(str(tmp_path / "r" / "bindings.R"))
+    assert nr == n
+    rcode = open(tmp_path / "r" / "bindings.R").read()
+    assert "ml_light_gbm_classifier" in rcode
