@@ -1,0 +1,91 @@
+#!/usr/bin/env python3
+"""ImageFeaturizer transfer-learning throughput (BASELINE config #4:
+ResNet-50 on CIFAR10-shaped 32x32x3, PyTorch-ROCm, DP via DDP/RCCL).
+One step = one fine-tune minibatch (forward+backward+optimizer);
+value = images/sec aggregate."""
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch", type=int, default=512)
+    ap.add_argument("--image-size", type=int, default=32)
+    args = ap.parse_args()
+
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    from mmlspark_amd.models.resnet import resnet50
+    from mmlspark_amd.parallel.comm import init_from_env
+
+    comm = init_from_env()
+    rank, world = comm.rank, comm.world_size
+    use_gpu = torch.cuda.is_available()
+    device = torch.device("cuda") if use_gpu else torch.device("cpu")
+    if use_gpu and world > 1:
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+        device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", rank)))
+    bs = args.batch if use_gpu else 16
+
+    torch.manual_seed(7 + rank)
+    net = resnet50(num_classes=10).to(device)
+    train_net = net
+    if comm.is_distributed:
+        train_net = torch.nn.parallel.DistributedDataParallel(
+            net, device_ids=[device.index] if use_gpu else None)
+    opt = torch.optim.SGD(train_net.parameters(), lr=0.1, momentum=0.9)
+    X = torch.randn(bs, 3, args.image_size, args.image_size, device=device)
+    y = torch.randint(0, 10, (bs,), device=device)
+
+    amp = torch.autocast(device_type="cuda", dtype=torch.bfloat16,
+                         enabled=use_gpu)
+
+    def step():
+        opt.zero_grad(set_to_none=True)
+        with amp:
+            loss = torch.nn.functional.cross_entropy(train_net(X), y)
+        loss.backward()
+        opt.step()
+
+    for _ in range(args.warmup):
+        step()
+    comm.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    if use_gpu:
+        torch.cuda.synchronize()
+    comm.barrier()
+    elapsed = time.perf_counter() - t0
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if comm.is_distributed:
+        comm.all_reduce(t.to(device) if use_gpu else t, op="max")
+    elapsed = float(t[0])
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "image_featurizer_images_per_sec",
+            "value": bs * world * args.steps / elapsed,
+            "unit": "images/s", "n_gpus": world, "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1e3,
+            "higher_is_better": True, "scaling": "weak",
+            "vs_baseline": None, "dtype": "bf16", "data": "synthetic",
+            "config": {"model": "ResNet50 fine-tune",
+                       "global_batch": bs * world,
+                       "image": f"{args.image_size}x{args.image_size}x3",
+                       "parallelism": f"dp{world} (DDP over RCCL)"},
+        }), flush=True)
+
+
+if __name__ == "__main__":
+    main()
