@@ -90,3 +90,102 @@ def test_distributed_gbdt_identical_models_and_quality():
     auc_d = roc_auc_score(y, p_dist)
     auc_s = roc_auc_score(y, p_single)
     assert abs(auc_d - auc_s) < 0.02, (auc_d, auc_s)
+
+
+def _worker_vw(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        torch.distributed.init_process_group(
+            "gloo", init_method=f"tcp://127.0.0.1:{port}",
+            rank=rank, world_size=world)
+        import pandas as pd
+        from mmlspark_amd.core.schema import SparseVector
+        from mmlspark_amd.models.vw.estimators import VowpalWabbitClassifier
+
+        rng = np.random.default_rng(rank)
+        size = 1 << 14
+        feat_ids = np.random.default_rng(0).integers(0, size, 100)
+        w_true = np.zeros(size)
+        w_true[feat_ids] = np.random.default_rng(1).normal(size=len(feat_ids))
+        rows, labels = [], []
+        for _ in range(1500):
+            idx = np.unique(rng.choice(feat_ids, size=20))
+            val = rng.normal(size=len(idx)).astype(np.float32)
+            rows.append(SparseVector(size, idx.astype(np.int32), val))
+            labels.append(1.0 if (w_true[idx] * val).sum() > 0 else 0.0)
+        df = pd.DataFrame({"features": rows, "label": labels})
+        m = VowpalWabbitClassifier(numPasses=3, numBits=14).fit(df)
+        q.put((rank, m.weights.tobytes()))
+        torch.distributed.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"ERROR: {e!r}"))
+
+
+@pytest.mark.timeout(180)
+def test_distributed_vw_weights_synchronized():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_vw, args=(r, 2, 29872, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, w = q.get(timeout=150)
+        results[rank] = w
+    for p in procs:
+        p.join(timeout=30)
+    assert not any(isinstance(w, str) and w.startswith("ERROR")
+                   for w in results.values()), results
+    # end-of-pass all_reduce must leave every rank with identical weights
+    w0 = np.frombuffer(results[0], dtype=np.float32)
+    w1 = np.frombuffer(results[1], dtype=np.float32)
+    np.testing.assert_allclose(w0, w1, rtol=1e-6, atol=1e-7)
+    assert np.abs(w0).max() > 0
+
+
+def _worker_ddp_vision(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        torch.distributed.init_process_group(
+            "gloo", init_method=f"tcp://127.0.0.1:{port}",
+            rank=rank, world_size=world)
+        import pandas as pd
+        from mmlspark_amd.models.image_featurizer import DeepVisionClassifier
+
+        rng = np.random.default_rng(rank)
+        imgs, ys = [], []
+        for i in range(16):
+            img = np.zeros((24, 24, 3), dtype=np.uint8)
+            c = i % 2
+            img[:, :, 0 if c == 0 else 2] = 200
+            imgs.append(img)
+            ys.append(c)
+        df = pd.DataFrame({"image": imgs, "label": ys})
+        torch.manual_seed(0)
+        m = DeepVisionClassifier(modelName="ResNet18", imageSize=24, epochs=2,
+                                 batchSize=8, device="cpu").fit(df)
+        sd = m.module.state_dict()
+        q.put((rank, float(sum(v.double().sum() for v in sd.values()))))
+        torch.distributed.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"ERROR: {e!r}"))
+
+
+@pytest.mark.timeout(240)
+def test_ddp_deep_vision_replicas_agree():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_ddp_vision, args=(r, 2, 29873, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, s = q.get(timeout=200)
+        results[rank] = s
+    for p in procs:
+        p.join(timeout=30)
+    assert not any(isinstance(s, str) for s in results.values()), results
+    # DDP gradient sync keeps replicas identical
+    assert abs(results[0] - results[1]) < 1e-3, results

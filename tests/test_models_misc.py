@@ -114,3 +114,32 @@ def test_ranking_evaluator():
     ev2 = RankingEvaluator(k=3, metricName="ndcgAt")
     v = ev2.evaluate(df)
     assert 0 < v < 1
+
+
+def test_ball_tree_matches_brute_force():
+    from mmlspark_amd.models.balltree import BallTree
+    rng = np.random.default_rng(7)
+    X = rng.normal(size=(500, 16)).astype(np.float32)
+    tree = BallTree(X, leaf_size=20)
+    for _ in range(10):
+        q = rng.normal(size=16).astype(np.float32)
+        got = tree.find_maximum_inner_products(q, k=5)
+        brute = np.argsort(X @ q)[::-1][:5]
+        assert [v for v, _ in got] == list(brute)
+
+
+def test_conditional_ball_tree():
+    from mmlspark_amd.models.balltree import ConditionalBallTree
+    rng = np.random.default_rng(8)
+    X = rng.normal(size=(400, 8)).astype(np.float32)
+    labels = [i % 3 for i in range(400)]
+    tree = ConditionalBallTree(X, labels, leaf_size=25)
+    q = rng.normal(size=8).astype(np.float32)
+    got = tree.find_maximum_inner_products(q, {1}, k=4)
+    # brute force restricted to label 1
+    mask = np.array(labels) == 1
+    cand = np.where(mask)[0]
+    brute = cand[np.argsort(X[cand] @ q)[::-1][:4]]
+    assert [v for v, _ in got] == list(brute)
+    for v, _ in got:
+        assert labels[v] == 1
