@@ -18,16 +18,21 @@ from ...parallel.comm import Comm
 
 
 class BinMapper:
-    """Per-feature quantile bin boundaries. upper_bounds[f, b] = largest value in bin b."""
+    """Per-feature quantile bin boundaries. upper_bounds[f, b] = largest value
+    in bin b.  Categorical features (categoricalSlotIndexes parity) are binned
+    by category id: bin = clamp(round(x), 0, n_bins-1)."""
 
-    def __init__(self, upper_bounds: torch.Tensor, n_bins: int):
+    def __init__(self, upper_bounds: torch.Tensor, n_bins: int,
+                 categorical: Optional[list] = None):
         self.upper_bounds = upper_bounds  # (nf, n_bins-1) float32, +inf padded
         self.n_bins = n_bins
         self.n_features = upper_bounds.shape[0]
+        self.categorical = sorted(set(categorical or []))
 
     @staticmethod
     def fit(X: torch.Tensor, n_bins: int = 255, sample_size: int = 200_000,
-            comm: Optional[Comm] = None, seed: int = 0) -> "BinMapper":
+            comm: Optional[Comm] = None, seed: int = 0,
+            categorical: Optional[list] = None) -> "BinMapper":
         n, nf = X.shape
         gen = torch.Generator(device="cpu").manual_seed(seed)
         k = min(n, max(1, sample_size // max(1, comm.world_size if comm else 1)))
@@ -46,11 +51,16 @@ class BinMapper:
         # strictly increasing boundaries; collapse duplicated quantiles
         ub = torch.cummax(ub, dim=1).values
         ub[:, -1] = float("inf")
-        return BinMapper(ub.contiguous(), n_bins)
+        return BinMapper(ub.contiguous(), n_bins, categorical)
 
     def transform(self, X: torch.Tensor) -> torch.Tensor:
         """(n, nf) float -> (ngroups, n, 4) uint8 feature-interleaved bins."""
-        return backend.bin_matrix(X, self.upper_bounds.to(X.device), self.n_bins)
+        out = backend.bin_matrix(X, self.upper_bounds.to(X.device), self.n_bins)
+        for f in self.categorical:  # category id IS the bin
+            bins = torch.nan_to_num(X[:, f], nan=0.0).round().clamp(
+                0, self.n_bins - 1).to(torch.uint8)
+            out[f // 4, :, f % 4] = bins
+        return out
 
     def bin_upper_value(self, feature: int, b: int) -> float:
         """Raw-value threshold for 'bin <= b' splits (used at predict time)."""
@@ -59,8 +69,10 @@ class BinMapper:
         return float(self.upper_bounds[feature, b])
 
     def state_dict(self):
-        return {"upper_bounds": self.upper_bounds.cpu(), "n_bins": self.n_bins}
+        return {"upper_bounds": self.upper_bounds.cpu(), "n_bins": self.n_bins,
+                "categorical": list(self.categorical)}
 
     @staticmethod
     def from_state(state):
-        return BinMapper(state["upper_bounds"], int(state["n_bins"]))
+        return BinMapper(state["upper_bounds"], int(state["n_bins"]),
+                         state.get("categorical"))

@@ -75,7 +75,8 @@ class Booster:
             f["feature"], f["threshold"], f["left"], f["right"], f["value"],
             f["offsets"], X.contiguous(), self.n_outputs, f["weights"],
             start_tree=start_iteration * self.n_outputs,
-            num_iteration=num_iteration)
+            num_iteration=num_iteration,
+            cat_offset=f.get("cat_offset"), cat_words=f.get("cat_words"))
         return raw + torch.from_numpy(self.base_score).to(X.device)
 
     def predict_prob(self, X: torch.Tensor) -> torch.Tensor:
@@ -92,7 +93,9 @@ class Booster:
         f = self._flat(X.device)
         return backend.predict_leaf(f["feature"], f["threshold"], f["left"],
                                     f["right"], f["leaf_index"], f["offsets"],
-                                    X.contiguous())
+                                    X.contiguous(),
+                                    cat_offset=f.get("cat_offset"),
+                                    cat_words=f.get("cat_words"))
 
     def predict_contrib(self, X: torch.Tensor) -> np.ndarray:
         """TreeSHAP contributions, (n, n_features+1) with expected value last."""
@@ -131,7 +134,8 @@ class Booster:
             "trees": [t.to_dict() for t in self.trees],
             "bin_mapper": (None if self.bin_mapper is None else {
                 "upper_bounds": self.bin_mapper.upper_bounds.cpu().numpy().tolist(),
-                "n_bins": self.bin_mapper.n_bins}),
+                "n_bins": self.bin_mapper.n_bins,
+                "categorical": list(self.bin_mapper.categorical)}),
         }
 
     def save_to_string(self) -> str:
@@ -144,7 +148,8 @@ class Booster:
         bm = None
         if d.get("bin_mapper"):
             ub = torch.tensor(d["bin_mapper"]["upper_bounds"], dtype=torch.float32)
-            bm = BinMapper(ub, int(d["bin_mapper"]["n_bins"]))
+            bm = BinMapper(ub, int(d["bin_mapper"]["n_bins"]),
+                           d["bin_mapper"].get("categorical"))
         return Booster(
             trees=[Tree.from_dict(t) for t in d["trees"]],
             objective=d["objective"], n_outputs=int(d["n_outputs"]),

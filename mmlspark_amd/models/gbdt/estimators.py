@@ -111,6 +111,7 @@ class _GBDTParams(Params):
             is_provide_training_metric=self.get("isProvideTrainingMetric"),
             metric=self.get("metric"),
             verbosity=self.get("verbosity"),
+            categorical_features=self.get("categoricalSlotIndexes"),
         )
 
     def _device(self):
@@ -168,11 +169,17 @@ class _GBDTEstimatorBase(_GBDTParams, Estimator):
             part = df if n_batches <= 1 else df.iloc[b]
             Xt, yt, w = self._extract(part, device)
             group = self._group_sizes(part, device)
+            init_score = None
+            isc = self.get("initScoreCol")
+            if isc and isc in part.columns:
+                init_score = torch.from_numpy(
+                    part[isc].to_numpy(dtype=np.float32)).to(device)
             booster, stats = train_booster(
                 Xt, yt, self._train_config(), self._make_objective(yt), comm,
                 weight=w, group_sizes=group, feature_names=names,
                 valid_sets=valid_sets, init_booster=booster,
-                metrics_fn=default_metrics_fn(self.get("metric") or None))
+                metrics_fn=default_metrics_fn(self.get("metric") or None),
+                init_score=init_score)
         model = self._model_class()(booster=booster)
         for p in ("labelCol", "featuresCol", "featureCols", "predictionCol"):
             model.set(p, self.get(p))

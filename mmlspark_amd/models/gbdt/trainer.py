@@ -70,6 +70,7 @@ class TrainConfig:
     verbosity: int = -1
     is_provide_training_metric: bool = False
     metric: str = ""
+    categorical_features: Optional[List[int]] = None
 
 
 @dataclass
@@ -94,7 +95,7 @@ class TrainingStats:
 
 class _Leaf:
     __slots__ = ("node_id", "rows", "hist", "G", "H", "C", "depth",
-                 "gain", "feat", "bin", "GL", "HL", "CL")
+                 "gain", "feat", "bin", "GL", "HL", "CL", "cats")
 
     def __init__(self, node_id, rows, hist, G, H, C, depth):
         self.node_id = node_id
@@ -103,6 +104,7 @@ class _Leaf:
         self.G, self.H, self.C = G, H, C
         self.depth = depth
         self.gain = NEG_INF
+        self.cats = None
 
     def __lt__(self, other):  # max-heap via negated gain at push site
         return False
@@ -138,6 +140,8 @@ class TreeGrower:
         self.n_global = n_global or binned_i4.shape[1]
         self.scale_g = 1.0
         self.scale_h = 1.0
+        self.cat_features = list(getattr(bin_mapper, "categorical", []) or [])
+        self.cat_smooth = 10.0
 
     _sync_timers = bool(__import__("os").environ.get("MMLSPARK_AMD_SYNC_TIMERS"))
 
@@ -189,20 +193,77 @@ class TreeGrower:
 
     def _scan(self, hists, feat_mask):
         """Fused split scan on stacked histograms; identical on all ranks
-        (input is the reduced histogram). One kernel pair + one readback."""
+        (input is the reduced histogram). One kernel pair + one readback.
+        Categorical features use a torch-side sorted one-vs-rest scan
+        (LightGBM categorical semantics) and compete with the numeric best."""
         t0 = time.perf_counter()
         cfg = self.cfg
+        histsf = self._to_float_hist(hists)
+        num_mask = feat_mask
+        if self.cat_features:
+            num_mask = (feat_mask.clone() if feat_mask is not None else
+                        torch.ones(self.nf_pad, dtype=torch.bool,
+                                   device=hists.device))
+            num_mask[self.cat_features] = False
         out = backend.split_scan(
-            self._to_float_hist(hists), cfg.max_bin, cfg.lambda_l1,
+            histsf, cfg.max_bin, cfg.lambda_l1,
             cfg.lambda_l2, float(cfg.min_data_in_leaf),
             cfg.min_sum_hessian_in_leaf, cfg.min_gain_to_split, self.nf,
-            feat_mask).cpu()
+            num_mask).cpu()
         res = []
-        for row in out.tolist():
+        for hi, row in enumerate(out.tolist()):
             bg, bf, bb, gl, hl, cl = row
-            res.append((bg, int(bf), int(bb), gl, hl, cl))
+            best = [bg, int(bf), int(bb), gl, hl, cl, None]
+            if self.cat_features:
+                cat = self._cat_scan(histsf[hi], feat_mask)
+                if cat is not None and cat[0] > best[0]:
+                    best = cat
+            res.append(tuple(best))
         self.stats.split_s += time.perf_counter() - t0
         return res
+
+    def _cat_scan(self, histf, feat_mask):
+        """Sorted one-vs-rest categorical split over present categories."""
+        cfg = self.cfg
+        best = None
+        hcpu = histf.cpu()
+        for f in self.cat_features:
+            if feat_mask is not None and not bool(feat_mask[f]):
+                continue
+            h = hcpu[f]  # (nb, 3): g, h, count
+            present = torch.nonzero(h[:, 2] > 0).squeeze(-1)
+            if present.numel() < 2:
+                continue
+            g = h[present, 0]
+            hh = h[present, 1]
+            c = h[present, 2]
+            order = torch.argsort(g / (hh + self.cat_smooth))
+            GL = g[order].cumsum(0)
+            HL = hh[order].cumsum(0)
+            CL = c[order].cumsum(0)
+            G, H, C = float(GL[-1]), float(HL[-1]), float(CL[-1])
+
+            def sc(Gs, Hs):
+                Ga = (Gs.abs() - cfg.lambda_l1).clamp_min(0)
+                return Ga * Ga / (Hs + cfg.lambda_l2 + 1e-32)
+
+            gain = sc(GL, HL) + sc(G - GL, H - HL) - sc(
+                torch.tensor(G), torch.tensor(H))
+            valid = ((CL >= cfg.min_data_in_leaf)
+                     & (C - CL >= cfg.min_data_in_leaf)
+                     & (HL >= cfg.min_sum_hessian_in_leaf)
+                     & (H - HL >= cfg.min_sum_hessian_in_leaf))
+            gain = torch.where(valid, gain, torch.full_like(gain, NEG_INF))
+            gain[-1] = NEG_INF
+            k = int(gain.argmax())
+            bg = float(gain[k])
+            if not np.isfinite(bg):
+                continue
+            if best is None or bg > best[0]:
+                cats = present[order[: k + 1]].tolist()
+                best = [bg, int(f), 0, float(GL[k]), float(HL[k]),
+                        float(CL[k]), cats]
+        return best
 
     def _best_split(self, hist, feat_mask):
         return self._scan(hist.unsqueeze(0), feat_mask)[0]
@@ -213,6 +274,7 @@ class TreeGrower:
         # node arrays (grown dynamically)
         feature, threshold, thr_bin = [], [], []
         left, right, value, count, gain_arr, leaf_idx = [], [], [], [], [], []
+        cat_off, cat_words = [], []
 
         def new_node():
             feature.append(-1)
@@ -224,14 +286,15 @@ class TreeGrower:
             count.append(0.0)
             gain_arr.append(0.0)
             leaf_idx.append(-1)
+            cat_off.append(-1)
             return len(feature) - 1
 
         self.set_scales(grad, hess)
         root_hist = self._hist(rows_root, grad, hess)
         G, H, C = self._sums(root_hist)
         root = _Leaf(new_node(), rows_root, root_hist, G, H, C, 0)
-        (root.gain, root.feat, root.bin, root.GL, root.HL, root.CL) = \
-            self._best_split(root_hist, feat_mask)
+        (root.gain, root.feat, root.bin, root.GL, root.HL, root.CL,
+         root.cats) = self._best_split(root_hist, feat_mask)
         count[root.node_id] = C
         value[root.node_id] = _leaf_output(G, H, cfg)
 
@@ -249,8 +312,16 @@ class TreeGrower:
             final_leaves.remove(leaf)
 
             t0 = time.perf_counter()
-            rows_l, rows_r = backend.partition_rows(
-                self.binned, leaf.rows, leaf.feat, leaf.bin)
+            if leaf.cats is not None:
+                gq, j = leaf.feat // 4, leaf.feat % 4
+                bins = self.binned[gq, leaf.rows.long(), j]
+                cats_t = torch.tensor(leaf.cats, dtype=bins.dtype,
+                                      device=bins.device)
+                mask = torch.isin(bins, cats_t)
+                rows_l, rows_r = leaf.rows[mask], leaf.rows[~mask]
+            else:
+                rows_l, rows_r = backend.partition_rows(
+                    self.binned, leaf.rows, leaf.feat, leaf.bin)
             self.stats.partition_s += time.perf_counter() - t0
 
             GL, HL, CL = leaf.GL, leaf.HL, leaf.CL
@@ -266,7 +337,16 @@ class TreeGrower:
             nid = leaf.node_id
             feature[nid] = leaf.feat
             thr_bin[nid] = leaf.bin
-            threshold[nid] = self.bin_mapper.bin_upper_value(leaf.feat, leaf.bin)
+            if leaf.cats is not None:
+                words = [0] * 8
+                for b in leaf.cats:
+                    words[b >> 5] |= 1 << (b & 31)
+                cat_off[nid] = len(cat_words) // 8
+                cat_words.extend(words)
+                threshold[nid] = float("nan")  # categorical marker for dumps
+            else:
+                threshold[nid] = self.bin_mapper.bin_upper_value(leaf.feat,
+                                                                 leaf.bin)
             gain_arr[nid] = -ngain
             lid, rid = new_node(), new_node()
             left[nid], right[nid] = lid, rid
@@ -280,7 +360,7 @@ class TreeGrower:
             # both children in ONE fused scan + readback
             pair = self._scan(torch.stack([hist_l, hist_r]), feat_mask)
             for ch, res in zip((lc, rc), pair):
-                (ch.gain, ch.feat, ch.bin, ch.GL, ch.HL, ch.CL) = res
+                (ch.gain, ch.feat, ch.bin, ch.GL, ch.HL, ch.CL, ch.cats) = res
                 heapq.heappush(heap, (-ch.gain, seq, ch))
                 seq += 1
                 final_leaves.append(ch)
@@ -292,7 +372,8 @@ class TreeGrower:
             lf.hist = None
 
         tree = Tree(feature, threshold, thr_bin, left, right, value, count,
-                    gain_arr, leaf_idx, shrinkage=1.0)
+                    gain_arr, leaf_idx, shrinkage=1.0,
+                    cat_offset=cat_off, cat_words=cat_words)
         return tree, final_leaves
 
 
@@ -305,6 +386,9 @@ def predict_tree_binned(tree: Tree, binned_i4: torch.Tensor,
     lft = torch.from_numpy(tree.left).to(device).long()
     rgt = torch.from_numpy(tree.right).to(device).long()
     val = torch.from_numpy(tree.value).to(device)
+    catoff = torch.from_numpy(tree.cat_offset).to(device).long()
+    catw = (torch.from_numpy(tree.cat_words.view(np.int32).copy()).to(device)
+            if len(tree.cat_words) else None)
     idx = torch.zeros(n, dtype=torch.long, device=device)
     active = feat[idx] >= 0
     flat = binned_i4.permute(0, 2, 1).reshape(-1, n)  # (nf_pad, n) view by feature
@@ -312,6 +396,16 @@ def predict_tree_binned(tree: Tree, binned_i4: torch.Tensor,
         f = feat[idx].clamp(min=0)
         bins = flat[f, torch.arange(n, device=device)]
         go_left = bins <= thrb[idx]
+        if catw is not None:
+            off = catoff[idx]
+            is_cat = off >= 0
+            if bool(is_cat.any()):
+                b = bins.long()
+                widx = (off.clamp(min=0) * 8 + (b >> 5)).clamp(
+                    max=catw.numel() - 1)
+                w32 = catw[widx].long() & 0xFFFFFFFF
+                bits = ((w32 >> (b & 31)) & 1).bool()
+                go_left = torch.where(is_cat, bits, go_left)
         nxt = torch.where(go_left, lft[idx], rgt[idx])
         idx = torch.where(active, nxt, idx)
         active = feat[idx] >= 0
@@ -351,7 +445,7 @@ class TrainingSession:
                  group_sizes: Optional[torch.Tensor] = None,
                  feature_names: Optional[List[str]] = None,
                  init_booster: Optional[Booster] = None,
-                 binned_cache=None):
+                 binned_cache=None, init_score: Optional[torch.Tensor] = None):
         self.X, self.y, self.cfg, self.objective = X, y, cfg, objective
         self.comm, self.weight = comm, weight
         self.feature_names = feature_names
@@ -366,7 +460,8 @@ class TrainingSession:
             self.bin_mapper, self.binned = binned_cache
         else:
             self.bin_mapper = BinMapper.fit(X, n_bins=cfg.max_bin, comm=comm,
-                                            seed=cfg.seed)
+                                            seed=cfg.seed,
+                                            categorical=cfg.categorical_features)
             self.binned = self.bin_mapper.transform(X)
 
         device, n, K = self.device, self.n, self.K
@@ -380,6 +475,9 @@ class TrainingSession:
             if base.numel() < K:
                 base = base.expand(K).contiguous()
             self.preds = base.unsqueeze(0).expand(n, K).clone()
+            if init_score is not None:  # initScoreCol warm-start margins
+                sc = init_score.to(device).float()
+                self.preds += sc.unsqueeze(-1) if sc.dim() == 1 else sc
             self.trees = []
             self.tree_w = []
         self.base = base
@@ -530,14 +628,15 @@ def train_booster(X: torch.Tensor, y: torch.Tensor, cfg: TrainConfig,
                   valid_sets: Optional[List[tuple]] = None,
                   init_booster: Optional[Booster] = None,
                   metrics_fn=None,
-                  binned_cache=None) -> (Booster, TrainingStats):
+                  binned_cache=None, init_score=None) -> (Booster, TrainingStats):
     """Full training loop with eval + early stopping over TrainingSession."""
     t_start = time.perf_counter()
     session = TrainingSession(X, y, cfg, objective, comm, weight=weight,
                               group_sizes=group_sizes,
                               feature_names=feature_names,
                               init_booster=init_booster,
-                              binned_cache=binned_cache)
+                              binned_cache=binned_cache,
+                              init_score=init_score)
     stats = session.stats
     best_score = None
     best_iter = -1

@@ -20,7 +20,8 @@ import numpy as np
 
 class Tree:
     def __init__(self, feature, threshold, thr_bin, left, right, value, count,
-                 gain, leaf_index, shrinkage: float = 1.0):
+                 gain, leaf_index, shrinkage: float = 1.0,
+                 cat_offset=None, cat_words=None):
         self.feature = np.asarray(feature, dtype=np.int32)
         self.threshold = np.asarray(threshold, dtype=np.float32)
         self.thr_bin = np.asarray(thr_bin, dtype=np.int32)
@@ -31,6 +32,23 @@ class Tree:
         self.gain = np.asarray(gain, dtype=np.float32)
         self.leaf_index = np.asarray(leaf_index, dtype=np.int32)
         self.shrinkage = float(shrinkage)
+        # categorical splits: cat_offset[node] >= 0 indexes an 8-word (256-bit)
+        # category bitset in cat_words; bit b set → category b goes LEFT
+        self.cat_offset = (np.full(len(self.feature), -1, dtype=np.int32)
+                           if cat_offset is None
+                           else np.asarray(cat_offset, dtype=np.int32))
+        self.cat_words = (np.zeros(0, dtype=np.uint32) if cat_words is None
+                          else np.asarray(cat_words, dtype=np.uint32))
+
+    def is_categorical_node(self, i: int) -> bool:
+        return self.cat_offset[i] >= 0
+
+    def cat_goes_left(self, i: int, category: float) -> bool:
+        b = int(category)
+        if not (0 <= b < 256) or np.isnan(category):
+            return True  # missing/out-of-range → left
+        w = self.cat_words[self.cat_offset[i] * 8 + (b >> 5)]
+        return bool((int(w) >> (b & 31)) & 1)
 
     @property
     def n_nodes(self) -> int:
@@ -52,13 +70,16 @@ class Tree:
             "gain": [float(x) for x in self.gain],
             "leaf_index": self.leaf_index.tolist(),
             "shrinkage": self.shrinkage,
+            "cat_offset": self.cat_offset.tolist(),
+            "cat_words": [int(w) for w in self.cat_words],
         }
 
     @staticmethod
     def from_dict(d: Dict) -> "Tree":
         return Tree(d["feature"], d["threshold"], d["thr_bin"], d["left"],
                     d["right"], d["value"], d["count"], d["gain"],
-                    d["leaf_index"], d.get("shrinkage", 1.0))
+                    d["leaf_index"], d.get("shrinkage", 1.0),
+                    d.get("cat_offset"), d.get("cat_words"))
 
     # ----------------------------------------------------------------- treeSHAP
     def shap_values(self, X: np.ndarray, scale: float = 1.0) -> np.ndarray:
@@ -157,8 +178,12 @@ class Tree:
                 return
             f = feature[node]
             x = row[f]
-            hot, cold = (left[node], right[node]) if (x <= threshold[node]
-                                                      or np.isnan(x)) else (right[node], left[node])
+            if self.cat_offset[node] >= 0:
+                goes_left = self.cat_goes_left(node, x)
+            else:
+                goes_left = x <= threshold[node] or np.isnan(x)
+            hot, cold = (left[node], right[node]) if goes_left \
+                else (right[node], left[node])
             iz, io = 1.0, 1.0
             k = -1
             for i in range(1, length):
@@ -187,12 +212,23 @@ def flatten_trees(trees: List[Tree]):
         return dict(feature=z, threshold=z.astype(np.float32), left=z, right=z,
                     value=z.astype(np.float32), leaf_index=z,
                     offsets=np.zeros(1, dtype=np.int64),
-                    weights=np.zeros(0, dtype=np.float32))
+                    weights=np.zeros(0, dtype=np.float32),
+                    cat_offset=z, cat_words=np.zeros(0, dtype=np.int32))
     offsets = np.zeros(len(trees) + 1, dtype=np.int64)
     for i, t in enumerate(trees):
         offsets[i + 1] = offsets[i] + t.n_nodes
     cat = lambda attr, dt: np.concatenate([getattr(t, attr).astype(dt) for t in trees])
     value = np.concatenate([(t.value * t.shrinkage).astype(np.float32) for t in trees])
+    # concatenate categorical bitsets, rebasing per-tree offsets
+    cat_off_parts = []
+    cat_word_parts = []
+    word_base = 0
+    for t in trees:
+        off = t.cat_offset.astype(np.int64).copy()
+        off[off >= 0] += word_base // 8
+        cat_off_parts.append(off.astype(np.int32))
+        cat_word_parts.append(t.cat_words.astype(np.uint32))
+        word_base += len(t.cat_words)
     return dict(
         feature=cat("feature", np.int32),
         threshold=cat("threshold", np.float32),
@@ -202,4 +238,7 @@ def flatten_trees(trees: List[Tree]):
         leaf_index=cat("leaf_index", np.int32),
         offsets=offsets,
         weights=np.ones(len(trees), dtype=np.float32),
+        cat_offset=np.concatenate(cat_off_parts),
+        cat_words=np.concatenate(cat_word_parts).view(np.int32)
+        if word_base else np.zeros(0, dtype=np.int32),
     )

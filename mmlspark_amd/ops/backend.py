@@ -73,7 +73,8 @@ def partition_rows(binned_i4, rows, feature, threshold_bin):
 
 def predict_forest(node_feature, node_threshold, node_left, node_right,
                    node_value, tree_offsets, X, n_outputs,
-                   tree_weights=None, start_tree=0, num_iteration=-1):
+                   tree_weights=None, start_tree=0, num_iteration=-1,
+                   cat_offset=None, cat_words=None):
     if X.is_cuda:
         n_trees = tree_offsets.numel() - 1
         end_tree = n_trees if num_iteration < 0 else min(
@@ -82,20 +83,24 @@ def predict_forest(node_feature, node_threshold, node_left, node_right,
             tree_weights = torch.ones(n_trees, dtype=torch.float32, device=X.device)
         return _require_ext().predict_forest(
             node_feature, node_threshold, node_left, node_right, node_value,
-            tree_offsets, tree_weights, X, n_outputs, start_tree, end_tree)
+            tree_offsets, tree_weights, X, n_outputs, start_tree, end_tree,
+            cat_offset, cat_words)
     return cpu_ref.predict_forest(node_feature, node_threshold, node_left,
                                   node_right, node_value, tree_offsets, X,
-                                  n_outputs, tree_weights, start_tree, num_iteration)
+                                  n_outputs, tree_weights, start_tree,
+                                  num_iteration, cat_offset, cat_words)
 
 
 def predict_leaf(node_feature, node_threshold, node_left, node_right,
-                 node_leaf_index, tree_offsets, X):
+                 node_leaf_index, tree_offsets, X,
+                 cat_offset=None, cat_words=None):
     if X.is_cuda:
         return _require_ext().predict_leaf(
             node_feature, node_threshold, node_left, node_right,
-            node_leaf_index, tree_offsets, X)
+            node_leaf_index, tree_offsets, X, cat_offset, cat_words)
     return cpu_ref.predict_leaf(node_feature, node_threshold, node_left,
-                                node_right, node_leaf_index, tree_offsets, X)
+                                node_right, node_leaf_index, tree_offsets, X,
+                                cat_offset, cat_words)
 
 
 def split_scan(hists, n_bins, l1, l2, min_data, min_hess, min_gain, nf_real,

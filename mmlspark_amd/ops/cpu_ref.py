@@ -41,12 +41,33 @@ def partition_rows(binned_i4: torch.Tensor, rows: torch.Tensor, feature: int,
     return rows[mask], rows[~mask]
 
 
+def _go_left(xv, idx, node_threshold, cat_offset, cat_words):
+    """Vectorized branch decision incl. categorical bitset nodes."""
+    gl = (xv <= node_threshold[idx]) | torch.isnan(xv)
+    if cat_offset is not None and cat_words is not None and cat_words.numel():
+        off = cat_offset[idx].long()
+        is_cat = off >= 0
+        if bool(is_cat.any()):
+            b = torch.nan_to_num(xv, nan=0.0).round().long().clamp(0, 255)
+            widx = (off.clamp(min=0) * 8 + (b >> 5)).clamp(
+                max=cat_words.numel() - 1)
+            w32 = cat_words[widx].long() & 0xFFFFFFFF
+            bits = (w32 >> (b & 31)) & 1
+            in_range = (xv >= 0) & (xv < 256) & ~torch.isnan(xv)
+            cat_left = torch.where(in_range, bits.bool(),
+                                   torch.ones_like(bits, dtype=torch.bool))
+            gl = torch.where(is_cat, cat_left, gl)
+    return gl
+
+
 def predict_forest(node_feature: torch.Tensor, node_threshold: torch.Tensor,
                    node_left: torch.Tensor, node_right: torch.Tensor,
                    node_value: torch.Tensor, tree_offsets: torch.Tensor,
                    X: torch.Tensor, n_outputs: int,
                    tree_weights: torch.Tensor = None,
-                   start_tree: int = 0, num_iteration: int = -1) -> torch.Tensor:
+                   start_tree: int = 0, num_iteration: int = -1,
+                   cat_offset: torch.Tensor = None,
+                   cat_words: torch.Tensor = None) -> torch.Tensor:
     """Sum of per-tree leaf values over the ensemble. Vectorized traversal.
 
     Flattened node arrays (concatenated trees); interior node: feature >= 0,
@@ -67,7 +88,7 @@ def predict_forest(node_feature: torch.Tensor, node_threshold: torch.Tensor,
         while bool(active.any()):
             f = node_feature[idx].clamp(min=0).long()
             xv = X[torch.arange(n, device=X.device), f]
-            go_left = (xv <= node_threshold[idx]) | torch.isnan(xv)
+            go_left = _go_left(xv, idx, node_threshold, cat_offset, cat_words)
             nxt = torch.where(go_left, node_left[idx], node_right[idx]).long()
             idx = torch.where(active, nxt + base, idx)
             active = node_feature[idx] >= 0
@@ -77,7 +98,8 @@ def predict_forest(node_feature: torch.Tensor, node_threshold: torch.Tensor,
 
 
 def predict_leaf(node_feature, node_threshold, node_left, node_right,
-                 node_leaf_index, tree_offsets, X) -> torch.Tensor:
+                 node_leaf_index, tree_offsets, X,
+                 cat_offset=None, cat_words=None) -> torch.Tensor:
     """Per-tree leaf index for each row: (n, n_trees) int32."""
     n = X.shape[0]
     n_trees = tree_offsets.numel() - 1
@@ -89,7 +111,7 @@ def predict_leaf(node_feature, node_threshold, node_left, node_right,
         while bool(active.any()):
             f = node_feature[idx].clamp(min=0).long()
             xv = X[torch.arange(n, device=X.device), f]
-            go_left = (xv <= node_threshold[idx]) | torch.isnan(xv)
+            go_left = _go_left(xv, idx, node_threshold, cat_offset, cat_words)
             nxt = torch.where(go_left, node_left[idx], node_right[idx]).long()
             idx = torch.where(active, nxt + base, idx)
             active = node_feature[idx] >= 0

@@ -14,10 +14,10 @@ void launch_hist_build(const void*, long, const int*, long, const float*,
 void launch_predict_forest(const int*, const float*, const int*, const int*,
                            const float*, const long*, const float*,
                            const float*, long, int, float*, int, int, int,
-                           hipStream_t);
+                           const int*, const unsigned*, hipStream_t);
 void launch_predict_leaf(const int*, const float*, const int*, const int*,
                          const int*, const long*, const float*, long, int,
-                         int*, int, hipStream_t);
+                         int*, int, const int*, const unsigned*, hipStream_t);
 void launch_bin_matrix(const float*, const float*, long, int, int, int, void*,
                        hipStream_t);
 void launch_hist_build_fixed(const void*, long, const int*, long, const float*,
@@ -76,39 +76,58 @@ torch::Tensor hist_build_fixed(torch::Tensor binned_i4, torch::Tensor rows,
   return hist;
 }
 
+static std::pair<const int*, const unsigned*> cat_ptrs(
+    const c10::optional<torch::Tensor>& cat_offset,
+    const c10::optional<torch::Tensor>& cat_words) {
+  const int* co = nullptr;
+  const unsigned* cw = nullptr;
+  if (cat_offset.has_value() && cat_words.has_value()
+      && cat_words->numel() > 0) {
+    co = cat_offset->data_ptr<int>();
+    cw = (const unsigned*)cat_words->data_ptr<int>();
+  }
+  return {co, cw};
+}
+
 torch::Tensor predict_forest(torch::Tensor feat, torch::Tensor thr,
                              torch::Tensor left, torch::Tensor right,
                              torch::Tensor val, torch::Tensor offsets,
                              torch::Tensor tw, torch::Tensor X,
-                             long n_outputs, long t0, long t1) {
+                             long n_outputs, long t0, long t1,
+                             c10::optional<torch::Tensor> cat_offset,
+                             c10::optional<torch::Tensor> cat_words) {
   CHECK_DEV(X); CHECK_CONTIG(X);
   const long n = X.size(0);
   const long nf = X.size(1);
   auto out = torch::zeros({n, n_outputs}, X.options().dtype(torch::kFloat32));
   TORCH_CHECK(offsets.dtype() == torch::kInt64, "offsets must be int64");
+  auto [co, cw] = cat_ptrs(cat_offset, cat_words);
   launch_predict_forest(feat.data_ptr<int>(), thr.data_ptr<float>(),
                         left.data_ptr<int>(), right.data_ptr<int>(),
                         val.data_ptr<float>(), offsets.data_ptr<long>(),
                         tw.data_ptr<float>(), X.data_ptr<float>(), n, (int)nf,
                         out.data_ptr<float>(), (int)n_outputs, (int)t0,
-                        (int)t1, cur_stream());
+                        (int)t1, co, cw, cur_stream());
   return out;
 }
 
 torch::Tensor predict_leaf(torch::Tensor feat, torch::Tensor thr,
                            torch::Tensor left, torch::Tensor right,
                            torch::Tensor leaf_index, torch::Tensor offsets,
-                           torch::Tensor X) {
+                           torch::Tensor X,
+                           c10::optional<torch::Tensor> cat_offset,
+                           c10::optional<torch::Tensor> cat_words) {
   CHECK_DEV(X); CHECK_CONTIG(X);
   const long n = X.size(0);
   const long nf = X.size(1);
   const long n_trees = offsets.numel() - 1;
   auto out = torch::zeros({n, n_trees}, X.options().dtype(torch::kInt32));
+  auto [co, cw] = cat_ptrs(cat_offset, cat_words);
   launch_predict_leaf(feat.data_ptr<int>(), thr.data_ptr<float>(),
                       left.data_ptr<int>(), right.data_ptr<int>(),
                       leaf_index.data_ptr<int>(), offsets.data_ptr<long>(),
                       X.data_ptr<float>(), n, (int)nf, out.data_ptr<int>(),
-                      (int)n_trees, cur_stream());
+                      (int)n_trees, co, cw, cur_stream());
   return out;
 }
 

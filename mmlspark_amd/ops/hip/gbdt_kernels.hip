@@ -328,7 +328,23 @@ extern "C" void launch_partition(const void* binned, long n_rows,
 }
 
 // ------------------------------------------------------------ forest predict
-// Flat node arrays across trees; per-thread row traversal.
+// Flat node arrays across trees; per-thread row traversal.  Categorical
+// nodes (cat_offset[idx] >= 0) test a 256-bit category bitset instead of the
+// numeric threshold (LightGBM categorical-split semantics).
+DEV_INLINE bool go_left_node(float xv, long idx, const float* __restrict__ thr,
+                             const int* __restrict__ catoff,
+                             const unsigned* __restrict__ catw) {
+  if (catoff != nullptr) {
+    const int off = catoff[idx];
+    if (off >= 0) {
+      if (isnan(xv) || xv < 0.0f || xv >= 256.0f) return true;
+      const int b = (int)rintf(xv);
+      return (catw[off * 8 + (b >> 5)] >> (b & 31)) & 1u;
+    }
+  }
+  return xv <= thr[idx] || isnan(xv);
+}
+
 __global__ void predict_forest_k(const int* __restrict__ feat,
                                  const float* __restrict__ thr,
                                  const int* __restrict__ left,
@@ -338,7 +354,9 @@ __global__ void predict_forest_k(const int* __restrict__ feat,
                                  const float* __restrict__ tw,
                                  const float* __restrict__ X, long n, int nf,
                                  float* __restrict__ out, int n_outputs,
-                                 int t0, int t1) {
+                                 int t0, int t1,
+                                 const int* __restrict__ catoff,
+                                 const unsigned* __restrict__ catw) {
   const long row0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
   for (long row = row0; row < n; row += stride) {
@@ -351,7 +369,8 @@ __global__ void predict_forest_k(const int* __restrict__ feat,
         int f = feat[idx];
         while (f >= 0) {
           const float xv = x[f];
-          idx = base + ((xv <= thr[idx] || isnan(xv)) ? left[idx] : right[idx]);
+          idx = base + (go_left_node(xv, idx, thr, catoff, catw) ? left[idx]
+                                                                 : right[idx]);
           f = feat[idx];
         }
         acc += tw[t] * val[idx];
@@ -364,7 +383,8 @@ __global__ void predict_forest_k(const int* __restrict__ feat,
         int f = feat[idx];
         while (f >= 0) {
           const float xv = x[f];
-          idx = base + ((xv <= thr[idx] || isnan(xv)) ? left[idx] : right[idx]);
+          idx = base + (go_left_node(xv, idx, thr, catoff, catw) ? left[idx]
+                                                                 : right[idx]);
           f = feat[idx];
         }
         out[row * n_outputs + (t % n_outputs)] += tw[t] * val[idx];
@@ -378,13 +398,15 @@ extern "C" void launch_predict_forest(const int* feat, const float* thr,
                                       const float* val, const long* offsets,
                                       const float* tw, const float* X, long n,
                                       int nf, float* out, int n_outputs,
-                                      int t0, int t1, hipStream_t stream) {
+                                      int t0, int t1, const int* catoff,
+                                      const unsigned* catw,
+                                      hipStream_t stream) {
   if (n == 0 || t1 <= t0) return;
   long blocks = (n + 255) / 256;
   if (blocks > 8192) blocks = 8192;
   hipLaunchKernelGGL(predict_forest_k, dim3((unsigned)blocks), dim3(256), 0,
                      stream, feat, thr, left, right, val, offsets, tw, X, n,
-                     nf, out, n_outputs, t0, t1);
+                     nf, out, n_outputs, t0, t1, catoff, catw);
 }
 
 // ------------------------------------------------------------- leaf indices
@@ -395,7 +417,9 @@ __global__ void predict_leaf_k(const int* __restrict__ feat,
                                const int* __restrict__ leaf_index,
                                const long* __restrict__ offsets,
                                const float* __restrict__ X, long n, int nf,
-                               int* __restrict__ out, int n_trees) {
+                               int* __restrict__ out, int n_trees,
+                               const int* __restrict__ catoff,
+                               const unsigned* __restrict__ catw) {
   const long row0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
   for (long row = row0; row < n; row += stride) {
@@ -406,7 +430,8 @@ __global__ void predict_leaf_k(const int* __restrict__ feat,
       int f = feat[idx];
       while (f >= 0) {
         const float xv = x[f];
-        idx = base + ((xv <= thr[idx] || isnan(xv)) ? left[idx] : right[idx]);
+        idx = base + (go_left_node(xv, idx, thr, catoff, catw) ? left[idx]
+                                                               : right[idx]);
         f = feat[idx];
       }
       out[row * n_trees + t] = leaf_index[idx];
@@ -418,13 +443,14 @@ extern "C" void launch_predict_leaf(const int* feat, const float* thr,
                                     const int* left, const int* right,
                                     const int* leaf_index, const long* offsets,
                                     const float* X, long n, int nf, int* out,
-                                    int n_trees, hipStream_t stream) {
+                                    int n_trees, const int* catoff,
+                                    const unsigned* catw, hipStream_t stream) {
   if (n == 0) return;
   long blocks = (n + 255) / 256;
   if (blocks > 8192) blocks = 8192;
   hipLaunchKernelGGL(predict_leaf_k, dim3((unsigned)blocks), dim3(256), 0,
                      stream, feat, thr, left, right, leaf_index, offsets, X, n,
-                     nf, out, n_trees);
+                     nf, out, n_trees, catoff, catw);
 }
 
 // ----------------------------------------------------------------- binning
