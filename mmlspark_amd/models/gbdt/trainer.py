@@ -320,8 +320,14 @@ class TreeGrower:
                 mask = torch.isin(bins, cats_t)
                 rows_l, rows_r = leaf.rows[mask], leaf.rows[~mask]
             else:
+                # counts above 2^24 lose exactness through the f32 scan
+                known = (int(leaf.CL) if (self.fixed
+                                          and not self.comm.is_distributed
+                                          and leaf.C < 1.6e7)
+                         else -1)
                 rows_l, rows_r = backend.partition_rows(
-                    self.binned, leaf.rows, leaf.feat, leaf.bin)
+                    self.binned, leaf.rows, leaf.feat, leaf.bin,
+                    known_left=known)
             self.stats.partition_s += time.perf_counter() - t0
 
             GL, HL, CL = leaf.GL, leaf.HL, leaf.CL

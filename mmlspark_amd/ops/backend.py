@@ -62,12 +62,13 @@ def hist_build_fixed(binned_i4, rows, grad, hess, n_bins, scale_g, scale_h):
     return (h.double() * scale).round().long()
 
 
-def partition_rows(binned_i4, rows, feature, threshold_bin):
+def partition_rows(binned_i4, rows, feature, threshold_bin, known_left=-1):
     if binned_i4.is_cuda:
-        # ordered 3-kernel partition, ONE device→host sync (torch's two
-        # masked_selects cost two nonzero() syncs per split)
+        # ordered 3-kernel partition; ONE device→host sync, or zero when the
+        # caller already knows the left count (single-rank training)
         return _require_ext().partition_rows(binned_i4, rows.contiguous(),
-                                             feature, threshold_bin)
+                                             feature, threshold_bin,
+                                             known_left)
     return cpu_ref.partition_rows(binned_i4, rows, feature, threshold_bin)
 
 

@@ -132,7 +132,8 @@ torch::Tensor predict_leaf(torch::Tensor feat, torch::Tensor thr,
 }
 
 std::tuple<torch::Tensor, torch::Tensor> partition_rows(
-    torch::Tensor binned_i4, torch::Tensor rows, long feature, long thr) {
+    torch::Tensor binned_i4, torch::Tensor rows, long feature, long thr,
+    long known_left) {
   CHECK_DEV(binned_i4); CHECK_CONTIG(binned_i4);
   CHECK_DEV(rows); CHECK_CONTIG(rows);
   TORCH_CHECK(rows.dtype() == torch::kInt32, "rows must be int32");
@@ -146,7 +147,9 @@ std::tuple<torch::Tensor, torch::Tensor> partition_rows(
                      out.data_ptr<int>(), scratch.data_ptr<int>(),
                      total.data_ptr<int>(), cur_stream());
   }
-  const long nl = total.item<int>();  // the ONE sync per split
+  // single-rank training already knows the exact left count from the split
+  // stats (integer histogram counts) — skip the device→host sync entirely
+  const long nl = known_left >= 0 ? known_left : total.item<int>();
   return {out.slice(0, 0, nl), out.slice(0, nl, m)};
 }
 
