@@ -112,6 +112,8 @@ class _GBDTParams(Params):
             metric=self.get("metric"),
             verbosity=self.get("verbosity"),
             categorical_features=self.get("categoricalSlotIndexes"),
+            parallelism=self.get("parallelism"),
+            top_k=self.get("topK"),
         )
 
     def _device(self):

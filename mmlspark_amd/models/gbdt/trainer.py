@@ -71,6 +71,8 @@ class TrainConfig:
     is_provide_training_metric: bool = False
     metric: str = ""
     categorical_features: Optional[List[int]] = None
+    parallelism: str = "data_parallel"
+    top_k: int = 20
 
 
 @dataclass
@@ -157,6 +159,13 @@ class TreeGrower:
         self.scale_g = (2.0 ** 61) / (max(self.n_global, 1) * gmax)
         self.scale_h = (2.0 ** 24) / hmax  # chunk ≤ 2^19 rows → fits 44 bits
 
+    @property
+    def voting(self) -> bool:
+        """Voting-parallel: reduce only globally-voted top-K
+        feature histograms instead of all of them."""
+        return (getattr(self.cfg, "parallelism", "data_parallel")
+                == "voting_parallel" and self.comm.is_distributed)
+
     def _hist(self, rows, grad, hess, reduce=True):
         t0 = time.perf_counter()
         if self.fixed:
@@ -171,9 +180,36 @@ class TreeGrower:
         t1 = time.perf_counter()
         self.stats.hist_s += t1 - t0
         if reduce:
-            self.comm.all_reduce(h)
+            if self.voting:
+                self._voting_reduce(h)
+            else:
+                self.comm.all_reduce(h)
             self.stats.comm_s += time.perf_counter() - t1
         return h
+
+    def _voting_reduce(self, h: torch.Tensor):
+        """Local top-K vote → all_gather of candidate feature ids →
+        all_reduce of the union's histograms only (LightGBM voting_parallel,
+        LightGBMConstants TopK=20).  Non-union features get their counts
+        zeroed so the split scan can never pick a locally-reduced feature."""
+        k = min(getattr(self.cfg, "top_k", 20), self.nf)
+        histf = self._to_float_hist(h.unsqueeze(0))[0]
+        # per-feature gains for the vote
+        g = histf[:, :, 0].cumsum(1)
+        hh = histf[:, :, 1].cumsum(1)
+        gains = (g[:, :-1] ** 2 / (hh[:, :-1] + 1e-6)
+                 + (g[:, -1:] - g[:, :-1]) ** 2
+                 / (hh[:, -1:] - hh[:, :-1] + 1e-6)).amax(dim=1)
+        gains[self.nf:] = float("-inf")
+        local_top = torch.topk(gains, k).indices.to(torch.int64)
+        cands = torch.cat(self.comm.all_gather(local_top.contiguous()))
+        union = torch.unique(cands)
+        sub = h.index_select(0, union).contiguous()
+        self.comm.all_reduce(sub)
+        keep = torch.zeros(h.shape[0], dtype=torch.bool, device=h.device)
+        keep[union] = True
+        h[~keep] = 0  # counts 0 ⇒ min_data_in_leaf rejects these features
+        h[union] = sub
 
     def _to_float_hist(self, hists: torch.Tensor) -> torch.Tensor:
         """Stacked histograms → float32 real units for the split scan."""
@@ -291,7 +327,15 @@ class TreeGrower:
 
         self.set_scales(grad, hess)
         root_hist = self._hist(rows_root, grad, hess)
-        G, H, C = self._sums(root_hist)
+        if self.voting:
+            t = torch.stack([grad[rows_root.long()].sum(),
+                             hess[rows_root.long()].sum(),
+                             torch.tensor(float(rows_root.numel()),
+                                          device=self.device)])
+            self.comm.all_reduce(t)
+            G, H, C = float(t[0]), float(t[1]), float(t[2])
+        else:
+            G, H, C = self._sums(root_hist)
         root = _Leaf(new_node(), rows_root, root_hist, G, H, C, 0)
         (root.gain, root.feat, root.bin, root.GL, root.HL, root.CL,
          root.cats) = self._best_split(root_hist, feat_mask)
@@ -332,13 +376,19 @@ class TreeGrower:
 
             GL, HL, CL = leaf.GL, leaf.HL, leaf.CL
             GR, HR, CR = leaf.G - GL, leaf.H - HL, leaf.C - CL
-            # reduce only the globally-smaller child; sibling by subtraction
-            left_small = CL <= CR
-            small_rows = rows_l if left_small else rows_r
-            hist_small = self._hist(small_rows, grad, hess)
-            hist_big = leaf.hist - hist_small
-            hist_l, hist_r = ((hist_small, hist_big) if left_small
-                              else (hist_big, hist_small))
+            if self.voting:
+                # voting mode reduces a per-leaf feature union, so the parent
+                # histogram is not globally complete — build both children
+                hist_l = self._hist(rows_l, grad, hess)
+                hist_r = self._hist(rows_r, grad, hess)
+            else:
+                # reduce only the globally-smaller child; sibling by subtraction
+                left_small = CL <= CR
+                small_rows = rows_l if left_small else rows_r
+                hist_small = self._hist(small_rows, grad, hess)
+                hist_big = leaf.hist - hist_small
+                hist_l, hist_r = ((hist_small, hist_big) if left_small
+                                  else (hist_big, hist_small))
 
             nid = leaf.node_id
             feature[nid] = leaf.feat

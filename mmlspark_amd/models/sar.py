@@ -205,3 +205,87 @@ class RankingEvaluator(Transformer):
 
     def _transform(self, df):
         return df
+
+
+@register
+class RankingAdapter(Transformer):
+    """Adapt a fitted recommender into (prediction-list, label-list) rows for
+    RankingEvaluator (core/.../recommendation/RankingAdapter parity)."""
+    recommenderModel = Param("recommenderModel", "fitted SARModel", None,
+                             is_complex=True)
+    userCol = Param("userCol", "user index column", "userIdx")
+    itemCol = Param("itemCol", "item index column", "itemIdx")
+    k = Param("k", "recommendations per user", 10, toInt)
+
+    def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
+        model = self.get("recommenderModel")
+        recs = model.recommendForAllUsers(self.get("k"), remove_seen=False)
+        ucol, icol = self.get("userCol"), self.get("itemCol")
+        truth = df.groupby(ucol)[icol].apply(list)
+        rec_map = {int(r[ucol]): [x[icol] for x in r["recommendations"]]
+                   for _, r in recs.iterrows()}
+        rows = []
+        for u, items in truth.items():
+            rows.append({ucol: u, "prediction": rec_map.get(int(u), []),
+                         "label": items})
+        return pd.DataFrame(rows)
+
+
+@register
+class RankingTrainValidationSplit(Estimator):
+    """Per-user temporal/random split + fit + ranking evaluation
+    (core/.../recommendation/RankingTrainValidationSplit parity, incl. the
+    min-ratings filtering that avoids user/item cold start)."""
+    estimator = Param("estimator", "recommender estimator (e.g. SAR)", None,
+                      is_complex=True)
+    userCol = Param("userCol", "user index column", "userIdx")
+    itemCol = Param("itemCol", "item index column", "itemIdx")
+    ratingCol = Param("ratingCol", "rating column", "rating")
+    trainRatio = Param("trainRatio", "train fraction per user", 0.75, toFloat)
+    minRatingsPerUser = Param("minRatingsPerUser", "drop colder users", 1, toInt)
+    minRatingsPerItem = Param("minRatingsPerItem", "drop colder items", 1, toInt)
+    k = Param("k", "eval cutoff", 10, toInt)
+    seed = Param("seed", "split seed", 0, toInt)
+
+    def _fit(self, df: pd.DataFrame):
+        rng = np.random.default_rng(self.get("seed"))
+        ucol, icol = self.get("userCol"), self.get("itemCol")
+        # cold-start filtering
+        uc = df[ucol].value_counts()
+        ic = df[icol].value_counts()
+        df = df[df[ucol].isin(uc[uc >= self.get("minRatingsPerUser")].index)
+                & df[icol].isin(ic[ic >= self.get("minRatingsPerItem")].index)]
+        train_idx = []
+        val_idx = []
+        for _, g in df.groupby(ucol, sort=False):
+            idx = g.index.to_numpy()
+            rng.shuffle(idx)
+            cut = max(1, int(len(idx) * self.get("trainRatio")))
+            train_idx.extend(idx[:cut])
+            val_idx.extend(idx[cut:])
+        train = df.loc[train_idx]
+        valid = df.loc[val_idx]
+        model = self.get("estimator").fit(train)
+        adapter = RankingAdapter(recommenderModel=model, userCol=ucol,
+                                 itemCol=icol, k=self.get("k"))
+        ranked = adapter.transform(valid) if len(valid) else pd.DataFrame(
+            columns=[ucol, "prediction", "label"])
+        ev = RankingEvaluator(k=self.get("k"), metricName="ndcgAt")
+        out = RankingTrainValidationSplitModel(best=model)
+        out.set("validationMetric",
+                ev.evaluate(ranked) if len(ranked) else 0.0)
+        return out
+
+
+@register
+class RankingTrainValidationSplitModel(Model):
+    bestModel = Param("bestModel", "fitted recommender", None, is_complex=True)
+    validationMetric = Param("validationMetric", "NDCG@k on validation", None)
+
+    def __init__(self, best=None, **kwargs):
+        super().__init__(**kwargs)
+        if best is not None:
+            self.set("bestModel", best)
+
+    def _transform(self, df):
+        return self.get("bestModel").transform(df)

@@ -189,3 +189,50 @@ def test_ddp_deep_vision_replicas_agree():
     assert not any(isinstance(s, str) for s in results.values()), results
     # DDP gradient sync keeps replicas identical
     assert abs(results[0] - results[1]) < 1e-3, results
+
+
+def _worker_voting(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        torch.distributed.init_process_group(
+            "gloo", init_method=f"tcp://127.0.0.1:{port}",
+            rank=rank, world_size=world)
+        from mmlspark_amd.models.gbdt.objectives import make_objective
+        from mmlspark_amd.models.gbdt.trainer import TrainConfig, train_booster
+        from mmlspark_amd.parallel.comm import Comm
+
+        X, y = _make_data(0, n=4000, nf=20)
+        sl = slice(rank * 2000, (rank + 1) * 2000)
+        cfg = TrainConfig(num_iterations=8, num_leaves=15, seed=7,
+                          parallelism="voting_parallel", top_k=5)
+        booster, _ = train_booster(torch.from_numpy(X[sl]),
+                                   torch.from_numpy(y[sl]), cfg,
+                                   make_objective("binary"), Comm())
+        q.put((rank, booster.save_to_string()))
+        torch.distributed.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"ERROR: {e!r}"))
+
+
+@pytest.mark.timeout(180)
+def test_voting_parallel_identical_and_accurate():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_voting, args=(r, 2, 29874, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, s = q.get(timeout=150)
+        results[rank] = s
+    for p in procs:
+        p.join(timeout=30)
+    assert not any(str(s).startswith("ERROR") for s in results.values()), results
+    assert results[0] == results[1]  # ranks agree bit-for-bit
+    from sklearn.metrics import roc_auc_score
+    from mmlspark_amd.models.gbdt.booster import Booster
+    X, y = _make_data(0, n=4000, nf=20)
+    b = Booster.load_from_string(results[0])
+    p = torch.sigmoid(b.predict_raw(torch.from_numpy(X)).squeeze(-1)).numpy()
+    assert roc_auc_score(y, p) > 0.9

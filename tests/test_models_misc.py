@@ -143,3 +143,22 @@ def test_conditional_ball_tree():
     assert [v for v, _ in got] == list(brute)
     for v, _ in got:
         assert labels[v] == 1
+
+
+def test_ranking_train_validation_split():
+    from mmlspark_amd.models.sar import RankingTrainValidationSplit, SAR
+    rng = np.random.default_rng(9)
+    rows = []
+    for u in range(30):
+        group = 0 if u < 15 else 1
+        items = rng.choice(np.arange(10) + group * 10, size=6, replace=False)
+        for it in items:
+            rows.append({"userIdx": u, "itemIdx": int(it), "rating": 1.0})
+    df = pd.DataFrame(rows)
+    tvs = RankingTrainValidationSplit(
+        estimator=SAR(supportThreshold=1), trainRatio=0.7, k=5,
+        minRatingsPerUser=2)
+    m = tvs.fit(df)
+    assert m.get("validationMetric") > 0.02
+    scored = m.transform(df.head(5))
+    assert "prediction" in scored.columns
