@@ -211,3 +211,53 @@ def test_categorical_gpu_matches_cpu():
     m.booster.invalidate_cache()
     raw_cpu = m.booster.predict_raw(Xt)
     assert torch.allclose(raw_gpu, raw_cpu, atol=1e-5)
+
+
+@requires_gpu
+def test_image_featurizer_gpu():
+    import pandas as pd
+    from mmlspark_amd.models.image_featurizer import ImageFeaturizer
+    rng = np.random.default_rng(20)
+    imgs = [rng.integers(0, 255, size=(32, 32, 3)).astype(np.uint8)
+            for _ in range(8)]
+    df = pd.DataFrame({"image": imgs})
+    f = ImageFeaturizer(modelName="ResNet50", imageSize=64, cutOutputLayers=1,
+                        device="cuda", batchSize=4)
+    feats = np.stack(f.transform(df)["features"].to_numpy())
+    assert feats.shape == (8, 2048)
+    assert np.isfinite(feats).all()
+
+
+@requires_gpu
+def test_tabular_shap_gpu_model(binary_df):
+    import pandas as pd
+    from mmlspark_amd.explainers.shap import TabularSHAP
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    X = np.stack(binary_df["features"].to_numpy())
+    cols = [f"c{i}" for i in range(X.shape[1])]
+    df = pd.DataFrame(X, columns=cols)
+    df["label"] = binary_df["label"].to_numpy()
+    model = LightGBMClassifier(featureCols=cols, numIterations=10,
+                               numLeaves=15, device="cuda").fit(df)
+    shap = TabularSHAP(inputCols=cols, model=model, targetCol="probability",
+                       targetClasses=[1], numSamples=128,
+                       backgroundData=df.head(100))
+    out = shap.transform(df.head(4))
+    exp = np.stack(out["explanation"].to_numpy())
+    probs = np.stack(model.transform(df.head(4))["probability"].to_numpy())[:, 1]
+    np.testing.assert_allclose(exp[:, 0, :].sum(axis=1), probs, atol=1e-3)
+
+
+@requires_gpu
+def test_serving_gpu_hipgraph(binary_df):
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    from mmlspark_amd.serving.server import LowLatencyGBDTScorer
+    model = LightGBMClassifier(numIterations=20, numLeaves=15,
+                               device="cuda").fit(binary_df)
+    scorer = LowLatencyGBDTScorer(model.booster, max_batch=4)
+    assert scorer.graph is not None, "hipGraph capture must succeed on GPU"
+    X = np.stack(binary_df["features"].to_numpy()[:3])
+    p = scorer.score(X)
+    ref = np.stack(model.transform(binary_df.head(3))["probability"]
+                   .to_numpy())[:, 1]
+    np.testing.assert_allclose(p[:, 0], ref, atol=1e-5)
