@@ -12,7 +12,7 @@ import numpy as np
 import pandas as pd
 
 from ..core.param import Param, toBool, toInt, toList, toString
-from ..core.pipeline import Transformer
+from ..core.pipeline import Estimator, Model, Transformer
 from ..core.registry import register
 
 
@@ -247,3 +247,57 @@ class UnicodeNormalize(Transformer):
         out = df.copy()
         out[self.get("outputCol")] = df[self.get("inputCol")].map(proc)
         return out
+
+
+@register
+class ClassBalancer(Estimator):
+    """Per-class balancing weights: weight = max(count)/count(class)
+    (stages/ClassBalancer.scala)."""
+    inputCol = Param("inputCol", "label column", "label")
+    outputCol = Param("outputCol", "weight column", "weight")
+    broadcastJoin = Param("broadcastJoin", "join strategy hint", True, toBool)
+
+    def _fit(self, df):
+        counts = df[self.get("inputCol")].value_counts()
+        mx = counts.max()
+        m = ClassBalancerModel()
+        m.set("weights", {str(k): float(mx / v) for k, v in counts.items()})
+        m.set("inputCol", self.get("inputCol"))
+        m.set("outputCol", self.get("outputCol"))
+        return m
+
+
+@register
+class ClassBalancerModel(Model):
+    inputCol = Param("inputCol", "label column", "label")
+    outputCol = Param("outputCol", "weight column", "weight")
+    weights = Param("weights", "class → weight", None, is_complex=True)
+
+    def _transform(self, df):
+        w = self.get("weights")
+        out = df.copy()
+        out[self.get("outputCol")] = [w.get(str(v), 1.0)
+                                      for v in df[self.get("inputCol")]]
+        return out
+
+
+@register
+class StratifiedRepartition(Transformer):
+    """Reorder rows so every equal-size shard (GPU rank) sees the same label
+    mix (stages/StratifiedRepartition.scala: class-balanced partitions)."""
+    labelCol = Param("labelCol", "label column", "label")
+    mode = Param("mode", "equal|original|mixed", "equal", toString)
+    seed = Param("seed", "shuffle seed", 0, toInt)
+
+    def _transform(self, df):
+        rng = np.random.default_rng(self.get("seed"))
+        groups = [g.sample(frac=1.0, random_state=int(rng.integers(1 << 31)))
+                  for _, g in df.groupby(self.get("labelCol"), sort=False)]
+        # round-robin interleave classes so contiguous shards are stratified
+        idx_lists = [list(g.index) for g in groups]
+        order = []
+        while any(idx_lists):
+            for lst in idx_lists:
+                if lst:
+                    order.append(lst.pop(0))
+        return df.loc[order].reset_index(drop=True)

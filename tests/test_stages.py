@@ -195,3 +195,20 @@ def test_find_best_model(mixed_class_df):
     fb = FindBestModel(models=[m1, m2], evaluationMetric="accuracy").fit(dff)
     assert fb.get("bestModel").uid == m2.uid
     assert len(fb.getEvaluationResults()) == 2
+
+
+def test_class_balancer():
+    from mmlspark_amd.stages.basic import ClassBalancer
+    df = pd.DataFrame({"label": [0, 0, 0, 1]})
+    m = ClassBalancer(inputCol="label").fit(df)
+    out = m.transform(df)
+    assert out["weight"].tolist() == [1.0, 1.0, 1.0, 3.0]
+
+
+def test_stratified_repartition():
+    from mmlspark_amd.stages.basic import StratifiedRepartition
+    df = pd.DataFrame({"label": [0] * 8 + [1] * 8, "x": range(16)})
+    out = StratifiedRepartition(labelCol="label").transform(df)
+    # every contiguous half (2-rank shard) sees both classes evenly
+    first, second = out.head(8), out.tail(8)
+    assert first["label"].sum() == 4 and second["label"].sum() == 4
