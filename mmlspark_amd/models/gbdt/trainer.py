@@ -304,8 +304,52 @@ class TreeGrower:
     def _best_split(self, hist, feat_mask):
         return self._scan(hist.unsqueeze(0), feat_mask)[0]
 
+    def _grow_native(self, rows_root, grad, hess, feat_mask):
+        """Whole leaf-wise loop in the C++ driver (ops/hip/gbdt_grower.cpp)
+        — one Python call per tree instead of ~6 per split."""
+        from ...ops import _hip_grower
+        cfg = self.cfg
+        reduce_fn = None
+        if self.comm.is_distributed:
+            comm = self.comm
+            reduce_fn = lambda t: comm.all_reduce(t)  # noqa: E731
+        t0 = time.perf_counter()
+        d = _hip_grower.grow_tree_native(
+            self.binned, rows_root.contiguous(), grad, hess, cfg.max_bin,
+            self.nf, self.scale_g, self.scale_h, cfg.lambda_l1, cfg.lambda_l2,
+            float(cfg.min_data_in_leaf), cfg.min_sum_hessian_in_leaf,
+            cfg.min_gain_to_split, cfg.max_delta_step, cfg.num_leaves,
+            cfg.max_depth, feat_mask, reduce_fn, self.comm.is_distributed)
+        self.stats.hist_s += time.perf_counter() - t0
+        feature = d["feature"].numpy()
+        thr_bin = d["thr_bin"].numpy()
+        threshold = np.array(
+            [self.bin_mapper.bin_upper_value(int(f), int(b)) if f >= 0 else 0.0
+             for f, b in zip(feature, thr_bin)], dtype=np.float32)
+        tree = Tree(feature, threshold, thr_bin, d["left"].numpy(),
+                    d["right"].numpy(), d["value"].numpy(),
+                    d["count"].numpy(), d["gain"].numpy(),
+                    d["leaf_index"].numpy(), shrinkage=1.0)
+        offs = d["leaf_offsets"].tolist()
+        nodes = d["leaf_nodes"].tolist()
+        leaves = []
+        for i, nid in enumerate(nodes):
+            lf = _Leaf(int(nid), d["leaf_rows"][offs[i]:offs[i + 1]], None,
+                       0, 0, 0, 0)
+            leaves.append(lf)
+        return tree, leaves
+
     def grow(self, rows_root: torch.Tensor, grad: torch.Tensor,
              hess: torch.Tensor, feat_mask) -> (Tree, List):
+        if (self.fixed and not self.cat_features and not self.voting
+                and not __import__("os").environ.get(
+                    "MMLSPARK_AMD_NO_NATIVE_GROWER")):
+            try:
+                from ...ops import _hip_grower  # noqa: F401
+                self.set_scales(grad, hess)
+                return self._grow_native(rows_root, grad, hess, feat_mask)
+            except ImportError:
+                pass
         cfg = self.cfg
         # node arrays (grown dynamically)
         feature, threshold, thr_bin = [], [], []

@@ -1,0 +1,304 @@
+// Native leaf-wise tree-growth driver — the C++ runtime piece of the GBDT.
+//
+// The Python grower pays ~150-250 µs of interpreter/dispatch overhead per
+// split (63 splits/tree); this driver runs the entire leaf-wise loop
+// (histogram → all_reduce → fused split scan → readback → ordered partition
+// → children) in C++, calling the same HIP launchers, and returns the
+// finished node arrays + per-leaf row segments.  Multi-rank histogram
+// reduction happens through an optional Python callback (one GIL hop per
+// reduce — the collective dominates).  Categorical features fall back to the
+// Python grower (models/gbdt/trainer.py).
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <cmath>
+#include <queue>
+#include <vector>
+
+extern "C" {
+void launch_hist_build_fixed(const void*, long, const int*, long, const float*,
+                             const float*, long long*, int, int, double,
+                             double, hipStream_t);
+void launch_split_scan(const float*, int, long, int, float, float, float,
+                       float, float, long, const bool*, float*, float*,
+                       hipStream_t);
+void launch_partition(const void*, long, const int*, long, int, int, int*,
+                      int*, int*, hipStream_t);
+}
+
+static hipStream_t grower_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+namespace {
+
+struct LeafCand {
+  double gain;
+  long seq;
+  int node_id;
+  int depth;
+  torch::Tensor rows;
+  torch::Tensor hist;   // int64 (nf_pad, nb, 3), globally reduced
+  double G, H, C;
+  double GL, HL, CL;
+  int feat, bin;
+};
+
+struct CandCmp {
+  bool operator()(const LeafCand& a, const LeafCand& b) const {
+    if (a.gain != b.gain) return a.gain < b.gain;  // max-heap by gain
+    return a.seq > b.seq;
+  }
+};
+
+struct GrowCtx {
+  torch::Tensor binned;
+  long n_rows;
+  int n_bins;
+  int ngroups;
+  long nf;
+  double scale_g, scale_h;
+  double l1, l2, min_data, min_hess, min_gain;
+  int num_leaves, max_depth;
+  torch::Tensor feat_mask;  // bool (nf_pad,) or undefined
+  py::object reduce_fn;     // callable(tensor) or None
+  bool has_reduce;
+  torch::Tensor inv_scales;  // double (3,)
+};
+
+torch::Tensor build_hist(GrowCtx& ctx, const torch::Tensor& rows,
+                         const torch::Tensor& grad, const torch::Tensor& hess) {
+  auto hist = torch::zeros({ctx.ngroups * 4, ctx.n_bins, 3},
+                           grad.options().dtype(torch::kInt64));
+  launch_hist_build_fixed(ctx.binned.data_ptr(), ctx.n_rows,
+                          rows.data_ptr<int>(), rows.numel(),
+                          grad.data_ptr<float>(), hess.data_ptr<float>(),
+                          (long long*)hist.data_ptr<int64_t>(), ctx.n_bins,
+                          ctx.ngroups, ctx.scale_g, ctx.scale_h,
+                          grower_stream());
+  if (ctx.has_reduce) {
+    py::gil_scoped_acquire gil;
+    ctx.reduce_fn(hist);
+  }
+  return hist;
+}
+
+// returns (gain, feat, bin, GL, HL, CL) per histogram in the stack
+std::vector<std::array<double, 6>> scan_pair(GrowCtx& ctx,
+                                             const torch::Tensor& hists_i64) {
+  auto histsf = (hists_i64.to(torch::kDouble) * ctx.inv_scales)
+                    .to(torch::kFloat32).contiguous();
+  const long nh = histsf.size(0);
+  const long nf_pad = histsf.size(1);
+  auto scratch = torch::empty({nh, nf_pad, 6}, histsf.options());
+  auto out = torch::empty({nh, 6}, histsf.options());
+  const bool* mask = ctx.feat_mask.defined()
+                         ? ctx.feat_mask.data_ptr<bool>() : nullptr;
+  launch_split_scan(histsf.data_ptr<float>(), (int)nh, nf_pad, ctx.n_bins,
+                    (float)ctx.l1, (float)ctx.l2, (float)ctx.min_data,
+                    (float)ctx.min_hess, (float)ctx.min_gain, ctx.nf, mask,
+                    scratch.data_ptr<float>(), out.data_ptr<float>(),
+                    grower_stream());
+  auto host = out.to(torch::kCPU);  // single sync per scan
+  auto acc = host.accessor<float, 2>();
+  std::vector<std::array<double, 6>> res((size_t)nh);
+  for (long i = 0; i < nh; ++i)
+    for (int j = 0; j < 6; ++j) res[i][j] = acc[i][j];
+  return res;
+}
+
+double leaf_output(double G, double H, double l1, double l2,
+                   double max_delta) {
+  double g = std::abs(G) - l1;
+  if (g <= 0) return 0.0;
+  double w = -std::copysign(g, G) / (H + l2);
+  if (max_delta > 0) w = std::clamp(w, -max_delta, max_delta);
+  return w;
+}
+
+}  // namespace
+
+// Returns dict with node arrays (CPU int32/f32 tensors), per-leaf rows
+// (device int32, concatenated) + offsets + leaf node ids + leaf values.
+py::dict grow_tree_native(torch::Tensor binned, torch::Tensor rows_root,
+                          torch::Tensor grad, torch::Tensor hess,
+                          long n_bins, long nf, double scale_g, double scale_h,
+                          double l1, double l2, double min_data,
+                          double min_hess, double min_gain, double max_delta,
+                          long num_leaves, long max_depth,
+                          c10::optional<torch::Tensor> feat_mask,
+                          py::object reduce_fn, bool distributed) {
+  GrowCtx ctx;
+  ctx.binned = binned;
+  ctx.n_rows = binned.size(1);
+  ctx.n_bins = (int)n_bins;
+  ctx.ngroups = (int)binned.size(0);
+  ctx.nf = nf;
+  ctx.scale_g = scale_g;
+  ctx.scale_h = scale_h;
+  ctx.l1 = l1; ctx.l2 = l2;
+  ctx.min_data = min_data; ctx.min_hess = min_hess; ctx.min_gain = min_gain;
+  ctx.num_leaves = (int)num_leaves;
+  ctx.max_depth = (int)max_depth;
+  if (feat_mask.has_value()) ctx.feat_mask = *feat_mask;
+  ctx.reduce_fn = reduce_fn;
+  ctx.has_reduce = !reduce_fn.is_none();
+  ctx.inv_scales = torch::tensor({1.0 / scale_g, 1.0 / scale_h, 1.0},
+                                 torch::TensorOptions()
+                                     .dtype(torch::kDouble)
+                                     .device(binned.device()));
+
+  py::gil_scoped_release nogil;
+
+  std::vector<int> feature_, thr_bin_, left_, right_, leaf_idx_;
+  std::vector<float> value_, count_, gain_;
+  auto new_node = [&]() {
+    feature_.push_back(-1); thr_bin_.push_back(0); left_.push_back(-1);
+    right_.push_back(-1); value_.push_back(0.f); count_.push_back(0.f);
+    gain_.push_back(0.f); leaf_idx_.push_back(-1);
+    return (int)feature_.size() - 1;
+  };
+
+  auto root_hist = build_hist(ctx, rows_root, grad, hess);
+  auto sums = root_hist.select(0, 0).sum(0).to(torch::kCPU);
+  auto sa = sums.accessor<int64_t, 1>();
+  const double G0 = (double)sa[0] / scale_g;
+  const double H0 = (double)sa[1] / scale_h;
+  const double C0 = (double)sa[2];
+
+  LeafCand root;
+  root.node_id = new_node();
+  root.depth = 0;
+  root.seq = 0;
+  root.rows = rows_root;
+  root.hist = root_hist;
+  root.G = G0; root.H = H0; root.C = C0;
+  {
+    auto r = scan_pair(ctx, root_hist.unsqueeze(0))[0];
+    root.gain = r[0]; root.feat = (int)r[1]; root.bin = (int)r[2];
+    root.GL = r[3]; root.HL = r[4]; root.CL = r[5];
+  }
+  count_[root.node_id] = (float)C0;
+  value_[root.node_id] = (float)leaf_output(G0, H0, l1, l2, max_delta);
+
+  std::priority_queue<LeafCand, std::vector<LeafCand>, CandCmp> heap;
+  std::vector<LeafCand> finals;
+  heap.push(root);
+  finals.push_back(root);
+  long seq = 1;
+  int n_leaves = 1;
+
+  auto scratch = torch::empty({4096},
+                              rows_root.options().dtype(torch::kInt32));
+  auto total = torch::zeros({1}, rows_root.options().dtype(torch::kInt32));
+
+  while (n_leaves < ctx.num_leaves && !heap.empty()) {
+    LeafCand leaf = heap.top();
+    heap.pop();
+    if (!(leaf.gain > ctx.min_gain) || !std::isfinite(leaf.gain)) continue;
+    if (ctx.max_depth > 0 && leaf.depth >= ctx.max_depth) continue;
+    // remove from finals
+    for (size_t i = 0; i < finals.size(); ++i)
+      if (finals[i].node_id == leaf.node_id) {
+        finals.erase(finals.begin() + i);
+        break;
+      }
+
+    const long m = leaf.rows.numel();
+    auto out_rows = torch::empty({m}, leaf.rows.options());
+    launch_partition(ctx.binned.data_ptr(), ctx.n_rows,
+                     leaf.rows.data_ptr<int>(), m, leaf.feat, leaf.bin,
+                     out_rows.data_ptr<int>(), scratch.data_ptr<int>(),
+                     total.data_ptr<int>(), grower_stream());
+    long nl;
+    if (!distributed && leaf.C < 1.6e7) {
+      nl = (long)leaf.CL;  // exact integer counts, no sync
+    } else {
+      nl = total.to(torch::kCPU).item<int>();
+    }
+    auto rows_l = out_rows.slice(0, 0, nl);
+    auto rows_r = out_rows.slice(0, nl, m);
+
+    const double GL = leaf.GL, HL = leaf.HL, CL = leaf.CL;
+    const double GR = leaf.G - GL, HR = leaf.H - HL, CR = leaf.C - CL;
+    const bool left_small = CL <= CR;
+    auto hist_small = build_hist(ctx, left_small ? rows_l : rows_r, grad, hess);
+    auto hist_big = leaf.hist - hist_small;
+    auto hist_l = left_small ? hist_small : hist_big;
+    auto hist_r = left_small ? hist_big : hist_small;
+
+    const int nid = leaf.node_id;
+    feature_[nid] = leaf.feat;
+    thr_bin_[nid] = leaf.bin;
+    gain_[nid] = (float)leaf.gain;
+    const int lid = new_node();
+    const int rid = new_node();
+    left_[nid] = lid;
+    right_[nid] = rid;
+    count_[lid] = (float)CL;
+    count_[rid] = (float)CR;
+    value_[lid] = (float)leaf_output(GL, HL, l1, l2, max_delta);
+    value_[rid] = (float)leaf_output(GR, HR, l1, l2, max_delta);
+
+    auto pair = scan_pair(ctx, torch::stack({hist_l, hist_r}));
+    LeafCand lc, rc;
+    lc.node_id = lid; rc.node_id = rid;
+    lc.depth = rc.depth = leaf.depth + 1;
+    lc.rows = rows_l; rc.rows = rows_r;
+    lc.hist = hist_l; rc.hist = hist_r;
+    lc.G = GL; lc.H = HL; lc.C = CL;
+    rc.G = GR; rc.H = HR; rc.C = CR;
+    lc.gain = pair[0][0]; lc.feat = (int)pair[0][1]; lc.bin = (int)pair[0][2];
+    lc.GL = pair[0][3]; lc.HL = pair[0][4]; lc.CL = pair[0][5];
+    rc.gain = pair[1][0]; rc.feat = (int)pair[1][1]; rc.bin = (int)pair[1][2];
+    rc.GL = pair[1][3]; rc.HL = pair[1][4]; rc.CL = pair[1][5];
+    lc.seq = seq++;
+    rc.seq = seq++;
+    heap.push(lc);
+    heap.push(rc);
+    finals.push_back(lc);
+    finals.push_back(rc);
+    n_leaves += 1;
+  }
+
+  // leaf ordinals in node order; collect row segments
+  std::sort(finals.begin(), finals.end(),
+            [](const LeafCand& a, const LeafCand& b) {
+              return a.node_id < b.node_id;
+            });
+  std::vector<torch::Tensor> segs;
+  std::vector<int64_t> seg_nodes;
+  for (size_t i = 0; i < finals.size(); ++i) {
+    leaf_idx_[finals[i].node_id] = (int)i;
+    segs.push_back(finals[i].rows);
+    seg_nodes.push_back(finals[i].node_id);
+  }
+  auto leaf_rows = segs.empty()
+                       ? torch::empty({0}, rows_root.options())
+                       : torch::cat(segs);
+  std::vector<int64_t> offs(1, 0);
+  for (auto& s : segs) offs.push_back(offs.back() + s.numel());
+
+  py::gil_scoped_acquire gil;
+  auto i32 = torch::TensorOptions().dtype(torch::kInt32);
+  auto f32 = torch::TensorOptions().dtype(torch::kFloat32);
+  py::dict d;
+  d["feature"] = torch::tensor(feature_, i32);
+  d["thr_bin"] = torch::tensor(thr_bin_, i32);
+  d["left"] = torch::tensor(left_, i32);
+  d["right"] = torch::tensor(right_, i32);
+  d["value"] = torch::tensor(value_, f32);
+  d["count"] = torch::tensor(count_, f32);
+  d["gain"] = torch::tensor(gain_, f32);
+  d["leaf_index"] = torch::tensor(leaf_idx_, i32);
+  d["leaf_rows"] = leaf_rows;
+  d["leaf_offsets"] = torch::tensor(offs, torch::kInt64);
+  d["leaf_nodes"] = torch::tensor(seg_nodes, torch::kInt64);
+  return d;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("grow_tree_native", &grow_tree_native,
+        "native leaf-wise GBDT tree growth (numeric features)");
+}

@@ -50,7 +50,15 @@ setup(
             extra_compile_args=["-O2", "-D__HIP_PLATFORM_AMD__=1",
                                 f"-I{ROCM}/include"],
             extra_link_args=[f"-L{ROCM}/lib", "-lamdhip64"],
-        )
+        ),
+        cpp_extension.CppExtension(
+            name="mmlspark_amd.ops._hip_grower",
+            sources=[os.path.join(HIP_DIR, "gbdt_grower.cpp")],
+            extra_objects=[o for o in extra_objects if "gbdt" in o],
+            extra_compile_args=["-O2", "-D__HIP_PLATFORM_AMD__=1",
+                                f"-I{ROCM}/include"],
+            extra_link_args=[f"-L{ROCM}/lib", "-lamdhip64"],
+        ),
     ],
     cmdclass={"build_ext": cpp_extension.BuildExtension.with_options(
         use_ninja=False)},
