@@ -172,3 +172,84 @@ class Booster:
         self.tree_weights = np.concatenate([self.tree_weights, other.tree_weights])
         self.invalidate_cache()
         return self
+
+
+def _to_lightgbm_text(booster: "Booster") -> str:
+    """Export in LightGBM's native text model format (v3) so models trained
+    here load in stock LightGBM tooling — the interop the reference's
+    saveNativeModel gives its users (LightGBMClassifier.scala:185-205).
+    Categorical bitset splits export with decision_type=1 (==) per LightGBM's
+    categorical encoding via the cat_boundaries/cat_threshold arrays."""
+    obj = {"binary": "binary sigmoid:1", "multiclass": "multiclass",
+           "regression": "regression"}.get(booster.objective,
+                                           booster.objective)
+    lines = ["tree", "version=v3",
+             f"num_class={booster.n_outputs}",
+             "num_tree_per_iteration=%d" % booster.n_outputs,
+             "label_index=0",
+             f"max_feature_idx={booster.n_features - 1}",
+             f"objective={obj}",
+             "feature_names=" + " ".join(booster.feature_names),
+             "feature_infos=" + " ".join(["none"] * booster.n_features), ""]
+    for ti, (tree, w) in enumerate(zip(booster.trees, booster.tree_weights)):
+        internal = np.nonzero(tree.feature >= 0)[0]
+        leaves = np.nonzero(tree.feature < 0)[0]
+        n_int = len(internal)
+        imap = {int(n): i for i, n in enumerate(internal)}
+        lmap = {int(n): int(tree.leaf_index[n]) for n in leaves}
+
+        def child_ref(n):
+            n = int(n)
+            return imap[n] if tree.feature[n] >= 0 else ~lmap[n]
+
+        split_feature, threshold, decision_type = [], [], []
+        left_child, right_child = [], []
+        internal_value, internal_count = [], []
+        cat_boundaries = [0]
+        cat_threshold = []
+        for n in internal:
+            split_feature.append(int(tree.feature[n]))
+            if tree.cat_offset[n] >= 0:
+                decision_type.append(1)  # categorical ==
+                threshold.append(float(len(cat_boundaries) - 1))
+                words = tree.cat_words[tree.cat_offset[n] * 8:
+                                       tree.cat_offset[n] * 8 + 8]
+                cat_threshold.extend(int(x) for x in words)
+                cat_boundaries.append(len(cat_threshold))
+            else:
+                decision_type.append(2)  # numerical <= with default-left
+                threshold.append(float(tree.threshold[n]))
+            left_child.append(child_ref(tree.left[n]))
+            right_child.append(child_ref(tree.right[n]))
+            internal_value.append(float(tree.value[n]))
+            internal_count.append(int(tree.count[n]))
+        leaf_sorted = sorted(leaves, key=lambda n: tree.leaf_index[n])
+        leaf_value = [float(tree.value[n] * tree.shrinkage * w)
+                      for n in leaf_sorted]
+        leaf_count = [int(tree.count[n]) for n in leaf_sorted]
+        lines += [f"Tree={ti}",
+                  f"num_leaves={len(leaves)}",
+                  "num_cat=%d" % (len(cat_boundaries) - 1),
+                  "split_feature=" + " ".join(map(str, split_feature)),
+                  "threshold=" + " ".join(f"{t:.17g}" for t in threshold),
+                  "decision_type=" + " ".join(map(str, decision_type)),
+                  "left_child=" + " ".join(map(str, left_child)),
+                  "right_child=" + " ".join(map(str, right_child)),
+                  "leaf_value=" + " ".join(f"{v:.17g}" for v in leaf_value),
+                  "leaf_count=" + " ".join(map(str, leaf_count)),
+                  "internal_value=" + " ".join(f"{v:.17g}"
+                                               for v in internal_value),
+                  "internal_count=" + " ".join(map(str, internal_count)),
+                  "shrinkage=%g" % booster.trees[ti].shrinkage]
+        if cat_threshold:
+            lines.insert(len(lines) - 1, "cat_boundaries=" +
+                         " ".join(map(str, cat_boundaries)))
+            lines.insert(len(lines) - 1, "cat_threshold=" +
+                         " ".join(map(str, cat_threshold)))
+        lines.append("")
+    lines.append("end of trees")
+    lines.append("")
+    return "\n".join(lines)
+
+
+Booster.to_lightgbm_text = _to_lightgbm_text

@@ -88,3 +88,18 @@ def test_fuzz_all_registered_stages_serialize(tmp_path):
                 assert v1 == v2, f"{name}.{pname}"
     # stages requiring constructor args are allowed, but most must construct
     assert len(skipped) <= max(2, len(all_stages()) // 4), skipped
+
+
+def test_model_equality_helper(tmp_path, binary_df):
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    from mmlspark_amd.utils.model_equality import assert_model_equality
+    m = LightGBMClassifier(numIterations=3, numLeaves=4).fit(binary_df)
+    a, b = os.path.join(tmp_path, "a"), os.path.join(tmp_path, "b")
+    m.save(a)
+    m.save(b)
+    assert_model_equality(a, b)
+    m2 = LightGBMClassifier(numIterations=4, numLeaves=4).fit(binary_df)
+    c = os.path.join(tmp_path, "c")
+    m2.save(c)
+    with pytest.raises(AssertionError):
+        assert_model_equality(a, c)

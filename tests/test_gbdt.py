@@ -220,3 +220,15 @@ def test_init_score_col():
     # LightGBM semantics: booster output EXCLUDES the init margin, so the
     # trees must have learned to compensate the +4 offset downward
     assert float(raw.mean()) < float(raw0.mean()) - 0.5
+
+
+def test_lightgbm_text_export(binary_df):
+    m = LightGBMClassifier(numIterations=5, numLeaves=7).fit(binary_df)
+    txt = m.booster.to_lightgbm_text()
+    assert txt.startswith("tree\nversion=v3")
+    assert txt.count("Tree=") == 5
+    assert "end of trees" in txt
+    assert f"max_feature_idx={10 - 1}" in txt
+    # every tree block carries the required arrays
+    for key in ("split_feature=", "threshold=", "left_child=", "leaf_value="):
+        assert txt.count(key) == 5
