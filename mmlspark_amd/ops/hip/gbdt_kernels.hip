@@ -101,7 +101,42 @@ __global__ void hist_build_fixed_k(const uchar4* __restrict__ binned,
   const long end = min(start + chunk, m);
   constexpr unsigned long long CNT_ONE = 1ull << 44;
 
-  for (long i = start + tid; i < end; i += blockDim.x) {
+  constexpr int ILP = 4;  // probe: +10% via deeper load batching
+  long i = start + tid;
+  for (; i + (ILP - 1) * (long)blockDim.x < end; i += ILP * blockDim.x) {
+    int r[ILP];
+    long long gq[ILP];
+    unsigned long long hq[ILP];
+#pragma unroll
+    for (int u = 0; u < ILP; ++u) r[u] = rows[i + u * blockDim.x];
+#pragma unroll
+    for (int u = 0; u < ILP; ++u) {
+      gq[u] = (long long)llrint((double)grad[r[u]] * scale_g);
+      hq[u] = CNT_ONE
+              | (unsigned long long)llrint((double)hess[r[u]] * scale_h);
+    }
+#pragma unroll
+    for (int q = 0; q < GPB; ++q) {
+      const int grp = gq0 + q;
+      if (grp >= ngroups) break;
+      uchar4 b4[ILP];
+#pragma unroll
+      for (int u = 0; u < ILP; ++u)
+        b4[u] = binned[(size_t)grp * n_rows + r[u]];
+#pragma unroll
+      for (int u = 0; u < ILP; ++u) {
+        const unsigned char bs[4] = {b4[u].x, b4[u].y, b4[u].z, b4[u].w};
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          unsigned long long* cell =
+              &lds64[((q * 4 + j) * n_bins + bs[j]) * 2];
+          atomicAdd(cell + 0, (unsigned long long)gq[u]);
+          atomicAdd(cell + 1, hq[u]);
+        }
+      }
+    }
+  }
+  for (; i < end; i += blockDim.x) {  // tail
     const int r = rows[i];
     const long long gq = (long long)llrint((double)grad[r] * scale_g);
     const unsigned long long hq =
@@ -150,7 +185,10 @@ extern "C" void launch_hist_build_fixed(const void* binned, long n_rows,
   const int n_fblocks = (ngroups + GPB - 1) / GPB;
   long chunks = (2048 + n_fblocks - 1) / n_fblocks;
   long chunk = (m + chunks - 1) / chunks;
-  if (chunk < 1024) chunk = 1024;
+  // every block flushes its whole LDS histogram with GLOBAL atomics, so
+  // small leaves must not shatter into hundreds of row-chunks (profiled:
+  // flush contention made small-leaf calls ~0.3 ms) — keep chunks coarse
+  if (chunk < 16384) chunk = 16384;
   if (chunk > (1l << 19)) chunk = 1l << 19;  // count-field bound (< 2^20)
   chunks = (m + chunk - 1) / chunk;
   dim3 grid((unsigned)chunks, (unsigned)n_fblocks);
