@@ -23,6 +23,9 @@ void launch_hist_build_fixed(const void*, long, const int*, long, const float*,
 void launch_split_scan(const float*, int, long, int, float, float, float,
                        float, float, long, const bool*, float*, float*,
                        hipStream_t);
+void launch_split_scan_fixed(const long long*, int, long, int, float, float,
+                             float, float, float, long, const bool*, float*,
+                             float*, double, double, hipStream_t);
 void launch_partition(const void*, long, const int*, long, int, int, int*,
                       int*, int*, hipStream_t);
 }
@@ -64,7 +67,6 @@ struct GrowCtx {
   torch::Tensor feat_mask;  // bool (nf_pad,) or undefined
   py::object reduce_fn;     // callable(tensor) or None
   bool has_reduce;
-  torch::Tensor inv_scales;  // double (3,)
 };
 
 torch::Tensor build_hist(GrowCtx& ctx, const torch::Tensor& rows,
@@ -87,19 +89,20 @@ torch::Tensor build_hist(GrowCtx& ctx, const torch::Tensor& rows,
 // returns (gain, feat, bin, GL, HL, CL) per histogram in the stack
 std::vector<std::array<double, 6>> scan_pair(GrowCtx& ctx,
                                              const torch::Tensor& hists_i64) {
-  auto histsf = (hists_i64.to(torch::kDouble) * ctx.inv_scales)
-                    .to(torch::kFloat32).contiguous();
-  const long nh = histsf.size(0);
-  const long nf_pad = histsf.size(1);
-  auto scratch = torch::empty({nh, nf_pad, 6}, histsf.options());
-  auto out = torch::empty({nh, 6}, histsf.options());
+  const long nh = hists_i64.size(0);
+  const long nf_pad = hists_i64.size(1);
+  auto fopt = torch::TensorOptions()
+                  .dtype(torch::kFloat32).device(hists_i64.device());
+  auto scratch = torch::empty({nh, nf_pad, 6}, fopt);
+  auto out = torch::empty({nh, 6}, fopt);
   const bool* mask = ctx.feat_mask.defined()
                          ? ctx.feat_mask.data_ptr<bool>() : nullptr;
-  launch_split_scan(histsf.data_ptr<float>(), (int)nh, nf_pad, ctx.n_bins,
-                    (float)ctx.l1, (float)ctx.l2, (float)ctx.min_data,
-                    (float)ctx.min_hess, (float)ctx.min_gain, ctx.nf, mask,
-                    scratch.data_ptr<float>(), out.data_ptr<float>(),
-                    grower_stream());
+  launch_split_scan_fixed(
+      (const long long*)hists_i64.data_ptr<int64_t>(), (int)nh, nf_pad,
+      ctx.n_bins, (float)ctx.l1, (float)ctx.l2, (float)ctx.min_data,
+      (float)ctx.min_hess, (float)ctx.min_gain, ctx.nf, mask,
+      scratch.data_ptr<float>(), out.data_ptr<float>(), 1.0 / ctx.scale_g,
+      1.0 / ctx.scale_h, grower_stream());
   auto host = out.to(torch::kCPU);  // single sync per scan
   auto acc = host.accessor<float, 2>();
   std::vector<std::array<double, 6>> res((size_t)nh);
@@ -144,10 +147,6 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor rows_root,
   if (feat_mask.has_value()) ctx.feat_mask = *feat_mask;
   ctx.reduce_fn = reduce_fn;
   ctx.has_reduce = !reduce_fn.is_none();
-  ctx.inv_scales = torch::tensor({1.0 / scale_g, 1.0 / scale_h, 1.0},
-                                 torch::TensorOptions()
-                                     .dtype(torch::kDouble)
-                                     .device(binned.device()));
 
   py::gil_scoped_release nogil;
 

@@ -539,11 +539,15 @@ extern "C" void launch_bin_matrix(const float* X, const float* ub, long n,
 // ~50 ms/iter vs 8 ms of GPU time).
 // hist: (n_hists, nf_pad, n_bins, 3); scratch: (n_hists, nf_pad, 6);
 // out: (n_hists, 6) = {gain, feat, bin, GL, HL, CL}.
-__global__ void split_scan_k(const float* __restrict__ hist, int n_bins,
+// FIXED=true reads int64 fixed-point histograms directly (scales applied
+// in-kernel) — saves the int64→float conversion launches per split.
+template <bool FIXED>
+__global__ void split_scan_k(const void* __restrict__ hist_v, int n_bins,
                              long nf_pad, float l1, float l2, float min_data,
                              float min_hess, float min_gain, long nf_real,
                              const bool* __restrict__ feat_mask,
-                             float* __restrict__ scratch) {
+                             float* __restrict__ scratch,
+                             double inv_g, double inv_h) {
   const int f = blockIdx.x;
   const int hi = blockIdx.y;
   const int tid = threadIdx.x;
@@ -555,12 +559,21 @@ __global__ void split_scan_k(const float* __restrict__ hist, int n_bins,
     }
     return;
   }
-  const float* H = hist + ((size_t)hi * nf_pad + f) * n_bins * 3;
   __shared__ float sg[256], sh[256], sc[256];
   const bool in = tid < n_bins;
-  sg[tid] = in ? H[tid * 3 + 0] : 0.0f;
-  sh[tid] = in ? H[tid * 3 + 1] : 0.0f;
-  sc[tid] = in ? H[tid * 3 + 2] : 0.0f;
+  if (FIXED) {
+    const long long* H =
+        (const long long*)hist_v + ((size_t)hi * nf_pad + f) * n_bins * 3;
+    sg[tid] = in ? (float)((double)H[tid * 3 + 0] * inv_g) : 0.0f;
+    sh[tid] = in ? (float)((double)H[tid * 3 + 1] * inv_h) : 0.0f;
+    sc[tid] = in ? (float)H[tid * 3 + 2] : 0.0f;
+  } else {
+    const float* H =
+        (const float*)hist_v + ((size_t)hi * nf_pad + f) * n_bins * 3;
+    sg[tid] = in ? H[tid * 3 + 0] : 0.0f;
+    sh[tid] = in ? H[tid * 3 + 1] : 0.0f;
+    sc[tid] = in ? H[tid * 3 + 2] : 0.0f;
+  }
   __syncthreads();
   // Hillis-Steele inclusive scan over 256 slots
 #pragma unroll
@@ -662,9 +675,25 @@ extern "C" void launch_split_scan(const float* hist, int n_hists, long nf_pad,
                                   const bool* feat_mask, float* scratch,
                                   float* out, hipStream_t stream) {
   dim3 grid1((unsigned)nf_pad, (unsigned)n_hists);
-  hipLaunchKernelGGL(split_scan_k, grid1, dim3(256), 0, stream, hist, n_bins,
-                     nf_pad, l1, l2, min_data, min_hess, min_gain, nf_real,
-                     feat_mask, scratch);
+  hipLaunchKernelGGL((split_scan_k<false>), grid1, dim3(256), 0, stream, hist,
+                     n_bins, nf_pad, l1, l2, min_data, min_hess, min_gain,
+                     nf_real, feat_mask, scratch, 1.0, 1.0);
+  hipLaunchKernelGGL(split_reduce_k, dim3((unsigned)n_hists), dim3(256), 0,
+                     stream, scratch, nf_pad, out);
+}
+
+extern "C" void launch_split_scan_fixed(const long long* hist, int n_hists,
+                                        long nf_pad, int n_bins, float l1,
+                                        float l2, float min_data,
+                                        float min_hess, float min_gain,
+                                        long nf_real, const bool* feat_mask,
+                                        float* scratch, float* out,
+                                        double inv_g, double inv_h,
+                                        hipStream_t stream) {
+  dim3 grid1((unsigned)nf_pad, (unsigned)n_hists);
+  hipLaunchKernelGGL((split_scan_k<true>), grid1, dim3(256), 0, stream, hist,
+                     n_bins, nf_pad, l1, l2, min_data, min_hess, min_gain,
+                     nf_real, feat_mask, scratch, inv_g, inv_h);
   hipLaunchKernelGGL(split_reduce_k, dim3((unsigned)n_hists), dim3(256), 0,
                      stream, scratch, nf_pad, out);
 }
