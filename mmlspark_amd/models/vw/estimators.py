@@ -88,6 +88,10 @@ class _VWParams(Params):
     batchSize = Param("batchSize", "SGD minibatch size (GPU hogwild window)",
                       4096, toInt)
     hashSeed = Param("hashSeed", "murmur seed", 0, toInt)
+    adaptive = Param("adaptive", "per-weight adaptive (AdaGrad) rates "
+                     "(--adaptive)", True, toBool)
+    normalized = Param("normalized", "scale updates by running max|x| per "
+                       "weight (--normalized, NAG-style)", False, toBool)
     initialModel = Param("initialModel", "warm-start weight table", None,
                          is_complex=True)
     passThroughArgs = Param("passThroughArgs",
@@ -105,10 +109,14 @@ class _VWParams(Params):
                    "--power_t": "powerT", "-b": "numBits", "--bit_precision":
                    "numBits", "--passes": "numPasses",
                    "--loss_function": "lossFunction"}
+        flags = {"--adaptive": "adaptive", "--normalized": "normalized"}
         while i < len(s):
             if s[i] in mapping and i + 1 < len(s):
                 self.set(mapping[s[i]], s[i + 1])
                 i += 2
+            elif s[i] in flags:
+                self.set(flags[s[i]], True)
+                i += 1
             else:
                 i += 1
 
@@ -145,10 +153,13 @@ class _VWBase(_VWParams, Estimator):
         else:
             w = torch.zeros(tbl, dtype=torch.float32, device=device)
             g = torch.zeros(tbl, dtype=torch.float32, device=device)
+        s_tbl = (torch.zeros(tbl, dtype=torch.float32, device=device)
+                 if self.get("normalized") else None)
 
         loss = self.get("lossFunction") or self._default_loss
         lr = self.get("learningRate")
         l1, l2 = self.get("l1"), self.get("l2")
+        power_t = self.get("powerT") if self.get("adaptive") else 0.0
         bs = self.get("batchSize")
         n = len(df)
         learn_s = 0.0
@@ -161,8 +172,8 @@ class _VWBase(_VWParams, Estimator):
                 sl = slice(int(off[s]), int(off[e]))
                 backend.vw_sgd_minibatch(
                     idx[sl], val[sl], o, labels[s:e], w, g, lr, l2,
-                    self.get("powerT"), loss,
-                    ex_w[s:e] if ex_w is not None else None)
+                    power_t, loss,
+                    ex_w[s:e] if ex_w is not None else None, s_tbl)
             learn_s += time.perf_counter() - t0
             # end-of-pass sync: RCCL all_reduce of weights + accumulators
             t0 = time.perf_counter()
@@ -171,6 +182,8 @@ class _VWBase(_VWParams, Estimator):
                 w /= comm.world_size
                 comm.all_reduce(g)
                 g /= comm.world_size
+                if s_tbl is not None:
+                    comm.all_reduce(s_tbl, op="max")
             if l1 > 0:  # proximal truncation
                 w.copy_(torch.sign(w) * (w.abs() - lr * l1).clamp_min(0))
             multipass_s += time.perf_counter() - t0

@@ -29,7 +29,7 @@ def vw_predict(indices, values, offsets, w_tbl):
 
 
 def vw_sgd_minibatch(indices, values, offsets, labels, w_tbl, g_tbl, lr, l2,
-                     power_t, loss: str, ex_weight=None):
+                     power_t, loss: str, ex_weight=None, s_tbl=None):
     """One pass over the minibatch: adaptive (AdaGrad) sparse updates.
     Collisions accumulate like the GPU kernel's atomics (index_add)."""
     n_ex = offsets.numel() - 1
@@ -48,5 +48,9 @@ def vw_sgd_minibatch(indices, values, offsets, labels, w_tbl, g_tbl, lr, l2,
         scale = torch.rsqrt(G + EPS)
     else:
         scale = (G + EPS) ** (-power_t)
+    if s_tbl is not None:  # --normalized: running max|x| per weight
+        s_tbl.scatter_reduce_(0, il, values.abs(), reduce="amax")
+        sn = s_tbl[il].clamp_min(EPS)
+        scale = scale / sn
     w_tbl.index_add_(0, il, -lr * g * scale)
     return preds

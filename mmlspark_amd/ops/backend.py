@@ -122,16 +122,17 @@ def bin_matrix(X, upper_bounds, n_bins):
 
 # --------------------------------------------------------------------------- VW
 def vw_sgd_minibatch(indices, values, offsets, labels, weights_tbl, adaptive_tbl,
-                     lr, l2, power_t, loss: str, ex_weight=None):
+                     lr, l2, power_t, loss: str, ex_weight=None,
+                     normalize_tbl=None):
     if weights_tbl.is_cuda:
         return _require_ext().vw_sgd_minibatch(
             indices, values, offsets, labels, weights_tbl, adaptive_tbl,
             lr, l2, power_t, {"squared": 0, "logistic": 1, "hinge": 2}[loss],
-            ex_weight)
+            ex_weight, normalize_tbl)
     from ..models.vw import sgd_ref
     return sgd_ref.vw_sgd_minibatch(indices, values, offsets, labels,
                                     weights_tbl, adaptive_tbl, lr, l2,
-                                    power_t, loss, ex_weight)
+                                    power_t, loss, ex_weight, normalize_tbl)
 
 
 def vw_predict(indices, values, offsets, weights_tbl):

@@ -29,8 +29,8 @@ void launch_split_scan(const float*, int, long, int, float, float, float,
                        float, float, long, const bool*, float*, float*,
                        hipStream_t);
 void launch_vw_sgd(const int*, const float*, const long*, const float*,
-                   const float*, float*, float*, float, float, float, int,
-                   long, float*, hipStream_t);
+                   const float*, float*, float*, float*, float, float, float,
+                   int, long, float*, hipStream_t);
 void launch_vw_predict(const int*, const float*, const long*, const float*,
                        long, float*, hipStream_t);
 }
@@ -194,16 +194,18 @@ torch::Tensor vw_sgd_minibatch(torch::Tensor idx, torch::Tensor val,
                                torch::Tensor w_tbl, torch::Tensor g_tbl,
                                double lr, double l2, double power_t,
                                long loss,
-                               c10::optional<torch::Tensor> ex_weight) {
+                               c10::optional<torch::Tensor> ex_weight,
+                               c10::optional<torch::Tensor> s_tbl) {
   CHECK_DEV(w_tbl); CHECK_CONTIG(w_tbl);
   const long n_ex = off.numel() - 1;
   auto preds = torch::zeros({n_ex}, w_tbl.options());
   const float* wptr = ex_weight.has_value() ? ex_weight->data_ptr<float>()
                                             : nullptr;
+  float* sptr = s_tbl.has_value() ? s_tbl->data_ptr<float>() : nullptr;
   launch_vw_sgd(idx.data_ptr<int>(), val.data_ptr<float>(),
                 off.data_ptr<long>(), label.data_ptr<float>(), wptr,
-                w_tbl.data_ptr<float>(), g_tbl.data_ptr<float>(), (float)lr,
-                (float)l2, (float)power_t, (int)loss, n_ex,
+                w_tbl.data_ptr<float>(), g_tbl.data_ptr<float>(), sptr,
+                (float)lr, (float)l2, (float)power_t, (int)loss, n_ex,
                 preds.data_ptr<float>(), cur_stream());
   return preds;
 }

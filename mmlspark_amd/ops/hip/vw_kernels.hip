@@ -31,12 +31,15 @@ __device__ __forceinline__ float dloss(int loss, float pred, float y) {
   return 0.0f;
 }
 
+// s_tbl (nullable): per-weight running max|x| — VW's --normalized scale.
+// atomicMax on the float bit pattern is order-correct for non-negative floats.
 __global__ void vw_sgd_k(const int* __restrict__ idx,
                          const float* __restrict__ val,
                          const long* __restrict__ off,
                          const float* __restrict__ label,
                          const float* __restrict__ ex_weight,
                          float* __restrict__ w_tbl, float* __restrict__ g_tbl,
+                         float* __restrict__ s_tbl,
                          float lr, float l2, float power_t, int loss,
                          long n_ex, float* __restrict__ preds_out) {
   const long wid0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
@@ -61,6 +64,12 @@ __global__ void vw_sgd_k(const int* __restrict__ idx,
       float scale;
       if (power_t == 0.5f) scale = __frsqrt_rn(G + 1e-10f);
       else scale = __powf(G + 1e-10f, -power_t);
+      if (s_tbl) {  // --normalized: divide by running max|x| per weight
+        const float ax = fabsf(x);
+        atomicMax((int*)&s_tbl[i], __float_as_int(ax));
+        const float sn = fmaxf(__int_as_float(((const int*)s_tbl)[i]), ax);
+        if (sn > 0.0f) scale /= sn;
+      }
       atomicAdd(&w_tbl[i], -lr * g * scale);
     }
   }
@@ -69,7 +78,8 @@ __global__ void vw_sgd_k(const int* __restrict__ idx,
 extern "C" void launch_vw_sgd(const int* idx, const float* val,
                               const long* off, const float* label,
                               const float* ex_weight,
-                              float* w_tbl, float* g_tbl, float lr, float l2,
+                              float* w_tbl, float* g_tbl, float* s_tbl,
+                              float lr, float l2,
                               float power_t, int loss, long n_ex,
                               float* preds_out, hipStream_t stream) {
   if (n_ex == 0) return;
@@ -77,8 +87,8 @@ extern "C" void launch_vw_sgd(const int* idx, const float* val,
   long blocks = (waves * WAVE + 255) / 256;
   if (blocks > 4096) blocks = 4096;
   hipLaunchKernelGGL(vw_sgd_k, dim3((unsigned)blocks), dim3(256), 0, stream,
-                     idx, val, off, label, ex_weight, w_tbl, g_tbl, lr, l2,
-                     power_t, loss, n_ex, preds_out);
+                     idx, val, off, label, ex_weight, w_tbl, g_tbl, s_tbl,
+                     lr, l2, power_t, loss, n_ex, preds_out);
 }
 
 __global__ void vw_predict_k(const int* __restrict__ idx,

@@ -155,3 +155,25 @@ def test_contextual_bandit():
     picked = out["prediction"].to_numpy() - 1
     ctx = df["ctx"].to_numpy()
     assert (picked == ctx).mean() > 0.9
+
+
+def test_vw_normalized_adaptive_flags():
+    from sklearn.metrics import roc_auc_score
+    df = _hashed_binary_data(n=3000)
+    # scale one feature's values up 100x — normalized updates should cope
+    for v in df["features"]:
+        v.values[:1] *= 100.0
+    m_norm = VowpalWabbitClassifier(numPasses=5, numBits=16,
+                                    normalized=True).fit(df)
+    p = np.stack(m_norm.transform(df)["probability"].to_numpy())[:, 1]
+    assert roc_auc_score(df["label"], p) > 0.8
+    # plain (non-adaptive) SGD path on well-scaled data
+    df2 = _hashed_binary_data(n=3000, seed=3)
+    m_plain = VowpalWabbitClassifier(numPasses=5, numBits=16, adaptive=False,
+                                     learningRate=0.05).fit(df2)
+    p2 = np.stack(m_plain.transform(df2)["probability"].to_numpy())[:, 1]
+    assert roc_auc_score(df2["label"], p2) > 0.75
+    # arg-string flags parse
+    est = VowpalWabbitClassifier(passThroughArgs="--normalized --adaptive")
+    est._parse_args()
+    assert est.get("normalized") and est.get("adaptive")
