@@ -11,8 +11,6 @@ single source of truth (the stage registry + Param declarations) are:
 from __future__ import annotations
 
 import os
-from typing import get_type_hints
-
 from .param import Param
 from .registry import all_stages
 

@@ -6,7 +6,6 @@ the batched HIP scoring kernels (the reference instead exploded rows into
 Spark and scored per-row UDFs)."""
 from __future__ import annotations
 
-from typing import List
 
 import numpy as np
 import pandas as pd

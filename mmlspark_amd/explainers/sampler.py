@@ -7,7 +7,7 @@ text-token, vector, tabular samplers) and KernelSHAPSampler.scala:44-129
 from __future__ import annotations
 
 import math
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import numpy as np
 

@@ -8,13 +8,12 @@ solve the constrained weighted least squares per target class.  Output per
 row: (n_classes, m+1) — [base_value, phi_1..phi_m] per class."""
 from __future__ import annotations
 
-from typing import List
 
 import numpy as np
 import pandas as pd
 import torch
 
-from ..core.param import Param, toInt, toList, toString
+from ..core.param import Param, toInt, toList
 from ..core.registry import register
 from ..core.schema import matrix_to_vector_column, vector_column_to_matrix
 from .base import LocalExplainer

@@ -12,12 +12,11 @@ from __future__ import annotations
 import concurrent.futures as cf
 import json
 import time
-from typing import Optional
 
 import numpy as np
 import pandas as pd
 
-from ..core.param import Param, toBool, toInt, toList, toString
+from ..core.param import Param, toBool, toInt, toString
 from ..core.pipeline import PipelineModel, Transformer
 from ..core.registry import register
 from .http_schema import HTTPRequestData, HTTPResponseData

@@ -11,7 +11,6 @@ live-Azure-only test strategy)."""
 from __future__ import annotations
 
 import json
-from typing import Optional
 
 import numpy as np
 import pandas as pd

@@ -10,12 +10,11 @@ import glob
 import json
 import os
 import struct
-from typing import Optional
 
 import numpy as np
 import pandas as pd
 
-from ..core.param import Param, toBool, toInt, toString
+from ..core.param import Param, toBool, toInt
 from ..core.pipeline import Transformer
 from ..core.registry import register
 

@@ -9,13 +9,13 @@ affinities; anomalous = low affinity, standardized per tenant so the output
 is in z-score units (higher = more anomalous)."""
 from __future__ import annotations
 
-from typing import Dict, Optional
+from typing import Dict
 
 import numpy as np
 import pandas as pd
 import torch
 
-from ..core.param import Param, toFloat, toInt, toString
+from ..core.param import Param, toFloat, toInt
 from ..core.pipeline import Estimator, Model, Transformer
 from ..core.registry import register
 

@@ -12,7 +12,7 @@ never row-at-a-time JNI like the reference's UDF scoring path.
 """
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import Optional
 
 import numpy as np
 import pandas as pd
@@ -21,8 +21,8 @@ import torch
 from ...core.param import Param, Params, toBool, toFloat, toInt, toString
 from ...core.pipeline import Estimator, Model
 from ...core.registry import register
-from ...core.schema import (features_matrix, find_unused_column,
-                            infer_feature_names, matrix_to_vector_column)
+from ...core.schema import (features_matrix, infer_feature_names,
+                            matrix_to_vector_column)
 from ...parallel.comm import get_comm
 from ...utils.devices import default_device
 from .booster import Booster

@@ -20,7 +20,7 @@ import pandas as pd
 import torch
 
 from ..core.param import Param, toBool, toFloat, toInt, toString
-from ..core.pipeline import Estimator, Model, Transformer
+from ..core.pipeline import Estimator, Model
 from ..core.registry import register
 from ..core.schema import matrix_to_vector_column
 from ..utils.devices import default_device

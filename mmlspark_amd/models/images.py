@@ -8,7 +8,6 @@ device is default, so the same code drives MIOpen-backed GPU kernels on
 MI355X (interpolate/conv2d) — no OpenCV JNI."""
 from __future__ import annotations
 
-from typing import List
 
 import numpy as np
 import pandas as pd

@@ -15,10 +15,10 @@ import numpy as np
 import pandas as pd
 import torch
 
-from ..core.param import Param, toInt, toList
+from ..core.param import Param, toInt
 from ..core.pipeline import Estimator, Model
 from ..core.registry import register
-from ..core.schema import features_matrix, vector_column_to_matrix
+from ..core.schema import features_matrix
 from ..utils.devices import default_device
 
 

@@ -9,7 +9,6 @@ from __future__ import annotations
 
 from typing import List
 
-import torch
 import torch.nn as nn
 
 

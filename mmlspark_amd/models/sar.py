@@ -8,7 +8,6 @@ Dense torch matmuls run on the matrix cores via rocBLAS."""
 from __future__ import annotations
 
 import math
-from typing import Optional
 
 import numpy as np
 import pandas as pd

@@ -4,13 +4,12 @@ FindBestModel/BestModel (FindBestModel.scala:50,134)."""
 from __future__ import annotations
 
 import concurrent.futures as cf
-from typing import List, Optional
 
 import numpy as np
 import pandas as pd
 
-from ..core.param import Param, toInt, toList, toString
-from ..core.pipeline import Estimator, Model, Transformer
+from ..core.param import Param, toInt, toString
+from ..core.pipeline import Estimator, Model
 from ..core.registry import register
 from .train import ComputeModelStatistics
 

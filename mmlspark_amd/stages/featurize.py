@@ -3,7 +3,6 @@ mixed columns → one vector), CleanMissingData, ValueIndexer, DataConversion,
 CountSelector."""
 from __future__ import annotations
 
-from typing import List
 
 import numpy as np
 import pandas as pd

@@ -2,12 +2,10 @@
 (tokenize → ngrams → TF(-IDF)), MultiNGram, PageSplitter."""
 from __future__ import annotations
 
-import math
 import re
 from typing import List
 
 import numpy as np
-import pandas as pd
 
 from ..core.param import Param, toBool, toInt, toList, toString
 from ..core.pipeline import Estimator, Model, Transformer

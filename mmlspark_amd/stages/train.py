@@ -4,12 +4,11 @@
 ComputePerInstanceStatistics."""
 from __future__ import annotations
 
-from typing import Optional
 
 import numpy as np
 import pandas as pd
 
-from ..core.param import Param, toBool, toInt, toList, toString
+from ..core.param import Param, toInt, toList, toString
 from ..core.pipeline import Estimator, Model, Transformer
 from ..core.registry import register
 from .featurize import Featurize
