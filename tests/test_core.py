@@ -103,3 +103,35 @@ def test_model_equality_helper(tmp_path, binary_df):
     m2.save(c)
     with pytest.raises(AssertionError):
         assert_model_equality(a, c)
+
+
+def test_telemetry_events(binary_df):
+    from mmlspark_amd.core.telemetry import recent_events
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    m = LightGBMClassifier(numIterations=2, numLeaves=4).fit(binary_df)
+    m.transform(binary_df.head(5))
+    evts = recent_events()
+    methods = {e["method"] for e in evts if e["className"].startswith("LightGBM")}
+    assert {"constructor", "fit", "transform"} <= methods
+    assert all("buildVersion" in e for e in evts[-3:])
+
+
+def test_timer_stage(binary_df):
+    from mmlspark_amd.core.telemetry import recent_events
+    from mmlspark_amd.stages.basic import DropColumns, Timer
+    t = Timer(stage=DropColumns(cols=["label"]))
+    out = t.transform(binary_df)
+    assert "label" not in out.columns
+    assert any(e["method"] == "timer" for e in recent_events())
+
+
+def test_pipeline_with_estimator_stage(binary_df):
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    from mmlspark_amd.stages.basic import RenameColumn
+    p = Pipeline(stages=[
+        RenameColumn(inputCol="label", outputCol="target"),
+        LightGBMClassifier(labelCol="target", numIterations=3, numLeaves=4),
+    ])
+    pm = p.fit(binary_df)
+    out = pm.transform(binary_df)
+    assert "prediction" in out.columns
