@@ -396,3 +396,43 @@ class AzureSearchWriter(CognitiveServicesBase):
             results[min(i // bs, len(results) - 1)].to_dict()
             for i in range(len(df))]
         return out
+
+
+@register
+class SpeechToText(CognitiveServicesBase):
+    """REST speech recognition (cognitive/.../SpeechToText.scala)."""
+    audioDataCol = Param("audioDataCol", "audio bytes column", "audio")
+    language = Param("language", "recognition language", "en-US")
+    format = Param("format", "simple|detailed", "simple", toString)
+
+    def _row_url(self, row):
+        url = self.get("url")
+        sep = "&" if "?" in url else "?"
+        return (f"{url}{sep}language={self._sv(row, 'language', 'en-US')}"
+                f"&format={self.get('format')}")
+
+    def _payload(self, row):
+        return None  # raw audio body
+
+    def _headers(self, row):
+        h = super()._headers(row)
+        h["Content-Type"] = "audio/wav; codecs=audio/pcm; samplerate=16000"
+        return h
+
+
+@register
+class DocumentTranslator(CognitiveServicesBase):
+    """Batch document translation (cognitive/.../DocumentTranslator.scala):
+    POST a batch job of {source, targets[]} entries."""
+    sourceUrlCol = Param("sourceUrlCol", "source container url column",
+                         "sourceUrl")
+    targetUrlCol = Param("targetUrlCol", "target container url column",
+                         "targetUrl")
+    targetLanguage = Param("targetLanguage", "target language", "en")
+
+    def _payload(self, row):
+        return {"inputs": [{
+            "source": {"sourceUrl": row[self.get("sourceUrlCol")]},
+            "targets": [{"targetUrl": row[self.get("targetUrlCol")],
+                         "language": self._sv(row, "targetLanguage", "en")}],
+        }]}
