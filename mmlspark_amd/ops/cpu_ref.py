@@ -79,21 +79,34 @@ def predict_forest(node_feature: torch.Tensor, node_threshold: torch.Tensor,
     """
     n = X.shape[0]
     n_trees = tree_offsets.numel() - 1
-    end_tree = n_trees if num_iteration < 0 else min(n_trees, start_tree + num_iteration * n_outputs)
-    out = torch.zeros(n, n_outputs, dtype=torch.float32, device=X.device)
-    for t in range(start_tree, end_tree):
-        base = int(tree_offsets[t])
-        idx = torch.full((n,), base, dtype=torch.long, device=X.device)
+    end_tree = n_trees if num_iteration < 0 else min(
+        n_trees, start_tree + num_iteration * n_outputs)
+    T = end_tree - start_tree
+    if T <= 0 or n == 0:
+        return torch.zeros(n, n_outputs, dtype=torch.float32, device=X.device)
+    # traverse ALL trees simultaneously: (n, T) node cursors, one vectorized
+    # step per tree level instead of a python loop per tree
+    bases = tree_offsets[start_tree:end_tree].long().to(X.device)  # (T,)
+    idx = bases.unsqueeze(0).expand(n, T).contiguous()
+    rows = torch.arange(n, device=X.device).unsqueeze(1).expand(n, T)
+    active = node_feature[idx] >= 0
+    while bool(active.any()):
+        f = node_feature[idx].clamp(min=0).long()
+        xv = X[rows, f]
+        go_left = _go_left(xv, idx, node_threshold, cat_offset, cat_words)
+        nxt = torch.where(go_left, node_left[idx], node_right[idx]).long()
+        idx = torch.where(active, nxt + bases.unsqueeze(0), idx)
         active = node_feature[idx] >= 0
-        while bool(active.any()):
-            f = node_feature[idx].clamp(min=0).long()
-            xv = X[torch.arange(n, device=X.device), f]
-            go_left = _go_left(xv, idx, node_threshold, cat_offset, cat_words)
-            nxt = torch.where(go_left, node_left[idx], node_right[idx]).long()
-            idx = torch.where(active, nxt + base, idx)
-            active = node_feature[idx] >= 0
-        w = 1.0 if tree_weights is None else float(tree_weights[t])
-        out[:, t % n_outputs] += w * node_value[idx]
+    vals = node_value[idx]
+    if tree_weights is not None:
+        vals = vals * tree_weights[start_tree:end_tree].unsqueeze(0)
+    out = torch.zeros(n, n_outputs, dtype=torch.float32, device=X.device)
+    if n_outputs == 1:
+        out[:, 0] = vals.sum(dim=1)
+    else:
+        cls = (torch.arange(start_tree, end_tree, device=X.device)
+               % n_outputs)
+        out.index_add_(1, cls, vals)
     return out
 
 
@@ -103,20 +116,20 @@ def predict_leaf(node_feature, node_threshold, node_left, node_right,
     """Per-tree leaf index for each row: (n, n_trees) int32."""
     n = X.shape[0]
     n_trees = tree_offsets.numel() - 1
-    out = torch.zeros(n, n_trees, dtype=torch.int32, device=X.device)
-    for t in range(n_trees):
-        base = int(tree_offsets[t])
-        idx = torch.full((n,), base, dtype=torch.long, device=X.device)
+    if n_trees == 0 or n == 0:
+        return torch.zeros(n, n_trees, dtype=torch.int32, device=X.device)
+    bases = tree_offsets[:-1].long().to(X.device)
+    idx = bases.unsqueeze(0).expand(n, n_trees).contiguous()
+    rows = torch.arange(n, device=X.device).unsqueeze(1).expand(n, n_trees)
+    active = node_feature[idx] >= 0
+    while bool(active.any()):
+        f = node_feature[idx].clamp(min=0).long()
+        xv = X[rows, f]
+        go_left = _go_left(xv, idx, node_threshold, cat_offset, cat_words)
+        nxt = torch.where(go_left, node_left[idx], node_right[idx]).long()
+        idx = torch.where(active, nxt + bases.unsqueeze(0), idx)
         active = node_feature[idx] >= 0
-        while bool(active.any()):
-            f = node_feature[idx].clamp(min=0).long()
-            xv = X[torch.arange(n, device=X.device), f]
-            go_left = _go_left(xv, idx, node_threshold, cat_offset, cat_words)
-            nxt = torch.where(go_left, node_left[idx], node_right[idx]).long()
-            idx = torch.where(active, nxt + base, idx)
-            active = node_feature[idx] >= 0
-        out[:, t] = node_leaf_index[idx]
-    return out
+    return node_leaf_index[idx].to(torch.int32)
 
 
 def bin_matrix(X: torch.Tensor, upper_bounds: torch.Tensor,
