@@ -296,3 +296,46 @@ class LowLatencyGBDTScorer:
                       for p in payloads])
         scores = self.score(X)
         return [{"score": s.tolist()} for s in scores]
+
+
+class DistributedServingServer:
+    """Multi-worker serving (the HTTPSourceV2 distributed shape): one
+    ServingServer per worker plus a head endpoint that serves the aggregated
+    ServiceInfo list for load-balancer discovery (DriverServiceUtils /
+    HTTPSourceStateHolder parity, HTTPSourceV2.scala:133-198,337).
+    Workers here are threads in one process (one per GPU rank in a real
+    deployment); requests go straight to workers, the head only discovers."""
+
+    def __init__(self, handler_factory, n_workers: int = 2,
+                 host: str = "127.0.0.1", base_port: int = 0,
+                 mode: str = "continuous", name: str = "mmlspark-serving"):
+        self.workers = [
+            ServingServer(handler_factory(i), host=host,
+                          port=(base_port + i if base_port else 0),
+                          mode=mode, name=f"{name}-{i}")
+            for i in range(n_workers)]
+        self.head: Optional[ServingServer] = None
+        self.name = name
+        self.host = host
+
+    def start(self):
+        for w in self.workers:
+            w.start()
+
+        def head_handler(payloads):
+            return [self.service_info() for _ in payloads]
+
+        self.head = ServingServer(head_handler, host=self.host, port=0,
+                                  mode="continuous",
+                                  name=f"{self.name}-head").start()
+        return self
+
+    def service_info(self):
+        return {"name": self.name,
+                "workers": [w.service_info() for w in self.workers]}
+
+    def stop(self):
+        for w in self.workers:
+            w.stop()
+        if self.head:
+            self.head.stop()

@@ -151,3 +151,29 @@ def test_serving_p50_latency_cpu(binary_df):
         assert p50 < 0.05, p50  # generous CPU bound; GPU bench asserts harder
     finally:
         srv.stop()
+
+
+def test_distributed_serving_server():
+    from mmlspark_amd.serving.server import DistributedServingServer
+
+    def factory(i):
+        return lambda payloads: [{"worker": i, "y": p["x"] + 1}
+                                 for p in payloads]
+
+    srv = DistributedServingServer(factory, n_workers=3).start()
+    try:
+        info = requests.get(
+            f"http://127.0.0.1:{srv.head.port}/__service_info",
+            timeout=5).json()
+        # head discovery endpoint is itself a server; query its handler
+        r = requests.post(f"http://127.0.0.1:{srv.head.port}/", json={},
+                          timeout=5).json()
+        assert len(r["workers"]) == 3
+        ports = [w["port"] for w in r["workers"]]
+        # every worker answers
+        for i, p in enumerate(ports):
+            rr = requests.post(f"http://127.0.0.1:{p}/", json={"x": i},
+                               timeout=5).json()
+            assert rr == {"worker": i, "y": i + 1}
+    finally:
+        srv.stop()
