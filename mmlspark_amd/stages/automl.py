@@ -67,7 +67,10 @@ class TuneHyperparameters(Estimator):
     pool (the reference runs parallel Spark jobs; TuneHyperparameters.fit:144)."""
     evaluationMetric = Param("evaluationMetric", "metric name", "AUC", toString)
     numFolds = Param("numFolds", "CV folds", 3, toInt)
-    numRuns = Param("numRuns", "sampled configurations", 8, toInt)
+    numRuns = Param("numRuns", "sampled configurations (random mode)", 8, toInt)
+    searchMode = Param("searchMode", "random|grid (grid enumerates the "
+                       "cartesian product of Discrete params)", "random",
+                       toString)
     parallelism = Param("parallelism", "concurrent fits", 4, toInt)
     seed = Param("seed", "sampling seed", 0, toInt)
     labelCol = Param("labelCol", "label column", "label")
@@ -97,11 +100,22 @@ class TuneHyperparameters(Estimator):
         fold_id = rng.integers(0, folds, size=n)
 
         configs = []
-        for _ in range(self.get("numRuns")):
-            base = models[int(rng.integers(0, len(models)))]
-            params = {k: dist.sample(rng) for k, dist in space.items()
-                      if base.hasParam(k)}
-            configs.append((base, params))
+        if self.get("searchMode") == "grid":
+            import itertools
+            keys = [k for k, d in space.items()
+                    if isinstance(d, DiscreteHyperParam)]
+            grids = [space[k].values for k in keys]
+            for base in models:
+                for combo in itertools.product(*grids) if grids else [()]:
+                    params = {k: v for k, v in zip(keys, combo)
+                              if base.hasParam(k)}
+                    configs.append((base, params))
+        else:
+            for _ in range(self.get("numRuns")):
+                base = models[int(rng.integers(0, len(models)))]
+                params = {k: dist.sample(rng) for k, dist in space.items()
+                          if base.hasParam(k)}
+                configs.append((base, params))
 
         def run(cfg):
             base, params = cfg
