@@ -97,8 +97,58 @@ class Booster:
                                     cat_offset=f.get("cat_offset"),
                                     cat_words=f.get("cat_words"))
 
+    def _tree_depth(self) -> int:
+        best = 0
+        for t in self.trees:
+            depth = np.zeros(t.n_nodes, dtype=np.int32)
+            for i in range(t.n_nodes):  # parents precede children
+                if t.feature[i] >= 0:
+                    depth[t.left[i]] = depth[i] + 1
+                    depth[t.right[i]] = depth[i] + 1
+            if t.n_nodes:
+                best = max(best, int(depth.max()))
+        return best
+
+    def expected_value(self) -> float:
+        """Cover-weighted mean output over all trees (+ base score)."""
+        total = float(self.base_score[0]) if self.base_score.size else 0.0
+        for t, w in zip(self.trees, self.tree_weights):
+            vals = t.value * t.shrinkage * float(w)
+            cover = t.count
+
+            def rec(i):
+                if t.feature[i] < 0:
+                    return vals[i]
+                cl = max(cover[t.left[i]], 1e-12)
+                cr = max(cover[t.right[i]], 1e-12)
+                return (rec(t.left[i]) * cl + rec(t.right[i]) * cr) / (cl + cr)
+
+            total += rec(0) if t.n_nodes else 0.0
+        return total
+
     def predict_contrib(self, X: torch.Tensor) -> np.ndarray:
-        """TreeSHAP contributions, (n, n_features+1) with expected value last."""
+        """TreeSHAP contributions, (n, n_features+1) with expected value last.
+        GPU path: tree_shap_k kernel (trees deeper than 24 fall back to CPU)."""
+        if isinstance(X, torch.Tensor) and X.is_cuda and self.trees                 and self._tree_depth() < 24:
+            from ...ops.backend import _require_ext
+            f = self._flat(X.device)
+            # fold per-tree weights into leaf values for the kernel
+            fw = flatten_trees(self.trees)
+            scaled = fw["value"].copy()
+            for t in range(len(self.trees)):
+                s, e = fw["offsets"][t], fw["offsets"][t + 1]
+                scaled[s:e] *= float(self.tree_weights[t])
+            val = torch.from_numpy(scaled).to(X.device)
+            cnt = torch.from_numpy(
+                np.concatenate([t.count for t in self.trees])
+                .astype(np.float32)).to(X.device)
+            out = _require_ext().tree_shap(
+                f["feature"], f["threshold"], f["left"], f["right"], val,
+                cnt, f["offsets"], X.float().contiguous(),
+                f.get("cat_offset"), f.get("cat_words"))
+            res = out.cpu().numpy().astype(np.float64)
+            res[:, -1] = self.expected_value()
+            return res.astype(np.float32)
         Xn = X.cpu().numpy() if isinstance(X, torch.Tensor) else np.asarray(X)
         Xn = Xn.astype(np.float32)
         out = np.zeros((Xn.shape[0], self.n_features + 1), dtype=np.float64)

@@ -33,6 +33,10 @@ void launch_vw_sgd(const int*, const float*, const long*, const float*,
                    int, long, float*, hipStream_t);
 void launch_vw_predict(const int*, const float*, const long*, const float*,
                        long, float*, hipStream_t);
+void launch_tree_shap(const int*, const float*, const int*, const int*,
+                      const float*, const float*, const long*, const int*,
+                      const unsigned*, const float*, long, int, int, float*,
+                      hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -221,6 +225,27 @@ torch::Tensor vw_predict(torch::Tensor idx, torch::Tensor val,
   return out;
 }
 
+torch::Tensor tree_shap(torch::Tensor feat, torch::Tensor thr,
+                        torch::Tensor left, torch::Tensor right,
+                        torch::Tensor val, torch::Tensor cnt,
+                        torch::Tensor offsets, torch::Tensor X,
+                        c10::optional<torch::Tensor> cat_offset,
+                        c10::optional<torch::Tensor> cat_words) {
+  CHECK_DEV(X); CHECK_CONTIG(X);
+  const long n = X.size(0);
+  const long nf = X.size(1);
+  const long n_trees = offsets.numel() - 1;
+  auto out = torch::zeros({n, nf + 1}, X.options().dtype(torch::kFloat32));
+  auto [co, cw] = cat_ptrs(cat_offset, cat_words);
+  launch_tree_shap(feat.data_ptr<int>(), thr.data_ptr<float>(),
+                   left.data_ptr<int>(), right.data_ptr<int>(),
+                   val.data_ptr<float>(), cnt.data_ptr<float>(),
+                   offsets.data_ptr<long>(), co, cw, X.data_ptr<float>(), n,
+                   (int)nf, (int)n_trees, out.data_ptr<float>(),
+                   cur_stream());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hist_build", &hist_build, "per-leaf (feature,bin) grad/hess/count histogram");
   m.def("hist_build_fixed", &hist_build_fixed,
@@ -233,4 +258,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "stable ordered row partition, single sync");
   m.def("vw_sgd_minibatch", &vw_sgd_minibatch, "adaptive sparse SGD minibatch");
   m.def("vw_predict", &vw_predict, "sparse linear predict");
+  m.def("tree_shap", &tree_shap, "path-dependent TreeSHAP contributions");
 }

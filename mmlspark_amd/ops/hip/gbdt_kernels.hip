@@ -697,3 +697,179 @@ extern "C" void launch_split_scan_fixed(const long long* hist, int n_hists,
   hipLaunchKernelGGL(split_reduce_k, dim3((unsigned)n_hists), dim3(256), 0,
                      stream, scratch, nf_pad, out);
 }
+
+// ------------------------------------------------------------- tree SHAP
+// Path-dependent TreeSHAP contributions on GPU (SURVEY hard-part #4: the
+// featuresShap column at scale).  Thread = one (row, tree) pair; the CPU
+// recursion (models/gbdt/tree.py shap_values — additivity-tested) is
+// expressed with an explicit frame stack; per-level path copies live in
+// scratch (bounded by TS_MAXD); contributions land via global f32 atomics.
+// Trees deeper than TS_MAXD fall back to the CPU implementation host-side.
+#define TS_MAXD 24
+
+struct TsPathEl {
+  int d;
+  float z, o, w;
+};
+
+DEV_INLINE void ts_extend(TsPathEl* p, int len, float pz, float po, int pi) {
+  p[len].d = pi;
+  p[len].z = pz;
+  p[len].o = po;
+  p[len].w = (len == 0) ? 1.0f : 0.0f;
+  for (int i = len - 1; i >= 0; --i) {
+    p[i + 1].w += po * p[i].w * (float)(i + 1) / (float)(len + 1);
+    p[i].w = pz * p[i].w * (float)(len - i) / (float)(len + 1);
+  }
+}
+
+DEV_INLINE void ts_unwind(TsPathEl* p, int len, int i) {
+  const float one = p[i].o;
+  const float zero = p[i].z;
+  float n = p[len].w;
+  for (int j = len - 1; j >= 0; --j) {
+    if (one != 0.0f) {
+      const float t = p[j].w;
+      p[j].w = n * (float)(len + 1) / ((float)(j + 1) * one);
+      n = t - p[j].w * zero * (float)(len - j) / (float)(len + 1);
+    } else {
+      p[j].w = p[j].w * (float)(len + 1) / (zero * (float)(len - j));
+    }
+  }
+  for (int j = i; j < len; ++j) {
+    p[j].d = p[j + 1].d;
+    p[j].z = p[j + 1].z;
+    p[j].o = p[j + 1].o;
+  }
+}
+
+DEV_INLINE float ts_unwound_sum(const TsPathEl* p, int len, int i) {
+  const float one = p[i].o;
+  const float zero = p[i].z;
+  float total = 0.0f;
+  float n = p[len].w;
+  for (int j = len - 1; j >= 0; --j) {
+    if (one != 0.0f) {
+      const float t = n * (float)(len + 1) / ((float)(j + 1) * one);
+      total += t;
+      n = p[j].w - t * zero * (float)(len - j) / (float)(len + 1);
+    } else {
+      total += p[j].w * (float)(len + 1) / (zero * (float)(len - j));
+    }
+  }
+  return total;
+}
+
+struct TsFrame {
+  int node;      // absolute node index
+  int len;       // path length BEFORE this node's extend
+  float pz, po;
+  int pi;
+  int phase;     // 0 = enter, 1 = hot done, 2 = cold done
+  float iz, io;
+  int hot, cold; // absolute child indices
+  int feat;
+  float rn;
+};
+
+__global__ void tree_shap_k(const int* __restrict__ feat,
+                            const float* __restrict__ thr,
+                            const int* __restrict__ left,
+                            const int* __restrict__ right,
+                            const float* __restrict__ val,   // leaf values ×w
+                            const float* __restrict__ cnt,
+                            const long* __restrict__ offsets,
+                            const int* __restrict__ catoff,
+                            const unsigned* __restrict__ catw,
+                            const float* __restrict__ X, long n, int nf,
+                            int n_trees, float* __restrict__ out /*(n,nf+1)*/) {
+  const long pair0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  const long n_pairs = n * (long)n_trees;
+  TsPathEl pstore[(TS_MAXD + 1) * (TS_MAXD + 1)];
+  TsFrame fst[TS_MAXD + 1];
+
+  for (long pair = pair0; pair < n_pairs; pair += stride) {
+    const long row = pair / n_trees;
+    const int t = (int)(pair % n_trees);
+    const long base = offsets[t];
+    const float* x = X + row * nf;
+    float* phi = out + row * (nf + 1);
+
+    int sp = 0;
+    fst[0] = {(int)base, 0, 1.0f, 1.0f, -1, 0, 0.f, 0.f, 0, 0, 0, 0.f};
+    while (sp >= 0) {
+      TsFrame& f = fst[sp];
+      TsPathEl* p = pstore + sp * (TS_MAXD + 1);
+      if (f.phase == 0) {
+        if (sp > 0) {  // copy parent's path
+          const TsPathEl* pp = pstore + (sp - 1) * (TS_MAXD + 1);
+          for (int i = 0; i < f.len; ++i) p[i] = pp[i];
+        }
+        ts_extend(p, f.len, f.pz, f.po, f.pi);
+        int len = f.len + 1;
+        const int nd = f.node;
+        const int ft = feat[nd];
+        if (ft < 0) {  // leaf: credit every feature on the path
+          for (int i = 1; i < len; ++i) {
+            const float w = ts_unwound_sum(p, len - 1, i);
+            atomicAdd(&phi[p[i].d], w * (p[i].o - p[i].z) * val[nd]);
+          }
+          --sp;
+          continue;
+        }
+        const float xv = x[ft];
+        const bool gl = go_left_node(xv, nd, thr, catoff, catw);
+        f.feat = ft;
+        f.hot = (int)base + (gl ? left[nd] : right[nd]);
+        f.cold = (int)base + (gl ? right[nd] : left[nd]);
+        f.iz = 1.0f;
+        f.io = 1.0f;
+        for (int k = 1; k < len; ++k) {
+          if (p[k].d == ft) {
+            f.iz = p[k].z;
+            f.io = p[k].o;
+            ts_unwind(p, len - 1, k);
+            --len;
+            break;
+          }
+        }
+        f.len = len;  // path length owned by this frame (post-unwind)
+        f.rn = fmaxf(cnt[nd], 1e-12f);
+        f.phase = 1;
+        if (sp + 1 <= TS_MAXD) {
+          fst[sp + 1] = {f.hot, len, f.iz * cnt[f.hot] / f.rn, f.io, ft,
+                         0, 0.f, 0.f, 0, 0, 0, 0.f};
+          ++sp;
+        }
+        continue;
+      }
+      if (f.phase == 1) {
+        f.phase = 2;
+        if (sp + 1 <= TS_MAXD) {
+          fst[sp + 1] = {f.cold, f.len, f.iz * cnt[f.cold] / f.rn, 0.0f,
+                         f.feat, 0, 0.f, 0.f, 0, 0, 0, 0.f};
+          ++sp;
+        }
+        continue;
+      }
+      --sp;
+    }
+  }
+}
+
+extern "C" void launch_tree_shap(const int* feat, const float* thr,
+                                 const int* left, const int* right,
+                                 const float* val, const float* cnt,
+                                 const long* offsets, const int* catoff,
+                                 const unsigned* catw, const float* X, long n,
+                                 int nf, int n_trees, float* out,
+                                 hipStream_t stream) {
+  if (n == 0 || n_trees == 0) return;
+  long pairs = n * (long)n_trees;
+  long blocks = (pairs + 255) / 256;
+  if (blocks > 8192) blocks = 8192;
+  hipLaunchKernelGGL(tree_shap_k, dim3((unsigned)blocks), dim3(256), 0,
+                     stream, feat, thr, left, right, val, cnt, offsets,
+                     catoff, catw, X, n, nf, n_trees, out);
+}

@@ -296,3 +296,17 @@ def test_native_grower_distributed_codepath(binary_df):
     assert (feature >= 0).sum() == 14  # 15 leaves → 14 internal nodes
     # leaf segments partition all rows exactly
     assert int(d["leaf_offsets"][-1]) == len(binary_df)
+
+
+@requires_gpu
+def test_tree_shap_gpu_matches_cpu(binary_df):
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    m = LightGBMClassifier(numIterations=10, numLeaves=15,
+                           device="cpu").fit(binary_df)
+    X = torch.from_numpy(np.stack(binary_df["features"].to_numpy()[:64]))
+    cpu = m.booster.predict_contrib(X)
+    gpu = m.booster.predict_contrib(X.cuda())
+    np.testing.assert_allclose(gpu, cpu, atol=2e-3, rtol=1e-3)
+    # additivity through the GPU path
+    raw = m.booster.predict_raw(X).squeeze(-1).numpy()
+    np.testing.assert_allclose(gpu.sum(axis=1), raw, atol=2e-3)
