@@ -35,8 +35,8 @@ void launch_vw_predict(const int*, const float*, const long*, const float*,
                        long, float*, hipStream_t);
 void launch_tree_shap(const int*, const float*, const int*, const int*,
                       const float*, const float*, const long*, const int*,
-                      const unsigned*, const float*, long, int, int, float*,
-                      hipStream_t);
+                      const unsigned*, const float*, long, int, int, int,
+                      float*, hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -229,6 +229,7 @@ torch::Tensor tree_shap(torch::Tensor feat, torch::Tensor thr,
                         torch::Tensor left, torch::Tensor right,
                         torch::Tensor val, torch::Tensor cnt,
                         torch::Tensor offsets, torch::Tensor X,
+                        long max_depth,
                         c10::optional<torch::Tensor> cat_offset,
                         c10::optional<torch::Tensor> cat_words) {
   CHECK_DEV(X); CHECK_CONTIG(X);
@@ -241,8 +242,8 @@ torch::Tensor tree_shap(torch::Tensor feat, torch::Tensor thr,
                    left.data_ptr<int>(), right.data_ptr<int>(),
                    val.data_ptr<float>(), cnt.data_ptr<float>(),
                    offsets.data_ptr<long>(), co, cw, X.data_ptr<float>(), n,
-                   (int)nf, (int)n_trees, out.data_ptr<float>(),
-                   cur_stream());
+                   (int)nf, (int)n_trees, (int)max_depth,
+                   out.data_ptr<float>(), cur_stream());
   return out;
 }
 

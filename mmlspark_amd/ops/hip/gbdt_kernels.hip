@@ -705,8 +705,6 @@ extern "C" void launch_split_scan_fixed(const long long* hist, int n_hists,
 // expressed with an explicit frame stack; per-level path copies live in
 // scratch (bounded by TS_MAXD); contributions land via global f32 atomics.
 // Trees deeper than TS_MAXD fall back to the CPU implementation host-side.
-#define TS_MAXD 24
-
 struct TsPathEl {
   int d;
   float z, o, w;
@@ -772,6 +770,9 @@ struct TsFrame {
   float rn;
 };
 
+// TS_MAXD bounds per-thread scratch ((D+1)^2 path elements); instantiated
+// at 8/16/24 and picked host-side from the ensemble's true max depth.
+template <int TS_MAXD>
 __global__ void tree_shap_k(const int* __restrict__ feat,
                             const float* __restrict__ thr,
                             const int* __restrict__ left,
@@ -863,13 +864,15 @@ extern "C" void launch_tree_shap(const int* feat, const float* thr,
                                  const float* val, const float* cnt,
                                  const long* offsets, const int* catoff,
                                  const unsigned* catw, const float* X, long n,
-                                 int nf, int n_trees, float* out,
-                                 hipStream_t stream) {
+                                 int nf, int n_trees, int max_depth,
+                                 float* out, hipStream_t stream) {
   if (n == 0 || n_trees == 0) return;
   long pairs = n * (long)n_trees;
   long blocks = (pairs + 255) / 256;
   if (blocks > 8192) blocks = 8192;
-  hipLaunchKernelGGL(tree_shap_k, dim3((unsigned)blocks), dim3(256), 0,
-                     stream, feat, thr, left, right, val, cnt, offsets,
-                     catoff, catw, X, n, nf, n_trees, out);
+#define TSLAUNCH(D)                                                        hipLaunchKernelGGL((tree_shap_k<D>), dim3((unsigned)blocks), dim3(256),                      0, stream, feat, thr, left, right, val, cnt, offsets,                      catoff, catw, X, n, nf, n_trees, out)
+  if (max_depth < 8) TSLAUNCH(8);
+  else if (max_depth < 16) TSLAUNCH(16);
+  else TSLAUNCH(24);
+#undef TSLAUNCH
 }
