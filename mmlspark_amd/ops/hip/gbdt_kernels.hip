@@ -401,7 +401,36 @@ __global__ void predict_forest_k(const int* __restrict__ feat,
     const float* x = X + row * nf;
     if (n_outputs == 1) {
       float acc = 0.0f;
-      for (int t = t0; t < t1; ++t) {
+      int t = t0;
+      // traversal is a serial dependent-load chain per tree — walk 4
+      // independent trees at once so their node loads overlap (ILP)
+      for (; t + 3 < t1; t += 4) {
+        long idx[4];
+        long base[4];
+        bool done[4] = {false, false, false, false};
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          base[u] = offsets[t + u];
+          idx[u] = base[u];
+          done[u] = feat[idx[u]] < 0;
+        }
+        while (!(done[0] & done[1] & done[2] & done[3])) {
+#pragma unroll
+          for (int u = 0; u < 4; ++u) {
+            if (!done[u]) {
+              const int f = feat[idx[u]];
+              const float xv = x[f];
+              idx[u] = base[u]
+                       + (go_left_node(xv, idx[u], thr, catoff, catw)
+                              ? left[idx[u]] : right[idx[u]]);
+              done[u] = feat[idx[u]] < 0;
+            }
+          }
+        }
+#pragma unroll
+        for (int u = 0; u < 4; ++u) acc += tw[t + u] * val[idx[u]];
+      }
+      for (; t < t1; ++t) {
         long idx = offsets[t];
         const long base = idx;
         int f = feat[idx];
