@@ -729,78 +729,12 @@ extern "C" void launch_split_scan_fixed(const long long* hist, int n_hists,
 
 // ------------------------------------------------------------- tree SHAP
 // Path-dependent TreeSHAP contributions on GPU (SURVEY hard-part #4: the
-// featuresShap column at scale).  Thread = one (row, tree) pair; the CPU
-// recursion (models/gbdt/tree.py shap_values — additivity-tested) is
-// expressed with an explicit frame stack; per-level path copies live in
-// scratch (bounded by TS_MAXD); contributions land via global f32 atomics.
-// Trees deeper than TS_MAXD fall back to the CPU implementation host-side.
-struct TsPathEl {
-  int d;
-  float z, o, w;
-};
-
-DEV_INLINE void ts_extend(TsPathEl* p, int len, float pz, float po, int pi) {
-  p[len].d = pi;
-  p[len].z = pz;
-  p[len].o = po;
-  p[len].w = (len == 0) ? 1.0f : 0.0f;
-  for (int i = len - 1; i >= 0; --i) {
-    p[i + 1].w += po * p[i].w * (float)(i + 1) / (float)(len + 1);
-    p[i].w = pz * p[i].w * (float)(len - i) / (float)(len + 1);
-  }
-}
-
-DEV_INLINE void ts_unwind(TsPathEl* p, int len, int i) {
-  const float one = p[i].o;
-  const float zero = p[i].z;
-  float n = p[len].w;
-  for (int j = len - 1; j >= 0; --j) {
-    if (one != 0.0f) {
-      const float t = p[j].w;
-      p[j].w = n * (float)(len + 1) / ((float)(j + 1) * one);
-      n = t - p[j].w * zero * (float)(len - j) / (float)(len + 1);
-    } else {
-      p[j].w = p[j].w * (float)(len + 1) / (zero * (float)(len - j));
-    }
-  }
-  for (int j = i; j < len; ++j) {
-    p[j].d = p[j + 1].d;
-    p[j].z = p[j + 1].z;
-    p[j].o = p[j + 1].o;
-  }
-}
-
-DEV_INLINE float ts_unwound_sum(const TsPathEl* p, int len, int i) {
-  const float one = p[i].o;
-  const float zero = p[i].z;
-  float total = 0.0f;
-  float n = p[len].w;
-  for (int j = len - 1; j >= 0; --j) {
-    if (one != 0.0f) {
-      const float t = n * (float)(len + 1) / ((float)(j + 1) * one);
-      total += t;
-      n = p[j].w - t * zero * (float)(len - j) / (float)(len + 1);
-    } else {
-      total += p[j].w * (float)(len + 1) / (zero * (float)(len - j));
-    }
-  }
-  return total;
-}
-
-struct TsFrame {
-  int node;      // absolute node index
-  int len;       // path length BEFORE this node's extend
-  float pz, po;
-  int pi;
-  int phase;     // 0 = enter, 1 = hot done, 2 = cold done
-  float iz, io;
-  int hot, cold; // absolute child indices
-  int feat;
-  float rn;
-};
-
-// TS_MAXD bounds per-thread scratch ((D+1)^2 path elements); instantiated
-// at 8/16/24 and picked host-side from the ensemble's true max depth.
+// featuresShap column at scale).  Thread = one (row, tree) pair doing a DFS
+// that keeps only the CURRENT root→node path (the Shapley path weights are
+// permutation-symmetric, so at each leaf we merge duplicate features and run
+// ONE extend over the merged elements — identical results to the recursive
+// formulation, verified against the CPU reference, with ~500 B of per-thread
+// state instead of per-level path copies).
 template <int TS_MAXD>
 __global__ void tree_shap_k(const int* __restrict__ feat,
                             const float* __restrict__ thr,
@@ -816,8 +750,14 @@ __global__ void tree_shap_k(const int* __restrict__ feat,
   const long pair0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
   const long n_pairs = n * (long)n_trees;
-  TsPathEl pstore[(TS_MAXD + 1) * (TS_MAXD + 1)];
-  TsFrame fst[TS_MAXD + 1];
+
+  // current path (edge elements root→node) + per-leaf merge scratch
+  int pd[TS_MAXD + 2];
+  float pz[TS_MAXD + 2], po[TS_MAXD + 2];
+  int md[TS_MAXD + 2];
+  float mz[TS_MAXD + 2], mo[TS_MAXD + 2], w[TS_MAXD + 2];
+  int nstack[TS_MAXD + 2];
+  signed char phase[TS_MAXD + 2];
 
   for (long pair = pair0; pair < n_pairs; pair += stride) {
     const long row = pair / n_trees;
@@ -826,59 +766,77 @@ __global__ void tree_shap_k(const int* __restrict__ feat,
     const float* x = X + row * nf;
     float* phi = out + row * (nf + 1);
 
-    int sp = 0;
-    fst[0] = {(int)base, 0, 1.0f, 1.0f, -1, 0, 0.f, 0.f, 0, 0, 0, 0.f};
+    int sp = 0;             // stack level == path length
+    nstack[0] = (int)base;
+    phase[0] = 0;
     while (sp >= 0) {
-      TsFrame& f = fst[sp];
-      TsPathEl* p = pstore + sp * (TS_MAXD + 1);
-      if (f.phase == 0) {
-        if (sp > 0) {  // copy parent's path
-          const TsPathEl* pp = pstore + (sp - 1) * (TS_MAXD + 1);
-          for (int i = 0; i < f.len; ++i) p[i] = pp[i];
-        }
-        ts_extend(p, f.len, f.pz, f.po, f.pi);
-        int len = f.len + 1;
-        const int nd = f.node;
-        const int ft = feat[nd];
-        if (ft < 0) {  // leaf: credit every feature on the path
-          for (int i = 1; i < len; ++i) {
-            const float w = ts_unwound_sum(p, len - 1, i);
-            atomicAdd(&phi[p[i].d], w * (p[i].o - p[i].z) * val[nd]);
+      const int nd = nstack[sp];
+      const int ft = feat[nd];
+      if (phase[sp] == 0) {
+        phase[sp] = 1;
+        if (ft < 0) {  // ---- leaf: merge path, extend once, credit features
+          int m = 0;
+          for (int i = 0; i < sp; ++i) {
+            int k = -1;
+            for (int j = 0; j < m; ++j)
+              if (md[j] == pd[i]) { k = j; break; }
+            if (k >= 0) { mz[k] *= pz[i]; mo[k] *= po[i]; }
+            else { md[m] = pd[i]; mz[m] = pz[i]; mo[m] = po[i]; ++m; }
+          }
+          // extend: element 0 is the root sentinel (z=1, o=1)
+          w[0] = 1.0f;
+          for (int e = 0; e < m; ++e) {  // append element e (len = e+1 after)
+            w[e + 1] = 0.0f;
+            for (int i = e; i >= 0; --i) {
+              w[i + 1] += mo[e] * w[i] * (float)(i + 1) / (float)(e + 2);
+              w[i] = mz[e] * w[i] * (float)(e + 1 - i) / (float)(e + 2);
+            }
+          }
+          const float leaf_v = val[nd];
+          for (int i = 0; i < m; ++i) {
+            // unwound sum: remove element i from the extended set
+            float total = 0.0f;
+            float nrun = w[m];
+            if (mo[i] != 0.0f) {
+              for (int j = m - 1; j >= 0; --j) {
+                const float tmp = nrun * (float)(m + 1) / ((float)(j + 1) * mo[i]);
+                total += tmp;
+                nrun = w[j] - tmp * mz[i] * (float)(m - j) / (float)(m + 1);
+              }
+            } else {
+              for (int j = m - 1; j >= 0; --j)
+                total += w[j] * (float)(m + 1) / (mz[i] * (float)(m - j));
+            }
+            atomicAdd(&phi[md[i]], total * (mo[i] - mz[i]) * leaf_v);
           }
           --sp;
           continue;
         }
+        // interior: descend hot child first
         const float xv = x[ft];
         const bool gl = go_left_node(xv, nd, thr, catoff, catw);
-        f.feat = ft;
-        f.hot = (int)base + (gl ? left[nd] : right[nd]);
-        f.cold = (int)base + (gl ? right[nd] : left[nd]);
-        f.iz = 1.0f;
-        f.io = 1.0f;
-        for (int k = 1; k < len; ++k) {
-          if (p[k].d == ft) {
-            f.iz = p[k].z;
-            f.io = p[k].o;
-            ts_unwind(p, len - 1, k);
-            --len;
-            break;
-          }
-        }
-        f.len = len;  // path length owned by this frame (post-unwind)
-        f.rn = fmaxf(cnt[nd], 1e-12f);
-        f.phase = 1;
-        if (sp + 1 <= TS_MAXD) {
-          fst[sp + 1] = {f.hot, len, f.iz * cnt[f.hot] / f.rn, f.io, ft,
-                         0, 0.f, 0.f, 0, 0, 0, 0.f};
+        const int hot = (int)base + (gl ? left[nd] : right[nd]);
+        if (sp < TS_MAXD) {
+          pd[sp] = ft;
+          pz[sp] = cnt[hot] / fmaxf(cnt[nd], 1e-12f);
+          po[sp] = 1.0f;
+          nstack[sp + 1] = hot;
+          phase[sp + 1] = 0;
           ++sp;
         }
         continue;
       }
-      if (f.phase == 1) {
-        f.phase = 2;
-        if (sp + 1 <= TS_MAXD) {
-          fst[sp + 1] = {f.cold, f.len, f.iz * cnt[f.cold] / f.rn, 0.0f,
-                         f.feat, 0, 0.f, 0.f, 0, 0, 0, 0.f};
+      if (phase[sp] == 1) {  // descend cold child
+        phase[sp] = 2;
+        const float xv = x[ft];
+        const bool gl = go_left_node(xv, nd, thr, catoff, catw);
+        const int cold = (int)base + (gl ? right[nd] : left[nd]);
+        if (sp < TS_MAXD) {
+          pd[sp] = ft;
+          pz[sp] = cnt[cold] / fmaxf(cnt[nd], 1e-12f);
+          po[sp] = 0.0f;
+          nstack[sp + 1] = cold;
+          phase[sp + 1] = 0;
           ++sp;
         }
         continue;
@@ -899,9 +857,12 @@ extern "C" void launch_tree_shap(const int* feat, const float* thr,
   long pairs = n * (long)n_trees;
   long blocks = (pairs + 255) / 256;
   if (blocks > 8192) blocks = 8192;
-#define TSLAUNCH(D)                                                        hipLaunchKernelGGL((tree_shap_k<D>), dim3((unsigned)blocks), dim3(256),                      0, stream, feat, thr, left, right, val, cnt, offsets,                      catoff, catw, X, n, nf, n_trees, out)
+#define TSLAUNCH(D)                                                      \
+  hipLaunchKernelGGL((tree_shap_k<D>), dim3((unsigned)blocks), dim3(256), \
+                     0, stream, feat, thr, left, right, val, cnt, offsets, \
+                     catoff, catw, X, n, nf, n_trees, out)
   if (max_depth < 8) TSLAUNCH(8);
   else if (max_depth < 16) TSLAUNCH(16);
-  else TSLAUNCH(24);
+  else TSLAUNCH(32);
 #undef TSLAUNCH
 }
