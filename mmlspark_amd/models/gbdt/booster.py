@@ -128,8 +128,10 @@ class Booster:
 
     def predict_contrib(self, X: torch.Tensor) -> np.ndarray:
         """TreeSHAP contributions, (n, n_features+1) with expected value last.
-        GPU path: tree_shap_k kernel (trees deeper than 24 fall back to CPU)."""
-        if isinstance(X, torch.Tensor) and X.is_cuda and self.trees                 and self._tree_depth() < 24:
+        GPU path: tree_shap_k kernel (trees deeper than 32 fall back to CPU)."""
+        depth = self._tree_depth() if self.trees else 0
+        if isinstance(X, torch.Tensor) and X.is_cuda and self.trees \
+                and depth < 32:
             from ...ops.backend import _require_ext
             f = self._flat(X.device)
             # fold per-tree weights into leaf values for the kernel
@@ -142,11 +144,15 @@ class Booster:
             cnt = torch.from_numpy(
                 np.concatenate([t.count for t in self.trees])
                 .astype(np.float32)).to(X.device)
-            out = _require_ext().tree_shap(
-                f["feature"], f["threshold"], f["left"], f["right"], val,
-                cnt, f["offsets"], X.float().contiguous(),
-                f.get("cat_offset"), f.get("cat_words"))
-            res = out.cpu().numpy().astype(np.float64)
+            Xc = X.float().contiguous()
+            outs = []
+            row_batch = 8192  # bound per-launch runtime
+            for s0 in range(0, Xc.shape[0], row_batch):
+                outs.append(_require_ext().tree_shap(
+                    f["feature"], f["threshold"], f["left"], f["right"], val,
+                    cnt, f["offsets"], Xc[s0:s0 + row_batch], depth,
+                    f.get("cat_offset"), f.get("cat_words")))
+            res = torch.cat(outs).cpu().numpy().astype(np.float64)
             res[:, -1] = self.expected_value()
             return res.astype(np.float32)
         Xn = X.cpu().numpy() if isinstance(X, torch.Tensor) else np.asarray(X)
