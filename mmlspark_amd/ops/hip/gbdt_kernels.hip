@@ -755,7 +755,9 @@ __global__ void tree_shap_k(const int* __restrict__ feat,
   int pd[TS_MAXD + 2];
   float pz[TS_MAXD + 2], po[TS_MAXD + 2];
   int md[TS_MAXD + 2];
-  float mz[TS_MAXD + 2], mo[TS_MAXD + 2], w[TS_MAXD + 2];
+  // deep paths need f64: the Shapley weights carry 1/C(m,k) factors that
+  // underflow f32 beyond ~20 path elements (CPU reference is float64 too)
+  double mz[TS_MAXD + 2], mo[TS_MAXD + 2], w[TS_MAXD + 2];
   int nstack[TS_MAXD + 2];
   signed char phase[TS_MAXD + 2];
 
@@ -784,30 +786,32 @@ __global__ void tree_shap_k(const int* __restrict__ feat,
             else { md[m] = pd[i]; mz[m] = pz[i]; mo[m] = po[i]; ++m; }
           }
           // extend: element 0 is the root sentinel (z=1, o=1)
-          w[0] = 1.0f;
+          w[0] = 1.0;
           for (int e = 0; e < m; ++e) {  // append element e (len = e+1 after)
-            w[e + 1] = 0.0f;
+            w[e + 1] = 0.0;
             for (int i = e; i >= 0; --i) {
-              w[i + 1] += mo[e] * w[i] * (float)(i + 1) / (float)(e + 2);
-              w[i] = mz[e] * w[i] * (float)(e + 1 - i) / (float)(e + 2);
+              w[i + 1] += mo[e] * w[i] * (double)(i + 1) / (double)(e + 2);
+              w[i] = mz[e] * w[i] * (double)(e + 1 - i) / (double)(e + 2);
             }
           }
-          const float leaf_v = val[nd];
+          const double leaf_v = val[nd];
           for (int i = 0; i < m; ++i) {
             // unwound sum: remove element i from the extended set
-            float total = 0.0f;
-            float nrun = w[m];
-            if (mo[i] != 0.0f) {
+            double total = 0.0;
+            double nrun = w[m];
+            if (mo[i] != 0.0) {
               for (int j = m - 1; j >= 0; --j) {
-                const float tmp = nrun * (float)(m + 1) / ((float)(j + 1) * mo[i]);
+                const double tmp =
+                    nrun * (double)(m + 1) / ((double)(j + 1) * mo[i]);
                 total += tmp;
-                nrun = w[j] - tmp * mz[i] * (float)(m - j) / (float)(m + 1);
+                nrun = w[j] - tmp * mz[i] * (double)(m - j) / (double)(m + 1);
               }
             } else {
               for (int j = m - 1; j >= 0; --j)
-                total += w[j] * (float)(m + 1) / (mz[i] * (float)(m - j));
+                total += w[j] * (double)(m + 1) / (mz[i] * (double)(m - j));
             }
-            atomicAdd(&phi[md[i]], total * (mo[i] - mz[i]) * leaf_v);
+            atomicAdd(&phi[md[i]],
+                      (float)(total * (mo[i] - mz[i]) * leaf_v));
           }
           --sp;
           continue;
