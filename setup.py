@@ -40,6 +40,14 @@ def compile_hip_kernels():
 
 extra_objects = compile_hip_kernels()
 
+# setuptools does not track extra_objects as dependencies: if only a .hip
+# kernel changed, the freshly compiled .o would NOT be relinked into the .so.
+# Force a relink by touching the binding sources when any kernel .o is newer.
+for _binding in ("ext.cpp", "gbdt_grower.cpp"):
+    _bp = os.path.join(HIP_DIR, _binding)
+    if any(os.path.getmtime(o) > os.path.getmtime(_bp) for o in extra_objects):
+        os.utime(_bp, None)
+
 setup(
     name="mmlspark_amd_hip_ops",
     ext_modules=[
