@@ -51,12 +51,21 @@ def main():
     if not use_gpu:  # CPU smoke: shrink so the default invocation finishes fast
         n = min(n, 200_000)
 
-    # synthetic Higgs-like binary data, random-init (no network/datasets here)
-    g = torch.Generator(device="cpu").manual_seed(1234 + rank)
-    X = torch.randn(n, nf, generator=g, dtype=torch.float32)
-    w = torch.randn(nf, generator=g) / (nf ** 0.5)
+    # synthetic Higgs-like binary data, random-init (no network/datasets
+    # here); generated directly on the device so an 8-rank launch does not
+    # stage 8×4 GB through host RAM
+    if use_gpu:
+        torch.cuda.manual_seed(1234 + rank)
+        X = torch.randn(n, nf, dtype=torch.float32, device=device)
+        w = torch.randn(nf, device=device) / (nf ** 0.5)
+        noise = 0.5 * torch.randn(n, device=device)
+    else:
+        g = torch.Generator(device="cpu").manual_seed(1234 + rank)
+        X = torch.randn(n, nf, generator=g, dtype=torch.float32)
+        w = torch.randn(nf, generator=g) / (nf ** 0.5)
+        noise = 0.5 * torch.randn(n, generator=g)
     logits = X @ w + 0.3 * torch.sin(3 * X[:, 0]) + 0.3 * X[:, 1] * X[:, 2]
-    y = (logits + 0.5 * torch.randn(n, generator=g) > 0).float()
+    y = (logits + noise > 0).float()
     X = X.to(device)
     y = y.to(device)
 
