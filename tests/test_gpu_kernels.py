@@ -310,3 +310,28 @@ def test_tree_shap_gpu_matches_cpu(binary_df):
     # additivity through the GPU path
     raw = m.booster.predict_raw(X).squeeze(-1).numpy()
     np.testing.assert_allclose(gpu.sum(axis=1), raw, atol=2e-3)
+
+
+@requires_gpu
+def test_contextual_bandit_gpu():
+    import pandas as pd
+    from mmlspark_amd.core.schema import SparseVector
+    from mmlspark_amd.models.vw.estimators import VowpalWabbitContextualBandit
+    rng = np.random.default_rng(2)
+    size = 1 << 14
+    rows = []
+    for _ in range(1500):
+        ctx = int(rng.integers(0, 3))
+        rows.append({
+            "shared": SparseVector(size, [100 + ctx], [1.0]),
+            "features": [SparseVector(size, [2000 + a], [1.0])
+                         for a in range(3)],
+            "chosenAction": int(rng.integers(0, 3)) + 1,
+            "cost": 0.0, "probability": 1 / 3, "ctx": ctx})
+        rows[-1]["cost"] = 0.0 if rows[-1]["chosenAction"] - 1 == ctx else 1.0
+    df = pd.DataFrame(rows)
+    cb = VowpalWabbitContextualBandit(numPasses=5, numBits=14,
+                                      learningRate=0.5, device="cuda").fit(df)
+    out = cb.transform(df)
+    acc = ((out["prediction"].to_numpy() - 1) == df["ctx"].to_numpy()).mean()
+    assert acc > 0.9
