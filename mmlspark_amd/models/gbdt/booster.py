@@ -62,9 +62,15 @@ class Booster:
     def num_iterations(self) -> int:
         return len(self.trees) // max(1, self.n_outputs)
 
+    best_iteration: int = -1  # set by early stopping; used at predict
+
     def predict_raw(self, X: torch.Tensor, start_iteration: int = 0,
                     num_iteration: int = -1) -> torch.Tensor:
-        """Raw margin scores (n, n_outputs)."""
+        """Raw margin scores (n, n_outputs). With early stopping, scoring
+        uses trees up to best_iteration (LightGBM semantics) unless
+        num_iteration overrides."""
+        if num_iteration < 0 and getattr(self, "best_iteration", -1) >= 0:
+            num_iteration = self.best_iteration + 1
         X = X if isinstance(X, torch.Tensor) else torch.as_tensor(X, dtype=torch.float32)
         X = X.float()
         if not self.trees:
@@ -187,6 +193,7 @@ class Booster:
             "feature_names": self.feature_names,
             "sigmoid": self.sigmoid,
             "tree_weights": [float(x) for x in self.tree_weights],
+            "best_iteration": int(getattr(self, "best_iteration", -1)),
             "trees": [t.to_dict() for t in self.trees],
             "bin_mapper": (None if self.bin_mapper is None else {
                 "upper_bounds": self.bin_mapper.upper_bounds.cpu().numpy().tolist(),
@@ -206,7 +213,7 @@ class Booster:
             ub = torch.tensor(d["bin_mapper"]["upper_bounds"], dtype=torch.float32)
             bm = BinMapper(ub, int(d["bin_mapper"]["n_bins"]),
                            d["bin_mapper"].get("categorical"))
-        return Booster(
+        b = Booster(
             trees=[Tree.from_dict(t) for t in d["trees"]],
             objective=d["objective"], n_outputs=int(d["n_outputs"]),
             base_score=np.array(d["base_score"], dtype=np.float32),
@@ -215,6 +222,8 @@ class Booster:
             bin_mapper=bm, sigmoid=d.get("sigmoid", 1.0),
             tree_weights=np.array(d.get("tree_weights", []), dtype=np.float32)
             if d.get("tree_weights") else None)
+        b.best_iteration = int(d.get("best_iteration", -1))
+        return b
 
     @staticmethod
     def load_from_string(s: str) -> "Booster":

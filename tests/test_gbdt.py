@@ -232,3 +232,23 @@ def test_lightgbm_text_export(binary_df):
     # every tree block carries the required arrays
     for key in ("split_feature=", "threshold=", "left_child=", "leaf_value="):
         assert txt.count(key) == 5
+
+
+def test_early_stopping_predicts_with_best_iteration(binary_df):
+    df = binary_df.copy()
+    rng = np.random.default_rng(9)
+    df["isVal"] = rng.random(len(df)) < 0.3
+    m = LightGBMClassifier(numIterations=60, numLeaves=31, learningRate=0.8,
+                           validationIndicatorCol="isVal",
+                           earlyStoppingRound=3).fit(df)
+    b = m.booster
+    assert b.best_iteration >= 0
+    X = torch.from_numpy(np.stack(binary_df["features"].to_numpy()[:50]))
+    default = b.predict_raw(X)
+    best = b.predict_raw(X, num_iteration=b.best_iteration + 1)
+    allt = b.predict_raw(X, num_iteration=b.num_trees)
+    assert torch.allclose(default, best)
+    assert not torch.allclose(default, allt)
+    # persists through native-model text
+    b2 = b.load_from_string(b.save_to_string())
+    assert b2.best_iteration == b.best_iteration
