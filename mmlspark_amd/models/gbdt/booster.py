@@ -115,10 +115,13 @@ class Booster:
                 best = max(best, int(depth.max()))
         return best
 
-    def expected_value(self) -> float:
-        """Cover-weighted mean output over all trees (+ base score)."""
-        total = float(self.base_score[0]) if self.base_score.size else 0.0
-        for t, w in zip(self.trees, self.tree_weights):
+    def expected_value(self) -> np.ndarray:
+        """Cover-weighted mean output per class (+ base score), shape (K,)."""
+        totals = self.base_score.astype(np.float64).copy() \
+            if self.base_score.size else np.zeros(self.n_outputs)
+        if totals.size < self.n_outputs:
+            totals = np.resize(totals, self.n_outputs)
+        for ti, (t, w) in enumerate(zip(self.trees, self.tree_weights)):
             vals = t.value * t.shrinkage * float(w)
             cover = t.count
 
@@ -129,13 +132,18 @@ class Booster:
                 cr = max(cover[t.right[i]], 1e-12)
                 return (rec(t.left[i]) * cl + rec(t.right[i]) * cr) / (cl + cr)
 
-            total += rec(0) if t.n_nodes else 0.0
-        return total
+            totals[ti % self.n_outputs] += rec(0) if t.n_nodes else 0.0
+        return totals
 
     def predict_contrib(self, X: torch.Tensor) -> np.ndarray:
-        """TreeSHAP contributions, (n, n_features+1) with expected value last.
+        """TreeSHAP contributions with the expected value last per class:
+        (n, n_features+1) for single-output models, (n, K*(n_features+1))
+        for multiclass (LightGBM contrib layout).
         GPU path: tree_shap_k kernel (trees deeper than 32 fall back to CPU)."""
         depth = self._tree_depth() if self.trees else 0
+        K = self.n_outputs
+        nf = self.n_features
+        expected = self.expected_value()
         if isinstance(X, torch.Tensor) and X.is_cuda and self.trees \
                 and depth < 32:
             from ...ops.backend import _require_ext
@@ -156,18 +164,21 @@ class Booster:
             for s0 in range(0, Xc.shape[0], row_batch):
                 outs.append(_require_ext().tree_shap(
                     f["feature"], f["threshold"], f["left"], f["right"], val,
-                    cnt, f["offsets"], Xc[s0:s0 + row_batch], depth,
+                    cnt, f["offsets"], Xc[s0:s0 + row_batch], K, depth,
                     f.get("cat_offset"), f.get("cat_words")))
             res = torch.cat(outs).cpu().numpy().astype(np.float64)
-            res[:, -1] = self.expected_value()
-            return res.astype(np.float32)
+            res = res.reshape(-1, K, nf + 1)
+            res[:, :, -1] = expected[None, :]
+            return res.reshape(-1, K * (nf + 1)).astype(np.float32) if K > 1 \
+                else res[:, 0].astype(np.float32)
         Xn = X.cpu().numpy() if isinstance(X, torch.Tensor) else np.asarray(X)
         Xn = Xn.astype(np.float32)
-        out = np.zeros((Xn.shape[0], self.n_features + 1), dtype=np.float64)
-        for t, w in zip(self.trees, self.tree_weights):
-            out += t.shap_values(Xn, scale=float(w))
-        out[:, -1] += float(self.base_score[0]) if self.base_score.size else 0.0
-        return out.astype(np.float32)
+        out = np.zeros((Xn.shape[0], K, nf + 1), dtype=np.float64)
+        for ti, (t, w) in enumerate(zip(self.trees, self.tree_weights)):
+            out[:, ti % K] += t.shap_values(Xn, scale=float(w))
+        out[:, :, -1] = expected[None, :]
+        return out.reshape(-1, K * (nf + 1)).astype(np.float32) if K > 1 \
+            else out[:, 0].astype(np.float32)
 
     # ------------------------------------------------------------- importances
     def feature_importances(self, importance_type: str = "split") -> np.ndarray:

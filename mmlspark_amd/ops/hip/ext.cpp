@@ -35,7 +35,7 @@ void launch_vw_predict(const int*, const float*, const long*, const float*,
                        long, float*, hipStream_t);
 void launch_tree_shap(const int*, const float*, const int*, const int*,
                       const float*, const float*, const long*, const int*,
-                      const unsigned*, const float*, long, int, int, int,
+                      const unsigned*, const float*, long, int, int, int, int,
                       float*, hipStream_t);
 }
 
@@ -229,20 +229,21 @@ torch::Tensor tree_shap(torch::Tensor feat, torch::Tensor thr,
                         torch::Tensor left, torch::Tensor right,
                         torch::Tensor val, torch::Tensor cnt,
                         torch::Tensor offsets, torch::Tensor X,
-                        long max_depth,
+                        long n_outputs, long max_depth,
                         c10::optional<torch::Tensor> cat_offset,
                         c10::optional<torch::Tensor> cat_words) {
   CHECK_DEV(X); CHECK_CONTIG(X);
   const long n = X.size(0);
   const long nf = X.size(1);
   const long n_trees = offsets.numel() - 1;
-  auto out = torch::zeros({n, nf + 1}, X.options().dtype(torch::kFloat32));
+  auto out = torch::zeros({n, n_outputs * (nf + 1)},
+                          X.options().dtype(torch::kFloat32));
   auto [co, cw] = cat_ptrs(cat_offset, cat_words);
   launch_tree_shap(feat.data_ptr<int>(), thr.data_ptr<float>(),
                    left.data_ptr<int>(), right.data_ptr<int>(),
                    val.data_ptr<float>(), cnt.data_ptr<float>(),
                    offsets.data_ptr<long>(), co, cw, X.data_ptr<float>(), n,
-                   (int)nf, (int)n_trees, (int)max_depth,
+                   (int)nf, (int)n_trees, (int)n_outputs, (int)max_depth,
                    out.data_ptr<float>(), cur_stream());
   return out;
 }

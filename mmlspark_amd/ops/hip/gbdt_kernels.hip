@@ -746,7 +746,8 @@ __global__ void tree_shap_k(const int* __restrict__ feat,
                             const int* __restrict__ catoff,
                             const unsigned* __restrict__ catw,
                             const float* __restrict__ X, long n, int nf,
-                            int n_trees, float* __restrict__ out /*(n,nf+1)*/) {
+                            int n_trees, int n_outputs,
+                            float* __restrict__ out /*(n, n_outputs*(nf+1))*/) {
   const long pair0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
   const long n_pairs = n * (long)n_trees;
@@ -766,7 +767,7 @@ __global__ void tree_shap_k(const int* __restrict__ feat,
     const int t = (int)(pair % n_trees);
     const long base = offsets[t];
     const float* x = X + row * nf;
-    float* phi = out + row * (nf + 1);
+    float* phi = out + (row * n_outputs + (t % n_outputs)) * (nf + 1);
 
     int sp = 0;             // stack level == path length
     nstack[0] = (int)base;
@@ -855,8 +856,9 @@ extern "C" void launch_tree_shap(const int* feat, const float* thr,
                                  const float* val, const float* cnt,
                                  const long* offsets, const int* catoff,
                                  const unsigned* catw, const float* X, long n,
-                                 int nf, int n_trees, int max_depth,
-                                 float* out, hipStream_t stream) {
+                                 int nf, int n_trees, int n_outputs,
+                                 int max_depth, float* out,
+                                 hipStream_t stream) {
   if (n == 0 || n_trees == 0) return;
   long pairs = n * (long)n_trees;
   long blocks = (pairs + 255) / 256;
@@ -864,7 +866,7 @@ extern "C" void launch_tree_shap(const int* feat, const float* thr,
 #define TSLAUNCH(D)                                                      \
   hipLaunchKernelGGL((tree_shap_k<D>), dim3((unsigned)blocks), dim3(256), \
                      0, stream, feat, thr, left, right, val, cnt, offsets, \
-                     catoff, catw, X, n, nf, n_trees, out)
+                     catoff, catw, X, n, nf, n_trees, n_outputs, out)
   if (max_depth < 8) TSLAUNCH(8);
   else if (max_depth < 16) TSLAUNCH(16);
   else TSLAUNCH(32);

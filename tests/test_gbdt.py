@@ -252,3 +252,20 @@ def test_early_stopping_predicts_with_best_iteration(binary_df):
     # persists through native-model text
     b2 = b.load_from_string(b.save_to_string())
     assert b2.best_iteration == b.best_iteration
+
+
+def test_multiclass_shap_contrib_layout():
+    rng = np.random.default_rng(3)
+    n = 1500
+    X = rng.normal(size=(n, 6)).astype(np.float32)
+    y = np.digitize(X[:, 0] + X[:, 1], [-1.0, 1.0]).astype(np.float32)
+    df = pd.DataFrame({"features": list(X), "label": y})
+    m = LightGBMClassifier(objective="multiclass", numIterations=8,
+                           numLeaves=7).fit(df)
+    contrib = m.booster.predict_contrib(torch.from_numpy(X[:20]))
+    K, nf = 3, 6
+    assert contrib.shape == (20, K * (nf + 1))
+    per = contrib.reshape(20, K, nf + 1)
+    raw = m.booster.predict_raw(torch.from_numpy(X[:20])).numpy()
+    # per-class additivity: sum of class contributions == class raw score
+    np.testing.assert_allclose(per.sum(axis=2), raw, atol=1e-3)
