@@ -128,3 +128,46 @@ def test_image_superpixel_and_lime():
     vals, counts = np.unique(corner_seg, return_counts=True)
     main_seg = int(vals[counts.argmax()])
     assert exp[main_seg] == exp.max()
+
+
+def test_vector_lime(model_and_data):
+    from mmlspark_amd.explainers.lime import VectorLIME
+    model, df, cols = model_and_data
+    X = df[cols].to_numpy(dtype=np.float32)
+    vdf = pd.DataFrame({"features": matrix_to_vector_column(X)})
+    vmodel = LightGBMClassifier(numIterations=10, numLeaves=7).fit(
+        pd.DataFrame({"features": matrix_to_vector_column(X),
+                      "label": df["label"]}))
+    lime = VectorLIME(model=vmodel, targetCol="probability", targetClasses=[1],
+                      numSamples=300,
+                      backgroundData=vdf.head(80))
+    out = lime.transform(vdf.head(3))
+    exp = np.stack(out["explanation"].to_numpy())
+    assert exp.shape == (3, 1, 5)
+    mean_abs = np.abs(exp[:, 0, :]).mean(axis=0)
+    assert mean_abs[:2].min() > mean_abs[2:].max()
+
+
+def test_image_shap():
+    from mmlspark_amd.explainers.shap import ImageSHAP
+
+    class _CornerModel:
+        def transform(self, df):
+            out = df.copy()
+            out["score"] = [float(np.asarray(v)[:8, :8].mean())
+                            for v in df["image"]]
+            return out
+
+    rng = np.random.default_rng(1)
+    img = rng.integers(0, 30, size=(24, 24, 3)).astype(np.uint8)
+    img[:8, :8] = 240
+    df = pd.DataFrame({"image": [img]})
+    shap = ImageSHAP(model=_CornerModel(), targetCol="score",
+                     targetClasses=[0], cellSize=8, numSamples=96)
+    out = shap.transform(df)
+    exp = out["explanation"].iloc[0][0]
+    segs = out["superpixels"].iloc[0]
+    vals, counts = np.unique(segs[:8, :8].flatten(), return_counts=True)
+    main_seg = int(vals[counts.argmax()])
+    # the bright-corner superpixel carries the largest contribution
+    assert exp[1 + main_seg] == exp[1:].max()
