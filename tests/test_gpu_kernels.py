@@ -335,3 +335,19 @@ def test_contextual_bandit_gpu():
     out = cb.transform(df)
     acc = ((out["prediction"].to_numpy() - 1) == df["ctx"].to_numpy()).mean()
     assert acc > 0.9
+
+
+@requires_gpu
+def test_tree_shap_gpu_multiclass():
+    import pandas as pd
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    rng = np.random.default_rng(4)
+    X = rng.normal(size=(1500, 6)).astype(np.float32)
+    y = np.digitize(X[:, 0] + X[:, 1], [-1.0, 1.0]).astype(np.float32)
+    df = pd.DataFrame({"features": list(X), "label": y})
+    m = LightGBMClassifier(objective="multiclass", numIterations=8,
+                           numLeaves=7, device="cpu").fit(df)
+    Xt = torch.from_numpy(X[:20])
+    cpu = m.booster.predict_contrib(Xt)
+    gpu = m.booster.predict_contrib(Xt.cuda())
+    np.testing.assert_allclose(gpu, cpu, atol=2e-3, rtol=1e-2)
