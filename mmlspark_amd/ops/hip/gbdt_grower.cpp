@@ -1,18 +1,22 @@
 // Native leaf-wise tree-growth driver — the C++ runtime piece of the GBDT.
 //
-// The Python grower pays ~150-250 µs of interpreter/dispatch overhead per
-// split (63 splits/tree); this driver runs the entire leaf-wise loop
-// (histogram → all_reduce → fused split scan → readback → ordered partition
-// → children) in C++, calling the same HIP launchers, and returns the
-// finished node arrays + per-leaf row segments.  Multi-rank histogram
-// reduction happens through an optional Python callback (one GIL hop per
-// reduce — the collective dominates).  Categorical features fall back to the
-// Python grower (models/gbdt/trainer.py).
+// Runs the entire leaf-wise loop (ordered partition → child histogram →
+// sibling subtraction → fused split scan → readback) in C++ with ONE Python
+// call per tree.  The split-processing of any heap leaf is independent of
+// processing order (its split was fixed when the leaf was created), so the
+// driver SPECULATIVELY launches the next-best candidate's whole chain while
+// the host waits on the current readback — results are cached on the leaf
+// and consumed when it is popped, keeping exact leaf-wise commit order while
+// pipelining the GPU work (per-job pinned scan buffers + HIP events).
+// Multi-rank mode stays serial (the histogram all_reduce callback and the
+// local-count readback are ordering barriers).  Categorical features fall
+// back to the Python grower (models/gbdt/trainer.py).
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 
 #include <cmath>
+#include <memory>
 #include <queue>
 #include <vector>
 
@@ -20,9 +24,6 @@ extern "C" {
 void launch_hist_build_fixed(const void*, long, const int*, long, const float*,
                              const float*, long long*, int, int, double,
                              double, hipStream_t);
-void launch_split_scan(const float*, int, long, int, float, float, float,
-                       float, float, long, const bool*, float*, float*,
-                       hipStream_t);
 void launch_split_scan_fixed(const long long*, int, long, int, float, float,
                              float, float, float, long, const bool*, float*,
                              float*, double, double, hipStream_t);
@@ -36,25 +37,6 @@ static hipStream_t grower_stream() {
 
 namespace {
 
-struct LeafCand {
-  double gain;
-  long seq;
-  int node_id;
-  int depth;
-  torch::Tensor rows;
-  torch::Tensor hist;   // int64 (nf_pad, nb, 3), globally reduced
-  double G, H, C;
-  double GL, HL, CL;
-  int feat, bin;
-};
-
-struct CandCmp {
-  bool operator()(const LeafCand& a, const LeafCand& b) const {
-    if (a.gain != b.gain) return a.gain < b.gain;  // max-heap by gain
-    return a.seq > b.seq;
-  }
-};
-
 struct GrowCtx {
   torch::Tensor binned;
   long n_rows;
@@ -67,6 +49,42 @@ struct GrowCtx {
   torch::Tensor feat_mask;  // bool (nf_pad,) or undefined
   py::object reduce_fn;     // callable(tensor) or None
   bool has_reduce;
+  bool distributed;
+  torch::Tensor scratch;    // partition block counters (stream-ordered reuse)
+  torch::Tensor total;      // partition left-count (stream-ordered reuse)
+};
+
+struct SplitJob {
+  torch::Tensor rows_l, rows_r, hist_l, hist_r;
+  torch::Tensor scan_host;  // (2,6) f32 pinned
+  hipEvent_t ev = nullptr;
+  bool done = false;
+
+  ~SplitJob() {
+    if (ev) (void)hipEventDestroy(ev);
+  }
+};
+
+struct LeafCand {
+  double gain = 0;
+  long seq = 0;
+  int node_id = 0;
+  int depth = 0;
+  torch::Tensor rows;
+  torch::Tensor hist;   // int64 (nf_pad, nb, 3), globally reduced
+  double G = 0, H = 0, C = 0;
+  double GL = 0, HL = 0, CL = 0;
+  int feat = -1, bin = 0;
+  std::shared_ptr<SplitJob> job;
+};
+
+using CandPtr = std::shared_ptr<LeafCand>;
+
+struct CandCmp {
+  bool operator()(const CandPtr& a, const CandPtr& b) const {
+    if (a->gain != b->gain) return a->gain < b->gain;  // max-heap by gain
+    return a->seq > b->seq;
+  }
 };
 
 torch::Tensor build_hist(GrowCtx& ctx, const torch::Tensor& rows,
@@ -86,9 +104,10 @@ torch::Tensor build_hist(GrowCtx& ctx, const torch::Tensor& rows,
   return hist;
 }
 
-// returns (gain, feat, bin, GL, HL, CL) per histogram in the stack
-std::vector<std::array<double, 6>> scan_pair(GrowCtx& ctx,
-                                             const torch::Tensor& hists_i64) {
+// launch the (2, nf_pad) scan of stacked child histograms; result lands in a
+// pinned host buffer behind an event — no stream sync
+void launch_scan_async(GrowCtx& ctx, const torch::Tensor& hists_i64,
+                       SplitJob& job) {
   const long nh = hists_i64.size(0);
   const long nf_pad = hists_i64.size(1);
   auto fopt = torch::TensorOptions()
@@ -103,12 +122,46 @@ std::vector<std::array<double, 6>> scan_pair(GrowCtx& ctx,
       (float)ctx.min_hess, (float)ctx.min_gain, ctx.nf, mask,
       scratch.data_ptr<float>(), out.data_ptr<float>(), 1.0 / ctx.scale_g,
       1.0 / ctx.scale_h, grower_stream());
-  auto host = out.to(torch::kCPU);  // single sync per scan
-  auto acc = host.accessor<float, 2>();
-  std::vector<std::array<double, 6>> res((size_t)nh);
-  for (long i = 0; i < nh; ++i)
-    for (int j = 0; j < 6; ++j) res[i][j] = acc[i][j];
-  return res;
+  job.scan_host = torch::empty({nh, 6}, torch::TensorOptions()
+                                            .dtype(torch::kFloat32)
+                                            .pinned_memory(true));
+  (void)hipMemcpyAsync(job.scan_host.data_ptr<float>(), out.data_ptr<float>(),
+                       nh * 6 * sizeof(float), hipMemcpyDeviceToHost,
+                       grower_stream());
+  (void)hipEventCreateWithFlags(&job.ev, hipEventDisableTiming);
+  (void)hipEventRecord(job.ev, grower_stream());
+}
+
+// launch a leaf's full split chain (partition → child hists → scan).
+// In distributed mode this blocks on the local left count and the histogram
+// reduce callback, so the caller never speculates there.
+void launch_job(GrowCtx& ctx, LeafCand& leaf, const torch::Tensor& grad,
+                const torch::Tensor& hess) {
+  auto job = std::make_shared<SplitJob>();
+  const long m = leaf.rows.numel();
+  auto out_rows = torch::empty({m}, leaf.rows.options());
+  launch_partition(ctx.binned.data_ptr(), ctx.n_rows,
+                   leaf.rows.data_ptr<int>(), m, leaf.feat, leaf.bin,
+                   out_rows.data_ptr<int>(), ctx.scratch.data_ptr<int>(),
+                   ctx.total.data_ptr<int>(), grower_stream());
+  long nl;
+  if (!ctx.distributed && leaf.C < 1.6e7) {
+    nl = (long)leaf.CL;  // exact integer counts, no sync
+  } else {
+    nl = ctx.total.to(torch::kCPU).item<int>();
+  }
+  job->rows_l = out_rows.slice(0, 0, nl);
+  job->rows_r = out_rows.slice(0, nl, m);
+
+  const double CL = leaf.CL, CR = leaf.C - leaf.CL;
+  const bool left_small = CL <= CR;
+  auto hist_small =
+      build_hist(ctx, left_small ? job->rows_l : job->rows_r, grad, hess);
+  auto hist_big = leaf.hist - hist_small;
+  job->hist_l = left_small ? hist_small : hist_big;
+  job->hist_r = left_small ? hist_big : hist_small;
+  launch_scan_async(ctx, torch::stack({job->hist_l, job->hist_r}), *job);
+  leaf.job = job;
 }
 
 double leaf_output(double G, double H, double l1, double l2,
@@ -120,10 +173,16 @@ double leaf_output(double G, double H, double l1, double l2,
   return w;
 }
 
+bool splittable(const GrowCtx& ctx, const LeafCand& c) {
+  if (!(c.gain > ctx.min_gain) || !std::isfinite(c.gain)) return false;
+  if (ctx.max_depth > 0 && c.depth >= ctx.max_depth) return false;
+  return true;
+}
+
 }  // namespace
 
 // Returns dict with node arrays (CPU int32/f32 tensors), per-leaf rows
-// (device int32, concatenated) + offsets + leaf node ids + leaf values.
+// (device int32, concatenated) + offsets + leaf node ids.
 py::dict grow_tree_native(torch::Tensor binned, torch::Tensor rows_root,
                           torch::Tensor grad, torch::Tensor hess,
                           long n_bins, long nf, double scale_g, double scale_h,
@@ -147,8 +206,12 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor rows_root,
   if (feat_mask.has_value()) ctx.feat_mask = *feat_mask;
   ctx.reduce_fn = reduce_fn;
   ctx.has_reduce = !reduce_fn.is_none();
+  ctx.distributed = distributed;
 
   py::gil_scoped_release nogil;
+
+  ctx.scratch = torch::empty({4096}, rows_root.options().dtype(torch::kInt32));
+  ctx.total = torch::zeros({1}, rows_root.options().dtype(torch::kInt32));
 
   std::vector<int> feature_, thr_bin_, left_, right_, leaf_idx_;
   std::vector<float> value_, count_, gain_;
@@ -166,71 +229,60 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor rows_root,
   const double H0 = (double)sa[1] / scale_h;
   const double C0 = (double)sa[2];
 
-  LeafCand root;
-  root.node_id = new_node();
-  root.depth = 0;
-  root.seq = 0;
-  root.rows = rows_root;
-  root.hist = root_hist;
-  root.G = G0; root.H = H0; root.C = C0;
+  auto root = std::make_shared<LeafCand>();
+  root->node_id = new_node();
+  root->rows = rows_root;
+  root->hist = root_hist;
+  root->G = G0; root->H = H0; root->C = C0;
   {
-    auto r = scan_pair(ctx, root_hist.unsqueeze(0))[0];
-    root.gain = r[0]; root.feat = (int)r[1]; root.bin = (int)r[2];
-    root.GL = r[3]; root.HL = r[4]; root.CL = r[5];
+    // one-time root scan through the same async machinery
+    SplitJob j;
+    launch_scan_async(ctx, root_hist.unsqueeze(0), j);
+    (void)hipEventSynchronize(j.ev);
+    auto a = j.scan_host.accessor<float, 2>();
+    root->gain = a[0][0]; root->feat = (int)a[0][1]; root->bin = (int)a[0][2];
+    root->GL = a[0][3]; root->HL = a[0][4]; root->CL = a[0][5];
   }
-  count_[root.node_id] = (float)C0;
-  value_[root.node_id] = (float)leaf_output(G0, H0, l1, l2, max_delta);
+  count_[root->node_id] = (float)C0;
+  value_[root->node_id] = (float)leaf_output(G0, H0, l1, l2, max_delta);
 
-  std::priority_queue<LeafCand, std::vector<LeafCand>, CandCmp> heap;
-  std::vector<LeafCand> finals;
+  std::priority_queue<CandPtr, std::vector<CandPtr>, CandCmp> heap;
+  std::vector<CandPtr> finals;
   heap.push(root);
   finals.push_back(root);
   long seq = 1;
   int n_leaves = 1;
 
-  auto scratch = torch::empty({4096},
-                              rows_root.options().dtype(torch::kInt32));
-  auto total = torch::zeros({1}, rows_root.options().dtype(torch::kInt32));
-
   while (n_leaves < ctx.num_leaves && !heap.empty()) {
-    LeafCand leaf = heap.top();
+    CandPtr leaf = heap.top();
     heap.pop();
-    if (!(leaf.gain > ctx.min_gain) || !std::isfinite(leaf.gain)) continue;
-    if (ctx.max_depth > 0 && leaf.depth >= ctx.max_depth) continue;
-    // remove from finals
+    if (!splittable(ctx, *leaf)) continue;
     for (size_t i = 0; i < finals.size(); ++i)
-      if (finals[i].node_id == leaf.node_id) {
+      if (finals[i]->node_id == leaf->node_id) {
         finals.erase(finals.begin() + i);
         break;
       }
 
-    const long m = leaf.rows.numel();
-    auto out_rows = torch::empty({m}, leaf.rows.options());
-    launch_partition(ctx.binned.data_ptr(), ctx.n_rows,
-                     leaf.rows.data_ptr<int>(), m, leaf.feat, leaf.bin,
-                     out_rows.data_ptr<int>(), scratch.data_ptr<int>(),
-                     total.data_ptr<int>(), grower_stream());
-    long nl;
-    if (!distributed && leaf.C < 1.6e7) {
-      nl = (long)leaf.CL;  // exact integer counts, no sync
-    } else {
-      nl = total.to(torch::kCPU).item<int>();
+    if (!leaf->job) launch_job(ctx, *leaf, grad, hess);
+    // speculate: pre-launch the next-best candidate's chain so its GPU work
+    // overlaps this readback (exact commit order preserved; the cache is
+    // consumed whenever that leaf is popped)
+    if (!ctx.distributed && n_leaves + 1 < ctx.num_leaves && !heap.empty()) {
+      CandPtr nxt = heap.top();
+      if (!nxt->job && splittable(ctx, *nxt))
+        launch_job(ctx, *nxt, grad, hess);
     }
-    auto rows_l = out_rows.slice(0, 0, nl);
-    auto rows_r = out_rows.slice(0, nl, m);
 
-    const double GL = leaf.GL, HL = leaf.HL, CL = leaf.CL;
-    const double GR = leaf.G - GL, HR = leaf.H - HL, CR = leaf.C - CL;
-    const bool left_small = CL <= CR;
-    auto hist_small = build_hist(ctx, left_small ? rows_l : rows_r, grad, hess);
-    auto hist_big = leaf.hist - hist_small;
-    auto hist_l = left_small ? hist_small : hist_big;
-    auto hist_r = left_small ? hist_big : hist_small;
+    SplitJob& job = *leaf->job;
+    (void)hipEventSynchronize(job.ev);
+    auto a = job.scan_host.accessor<float, 2>();
 
-    const int nid = leaf.node_id;
-    feature_[nid] = leaf.feat;
-    thr_bin_[nid] = leaf.bin;
-    gain_[nid] = (float)leaf.gain;
+    const double GL = leaf->GL, HL = leaf->HL, CL = leaf->CL;
+    const double GR = leaf->G - GL, HR = leaf->H - HL, CR = leaf->C - CL;
+    const int nid = leaf->node_id;
+    feature_[nid] = leaf->feat;
+    thr_bin_[nid] = leaf->bin;
+    gain_[nid] = (float)leaf->gain;
     const int lid = new_node();
     const int rid = new_node();
     left_[nid] = lid;
@@ -240,20 +292,22 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor rows_root,
     value_[lid] = (float)leaf_output(GL, HL, l1, l2, max_delta);
     value_[rid] = (float)leaf_output(GR, HR, l1, l2, max_delta);
 
-    auto pair = scan_pair(ctx, torch::stack({hist_l, hist_r}));
-    LeafCand lc, rc;
-    lc.node_id = lid; rc.node_id = rid;
-    lc.depth = rc.depth = leaf.depth + 1;
-    lc.rows = rows_l; rc.rows = rows_r;
-    lc.hist = hist_l; rc.hist = hist_r;
-    lc.G = GL; lc.H = HL; lc.C = CL;
-    rc.G = GR; rc.H = HR; rc.C = CR;
-    lc.gain = pair[0][0]; lc.feat = (int)pair[0][1]; lc.bin = (int)pair[0][2];
-    lc.GL = pair[0][3]; lc.HL = pair[0][4]; lc.CL = pair[0][5];
-    rc.gain = pair[1][0]; rc.feat = (int)pair[1][1]; rc.bin = (int)pair[1][2];
-    rc.GL = pair[1][3]; rc.HL = pair[1][4]; rc.CL = pair[1][5];
-    lc.seq = seq++;
-    rc.seq = seq++;
+    auto lc = std::make_shared<LeafCand>();
+    auto rc = std::make_shared<LeafCand>();
+    lc->node_id = lid; rc->node_id = rid;
+    lc->depth = rc->depth = leaf->depth + 1;
+    lc->rows = job.rows_l; rc->rows = job.rows_r;
+    lc->hist = job.hist_l; rc->hist = job.hist_r;
+    lc->G = GL; lc->H = HL; lc->C = CL;
+    rc->G = GR; rc->H = HR; rc->C = CR;
+    lc->gain = a[0][0]; lc->feat = (int)a[0][1]; lc->bin = (int)a[0][2];
+    lc->GL = a[0][3]; lc->HL = a[0][4]; lc->CL = a[0][5];
+    rc->gain = a[1][0]; rc->feat = (int)a[1][1]; rc->bin = (int)a[1][2];
+    rc->GL = a[1][3]; rc->HL = a[1][4]; rc->CL = a[1][5];
+    lc->seq = seq++;
+    rc->seq = seq++;
+    leaf->job.reset();
+    leaf->hist = torch::Tensor();  // free parent histogram
     heap.push(lc);
     heap.push(rc);
     finals.push_back(lc);
@@ -263,15 +317,15 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor rows_root,
 
   // leaf ordinals in node order; collect row segments
   std::sort(finals.begin(), finals.end(),
-            [](const LeafCand& a, const LeafCand& b) {
-              return a.node_id < b.node_id;
+            [](const CandPtr& a, const CandPtr& b) {
+              return a->node_id < b->node_id;
             });
   std::vector<torch::Tensor> segs;
   std::vector<int64_t> seg_nodes;
   for (size_t i = 0; i < finals.size(); ++i) {
-    leaf_idx_[finals[i].node_id] = (int)i;
-    segs.push_back(finals[i].rows);
-    seg_nodes.push_back(finals[i].node_id);
+    leaf_idx_[finals[i]->node_id] = (int)i;
+    segs.push_back(finals[i]->rows);
+    seg_nodes.push_back(finals[i]->node_id);
   }
   auto leaf_rows = segs.empty()
                        ? torch::empty({0}, rows_root.options())
