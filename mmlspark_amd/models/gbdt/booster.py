@@ -238,7 +238,13 @@ class Booster:
 
     @staticmethod
     def load_from_string(s: str) -> "Booster":
-        return Booster.from_dict(json.loads(s))
+        """Accepts either this framework's JSON model or stock LightGBM
+        native model text (``tree\\nversion=v3...``) — the reference's
+        setModelString takes LightGBM text (LightGBMClassifier.scala:150),
+        so models trained elsewhere score here unchanged."""
+        if s.lstrip().startswith("{"):
+            return Booster.from_dict(json.loads(s))
+        return _from_lightgbm_text(s)
 
     def merge(self, other: "Booster") -> "Booster":
         """Append another booster's trees (mergeBooster analog,
@@ -300,7 +306,12 @@ def _to_lightgbm_text(booster: "Booster") -> str:
             internal_value.append(float(tree.value[n]))
             internal_count.append(int(tree.count[n]))
         leaf_sorted = sorted(leaves, key=lambda n: tree.leaf_index[n])
-        leaf_value = [float(tree.value[n] * tree.shrinkage * w)
+        # stock LightGBM has no base-score field: it bakes the init score
+        # into the first iteration's leaves (GBDT::AddBias on tree 0) — do
+        # the same so text predictions match predict_raw everywhere
+        bias = (float(booster.base_score[ti % booster.n_outputs])
+                if ti < booster.n_outputs else 0.0)
+        leaf_value = [float(tree.value[n] * tree.shrinkage * w) + bias
                       for n in leaf_sorted]
         leaf_count = [int(tree.count[n]) for n in leaf_sorted]
         lines += [f"Tree={ti}",
@@ -326,6 +337,128 @@ def _to_lightgbm_text(booster: "Booster") -> str:
     lines.append("end of trees")
     lines.append("")
     return "\n".join(lines)
+
+
+def _from_lightgbm_text(s: str) -> "Booster":
+    """Parse LightGBM native model text (v2/v3) into a Booster — the inverse
+    of to_lightgbm_text, and the interop the reference's setModelString
+    expects (LightGBMClassifier.scala:150: users pass stock LightGBM model
+    strings for warm start / scoring).  Internal nodes keep LightGBM's
+    node numbering (node i = internal i), leaves are appended after the
+    internal nodes.  Categorical bitsets wider than 256 bits are rejected
+    (our device bitset is 8x32 bits; max_bin here is <= 255)."""
+    header = {}
+    tree_blocks = []
+    cur = None
+    for line in s.splitlines():
+        line = line.strip()
+        if not line or line == "end of trees":
+            continue
+        if line.startswith("Tree="):
+            cur = {}
+            tree_blocks.append(cur)
+            continue
+        if "=" not in line:
+            if cur is None:
+                header[line] = True  # bare "tree" magic
+            continue
+        k, v = line.split("=", 1)
+        (header if cur is None else cur)[k] = v
+
+    n_outputs = int(header.get("num_class", 1))
+    n_features = int(header["max_feature_idx"]) + 1
+    obj_parts = header.get("objective", "regression").split()
+    objective = obj_parts[0]
+    sigmoid = 1.0
+    for p in obj_parts[1:]:
+        if p.startswith("sigmoid:"):
+            sigmoid = float(p.split(":")[1])
+    feature_names = header.get(
+        "feature_names", " ".join(f"f{i}" for i in range(n_features))).split()
+
+    def ints(block, key, default=""):
+        v = block.get(key, default)
+        return [int(float(x)) for x in v.split()] if v else []
+
+    def floats(block, key, default=""):
+        v = block.get(key, default)
+        return [float(x) for x in v.split()] if v else []
+
+    trees = []
+    for tb in tree_blocks:
+        split_feature = ints(tb, "split_feature")
+        threshold = floats(tb, "threshold")
+        decision_type = ints(tb, "decision_type")
+        lchild = ints(tb, "left_child")
+        rchild = ints(tb, "right_child")
+        leaf_value = floats(tb, "leaf_value")
+        leaf_count = ints(tb, "leaf_count")
+        internal_value = floats(tb, "internal_value")
+        internal_count = ints(tb, "internal_count")
+        split_gain = floats(tb, "split_gain")
+        cat_boundaries = ints(tb, "cat_boundaries")
+        cat_threshold = ints(tb, "cat_threshold")
+        n_int = len(split_feature)
+        n_leaf = len(leaf_value)
+        if n_int == 0:  # single-leaf (stump) tree
+            trees.append(Tree([-1], [0.0], [0], [-1], [-1],
+                              [leaf_value[0] if leaf_value else 0.0],
+                              [leaf_count[0] if leaf_count else 0.0],
+                              [0.0], [0]))
+            continue
+        n_nodes = n_int + n_leaf
+
+        def ref(r):  # LightGBM child ref: >=0 internal, <0 → leaf ~r
+            return r if r >= 0 else n_int + (~r)
+
+        feature = np.full(n_nodes, -1, dtype=np.int32)
+        thr = np.zeros(n_nodes, dtype=np.float32)
+        left = np.full(n_nodes, -1, dtype=np.int32)
+        right = np.full(n_nodes, -1, dtype=np.int32)
+        value = np.zeros(n_nodes, dtype=np.float32)
+        count = np.zeros(n_nodes, dtype=np.float32)
+        gain = np.zeros(n_nodes, dtype=np.float32)
+        leaf_index = np.full(n_nodes, -1, dtype=np.int32)
+        cat_offset = np.full(n_nodes, -1, dtype=np.int32)
+        cat_words = []
+        for i in range(n_int):
+            feature[i] = split_feature[i]
+            left[i] = ref(lchild[i])
+            right[i] = ref(rchild[i])
+            if internal_value:
+                value[i] = internal_value[i]
+            if internal_count:
+                count[i] = internal_count[i]
+            if split_gain:
+                gain[i] = split_gain[i]
+            if decision_type and (decision_type[i] & 1):  # categorical
+                ci = int(threshold[i])
+                lo, hi = cat_boundaries[ci], cat_boundaries[ci + 1]
+                if hi - lo > 8:
+                    raise ValueError(
+                        "categorical bitset wider than 256 bits unsupported")
+                words = cat_threshold[lo:hi] + [0] * (8 - (hi - lo))
+                cat_offset[i] = len(cat_words) // 8
+                cat_words.extend(words)
+            else:
+                thr[i] = threshold[i]
+        for j in range(n_leaf):
+            value[n_int + j] = leaf_value[j]
+            if leaf_count:
+                count[n_int + j] = leaf_count[j]
+            leaf_index[n_int + j] = j
+        trees.append(Tree(feature, thr, np.zeros(n_nodes, dtype=np.int32),
+                          left, right, value, count, gain, leaf_index,
+                          shrinkage=1.0,  # leaf_value already includes it
+                          cat_offset=cat_offset,
+                          cat_words=np.array(cat_words, dtype=np.uint64)
+                          .astype(np.uint32) if cat_words else None))
+
+    return Booster(trees=trees, objective=objective, n_outputs=n_outputs,
+                   base_score=np.zeros(n_outputs, dtype=np.float32),
+                   n_features=n_features, feature_names=feature_names,
+                   bin_mapper=None, sigmoid=sigmoid,
+                   tree_weights=np.ones(len(trees), dtype=np.float32))
 
 
 Booster.to_lightgbm_text = _to_lightgbm_text

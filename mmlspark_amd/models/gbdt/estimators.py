@@ -227,6 +227,19 @@ class _GBDTModelBase(Model):
     def getNativeModel(self) -> str:
         return self.booster.save_to_string()
 
+    @classmethod
+    def loadNativeModelFromString(cls, s: str, **kwargs):
+        """Build a scoring model from a model string — ours (JSON) or stock
+        LightGBM native text (LightGBMClassificationModel.
+        loadNativeModelFromString, LightGBMClassifier.scala:247)."""
+        return cls(booster=Booster.load_from_string(s), **kwargs)
+
+    @classmethod
+    def loadNativeModelFromFile(cls, path: str, **kwargs):
+        """Analog of loadNativeModelFromFile (LightGBMClassifier.scala:240)."""
+        with open(path) as f:
+            return cls(booster=Booster.load_from_string(f.read()), **kwargs)
+
     def saveNativeModel(self, path: str):
         """Analog of saveNativeModel (LightGBMClassifier.scala:185-205)."""
         with open(path, "w") as f:

@@ -234,6 +234,66 @@ def test_lightgbm_text_export(binary_df):
         assert txt.count(key) == 5
 
 
+def test_lightgbm_text_roundtrip(binary_df):
+    """to_lightgbm_text → load_from_string gives identical predictions (the
+    setModelString interop path: stock LightGBM model text loads here)."""
+    from mmlspark_amd.models.gbdt.booster import Booster
+    m = LightGBMClassifier(numIterations=8, numLeaves=15).fit(binary_df)
+    X = torch.from_numpy(np.stack(binary_df["features"].to_numpy()[:200]))
+    txt = m.booster.to_lightgbm_text()
+    b2 = Booster.load_from_string(txt)
+    assert b2.n_features == m.booster.n_features
+    assert b2.num_trees == m.booster.num_trees
+    np.testing.assert_allclose(b2.predict_raw(X).numpy(),
+                               m.booster.predict_raw(X).numpy(),
+                               rtol=1e-5, atol=1e-5)
+
+
+def test_lightgbm_text_roundtrip_multiclass():
+    rng = np.random.default_rng(3)
+    X = rng.normal(size=(1500, 6)).astype(np.float32)
+    y = np.digitize(X[:, 0] + 0.5 * X[:, 1], [-0.5, 0.5]).astype(np.float32)
+    df = pd.DataFrame({"features": list(X), "label": y})
+    from mmlspark_amd.models.gbdt.booster import Booster
+    m = LightGBMClassifier(numIterations=5, numLeaves=7,
+                           objective="multiclass").fit(df)
+    txt = m.booster.to_lightgbm_text()
+    b2 = Booster.load_from_string(txt)
+    assert b2.n_outputs == 3
+    Xt = torch.from_numpy(X[:100])
+    np.testing.assert_allclose(b2.predict_raw(Xt).numpy(),
+                               m.booster.predict_raw(Xt).numpy(),
+                               rtol=1e-5, atol=1e-5)
+
+
+def test_model_string_accepts_stock_lightgbm_text(binary_df):
+    """A hand-written stock-LightGBM v3 model string (the reference's
+    setModelString input format) scores through the full estimator path."""
+    txt = "\n".join([
+        "tree", "version=v3", "num_class=1", "num_tree_per_iteration=1",
+        "label_index=0", "max_feature_idx=9",
+        "objective=binary sigmoid:1",
+        "feature_names=" + " ".join(f"f{i}" for i in range(10)),
+        "feature_infos=" + " ".join(["none"] * 10), "",
+        "Tree=0", "num_leaves=3", "num_cat=0",
+        "split_feature=0 1", "threshold=0.25 -0.5", "decision_type=2 2",
+        "left_child=1 -1", "right_child=-3 -2",
+        "leaf_value=-1.5 0.5 2.0", "leaf_count=40 30 30",
+        "internal_value=0 -0.4", "internal_count=100 70",
+        "shrinkage=0.1", "",
+        "end of trees", ""])
+    m = LightGBMClassificationModel.loadNativeModelFromString(txt)
+    X = np.array([[0.0, -1.0] + [0.0] * 8,    # f0<=.25, f1<=-.5 → leaf0 -1.5
+                  [0.0, 0.0] + [0.0] * 8,     # f0<=.25, f1>-.5  → leaf1 0.5
+                  [1.0, 0.0] + [0.0] * 8],    # f0>.25           → leaf2 2.0
+                 dtype=np.float32)
+    df = pd.DataFrame({"features": list(X)})
+    out = m.transform(df)
+    raw = np.array([v[1] for v in out["rawPrediction"]])
+    np.testing.assert_allclose(raw, [-1.5, 0.5, 2.0], atol=1e-6)
+    assert out["prediction"].tolist() == [0.0, 1.0, 1.0]
+
+
 def test_early_stopping_predicts_with_best_iteration(binary_df):
     df = binary_df.copy()
     rng = np.random.default_rng(9)
