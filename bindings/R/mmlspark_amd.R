@@ -2,6 +2,133 @@
 library(reticulate)
 mmlspark_amd <- import("mmlspark_amd")
 
+ml_access_anomaly <- function(tenantCol = NULL, indexedUserCol = NULL, indexedResCol = NULL, rankParam = NULL, regParam = NULL, maxIter = NULL, complementsetFactor = NULL, negScore = NULL, outputCol = NULL, seed = NULL) {
+  stage <- mmlspark_amd$models$cyber$AccessAnomaly()
+  if (!is.null(tenantCol)) stage$set("tenantCol", tenantCol)
+  if (!is.null(indexedUserCol)) stage$set("indexedUserCol", indexedUserCol)
+  if (!is.null(indexedResCol)) stage$set("indexedResCol", indexedResCol)
+  if (!is.null(rankParam)) stage$set("rankParam", rankParam)
+  if (!is.null(regParam)) stage$set("regParam", regParam)
+  if (!is.null(maxIter)) stage$set("maxIter", maxIter)
+  if (!is.null(complementsetFactor)) stage$set("complementsetFactor", complementsetFactor)
+  if (!is.null(negScore)) stage$set("negScore", negScore)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(seed)) stage$set("seed", seed)
+  stage
+}
+
+ml_access_anomaly_model <- function(tenantCol = NULL, indexedUserCol = NULL, indexedResCol = NULL, outputCol = NULL, userFactors = NULL, resFactors = NULL, scalerStats = NULL) {
+  stage <- mmlspark_amd$models$cyber$AccessAnomalyModel()
+  if (!is.null(tenantCol)) stage$set("tenantCol", tenantCol)
+  if (!is.null(indexedUserCol)) stage$set("indexedUserCol", indexedUserCol)
+  if (!is.null(indexedResCol)) stage$set("indexedResCol", indexedResCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(userFactors)) stage$set("userFactors", userFactors)
+  if (!is.null(resFactors)) stage$set("resFactors", resFactors)
+  if (!is.null(scalerStats)) stage$set("scalerStats", scalerStats)
+  stage
+}
+
+ml_analyze_business_cards <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$AnalyzeBusinessCards()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(imageUrlCol)) stage$set("imageUrlCol", imageUrlCol)
+  if (!is.null(imageBytesCol)) stage$set("imageBytesCol", imageBytesCol)
+  stage
+}
+
+ml_analyze_id_documents <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$AnalyzeIDDocuments()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(imageUrlCol)) stage$set("imageUrlCol", imageUrlCol)
+  if (!is.null(imageBytesCol)) stage$set("imageBytesCol", imageBytesCol)
+  stage
+}
+
+ml_analyze_image <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL, visualFeatures = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$AnalyzeImage()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(imageUrlCol)) stage$set("imageUrlCol", imageUrlCol)
+  if (!is.null(imageBytesCol)) stage$set("imageBytesCol", imageBytesCol)
+  if (!is.null(visualFeatures)) stage$set("visualFeatures", visualFeatures)
+  stage
+}
+
+ml_analyze_invoices <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$AnalyzeInvoices()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(imageUrlCol)) stage$set("imageUrlCol", imageUrlCol)
+  if (!is.null(imageBytesCol)) stage$set("imageBytesCol", imageBytesCol)
+  stage
+}
+
+ml_analyze_layout <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$AnalyzeLayout()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(imageUrlCol)) stage$set("imageUrlCol", imageUrlCol)
+  if (!is.null(imageBytesCol)) stage$set("imageBytesCol", imageBytesCol)
+  stage
+}
+
+ml_analyze_receipts <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$AnalyzeReceipts()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(imageUrlCol)) stage$set("imageUrlCol", imageUrlCol)
+  if (!is.null(imageBytesCol)) stage$set("imageBytesCol", imageBytesCol)
+  stage
+}
+
+ml_azure_search_writer <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, indexDocsCol = NULL, actionType = NULL, batchSize = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$AzureSearchWriter()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(indexDocsCol)) stage$set("indexDocsCol", indexDocsCol)
+  if (!is.null(actionType)) stage$set("actionType", actionType)
+  if (!is.null(batchSize)) stage$set("batchSize", batchSize)
+  stage
+}
+
 ml_best_model <- function(bestModel = NULL, allModelMetrics = NULL, bestModelMetrics = NULL) {
   stage <- mmlspark_amd$stages$automl$BestModel()
   if (!is.null(bestModel)) stage$set("bestModel", bestModel)
@@ -10,9 +137,60 @@ ml_best_model <- function(bestModel = NULL, allModelMetrics = NULL, bestModelMet
   stage
 }
 
+ml_binary_file_reader <- function(pathCol = NULL, bytesCol = NULL) {
+  stage <- mmlspark_amd$io_http$files$BinaryFileReader()
+  if (!is.null(pathCol)) stage$set("pathCol", pathCol)
+  if (!is.null(bytesCol)) stage$set("bytesCol", bytesCol)
+  stage
+}
+
+ml_bing_image_search <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, qCol = NULL, count = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$BingImageSearch()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(qCol)) stage$set("qCol", qCol)
+  if (!is.null(count)) stage$set("count", count)
+  stage
+}
+
+ml_break_sentence <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, toLanguage = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$BreakSentence()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(textCol)) stage$set("textCol", textCol)
+  if (!is.null(toLanguage)) stage$set("toLanguage", toLanguage)
+  stage
+}
+
 ml_cacher <- function(disable = NULL) {
   stage <- mmlspark_amd$stages$basic$Cacher()
   if (!is.null(disable)) stage$set("disable", disable)
+  stage
+}
+
+ml_class_balancer <- function(inputCol = NULL, outputCol = NULL, broadcastJoin = NULL) {
+  stage <- mmlspark_amd$stages$basic$ClassBalancer()
+  if (!is.null(inputCol)) stage$set("inputCol", inputCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(broadcastJoin)) stage$set("broadcastJoin", broadcastJoin)
+  stage
+}
+
+ml_class_balancer_model <- function(inputCol = NULL, outputCol = NULL, weights = NULL) {
+  stage <- mmlspark_amd$stages$basic$ClassBalancerModel()
+  if (!is.null(inputCol)) stage$set("inputCol", inputCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(weights)) stage$set("weights", weights)
   stage
 }
 
@@ -30,6 +208,16 @@ ml_clean_missing_data_model <- function(inputCols = NULL, outputCols = NULL, fil
   if (!is.null(inputCols)) stage$set("inputCols", inputCols)
   if (!is.null(outputCols)) stage$set("outputCols", outputCols)
   if (!is.null(fillValues)) stage$set("fillValues", fillValues)
+  stage
+}
+
+ml_complement_access_transformer <- function(tenantCol = NULL, indexedUserCol = NULL, indexedResCol = NULL, complementsetFactor = NULL, seed = NULL) {
+  stage <- mmlspark_amd$models$cyber$ComplementAccessTransformer()
+  if (!is.null(tenantCol)) stage$set("tenantCol", tenantCol)
+  if (!is.null(indexedUserCol)) stage$set("indexedUserCol", indexedUserCol)
+  if (!is.null(indexedResCol)) stage$set("indexedResCol", indexedResCol)
+  if (!is.null(complementsetFactor)) stage$set("complementsetFactor", complementsetFactor)
+  if (!is.null(seed)) stage$set("seed", seed)
   stage
 }
 
@@ -136,6 +324,94 @@ ml_deep_vision_model <- function(inputCol = NULL, outputCol = NULL, batchSize = 
   stage
 }
 
+ml_describe_image <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$DescribeImage()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(imageUrlCol)) stage$set("imageUrlCol", imageUrlCol)
+  if (!is.null(imageBytesCol)) stage$set("imageBytesCol", imageBytesCol)
+  stage
+}
+
+ml_detect <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, toLanguage = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$Detect()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(textCol)) stage$set("textCol", textCol)
+  if (!is.null(toLanguage)) stage$set("toLanguage", toLanguage)
+  stage
+}
+
+ml_detect_entire_series <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, seriesCol = NULL, granularity = NULL, sensitivity = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$DetectEntireSeries()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(seriesCol)) stage$set("seriesCol", seriesCol)
+  if (!is.null(granularity)) stage$set("granularity", granularity)
+  if (!is.null(sensitivity)) stage$set("sensitivity", sensitivity)
+  stage
+}
+
+ml_detect_face <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL, returnFaceAttributes = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$DetectFace()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(imageUrlCol)) stage$set("imageUrlCol", imageUrlCol)
+  if (!is.null(imageBytesCol)) stage$set("imageBytesCol", imageBytesCol)
+  if (!is.null(returnFaceAttributes)) stage$set("returnFaceAttributes", returnFaceAttributes)
+  stage
+}
+
+ml_detect_last_anomaly <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, seriesCol = NULL, granularity = NULL, sensitivity = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$DetectLastAnomaly()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(seriesCol)) stage$set("seriesCol", seriesCol)
+  if (!is.null(granularity)) stage$set("granularity", granularity)
+  if (!is.null(sensitivity)) stage$set("sensitivity", sensitivity)
+  stage
+}
+
+ml_document_translator <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, sourceUrlCol = NULL, targetUrlCol = NULL, targetLanguage = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$DocumentTranslator()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(sourceUrlCol)) stage$set("sourceUrlCol", sourceUrlCol)
+  if (!is.null(targetUrlCol)) stage$set("targetUrlCol", targetUrlCol)
+  if (!is.null(targetLanguage)) stage$set("targetLanguage", targetLanguage)
+  stage
+}
+
 ml_drop_columns <- function(cols = NULL) {
   stage <- mmlspark_amd$stages$basic$DropColumns()
   if (!is.null(cols)) stage$set("cols", cols)
@@ -161,6 +437,21 @@ ml_ensemble_by_key <- function(keys = NULL, cols = NULL, strategy = NULL, collap
   if (!is.null(cols)) stage$set("cols", cols)
   if (!is.null(strategy)) stage$set("strategy", strategy)
   if (!is.null(collapseGroup)) stage$set("collapseGroup", collapseGroup)
+  stage
+}
+
+ml_entity_detector <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, language = NULL, languageCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$EntityDetector()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(textCol)) stage$set("textCol", textCol)
+  if (!is.null(language)) stage$set("language", language)
+  if (!is.null(languageCol)) stage$set("languageCol", languageCol)
   stage
 }
 
@@ -195,6 +486,20 @@ ml_find_best_model <- function(evaluationMetric = NULL, labelCol = NULL, models 
   stage
 }
 
+ml_find_similar_face <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, faceIdCol = NULL, faceIdsCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$FindSimilarFace()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(faceIdCol)) stage$set("faceIdCol", faceIdCol)
+  if (!is.null(faceIdsCol)) stage$set("faceIdsCol", faceIdsCol)
+  stage
+}
+
 ml_fixed_mini_batch_transformer <- function(batchSize = NULL, maxBufferSize = NULL) {
   stage <- mmlspark_amd$stages$batching$FixedMiniBatchTransformer()
   if (!is.null(batchSize)) stage$set("batchSize", batchSize)
@@ -208,12 +513,74 @@ ml_flatten_batch <- function() {
   stage
 }
 
+ml_generate_thumbnails <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL, width = NULL, height = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$GenerateThumbnails()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(imageUrlCol)) stage$set("imageUrlCol", imageUrlCol)
+  if (!is.null(imageBytesCol)) stage$set("imageBytesCol", imageBytesCol)
+  if (!is.null(width)) stage$set("width", width)
+  if (!is.null(height)) stage$set("height", height)
+  stage
+}
+
+ml_group_faces <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, faceIdsCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$GroupFaces()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(faceIdsCol)) stage$set("faceIdsCol", faceIdsCol)
+  stage
+}
+
 ml_http_transformer <- function(inputCol = NULL, outputCol = NULL, concurrency = NULL, timeout = NULL) {
   stage <- mmlspark_amd$io_http$client$HTTPTransformer()
   if (!is.null(inputCol)) stage$set("inputCol", inputCol)
   if (!is.null(outputCol)) stage$set("outputCol", outputCol)
   if (!is.null(concurrency)) stage$set("concurrency", concurrency)
   if (!is.null(timeout)) stage$set("timeout", timeout)
+  stage
+}
+
+ml_id_indexer <- function(inputCol = NULL, partitionKey = NULL, outputCol = NULL, resetPerPartition = NULL) {
+  stage <- mmlspark_amd$models$cyber$IdIndexer()
+  if (!is.null(inputCol)) stage$set("inputCol", inputCol)
+  if (!is.null(partitionKey)) stage$set("partitionKey", partitionKey)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(resetPerPartition)) stage$set("resetPerPartition", resetPerPartition)
+  stage
+}
+
+ml_id_indexer_model <- function(inputCol = NULL, partitionKey = NULL, outputCol = NULL, resetPerPartition = NULL, idMaps = NULL) {
+  stage <- mmlspark_amd$models$cyber$IdIndexerModel()
+  if (!is.null(inputCol)) stage$set("inputCol", inputCol)
+  if (!is.null(partitionKey)) stage$set("partitionKey", partitionKey)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(resetPerPartition)) stage$set("resetPerPartition", resetPerPartition)
+  if (!is.null(idMaps)) stage$set("idMaps", idMaps)
+  stage
+}
+
+ml_identify_faces <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, faceIdsCol = NULL, personGroupId = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$IdentifyFaces()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(faceIdsCol)) stage$set("faceIdsCol", faceIdsCol)
+  if (!is.null(personGroupId)) stage$set("personGroupId", personGroupId)
   stage
 }
 
@@ -247,6 +614,14 @@ ml_image_lime <- function(model = NULL, targetCol = NULL, targetClasses = NULL, 
   if (!is.null(cellSize)) stage$set("cellSize", cellSize)
   if (!is.null(modifier)) stage$set("modifier", modifier)
   if (!is.null(superpixelCol)) stage$set("superpixelCol", superpixelCol)
+  stage
+}
+
+ml_image_reader <- function(bytesCol = NULL, imageCol = NULL, dropInvalid = NULL) {
+  stage <- mmlspark_amd$io_http$files$ImageReader()
+  if (!is.null(bytesCol)) stage$set("bytesCol", bytesCol)
+  if (!is.null(imageCol)) stage$set("imageCol", imageCol)
+  if (!is.null(dropInvalid)) stage$set("dropInvalid", dropInvalid)
   stage
 }
 
@@ -357,9 +732,39 @@ ml_knn_model <- function(featuresCol = NULL, valuesCol = NULL, outputCol = NULL,
   stage
 }
 
+ml_key_phrase_extractor <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, language = NULL, languageCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$KeyPhraseExtractor()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(textCol)) stage$set("textCol", textCol)
+  if (!is.null(language)) stage$set("language", language)
+  if (!is.null(languageCol)) stage$set("languageCol", languageCol)
+  stage
+}
+
 ml_lambda <- function() {
   stage <- mmlspark_amd$stages$basic$Lambda()
 
+  stage
+}
+
+ml_language_detector <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, language = NULL, languageCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$LanguageDetector()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(textCol)) stage$set("textCol", textCol)
+  if (!is.null(language)) stage$set("language", language)
+  if (!is.null(languageCol)) stage$set("languageCol", languageCol)
   stage
 }
 
@@ -551,6 +956,27 @@ ml_light_gbm_regressor <- function(labelCol = NULL, featuresCol = NULL, featureC
   stage
 }
 
+ml_linear_scalar_scaler <- function(inputCol = NULL, partitionKey = NULL, outputCol = NULL, minRequiredValue = NULL, maxRequiredValue = NULL) {
+  stage <- mmlspark_amd$models$cyber$LinearScalarScaler()
+  if (!is.null(inputCol)) stage$set("inputCol", inputCol)
+  if (!is.null(partitionKey)) stage$set("partitionKey", partitionKey)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(minRequiredValue)) stage$set("minRequiredValue", minRequiredValue)
+  if (!is.null(maxRequiredValue)) stage$set("maxRequiredValue", maxRequiredValue)
+  stage
+}
+
+ml_linear_scalar_scaler_model <- function(inputCol = NULL, partitionKey = NULL, outputCol = NULL, minRequiredValue = NULL, maxRequiredValue = NULL, stats = NULL) {
+  stage <- mmlspark_amd$models$cyber$LinearScalarScalerModel()
+  if (!is.null(inputCol)) stage$set("inputCol", inputCol)
+  if (!is.null(partitionKey)) stage$set("partitionKey", partitionKey)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(minRequiredValue)) stage$set("minRequiredValue", minRequiredValue)
+  if (!is.null(maxRequiredValue)) stage$set("maxRequiredValue", maxRequiredValue)
+  if (!is.null(stats)) stage$set("stats", stats)
+  stage
+}
+
 ml_multi_column_adapter <- function(baseStage = NULL, inputCols = NULL, outputCols = NULL) {
   stage <- mmlspark_amd$stages$basic$MultiColumnAdapter()
   if (!is.null(baseStage)) stage$set("baseStage", baseStage)
@@ -564,6 +990,35 @@ ml_multi_n_gram <- function(inputCol = NULL, outputCol = NULL, lengths = NULL) {
   if (!is.null(inputCol)) stage$set("inputCol", inputCol)
   if (!is.null(outputCol)) stage$set("outputCol", outputCol)
   if (!is.null(lengths)) stage$set("lengths", lengths)
+  stage
+}
+
+ml_ner <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, language = NULL, languageCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$NER()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(textCol)) stage$set("textCol", textCol)
+  if (!is.null(language)) stage$set("language", language)
+  if (!is.null(languageCol)) stage$set("languageCol", languageCol)
+  stage
+}
+
+ml_ocr <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$OCR()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(imageUrlCol)) stage$set("imageUrlCol", imageUrlCol)
+  if (!is.null(imageBytesCol)) stage$set("imageBytesCol", imageBytesCol)
   stage
 }
 
@@ -593,6 +1048,14 @@ ml_pipeline <- function(stages = NULL) {
 ml_pipeline_model <- function(stages = NULL) {
   stage <- mmlspark_amd$core$pipeline$PipelineModel()
   if (!is.null(stages)) stage$set("stages", stages)
+  stage
+}
+
+ml_power_bi_writer <- function(url = NULL, batchSize = NULL, concurrency = NULL) {
+  stage <- mmlspark_amd$io_http$files$PowerBIWriter()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(batchSize)) stage$set("batchSize", batchSize)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
   stage
 }
 
@@ -632,6 +1095,35 @@ ml_ranking_train_validation_split_model <- function(bestModel = NULL, validation
   stage <- mmlspark_amd$models$sar$RankingTrainValidationSplitModel()
   if (!is.null(bestModel)) stage$set("bestModel", bestModel)
   if (!is.null(validationMetric)) stage$set("validationMetric", validationMetric)
+  stage
+}
+
+ml_recognize_domain_specific_content <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL, model = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$RecognizeDomainSpecificContent()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(imageUrlCol)) stage$set("imageUrlCol", imageUrlCol)
+  if (!is.null(imageBytesCol)) stage$set("imageBytesCol", imageBytesCol)
+  if (!is.null(model)) stage$set("model", model)
+  stage
+}
+
+ml_recognize_text <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$RecognizeText()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(imageUrlCol)) stage$set("imageUrlCol", imageUrlCol)
+  if (!is.null(imageBytesCol)) stage$set("imageBytesCol", imageBytesCol)
   stage
 }
 
@@ -708,6 +1200,62 @@ ml_simple_http_transformer <- function(inputCol = NULL, outputCol = NULL, url = 
   stage
 }
 
+ml_speech_to_text <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, audioDataCol = NULL, language = NULL, format = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$SpeechToText()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(audioDataCol)) stage$set("audioDataCol", audioDataCol)
+  if (!is.null(language)) stage$set("language", language)
+  if (!is.null(format)) stage$set("format", format)
+  stage
+}
+
+ml_speech_to_text_sdk <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, audioBytesCol = NULL, format = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$SpeechToTextSDK()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(audioBytesCol)) stage$set("audioBytesCol", audioBytesCol)
+  if (!is.null(format)) stage$set("format", format)
+  stage
+}
+
+ml_standard_scalar_scaler <- function(inputCol = NULL, partitionKey = NULL, outputCol = NULL, coefficientFactor = NULL) {
+  stage <- mmlspark_amd$models$cyber$StandardScalarScaler()
+  if (!is.null(inputCol)) stage$set("inputCol", inputCol)
+  if (!is.null(partitionKey)) stage$set("partitionKey", partitionKey)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(coefficientFactor)) stage$set("coefficientFactor", coefficientFactor)
+  stage
+}
+
+ml_standard_scalar_scaler_model <- function(inputCol = NULL, partitionKey = NULL, outputCol = NULL, coefficientFactor = NULL, stats = NULL) {
+  stage <- mmlspark_amd$models$cyber$StandardScalarScalerModel()
+  if (!is.null(inputCol)) stage$set("inputCol", inputCol)
+  if (!is.null(partitionKey)) stage$set("partitionKey", partitionKey)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(coefficientFactor)) stage$set("coefficientFactor", coefficientFactor)
+  if (!is.null(stats)) stage$set("stats", stats)
+  stage
+}
+
+ml_stratified_repartition <- function(labelCol = NULL, mode = NULL, seed = NULL) {
+  stage <- mmlspark_amd$stages$basic$StratifiedRepartition()
+  if (!is.null(labelCol)) stage$set("labelCol", labelCol)
+  if (!is.null(mode)) stage$set("mode", mode)
+  if (!is.null(seed)) stage$set("seed", seed)
+  stage
+}
+
 ml_string_output_parser <- function(inputCol = NULL, outputCol = NULL) {
   stage <- mmlspark_amd$io_http$client$StringOutputParser()
   if (!is.null(inputCol)) stage$set("inputCol", inputCol)
@@ -770,6 +1318,20 @@ ml_tabular_shap <- function(model = NULL, targetCol = NULL, targetClasses = NULL
   if (!is.null(rowBatch)) stage$set("rowBatch", rowBatch)
   if (!is.null(inputCols)) stage$set("inputCols", inputCols)
   if (!is.null(backgroundData)) stage$set("backgroundData", backgroundData)
+  stage
+}
+
+ml_tag_image <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$TagImage()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(imageUrlCol)) stage$set("imageUrlCol", imageUrlCol)
+  if (!is.null(imageBytesCol)) stage$set("imageBytesCol", imageBytesCol)
   stage
 }
 
@@ -845,6 +1407,21 @@ ml_text_shap <- function(model = NULL, targetCol = NULL, targetClasses = NULL, o
   stage
 }
 
+ml_text_sentiment <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, language = NULL, languageCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$TextSentiment()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(textCol)) stage$set("textCol", textCol)
+  if (!is.null(language)) stage$set("language", language)
+  if (!is.null(languageCol)) stage$set("languageCol", languageCol)
+  stage
+}
+
 ml_time_interval_mini_batch_transformer <- function(millisToWait = NULL, maxBatchSize = NULL) {
   stage <- mmlspark_amd$stages$batching$TimeIntervalMiniBatchTransformer()
   if (!is.null(millisToWait)) stage$set("millisToWait", millisToWait)
@@ -904,11 +1481,42 @@ ml_trained_regressor_model <- function(labelCol = NULL, featurizerModel = NULL, 
   stage
 }
 
-ml_tune_hyperparameters <- function(evaluationMetric = NULL, numFolds = NULL, numRuns = NULL, parallelism = NULL, seed = NULL, labelCol = NULL, models = NULL, paramSpace = NULL) {
+ml_translate <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, toLanguage = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$Translate()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(textCol)) stage$set("textCol", textCol)
+  if (!is.null(toLanguage)) stage$set("toLanguage", toLanguage)
+  stage
+}
+
+ml_transliterate <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, toLanguage = NULL, fromScript = NULL, toScript = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$Transliterate()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(textCol)) stage$set("textCol", textCol)
+  if (!is.null(toLanguage)) stage$set("toLanguage", toLanguage)
+  if (!is.null(fromScript)) stage$set("fromScript", fromScript)
+  if (!is.null(toScript)) stage$set("toScript", toScript)
+  stage
+}
+
+ml_tune_hyperparameters <- function(evaluationMetric = NULL, numFolds = NULL, numRuns = NULL, searchMode = NULL, parallelism = NULL, seed = NULL, labelCol = NULL, models = NULL, paramSpace = NULL) {
   stage <- mmlspark_amd$stages$automl$TuneHyperparameters()
   if (!is.null(evaluationMetric)) stage$set("evaluationMetric", evaluationMetric)
   if (!is.null(numFolds)) stage$set("numFolds", numFolds)
   if (!is.null(numRuns)) stage$set("numRuns", numRuns)
+  if (!is.null(searchMode)) stage$set("searchMode", searchMode)
   if (!is.null(parallelism)) stage$set("parallelism", parallelism)
   if (!is.null(seed)) stage$set("seed", seed)
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
@@ -989,7 +1597,21 @@ ml_vector_shap <- function(model = NULL, targetCol = NULL, targetClasses = NULL,
   stage
 }
 
-ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, weightsArrays = NULL, rawPredictionCol = NULL, probabilityCol = NULL) {
+ml_verify_faces <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, faceId1Col = NULL, faceId2Col = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$VerifyFaces()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(faceId1Col)) stage$set("faceId1Col", faceId1Col)
+  if (!is.null(faceId2Col)) stage$set("faceId2Col", faceId2Col)
+  stage
+}
+
+ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, weightsArrays = NULL, rawPredictionCol = NULL, probabilityCol = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitClassificationModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1005,6 +1627,8 @@ ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol =
   if (!is.null(lossFunction)) stage$set("lossFunction", lossFunction)
   if (!is.null(batchSize)) stage$set("batchSize", batchSize)
   if (!is.null(hashSeed)) stage$set("hashSeed", hashSeed)
+  if (!is.null(adaptive)) stage$set("adaptive", adaptive)
+  if (!is.null(normalized)) stage$set("normalized", normalized)
   if (!is.null(initialModel)) stage$set("initialModel", initialModel)
   if (!is.null(passThroughArgs)) stage$set("passThroughArgs", passThroughArgs)
   if (!is.null(device)) stage$set("device", device)
@@ -1014,7 +1638,7 @@ ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol =
   stage
 }
 
-ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL) {
+ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitClassifier()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1030,6 +1654,8 @@ ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, add
   if (!is.null(lossFunction)) stage$set("lossFunction", lossFunction)
   if (!is.null(batchSize)) stage$set("batchSize", batchSize)
   if (!is.null(hashSeed)) stage$set("hashSeed", hashSeed)
+  if (!is.null(adaptive)) stage$set("adaptive", adaptive)
+  if (!is.null(normalized)) stage$set("normalized", normalized)
   if (!is.null(initialModel)) stage$set("initialModel", initialModel)
   if (!is.null(passThroughArgs)) stage$set("passThroughArgs", passThroughArgs)
   if (!is.null(device)) stage$set("device", device)
@@ -1038,7 +1664,7 @@ ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, add
   stage
 }
 
-ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, sharedCol = NULL, chosenActionCol = NULL, probabilityCol = NULL, epsilon = NULL) {
+ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, sharedCol = NULL, chosenActionCol = NULL, probabilityCol = NULL, epsilon = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitContextualBandit()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1054,6 +1680,8 @@ ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NU
   if (!is.null(lossFunction)) stage$set("lossFunction", lossFunction)
   if (!is.null(batchSize)) stage$set("batchSize", batchSize)
   if (!is.null(hashSeed)) stage$set("hashSeed", hashSeed)
+  if (!is.null(adaptive)) stage$set("adaptive", adaptive)
+  if (!is.null(normalized)) stage$set("normalized", normalized)
   if (!is.null(initialModel)) stage$set("initialModel", initialModel)
   if (!is.null(passThroughArgs)) stage$set("passThroughArgs", passThroughArgs)
   if (!is.null(device)) stage$set("device", device)
@@ -1064,7 +1692,7 @@ ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NU
   stage
 }
 
-ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, weightsArrays = NULL, sharedCol = NULL, epsilon = NULL) {
+ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, weightsArrays = NULL, sharedCol = NULL, epsilon = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitContextualBanditModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1080,6 +1708,8 @@ ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCo
   if (!is.null(lossFunction)) stage$set("lossFunction", lossFunction)
   if (!is.null(batchSize)) stage$set("batchSize", batchSize)
   if (!is.null(hashSeed)) stage$set("hashSeed", hashSeed)
+  if (!is.null(adaptive)) stage$set("adaptive", adaptive)
+  if (!is.null(normalized)) stage$set("normalized", normalized)
   if (!is.null(initialModel)) stage$set("initialModel", initialModel)
   if (!is.null(passThroughArgs)) stage$set("passThroughArgs", passThroughArgs)
   if (!is.null(device)) stage$set("device", device)
@@ -1109,7 +1739,7 @@ ml_vowpal_wabbit_interactions <- function(inputCols = NULL, outputCol = NULL, nu
   stage
 }
 
-ml_vowpal_wabbit_regressor <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL) {
+ml_vowpal_wabbit_regressor <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitRegressor()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1125,13 +1755,15 @@ ml_vowpal_wabbit_regressor <- function(labelCol = NULL, featuresCol = NULL, addi
   if (!is.null(lossFunction)) stage$set("lossFunction", lossFunction)
   if (!is.null(batchSize)) stage$set("batchSize", batchSize)
   if (!is.null(hashSeed)) stage$set("hashSeed", hashSeed)
+  if (!is.null(adaptive)) stage$set("adaptive", adaptive)
+  if (!is.null(normalized)) stage$set("normalized", normalized)
   if (!is.null(initialModel)) stage$set("initialModel", initialModel)
   if (!is.null(passThroughArgs)) stage$set("passThroughArgs", passThroughArgs)
   if (!is.null(device)) stage$set("device", device)
   stage
 }
 
-ml_vowpal_wabbit_regressor_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, weightsArrays = NULL) {
+ml_vowpal_wabbit_regressor_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, weightsArrays = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitRegressorModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1147,6 +1779,8 @@ ml_vowpal_wabbit_regressor_model <- function(labelCol = NULL, featuresCol = NULL
   if (!is.null(lossFunction)) stage$set("lossFunction", lossFunction)
   if (!is.null(batchSize)) stage$set("batchSize", batchSize)
   if (!is.null(hashSeed)) stage$set("hashSeed", hashSeed)
+  if (!is.null(adaptive)) stage$set("adaptive", adaptive)
+  if (!is.null(normalized)) stage$set("normalized", normalized)
   if (!is.null(initialModel)) stage$set("initialModel", initialModel)
   if (!is.null(passThroughArgs)) stage$set("passThroughArgs", passThroughArgs)
   if (!is.null(device)) stage$set("device", device)

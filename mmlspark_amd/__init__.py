@@ -44,6 +44,10 @@ def _register_all():
         "mmlspark_amd.stages.train",
         "mmlspark_amd.stages.automl",
         "mmlspark_amd.io_http.client",
+        "mmlspark_amd.io_http.cognitive",
+        "mmlspark_amd.io_http.files",
+        "mmlspark_amd.models.cyber",
+        "mmlspark_amd.stages.udfs",
     ):
         try:
             import_module(mod)

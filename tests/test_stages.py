@@ -263,3 +263,30 @@ def test_tune_hyperparameters_grid(mixed_class_df):
     best = tuner.fit(dff)
     assert best.get("bestMetric") > 0.8
     assert best.get("bestParams")["numLeaves"] in (4, 8)
+
+
+def test_plot_helpers(tmp_path):
+    """plot/ module (plot.py:17,45 analog): AUC math vs sklearn oracle and
+    confusion-matrix counts; renders to an Agg canvas (no display)."""
+    import matplotlib
+    matplotlib.use("Agg")
+    import matplotlib.pyplot as plt
+    import pandas as pd
+    from sklearn.metrics import roc_auc_score
+    from mmlspark_amd.plot import confusionMatrix, roc
+
+    rng = np.random.default_rng(0)
+    y = rng.integers(0, 2, 500)
+    score = y * 0.6 + rng.random(500) * 0.7
+    df = pd.DataFrame({"label": y.astype(float), "score": score,
+                       "pred": (score > 0.65).astype(float)})
+    plt.figure()
+    auc = roc(df, "label", "score")
+    plt.savefig(tmp_path / "roc.png")
+    assert abs(auc - roc_auc_score(y, score)) < 1e-9
+    plt.figure()
+    cm = confusionMatrix(df, "label", "pred", ["neg", "pos"])
+    plt.savefig(tmp_path / "cm.png")
+    assert cm.sum() == 500
+    assert cm[1, 1] > cm[1, 0]  # classifier is informative
+    plt.close("all")
