@@ -92,6 +92,10 @@ class _VWParams(Params):
                      "(--adaptive)", True, toBool)
     normalized = Param("normalized", "scale updates by running max|x| per "
                        "weight (--normalized, NAG-style)", False, toBool)
+    invariant = Param("invariant", "importance-weight-invariant updates "
+                      "(--invariant): closed-form integration of the per-"
+                      "example gradient flow, safe for large weights", False,
+                      toBool)
     initialModel = Param("initialModel", "warm-start weight table", None,
                          is_complex=True)
     passThroughArgs = Param("passThroughArgs",
@@ -109,7 +113,8 @@ class _VWParams(Params):
                    "--power_t": "powerT", "-b": "numBits", "--bit_precision":
                    "numBits", "--passes": "numPasses",
                    "--loss_function": "lossFunction"}
-        flags = {"--adaptive": "adaptive", "--normalized": "normalized"}
+        flags = {"--adaptive": "adaptive", "--normalized": "normalized",
+                 "--invariant": "invariant"}
         while i < len(s):
             if s[i] in mapping and i + 1 < len(s):
                 self.set(mapping[s[i]], s[i + 1])
@@ -173,7 +178,8 @@ class _VWBase(_VWParams, Estimator):
                 backend.vw_sgd_minibatch(
                     idx[sl], val[sl], o, labels[s:e], w, g, lr, l2,
                     power_t, loss,
-                    ex_w[s:e] if ex_w is not None else None, s_tbl)
+                    ex_w[s:e] if ex_w is not None else None, s_tbl,
+                    invariant=self.get("invariant"))
             learn_s += time.perf_counter() - t0
             # end-of-pass sync: RCCL all_reduce of weights + accumulators
             t0 = time.perf_counter()
@@ -381,9 +387,12 @@ class VowpalWabbitContextualBandit(_VWBase):
                 e = min(s + bs, n)
                 o = off_t[s:e + 1] - off_t[s]
                 sl = slice(int(off_t[s]), int(off_t[e]))
+                # IPS weights 1/prob can be huge → the invariant update is
+                # what keeps a rare action's example from blowing up the step
                 backend.vw_sgd_minibatch(idx_t[sl], val_t[sl], o, y_t[s:e],
                                          w, g, lr, l2, pt, "squared",
-                                         w_t[s:e])
+                                         w_t[s:e],
+                                         invariant=self.get("invariant"))
         model = VowpalWabbitContextualBanditModel(weights=w.cpu().numpy(),
                                                   adaptive=g.cpu().numpy())
         for p in ("sharedCol", "featuresCol", "predictionCol", "numBits",

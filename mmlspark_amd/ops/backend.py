@@ -123,16 +123,17 @@ def bin_matrix(X, upper_bounds, n_bins):
 # --------------------------------------------------------------------------- VW
 def vw_sgd_minibatch(indices, values, offsets, labels, weights_tbl, adaptive_tbl,
                      lr, l2, power_t, loss: str, ex_weight=None,
-                     normalize_tbl=None):
+                     normalize_tbl=None, invariant=False):
     if weights_tbl.is_cuda:
         return _require_ext().vw_sgd_minibatch(
             indices, values, offsets, labels, weights_tbl, adaptive_tbl,
             lr, l2, power_t, {"squared": 0, "logistic": 1, "hinge": 2}[loss],
-            ex_weight, normalize_tbl)
+            ex_weight, normalize_tbl, invariant)
     from ..models.vw import sgd_ref
     return sgd_ref.vw_sgd_minibatch(indices, values, offsets, labels,
                                     weights_tbl, adaptive_tbl, lr, l2,
-                                    power_t, loss, ex_weight, normalize_tbl)
+                                    power_t, loss, ex_weight, normalize_tbl,
+                                    invariant)
 
 
 def vw_predict(indices, values, offsets, weights_tbl):

@@ -30,7 +30,7 @@ void launch_split_scan(const float*, int, long, int, float, float, float,
                        hipStream_t);
 void launch_vw_sgd(const int*, const float*, const long*, const float*,
                    const float*, float*, float*, float*, float, float, float,
-                   int, long, float*, hipStream_t);
+                   int, int, long, float*, hipStream_t);
 void launch_vw_predict(const int*, const float*, const long*, const float*,
                        long, float*, hipStream_t);
 void launch_tree_shap(const int*, const float*, const int*, const int*,
@@ -199,7 +199,8 @@ torch::Tensor vw_sgd_minibatch(torch::Tensor idx, torch::Tensor val,
                                double lr, double l2, double power_t,
                                long loss,
                                c10::optional<torch::Tensor> ex_weight,
-                               c10::optional<torch::Tensor> s_tbl) {
+                               c10::optional<torch::Tensor> s_tbl,
+                               bool invariant) {
   CHECK_DEV(w_tbl); CHECK_CONTIG(w_tbl);
   const long n_ex = off.numel() - 1;
   auto preds = torch::zeros({n_ex}, w_tbl.options());
@@ -209,7 +210,8 @@ torch::Tensor vw_sgd_minibatch(torch::Tensor idx, torch::Tensor val,
   launch_vw_sgd(idx.data_ptr<int>(), val.data_ptr<float>(),
                 off.data_ptr<long>(), label.data_ptr<float>(), wptr,
                 w_tbl.data_ptr<float>(), g_tbl.data_ptr<float>(), sptr,
-                (float)lr, (float)l2, (float)power_t, (int)loss, n_ex,
+                (float)lr, (float)l2, (float)power_t, (int)loss,
+                invariant ? 1 : 0, n_ex,
                 preds.data_ptr<float>(), cur_stream());
   return preds;
 }

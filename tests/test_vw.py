@@ -177,3 +177,63 @@ def test_vw_normalized_adaptive_flags():
     est = VowpalWabbitClassifier(passThroughArgs="--normalized --adaptive")
     est._parse_args()
     assert est.get("normalized") and est.get("adaptive")
+
+
+def test_invariant_update_semantics():
+    """--invariant (VowpalWabbitBase.scala arg surface): a weight-h update
+    equals h sequential weight-1 updates as h*eta -> 0, and never overshoots
+    the label for ANY h (closed-form gradient-flow integration)."""
+    import torch
+    from mmlspark_amd.models.vw import sgd_ref
+
+    def one_example(h, invariant, loss="squared", n_upd=1, lr=0.5,
+                    power_t=0.5):
+        idx = torch.tensor([0, 1], dtype=torch.int32)
+        val = torch.tensor([1.0, 2.0])
+        off = torch.tensor([0, 2], dtype=torch.int64)
+        y = torch.tensor([3.0 if loss == "squared" else 1.0])
+        w = torch.zeros(8)
+        g = torch.zeros(8)
+        for _ in range(n_upd):
+            sgd_ref.vw_sgd_minibatch(
+                idx, val, off, y, w, g, lr, 0.0, power_t, loss,
+                ex_weight=torch.tensor([float(h)]), invariant=invariant)
+        pred = float(w[0] * 1.0 + w[1] * 2.0)
+        return pred
+
+    # huge importance weight: invariant converges exactly to the label;
+    # plain AdaGrad self-normalizes to a fixed lr-sized step far from it
+    p_inv = one_example(1e6, True)
+    p_plain = one_example(1e6, False)
+    assert abs(p_inv - 3.0) < 1e-3          # asymptote = label, no overshoot
+    assert abs(p_plain - 3.0) > 1.0
+    # invariance: with constant rates (power_t=0) one h=4 update is EXACTLY
+    # 4 sequential h=1 updates (exp decay composes); AdaGrad breaks exactness
+    p4 = one_example(4, True, lr=0.01, power_t=0.0)
+    pseq = one_example(1, True, lr=0.01, n_upd=4, power_t=0.0)
+    assert abs(p4 - pseq) < 1e-6
+    # logistic: q = y*p strictly grows toward +inf but by a bounded step
+    pl = one_example(50, True, loss="logistic", lr=0.1)
+    assert 0 < pl < 20 and np.isfinite(pl)
+    # hinge: lands exactly ON the margin (y*p == 1), never past
+    ph = one_example(1e5, True, loss="hinge", lr=0.5)
+    assert abs(ph - 1.0) < 1e-5
+
+
+def test_invariant_flag_parses_and_trains():
+    rng = np.random.default_rng(5)
+    n = 4000
+    X = rng.normal(size=(n, 6)).astype(np.float32)
+    wstar = np.array([2, -1.5, 1, 0, 0, 0], dtype=np.float32)
+    y = (X @ wstar + rng.normal(size=n) * 0.2 > 0).astype(np.float32)
+    df = pd.DataFrame({"text": [
+        " ".join(f"f{j}:{X[i, j]:.4f}" for j in range(6)) for i in range(n)],
+        "label": y})
+    feat = VowpalWabbitFeaturizer(inputCols=["text"], outputCol="features")
+    fdf = feat.transform(df)
+    m = VowpalWabbitClassifier(
+        passThroughArgs="--invariant --adaptive --loss_function logistic",
+        numPasses=8, learningRate=0.5).fit(fdf)
+    out = m.transform(fdf)
+    acc = float((out["prediction"].to_numpy() == y).mean())
+    assert acc > 0.9
