@@ -1611,7 +1611,7 @@ ml_verify_faces <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyC
   stage
 }
 
-ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, weightsArrays = NULL, rawPredictionCol = NULL, probabilityCol = NULL) {
+ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, weightsArrays = NULL, rawPredictionCol = NULL, probabilityCol = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitClassificationModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1629,6 +1629,7 @@ ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol =
   if (!is.null(hashSeed)) stage$set("hashSeed", hashSeed)
   if (!is.null(adaptive)) stage$set("adaptive", adaptive)
   if (!is.null(normalized)) stage$set("normalized", normalized)
+  if (!is.null(invariant)) stage$set("invariant", invariant)
   if (!is.null(initialModel)) stage$set("initialModel", initialModel)
   if (!is.null(passThroughArgs)) stage$set("passThroughArgs", passThroughArgs)
   if (!is.null(device)) stage$set("device", device)
@@ -1638,7 +1639,7 @@ ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol =
   stage
 }
 
-ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL) {
+ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitClassifier()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1656,6 +1657,7 @@ ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, add
   if (!is.null(hashSeed)) stage$set("hashSeed", hashSeed)
   if (!is.null(adaptive)) stage$set("adaptive", adaptive)
   if (!is.null(normalized)) stage$set("normalized", normalized)
+  if (!is.null(invariant)) stage$set("invariant", invariant)
   if (!is.null(initialModel)) stage$set("initialModel", initialModel)
   if (!is.null(passThroughArgs)) stage$set("passThroughArgs", passThroughArgs)
   if (!is.null(device)) stage$set("device", device)
@@ -1664,7 +1666,7 @@ ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, add
   stage
 }
 
-ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, sharedCol = NULL, chosenActionCol = NULL, probabilityCol = NULL, epsilon = NULL) {
+ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, sharedCol = NULL, chosenActionCol = NULL, probabilityCol = NULL, epsilon = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitContextualBandit()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1682,6 +1684,7 @@ ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NU
   if (!is.null(hashSeed)) stage$set("hashSeed", hashSeed)
   if (!is.null(adaptive)) stage$set("adaptive", adaptive)
   if (!is.null(normalized)) stage$set("normalized", normalized)
+  if (!is.null(invariant)) stage$set("invariant", invariant)
   if (!is.null(initialModel)) stage$set("initialModel", initialModel)
   if (!is.null(passThroughArgs)) stage$set("passThroughArgs", passThroughArgs)
   if (!is.null(device)) stage$set("device", device)
@@ -1692,7 +1695,7 @@ ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NU
   stage
 }
 
-ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, weightsArrays = NULL, sharedCol = NULL, epsilon = NULL) {
+ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, weightsArrays = NULL, sharedCol = NULL, epsilon = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitContextualBanditModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1710,6 +1713,7 @@ ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCo
   if (!is.null(hashSeed)) stage$set("hashSeed", hashSeed)
   if (!is.null(adaptive)) stage$set("adaptive", adaptive)
   if (!is.null(normalized)) stage$set("normalized", normalized)
+  if (!is.null(invariant)) stage$set("invariant", invariant)
   if (!is.null(initialModel)) stage$set("initialModel", initialModel)
   if (!is.null(passThroughArgs)) stage$set("passThroughArgs", passThroughArgs)
   if (!is.null(device)) stage$set("device", device)
@@ -1739,7 +1743,7 @@ ml_vowpal_wabbit_interactions <- function(inputCols = NULL, outputCol = NULL, nu
   stage
 }
 
-ml_vowpal_wabbit_regressor <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL) {
+ml_vowpal_wabbit_regressor <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitRegressor()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1757,13 +1761,14 @@ ml_vowpal_wabbit_regressor <- function(labelCol = NULL, featuresCol = NULL, addi
   if (!is.null(hashSeed)) stage$set("hashSeed", hashSeed)
   if (!is.null(adaptive)) stage$set("adaptive", adaptive)
   if (!is.null(normalized)) stage$set("normalized", normalized)
+  if (!is.null(invariant)) stage$set("invariant", invariant)
   if (!is.null(initialModel)) stage$set("initialModel", initialModel)
   if (!is.null(passThroughArgs)) stage$set("passThroughArgs", passThroughArgs)
   if (!is.null(device)) stage$set("device", device)
   stage
 }
 
-ml_vowpal_wabbit_regressor_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, weightsArrays = NULL) {
+ml_vowpal_wabbit_regressor_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, weightsArrays = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitRegressorModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1781,6 +1786,7 @@ ml_vowpal_wabbit_regressor_model <- function(labelCol = NULL, featuresCol = NULL
   if (!is.null(hashSeed)) stage$set("hashSeed", hashSeed)
   if (!is.null(adaptive)) stage$set("adaptive", adaptive)
   if (!is.null(normalized)) stage$set("normalized", normalized)
+  if (!is.null(invariant)) stage$set("invariant", invariant)
   if (!is.null(initialModel)) stage$set("initialModel", initialModel)
   if (!is.null(passThroughArgs)) stage$set("passThroughArgs", passThroughArgs)
   if (!is.null(device)) stage$set("device", device)
