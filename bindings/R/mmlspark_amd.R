@@ -43,6 +43,21 @@ ml_analyze_business_cards <- function(url = NULL, subscriptionKey = NULL, subscr
   stage
 }
 
+ml_analyze_custom_model <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL, modelId = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$AnalyzeCustomModel()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(imageUrlCol)) stage$set("imageUrlCol", imageUrlCol)
+  if (!is.null(imageBytesCol)) stage$set("imageBytesCol", imageBytesCol)
+  if (!is.null(modelId)) stage$set("modelId", modelId)
+  stage
+}
+
 ml_analyze_id_documents <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL) {
   stage <- mmlspark_amd$io_http$cognitive$AnalyzeIDDocuments()
   if (!is.null(url)) stage$set("url", url)
@@ -262,6 +277,20 @@ ml_conditional_knn_model <- function(featuresCol = NULL, valuesCol = NULL, outpu
   stage
 }
 
+ml_conversation_transcription <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, audioBytesCol = NULL, format = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$ConversationTranscription()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(audioBytesCol)) stage$set("audioBytesCol", audioBytesCol)
+  if (!is.null(format)) stage$set("format", format)
+  stage
+}
+
 ml_count_selector <- function(inputCol = NULL, outputCol = NULL) {
   stage <- mmlspark_amd$stages$featurize$CountSelector()
   if (!is.null(inputCol)) stage$set("inputCol", inputCol)
@@ -397,6 +426,35 @@ ml_detect_last_anomaly <- function(url = NULL, subscriptionKey = NULL, subscript
   stage
 }
 
+ml_dictionary_examples <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, toLanguage = NULL, translationCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$DictionaryExamples()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(textCol)) stage$set("textCol", textCol)
+  if (!is.null(toLanguage)) stage$set("toLanguage", toLanguage)
+  if (!is.null(translationCol)) stage$set("translationCol", translationCol)
+  stage
+}
+
+ml_dictionary_lookup <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, toLanguage = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$DictionaryLookup()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(textCol)) stage$set("textCol", textCol)
+  if (!is.null(toLanguage)) stage$set("toLanguage", toLanguage)
+  stage
+}
+
 ml_document_translator <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, sourceUrlCol = NULL, targetUrlCol = NULL, targetLanguage = NULL) {
   stage <- mmlspark_amd$io_http$cognitive$DocumentTranslator()
   if (!is.null(url)) stage$set("url", url)
@@ -442,6 +500,21 @@ ml_ensemble_by_key <- function(keys = NULL, cols = NULL, strategy = NULL, collap
 
 ml_entity_detector <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, language = NULL, languageCol = NULL) {
   stage <- mmlspark_amd$io_http$cognitive$EntityDetector()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(textCol)) stage$set("textCol", textCol)
+  if (!is.null(language)) stage$set("language", language)
+  if (!is.null(languageCol)) stage$set("languageCol", languageCol)
+  stage
+}
+
+ml_entity_detector_v2 <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, language = NULL, languageCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$EntityDetectorV2()
   if (!is.null(url)) stage$set("url", url)
   if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
   if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
@@ -526,6 +599,21 @@ ml_generate_thumbnails <- function(url = NULL, subscriptionKey = NULL, subscript
   if (!is.null(imageBytesCol)) stage$set("imageBytesCol", imageBytesCol)
   if (!is.null(width)) stage$set("width", width)
   if (!is.null(height)) stage$set("height", height)
+  stage
+}
+
+ml_get_custom_model <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL, modelId = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$GetCustomModel()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(imageUrlCol)) stage$set("imageUrlCol", imageUrlCol)
+  if (!is.null(imageBytesCol)) stage$set("imageBytesCol", imageBytesCol)
+  if (!is.null(modelId)) stage$set("modelId", modelId)
   stage
 }
 
@@ -747,6 +835,21 @@ ml_key_phrase_extractor <- function(url = NULL, subscriptionKey = NULL, subscrip
   stage
 }
 
+ml_key_phrase_extractor_v2 <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, language = NULL, languageCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$KeyPhraseExtractorV2()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(textCol)) stage$set("textCol", textCol)
+  if (!is.null(language)) stage$set("language", language)
+  if (!is.null(languageCol)) stage$set("languageCol", languageCol)
+  stage
+}
+
 ml_lambda <- function() {
   stage <- mmlspark_amd$stages$basic$Lambda()
 
@@ -755,6 +858,21 @@ ml_lambda <- function() {
 
 ml_language_detector <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, language = NULL, languageCol = NULL) {
   stage <- mmlspark_amd$io_http$cognitive$LanguageDetector()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(textCol)) stage$set("textCol", textCol)
+  if (!is.null(language)) stage$set("language", language)
+  if (!is.null(languageCol)) stage$set("languageCol", languageCol)
+  stage
+}
+
+ml_language_detector_v2 <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, language = NULL, languageCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$LanguageDetectorV2()
   if (!is.null(url)) stage$set("url", url)
   if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
   if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
@@ -977,6 +1095,20 @@ ml_linear_scalar_scaler_model <- function(inputCol = NULL, partitionKey = NULL, 
   stage
 }
 
+ml_list_custom_models <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$ListCustomModels()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(imageUrlCol)) stage$set("imageUrlCol", imageUrlCol)
+  if (!is.null(imageBytesCol)) stage$set("imageBytesCol", imageBytesCol)
+  stage
+}
+
 ml_multi_column_adapter <- function(baseStage = NULL, inputCols = NULL, outputCols = NULL) {
   stage <- mmlspark_amd$stages$basic$MultiColumnAdapter()
   if (!is.null(baseStage)) stage$set("baseStage", baseStage)
@@ -995,6 +1127,21 @@ ml_multi_n_gram <- function(inputCol = NULL, outputCol = NULL, lengths = NULL) {
 
 ml_ner <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, language = NULL, languageCol = NULL) {
   stage <- mmlspark_amd$io_http$cognitive$NER()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(textCol)) stage$set("textCol", textCol)
+  if (!is.null(language)) stage$set("language", language)
+  if (!is.null(languageCol)) stage$set("languageCol", languageCol)
+  stage
+}
+
+ml_nerv2 <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, language = NULL, languageCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$NERV2()
   if (!is.null(url)) stage$set("url", url)
   if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
   if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
@@ -1098,6 +1245,20 @@ ml_ranking_train_validation_split_model <- function(bestModel = NULL, validation
   stage
 }
 
+ml_read_image <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$ReadImage()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(imageUrlCol)) stage$set("imageUrlCol", imageUrlCol)
+  if (!is.null(imageBytesCol)) stage$set("imageBytesCol", imageBytesCol)
+  stage
+}
+
 ml_recognize_domain_specific_content <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, imageUrlCol = NULL, imageBytesCol = NULL, model = NULL) {
   stage <- mmlspark_amd$io_http$cognitive$RecognizeDomainSpecificContent()
   if (!is.null(url)) stage$set("url", url)
@@ -1186,6 +1347,24 @@ ml_sar_model <- function(userCol = NULL, itemCol = NULL, ratingCol = NULL, predi
 ml_select_columns <- function(cols = NULL) {
   stage <- mmlspark_amd$stages$basic$SelectColumns()
   if (!is.null(cols)) stage$set("cols", cols)
+  stage
+}
+
+ml_simple_detect_anomalies <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, seriesCol = NULL, granularity = NULL, sensitivity = NULL, groupbyCol = NULL, timestampCol = NULL, valueCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$SimpleDetectAnomalies()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(seriesCol)) stage$set("seriesCol", seriesCol)
+  if (!is.null(granularity)) stage$set("granularity", granularity)
+  if (!is.null(sensitivity)) stage$set("sensitivity", sensitivity)
+  if (!is.null(groupbyCol)) stage$set("groupbyCol", groupbyCol)
+  if (!is.null(timestampCol)) stage$set("timestampCol", timestampCol)
+  if (!is.null(valueCol)) stage$set("valueCol", valueCol)
   stage
 }
 
@@ -1409,6 +1588,21 @@ ml_text_shap <- function(model = NULL, targetCol = NULL, targetClasses = NULL, o
 
 ml_text_sentiment <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, language = NULL, languageCol = NULL) {
   stage <- mmlspark_amd$io_http$cognitive$TextSentiment()
+  if (!is.null(url)) stage$set("url", url)
+  if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
+  if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(errorCol)) stage$set("errorCol", errorCol)
+  if (!is.null(concurrency)) stage$set("concurrency", concurrency)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(textCol)) stage$set("textCol", textCol)
+  if (!is.null(language)) stage$set("language", language)
+  if (!is.null(languageCol)) stage$set("languageCol", languageCol)
+  stage
+}
+
+ml_text_sentiment_v2 <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol = NULL, outputCol = NULL, errorCol = NULL, concurrency = NULL, timeout = NULL, textCol = NULL, language = NULL, languageCol = NULL) {
+  stage <- mmlspark_amd$io_http$cognitive$TextSentimentV2()
   if (!is.null(url)) stage$set("url", url)
   if (!is.null(subscriptionKey)) stage$set("subscriptionKey", subscriptionKey)
   if (!is.null(subscriptionKeyCol)) stage$set("subscriptionKeyCol", subscriptionKeyCol)

@@ -138,6 +138,33 @@ class EntityDetector(_TextAnalyticsBase):
     pass
 
 
+# v2-endpoint variants (TextAnalytics.scala TextSentimentV2/NERV2/… — same
+# request shape, older service route; kept as distinct stages for parity)
+@register
+class TextSentimentV2(_TextAnalyticsBase):
+    pass
+
+
+@register
+class NERV2(_TextAnalyticsBase):
+    pass
+
+
+@register
+class KeyPhraseExtractorV2(_TextAnalyticsBase):
+    pass
+
+
+@register
+class EntityDetectorV2(_TextAnalyticsBase):
+    pass
+
+
+@register
+class LanguageDetectorV2(LanguageDetector):
+    pass
+
+
 # --------------------------------------------------------- computer vision
 class _ImageServiceBase(CognitiveServicesBase):
     imageUrlCol = Param("imageUrlCol", "image url column", None)
@@ -267,6 +294,33 @@ class DetectEntireSeries(_AnomalyBase):
     pass
 
 
+@register
+class SimpleDetectAnomalies(_AnomalyBase):
+    """Grouped per-series anomaly detection (AnomalyDetection.scala
+    SimpleDetectAnomalies): rows carry (group, timestamp, value); each
+    group's series is assembled then posted like DetectEntireSeries."""
+    groupbyCol = Param("groupbyCol", "series grouping column", "group")
+    timestampCol = Param("timestampCol", "timestamp column", "timestamp")
+    valueCol = Param("valueCol", "value column", "value")
+
+    def _transform(self, df):
+        gcol = self.get("groupbyCol")
+        heads = []
+        for _, gdf in df.groupby(gcol, sort=False):
+            series = [{"timestamp": str(t), "value": float(v)}
+                      for t, v in zip(gdf[self.get("timestampCol")],
+                                      gdf[self.get("valueCol")])]
+            head = gdf.iloc[0].copy()
+            head[self.get("seriesCol")] = series
+            heads.append(head)
+        per_group = super()._transform(pd.DataFrame(heads))
+        resp_by_group = dict(zip(per_group[gcol],
+                                 per_group[self.get("outputCol")]))
+        out = df.copy()
+        out[self.get("outputCol")] = df[gcol].map(resp_by_group)
+        return out
+
+
 # ------------------------------------------------------------ translator
 @register
 class Translate(CognitiveServicesBase):
@@ -304,6 +358,34 @@ class Transliterate(Translate):
     toScript = Param("toScript", "target script", "Latn")
 
 
+@register
+class DictionaryLookup(Translate):
+    """Translator.scala DictionaryLookup: alternative translations per term."""
+    def _row_url(self, row):
+        url = self.get("url")
+        langs = self.get("toLanguage") or ["en"]
+        sep = "&" if "?" in url else "?"
+        return url + sep + f"to={langs[0]}"
+
+
+@register
+class DictionaryExamples(Translate):
+    """Translator.scala DictionaryExamples: usage examples for a
+    (text, translation) pair per row."""
+    translationCol = Param("translationCol", "translation column",
+                           "translation")
+
+    def _payload(self, row):
+        texts = row[self.get("textCol")]
+        trans = row[self.get("translationCol")]
+        if isinstance(texts, str):
+            texts = [texts]
+        if isinstance(trans, str):
+            trans = [trans]
+        return [{"Text": t, "Translation": tr}
+                for t, tr in zip(texts, trans)]
+
+
 # -------------------------------------------------------- form recognizer
 class _FormRecognizerBase(_ImageServiceBase):
     pass
@@ -334,6 +416,49 @@ class AnalyzeIDDocuments(_FormRecognizerBase):
     pass
 
 
+@register
+class AnalyzeCustomModel(_FormRecognizerBase):
+    """FormRecognizer.scala AnalyzeCustomModel: analyze with a trained
+    custom model id spliced into the route."""
+    modelId = Param("modelId", "custom model id", None)
+
+    def _row_url(self, row):
+        url = self.get("url")
+        return url.replace("{modelId}", str(self._sv(row, "modelId", "")))
+
+
+@register
+class GetCustomModel(_FormRecognizerBase):
+    """FormRecognizer.scala GetCustomModel (GET of model metadata)."""
+    modelId = Param("modelId", "custom model id", None)
+
+    def _http_method(self):
+        return "GET"
+
+    def _payload(self, row):
+        return None
+
+    def _row_url(self, row):
+        url = self.get("url")
+        return url.replace("{modelId}", str(self._sv(row, "modelId", "")))
+
+
+@register
+class ListCustomModels(_FormRecognizerBase):
+    """FormRecognizer.scala ListCustomModels (GET of the model list)."""
+
+    def _http_method(self):
+        return "GET"
+
+    def _payload(self, row):
+        return None
+
+
+@register
+class ReadImage(_ImageServiceBase):
+    """ComputerVision.scala Read: async v3 Read API (document OCR)."""
+
+
 # ------------------------------------------------------------ speech/search
 @register
 class SpeechToTextSDK(CognitiveServicesBase):
@@ -347,6 +472,12 @@ class SpeechToTextSDK(CognitiveServicesBase):
         h = super()._headers(row)
         h["Content-Type"] = "audio/wav"
         return h
+
+
+@register
+class ConversationTranscription(SpeechToTextSDK):
+    """SpeechToTextSDK.scala ConversationTranscription: same audio-post
+    surface, speaker-attributed transcript endpoint."""
 
 
 @register

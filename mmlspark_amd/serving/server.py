@@ -90,9 +90,10 @@ class ServingServer:
                     self.end_headers()
                     self.wfile.write(info)
                 else:
-                    self.send_response(404)
-                    self.send_header("Content-Length", "0")
-                    self.end_headers()
+                    # GET = bodyless request routed like any other (the
+                    # reference's HTTPSourceV2 passes method through in
+                    # HTTPRequestData rather than rejecting non-POST)
+                    self._serve({"__method": "GET", "__path": self.path})
 
             def do_POST(self):
                 n = int(self.headers.get("Content-Length", 0))
@@ -104,6 +105,9 @@ class ServingServer:
                     self.send_header("Content-Length", "0")
                     self.end_headers()
                     return
+                self._serve(payload)
+
+            def _serve(self, payload):
                 if server.mode == "continuous":
                     try:
                         reply = server.handler([payload])[0]

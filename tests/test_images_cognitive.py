@@ -116,3 +116,40 @@ def test_bing_image_search_get(mock_service):
     out = b.transform(df)
     # mock returns a JSON body for GET-with-params too (echo handler tolerates)
     assert "response" in out.columns
+
+
+def test_simple_detect_anomalies_grouped(mock_service):
+    """SimpleDetectAnomalies (AnomalyDetection.scala): one request per
+    group, response joined back onto every member row."""
+    from mmlspark_amd.io_http.cognitive import SimpleDetectAnomalies
+    df = pd.DataFrame({
+        "group": ["a", "a", "a", "b", "b"],
+        "timestamp": [f"2024-01-0{i}T00:00:00Z" for i in (1, 2, 3, 1, 2)],
+        "value": [1.0, 2.0, 50.0, 5.0, 5.5],
+    })
+    det = SimpleDetectAnomalies(url=mock_service, outputCol="anom")
+    out = det.transform(df)
+    assert len(out) == 5
+    # all rows of one group share the same (single) service response
+    a = out[out["group"] == "a"]["anom"].tolist()
+    assert all(r == a[0] for r in a)
+    assert out["anom"].notna().all()
+
+
+def test_get_custom_model_uses_get(mock_service):
+    from mmlspark_amd.io_http.cognitive import GetCustomModel, ListCustomModels
+    df = pd.DataFrame({"x": [1]})
+    g = GetCustomModel(url=mock_service + "models/{modelId}", modelId="m-7",
+                       outputCol="meta")
+    out = g.transform(df)
+    assert out["meta"].iloc[0] is not None
+    ls = ListCustomModels(url=mock_service, outputCol="models")
+    assert ls.transform(df)["models"].iloc[0] is not None
+
+
+def test_dictionary_examples_payload():
+    from mmlspark_amd.io_http.cognitive import DictionaryExamples
+    d = DictionaryExamples(textCol="t", translationCol="tr")
+    row = pd.Series({"t": "hello", "tr": "hola"})
+    p = d._payload(row)
+    assert p == [{"Text": "hello", "Translation": "hola"}]
