@@ -886,7 +886,7 @@ ml_language_detector_v2 <- function(url = NULL, subscriptionKey = NULL, subscrip
   stage
 }
 
-ml_light_gbm_classification_model <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, predictionCol = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, boosterModelStr = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL, thresholds = NULL) {
+ml_light_gbm_classification_model <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, predictionCol = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, boosterModelStr = NULL, startIteration = NULL, numIterations = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL, thresholds = NULL) {
   stage <- mmlspark_amd$models$gbdt$estimators$LightGBMClassificationModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -895,6 +895,8 @@ ml_light_gbm_classification_model <- function(labelCol = NULL, featuresCol = NUL
   if (!is.null(leafPredictionCol)) stage$set("leafPredictionCol", leafPredictionCol)
   if (!is.null(featuresShapCol)) stage$set("featuresShapCol", featuresShapCol)
   if (!is.null(boosterModelStr)) stage$set("boosterModelStr", boosterModelStr)
+  if (!is.null(startIteration)) stage$set("startIteration", startIteration)
+  if (!is.null(numIterations)) stage$set("numIterations", numIterations)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(rawPredictionCol)) stage$set("rawPredictionCol", rawPredictionCol)
   if (!is.null(probabilityCol)) stage$set("probabilityCol", probabilityCol)
@@ -902,7 +904,7 @@ ml_light_gbm_classification_model <- function(labelCol = NULL, featuresCol = NUL
   stage
 }
 
-ml_light_gbm_classifier <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, modelString = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL) {
+ml_light_gbm_classifier <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, categoricalSlotNames = NULL, slotNames = NULL, modelString = NULL, lightGBMBooster = NULL, fobj = NULL, boostFromAverage = NULL, improvementTolerance = NULL, posBaggingFraction = NULL, negBaggingFraction = NULL, binSampleCount = NULL, maxBinByFeature = NULL, uniformDrop = NULL, xgboostDartMode = NULL, startIteration = NULL, chunkSize = NULL, defaultListenPort = NULL, driverListenPort = NULL, timeout = NULL, numTasks = NULL, numThreads = NULL, useSingleDatasetMode = NULL, matrixType = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL, isUnbalance = NULL) {
   stage <- mmlspark_amd$models$gbdt$estimators$LightGBMClassifier()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -943,14 +945,36 @@ ml_light_gbm_classifier <- function(labelCol = NULL, featuresCol = NULL, feature
   if (!is.null(parallelism)) stage$set("parallelism", parallelism)
   if (!is.null(topK)) stage$set("topK", topK)
   if (!is.null(categoricalSlotIndexes)) stage$set("categoricalSlotIndexes", categoricalSlotIndexes)
+  if (!is.null(categoricalSlotNames)) stage$set("categoricalSlotNames", categoricalSlotNames)
+  if (!is.null(slotNames)) stage$set("slotNames", slotNames)
   if (!is.null(modelString)) stage$set("modelString", modelString)
+  if (!is.null(lightGBMBooster)) stage$set("lightGBMBooster", lightGBMBooster)
+  if (!is.null(fobj)) stage$set("fobj", fobj)
+  if (!is.null(boostFromAverage)) stage$set("boostFromAverage", boostFromAverage)
+  if (!is.null(improvementTolerance)) stage$set("improvementTolerance", improvementTolerance)
+  if (!is.null(posBaggingFraction)) stage$set("posBaggingFraction", posBaggingFraction)
+  if (!is.null(negBaggingFraction)) stage$set("negBaggingFraction", negBaggingFraction)
+  if (!is.null(binSampleCount)) stage$set("binSampleCount", binSampleCount)
+  if (!is.null(maxBinByFeature)) stage$set("maxBinByFeature", maxBinByFeature)
+  if (!is.null(uniformDrop)) stage$set("uniformDrop", uniformDrop)
+  if (!is.null(xgboostDartMode)) stage$set("xgboostDartMode", xgboostDartMode)
+  if (!is.null(startIteration)) stage$set("startIteration", startIteration)
+  if (!is.null(chunkSize)) stage$set("chunkSize", chunkSize)
+  if (!is.null(defaultListenPort)) stage$set("defaultListenPort", defaultListenPort)
+  if (!is.null(driverListenPort)) stage$set("driverListenPort", driverListenPort)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(numTasks)) stage$set("numTasks", numTasks)
+  if (!is.null(numThreads)) stage$set("numThreads", numThreads)
+  if (!is.null(useSingleDatasetMode)) stage$set("useSingleDatasetMode", useSingleDatasetMode)
+  if (!is.null(matrixType)) stage$set("matrixType", matrixType)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(rawPredictionCol)) stage$set("rawPredictionCol", rawPredictionCol)
   if (!is.null(probabilityCol)) stage$set("probabilityCol", probabilityCol)
+  if (!is.null(isUnbalance)) stage$set("isUnbalance", isUnbalance)
   stage
 }
 
-ml_light_gbm_ranker <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, modelString = NULL, device = NULL, groupCol = NULL, labelGain = NULL, maxPosition = NULL, evalAt = NULL) {
+ml_light_gbm_ranker <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, categoricalSlotNames = NULL, slotNames = NULL, modelString = NULL, lightGBMBooster = NULL, fobj = NULL, boostFromAverage = NULL, improvementTolerance = NULL, posBaggingFraction = NULL, negBaggingFraction = NULL, binSampleCount = NULL, maxBinByFeature = NULL, uniformDrop = NULL, xgboostDartMode = NULL, startIteration = NULL, chunkSize = NULL, defaultListenPort = NULL, driverListenPort = NULL, timeout = NULL, numTasks = NULL, numThreads = NULL, useSingleDatasetMode = NULL, matrixType = NULL, device = NULL, groupCol = NULL, labelGain = NULL, maxPosition = NULL, evalAt = NULL) {
   stage <- mmlspark_amd$models$gbdt$estimators$LightGBMRanker()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -991,7 +1015,28 @@ ml_light_gbm_ranker <- function(labelCol = NULL, featuresCol = NULL, featureCols
   if (!is.null(parallelism)) stage$set("parallelism", parallelism)
   if (!is.null(topK)) stage$set("topK", topK)
   if (!is.null(categoricalSlotIndexes)) stage$set("categoricalSlotIndexes", categoricalSlotIndexes)
+  if (!is.null(categoricalSlotNames)) stage$set("categoricalSlotNames", categoricalSlotNames)
+  if (!is.null(slotNames)) stage$set("slotNames", slotNames)
   if (!is.null(modelString)) stage$set("modelString", modelString)
+  if (!is.null(lightGBMBooster)) stage$set("lightGBMBooster", lightGBMBooster)
+  if (!is.null(fobj)) stage$set("fobj", fobj)
+  if (!is.null(boostFromAverage)) stage$set("boostFromAverage", boostFromAverage)
+  if (!is.null(improvementTolerance)) stage$set("improvementTolerance", improvementTolerance)
+  if (!is.null(posBaggingFraction)) stage$set("posBaggingFraction", posBaggingFraction)
+  if (!is.null(negBaggingFraction)) stage$set("negBaggingFraction", negBaggingFraction)
+  if (!is.null(binSampleCount)) stage$set("binSampleCount", binSampleCount)
+  if (!is.null(maxBinByFeature)) stage$set("maxBinByFeature", maxBinByFeature)
+  if (!is.null(uniformDrop)) stage$set("uniformDrop", uniformDrop)
+  if (!is.null(xgboostDartMode)) stage$set("xgboostDartMode", xgboostDartMode)
+  if (!is.null(startIteration)) stage$set("startIteration", startIteration)
+  if (!is.null(chunkSize)) stage$set("chunkSize", chunkSize)
+  if (!is.null(defaultListenPort)) stage$set("defaultListenPort", defaultListenPort)
+  if (!is.null(driverListenPort)) stage$set("driverListenPort", driverListenPort)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(numTasks)) stage$set("numTasks", numTasks)
+  if (!is.null(numThreads)) stage$set("numThreads", numThreads)
+  if (!is.null(useSingleDatasetMode)) stage$set("useSingleDatasetMode", useSingleDatasetMode)
+  if (!is.null(matrixType)) stage$set("matrixType", matrixType)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(groupCol)) stage$set("groupCol", groupCol)
   if (!is.null(labelGain)) stage$set("labelGain", labelGain)
@@ -1000,7 +1045,7 @@ ml_light_gbm_ranker <- function(labelCol = NULL, featuresCol = NULL, featureCols
   stage
 }
 
-ml_light_gbm_ranker_model <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, predictionCol = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, boosterModelStr = NULL, device = NULL) {
+ml_light_gbm_ranker_model <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, predictionCol = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, boosterModelStr = NULL, startIteration = NULL, numIterations = NULL, device = NULL) {
   stage <- mmlspark_amd$models$gbdt$estimators$LightGBMRankerModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1009,11 +1054,13 @@ ml_light_gbm_ranker_model <- function(labelCol = NULL, featuresCol = NULL, featu
   if (!is.null(leafPredictionCol)) stage$set("leafPredictionCol", leafPredictionCol)
   if (!is.null(featuresShapCol)) stage$set("featuresShapCol", featuresShapCol)
   if (!is.null(boosterModelStr)) stage$set("boosterModelStr", boosterModelStr)
+  if (!is.null(startIteration)) stage$set("startIteration", startIteration)
+  if (!is.null(numIterations)) stage$set("numIterations", numIterations)
   if (!is.null(device)) stage$set("device", device)
   stage
 }
 
-ml_light_gbm_regression_model <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, predictionCol = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, boosterModelStr = NULL, device = NULL) {
+ml_light_gbm_regression_model <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, predictionCol = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, boosterModelStr = NULL, startIteration = NULL, numIterations = NULL, device = NULL) {
   stage <- mmlspark_amd$models$gbdt$estimators$LightGBMRegressionModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1022,11 +1069,13 @@ ml_light_gbm_regression_model <- function(labelCol = NULL, featuresCol = NULL, f
   if (!is.null(leafPredictionCol)) stage$set("leafPredictionCol", leafPredictionCol)
   if (!is.null(featuresShapCol)) stage$set("featuresShapCol", featuresShapCol)
   if (!is.null(boosterModelStr)) stage$set("boosterModelStr", boosterModelStr)
+  if (!is.null(startIteration)) stage$set("startIteration", startIteration)
+  if (!is.null(numIterations)) stage$set("numIterations", numIterations)
   if (!is.null(device)) stage$set("device", device)
   stage
 }
 
-ml_light_gbm_regressor <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, modelString = NULL, device = NULL, alpha = NULL, tweedieVariancePower = NULL) {
+ml_light_gbm_regressor <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, categoricalSlotNames = NULL, slotNames = NULL, modelString = NULL, lightGBMBooster = NULL, fobj = NULL, boostFromAverage = NULL, improvementTolerance = NULL, posBaggingFraction = NULL, negBaggingFraction = NULL, binSampleCount = NULL, maxBinByFeature = NULL, uniformDrop = NULL, xgboostDartMode = NULL, startIteration = NULL, chunkSize = NULL, defaultListenPort = NULL, driverListenPort = NULL, timeout = NULL, numTasks = NULL, numThreads = NULL, useSingleDatasetMode = NULL, matrixType = NULL, device = NULL, alpha = NULL, tweedieVariancePower = NULL) {
   stage <- mmlspark_amd$models$gbdt$estimators$LightGBMRegressor()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1067,7 +1116,28 @@ ml_light_gbm_regressor <- function(labelCol = NULL, featuresCol = NULL, featureC
   if (!is.null(parallelism)) stage$set("parallelism", parallelism)
   if (!is.null(topK)) stage$set("topK", topK)
   if (!is.null(categoricalSlotIndexes)) stage$set("categoricalSlotIndexes", categoricalSlotIndexes)
+  if (!is.null(categoricalSlotNames)) stage$set("categoricalSlotNames", categoricalSlotNames)
+  if (!is.null(slotNames)) stage$set("slotNames", slotNames)
   if (!is.null(modelString)) stage$set("modelString", modelString)
+  if (!is.null(lightGBMBooster)) stage$set("lightGBMBooster", lightGBMBooster)
+  if (!is.null(fobj)) stage$set("fobj", fobj)
+  if (!is.null(boostFromAverage)) stage$set("boostFromAverage", boostFromAverage)
+  if (!is.null(improvementTolerance)) stage$set("improvementTolerance", improvementTolerance)
+  if (!is.null(posBaggingFraction)) stage$set("posBaggingFraction", posBaggingFraction)
+  if (!is.null(negBaggingFraction)) stage$set("negBaggingFraction", negBaggingFraction)
+  if (!is.null(binSampleCount)) stage$set("binSampleCount", binSampleCount)
+  if (!is.null(maxBinByFeature)) stage$set("maxBinByFeature", maxBinByFeature)
+  if (!is.null(uniformDrop)) stage$set("uniformDrop", uniformDrop)
+  if (!is.null(xgboostDartMode)) stage$set("xgboostDartMode", xgboostDartMode)
+  if (!is.null(startIteration)) stage$set("startIteration", startIteration)
+  if (!is.null(chunkSize)) stage$set("chunkSize", chunkSize)
+  if (!is.null(defaultListenPort)) stage$set("defaultListenPort", defaultListenPort)
+  if (!is.null(driverListenPort)) stage$set("driverListenPort", driverListenPort)
+  if (!is.null(timeout)) stage$set("timeout", timeout)
+  if (!is.null(numTasks)) stage$set("numTasks", numTasks)
+  if (!is.null(numThreads)) stage$set("numThreads", numThreads)
+  if (!is.null(useSingleDatasetMode)) stage$set("useSingleDatasetMode", useSingleDatasetMode)
+  if (!is.null(matrixType)) stage$set("matrixType", matrixType)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(alpha)) stage$set("alpha", alpha)
   if (!is.null(tweedieVariancePower)) stage$set("tweedieVariancePower", tweedieVariancePower)

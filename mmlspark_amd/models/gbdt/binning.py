@@ -32,7 +32,8 @@ class BinMapper:
     @staticmethod
     def fit(X: torch.Tensor, n_bins: int = 255, sample_size: int = 200_000,
             comm: Optional[Comm] = None, seed: int = 0,
-            categorical: Optional[list] = None) -> "BinMapper":
+            categorical: Optional[list] = None,
+            max_bin_by_feature: Optional[list] = None) -> "BinMapper":
         n, nf = X.shape
         gen = torch.Generator(device="cpu").manual_seed(seed)
         k = min(n, max(1, sample_size // max(1, comm.world_size if comm else 1)))
@@ -51,6 +52,16 @@ class BinMapper:
         # strictly increasing boundaries; collapse duplicated quantiles
         ub = torch.cummax(ub, dim=1).values
         ub[:, -1] = float("inf")
+        if max_bin_by_feature:  # maxBinByFeature: per-feature bin caps —
+            # repeat the capped feature's quantile grid so only `cap` distinct
+            # boundaries survive (duplicates collapse to one effective bin)
+            for f, cap in enumerate(max_bin_by_feature[: ub.shape[0]]):
+                if cap and 1 < cap < n_bins:
+                    src = torch.linspace(0, ub.shape[1] - 1, cap - 1).long()
+                    ub[f] = ub[f, src].repeat_interleave(
+                        (ub.shape[1] + cap - 2) // (cap - 1))[: ub.shape[1]]
+                    ub[f] = torch.cummax(ub[f], dim=0).values
+                    ub[f, -1] = float("inf")
         return BinMapper(ub.contiguous(), n_bins, categorical)
 
     def transform(self, X: torch.Tensor) -> torch.Tensor:

@@ -85,8 +85,9 @@ class Booster:
             cat_offset=f.get("cat_offset"), cat_words=f.get("cat_words"))
         return raw + torch.from_numpy(self.base_score).to(X.device)
 
-    def predict_prob(self, X: torch.Tensor) -> torch.Tensor:
-        raw = self.predict_raw(X)
+    def predict_prob(self, X: torch.Tensor, start_iteration: int = 0,
+                     num_iteration: int = -1) -> torch.Tensor:
+        raw = self.predict_raw(X, start_iteration, num_iteration)
         if self.objective == "binary":
             p1 = torch.sigmoid(self.sigmoid * raw)
             return torch.cat([1 - p1, p1], dim=-1)

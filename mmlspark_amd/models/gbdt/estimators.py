@@ -81,7 +81,53 @@ class _GBDTParams(Params):
     topK = Param("topK", "voting-parallel top-K", 20, toInt)
     categoricalSlotIndexes = Param("categoricalSlotIndexes",
                                    "indexes of categorical features", None)
+    categoricalSlotNames = Param("categoricalSlotNames",
+                                 "names of categorical features (resolved "
+                                 "against feature names)", None)
+    slotNames = Param("slotNames", "override feature slot names", None)
     modelString = Param("modelString", "warm-start model text", "", toString)
+    lightGBMBooster = Param("lightGBMBooster", "warm-start Booster object "
+                            "(LightGBMParams.scala lightGBMBooster)", None,
+                            is_complex=True)
+    fobj = Param("fobj", "custom objective callable (preds, label, weight) "
+                 "-> (grad, hess) — FObjTrait analog", None, is_complex=True)
+    boostFromAverage = Param("boostFromAverage", "start boosting from the "
+                             "global label mean", True, toBool)
+    improvementTolerance = Param("improvementTolerance",
+                                 "min metric delta that counts as an early-"
+                                 "stopping improvement", 0.0, toFloat)
+    posBaggingFraction = Param("posBaggingFraction",
+                               "positive-row bagging fraction", 1.0, toFloat)
+    negBaggingFraction = Param("negBaggingFraction",
+                               "negative-row bagging fraction", 1.0, toFloat)
+    binSampleCount = Param("binSampleCount", "rows sampled for quantile bin "
+                           "boundaries", 200000, toInt)
+    maxBinByFeature = Param("maxBinByFeature", "per-feature max bin caps", None)
+    uniformDrop = Param("uniformDrop", "DART: uniform drop selection", True,
+                        toBool)
+    xgboostDartMode = Param("xgboostDartMode", "DART: xgboost normalization "
+                            "mode", False, toBool)
+    startIteration = Param("startIteration", "first iteration used at "
+                           "predict time", 0, toInt)
+    # Spark/JVM topology params accepted for API compatibility; topology is
+    # one-process-per-GPU with RCCL here, so they change nothing (documented
+    # in PARITY.md — LightGBMParams.scala:54-100)
+    chunkSize = Param("chunkSize", "ingestion chunk size (obsolete: data is "
+                      "device-resident)", 10000, toInt)
+    defaultListenPort = Param("defaultListenPort", "obsolete (RCCL "
+                              "rendezvous via MASTER_ADDR)", 12400, toInt)
+    driverListenPort = Param("driverListenPort", "obsolete", 0, toInt)
+    timeout = Param("timeout", "training timeout seconds (driver-level here)",
+                    1200.0, toFloat)
+    numTasks = Param("numTasks", "obsolete: world size = launched ranks", 0,
+                     toInt)
+    numThreads = Param("numThreads", "obsolete: GPU kernels replace the "
+                       "OpenMP pool", 0, toInt)
+    useSingleDatasetMode = Param("useSingleDatasetMode", "obsolete: one "
+                                 "process per GPU owns its shard", False,
+                                 toBool)
+    matrixType = Param("matrixType", "auto|dense|sparse (binned storage is "
+                       "always dense uint8)", "auto", toString)
     device = Param("device", "cpu|cuda|auto", "auto", toString)
 
     def _train_config(self) -> TrainConfig:
@@ -114,6 +160,14 @@ class _GBDTParams(Params):
             categorical_features=self.get("categoricalSlotIndexes"),
             parallelism=self.get("parallelism"),
             top_k=self.get("topK"),
+            boost_from_average=self.get("boostFromAverage"),
+            improvement_tolerance=self.get("improvementTolerance"),
+            pos_bagging_fraction=self.get("posBaggingFraction"),
+            neg_bagging_fraction=self.get("negBaggingFraction"),
+            bin_sample_count=self.get("binSampleCount"),
+            max_bin_by_feature=self.get("maxBinByFeature"),
+            uniform_drop=self.get("uniformDrop"),
+            xgboost_dart_mode=self.get("xgboostDartMode"),
         )
 
     def _device(self):
@@ -126,7 +180,8 @@ class _GBDTEstimatorBase(_GBDTParams, Estimator):
     def _make_objective(self, y: torch.Tensor):
         name = self.get("objective") or self._default_objective
         num_class = int(y.max().item()) + 1 if name in ("multiclass", "softmax") else 2
-        return make_objective(name, num_class=num_class)
+        return make_objective(name, num_class=num_class,
+                              fobj=self.get("fobj"))
 
     def _extract(self, df: pd.DataFrame, device):
         X = features_matrix(df, self.get("featuresCol"), self.get("featureCols"))
@@ -144,6 +199,14 @@ class _GBDTEstimatorBase(_GBDTParams, Estimator):
         comm = get_comm()
         names = infer_feature_names(df, self.get("featuresCol"),
                                     self.get("featureCols"))
+        if self.get("slotNames"):  # slotNames overrides inferred names
+            sn = list(self.get("slotNames"))
+            names = sn + names[len(sn):]
+        csn = self.get("categoricalSlotNames")
+        if csn:  # resolve categorical names -> slot indexes
+            cat = list(self.get("categoricalSlotIndexes") or [])
+            cat += [names.index(nm) for nm in csn if nm in names]
+            self.set("categoricalSlotIndexes", sorted(set(cat)))
         valid_df = None
         vic = self.get("validationIndicatorCol")
         if vic and vic in df.columns:
@@ -158,6 +221,8 @@ class _GBDTEstimatorBase(_GBDTParams, Estimator):
         ms = self.get("modelString")
         if ms:
             init = Booster.load_from_string(ms)
+        elif self.get("lightGBMBooster") is not None:
+            init = self.get("lightGBMBooster")  # warm start from object
 
         valid_sets = None
         if valid_df is not None and len(valid_df):
@@ -170,6 +235,7 @@ class _GBDTEstimatorBase(_GBDTParams, Estimator):
         for b in batches:
             part = df if n_batches <= 1 else df.iloc[b]
             Xt, yt, w = self._extract(part, device)
+            w = self._adjust_weights(yt, w)
             group = self._group_sizes(part, device)
             init_score = None
             isc = self.get("initScoreCol")
@@ -191,6 +257,9 @@ class _GBDTEstimatorBase(_GBDTParams, Estimator):
     def _group_sizes(self, df, device):
         return None
 
+    def _adjust_weights(self, yt, w):
+        return w
+
     def _model_class(self):
         raise NotImplementedError
 
@@ -206,6 +275,10 @@ class _GBDTModelBase(Model):
                             "output column for SHAP contributions", None)
     boosterModelStr = Param("boosterModelStr", "serialized booster", None,
                             is_complex=True)
+    startIteration = Param("startIteration", "first iteration used at "
+                           "predict time", 0, toInt)
+    numIterations = Param("numIterations", "iterations used at predict time "
+                          "(-1 = best/all)", -1, toInt)
     device = Param("device", "cpu|cuda|auto", "auto", toString)
 
     def __init__(self, booster: Optional[Booster] = None, **kwargs):
@@ -268,6 +341,19 @@ class LightGBMClassifier(_GBDTEstimatorBase):
     _default_objective = "binary"
     rawPredictionCol = Param("rawPredictionCol", "raw margin column", "rawPrediction")
     probabilityCol = Param("probabilityCol", "probability column", "probability")
+    isUnbalance = Param("isUnbalance", "re-weight positives by n_neg/n_pos "
+                        "(binary only)", False, toBool)
+
+    def _adjust_weights(self, yt, w):
+        if not self.get("isUnbalance"):
+            return w
+        pos = yt > 0
+        n_pos = float(pos.sum())
+        n_neg = float(yt.numel() - n_pos)
+        if n_pos == 0 or n_neg == 0:
+            return w
+        scale = torch.where(pos, n_neg / n_pos, 1.0)
+        return scale if w is None else w * scale
 
     def _model_class(self):
         return LightGBMClassificationModel
@@ -289,8 +375,9 @@ class LightGBMClassificationModel(_GBDTModelBase):
         device = default_device(self.get("device"))
         X = self._X(df, device)
         b = self.booster
-        raw = b.predict_raw(X)
-        prob = b.predict_prob(X)
+        si, ni = self.get("startIteration"), self.get("numIterations")
+        raw = b.predict_raw(X, si, ni)
+        prob = b.predict_prob(X, si, ni)
         out = df.copy()
         if b.objective == "binary":
             raw2 = torch.cat([-raw, raw], dim=-1)
@@ -335,7 +422,8 @@ class LightGBMRegressionModel(_GBDTModelBase):
     def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
         device = default_device(self.get("device"))
         X = self._X(df, device)
-        raw = self.booster.predict_raw(X)
+        raw = self.booster.predict_raw(X, self.get("startIteration"),
+                                       self.get("numIterations"))
         if self.booster.objective in ("poisson", "tweedie"):
             raw = torch.exp(raw)
         out = df.copy()
@@ -376,7 +464,8 @@ class LightGBMRankerModel(_GBDTModelBase):
     def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
         device = default_device(self.get("device"))
         X = self._X(df, device)
-        raw = self.booster.predict_raw(X)
+        raw = self.booster.predict_raw(X, self.get("startIteration"),
+                                       self.get("numIterations"))
         out = df.copy()
         out[self.get("predictionCol")] = raw.squeeze(-1).cpu().numpy().astype(np.float64)
         return self._maybe_extra_cols(df, out, X)

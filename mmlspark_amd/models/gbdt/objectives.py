@@ -280,9 +280,23 @@ class CustomObjective(Objective):
     """
     name = "custom"
 
-    def __init__(self, fobj: Callable, n_outputs: int = 1):
+    def __init__(self, fobj: Callable, base: Optional[Objective] = None):
         self.fobj = fobj
-        self.n_outputs = n_outputs
+        # objective param still names the output transform (ObjectiveParams
+        # semantics: fobj drives training, objective drives predict scale)
+        self.base = base
+        if base is not None:
+            self.name = base.name
+            self.n_outputs = base.n_outputs
+            self.higher_better_metric = base.higher_better_metric
+
+    def init_score(self, label, weight):
+        if self.base is not None:
+            return self.base.init_score(label, weight)
+        return super().init_score(label, weight)
+
+    def transform(self, raw):
+        return self.base.transform(raw) if self.base is not None else raw
 
     def grad_hess(self, preds, label, weight):
         g, h = self.fobj(preds, label, weight)
@@ -296,7 +310,13 @@ def make_objective(name: str, *, num_class: int = 2, sigmoid: float = 1.0,
                    tweedie_variance_power: float = 1.5,
                    label_gain=None, fobj: Optional[Callable] = None) -> Objective:
     if fobj is not None:
-        return CustomObjective(fobj)
+        base = None
+        if name:
+            base = make_objective(name, num_class=num_class, sigmoid=sigmoid,
+                                  alpha=alpha, fair_c=fair_c,
+                                  tweedie_variance_power=tweedie_variance_power,
+                                  label_gain=label_gain)
+        return CustomObjective(fobj, base)
     name = (name or "regression").lower()
     table = {
         "binary": lambda: BinaryObjective(sigmoid),

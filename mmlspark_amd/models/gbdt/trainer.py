@@ -73,6 +73,14 @@ class TrainConfig:
     categorical_features: Optional[List[int]] = None
     parallelism: str = "data_parallel"
     top_k: int = 20
+    boost_from_average: bool = True
+    improvement_tolerance: float = 0.0
+    pos_bagging_fraction: float = 1.0   # stratified bagging (binary labels)
+    neg_bagging_fraction: float = 1.0
+    bin_sample_count: int = 200_000     # binSampleCount
+    max_bin_by_feature: Optional[List[int]] = None
+    uniform_drop: bool = True           # dart: uniform tree-drop selection
+    xgboost_dart_mode: bool = False
 
 
 @dataclass
@@ -561,7 +569,9 @@ class TrainingSession:
         else:
             self.bin_mapper = BinMapper.fit(X, n_bins=cfg.max_bin, comm=comm,
                                             seed=cfg.seed,
-                                            categorical=cfg.categorical_features)
+                                            sample_size=cfg.bin_sample_count,
+                                            categorical=cfg.categorical_features,
+                                            max_bin_by_feature=cfg.max_bin_by_feature)
             self.binned = self.bin_mapper.transform(X)
 
         device, n, K = self.device, self.n, self.K
@@ -596,6 +606,8 @@ class TrainingSession:
     def _global_init_score(self):
         y, weight, device = self.y, self.weight, self.device
         obj, comm = self.objective, self.comm
+        if not self.cfg.boost_from_average:  # boostFromAverage=False
+            return torch.zeros(1, device=device)
         if comm.is_distributed:
             w = weight if weight is not None else torch.ones_like(
                 y, dtype=torch.float32)
@@ -624,7 +636,13 @@ class TrainingSession:
             gen.manual_seed(cfg.seed * 7919 + it)
             if float(torch.rand(1, generator=gen)) >= cfg.skip_drop:
                 cand = list(range(self.n_start_trees, len(trees)))
-                mask = torch.rand(len(cand), generator=gen) < cfg.drop_rate
+                if cfg.uniform_drop:
+                    pdrop = torch.full((len(cand),), cfg.drop_rate)
+                else:  # drop probability proportional to tree weight
+                    tw = torch.tensor([abs(tree_w[t]) for t in cand])
+                    pdrop = (cfg.drop_rate * len(cand)
+                             * tw / tw.sum().clamp_min(1e-12)).clamp(0, 1)
+                mask = torch.rand(len(cand), generator=gen) < pdrop
                 dropped = [cand[i] for i in range(len(cand)) if bool(mask[i])]
                 dropped = dropped[: cfg.max_drop]
         if dropped:
@@ -645,6 +663,17 @@ class TrainingSession:
         gen.manual_seed(cfg.seed * 104729 + it * 31 + self.comm.rank)
         if goss_mode and it >= 1:
             rows_root = _goss_sample(grad, hess, cfg, gen)
+        elif (cfg.pos_bagging_fraction < 1.0 or cfg.neg_bagging_fraction < 1.0) \
+                and cfg.bagging_freq > 0 and it % cfg.bagging_freq == 0:
+            # posBaggingFraction/negBaggingFraction: stratified row bagging
+            pos = (self.y > 0).cpu()
+            keep = torch.rand(self.n, generator=gen) < torch.where(
+                pos, torch.tensor(cfg.pos_bagging_fraction),
+                torch.tensor(cfg.neg_bagging_fraction))
+            rows_root = keep.nonzero(as_tuple=True)[0].to(
+                device, torch.int32).sort().values
+            if rows_root.numel() == 0:
+                rows_root = self.all_rows
         elif (cfg.bagging_freq > 0 and cfg.bagging_fraction < 1.0
               and it % cfg.bagging_freq == 0) or rf_mode:
             frac = cfg.bagging_fraction if cfg.bagging_fraction < 1.0 else 0.632
@@ -688,9 +717,15 @@ class TrainingSession:
             new_trees.append(tree)
 
         if dropped:
-            # DART normalization: dropped ×k/(k+1), new tree ×1/(k+1)
+            # DART normalization — default: dropped ×k/(k+1), new ×1/(k+1);
+            # xgboostDartMode: dropped ×k/(k+lr), new ×lr/(k+lr)
             kdrop = len(dropped)
-            factor = kdrop / (kdrop + 1.0)
+            if cfg.xgboost_dart_mode:
+                factor = kdrop / (kdrop + cfg.learning_rate)
+                new_scale = cfg.learning_rate / (kdrop + cfg.learning_rate)
+            else:
+                factor = kdrop / (kdrop + 1.0)
+                new_scale = 1.0 / (kdrop + 1.0)
             for t in dropped:
                 k = (t - self.n_start_trees) % K
                 delta = (factor - 1.0) * tree_w[t]
@@ -698,9 +733,9 @@ class TrainingSession:
                                                            device)
                 tree_w[t] *= factor
             for k, tr in enumerate(new_trees):
-                excess = (1.0 - 1.0 / (kdrop + 1.0)) * tr.shrinkage
+                excess = (1.0 - new_scale) * tr.shrinkage
                 preds[:, k] -= excess * predict_tree_binned(tr, self.binned, device)
-                tr.shrinkage *= 1.0 / (kdrop + 1.0)
+                tr.shrinkage *= new_scale
 
         for tr in new_trees:
             trees.append(tr)
@@ -757,9 +792,11 @@ def train_booster(X: torch.Tensor, y: torch.Tensor, cfg: TrainConfig,
             stats.evals.append(entry)
             stats.eval_s += time.perf_counter() - t0
             if score is not None and cfg.early_stopping_round > 0:
+                tol = cfg.improvement_tolerance  # improvementTolerance
                 better = (best_score is None
-                          or (score > best_score if objective.higher_better_metric
-                              else score < best_score))
+                          or (score > best_score + tol
+                              if objective.higher_better_metric
+                              else score < best_score - tol))
                 if better:
                     best_score, best_iter = score, it
                     rounds_no_improve = 0

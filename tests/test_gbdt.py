@@ -329,3 +329,74 @@ def test_multiclass_shap_contrib_layout():
     raw = m.booster.predict_raw(torch.from_numpy(X[:20])).numpy()
     # per-class additivity: sum of class contributions == class raw score
     np.testing.assert_allclose(per.sum(axis=2), raw, atol=1e-3)
+
+
+def test_custom_fobj(binary_df):
+    """fobj custom objective (FObjTrait analog): hand-written logistic
+    grad/hess trains as well as the built-in binary objective."""
+    def logistic_fobj(preds, label, weight):
+        z = torch.sigmoid(preds.squeeze(-1))
+        g = (z - label.float()).unsqueeze(-1)
+        h = (z * (1 - z)).clamp_min(1e-16).unsqueeze(-1)
+        return g, h
+
+    m = LightGBMClassifier(numIterations=20, numLeaves=15, fobj=logistic_fobj,
+                           objective="binary").fit(binary_df)
+    prob = np.stack(m.transform(binary_df)["probability"].to_numpy())[:, 1]
+    assert _auc(binary_df["label"].to_numpy(), prob) > 0.93
+
+
+def test_is_unbalance_reweights(binary_df):
+    """isUnbalance raises recall on the rare class via n_neg/n_pos weights."""
+    df = binary_df.copy()
+    pos = df[df.label == 1].head(40)  # make positives rare
+    df = pd.concat([df[df.label == 0], pos]).reset_index(drop=True)
+    base = LightGBMClassifier(numIterations=20, numLeaves=7).fit(df)
+    bal = LightGBMClassifier(numIterations=20, numLeaves=7,
+                             isUnbalance=True).fit(df)
+    pb = np.stack(base.transform(df)["probability"].to_numpy())[:, 1]
+    pw = np.stack(bal.transform(df)["probability"].to_numpy())[:, 1]
+    y = df["label"].to_numpy()
+    # weighted model shifts scores up on true positives
+    assert pw[y == 1].mean() > pb[y == 1].mean()
+
+
+def test_boost_from_average_flag(binary_df):
+    m0 = LightGBMClassifier(numIterations=1, numLeaves=3, learningRate=0.0,
+                            boostFromAverage=False).fit(binary_df)
+    raw = m0.booster.predict_raw(
+        torch.from_numpy(np.stack(binary_df["features"].to_numpy()[:4])))
+    np.testing.assert_allclose(raw.numpy(), 0.0, atol=1e-6)  # zero init, lr=0
+
+
+def test_start_iteration_predict_window(binary_df):
+    m = LightGBMClassifier(numIterations=10, numLeaves=7).fit(binary_df)
+    model = m
+    X = torch.from_numpy(np.stack(binary_df["features"].to_numpy()[:50]))
+    full = m.booster.predict_raw(X, 0, -1)
+    tail = m.booster.predict_raw(X, 5, -1)   # trees 5..9 only
+    head = m.booster.predict_raw(X, 0, 5)    # trees 0..4 only
+    base = torch.from_numpy(m.booster.base_score)
+    np.testing.assert_allclose((head + tail - base).numpy(), full.numpy(),
+                               rtol=1e-4, atol=1e-4)
+    model.set("startIteration", 5)
+    out = model.transform(binary_df.head(50))
+    raw_col = np.stack(out["rawPrediction"].to_numpy())[:, 1]
+    np.testing.assert_allclose(raw_col, tail.squeeze(-1).numpy(), rtol=1e-4,
+                               atol=1e-4)
+
+
+def test_max_bin_by_feature_and_pos_bagging(binary_df):
+    m = LightGBMClassifier(numIterations=10, numLeaves=7,
+                           maxBinByFeature=[4] * 10,
+                           posBaggingFraction=0.8, negBaggingFraction=0.5,
+                           baggingFreq=1).fit(binary_df)
+    prob = np.stack(m.transform(binary_df)["probability"].to_numpy())[:, 1]
+    assert _auc(binary_df["label"].to_numpy(), prob) > 0.8  # still learns
+
+
+def test_warm_start_from_booster_object(binary_df):
+    m1 = LightGBMClassifier(numIterations=5, numLeaves=7).fit(binary_df)
+    m2 = LightGBMClassifier(numIterations=5, numLeaves=7,
+                            lightGBMBooster=m1.booster).fit(binary_df)
+    assert m2.booster.num_trees == 10  # 5 warm + 5 new
