@@ -1875,7 +1875,7 @@ ml_verify_faces <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyC
   stage
 }
 
-ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, weightsArrays = NULL, rawPredictionCol = NULL, probabilityCol = NULL) {
+ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, device = NULL, weightsArrays = NULL, rawPredictionCol = NULL, probabilityCol = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitClassificationModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1896,6 +1896,11 @@ ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol =
   if (!is.null(invariant)) stage$set("invariant", invariant)
   if (!is.null(initialModel)) stage$set("initialModel", initialModel)
   if (!is.null(passThroughArgs)) stage$set("passThroughArgs", passThroughArgs)
+  if (!is.null(args)) stage$set("args", args)
+  if (!is.null(testArgs)) stage$set("testArgs", testArgs)
+  if (!is.null(interactions)) stage$set("interactions", interactions)
+  if (!is.null(ignoreNamespaces)) stage$set("ignoreNamespaces", ignoreNamespaces)
+  if (!is.null(useBarrierExecutionMode)) stage$set("useBarrierExecutionMode", useBarrierExecutionMode)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(weightsArrays)) stage$set("weightsArrays", weightsArrays)
   if (!is.null(rawPredictionCol)) stage$set("rawPredictionCol", rawPredictionCol)
@@ -1903,7 +1908,7 @@ ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol =
   stage
 }
 
-ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL) {
+ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL, labelConversion = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitClassifier()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1924,13 +1929,19 @@ ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, add
   if (!is.null(invariant)) stage$set("invariant", invariant)
   if (!is.null(initialModel)) stage$set("initialModel", initialModel)
   if (!is.null(passThroughArgs)) stage$set("passThroughArgs", passThroughArgs)
+  if (!is.null(args)) stage$set("args", args)
+  if (!is.null(testArgs)) stage$set("testArgs", testArgs)
+  if (!is.null(interactions)) stage$set("interactions", interactions)
+  if (!is.null(ignoreNamespaces)) stage$set("ignoreNamespaces", ignoreNamespaces)
+  if (!is.null(useBarrierExecutionMode)) stage$set("useBarrierExecutionMode", useBarrierExecutionMode)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(rawPredictionCol)) stage$set("rawPredictionCol", rawPredictionCol)
   if (!is.null(probabilityCol)) stage$set("probabilityCol", probabilityCol)
+  if (!is.null(labelConversion)) stage$set("labelConversion", labelConversion)
   stage
 }
 
-ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, sharedCol = NULL, chosenActionCol = NULL, probabilityCol = NULL, epsilon = NULL) {
+ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, device = NULL, sharedCol = NULL, chosenActionCol = NULL, probabilityCol = NULL, epsilon = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitContextualBandit()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1951,6 +1962,11 @@ ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NU
   if (!is.null(invariant)) stage$set("invariant", invariant)
   if (!is.null(initialModel)) stage$set("initialModel", initialModel)
   if (!is.null(passThroughArgs)) stage$set("passThroughArgs", passThroughArgs)
+  if (!is.null(args)) stage$set("args", args)
+  if (!is.null(testArgs)) stage$set("testArgs", testArgs)
+  if (!is.null(interactions)) stage$set("interactions", interactions)
+  if (!is.null(ignoreNamespaces)) stage$set("ignoreNamespaces", ignoreNamespaces)
+  if (!is.null(useBarrierExecutionMode)) stage$set("useBarrierExecutionMode", useBarrierExecutionMode)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(sharedCol)) stage$set("sharedCol", sharedCol)
   if (!is.null(chosenActionCol)) stage$set("chosenActionCol", chosenActionCol)
@@ -1959,7 +1975,7 @@ ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NU
   stage
 }
 
-ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, weightsArrays = NULL, sharedCol = NULL, epsilon = NULL) {
+ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, device = NULL, weightsArrays = NULL, sharedCol = NULL, epsilon = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitContextualBanditModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1980,6 +1996,11 @@ ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCo
   if (!is.null(invariant)) stage$set("invariant", invariant)
   if (!is.null(initialModel)) stage$set("initialModel", initialModel)
   if (!is.null(passThroughArgs)) stage$set("passThroughArgs", passThroughArgs)
+  if (!is.null(args)) stage$set("args", args)
+  if (!is.null(testArgs)) stage$set("testArgs", testArgs)
+  if (!is.null(interactions)) stage$set("interactions", interactions)
+  if (!is.null(ignoreNamespaces)) stage$set("ignoreNamespaces", ignoreNamespaces)
+  if (!is.null(useBarrierExecutionMode)) stage$set("useBarrierExecutionMode", useBarrierExecutionMode)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(weightsArrays)) stage$set("weightsArrays", weightsArrays)
   if (!is.null(sharedCol)) stage$set("sharedCol", sharedCol)
@@ -1987,13 +2008,15 @@ ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCo
   stage
 }
 
-ml_vowpal_wabbit_featurizer <- function(inputCols = NULL, outputCol = NULL, numBits = NULL, sumCollisions = NULL, stringSplitInputCols = NULL, seed = NULL) {
+ml_vowpal_wabbit_featurizer <- function(inputCols = NULL, outputCol = NULL, numBits = NULL, sumCollisions = NULL, stringSplitInputCols = NULL, prefixStringsWithColumnName = NULL, preserveOrderNumBits = NULL, seed = NULL) {
   stage <- mmlspark_amd$models$vw$featurizer$VowpalWabbitFeaturizer()
   if (!is.null(inputCols)) stage$set("inputCols", inputCols)
   if (!is.null(outputCol)) stage$set("outputCol", outputCol)
   if (!is.null(numBits)) stage$set("numBits", numBits)
   if (!is.null(sumCollisions)) stage$set("sumCollisions", sumCollisions)
   if (!is.null(stringSplitInputCols)) stage$set("stringSplitInputCols", stringSplitInputCols)
+  if (!is.null(prefixStringsWithColumnName)) stage$set("prefixStringsWithColumnName", prefixStringsWithColumnName)
+  if (!is.null(preserveOrderNumBits)) stage$set("preserveOrderNumBits", preserveOrderNumBits)
   if (!is.null(seed)) stage$set("seed", seed)
   stage
 }
@@ -2007,7 +2030,7 @@ ml_vowpal_wabbit_interactions <- function(inputCols = NULL, outputCol = NULL, nu
   stage
 }
 
-ml_vowpal_wabbit_regressor <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL) {
+ml_vowpal_wabbit_regressor <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, device = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitRegressor()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -2028,11 +2051,16 @@ ml_vowpal_wabbit_regressor <- function(labelCol = NULL, featuresCol = NULL, addi
   if (!is.null(invariant)) stage$set("invariant", invariant)
   if (!is.null(initialModel)) stage$set("initialModel", initialModel)
   if (!is.null(passThroughArgs)) stage$set("passThroughArgs", passThroughArgs)
+  if (!is.null(args)) stage$set("args", args)
+  if (!is.null(testArgs)) stage$set("testArgs", testArgs)
+  if (!is.null(interactions)) stage$set("interactions", interactions)
+  if (!is.null(ignoreNamespaces)) stage$set("ignoreNamespaces", ignoreNamespaces)
+  if (!is.null(useBarrierExecutionMode)) stage$set("useBarrierExecutionMode", useBarrierExecutionMode)
   if (!is.null(device)) stage$set("device", device)
   stage
 }
 
-ml_vowpal_wabbit_regressor_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, device = NULL, weightsArrays = NULL) {
+ml_vowpal_wabbit_regressor_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, device = NULL, weightsArrays = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitRegressorModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -2053,6 +2081,11 @@ ml_vowpal_wabbit_regressor_model <- function(labelCol = NULL, featuresCol = NULL
   if (!is.null(invariant)) stage$set("invariant", invariant)
   if (!is.null(initialModel)) stage$set("initialModel", initialModel)
   if (!is.null(passThroughArgs)) stage$set("passThroughArgs", passThroughArgs)
+  if (!is.null(args)) stage$set("args", args)
+  if (!is.null(testArgs)) stage$set("testArgs", testArgs)
+  if (!is.null(interactions)) stage$set("interactions", interactions)
+  if (!is.null(ignoreNamespaces)) stage$set("ignoreNamespaces", ignoreNamespaces)
+  if (!is.null(useBarrierExecutionMode)) stage$set("useBarrierExecutionMode", useBarrierExecutionMode)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(weightsArrays)) stage$set("weightsArrays", weightsArrays)
   stage

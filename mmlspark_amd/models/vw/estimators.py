@@ -102,12 +102,46 @@ class _VWParams(Params):
                             "VW-style arg string (subset parsed: --l1 --l2 "
                             "--learning_rate --power_t -b --passes "
                             "--loss_function)", "", toString)
+    args = Param("args", "VW arg string (alias of passThroughArgs — the "
+                 "reference's primary arg surface)", "", toString)
+    testArgs = Param("testArgs", "prediction-time VW args (accepted; "
+                     "scoring here has no mutable flags)", "", toString)
+    interactions = Param("interactions", "feature-column groups to cross "
+                         "with VW's FNV hash-combine (estimator-level -q): "
+                         "each entry a list or 'colA,colB' string", None)
+    ignoreNamespaces = Param("ignoreNamespaces", "first-letter namespaces "
+                             "whose additionalFeatures columns are dropped",
+                             None)
+    useBarrierExecutionMode = Param("useBarrierExecutionMode",
+                                    "gang-schedule ranks (no-op: ranks are "
+                                    "always gang-launched here)", False,
+                                    toBool)
     device = Param("device", "cpu|cuda|auto", "auto", toString)
+
+    def _feature_frame(self, df: pd.DataFrame):
+        """Apply ignoreNamespaces + estimator-level interactions; returns
+        (df_with_interaction_cols, effective extra feature columns)."""
+        extra = list(self.get("additionalFeatures") or [])
+        ig = self.get("ignoreNamespaces") or []
+        if ig:
+            extra = [c for c in extra if c[:1] not in ig]
+        specs = self.get("interactions") or []
+        from .featurizer import VowpalWabbitInteractions
+        for k, spec in enumerate(specs):
+            cols = (list(spec) if isinstance(spec, (list, tuple))
+                    else [c.strip() for c in str(spec).split(",")])
+            out = f"__interact_{k}"
+            df = VowpalWabbitInteractions(
+                inputCols=cols, outputCol=out,
+                numBits=self.get("numBits")).transform(df)
+            extra.append(out)
+        return df, extra
 
     def _parse_args(self):
         """Apply passThroughArgs (analog of the reference building the VW arg
         string, VowpalWabbitBase.scala:531-543 — here parsed back to Params)."""
-        s = (self.get("passThroughArgs") or "").split()
+        s = ((self.get("passThroughArgs") or "") + " "
+             + (self.get("args") or "")).split()
         i = 0
         mapping = {"--l1": "l1", "--l2": "l2", "--learning_rate": "learningRate",
                    "--power_t": "powerT", "-b": "numBits", "--bit_precision":
@@ -135,10 +169,11 @@ class _VWBase(_VWParams, Estimator):
         comm = get_comm()
         device = default_device(self.get("device"))
         t_ingest = time.perf_counter()
+        df, extra_cols = self._feature_frame(df)
         idx, val, off, _ = _extract_csr(df, self.get("featuresCol"),
-                                        self.get("additionalFeatures"), device)
+                                        extra_cols, device)
         y = df[self.get("labelCol")].to_numpy(dtype=np.float32)
-        if self._binary_labels:
+        if self._binary_labels and getattr(self, "_convert_labels", True):
             y = np.where(y > 0, 1.0, -1.0).astype(np.float32)
         labels = torch.from_numpy(y).to(device)
         ex_w = None
@@ -197,7 +232,8 @@ class _VWBase(_VWParams, Estimator):
         model = self._model_class()(weights=w.cpu().numpy(),
                                     adaptive=g.cpu().numpy())
         for pname in ("labelCol", "featuresCol", "additionalFeatures",
-                      "predictionCol", "numBits"):
+                      "predictionCol", "numBits", "interactions",
+                      "ignoreNamespaces"):
             model.set(pname, self.get(pname))
         model.set("lossFunction", loss)
         model._stats = pd.DataFrame([{
@@ -243,8 +279,9 @@ class _VWModelBase(_VWParams, Model):
 
     def _raw(self, df: pd.DataFrame) -> np.ndarray:
         device = default_device(self.get("device"))
+        df, extra_cols = self._feature_frame(df)
         idx, val, off, _ = _extract_csr(df, self.get("featuresCol"),
-                                        self.get("additionalFeatures"), device)
+                                        extra_cols, device)
         w = torch.from_numpy(self.weights).to(device)
         return backend.vw_predict(idx, val, off, w).cpu().numpy()
 
@@ -271,6 +308,13 @@ class VowpalWabbitClassifier(_VWBase):
     _binary_labels = True
     rawPredictionCol = Param("rawPredictionCol", "margin column", "rawPrediction")
     probabilityCol = Param("probabilityCol", "probability column", "probability")
+    labelConversion = Param("labelConversion", "convert {0,1} labels to "
+                            "{-1,+1} before training "
+                            "(VowpalWabbitClassifier.scala)", True, toBool)
+
+    @property
+    def _convert_labels(self):
+        return self.get("labelConversion")
 
     def _fit(self, df):
         model = super()._fit(df)

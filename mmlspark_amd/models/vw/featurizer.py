@@ -46,6 +46,12 @@ class VowpalWabbitFeaturizer(Transformer):
     stringSplitInputCols = Param("stringSplitInputCols",
                                  "string columns split on whitespace into "
                                  "individual features", None, toList)
+    prefixStringsWithColumnName = Param(
+        "prefixStringsWithColumnName", "hash '<col><value>' instead of the "
+        "bare string value (VowpalWabbitFeaturizer.scala)", True, toBool)
+    preserveOrderNumBits = Param(
+        "preserveOrderNumBits", "reserve this many top index bits for the "
+        "column ordinal so feature order survives hashing", 0, toInt)
     seed = Param("seed", "murmur seed", 0, toInt)
 
     def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
@@ -56,6 +62,11 @@ class VowpalWabbitFeaturizer(Transformer):
         size = 1 << bits
         seed = self.get("seed")
         sum_c = self.get("sumCollisions")
+        prefix = self.get("prefixStringsWithColumnName")
+        order_bits = self.get("preserveOrderNumBits")
+        col_order = {c: i for i, c in enumerate(list(cols) + list(split_cols))}
+        if order_bits:
+            mask = (1 << (bits - order_bits)) - 1
 
         # per-column namespace seeds (column name = VW namespace)
         col_seed = {c: hash_string(c, seed) for c in list(cols) + list(split_cols)}
@@ -81,6 +92,8 @@ class VowpalWabbitFeaturizer(Transformer):
         for _, row in df.iterrows():
             idx: List[int] = []
             val: List[float] = []
+            pref_ord: List[int] = []
+            n_before = 0
             for kind, c in plans:
                 v = row[c]
                 ns = col_seed[c]
@@ -95,7 +108,8 @@ class VowpalWabbitFeaturizer(Transformer):
                         val.append(1.0)
                 elif kind == "str":
                     if v:
-                        idx.append(hash_string(f"{c}{v}", ns) & mask)
+                        key = f"{c}{v}" if prefix else str(v)
+                        idx.append(hash_string(key, ns) & mask)
                         val.append(1.0)
                 elif kind == "split":
                     if v:
@@ -118,8 +132,15 @@ class VowpalWabbitFeaturizer(Transformer):
                         for tok in v:
                             idx.append(hash_string(str(tok), ns) & mask)
                             val.append(1.0)
+                if order_bits:
+                    pref_ord.extend([col_order[c]] * (len(idx) - n_before))
+                    n_before = len(idx)
             ia = np.asarray(idx, dtype=np.int64)
             va = np.asarray(val, dtype=np.float32)
+            if order_bits:
+                # order-preserving bit prefix (preserveOrderNumBits)
+                pref = np.asarray(pref_ord, dtype=np.int64)
+                ia = (pref << (bits - order_bits)) | ia
             if sum_c:
                 ia, va = _sum_collisions(ia, va)
             else:
@@ -151,8 +172,13 @@ class VowpalWabbitInteractions(Transformer):
             idx = np.zeros(1, dtype=np.int64)
             val = np.ones(1, dtype=np.float64)
             for v in vecs:
-                vi = v.indices.astype(np.int64)
-                vv = v.values.astype(np.float64)
+                if isinstance(v, SparseVector):
+                    vi = v.indices.astype(np.int64)
+                    vv = v.values.astype(np.float64)
+                else:  # dense vector column
+                    dense = np.asarray(v, dtype=np.float64)
+                    vi = np.nonzero(dense)[0].astype(np.int64)
+                    vv = dense[vi]
                 # VW interaction hash: h = h_prev * FNV_PRIME ^ h_feature
                 idx = ((idx[:, None] * FNV_PRIME) ^ vi[None, :]).reshape(-1) & _M32
                 val = (val[:, None] * vv[None, :]).reshape(-1)

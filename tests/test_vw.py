@@ -237,3 +237,56 @@ def test_invariant_flag_parses_and_trains():
     out = m.transform(fdf)
     acc = float((out["prediction"].to_numpy() == y).mean())
     assert acc > 0.9
+
+
+def test_interactions_estimator_level():
+    """Estimator-level interactions (-q analog): crossing two informative
+    namespaces lets a linear model learn an XOR-ish product signal."""
+    rng = np.random.default_rng(7)
+    n = 3000
+    a = rng.choice([-1.0, 1.0], size=n).astype(np.float32)
+    b = rng.choice([-1.0, 1.0], size=n).astype(np.float32)
+    y = (a * b > 0).astype(np.float32)  # pure interaction, no main effects
+    df = pd.DataFrame({"fa": list(np.eye(2, dtype=np.float32)[(a > 0).astype(int)]),
+                       "fb": list(np.eye(2, dtype=np.float32)[(b > 0).astype(int)]),
+                       "label": y})
+    plain = VowpalWabbitClassifier(featuresCol="fa", additionalFeatures=["fb"],
+                                   numPasses=6).fit(df)
+    crossed = VowpalWabbitClassifier(featuresCol="fa", additionalFeatures=["fb"],
+                                     interactions=["fa,fb"],
+                                     numPasses=6).fit(df)
+    acc_p = (plain.transform(df)["prediction"].to_numpy() == y).mean()
+    acc_c = (crossed.transform(df)["prediction"].to_numpy() == y).mean()
+    assert acc_c > 0.95
+    assert acc_p < 0.7  # no main effects: plain linear model can't learn it
+
+
+def test_featurizer_order_bits_and_prefix():
+    df = pd.DataFrame({"colA": ["x"], "colB": ["x"]})
+    f = VowpalWabbitFeaturizer(inputCols=["colA", "colB"], numBits=10,
+                               preserveOrderNumBits=2)
+    v = f.transform(df)["features"].iloc[0]
+    # order prefix puts colA features in the bottom quarter, colB in the next
+    assert all(i < 256 for i in v.indices[:1])
+    assert any(256 <= i < 512 for i in v.indices)
+    # prefix off: same token in two columns hashes differently only via seed
+    f2 = VowpalWabbitFeaturizer(inputCols=["colA"],
+                                prefixStringsWithColumnName=False, numBits=10)
+    f3 = VowpalWabbitFeaturizer(inputCols=["colA"], numBits=10)
+    i2 = f2.transform(df)["features"].iloc[0].indices[0]
+    i3 = f3.transform(df)["features"].iloc[0].indices[0]
+    assert i2 != i3  # "x" vs "colAx" under the same namespace seed
+
+
+def test_args_alias_and_label_conversion():
+    rng = np.random.default_rng(8)
+    n = 2000
+    X = rng.normal(size=(n, 5)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.float32)
+    df = pd.DataFrame({"features": list(X), "label": y})
+    m = VowpalWabbitClassifier(args="--learning_rate 0.7 --passes 4").fit(df)
+    assert m.get("numBits") == 18
+    acc = (m.transform(df)["prediction"].to_numpy() == y).mean()
+    assert acc > 0.9
+    m2 = VowpalWabbitClassifier(labelConversion=False, numPasses=4).fit(df)
+    assert (m2.transform(df)["prediction"].to_numpy() == y).mean() > 0.7
