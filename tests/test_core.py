@@ -135,3 +135,48 @@ def test_pipeline_with_estimator_stage(binary_df):
     pm = p.fit(binary_df)
     out = pm.transform(binary_df)
     assert "prediction" in out.columns
+
+
+def test_fault_utils():
+    """retry_with_timeout / StopWatch / async_map / using / SharedVariable
+    (FaultToleranceUtils.scala:33, StopWatch.scala, SharedVariable.scala:18)."""
+    import time as _time
+    from mmlspark_amd.utils.fault import (
+        StopWatch, SharedVariable, async_map, retry_with_timeout, using)
+
+    calls = []
+
+    def flaky():
+        calls.append(1)
+        if len(calls) < 3:
+            raise RuntimeError("transient")
+        return 42
+
+    assert retry_with_timeout(flaky, timeout_s=5, retries=4,
+                              backoff_s=0.001) == 42
+    assert len(calls) == 3
+
+    with pytest.raises(TimeoutError):
+        retry_with_timeout(lambda: _time.sleep(2), timeout_s=0.05, retries=1)
+
+    sw = StopWatch()
+    with sw.measure():
+        _time.sleep(0.01)
+    assert sw.elapsed_s >= 0.01
+
+    assert async_map(lambda x: x * x, range(10), concurrency=4) == \
+        [x * x for x in range(10)]
+
+    class _Res:
+        closed = False
+        def close(self):
+            self.closed = True
+    r = _Res()
+    with using(r):
+        pass
+    assert r.closed
+
+    built = []
+    sv = SharedVariable(lambda: built.append(1) or {"n": 0})
+    assert sv.get() is sv.get()
+    assert built == [1]
