@@ -186,9 +186,16 @@ extern "C" void launch_hist_build_fixed(const void* binned, long n_rows,
   long chunks = (2048 + n_fblocks - 1) / n_fblocks;
   long chunk = (m + chunks - 1) / chunks;
   // every block flushes its whole LDS histogram with GLOBAL atomics, so
-  // small leaves must not shatter into hundreds of row-chunks (profiled:
-  // flush contention made small-leaf calls ~0.3 ms) — keep chunks coarse
-  if (chunk < 16384) chunk = 16384;
+  // large leaves must not shatter into hundreds of row-chunks (profiled:
+  // flush contention made calls ~0.3 ms at ~157 chunks) — but a SMALL leaf
+  // at one coarse chunk uses only n_fblocks workgroups of a 256-CU chip, so
+  // give small leaves at least 8 row-chunks (8-way flush contention is
+  // negligible; measured floor drops ~3x)
+  if (chunk < 16384) {
+    chunk = (m + 7) / 8;
+    if (chunk < 2048) chunk = 2048;
+    if (chunk > 16384) chunk = 16384;
+  }
   if (chunk > (1l << 19)) chunk = 1l << 19;  // count-field bound (< 2^20)
   chunks = (m + chunk - 1) / chunk;
   dim3 grid((unsigned)chunks, (unsigned)n_fblocks);

@@ -44,6 +44,14 @@ class HyperparamBuilder:
         self.space[name] = dist
         return self
 
+    def addRange(self, name: str, lo, hi):
+        """Convenience for Int/DoubleRangeHyperParam (DefaultHyperparams
+        usage, automl/DefaultHyperparams.scala:20-34)."""
+        return self.addHyperparam(name, RangeHyperParam(lo, hi))
+
+    def addDiscrete(self, name: str, values):
+        return self.addHyperparam(name, DiscreteHyperParam(values))
+
     def build(self):
         return self.space
 
@@ -156,6 +164,12 @@ class TuneHyperparametersModel(Model):
 
     def _transform(self, df):
         return self.get("bestModel").transform(df)
+
+    def getBestModelInfo(self):
+        """Winning hyperparameters + CV metric (TuneHyperparameters.scala
+        getBestModelInfo)."""
+        return {"params": self.get("bestParams"),
+                "metric": self.get("bestMetric")}
 
 
 @register
