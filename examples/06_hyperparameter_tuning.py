@@ -36,7 +36,7 @@ print("best params:", tuned.getBestModelInfo())
 candidates = [LightGBMClassifier(numLeaves=nl, numIterations=25).fit(df)
               for nl in (7, 31)]
 best = FindBestModel(models=candidates, evaluationMetric="AUC").fit(df)
-print("FindBestModel AUC:", round(best.getBestModelMetrics()["AUC"], 4))
+print("FindBestModel AUC:", round(best.getBestModelMetrics(), 4))
 
 # interop: export the winner as stock LightGBM native text + reload
 txt = best.getBestModel().booster.to_lightgbm_text()

@@ -23,7 +23,9 @@ class DiscreteHyperParam:
 
 
 class RangeHyperParam:
-    def __init__(self, lo, hi, is_int=False, log=False):
+    def __init__(self, lo, hi, is_int=None, log=False):
+        if is_int is None:  # Int vs DoubleRangeHyperParam inferred from bounds
+            is_int = isinstance(lo, int) and isinstance(hi, int)
         self.lo, self.hi, self.is_int, self.log = lo, hi, is_int, log
 
     def sample(self, rng):
