@@ -327,7 +327,9 @@ class TreeGrower:
             self.nf, self.scale_g, self.scale_h, cfg.lambda_l1, cfg.lambda_l2,
             float(cfg.min_data_in_leaf), cfg.min_sum_hessian_in_leaf,
             cfg.min_gain_to_split, cfg.max_delta_step, cfg.num_leaves,
-            cfg.max_depth, feat_mask, reduce_fn, self.comm.is_distributed)
+            cfg.max_depth, feat_mask, reduce_fn,
+            self.comm.is_distributed or bool(__import__("os").environ.get(
+                "MMLSPARK_AMD_FORCE_DIST_GROWER")))
         self.stats.hist_s += time.perf_counter() - t0
         feature = d["feature"].numpy()
         thr_bin = d["thr_bin"].numpy()

@@ -24,6 +24,10 @@ extern "C" {
 void launch_hist_build_fixed(const void*, long, const int*, long, const float*,
                              const float*, long long*, int, int, double,
                              double, hipStream_t);
+void launch_hist_build_fixed_child(const void*, long, const int*, long,
+                                   const float*, const float*, long long*,
+                                   int, int, double, double, const int*, int,
+                                   hipStream_t);
 void launch_split_scan_fixed(const long long*, int, long, int, float, float,
                              float, float, float, long, const bool*, float*,
                              float*, double, double, hipStream_t);
@@ -55,10 +59,12 @@ struct GrowCtx {
 };
 
 struct SplitJob {
-  torch::Tensor rows_l, rows_r, hist_l, hist_r;
+  torch::Tensor out_rows;   // partitioned rows (left | right), device
+  torch::Tensor hist_l, hist_r;
   torch::Tensor scan_host;  // (2,6) f32 pinned
+  torch::Tensor nl_host;    // (1,) i32 pinned — local left count readback
+  long nl_known = -1;       // host-known left count (single-rank fast path)
   hipEvent_t ev = nullptr;
-  bool done = false;
 
   ~SplitJob() {
     if (ev) (void)hipEventDestroy(ev);
@@ -88,15 +94,26 @@ struct CandCmp {
 };
 
 torch::Tensor build_hist(GrowCtx& ctx, const torch::Tensor& rows,
-                         const torch::Tensor& grad, const torch::Tensor& hess) {
+                         const torch::Tensor& grad, const torch::Tensor& hess,
+                         const int* nl_dev = nullptr, int side = -1) {
   auto hist = torch::zeros({ctx.ngroups * 4, ctx.n_bins, 3},
                            grad.options().dtype(torch::kInt64));
-  launch_hist_build_fixed(ctx.binned.data_ptr(), ctx.n_rows,
-                          rows.data_ptr<int>(), rows.numel(),
-                          grad.data_ptr<float>(), hess.data_ptr<float>(),
-                          (long long*)hist.data_ptr<int64_t>(), ctx.n_bins,
-                          ctx.ngroups, ctx.scale_g, ctx.scale_h,
-                          grower_stream());
+  if (side >= 0) {
+    // child-of-partition mode: rows = parent partition buffer, the child
+    // range [0,nl) or [nl,m) is resolved on DEVICE — no host readback
+    launch_hist_build_fixed_child(
+        ctx.binned.data_ptr(), ctx.n_rows, rows.data_ptr<int>(), rows.numel(),
+        grad.data_ptr<float>(), hess.data_ptr<float>(),
+        (long long*)hist.data_ptr<int64_t>(), ctx.n_bins, ctx.ngroups,
+        ctx.scale_g, ctx.scale_h, nl_dev, side, grower_stream());
+  } else {
+    launch_hist_build_fixed(ctx.binned.data_ptr(), ctx.n_rows,
+                            rows.data_ptr<int>(), rows.numel(),
+                            grad.data_ptr<float>(), hess.data_ptr<float>(),
+                            (long long*)hist.data_ptr<int64_t>(), ctx.n_bins,
+                            ctx.ngroups, ctx.scale_g, ctx.scale_h,
+                            grower_stream());
+  }
   if (ctx.has_reduce) {
     py::gil_scoped_acquire gil;
     ctx.reduce_fn(hist);
@@ -132,31 +149,42 @@ void launch_scan_async(GrowCtx& ctx, const torch::Tensor& hists_i64,
   (void)hipEventRecord(job.ev, grower_stream());
 }
 
-// launch a leaf's full split chain (partition → child hists → scan).
-// In distributed mode this blocks on the local left count and the histogram
-// reduce callback, so the caller never speculates there.
+// launch a leaf's full split chain (partition → smaller-child hist →
+// sibling subtraction → scan → async readbacks).  Never blocks the host:
+// the child row range is resolved on device (child-mode hist kernel) and
+// the local left count comes back through the same pinned-buffer + event
+// as the scan result, so speculation works in distributed mode too (the
+// candidate order — and therefore the all_reduce order — is identical on
+// every rank by construction).
 void launch_job(GrowCtx& ctx, LeafCand& leaf, const torch::Tensor& grad,
                 const torch::Tensor& hess) {
   auto job = std::make_shared<SplitJob>();
   const long m = leaf.rows.numel();
-  auto out_rows = torch::empty({m}, leaf.rows.options());
+  job->out_rows = torch::empty({m}, leaf.rows.options());
   launch_partition(ctx.binned.data_ptr(), ctx.n_rows,
                    leaf.rows.data_ptr<int>(), m, leaf.feat, leaf.bin,
-                   out_rows.data_ptr<int>(), ctx.scratch.data_ptr<int>(),
+                   job->out_rows.data_ptr<int>(), ctx.scratch.data_ptr<int>(),
                    ctx.total.data_ptr<int>(), grower_stream());
-  long nl;
-  if (!ctx.distributed && leaf.C < 1.6e7) {
-    nl = (long)leaf.CL;  // exact integer counts, no sync
-  } else {
-    nl = ctx.total.to(torch::kCPU).item<int>();
-  }
-  job->rows_l = out_rows.slice(0, 0, nl);
-  job->rows_r = out_rows.slice(0, nl, m);
 
   const double CL = leaf.CL, CR = leaf.C - leaf.CL;
-  const bool left_small = CL <= CR;
-  auto hist_small =
-      build_hist(ctx, left_small ? job->rows_l : job->rows_r, grad, hess);
+  const bool left_small = CL <= CR;  // by GLOBAL counts: same on all ranks
+  torch::Tensor hist_small;
+  if (!ctx.distributed && leaf.C < 1.6e7) {
+    job->nl_known = (long)leaf.CL;  // exact integer counts on this rank
+    auto rows_small = left_small
+                          ? job->out_rows.slice(0, 0, job->nl_known)
+                          : job->out_rows.slice(0, job->nl_known, m);
+    hist_small = build_hist(ctx, rows_small, grad, hess);
+  } else {
+    hist_small = build_hist(ctx, job->out_rows, grad, hess,
+                            ctx.total.data_ptr<int>(), left_small ? 0 : 1);
+    job->nl_host = torch::empty({1}, torch::TensorOptions()
+                                         .dtype(torch::kInt32)
+                                         .pinned_memory(true));
+    (void)hipMemcpyAsync(job->nl_host.data_ptr<int>(),
+                         ctx.total.data_ptr<int>(), sizeof(int),
+                         hipMemcpyDeviceToHost, grower_stream());
+  }
   auto hist_big = leaf.hist - hist_small;
   job->hist_l = left_small ? hist_small : hist_big;
   job->hist_r = left_small ? hist_big : hist_small;
@@ -267,7 +295,7 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor rows_root,
     // speculate: pre-launch the next-best candidate's chain so its GPU work
     // overlaps this readback (exact commit order preserved; the cache is
     // consumed whenever that leaf is popped)
-    if (!ctx.distributed && n_leaves + 1 < ctx.num_leaves && !heap.empty()) {
+    if (n_leaves + 1 < ctx.num_leaves && !heap.empty()) {
       CandPtr nxt = heap.top();
       if (!nxt->job && splittable(ctx, *nxt))
         launch_job(ctx, *nxt, grad, hess);
@@ -276,6 +304,12 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor rows_root,
     SplitJob& job = *leaf->job;
     (void)hipEventSynchronize(job.ev);
     auto a = job.scan_host.accessor<float, 2>();
+    const long m_parent = job.out_rows.numel();
+    const long nl = job.nl_known >= 0
+                        ? job.nl_known
+                        : (long)job.nl_host.data_ptr<int>()[0];
+    auto rows_l = job.out_rows.slice(0, 0, nl);
+    auto rows_r = job.out_rows.slice(0, nl, m_parent);
 
     const double GL = leaf->GL, HL = leaf->HL, CL = leaf->CL;
     const double GR = leaf->G - GL, HR = leaf->H - HL, CR = leaf->C - CL;
@@ -296,7 +330,7 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor rows_root,
     auto rc = std::make_shared<LeafCand>();
     lc->node_id = lid; rc->node_id = rid;
     lc->depth = rc->depth = leaf->depth + 1;
-    lc->rows = job.rows_l; rc->rows = job.rows_r;
+    lc->rows = rows_l; rc->rows = rows_r;
     lc->hist = job.hist_l; rc->hist = job.hist_r;
     lc->G = GL; lc->H = HL; lc->C = CL;
     rc->G = GR; rc->H = HR; rc->C = CR;

@@ -88,7 +88,21 @@ __global__ void hist_build_fixed_k(const uchar4* __restrict__ binned,
                                    const float* __restrict__ hess,
                                    long long* __restrict__ hist, int n_bins,
                                    int ngroups, long chunk, double scale_g,
-                                   double scale_h) {
+                                   double scale_h,
+                                   const int* __restrict__ nl_dev, int side) {
+  // side >= 0: child-of-partition mode — m is the PARENT row count (an
+  // upper bound for grid sizing) and the actual child range comes from the
+  // device-side left count nl_dev[0]: left = rows[0, nl), right = rows[nl,
+  // m).  Lets the distributed grower enqueue the smaller child's histogram
+  // without a host readback of the local partition count.
+  long base = 0, m_eff = m;
+  if (side >= 0) {
+    const long nl = nl_dev[0];
+    m_eff = (side == 0) ? nl : m - nl;
+    base = (side == 0) ? 0 : nl;
+  }
+  if ((long)blockIdx.x * chunk >= m_eff) return;  // surplus block: no work
+  rows += base;
   extern __shared__ unsigned long long lds64[];  // [GPB*4][n_bins][2]
   const int tid = threadIdx.x;
   const int nfb = GPB * 4;
@@ -98,7 +112,7 @@ __global__ void hist_build_fixed_k(const uchar4* __restrict__ binned,
 
   const int gq0 = blockIdx.y * GPB;
   const long start = (long)blockIdx.x * chunk;
-  const long end = min(start + chunk, m);
+  const long end = min(start + chunk, m_eff);
   constexpr unsigned long long CNT_ONE = 1ull << 44;
 
   constexpr int ILP = 4;  // probe: +10% via deeper load batching
@@ -174,6 +188,31 @@ __global__ void hist_build_fixed_k(const uchar4* __restrict__ binned,
   }
 }
 
+extern "C" void launch_hist_build_fixed_child(
+    const void* binned, long n_rows, const int* rows, long m,
+    const float* grad, const float* hess, long long* hist, int n_bins,
+    int ngroups, double scale_g, double scale_h, const int* nl_dev, int side,
+    hipStream_t stream) {
+  if (m == 0) return;
+  constexpr int GPB = 2;
+  const int n_fblocks = (ngroups + GPB - 1) / GPB;
+  long chunks = (2048 + n_fblocks - 1) / n_fblocks;
+  long chunk = (m + chunks - 1) / chunks;
+  if (chunk < 16384) {
+    chunk = (m + 7) / 8;
+    if (chunk < 2048) chunk = 2048;
+    if (chunk > 16384) chunk = 16384;
+  }
+  if (chunk > (1l << 19)) chunk = 1l << 19;
+  chunks = (m + chunk - 1) / chunk;  // m = parent size: grid upper bound
+  dim3 grid((unsigned)chunks, (unsigned)n_fblocks);
+  const size_t lds_bytes = (size_t)GPB * 4 * n_bins * 2 * sizeof(long long);
+  hipLaunchKernelGGL((hist_build_fixed_k<GPB>), grid, dim3(256), lds_bytes,
+                     stream, (const uchar4*)binned, n_rows, rows, m, grad,
+                     hess, hist, n_bins, ngroups, chunk, scale_g, scale_h,
+                     nl_dev, side);
+}
+
 extern "C" void launch_hist_build_fixed(const void* binned, long n_rows,
                                         const int* rows, long m,
                                         const float* grad, const float* hess,
@@ -202,7 +241,8 @@ extern "C" void launch_hist_build_fixed(const void* binned, long n_rows,
   const size_t lds_bytes = (size_t)GPB * 4 * n_bins * 2 * sizeof(long long);
   hipLaunchKernelGGL((hist_build_fixed_k<GPB>), grid, dim3(256), lds_bytes,
                      stream, (const uchar4*)binned, n_rows, rows, m, grad,
-                     hess, hist, n_bins, ngroups, chunk, scale_g, scale_h);
+                     hess, hist, n_bins, ngroups, chunk, scale_g, scale_h,
+                     nullptr, -1);
 }
 
 static int hist_gpb_env() {
