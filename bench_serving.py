@@ -20,9 +20,6 @@ def main():
     ap.add_argument("--features", type=int, default=28)
     args = ap.parse_args()
 
-    import requests as rq
-    import torch
-
     from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
     from mmlspark_amd.serving.server import LowLatencyGBDTScorer, ServingServer
 

@@ -9,7 +9,6 @@ import os
 import sys
 import time
 
-import numpy as np
 import torch
 
 

@@ -8,7 +8,6 @@ the analog of FObjTrait (lightgbm/.../params/FObjTrait.scala).
 """
 from __future__ import annotations
 
-import math
 from typing import Callable, Optional, Tuple
 
 import torch

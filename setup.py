@@ -9,7 +9,6 @@ the kernel objects + libamdhip64.
 """
 import os
 import subprocess
-import sys
 
 from setuptools import setup
 from torch.utils import cpp_extension
