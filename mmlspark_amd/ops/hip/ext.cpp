@@ -23,6 +23,10 @@ void launch_bin_matrix(const float*, const float*, long, int, int, int, void*,
 void launch_hist_build_fixed(const void*, long, const int*, long, const float*,
                              const float*, long long*, int, int, double,
                              double, hipStream_t);
+void launch_hist_build_fixed_pair(const void*, long, const int*, long,
+                                  const float*, const float*, long long*,
+                                  int, int, double, double, const int*, int,
+                                  hipStream_t);
 void launch_partition(const void*, long, const int*, long, int, int, int*,
                       int*, int*, hipStream_t);
 void launch_split_scan(const float*, int, long, int, float, float, float,
@@ -59,6 +63,27 @@ torch::Tensor hist_build(torch::Tensor binned_i4, torch::Tensor rows,
                     rows.numel(), grad.data_ptr<float>(),
                     hess.data_ptr<float>(), hist.data_ptr<float>(),
                     (int)n_bins, (int)ngroups, cur_stream());
+  return hist;
+}
+
+torch::Tensor hist_build_fixed_pair(torch::Tensor binned_pair,
+                                    torch::Tensor rows, torch::Tensor grad,
+                                    torch::Tensor hess, long n_bins,
+                                    double scale_g, double scale_h) {
+  CHECK_DEV(binned_pair); CHECK_CONTIG(binned_pair);
+  CHECK_DEV(rows); CHECK_CONTIG(rows);
+  TORCH_CHECK(rows.dtype() == torch::kInt32, "rows must be int32");
+  TORCH_CHECK(binned_pair.dtype() == torch::kInt64, "paired binned is int64");
+  const long npairs = binned_pair.size(0);
+  const long n_rows = binned_pair.size(1);
+  auto hist = torch::zeros({npairs * 8, n_bins, 3},
+                           grad.options().dtype(torch::kInt64));
+  launch_hist_build_fixed_pair(binned_pair.data_ptr(), n_rows,
+                               rows.data_ptr<int>(), rows.numel(),
+                               grad.data_ptr<float>(), hess.data_ptr<float>(),
+                               (long long*)hist.data_ptr<int64_t>(),
+                               (int)n_bins, (int)npairs, scale_g, scale_h,
+                               nullptr, -1, cur_stream());
   return hist;
 }
 
@@ -252,6 +277,8 @@ torch::Tensor tree_shap(torch::Tensor feat, torch::Tensor thr,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hist_build", &hist_build, "per-leaf (feature,bin) grad/hess/count histogram");
+  m.def("hist_build_fixed_pair", &hist_build_fixed_pair,
+        "fixed-point histogram over paired (8-feature) planes");
   m.def("hist_build_fixed", &hist_build_fixed,
         "fixed-point u64 histogram (fast LDS integer atomics)");
   m.def("predict_forest", &predict_forest, "GBDT ensemble raw scores");

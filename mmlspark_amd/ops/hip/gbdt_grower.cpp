@@ -24,10 +24,10 @@ extern "C" {
 void launch_hist_build_fixed(const void*, long, const int*, long, const float*,
                              const float*, long long*, int, int, double,
                              double, hipStream_t);
-void launch_hist_build_fixed_child(const void*, long, const int*, long,
-                                   const float*, const float*, long long*,
-                                   int, int, double, double, const int*, int,
-                                   hipStream_t);
+void launch_hist_build_fixed_pair(const void*, long, const int*, long,
+                                  const float*, const float*, long long*,
+                                  int, int, double, double, const int*, int,
+                                  hipStream_t);
 void launch_split_scan_fixed(const long long*, int, long, int, float, float,
                              float, float, float, long, const bool*, float*,
                              float*, double, double, hipStream_t);
@@ -42,10 +42,11 @@ static hipStream_t grower_stream() {
 namespace {
 
 struct GrowCtx {
-  torch::Tensor binned;
+  torch::Tensor binned;       // (ngroups, n) uchar4 planes — partition kernel
+  torch::Tensor binned_pair;  // (npairs, n) u64 paired planes — hist kernel
   long n_rows;
   int n_bins;
-  int ngroups;
+  int npairs;
   long nf;
   double scale_g, scale_h;
   double l1, l2, min_data, min_hess, min_gain;
@@ -96,24 +97,16 @@ struct CandCmp {
 torch::Tensor build_hist(GrowCtx& ctx, const torch::Tensor& rows,
                          const torch::Tensor& grad, const torch::Tensor& hess,
                          const int* nl_dev = nullptr, int side = -1) {
-  auto hist = torch::zeros({ctx.ngroups * 4, ctx.n_bins, 3},
+  auto hist = torch::zeros({ctx.npairs * 8, ctx.n_bins, 3},
                            grad.options().dtype(torch::kInt64));
-  if (side >= 0) {
-    // child-of-partition mode: rows = parent partition buffer, the child
-    // range [0,nl) or [nl,m) is resolved on DEVICE — no host readback
-    launch_hist_build_fixed_child(
-        ctx.binned.data_ptr(), ctx.n_rows, rows.data_ptr<int>(), rows.numel(),
-        grad.data_ptr<float>(), hess.data_ptr<float>(),
-        (long long*)hist.data_ptr<int64_t>(), ctx.n_bins, ctx.ngroups,
-        ctx.scale_g, ctx.scale_h, nl_dev, side, grower_stream());
-  } else {
-    launch_hist_build_fixed(ctx.binned.data_ptr(), ctx.n_rows,
-                            rows.data_ptr<int>(), rows.numel(),
-                            grad.data_ptr<float>(), hess.data_ptr<float>(),
-                            (long long*)hist.data_ptr<int64_t>(), ctx.n_bins,
-                            ctx.ngroups, ctx.scale_g, ctx.scale_h,
-                            grower_stream());
-  }
+  // paired planes: ONE 8-byte load per row per feature-chunk block (halves
+  // the gather line-fetches of sparse child rows); side >= 0 resolves the
+  // child range [0,nl)/[nl,m) from the DEVICE count — no host readback
+  launch_hist_build_fixed_pair(
+      ctx.binned_pair.data_ptr(), ctx.n_rows, rows.data_ptr<int>(),
+      rows.numel(), grad.data_ptr<float>(), hess.data_ptr<float>(),
+      (long long*)hist.data_ptr<int64_t>(), ctx.n_bins, ctx.npairs,
+      ctx.scale_g, ctx.scale_h, nl_dev, side, grower_stream());
   if (ctx.has_reduce) {
     py::gil_scoped_acquire gil;
     ctx.reduce_fn(hist);
@@ -211,7 +204,8 @@ bool splittable(const GrowCtx& ctx, const LeafCand& c) {
 
 // Returns dict with node arrays (CPU int32/f32 tensors), per-leaf rows
 // (device int32, concatenated) + offsets + leaf node ids.
-py::dict grow_tree_native(torch::Tensor binned, torch::Tensor rows_root,
+py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
+                          torch::Tensor rows_root,
                           torch::Tensor grad, torch::Tensor hess,
                           long n_bins, long nf, double scale_g, double scale_h,
                           double l1, double l2, double min_data,
@@ -221,9 +215,10 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor rows_root,
                           py::object reduce_fn, bool distributed) {
   GrowCtx ctx;
   ctx.binned = binned;
+  ctx.binned_pair = binned_pair;
   ctx.n_rows = binned.size(1);
   ctx.n_bins = (int)n_bins;
-  ctx.ngroups = (int)binned.size(0);
+  ctx.npairs = (int)binned_pair.size(0);
   ctx.nf = nf;
   ctx.scale_g = scale_g;
   ctx.scale_h = scale_h;

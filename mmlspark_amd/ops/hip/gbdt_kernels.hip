@@ -188,6 +188,122 @@ __global__ void hist_build_fixed_k(const uchar4* __restrict__ binned,
   }
 }
 
+// --------------------------------------------- paired-plane fixed histogram
+// binned_pair: (npairs, n_rows) u64, low 4B = group 2p's uchar4, high 4B =
+// group 2p+1's.  A gathered row costs ONE cacheline per feature-chunk block
+// instead of two (the two planes of a GPB=2 block are ~40 MB apart in the
+// plane-major layout, so every sparse row paid two line fetches; measured
+// 2.3x gather penalty on mid-size leaves, tools/hist_gather_probe.py).
+// Trailing-group padding (odd ngroups) accumulates into pad features that
+// the split scan masks out via nf_real.
+__global__ void hist_build_fixed_pair_k(
+    const unsigned long long* __restrict__ binned_pair, long n_rows,
+    const int* __restrict__ rows, long m, const float* __restrict__ grad,
+    const float* __restrict__ hess, long long* __restrict__ hist, int n_bins,
+    int npairs, long chunk, double scale_g, double scale_h,
+    const int* __restrict__ nl_dev, int side) {
+  long base = 0, m_eff = m;
+  if (side >= 0) {
+    const long nl = nl_dev[0];
+    m_eff = (side == 0) ? nl : m - nl;
+    base = (side == 0) ? 0 : nl;
+  }
+  if ((long)blockIdx.x * chunk >= m_eff) return;
+  rows += base;
+  extern __shared__ unsigned long long lds64[];  // [8][n_bins][2]
+  const int tid = threadIdx.x;
+  const int lds_elems = 8 * n_bins * 2;
+  for (int i = tid; i < lds_elems; i += blockDim.x) lds64[i] = 0ull;
+  __syncthreads();
+
+  const int pair = blockIdx.y;
+  const unsigned long long* plane = binned_pair + (size_t)pair * n_rows;
+  const long start = (long)blockIdx.x * chunk;
+  const long end = min(start + chunk, m_eff);
+  constexpr unsigned long long CNT_ONE = 1ull << 44;
+
+  constexpr int ILP = 4;
+  long i = start + tid;
+  for (; i + (ILP - 1) * (long)blockDim.x < end; i += ILP * blockDim.x) {
+    int r[ILP];
+    long long gq[ILP];
+    unsigned long long hq[ILP], v[ILP];
+#pragma unroll
+    for (int u = 0; u < ILP; ++u) r[u] = rows[i + u * blockDim.x];
+#pragma unroll
+    for (int u = 0; u < ILP; ++u) v[u] = plane[r[u]];
+#pragma unroll
+    for (int u = 0; u < ILP; ++u) {
+      gq[u] = (long long)llrint((double)grad[r[u]] * scale_g);
+      hq[u] = CNT_ONE
+              | (unsigned long long)llrint((double)hess[r[u]] * scale_h);
+    }
+#pragma unroll
+    for (int u = 0; u < ILP; ++u) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int b = (int)((v[u] >> (8 * j)) & 0xffull);
+        unsigned long long* cell = &lds64[(j * n_bins + b) * 2];
+        atomicAdd(cell + 0, (unsigned long long)gq[u]);
+        atomicAdd(cell + 1, hq[u]);
+      }
+    }
+  }
+  for (; i < end; i += blockDim.x) {  // tail
+    const int r = rows[i];
+    const unsigned long long v = plane[r];
+    const long long gq = (long long)llrint((double)grad[r] * scale_g);
+    const unsigned long long hq =
+        CNT_ONE | (unsigned long long)llrint((double)hess[r] * scale_h);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int b = (int)((v >> (8 * j)) & 0xffull);
+      unsigned long long* cell = &lds64[(j * n_bins + b) * 2];
+      atomicAdd(cell + 0, (unsigned long long)gq);
+      atomicAdd(cell + 1, hq);
+    }
+  }
+  __syncthreads();
+
+  // flush to (npairs*8, n_bins, 3) int64
+  for (int i = tid; i < 8 * n_bins; i += blockDim.x) {
+    const int f = i / n_bins;
+    const int b = i % n_bins;
+    const unsigned long long gsum = lds64[(f * n_bins + b) * 2 + 0];
+    const unsigned long long hpacked = lds64[(f * n_bins + b) * 2 + 1];
+    if (gsum == 0ull && hpacked == 0ull) continue;
+    const unsigned long long cnt = hpacked >> 44;
+    const unsigned long long hsum = hpacked & ((1ull << 44) - 1ull);
+    long long* out = hist + ((size_t)(pair * 8 + f) * n_bins + b) * 3;
+    atomicAdd((unsigned long long*)(out + 0), gsum);
+    atomicAdd((unsigned long long*)(out + 1), hsum);
+    atomicAdd((unsigned long long*)(out + 2), cnt);
+  }
+}
+
+extern "C" void launch_hist_build_fixed_pair(
+    const void* binned_pair, long n_rows, const int* rows, long m,
+    const float* grad, const float* hess, long long* hist, int n_bins,
+    int npairs, double scale_g, double scale_h, const int* nl_dev, int side,
+    hipStream_t stream) {
+  if (m == 0) return;
+  long chunks = (2048 + npairs - 1) / npairs;
+  long chunk = (m + chunks - 1) / chunks;
+  if (chunk < 16384) {
+    chunk = (m + 7) / 8;
+    if (chunk < 2048) chunk = 2048;
+    if (chunk > 16384) chunk = 16384;
+  }
+  if (chunk > (1l << 19)) chunk = 1l << 19;
+  chunks = (m + chunk - 1) / chunk;
+  dim3 grid((unsigned)chunks, (unsigned)npairs);
+  const size_t lds_bytes = (size_t)8 * n_bins * 2 * sizeof(long long);
+  hipLaunchKernelGGL(hist_build_fixed_pair_k, grid, dim3(256), lds_bytes,
+                     stream, (const unsigned long long*)binned_pair, n_rows, rows, m,
+                     grad, hess, hist, n_bins, npairs, chunk, scale_g,
+                     scale_h, nl_dev, side);
+}
+
 extern "C" void launch_hist_build_fixed_child(
     const void* binned, long n_rows, const int* rows, long m,
     const float* grad, const float* hess, long long* hist, int n_bins,

@@ -398,3 +398,34 @@ def test_native_grower_device_count_path(binary_df):
     p1 = m1.booster.predict_raw(X).cpu()
     p2 = m2.booster.predict_raw(X).cpu()
     assert torch.equal(p1, p2), float((p1 - p2).abs().max())
+
+
+@requires_gpu
+def test_hist_build_fixed_pair_matches_unpaired():
+    """Paired-plane histogram == unpaired fixed-point histogram exactly
+    (same fixed-point integers, one 8B load per row per block)."""
+    from mmlspark_amd.ops import backend
+    g = torch.Generator().manual_seed(13)
+    n, nf, nb = 300_000, 26, 255  # odd group count → zero-padded high plane
+    ngroups = (nf + 3) // 4
+    binned = torch.randint(0, nb, (ngroups, n, 4), generator=g,
+                           dtype=torch.uint8).cuda()
+    g32 = binned.view(torch.int32).reshape(ngroups, -1)
+    if ngroups % 2:
+        g32 = torch.cat([g32, torch.zeros_like(g32[:1])])
+    lo = g32[0::2].to(torch.int64) & 0xFFFFFFFF
+    hi = g32[1::2].to(torch.int64) & 0xFFFFFFFF
+    pair = (lo | (hi << 32)).contiguous()
+    rows = torch.randperm(n, generator=g)[: n // 4].to(
+        torch.int32).sort().values.cuda()
+    grad = torch.randn(n, generator=g).cuda()
+    hess = (torch.rand(n, generator=g) + 0.1).cuda()
+    sg, sh = 2.0 ** 40 / n, 2.0 ** 24
+    ref = backend.hist_build_fixed(binned, rows, grad, hess, nb, sg, sh)
+    out = backend.hist_build_fixed_pair(pair, rows, grad, hess, nb, sg, sh)
+    nf_pad = ngroups * 4
+    assert torch.equal(out[:nf_pad].cpu(), ref.cpu())
+    # zero-padded plane: every row lands in bin 0 of the pad features
+    if out.shape[0] > nf_pad:
+        assert int(out[nf_pad, 0, 2]) == rows.numel()
+        assert int(out[nf_pad, 1:, 2].sum()) == 0
