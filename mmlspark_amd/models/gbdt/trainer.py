@@ -159,6 +159,7 @@ class TreeGrower:
             hi = g32[1::2].to(torch.int64) & 0xFFFFFFFF
             self.binned_pair = (lo | (hi << 32)).contiguous()
             self.nf_pad = self.binned_pair.shape[0] * 8
+            self.tail_bytes = G * 4 - (self.binned_pair.shape[0] - 1) * 8
         self.n_global = n_global or binned_i4.shape[1]
         self.scale_g = 1.0
         self.scale_h = 1.0
@@ -191,6 +192,7 @@ class TreeGrower:
         if self.fixed:
             h = backend.hist_build_fixed_pair(self.binned_pair, rows, grad,
                                               hess, self.cfg.max_bin,
+                                              self.tail_bytes,
                                               self.scale_g, self.scale_h)
         else:
             h = backend.hist_build(self.binned, rows, grad, hess,

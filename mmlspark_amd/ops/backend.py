@@ -51,14 +51,15 @@ def hist_build(binned_i4, rows, grad, hess, n_bins):
     return cpu_ref.hist_build(binned_i4, rows, grad, hess, n_bins)
 
 
-def hist_build_fixed_pair(binned_pair, rows, grad, hess, n_bins, scale_g,
-                          scale_h):
+def hist_build_fixed_pair(binned_pair, rows, grad, hess, n_bins, tail_bytes,
+                          scale_g, scale_h):
     """Fixed-point histogram over PAIRED planes ((npairs, n) int64: two
     uchar4 feature groups per element — one cacheline per gathered row).
+    tail_bytes = valid feature-bytes in the last pair (zero-pad skipped).
     GPU-only (the pairing exists only on the device path)."""
     return _require_ext().hist_build_fixed_pair(binned_pair, rows, grad,
-                                                hess, n_bins, scale_g,
-                                                scale_h)
+                                                hess, n_bins, tail_bytes,
+                                                scale_g, scale_h)
 
 
 def hist_build_fixed(binned_i4, rows, grad, hess, n_bins, scale_g, scale_h):

@@ -25,8 +25,8 @@ void launch_hist_build_fixed(const void*, long, const int*, long, const float*,
                              double, hipStream_t);
 void launch_hist_build_fixed_pair(const void*, long, const int*, long,
                                   const float*, const float*, long long*,
-                                  int, int, double, double, const int*, int,
-                                  hipStream_t);
+                                  int, int, int, double, double, const int*,
+                                  int, hipStream_t);
 void launch_partition(const void*, long, const int*, long, int, int, int*,
                       int*, int*, hipStream_t);
 void launch_split_scan(const float*, int, long, int, float, float, float,
@@ -69,6 +69,7 @@ torch::Tensor hist_build(torch::Tensor binned_i4, torch::Tensor rows,
 torch::Tensor hist_build_fixed_pair(torch::Tensor binned_pair,
                                     torch::Tensor rows, torch::Tensor grad,
                                     torch::Tensor hess, long n_bins,
+                                    long tail_bytes,
                                     double scale_g, double scale_h) {
   CHECK_DEV(binned_pair); CHECK_CONTIG(binned_pair);
   CHECK_DEV(rows); CHECK_CONTIG(rows);
@@ -82,8 +83,8 @@ torch::Tensor hist_build_fixed_pair(torch::Tensor binned_pair,
                                rows.data_ptr<int>(), rows.numel(),
                                grad.data_ptr<float>(), hess.data_ptr<float>(),
                                (long long*)hist.data_ptr<int64_t>(),
-                               (int)n_bins, (int)npairs, scale_g, scale_h,
-                               nullptr, -1, cur_stream());
+                               (int)n_bins, (int)npairs, (int)tail_bytes,
+                               scale_g, scale_h, nullptr, -1, cur_stream());
   return hist;
 }
 

@@ -26,8 +26,8 @@ void launch_hist_build_fixed(const void*, long, const int*, long, const float*,
                              double, hipStream_t);
 void launch_hist_build_fixed_pair(const void*, long, const int*, long,
                                   const float*, const float*, long long*,
-                                  int, int, double, double, const int*, int,
-                                  hipStream_t);
+                                  int, int, int, double, double, const int*,
+                                  int, hipStream_t);
 void launch_split_scan_fixed(const long long*, int, long, int, float, float,
                              float, float, float, long, const bool*, float*,
                              float*, double, double, hipStream_t);
@@ -47,6 +47,7 @@ struct GrowCtx {
   long n_rows;
   int n_bins;
   int npairs;
+  int tail_bytes;  // valid feature-bytes in the last pair (zero-pad skipped)
   long nf;
   double scale_g, scale_h;
   double l1, l2, min_data, min_hess, min_gain;
@@ -106,7 +107,8 @@ torch::Tensor build_hist(GrowCtx& ctx, const torch::Tensor& rows,
       ctx.binned_pair.data_ptr(), ctx.n_rows, rows.data_ptr<int>(),
       rows.numel(), grad.data_ptr<float>(), hess.data_ptr<float>(),
       (long long*)hist.data_ptr<int64_t>(), ctx.n_bins, ctx.npairs,
-      ctx.scale_g, ctx.scale_h, nl_dev, side, grower_stream());
+      ctx.tail_bytes, ctx.scale_g, ctx.scale_h, nl_dev, side,
+      grower_stream());
   if (ctx.has_reduce) {
     py::gil_scoped_acquire gil;
     ctx.reduce_fn(hist);
@@ -219,6 +221,7 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
   ctx.n_rows = binned.size(1);
   ctx.n_bins = (int)n_bins;
   ctx.npairs = (int)binned_pair.size(0);
+  ctx.tail_bytes = (int)(binned.size(0) * 4 - (ctx.npairs - 1) * 8);
   ctx.nf = nf;
   ctx.scale_g = scale_g;
   ctx.scale_h = scale_h;

@@ -200,7 +200,7 @@ __global__ void hist_build_fixed_pair_k(
     const unsigned long long* __restrict__ binned_pair, long n_rows,
     const int* __restrict__ rows, long m, const float* __restrict__ grad,
     const float* __restrict__ hess, long long* __restrict__ hist, int n_bins,
-    int npairs, long chunk, double scale_g, double scale_h,
+    int npairs, int tail_bytes, long chunk, double scale_g, double scale_h,
     const int* __restrict__ nl_dev, int side) {
   long base = 0, m_eff = m;
   if (side >= 0) {
@@ -217,6 +217,9 @@ __global__ void hist_build_fixed_pair_k(
   __syncthreads();
 
   const int pair = blockIdx.y;
+  // zero-padded bytes of the last pair would funnel EVERY row into bin 0 of
+  // the pad features — a same-address LDS-atomic hotspot — so skip them
+  const int jmax = (pair == npairs - 1) ? tail_bytes : 8;
   const unsigned long long* plane = binned_pair + (size_t)pair * n_rows;
   const long start = (long)blockIdx.x * chunk;
   const long end = min(start + chunk, m_eff);
@@ -242,6 +245,7 @@ __global__ void hist_build_fixed_pair_k(
     for (int u = 0; u < ILP; ++u) {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
+        if (j >= jmax) break;
         const int b = (int)((v[u] >> (8 * j)) & 0xffull);
         unsigned long long* cell = &lds64[(j * n_bins + b) * 2];
         atomicAdd(cell + 0, (unsigned long long)gq[u]);
@@ -257,6 +261,7 @@ __global__ void hist_build_fixed_pair_k(
         CNT_ONE | (unsigned long long)llrint((double)hess[r] * scale_h);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
+      if (j >= jmax) break;
       const int b = (int)((v >> (8 * j)) & 0xffull);
       unsigned long long* cell = &lds64[(j * n_bins + b) * 2];
       atomicAdd(cell + 0, (unsigned long long)gq);
@@ -266,7 +271,7 @@ __global__ void hist_build_fixed_pair_k(
   __syncthreads();
 
   // flush to (npairs*8, n_bins, 3) int64
-  for (int i = tid; i < 8 * n_bins; i += blockDim.x) {
+  for (int i = tid; i < jmax * n_bins; i += blockDim.x) {
     const int f = i / n_bins;
     const int b = i % n_bins;
     const unsigned long long gsum = lds64[(f * n_bins + b) * 2 + 0];
@@ -284,8 +289,8 @@ __global__ void hist_build_fixed_pair_k(
 extern "C" void launch_hist_build_fixed_pair(
     const void* binned_pair, long n_rows, const int* rows, long m,
     const float* grad, const float* hess, long long* hist, int n_bins,
-    int npairs, double scale_g, double scale_h, const int* nl_dev, int side,
-    hipStream_t stream) {
+    int npairs, int tail_bytes, double scale_g, double scale_h,
+    const int* nl_dev, int side, hipStream_t stream) {
   if (m == 0) return;
   long chunks = (2048 + npairs - 1) / npairs;
   long chunk = (m + chunks - 1) / chunks;
@@ -299,9 +304,9 @@ extern "C" void launch_hist_build_fixed_pair(
   dim3 grid((unsigned)chunks, (unsigned)npairs);
   const size_t lds_bytes = (size_t)8 * n_bins * 2 * sizeof(long long);
   hipLaunchKernelGGL(hist_build_fixed_pair_k, grid, dim3(256), lds_bytes,
-                     stream, (const unsigned long long*)binned_pair, n_rows, rows, m,
-                     grad, hess, hist, n_bins, npairs, chunk, scale_g,
-                     scale_h, nl_dev, side);
+                     stream, (const unsigned long long*)binned_pair, n_rows,
+                     rows, m, grad, hess, hist, n_bins, npairs, tail_bytes,
+                     chunk, scale_g, scale_h, nl_dev, side);
 }
 
 extern "C" void launch_hist_build_fixed_child(
