@@ -24,9 +24,9 @@ void launch_hist_build_fixed(const void*, long, const int*, long, const float*,
                              const float*, long long*, int, int, double,
                              double, hipStream_t);
 void launch_hist_build_fixed_pair(const void*, long, const int*, long,
-                                  const float*, const float*, long long*,
-                                  int, int, int, double, double, const int*,
-                                  int, hipStream_t);
+                                  const float*, const float*, const void*,
+                                  long long*, int, int, int, double, double,
+                                  const int*, int, hipStream_t);
 void launch_partition(const void*, long, const int*, long, int, int, int*,
                       int*, int*, hipStream_t);
 void launch_split_scan(const float*, int, long, int, float, float, float,
@@ -82,6 +82,7 @@ torch::Tensor hist_build_fixed_pair(torch::Tensor binned_pair,
   launch_hist_build_fixed_pair(binned_pair.data_ptr(), n_rows,
                                rows.data_ptr<int>(), rows.numel(),
                                grad.data_ptr<float>(), hess.data_ptr<float>(),
+                               nullptr,
                                (long long*)hist.data_ptr<int64_t>(),
                                (int)n_bins, (int)npairs, (int)tail_bytes,
                                scale_g, scale_h, nullptr, -1, cur_stream());

@@ -25,9 +25,11 @@ void launch_hist_build_fixed(const void*, long, const int*, long, const float*,
                              const float*, long long*, int, int, double,
                              double, hipStream_t);
 void launch_hist_build_fixed_pair(const void*, long, const int*, long,
-                                  const float*, const float*, long long*,
-                                  int, int, int, double, double, const int*,
-                                  int, hipStream_t);
+                                  const float*, const float*, const void*,
+                                  long long*, int, int, int, double, double,
+                                  const int*, int, hipStream_t);
+void launch_quantize_gh(const float*, const float*, long, double, double,
+                        void*, hipStream_t);
 void launch_split_scan_fixed(const long long*, int, long, int, float, float,
                              float, float, float, long, const bool*, float*,
                              float*, double, double, hipStream_t);
@@ -48,6 +50,7 @@ struct GrowCtx {
   int n_bins;
   int npairs;
   int tail_bytes;  // valid feature-bytes in the last pair (zero-pad skipped)
+  torch::Tensor ghq;  // (n, 2) i64 per-row pre-quantized (gq, CNT|hq)
   long nf;
   double scale_g, scale_h;
   double l1, l2, min_data, min_hess, min_gain;
@@ -106,6 +109,7 @@ torch::Tensor build_hist(GrowCtx& ctx, const torch::Tensor& rows,
   launch_hist_build_fixed_pair(
       ctx.binned_pair.data_ptr(), ctx.n_rows, rows.data_ptr<int>(),
       rows.numel(), grad.data_ptr<float>(), hess.data_ptr<float>(),
+      ctx.ghq.data_ptr(),
       (long long*)hist.data_ptr<int64_t>(), ctx.n_bins, ctx.npairs,
       ctx.tail_bytes, ctx.scale_g, ctx.scale_h, nl_dev, side,
       grower_stream());
@@ -238,6 +242,13 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
 
   ctx.scratch = torch::empty({4096}, rows_root.options().dtype(torch::kInt32));
   ctx.total = torch::zeros({1}, rows_root.options().dtype(torch::kInt32));
+  // pre-quantize grad/hess once per tree: the hist kernel then gathers ONE
+  // 16B record per row instead of two float arrays
+  ctx.ghq = torch::empty({ctx.n_rows, 2},
+                         rows_root.options().dtype(torch::kInt64));
+  launch_quantize_gh(grad.data_ptr<float>(), hess.data_ptr<float>(),
+                     ctx.n_rows, ctx.scale_g, ctx.scale_h,
+                     ctx.ghq.data_ptr(), grower_stream());
 
   std::vector<int> feature_, thr_bin_, left_, right_, leaf_idx_;
   std::vector<float> value_, count_, gain_;

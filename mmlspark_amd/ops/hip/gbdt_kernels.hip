@@ -196,10 +196,17 @@ __global__ void hist_build_fixed_k(const uchar4* __restrict__ binned,
 // 2.3x gather penalty on mid-size leaves, tools/hist_gather_probe.py).
 // Trailing-group padding (odd ngroups) accumulates into pad features that
 // the split scan masks out via nf_real.
+// GHQ=true: grad/hess come pre-quantized as one 16B record per row
+// ((gq:i64, CNT|hq:u64) in ghq) — each pair-block then gathers 2 cachelines
+// per sparse row (binned + ghq) instead of 3 (binned + grad + hess), and
+// the double->fixed conversion runs once per row instead of once per
+// (row, pair-block).
+template <bool GHQ>
 __global__ void hist_build_fixed_pair_k(
     const unsigned long long* __restrict__ binned_pair, long n_rows,
     const int* __restrict__ rows, long m, const float* __restrict__ grad,
-    const float* __restrict__ hess, long long* __restrict__ hist, int n_bins,
+    const float* __restrict__ hess, const longlong2* __restrict__ ghq,
+    long long* __restrict__ hist, int n_bins,
     int npairs, int tail_bytes, long chunk, double scale_g, double scale_h,
     const int* __restrict__ nl_dev, int side) {
   long base = 0, m_eff = m;
@@ -237,9 +244,15 @@ __global__ void hist_build_fixed_pair_k(
     for (int u = 0; u < ILP; ++u) v[u] = plane[r[u]];
 #pragma unroll
     for (int u = 0; u < ILP; ++u) {
-      gq[u] = (long long)llrint((double)grad[r[u]] * scale_g);
-      hq[u] = CNT_ONE
-              | (unsigned long long)llrint((double)hess[r[u]] * scale_h);
+      if (GHQ) {
+        const longlong2 q = ghq[r[u]];
+        gq[u] = q.x;
+        hq[u] = (unsigned long long)q.y;
+      } else {
+        gq[u] = (long long)llrint((double)grad[r[u]] * scale_g);
+        hq[u] = CNT_ONE
+                | (unsigned long long)llrint((double)hess[r[u]] * scale_h);
+      }
     }
 #pragma unroll
     for (int u = 0; u < ILP; ++u) {
@@ -256,9 +269,16 @@ __global__ void hist_build_fixed_pair_k(
   for (; i < end; i += blockDim.x) {  // tail
     const int r = rows[i];
     const unsigned long long v = plane[r];
-    const long long gq = (long long)llrint((double)grad[r] * scale_g);
-    const unsigned long long hq =
-        CNT_ONE | (unsigned long long)llrint((double)hess[r] * scale_h);
+    long long gq;
+    unsigned long long hq;
+    if (GHQ) {
+      const longlong2 q = ghq[r];
+      gq = q.x;
+      hq = (unsigned long long)q.y;
+    } else {
+      gq = (long long)llrint((double)grad[r] * scale_g);
+      hq = CNT_ONE | (unsigned long long)llrint((double)hess[r] * scale_h);
+    }
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       if (j >= jmax) break;
@@ -286,10 +306,34 @@ __global__ void hist_build_fixed_pair_k(
   }
 }
 
+// one-shot per tree: quantize grad/hess to (gq, CNT|hq) 16B records
+__global__ void quantize_gh_k(const float* __restrict__ grad,
+                              const float* __restrict__ hess, long n,
+                              double scale_g, double scale_h,
+                              longlong2* __restrict__ out) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  longlong2 q;
+  q.x = (long long)llrint((double)grad[i] * scale_g);
+  q.y = (long long)((1ull << 44)
+                    | (unsigned long long)llrint((double)hess[i] * scale_h));
+  out[i] = q;
+}
+
+extern "C" void launch_quantize_gh(const float* grad, const float* hess,
+                                   long n, double scale_g, double scale_h,
+                                   void* out, hipStream_t stream) {
+  if (n == 0) return;
+  const long blocks = (n + 255) / 256;
+  hipLaunchKernelGGL(quantize_gh_k, dim3((unsigned)blocks), dim3(256), 0,
+                     stream, grad, hess, n, scale_g, scale_h,
+                     (longlong2*)out);
+}
+
 extern "C" void launch_hist_build_fixed_pair(
     const void* binned_pair, long n_rows, const int* rows, long m,
-    const float* grad, const float* hess, long long* hist, int n_bins,
-    int npairs, int tail_bytes, double scale_g, double scale_h,
+    const float* grad, const float* hess, const void* ghq, long long* hist,
+    int n_bins, int npairs, int tail_bytes, double scale_g, double scale_h,
     const int* nl_dev, int side, hipStream_t stream) {
   if (m == 0) return;
   long chunks = (2048 + npairs - 1) / npairs;
@@ -303,10 +347,20 @@ extern "C" void launch_hist_build_fixed_pair(
   chunks = (m + chunk - 1) / chunk;
   dim3 grid((unsigned)chunks, (unsigned)npairs);
   const size_t lds_bytes = (size_t)8 * n_bins * 2 * sizeof(long long);
-  hipLaunchKernelGGL(hist_build_fixed_pair_k, grid, dim3(256), lds_bytes,
-                     stream, (const unsigned long long*)binned_pair, n_rows,
-                     rows, m, grad, hess, hist, n_bins, npairs, tail_bytes,
-                     chunk, scale_g, scale_h, nl_dev, side);
+  if (ghq) {
+    hipLaunchKernelGGL(hist_build_fixed_pair_k<true>, grid, dim3(256),
+                       lds_bytes, stream,
+                       (const unsigned long long*)binned_pair, n_rows, rows,
+                       m, grad, hess, (const longlong2*)ghq, hist, n_bins,
+                       npairs, tail_bytes, chunk, scale_g, scale_h, nl_dev,
+                       side);
+  } else {
+    hipLaunchKernelGGL(hist_build_fixed_pair_k<false>, grid, dim3(256),
+                       lds_bytes, stream,
+                       (const unsigned long long*)binned_pair, n_rows, rows,
+                       m, grad, hess, nullptr, hist, n_bins, npairs,
+                       tail_bytes, chunk, scale_g, scale_h, nl_dev, side);
+  }
 }
 
 extern "C" void launch_hist_build_fixed_child(
