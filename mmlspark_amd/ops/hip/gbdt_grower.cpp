@@ -1,16 +1,32 @@
 // Native leaf-wise tree-growth driver — the C++ runtime piece of the GBDT.
 //
-// Runs the entire leaf-wise loop (ordered partition → child histogram →
-// sibling subtraction → fused split scan → readback) in C++ with ONE Python
-// call per tree.  The split-processing of any heap leaf is independent of
-// processing order (its split was fixed when the leaf was created), so the
-// driver SPECULATIVELY launches the next-best candidate's whole chain while
-// the host waits on the current readback — results are cached on the leaf
-// and consumed when it is popped, keeping exact leaf-wise commit order while
-// pipelining the GPU work (per-job pinned scan buffers + HIP events).
-// Multi-rank mode stays serial (the histogram all_reduce callback and the
-// local-count readback are ordering barriers).  Categorical features fall
-// back to the Python grower (models/gbdt/trainer.py).
+// Runs the entire leaf-wise loop in C++ with ONE Python call per tree,
+// organized around two ideas:
+//
+//  * ARENA MODE: the tree's working set (paired bin planes + pre-quantized
+//    (gq,hq) records + original row ids) lives in a per-tree DOUBLE-BUFFERED
+//    arena.  Every split scatters the parent's segment into the other buffer
+//    (stable left|right), so child histograms read CONTIGUOUS rows — the
+//    sparse row gather that bounded child hist builds (measured 2.3x,
+//    tools/hist_gather_probe.py) is gone; the cost is streaming ~124 B/row
+//    of arena payload per split, which HBM3E absorbs.  Live leaves always
+//    own disjoint row ranges, so two in-flight splits never overlap even
+//    across buffers.
+//
+//  * SPECULATION: a popped leaf's split work is independent of processing
+//    order (its split was fixed at creation), so the driver pre-launches the
+//    next-best candidate's whole chain (partition → smaller-child hist →
+//    sibling subtraction → scan → pinned readback behind a HIP event) while
+//    the host waits on the current readback.  Commit order stays exactly
+//    leaf-wise.  The chain never blocks the host: child ranges resolve from
+//    the DEVICE-side left count, which rides back with the scan result.
+//    Candidate order is deterministic across ranks, so the histogram
+//    all_reduce order matches and speculation is safe in multi-rank mode.
+//
+// Categorical features fall back to the Python grower (models/gbdt/
+// trainer.py).  Replaces the growth loop the reference delegates to
+// LightGBM's serial_tree_learner behind LGBM_BoosterUpdateOneIter
+// (SURVEY §2.1).
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
@@ -21,20 +37,18 @@
 #include <vector>
 
 extern "C" {
-void launch_hist_build_fixed(const void*, long, const int*, long, const float*,
-                             const float*, long long*, int, int, double,
-                             double, hipStream_t);
-void launch_hist_build_fixed_pair(const void*, long, const int*, long,
-                                  const float*, const float*, const void*,
-                                  long long*, int, int, int, double, double,
-                                  const int*, int, hipStream_t);
-void launch_quantize_gh(const float*, const float*, long, double, double,
-                        void*, hipStream_t);
+void launch_arena_gather(const void*, long, const float*, const float*,
+                         const int*, long, int, double, double, void*, void*,
+                         int*, hipStream_t);
+void launch_partition_arena(const void*, const void*, const int*, void*,
+                            void*, int*, long, int, long, long, int, int,
+                            int*, int*, hipStream_t);
+void launch_hist_pair_range(const void*, const void*, long, long, long,
+                            long long*, int, int, int, const int*, int,
+                            hipStream_t);
 void launch_split_scan_fixed(const long long*, int, long, int, float, float,
                              float, float, float, long, const bool*, float*,
                              float*, double, double, hipStream_t);
-void launch_partition(const void*, long, const int*, long, int, int, int*,
-                      int*, int*, hipStream_t);
 }
 
 static hipStream_t grower_stream() {
@@ -44,13 +58,14 @@ static hipStream_t grower_stream() {
 namespace {
 
 struct GrowCtx {
-  torch::Tensor binned;       // (ngroups, n) uchar4 planes — partition kernel
-  torch::Tensor binned_pair;  // (npairs, n) u64 paired planes — hist kernel
-  long n_rows;
+  long n_arena;    // rows in the arena (= rows_root size after bagging)
   int n_bins;
   int npairs;
   int tail_bytes;  // valid feature-bytes in the last pair (zero-pad skipped)
-  torch::Tensor ghq;  // (n, 2) i64 per-row pre-quantized (gq, CNT|hq)
+  // double-buffered arena: [0]/[1]
+  torch::Tensor pair[2];   // (npairs, n_arena) i64 paired bin planes
+  torch::Tensor ghq[2];    // (n_arena, 2) i64 (gq, CNT|hq)
+  torch::Tensor rowid[2];  // (n_arena,) i32 original row ids
   long nf;
   double scale_g, scale_h;
   double l1, l2, min_data, min_hess, min_gain;
@@ -64,7 +79,6 @@ struct GrowCtx {
 };
 
 struct SplitJob {
-  torch::Tensor out_rows;   // partitioned rows (left | right), device
   torch::Tensor hist_l, hist_r;
   torch::Tensor scan_host;  // (2,6) f32 pinned
   torch::Tensor nl_host;    // (1,) i32 pinned — local left count readback
@@ -81,7 +95,8 @@ struct LeafCand {
   long seq = 0;
   int node_id = 0;
   int depth = 0;
-  torch::Tensor rows;
+  long lo = 0, hi = 0;  // arena row range
+  int buf = 0;          // which arena buffer holds this leaf's rows
   torch::Tensor hist;   // int64 (nf_pad, nb, 3), globally reduced
   double G = 0, H = 0, C = 0;
   double GL = 0, HL = 0, CL = 0;
@@ -98,21 +113,16 @@ struct CandCmp {
   }
 };
 
-torch::Tensor build_hist(GrowCtx& ctx, const torch::Tensor& rows,
-                         const torch::Tensor& grad, const torch::Tensor& hess,
+// histogram over an arena segment [lo+?, ...) of buffer `buf`; side >= 0
+// resolves the child sub-range from the device left count (no host sync)
+torch::Tensor build_hist(GrowCtx& ctx, int buf, long lo, long m,
                          const int* nl_dev = nullptr, int side = -1) {
   auto hist = torch::zeros({ctx.npairs * 8, ctx.n_bins, 3},
-                           grad.options().dtype(torch::kInt64));
-  // paired planes: ONE 8-byte load per row per feature-chunk block (halves
-  // the gather line-fetches of sparse child rows); side >= 0 resolves the
-  // child range [0,nl)/[nl,m) from the DEVICE count — no host readback
-  launch_hist_build_fixed_pair(
-      ctx.binned_pair.data_ptr(), ctx.n_rows, rows.data_ptr<int>(),
-      rows.numel(), grad.data_ptr<float>(), hess.data_ptr<float>(),
-      ctx.ghq.data_ptr(),
+                           ctx.pair[buf].options());
+  launch_hist_pair_range(
+      ctx.pair[buf].data_ptr(), ctx.ghq[buf].data_ptr(), ctx.n_arena, lo, m,
       (long long*)hist.data_ptr<int64_t>(), ctx.n_bins, ctx.npairs,
-      ctx.tail_bytes, ctx.scale_g, ctx.scale_h, nl_dev, side,
-      grower_stream());
+      ctx.tail_bytes, nl_dev, side, grower_stream());
   if (ctx.has_reduce) {
     py::gil_scoped_acquire gil;
     ctx.reduce_fn(hist);
@@ -148,34 +158,28 @@ void launch_scan_async(GrowCtx& ctx, const torch::Tensor& hists_i64,
   (void)hipEventRecord(job.ev, grower_stream());
 }
 
-// launch a leaf's full split chain (partition → smaller-child hist →
-// sibling subtraction → scan → async readbacks).  Never blocks the host:
-// the child row range is resolved on device (child-mode hist kernel) and
-// the local left count comes back through the same pinned-buffer + event
-// as the scan result, so speculation works in distributed mode too (the
-// candidate order — and therefore the all_reduce order — is identical on
-// every rank by construction).
-void launch_job(GrowCtx& ctx, LeafCand& leaf, const torch::Tensor& grad,
-                const torch::Tensor& hess) {
+void launch_job(GrowCtx& ctx, LeafCand& leaf) {
   auto job = std::make_shared<SplitJob>();
-  const long m = leaf.rows.numel();
-  job->out_rows = torch::empty({m}, leaf.rows.options());
-  launch_partition(ctx.binned.data_ptr(), ctx.n_rows,
-                   leaf.rows.data_ptr<int>(), m, leaf.feat, leaf.bin,
-                   job->out_rows.data_ptr<int>(), ctx.scratch.data_ptr<int>(),
-                   ctx.total.data_ptr<int>(), grower_stream());
+  const long m = leaf.hi - leaf.lo;
+  const int src = leaf.buf, dst = leaf.buf ^ 1;
+  launch_partition_arena(
+      ctx.pair[src].data_ptr(), ctx.ghq[src].data_ptr(),
+      ctx.rowid[src].data_ptr<int>(), ctx.pair[dst].data_ptr(),
+      ctx.ghq[dst].data_ptr(), ctx.rowid[dst].data_ptr<int>(), ctx.n_arena,
+      ctx.npairs, leaf.lo, m, leaf.feat, leaf.bin,
+      ctx.scratch.data_ptr<int>(), ctx.total.data_ptr<int>(),
+      grower_stream());
 
   const double CL = leaf.CL, CR = leaf.C - leaf.CL;
   const bool left_small = CL <= CR;  // by GLOBAL counts: same on all ranks
   torch::Tensor hist_small;
   if (!ctx.distributed && leaf.C < 1.6e7) {
     job->nl_known = (long)leaf.CL;  // exact integer counts on this rank
-    auto rows_small = left_small
-                          ? job->out_rows.slice(0, 0, job->nl_known)
-                          : job->out_rows.slice(0, job->nl_known, m);
-    hist_small = build_hist(ctx, rows_small, grad, hess);
+    const long lo_s = left_small ? leaf.lo : leaf.lo + job->nl_known;
+    const long m_s = left_small ? job->nl_known : m - job->nl_known;
+    hist_small = build_hist(ctx, dst, lo_s, m_s);
   } else {
-    hist_small = build_hist(ctx, job->out_rows, grad, hess,
+    hist_small = build_hist(ctx, dst, leaf.lo, m,
                             ctx.total.data_ptr<int>(), left_small ? 0 : 1);
     job->nl_host = torch::empty({1}, torch::TensorOptions()
                                          .dtype(torch::kInt32)
@@ -220,9 +224,8 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
                           c10::optional<torch::Tensor> feat_mask,
                           py::object reduce_fn, bool distributed) {
   GrowCtx ctx;
-  ctx.binned = binned;
-  ctx.binned_pair = binned_pair;
-  ctx.n_rows = binned.size(1);
+  const long n_full = binned_pair.size(1);
+  ctx.n_arena = rows_root.numel();
   ctx.n_bins = (int)n_bins;
   ctx.npairs = (int)binned_pair.size(0);
   ctx.tail_bytes = (int)(binned.size(0) * 4 - (ctx.npairs - 1) * 8);
@@ -242,13 +245,21 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
 
   ctx.scratch = torch::empty({4096}, rows_root.options().dtype(torch::kInt32));
   ctx.total = torch::zeros({1}, rows_root.options().dtype(torch::kInt32));
-  // pre-quantize grad/hess once per tree: the hist kernel then gathers ONE
-  // 16B record per row instead of two float arrays
-  ctx.ghq = torch::empty({ctx.n_rows, 2},
-                         rows_root.options().dtype(torch::kInt64));
-  launch_quantize_gh(grad.data_ptr<float>(), hess.data_ptr<float>(),
-                     ctx.n_rows, ctx.scale_g, ctx.scale_h,
-                     ctx.ghq.data_ptr(), grower_stream());
+  auto i64d = rows_root.options().dtype(torch::kInt64);
+  for (int b = 0; b < 2; ++b) {
+    ctx.pair[b] = torch::empty({ctx.npairs, ctx.n_arena}, i64d);
+    ctx.ghq[b] = torch::empty({ctx.n_arena, 2}, i64d);
+    ctx.rowid[b] = torch::empty({ctx.n_arena},
+                                rows_root.options().dtype(torch::kInt32));
+  }
+  // materialize arena buffer 0: bin pairs + quantized (gq,hq) + row ids for
+  // the (possibly bagged) root row set
+  launch_arena_gather(binned_pair.data_ptr(), n_full,
+                      grad.data_ptr<float>(), hess.data_ptr<float>(),
+                      rows_root.data_ptr<int>(), ctx.n_arena, ctx.npairs,
+                      ctx.scale_g, ctx.scale_h, ctx.pair[0].data_ptr(),
+                      ctx.ghq[0].data_ptr(), ctx.rowid[0].data_ptr<int>(),
+                      grower_stream());
 
   std::vector<int> feature_, thr_bin_, left_, right_, leaf_idx_;
   std::vector<float> value_, count_, gain_;
@@ -259,7 +270,7 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
     return (int)feature_.size() - 1;
   };
 
-  auto root_hist = build_hist(ctx, rows_root, grad, hess);
+  auto root_hist = build_hist(ctx, 0, 0, ctx.n_arena);
   auto sums = root_hist.select(0, 0).sum(0).to(torch::kCPU);
   auto sa = sums.accessor<int64_t, 1>();
   const double G0 = (double)sa[0] / scale_g;
@@ -268,7 +279,7 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
 
   auto root = std::make_shared<LeafCand>();
   root->node_id = new_node();
-  root->rows = rows_root;
+  root->lo = 0; root->hi = ctx.n_arena; root->buf = 0;
   root->hist = root_hist;
   root->G = G0; root->H = H0; root->C = C0;
   {
@@ -300,25 +311,23 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
         break;
       }
 
-    if (!leaf->job) launch_job(ctx, *leaf, grad, hess);
+    if (!leaf->job) launch_job(ctx, *leaf);
     // speculate: pre-launch the next-best candidate's chain so its GPU work
     // overlaps this readback (exact commit order preserved; the cache is
-    // consumed whenever that leaf is popped)
+    // consumed whenever that leaf is popped; live leaves own disjoint row
+    // ranges so in-flight scatters never collide)
     if (n_leaves + 1 < ctx.num_leaves && !heap.empty()) {
       CandPtr nxt = heap.top();
-      if (!nxt->job && splittable(ctx, *nxt))
-        launch_job(ctx, *nxt, grad, hess);
+      if (!nxt->job && splittable(ctx, *nxt)) launch_job(ctx, *nxt);
     }
 
     SplitJob& job = *leaf->job;
     (void)hipEventSynchronize(job.ev);
     auto a = job.scan_host.accessor<float, 2>();
-    const long m_parent = job.out_rows.numel();
+    const long m_parent = leaf->hi - leaf->lo;
     const long nl = job.nl_known >= 0
                         ? job.nl_known
                         : (long)job.nl_host.data_ptr<int>()[0];
-    auto rows_l = job.out_rows.slice(0, 0, nl);
-    auto rows_r = job.out_rows.slice(0, nl, m_parent);
 
     const double GL = leaf->GL, HL = leaf->HL, CL = leaf->CL;
     const double GR = leaf->G - GL, HR = leaf->H - HL, CR = leaf->C - CL;
@@ -339,7 +348,9 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
     auto rc = std::make_shared<LeafCand>();
     lc->node_id = lid; rc->node_id = rid;
     lc->depth = rc->depth = leaf->depth + 1;
-    lc->rows = rows_l; rc->rows = rows_r;
+    lc->buf = rc->buf = leaf->buf ^ 1;
+    lc->lo = leaf->lo; lc->hi = leaf->lo + nl;
+    rc->lo = leaf->lo + nl; rc->hi = leaf->lo + m_parent;
     lc->hist = job.hist_l; rc->hist = job.hist_r;
     lc->G = GL; lc->H = HL; lc->C = CL;
     rc->G = GR; rc->H = HR; rc->C = CR;
@@ -358,7 +369,8 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
     n_leaves += 1;
   }
 
-  // leaf ordinals in node order; collect row segments
+  // leaf ordinals in node order; collect original-row-id segments from each
+  // leaf's own arena buffer
   std::sort(finals.begin(), finals.end(),
             [](const CandPtr& a, const CandPtr& b) {
               return a->node_id < b->node_id;
@@ -367,7 +379,8 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
   std::vector<int64_t> seg_nodes;
   for (size_t i = 0; i < finals.size(); ++i) {
     leaf_idx_[finals[i]->node_id] = (int)i;
-    segs.push_back(finals[i]->rows);
+    segs.push_back(ctx.rowid[finals[i]->buf].slice(0, finals[i]->lo,
+                                                   finals[i]->hi));
     seg_nodes.push_back(finals[i]->node_id);
   }
   auto leaf_rows = segs.empty()
@@ -396,5 +409,5 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("grow_tree_native", &grow_tree_native,
-        "native leaf-wise GBDT tree growth (numeric features)");
+        "native leaf-wise GBDT tree growth (numeric features, arena mode)");
 }

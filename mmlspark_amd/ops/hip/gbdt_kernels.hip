@@ -565,6 +565,261 @@ __global__ void part_scatter_k(const uchar4* __restrict__ binned, long n_rows,
   }
 }
 
+// ------------------------------------------------------------- arena mode
+// Per-tree physical repartitioning: the tree's working set lives in a
+// double-buffered ARENA of paired bin planes + pre-quantized (gq,hq)
+// records + original row ids.  Every split SCATTERS the parent's segment
+// into the other buffer (left|right, stable), so child histograms read
+// CONTIGUOUS rows — the sparse row gather that bounded child hist builds
+// (tools/hist_gather_probe.py: 2.3x) disappears entirely, at the cost of
+// moving ~124 B/row/split of arena payload (HBM streaming, ~8 TB/s).
+
+__global__ void arena_gather_k(const unsigned long long* __restrict__ pair_src,
+                               long n_src, const float* __restrict__ grad,
+                               const float* __restrict__ hess,
+                               const int* __restrict__ rows, long m,
+                               int npairs, double scale_g, double scale_h,
+                               unsigned long long* __restrict__ pair_dst,
+                               longlong2* __restrict__ ghq_dst,
+                               int* __restrict__ rowid_dst) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= m) return;
+  const int r = rows[i];
+  rowid_dst[i] = r;
+  longlong2 q;
+  q.x = (long long)llrint((double)grad[r] * scale_g);
+  q.y = (long long)((1ull << 44)
+                    | (unsigned long long)llrint((double)hess[r] * scale_h));
+  ghq_dst[i] = q;
+  for (int p = 0; p < npairs; ++p)
+    pair_dst[(size_t)p * m + i] = pair_src[(size_t)p * n_src + r];
+}
+
+extern "C" void launch_arena_gather(const void* pair_src, long n_src,
+                                    const float* grad, const float* hess,
+                                    const int* rows, long m, int npairs,
+                                    double scale_g, double scale_h,
+                                    void* pair_dst, void* ghq_dst,
+                                    int* rowid_dst, hipStream_t stream) {
+  if (m == 0) return;
+  const long blocks = (m + 255) / 256;
+  hipLaunchKernelGGL(arena_gather_k, dim3((unsigned)blocks), dim3(256), 0,
+                     stream, (const unsigned long long*)pair_src, n_src, grad,
+                     hess, rows, m, npairs, scale_g, scale_h,
+                     (unsigned long long*)pair_dst, (longlong2*)ghq_dst,
+                     rowid_dst);
+}
+
+__global__ void part_arena_count_k(const unsigned long long* __restrict__ pair,
+                                   long n_arena, long lo, long m, int pf,
+                                   int jbyte, int thr, long chunk,
+                                   int* __restrict__ block_counts) {
+  const long start = (long)blockIdx.x * chunk;
+  const long end = min(start + chunk, m);
+  const unsigned long long* plane = pair + (size_t)pf * n_arena + lo;
+  int cnt = 0;
+  for (long i = start + threadIdx.x; i < end; i += blockDim.x)
+    cnt += (int)(((plane[i] >> (8 * jbyte)) & 0xffull) <= (unsigned)thr);
+  __shared__ int sh[256];
+  sh[threadIdx.x] = cnt;
+  __syncthreads();
+  for (int d = 128; d > 0; d >>= 1) {
+    if (threadIdx.x < d) sh[threadIdx.x] += sh[threadIdx.x + d];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) block_counts[blockIdx.x] = sh[0];
+}
+
+__global__ void part_arena_scatter_k(
+    const unsigned long long* __restrict__ pair_src,
+    const longlong2* __restrict__ ghq_src, const int* __restrict__ rowid_src,
+    unsigned long long* __restrict__ pair_dst,
+    longlong2* __restrict__ ghq_dst, int* __restrict__ rowid_dst,
+    long n_arena, int npairs, long lo, long m, int pf, int jbyte, int thr,
+    long chunk, const int* __restrict__ block_offsets,
+    const int* __restrict__ total_left) {
+  const long start = (long)blockIdx.x * chunk;
+  const long end = min(start + chunk, m);
+  const long nl_total = *total_left;
+  __shared__ long base_l, base_r;
+  __shared__ int wave_l[4], wave_r[4];
+  if (threadIdx.x == 0) {
+    base_l = lo + block_offsets[blockIdx.x];
+    base_r = lo + nl_total + (start - block_offsets[blockIdx.x]);
+  }
+  __syncthreads();
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const unsigned long long* split_plane = pair_src + (size_t)pf * n_arena + lo;
+  for (long i0 = start; i0 < end; i0 += blockDim.x) {
+    const long i = i0 + threadIdx.x;
+    bool valid = i < end, left = false;
+    if (valid)
+      left = ((split_plane[i] >> (8 * jbyte)) & 0xffull) <= (unsigned)thr;
+    const unsigned long long mask_l = __ballot(valid && left);
+    const unsigned long long mask_r = __ballot(valid && !left);
+    const unsigned long long lt = (1ull << lane) - 1ull;
+    if (lane == 0) {
+      wave_l[wid] = __popcll(mask_l);
+      wave_r[wid] = __popcll(mask_r);
+    }
+    __syncthreads();
+    long wl = base_l, wr = base_r;
+    for (int w = 0; w < wid; ++w) {
+      wl += wave_l[w];
+      wr += wave_r[w];
+    }
+    if (valid) {
+      const long src = lo + i;
+      const long dst = left ? wl + __popcll(mask_l & lt)
+                            : wr + __popcll(mask_r & lt);
+      rowid_dst[dst] = rowid_src[src];
+      ghq_dst[dst] = ghq_src[src];
+      for (int p = 0; p < npairs; ++p)
+        pair_dst[(size_t)p * n_arena + dst] =
+            pair_src[(size_t)p * n_arena + src];
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      long al = 0, ar = 0;
+      for (int w = 0; w < 4; ++w) {
+        al += wave_l[w];
+        ar += wave_r[w];
+      }
+      base_l += al;
+      base_r += ar;
+    }
+    __syncthreads();
+  }
+}
+
+extern "C" void launch_partition_arena(
+    const void* pair_src, const void* ghq_src, const int* rowid_src,
+    void* pair_dst, void* ghq_dst, int* rowid_dst, long n_arena, int npairs,
+    long lo, long m, int feature, int thr, int* scratch, int* total_left,
+    hipStream_t stream) {
+  if (m == 0) return;
+  long chunk = 4096;
+  long blocks = (m + chunk - 1) / chunk;
+  if (blocks > 4096) {
+    chunk = (m + 4095) / 4096;
+    blocks = (m + chunk - 1) / chunk;
+  }
+  const int pf = feature / 8, jbyte = feature % 8;
+  hipLaunchKernelGGL(part_arena_count_k, dim3((unsigned)blocks), dim3(256),
+                     0, stream, (const unsigned long long*)pair_src, n_arena,
+                     lo, m, pf, jbyte, thr, chunk, scratch);
+  hipLaunchKernelGGL(part_scan_k, dim3(1), dim3(256), 0, stream, scratch,
+                     (int)blocks, total_left);
+  hipLaunchKernelGGL(part_arena_scatter_k, dim3((unsigned)blocks), dim3(256),
+                     0, stream, (const unsigned long long*)pair_src,
+                     (const longlong2*)ghq_src, rowid_src,
+                     (unsigned long long*)pair_dst, (longlong2*)ghq_dst,
+                     rowid_dst, n_arena, npairs, lo, m, pf, jbyte, thr, chunk,
+                     scratch, total_left);
+}
+
+// range-mode histogram over arena segments: rows are CONTIGUOUS [lo+base,
+// lo+base+m_eff) of a buffer — no index gather at all.  side resolves the
+// child range from the device left count exactly like the rows-list kernel.
+__global__ void hist_pair_range_k(
+    const unsigned long long* __restrict__ pair, const longlong2* __restrict__ ghq,
+    long n_arena, long lo, long m, long long* __restrict__ hist, int n_bins,
+    int npairs, int tail_bytes, long chunk,
+    const int* __restrict__ nl_dev, int side) {
+  long base = 0, m_eff = m;
+  if (side >= 0) {
+    const long nl = nl_dev[0];
+    m_eff = (side == 0) ? nl : m - nl;
+    base = (side == 0) ? 0 : nl;
+  }
+  if ((long)blockIdx.x * chunk >= m_eff) return;
+  extern __shared__ unsigned long long lds64[];
+  const int tid = threadIdx.x;
+  const int lds_elems = 8 * n_bins * 2;
+  for (int i = tid; i < lds_elems; i += blockDim.x) lds64[i] = 0ull;
+  __syncthreads();
+
+  const int pair_id = blockIdx.y;
+  const int jmax = (pair_id == npairs - 1) ? tail_bytes : 8;
+  const unsigned long long* plane =
+      pair + (size_t)pair_id * n_arena + lo + base;
+  const longlong2* gh = ghq + lo + base;
+  const long start = (long)blockIdx.x * chunk;
+  const long end = min(start + chunk, m_eff);
+
+  constexpr int ILP = 4;
+  long i = start + tid;
+  for (; i + (ILP - 1) * (long)blockDim.x < end; i += ILP * blockDim.x) {
+    unsigned long long v[ILP];
+    longlong2 q[ILP];
+#pragma unroll
+    for (int u = 0; u < ILP; ++u) v[u] = plane[i + u * blockDim.x];
+#pragma unroll
+    for (int u = 0; u < ILP; ++u) q[u] = gh[i + u * blockDim.x];
+#pragma unroll
+    for (int u = 0; u < ILP; ++u) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        if (j >= jmax) break;
+        const int b = (int)((v[u] >> (8 * j)) & 0xffull);
+        unsigned long long* cell = &lds64[(j * n_bins + b) * 2];
+        atomicAdd(cell + 0, (unsigned long long)q[u].x);
+        atomicAdd(cell + 1, (unsigned long long)q[u].y);
+      }
+    }
+  }
+  for (; i < end; i += blockDim.x) {
+    const unsigned long long v = plane[i];
+    const longlong2 q = gh[i];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      if (j >= jmax) break;
+      const int b = (int)((v >> (8 * j)) & 0xffull);
+      unsigned long long* cell = &lds64[(j * n_bins + b) * 2];
+      atomicAdd(cell + 0, (unsigned long long)q.x);
+      atomicAdd(cell + 1, (unsigned long long)q.y);
+    }
+  }
+  __syncthreads();
+
+  for (int i2 = tid; i2 < jmax * n_bins; i2 += blockDim.x) {
+    const int f = i2 / n_bins;
+    const int b = i2 % n_bins;
+    const unsigned long long gsum = lds64[(f * n_bins + b) * 2 + 0];
+    const unsigned long long hpacked = lds64[(f * n_bins + b) * 2 + 1];
+    if (gsum == 0ull && hpacked == 0ull) continue;
+    const unsigned long long cnt = hpacked >> 44;
+    const unsigned long long hsum = hpacked & ((1ull << 44) - 1ull);
+    long long* out = hist + ((size_t)(pair_id * 8 + f) * n_bins + b) * 3;
+    atomicAdd((unsigned long long*)(out + 0), gsum);
+    atomicAdd((unsigned long long*)(out + 1), hsum);
+    atomicAdd((unsigned long long*)(out + 2), cnt);
+  }
+}
+
+extern "C" void launch_hist_pair_range(
+    const void* pair, const void* ghq, long n_arena, long lo, long m,
+    long long* hist, int n_bins, int npairs, int tail_bytes,
+    const int* nl_dev, int side, hipStream_t stream) {
+  if (m == 0) return;
+  long chunks = (2048 + npairs - 1) / npairs;
+  long chunk = (m + chunks - 1) / chunks;
+  if (chunk < 16384) {
+    chunk = (m + 7) / 8;
+    if (chunk < 2048) chunk = 2048;
+    if (chunk > 16384) chunk = 16384;
+  }
+  if (chunk > (1l << 19)) chunk = 1l << 19;
+  chunks = (m + chunk - 1) / chunk;
+  dim3 grid((unsigned)chunks, (unsigned)npairs);
+  const size_t lds_bytes = (size_t)8 * n_bins * 2 * sizeof(long long);
+  hipLaunchKernelGGL(hist_pair_range_k, grid, dim3(256), lds_bytes, stream,
+                     (const unsigned long long*)pair, (const longlong2*)ghq,
+                     n_arena, lo, m, hist, n_bins, npairs, tail_bytes, chunk,
+                     nl_dev, side);
+}
+
 extern "C" void launch_partition(const void* binned, long n_rows,
                                  const int* rows, long m, int feature,
                                  int thr, int* out, int* scratch,
