@@ -41,8 +41,8 @@ void launch_arena_gather(const void*, long, const float*, const float*,
                          const int*, long, int, double, double, void*, void*,
                          int*, hipStream_t);
 void launch_partition_arena(const void*, const void*, const int*, void*,
-                            void*, int*, long, int, long, long, int, int,
-                            int*, int*, hipStream_t);
+                            void*, int*, int*, long, int, long, long, int,
+                            int, int*, int*, hipStream_t);
 void launch_hist_pair_range(const void*, const void*, long, long, long,
                             long long*, int, int, int, const int*, int,
                             hipStream_t);
@@ -76,6 +76,7 @@ struct GrowCtx {
   bool distributed;
   torch::Tensor scratch;    // partition block counters (stream-ordered reuse)
   torch::Tensor total;      // partition left-count (stream-ordered reuse)
+  torch::Tensor dst_idx;    // (n_arena,) i32 per-split dest ranks (reused)
 };
 
 struct SplitJob {
@@ -165,7 +166,8 @@ void launch_job(GrowCtx& ctx, LeafCand& leaf) {
   launch_partition_arena(
       ctx.pair[src].data_ptr(), ctx.ghq[src].data_ptr(),
       ctx.rowid[src].data_ptr<int>(), ctx.pair[dst].data_ptr(),
-      ctx.ghq[dst].data_ptr(), ctx.rowid[dst].data_ptr<int>(), ctx.n_arena,
+      ctx.ghq[dst].data_ptr(), ctx.rowid[dst].data_ptr<int>(),
+      ctx.dst_idx.data_ptr<int>() + leaf.lo, ctx.n_arena,
       ctx.npairs, leaf.lo, m, leaf.feat, leaf.bin,
       ctx.scratch.data_ptr<int>(), ctx.total.data_ptr<int>(),
       grower_stream());
@@ -245,6 +247,10 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
 
   ctx.scratch = torch::empty({4096}, rows_root.options().dtype(torch::kInt32));
   ctx.total = torch::zeros({1}, rows_root.options().dtype(torch::kInt32));
+  // per-leaf segments are disjoint, so one dst_idx array serves concurrent
+  // in-flight splits (indexed at leaf.lo)
+  ctx.dst_idx = torch::empty({ctx.n_arena},
+                             rows_root.options().dtype(torch::kInt32));
   auto i64d = rows_root.options().dtype(torch::kInt64);
   for (int b = 0; b < 2; ++b) {
     ctx.pair[b] = torch::empty({ctx.npairs, ctx.n_arena}, i64d);

@@ -630,13 +630,14 @@ __global__ void part_arena_count_k(const unsigned long long* __restrict__ pair,
   if (threadIdx.x == 0) block_counts[blockIdx.x] = sh[0];
 }
 
-__global__ void part_arena_scatter_k(
+// pass 1: per-row destination index (stable left|right ranks) + move the
+// small payloads (row id + quantized gh); dst_idx drives the plane copies
+__global__ void part_arena_index_k(
     const unsigned long long* __restrict__ pair_src,
     const longlong2* __restrict__ ghq_src, const int* __restrict__ rowid_src,
-    unsigned long long* __restrict__ pair_dst,
     longlong2* __restrict__ ghq_dst, int* __restrict__ rowid_dst,
-    long n_arena, int npairs, long lo, long m, int pf, int jbyte, int thr,
-    long chunk, const int* __restrict__ block_offsets,
+    int* __restrict__ dst_idx, long n_arena, long lo, long m, int pf,
+    int jbyte, int thr, long chunk, const int* __restrict__ block_offsets,
     const int* __restrict__ total_left) {
   const long start = (long)blockIdx.x * chunk;
   const long end = min(start + chunk, m);
@@ -644,8 +645,8 @@ __global__ void part_arena_scatter_k(
   __shared__ long base_l, base_r;
   __shared__ int wave_l[4], wave_r[4];
   if (threadIdx.x == 0) {
-    base_l = lo + block_offsets[blockIdx.x];
-    base_r = lo + nl_total + (start - block_offsets[blockIdx.x]);
+    base_l = block_offsets[blockIdx.x];
+    base_r = nl_total + (start - block_offsets[blockIdx.x]);
   }
   __syncthreads();
   const int wid = threadIdx.x >> 6;
@@ -670,14 +671,11 @@ __global__ void part_arena_scatter_k(
       wr += wave_r[w];
     }
     if (valid) {
-      const long src = lo + i;
-      const long dst = left ? wl + __popcll(mask_l & lt)
-                            : wr + __popcll(mask_r & lt);
-      rowid_dst[dst] = rowid_src[src];
-      ghq_dst[dst] = ghq_src[src];
-      for (int p = 0; p < npairs; ++p)
-        pair_dst[(size_t)p * n_arena + dst] =
-            pair_src[(size_t)p * n_arena + src];
+      const long d = left ? wl + __popcll(mask_l & lt)
+                          : wr + __popcll(mask_r & lt);  // relative to lo
+      dst_idx[i] = (int)d;
+      rowid_dst[lo + d] = rowid_src[lo + i];
+      ghq_dst[lo + d] = ghq_src[lo + i];
     }
     __syncthreads();
     if (threadIdx.x == 0) {
@@ -693,11 +691,26 @@ __global__ void part_arena_scatter_k(
   }
 }
 
+// pass 2: plane copies driven by dst_idx — 2-D grid (row-chunk, plane) so
+// all npairs planes stream in parallel with coalesced reads
+__global__ void part_arena_copy_k(
+    const unsigned long long* __restrict__ pair_src,
+    unsigned long long* __restrict__ pair_dst,
+    const int* __restrict__ dst_idx, long n_arena, long lo, long m,
+    long chunk) {
+  const long start = (long)blockIdx.x * chunk;
+  const long end = min(start + chunk, m);
+  const unsigned long long* src = pair_src + (size_t)blockIdx.y * n_arena + lo;
+  unsigned long long* dst = pair_dst + (size_t)blockIdx.y * n_arena + lo;
+  for (long i = start + threadIdx.x; i < end; i += blockDim.x)
+    dst[dst_idx[i]] = src[i];
+}
+
 extern "C" void launch_partition_arena(
     const void* pair_src, const void* ghq_src, const int* rowid_src,
-    void* pair_dst, void* ghq_dst, int* rowid_dst, long n_arena, int npairs,
-    long lo, long m, int feature, int thr, int* scratch, int* total_left,
-    hipStream_t stream) {
+    void* pair_dst, void* ghq_dst, int* rowid_dst, int* dst_idx,
+    long n_arena, int npairs, long lo, long m, int feature, int thr,
+    int* scratch, int* total_left, hipStream_t stream) {
   if (m == 0) return;
   long chunk = 4096;
   long blocks = (m + chunk - 1) / chunk;
@@ -711,12 +724,22 @@ extern "C" void launch_partition_arena(
                      lo, m, pf, jbyte, thr, chunk, scratch);
   hipLaunchKernelGGL(part_scan_k, dim3(1), dim3(256), 0, stream, scratch,
                      (int)blocks, total_left);
-  hipLaunchKernelGGL(part_arena_scatter_k, dim3((unsigned)blocks), dim3(256),
+  hipLaunchKernelGGL(part_arena_index_k, dim3((unsigned)blocks), dim3(256),
                      0, stream, (const unsigned long long*)pair_src,
                      (const longlong2*)ghq_src, rowid_src,
-                     (unsigned long long*)pair_dst, (longlong2*)ghq_dst,
-                     rowid_dst, n_arena, npairs, lo, m, pf, jbyte, thr, chunk,
-                     scratch, total_left);
+                     (longlong2*)ghq_dst, rowid_dst, dst_idx, n_arena, lo, m,
+                     pf, jbyte, thr, chunk, scratch, total_left);
+  long cchunk = 2048;
+  long cblocks = (m + cchunk - 1) / cchunk;
+  if (cblocks > 4096) {
+    cchunk = (m + 4095) / 4096;
+    cblocks = (m + cchunk - 1) / cchunk;
+  }
+  hipLaunchKernelGGL(part_arena_copy_k, dim3((unsigned)cblocks,
+                                             (unsigned)npairs), dim3(256), 0,
+                     stream, (const unsigned long long*)pair_src,
+                     (unsigned long long*)pair_dst, dst_idx, n_arena, lo, m,
+                     cchunk);
 }
 
 // range-mode histogram over arena segments: rows are CONTIGUOUS [lo+base,
