@@ -829,11 +829,8 @@ extern "C" void launch_hist_pair_range(
   long chunks = (2048 + npairs - 1) / npairs;
   long chunk = (m + chunks - 1) / chunks;
   if (chunk < 16384) {
-    // contiguous reads: small leaves can shatter finer than the gather
-    // kernels (32 row-chunks x npairs blocks; 32-way flush contention on
-    // 2040 addresses is noise) to fill more of the 256 CUs
-    chunk = (m + 31) / 32;
-    if (chunk < 512) chunk = 512;
+    chunk = (m + 7) / 8;
+    if (chunk < 2048) chunk = 2048;
     if (chunk > 16384) chunk = 16384;
   }
   if (chunk > (1l << 19)) chunk = 1l << 19;
