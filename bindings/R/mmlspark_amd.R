@@ -1875,7 +1875,7 @@ ml_verify_faces <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyC
   stage
 }
 
-ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, device = NULL, weightsArrays = NULL, rawPredictionCol = NULL, probabilityCol = NULL) {
+ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, maxIterBfgs = NULL, device = NULL, weightsArrays = NULL, rawPredictionCol = NULL, probabilityCol = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitClassificationModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1901,6 +1901,8 @@ ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol =
   if (!is.null(interactions)) stage$set("interactions", interactions)
   if (!is.null(ignoreNamespaces)) stage$set("ignoreNamespaces", ignoreNamespaces)
   if (!is.null(useBarrierExecutionMode)) stage$set("useBarrierExecutionMode", useBarrierExecutionMode)
+  if (!is.null(bfgs)) stage$set("bfgs", bfgs)
+  if (!is.null(maxIterBfgs)) stage$set("maxIterBfgs", maxIterBfgs)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(weightsArrays)) stage$set("weightsArrays", weightsArrays)
   if (!is.null(rawPredictionCol)) stage$set("rawPredictionCol", rawPredictionCol)
@@ -1908,7 +1910,7 @@ ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol =
   stage
 }
 
-ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL, labelConversion = NULL) {
+ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, maxIterBfgs = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL, labelConversion = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitClassifier()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1934,6 +1936,8 @@ ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, add
   if (!is.null(interactions)) stage$set("interactions", interactions)
   if (!is.null(ignoreNamespaces)) stage$set("ignoreNamespaces", ignoreNamespaces)
   if (!is.null(useBarrierExecutionMode)) stage$set("useBarrierExecutionMode", useBarrierExecutionMode)
+  if (!is.null(bfgs)) stage$set("bfgs", bfgs)
+  if (!is.null(maxIterBfgs)) stage$set("maxIterBfgs", maxIterBfgs)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(rawPredictionCol)) stage$set("rawPredictionCol", rawPredictionCol)
   if (!is.null(probabilityCol)) stage$set("probabilityCol", probabilityCol)
@@ -1941,7 +1945,7 @@ ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, add
   stage
 }
 
-ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, device = NULL, sharedCol = NULL, chosenActionCol = NULL, probabilityCol = NULL, epsilon = NULL) {
+ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, maxIterBfgs = NULL, device = NULL, sharedCol = NULL, chosenActionCol = NULL, probabilityCol = NULL, epsilon = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitContextualBandit()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1967,6 +1971,8 @@ ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NU
   if (!is.null(interactions)) stage$set("interactions", interactions)
   if (!is.null(ignoreNamespaces)) stage$set("ignoreNamespaces", ignoreNamespaces)
   if (!is.null(useBarrierExecutionMode)) stage$set("useBarrierExecutionMode", useBarrierExecutionMode)
+  if (!is.null(bfgs)) stage$set("bfgs", bfgs)
+  if (!is.null(maxIterBfgs)) stage$set("maxIterBfgs", maxIterBfgs)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(sharedCol)) stage$set("sharedCol", sharedCol)
   if (!is.null(chosenActionCol)) stage$set("chosenActionCol", chosenActionCol)
@@ -1975,7 +1981,7 @@ ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NU
   stage
 }
 
-ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, device = NULL, weightsArrays = NULL, sharedCol = NULL, epsilon = NULL) {
+ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, maxIterBfgs = NULL, device = NULL, weightsArrays = NULL, sharedCol = NULL, epsilon = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitContextualBanditModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -2001,6 +2007,8 @@ ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCo
   if (!is.null(interactions)) stage$set("interactions", interactions)
   if (!is.null(ignoreNamespaces)) stage$set("ignoreNamespaces", ignoreNamespaces)
   if (!is.null(useBarrierExecutionMode)) stage$set("useBarrierExecutionMode", useBarrierExecutionMode)
+  if (!is.null(bfgs)) stage$set("bfgs", bfgs)
+  if (!is.null(maxIterBfgs)) stage$set("maxIterBfgs", maxIterBfgs)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(weightsArrays)) stage$set("weightsArrays", weightsArrays)
   if (!is.null(sharedCol)) stage$set("sharedCol", sharedCol)
@@ -2030,7 +2038,7 @@ ml_vowpal_wabbit_interactions <- function(inputCols = NULL, outputCol = NULL, nu
   stage
 }
 
-ml_vowpal_wabbit_regressor <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, device = NULL) {
+ml_vowpal_wabbit_regressor <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, maxIterBfgs = NULL, device = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitRegressor()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -2056,11 +2064,13 @@ ml_vowpal_wabbit_regressor <- function(labelCol = NULL, featuresCol = NULL, addi
   if (!is.null(interactions)) stage$set("interactions", interactions)
   if (!is.null(ignoreNamespaces)) stage$set("ignoreNamespaces", ignoreNamespaces)
   if (!is.null(useBarrierExecutionMode)) stage$set("useBarrierExecutionMode", useBarrierExecutionMode)
+  if (!is.null(bfgs)) stage$set("bfgs", bfgs)
+  if (!is.null(maxIterBfgs)) stage$set("maxIterBfgs", maxIterBfgs)
   if (!is.null(device)) stage$set("device", device)
   stage
 }
 
-ml_vowpal_wabbit_regressor_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, device = NULL, weightsArrays = NULL) {
+ml_vowpal_wabbit_regressor_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, maxIterBfgs = NULL, device = NULL, weightsArrays = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitRegressorModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -2086,6 +2096,8 @@ ml_vowpal_wabbit_regressor_model <- function(labelCol = NULL, featuresCol = NULL
   if (!is.null(interactions)) stage$set("interactions", interactions)
   if (!is.null(ignoreNamespaces)) stage$set("ignoreNamespaces", ignoreNamespaces)
   if (!is.null(useBarrierExecutionMode)) stage$set("useBarrierExecutionMode", useBarrierExecutionMode)
+  if (!is.null(bfgs)) stage$set("bfgs", bfgs)
+  if (!is.null(maxIterBfgs)) stage$set("maxIterBfgs", maxIterBfgs)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(weightsArrays)) stage$set("weightsArrays", weightsArrays)
   stage

@@ -116,6 +116,9 @@ class _VWParams(Params):
                                     "gang-schedule ranks (no-op: ranks are "
                                     "always gang-launched here)", False,
                                     toBool)
+    bfgs = Param("bfgs", "second-order full-batch L-BFGS optimization "
+                 "(--bfgs) instead of online SGD", False, toBool)
+    maxIterBfgs = Param("maxIterBfgs", "L-BFGS iteration budget", 100, toInt)
     device = Param("device", "cpu|cuda|auto", "auto", toString)
 
     def _feature_frame(self, df: pd.DataFrame):
@@ -148,7 +151,7 @@ class _VWParams(Params):
                    "numBits", "--passes": "numPasses",
                    "--loss_function": "lossFunction"}
         flags = {"--adaptive": "adaptive", "--normalized": "normalized",
-                 "--invariant": "invariant"}
+                 "--invariant": "invariant", "--bfgs": "bfgs"}
         while i < len(s):
             if s[i] in mapping and i + 1 < len(s):
                 self.set(mapping[s[i]], s[i + 1])
@@ -163,6 +166,46 @@ class _VWParams(Params):
 class _VWBase(_VWParams, Estimator):
     _default_loss = "squared"
     _binary_labels = False
+
+    def _fit_bfgs(self, idx, val, off, labels, ex_w, w0, loss, l2, comm):
+        n_ex = off.numel() - 1
+        counts = off[1:] - off[:-1]
+        seg = torch.repeat_interleave(
+            torch.arange(n_ex, device=idx.device), counts)
+        il = idx.long()
+        ew = ex_w if ex_w is not None else torch.ones_like(labels)
+        w = w0.clone().requires_grad_(True)
+
+        def forward():
+            contrib = w[il] * val
+            preds = torch.zeros(n_ex, device=w.device).index_add(
+                0, seg, contrib)
+            if loss == "logistic":
+                per = torch.nn.functional.softplus(-labels * preds)
+            elif loss == "hinge":
+                per = torch.relu(1.0 - labels * preds)
+            else:
+                per = 0.5 * (preds - labels) ** 2
+            out = (per * ew).sum() / ew.sum()
+            if l2 > 0:
+                out = out + l2 * (w * w).sum()
+            return out
+
+        opt = torch.optim.LBFGS([w], max_iter=self.get("maxIterBfgs"),
+                                history_size=10,
+                                line_search_fn="strong_wolfe")
+
+        def closure():
+            opt.zero_grad()
+            ls = forward()
+            ls.backward()
+            if comm.is_distributed:  # global batch gradient
+                comm.all_reduce(w.grad)
+                w.grad /= comm.world_size
+            return ls
+
+        opt.step(closure)
+        return w.detach()
 
     def _fit(self, df: pd.DataFrame):
         self._parse_args()
@@ -204,7 +247,17 @@ class _VWBase(_VWParams, Estimator):
         n = len(df)
         learn_s = 0.0
         multipass_s = 0.0
-        for p in range(self.get("numPasses")):
+        if self.get("bfgs"):
+            # --bfgs: full-batch second-order optimization over the hashed
+            # weight table (VowpalWabbitBase.scala arg surface).  The
+            # forward is a differentiable gather + segment-sum, so torch's
+            # L-BFGS (with line search) runs it entirely on device.
+            t0 = time.perf_counter()
+            w = self._fit_bfgs(idx, val, off, labels, ex_w, w, loss, l2,
+                               comm)
+            learn_s = time.perf_counter() - t0
+        for p in (range(0) if self.get("bfgs")
+                  else range(self.get("numPasses"))):
             t0 = time.perf_counter()
             for s in range(0, n, bs):
                 e = min(s + bs, n)
