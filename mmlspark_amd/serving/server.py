@@ -249,6 +249,9 @@ class LowLatencyGBDTScorer:
                        else torch.device("cpu"))
         self.nf = booster.n_features
         self.max_batch = max_batch
+        # continuous-mode HTTP threads share this scorer; the preallocated
+        # input buffer + captured graph are single-flight
+        self._lock = threading.Lock()
         self.inp = torch.zeros(max_batch, self.nf, device=self.device)
         self.flat = booster._flat(self.device)
         self.graph = None
@@ -276,6 +279,10 @@ class LowLatencyGBDTScorer:
             self.out = self._raw()
 
     def score(self, X: np.ndarray) -> np.ndarray:
+        with self._lock:
+            return self._score_locked(X)
+
+    def _score_locked(self, X: np.ndarray) -> np.ndarray:
         torch = self.torch
         n = X.shape[0]
         assert n <= self.max_batch
