@@ -147,41 +147,53 @@ class IsolationForestModel(Model):
                 "sub_n": np.asarray([sub_n], np.int64),
             })
 
+    def _leaf_depth_values(self, f):
+        """Per-node value = depth + avg_path(leaf size) for leaves (0 for
+        internal nodes): the whole path-length estimate becomes ONE forest
+        traversal through the batched HIP predict kernel."""
+        feat, lft, rgt = f["feature"], f["left"], f["right"]
+        size, offs = f["size"], f["offsets"]
+        val = np.zeros(len(feat), dtype=np.float32)
+        for t in range(len(offs) - 1):
+            base, end = int(offs[t]), int(offs[t + 1])
+            depth = np.zeros(end - base, dtype=np.float32)
+            for i in range(base, end):  # parents precede children
+                if feat[i] >= 0:
+                    depth[lft[i] ] = depth[i - base] + 1
+                    depth[rgt[i] ] = depth[i - base] + 1
+            for i in range(base, end):
+                if feat[i] < 0:
+                    val[i] = depth[i - base] + _avg_path(
+                        max(float(size[i]), 1.0))
+        return val
+
     def _scores(self, X: np.ndarray) -> np.ndarray:
+        from ..ops import backend
         f = self.get("forestArrays")
+        if "leaf_value" not in f:
+            f = dict(f)
+            f["leaf_value"] = self._leaf_depth_values(f)
+            self.set("forestArrays", f)
         device = default_device("auto")
-        feat = torch.from_numpy(f["feature"]).to(device).long()
-        thr = torch.from_numpy(f["threshold"]).to(device)
-        lft = torch.from_numpy(f["left"]).to(device).long()
-        rgt = torch.from_numpy(f["right"]).to(device).long()
-        size = torch.from_numpy(f["size"]).to(device)
-        offs = f["offsets"]
         sub_n = int(f["sub_n"][0])
+        n_trees = len(f["offsets"]) - 1
         Xt = torch.from_numpy(np.ascontiguousarray(X)).to(device)
-        n = Xt.shape[0]
-        n_trees = len(offs) - 1
-        depth_sum = torch.zeros(n, device=device)
-        ar = torch.arange(n, device=device)
-        for t in range(n_trees):
-            base = int(offs[t])
-            idx = torch.full((n,), base, dtype=torch.long, device=device)
-            depth = torch.zeros(n, device=device)
-            active = feat[idx] >= 0
-            while bool(active.any()):
-                ff = feat[idx].clamp(min=0)
-                xv = Xt[ar, ff]
-                nxt = torch.where(xv < thr[idx], lft[idx], rgt[idx]) + base
-                idx = torch.where(active, nxt, idx)
-                depth = depth + active.float()
-                active = feat[idx] >= 0
-            # adjustment for unsplit leaves with >1 point
-            adj = torch.tensor([_avg_path(max(float(s), 1.0))
-                                for s in size[idx].cpu().tolist()],
-                               device=device)
-            depth_sum += depth + adj
-        avg_depth = depth_sum / n_trees
+        dev = lambda a, dt: torch.from_numpy(  # noqa: E731
+            np.ascontiguousarray(a)).to(device).to(dt)
+        # shift thresholds so the kernel's `x <= thr` matches iforest's
+        # `x < thr` (thresholds are continuous uniforms: exact ties are
+        # measure-zero; nextafter keeps even those consistent)
+        thr = np.nextafter(f["threshold"], -np.inf).astype(np.float32)
+        depth_sum = backend.predict_forest(
+            dev(f["feature"], torch.int32), dev(thr, torch.float32),
+            dev(f["left"], torch.int32), dev(f["right"], torch.int32),
+            dev(f["leaf_value"], torch.float32),
+            torch.from_numpy(f["offsets"]).to(device), Xt.float(), 1,
+            torch.ones(n_trees, device=device))
+        avg_depth = depth_sum.reshape(-1) / n_trees
         c = _avg_path(sub_n)
-        return (2.0 ** (-avg_depth.cpu().numpy() / max(c, 1e-9))).astype(np.float64)
+        return (2.0 ** (-avg_depth.cpu().numpy()
+                        / max(c, 1e-9))).astype(np.float64)
 
     def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
         X = features_matrix(df, self.get("featuresCol"), self.get("featureCols"))
