@@ -1,0 +1,2 @@
+from .server import (DistributedServingServer, LowLatencyGBDTScorer,  # noqa: F401
+                     ServingServer)
