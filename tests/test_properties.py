@@ -1,0 +1,80 @@
+"""Property-based invariants (hypothesis) for foundational pieces: hashing,
+sparse-vector collision handling, quantile binning, partition stability,
+and the invariant-update safety bound."""
+import numpy as np
+import torch
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from mmlspark_amd.models.vw.murmur import hash_string
+
+
+@given(st.text(min_size=0, max_size=64), st.integers(0, 2**31 - 1))
+@settings(max_examples=200, deadline=None)
+def test_murmur_deterministic_and_seeded(s, seed):
+    a = hash_string(s, seed)
+    assert a == hash_string(s, seed)          # deterministic
+    assert 0 <= a < 2**32                     # 32-bit range
+
+
+@given(st.lists(st.tuples(st.integers(0, 255), st.floats(-10, 10)),
+                min_size=0, max_size=64))
+@settings(max_examples=100, deadline=None)
+def test_sum_collisions_preserves_mass(pairs):
+    from mmlspark_amd.models.vw.featurizer import _sum_collisions
+    idx = np.array([p[0] for p in pairs], dtype=np.int64)
+    val = np.array([p[1] for p in pairs], dtype=np.float32)
+    ia, va = _sum_collisions(idx, val)
+    assert len(ia) == len(np.unique(idx))     # one entry per distinct index
+    assert np.all(np.diff(ia) > 0)            # sorted strictly
+    np.testing.assert_allclose(va.sum(), val.sum(), rtol=1e-4, atol=1e-4)
+
+
+@given(st.integers(2, 64), st.integers(100, 2000))
+@settings(max_examples=20, deadline=None)
+def test_binning_is_monotone_and_bounded(n_bins, n):
+    from mmlspark_amd.models.gbdt.binning import BinMapper
+    g = torch.Generator().manual_seed(n)
+    X = torch.randn(n, 3, generator=g)
+    bm = BinMapper.fit(X, n_bins=n_bins, sample_size=n)
+    binned = bm.transform(X)
+    flat = binned.permute(0, 2, 1).reshape(-1, n)[:3]
+    assert int(flat.max()) < n_bins
+    # monotone: larger raw value never gets a smaller bin
+    for f in range(3):
+        order = torch.argsort(X[:, f])
+        b = flat[f][order].to(torch.int64)
+        assert bool((b[1:] >= b[:-1]).all())
+
+
+@given(st.integers(0, 254), st.integers(50, 400))
+@settings(max_examples=25, deadline=None)
+def test_partition_is_stable_and_exact(thr, n):
+    from mmlspark_amd.ops import cpu_ref
+    g = torch.Generator().manual_seed(thr * 1000 + n)
+    binned = torch.randint(0, 255, (1, n, 4), generator=g, dtype=torch.uint8)
+    rows = torch.randperm(n, generator=g)[: n // 2].to(torch.int32)
+    left, right = cpu_ref.partition_rows(binned, rows, 2, thr)
+    bins = binned[0, :, 2]
+    assert all(int(bins[r]) <= thr for r in left.tolist())
+    assert all(int(bins[r]) > thr for r in right.tolist())
+    # stability: relative order within each side preserved
+    order = {int(r): i for i, r in enumerate(rows.tolist())}
+    for side in (left.tolist(), right.tolist()):
+        pos = [order[r] for r in side]
+        assert pos == sorted(pos)
+    assert len(left) + len(right) == len(rows)
+
+
+@given(st.floats(0.1, 1e6), st.floats(0.01, 5.0))
+@settings(max_examples=100, deadline=None)
+def test_invariant_update_never_overshoots(h, lr):
+    """For ANY importance weight and rate, the squared-loss invariant update
+    moves the prediction toward the label and never past it."""
+    from mmlspark_amd.models.vw.sgd_ref import _invariant_dp
+    pred = torch.tensor([0.0])
+    y = torch.tensor([3.0])
+    x_norm = 1.7
+    dp = _invariant_dp("squared", pred, y, torch.tensor([h * lr * x_norm]))
+    new = float(pred + dp)
+    assert 0.0 <= new <= 3.0 + 1e-5
