@@ -43,7 +43,7 @@ def write_binary_files(df: pd.DataFrame, out_dir: str, path_col: str = "path",
 
 # ------------------------------------------------------------- image codecs
 def decode_image(data: bytes) -> np.ndarray:
-    """bytes → HWC uint8 array. BMP/PPM/PGM/NPY supported offline."""
+    """bytes → HWC uint8 array. BMP/PPM/PGM/NPY/PNG supported offline."""
     if data[:2] == b"BM":
         return _decode_bmp(data)
     if data[:2] in (b"P6", b"P5", b"P3", b"P2"):
@@ -55,8 +55,8 @@ def decode_image(data: bytes) -> np.ndarray:
         raise ValueError("JPEG decoding requires a codec not present in this "
                          "offline image; re-encode as BMP/PPM/NPY")
     if data[:8] == b"\x89PNG\r\n\x1a\n":
-        raise ValueError("PNG decoding requires a codec not present in this "
-                         "offline image; re-encode as BMP/PPM/NPY")
+        from .png_codec import decode_png
+        return decode_png(data)
     raise ValueError("unrecognized image format")
 
 
@@ -74,6 +74,9 @@ def encode_image(img: np.ndarray, fmt: str = "ppm") -> bytes:
         buf = io.BytesIO()
         np.save(buf, img)
         return buf.getvalue()
+    if fmt == "png":
+        from .png_codec import encode_png
+        return encode_png(img)
     raise ValueError(f"unsupported encode format {fmt}")
 
 

@@ -87,3 +87,71 @@ def test_codegen_outputs(tmp_path):
     assert nr == n
     rcode = open(tmp_path / "r" / "bindings.R").read()
     assert "ml_light_gbm_classifier" in rcode
+
+
+def test_png_codec_roundtrip_and_filters():
+    """Pure-numpy PNG codec (ImageUtils analog): encode→decode round trip
+    for gray/RGB/RGBA, all five scanline filters, and palette images."""
+    import struct
+    import zlib
+    from mmlspark_amd.io_http.files import decode_image, encode_image
+    from mmlspark_amd.io_http.png_codec import decode_png
+
+    rng = np.random.default_rng(0)
+    for shape in [(17, 23), (16, 16, 3), (9, 5, 4)]:
+        img = rng.integers(0, 256, size=shape).astype(np.uint8)
+        back = decode_image(encode_image(img, "png"))
+        assert back.shape == img.shape and (back == img).all()
+
+    # decode with every filter type (hand-encoded rows 0..4)
+    w, h = 8, 5
+    img = rng.integers(0, 256, size=(h, w, 3)).astype(np.uint8)
+    raws = bytearray()
+    prev = [0] * (w * 3)
+    for y, f in enumerate([0, 1, 2, 3, 4]):
+        line = [int(v) for v in img[y].reshape(-1)]
+        enc = []
+        for x in range(w * 3):
+            a = line[x - 3] if x >= 3 else 0
+            b = prev[x]
+            c = prev[x - 3] if x >= 3 else 0
+            if f == 0:
+                enc.append(line[x])
+            elif f == 1:
+                enc.append((line[x] - a) & 255)
+            elif f == 2:
+                enc.append((line[x] - b) & 255)
+            elif f == 3:
+                enc.append((line[x] - (a + b) // 2) & 255)
+            else:
+                p = a + b - c
+                pa, pb, pc = abs(p - a), abs(p - b), abs(p - c)
+                pr = a if pa <= pb and pa <= pc else (b if pb <= pc else c)
+                enc.append((line[x] - pr) & 255)
+        raws.append(f)
+        raws.extend(bytes(enc))
+        prev = line
+
+    def chunk(t, p):
+        return (struct.pack(">I", len(p)) + t + p
+                + struct.pack(">I", zlib.crc32(t + p) & 0xFFFFFFFF))
+
+    png = (b"\x89PNG\r\n\x1a\n"
+           + chunk(b"IHDR", struct.pack(">IIBBBBB", w, h, 8, 2, 0, 0, 0))
+           + chunk(b"IDAT", zlib.compress(bytes(raws)))
+           + chunk(b"IEND", b""))
+    assert (decode_png(png) == img).all()
+
+    # palette (color type 3)
+    pal = rng.integers(0, 256, size=(4, 3)).astype(np.uint8)
+    idx = rng.integers(0, 4, size=(6, 7)).astype(np.uint8)
+    raws = bytearray()
+    for y in range(6):
+        raws.append(0)
+        raws.extend(idx[y].tobytes())
+    png = (b"\x89PNG\r\n\x1a\n"
+           + chunk(b"IHDR", struct.pack(">IIBBBBB", 7, 6, 8, 3, 0, 0, 0))
+           + chunk(b"PLTE", pal.tobytes())
+           + chunk(b"IDAT", zlib.compress(bytes(raws)))
+           + chunk(b"IEND", b""))
+    assert (decode_png(png) == pal[idx]).all()
