@@ -43,7 +43,7 @@ def write_binary_files(df: pd.DataFrame, out_dir: str, path_col: str = "path",
 
 # ------------------------------------------------------------- image codecs
 def decode_image(data: bytes) -> np.ndarray:
-    """bytes → HWC uint8 array. BMP/PPM/PGM/NPY/PNG supported offline."""
+    """bytes → HWC uint8 array. BMP/PPM/PGM/NPY/PNG/JPEG supported offline."""
     if data[:2] == b"BM":
         return _decode_bmp(data)
     if data[:2] in (b"P6", b"P5", b"P3", b"P2"):
@@ -52,8 +52,8 @@ def decode_image(data: bytes) -> np.ndarray:
         import io
         return np.load(io.BytesIO(data))
     if data[:2] == b"\xff\xd8":
-        raise ValueError("JPEG decoding requires a codec not present in this "
-                         "offline image; re-encode as BMP/PPM/NPY")
+        from .jpeg_codec import decode_jpeg
+        return decode_jpeg(data)
     if data[:8] == b"\x89PNG\r\n\x1a\n":
         from .png_codec import decode_png
         return decode_png(data)
@@ -77,6 +77,9 @@ def encode_image(img: np.ndarray, fmt: str = "ppm") -> bytes:
     if fmt == "png":
         from .png_codec import encode_png
         return encode_png(img)
+    if fmt in ("jpg", "jpeg"):
+        from .jpeg_codec import encode_jpeg
+        return encode_jpeg(img)
     raise ValueError(f"unsupported encode format {fmt}")
 
 

@@ -155,3 +155,24 @@ def test_png_codec_roundtrip_and_filters():
            + chunk(b"IDAT", zlib.compress(bytes(raws)))
            + chunk(b"IEND", b""))
     assert (decode_png(png) == pal[idx]).all()
+
+
+def test_jpeg_codec_roundtrip():
+    """Baseline JPEG codec: encode→decode stays within quantization error
+    (high PSNR on smooth content, exact on flat blocks); gray + RGB."""
+    from mmlspark_amd.io_http.files import decode_image, encode_image
+    yy, xx = np.mgrid[0:40, 0:56]
+    img = np.dstack([(yy * 2) % 256, (xx * 3) % 256,
+                     ((yy + xx) * 2) % 256]).astype(np.uint8)
+    dec = decode_image(encode_image(img, "jpg"))
+    assert dec.shape == img.shape
+    err = (dec.astype(int) - img.astype(int)).astype(float)
+    psnr = 10 * np.log10(255 ** 2 / max((err ** 2).mean(), 1e-9))
+    assert psnr > 40, psnr
+    flat = np.full((16, 24, 3), 201, np.uint8)
+    dflat = decode_image(encode_image(flat, "jpeg"))
+    assert np.abs(dflat.astype(int) - 201).max() <= 1
+    gray = ((yy * 5) % 256).astype(np.uint8)
+    dgray = decode_image(encode_image(gray, "jpg"))
+    assert dgray.shape == gray.shape
+    assert np.abs(dgray.astype(int) - gray.astype(int)).max() <= 4
