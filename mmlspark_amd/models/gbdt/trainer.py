@@ -25,6 +25,7 @@ from __future__ import annotations
 
 import heapq
 import math
+import os
 import time
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional
@@ -166,7 +167,7 @@ class TreeGrower:
         self.cat_features = list(getattr(bin_mapper, "categorical", []) or [])
         self.cat_smooth = 10.0
 
-    _sync_timers = bool(__import__("os").environ.get("MMLSPARK_AMD_SYNC_TIMERS"))
+    _sync_timers = bool(os.environ.get("MMLSPARK_AMD_SYNC_TIMERS"))
 
     def set_scales(self, grad: torch.Tensor, hess: torch.Tensor):
         """Per-tree fixed-point scales from GLOBAL max|g|, max h (identical on
@@ -343,7 +344,7 @@ class TreeGrower:
             float(cfg.min_data_in_leaf), cfg.min_sum_hessian_in_leaf,
             cfg.min_gain_to_split, cfg.max_delta_step, cfg.num_leaves,
             cfg.max_depth, feat_mask, reduce_fn,
-            self.comm.is_distributed or bool(__import__("os").environ.get(
+            self.comm.is_distributed or bool(os.environ.get(
                 "MMLSPARK_AMD_FORCE_DIST_GROWER")))
         self.stats.hist_s += time.perf_counter() - t0
         feature = d["feature"].numpy()
@@ -367,7 +368,7 @@ class TreeGrower:
     def grow(self, rows_root: torch.Tensor, grad: torch.Tensor,
              hess: torch.Tensor, feat_mask) -> (Tree, List):
         if (self.fixed and not self.cat_features and not self.voting
-                and not __import__("os").environ.get(
+                and not os.environ.get(
                     "MMLSPARK_AMD_NO_NATIVE_GROWER")):
             try:
                 from ...ops import _hip_grower  # noqa: F401
