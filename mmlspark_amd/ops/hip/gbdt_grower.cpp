@@ -325,6 +325,12 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
     if (n_leaves + 1 < ctx.num_leaves && !heap.empty()) {
       CandPtr nxt = heap.top();
       if (!nxt->job && splittable(ctx, *nxt)) launch_job(ctx, *nxt);
+      if (n_leaves + 2 < ctx.num_leaves && heap.size() >= 2) {
+        heap.pop();  // peek the second-best, then restore
+        CandPtr nxt2 = heap.top();
+        heap.push(nxt);
+        if (!nxt2->job && splittable(ctx, *nxt2)) launch_job(ctx, *nxt2);
+      }
     }
 
     SplitJob& job = *leaf->job;
