@@ -43,11 +43,17 @@ def write_binary_files(df: pd.DataFrame, out_dir: str, path_col: str = "path",
 
 # ------------------------------------------------------------- image codecs
 def decode_image(data: bytes) -> np.ndarray:
-    """bytes → HWC uint8 array. BMP/PPM/PGM/NPY/PNG/JPEG supported offline."""
-    if data[:2] == b"BM":
-        return _decode_bmp(data)
-    if data[:2] in (b"P6", b"P5", b"P3", b"P2"):
-        return _decode_pnm(data)
+    """bytes → HWC uint8 array. BMP/PPM/PGM/NPY/PNG/JPEG supported offline.
+    Malformed input raises ValueError (never a struct/parse internal error)."""
+    try:
+        if data[:2] == b"BM":
+            return _decode_bmp(data)
+        if data[:2] in (b"P6", b"P5", b"P3", b"P2"):
+            return _decode_pnm(data)
+    except ValueError:
+        raise
+    except Exception as e:  # truncated/garbled container
+        raise ValueError(f"malformed image data: {e}") from e
     if data[:6] == b"\x93NUMPY":
         import io
         return np.load(io.BytesIO(data))
