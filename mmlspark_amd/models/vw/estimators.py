@@ -199,9 +199,15 @@ class _VWBase(_VWParams, Estimator):
             opt.zero_grad()
             ls = forward()
             ls.backward()
-            if comm.is_distributed:  # global batch gradient
+            if comm.is_distributed:
+                # ranks must see the SAME loss value and gradient, or the
+                # strong-Wolfe line searches diverge across ranks
                 comm.all_reduce(w.grad)
                 w.grad /= comm.world_size
+                with torch.no_grad():
+                    lsg = ls.detach().clone()
+                    comm.all_reduce(lsg)
+                    return lsg / comm.world_size
             return ls
 
         opt.step(closure)

@@ -236,3 +236,56 @@ def test_voting_parallel_identical_and_accurate():
     b = Booster.load_from_string(results[0])
     p = torch.sigmoid(b.predict_raw(torch.from_numpy(X)).squeeze(-1)).numpy()
     assert roc_auc_score(y, p) > 0.9
+
+
+def _free_port():
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def _bfgs_worker(rank, world, port, out):
+    import os
+    import numpy as np
+    import pandas as pd
+    import torch.distributed as dist
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from mmlspark_amd.models.vw.estimators import VowpalWabbitClassifier
+        rng = np.random.default_rng(100 + rank)  # DIFFERENT shard per rank
+        X = rng.normal(size=(1500, 6)).astype(np.float32)
+        w = np.array([2, -1, 1, 0, 0, 0], dtype=np.float32)
+        y = (X @ w > 0).astype(np.float32)
+        df = pd.DataFrame({"features": list(X), "label": y})
+        m = VowpalWabbitClassifier(bfgs=True, lossFunction="logistic",
+                                   maxIterBfgs=40).fit(df)
+        out[rank] = (m.weights.copy(),
+                     (m.transform(df)["prediction"].to_numpy() == y).mean())
+    finally:
+        dist.destroy_process_group()
+
+
+def test_distributed_bfgs_identical_weights():
+    """--bfgs multi-rank: averaged loss + gradient keep every rank's L-BFGS
+    trajectory in lockstep — final weight tables identical, model good."""
+    import multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        out = mgr.dict()
+        port = _free_port()
+        ps = [ctx.Process(target=_bfgs_worker, args=(r, 2, port, out))
+              for r in range(2)]
+        for p in ps:
+            p.start()
+        for p in ps:
+            p.join(240)
+        assert all(p.exitcode == 0 for p in ps), [p.exitcode for p in ps]
+        w0, acc0 = out[0]
+        w1, acc1 = out[1]
+    np.testing.assert_allclose(w0, w1, rtol=0, atol=0)  # bit-identical
+    assert acc0 > 0.9 and acc1 > 0.9
