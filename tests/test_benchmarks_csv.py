@@ -62,3 +62,31 @@ def test_vw_benchmark_csv():
                higher_is_better=False)
     problems = runner.compare()
     assert not problems, problems
+
+
+def test_train_classifier_benchmark_csv():
+    """TrainClassifier AUROC/AUPR bars (benchmarks_VerifyTrainClassifier.csv
+    parity: PimaIndian-shaped 8-feature binary data, auto-featurized)."""
+    from mmlspark_amd.stages.train import (ComputeModelStatistics,
+                                           TrainClassifier)
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+
+    rng = np.random.default_rng(7)
+    n = 768  # PimaIndian size
+    X = rng.normal(size=(n, 8)).astype(np.float32)
+    w = rng.normal(size=8)
+    y = ((X @ w + rng.normal(size=n) * 1.5) > 0).astype(np.float32)
+    df = pd.DataFrame({f"f{i}": X[:, i] for i in range(8)})
+    df["label"] = y
+    model = TrainClassifier(model=LightGBMClassifier(numIterations=20,
+                                                     numLeaves=7, seed=0),
+                            labelCol="label").fit(df)
+    stats = ComputeModelStatistics(labelCol="label").transform(
+        model.transform(df))
+    runner = BenchmarkRunner("VerifyTrainClassifier", RESOURCE_DIR)
+    runner.add("auroc_pima_shaped", float(stats.iloc[0]["AUC"]),
+               precision=0.05)
+    runner.add("aupr_pima_shaped", float(stats.iloc[0].get(
+        "AUPR", stats.iloc[0].get("precision", 0.0))), precision=0.08)
+    problems = runner.compare()
+    assert not problems, problems
