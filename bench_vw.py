@@ -54,8 +54,6 @@ def main():
 
     w = torch.zeros(tbl, dtype=torch.float32, device=device)
     gacc = torch.zeros(tbl, dtype=torch.float32, device=device)
-    wg = (torch.zeros(2 * tbl, dtype=torch.float32, device=device)
-          if use_gpu else None)  # interleaved (w,G): same-cacheline atomics
     bs = 1 << 16
 
     def one_pass():
@@ -63,22 +61,13 @@ def main():
             e = min(s + bs, n)
             o = off[s:e + 1] - off[s]
             sl = slice(int(off[s]), int(off[e]))
-            if use_gpu:
-                backend.vw_sgd_minibatch_packed(idx[sl], val[sl], o,
-                                                labels[s:e], wg,
-                                                0.5, 0.0, 0.5, "logistic")
-            else:
-                backend.vw_sgd_minibatch(idx[sl], val[sl], o, labels[s:e],
-                                         w, gacc, 0.5, 0.0, 0.5, "logistic")
+            backend.vw_sgd_minibatch(idx[sl], val[sl], o, labels[s:e], w, gacc,
+                                     0.5, 0.0, 0.5, "logistic")
         if comm.is_distributed:
-            if use_gpu:
-                comm.all_reduce(wg)
-                wg.div_(world)
-            else:
-                comm.all_reduce(w)
-                w.div_(world)
-                comm.all_reduce(gacc)
-                gacc.div_(world)
+            comm.all_reduce(w)
+            w.div_(world)
+            comm.all_reduce(gacc)
+            gacc.div_(world)
 
     for _ in range(args.warmup):
         one_pass()
@@ -98,8 +87,6 @@ def main():
     elapsed = float(t[0])
 
     if rank == 0:
-        if use_gpu:
-            w = wg.view(-1, 2)[:, 0].contiguous()
         preds = backend.vw_predict(idx, val, off, w)
         acc = float((preds.sign() == labels).float().mean())
         print(json.dumps({

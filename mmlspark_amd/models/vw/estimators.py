@@ -253,11 +253,6 @@ class _VWBase(_VWParams, Estimator):
         n = len(df)
         learn_s = 0.0
         multipass_s = 0.0
-        use_packed = device.type == "cuda"
-        if use_packed and not self.get("bfgs"):
-            # interleave (w, G): both per-feature atomics land on one
-            # cacheline (the kernel is line-RMW bound — ops/hip/vw_kernels)
-            wg = torch.stack([w, g], dim=1).reshape(-1).contiguous()
         if self.get("bfgs"):
             # --bfgs: full-batch second-order optimization over the hashed
             # weight table (VowpalWabbitBase.scala arg surface).  The
@@ -274,41 +269,25 @@ class _VWBase(_VWParams, Estimator):
                 e = min(s + bs, n)
                 o = off[s:e + 1] - off[s]
                 sl = slice(int(off[s]), int(off[e]))
-                if use_packed:
-                    backend.vw_sgd_minibatch_packed(
-                        idx[sl], val[sl], o, labels[s:e], wg, lr, l2,
-                        power_t, loss,
-                        ex_w[s:e] if ex_w is not None else None, s_tbl,
-                        invariant=self.get("invariant"))
-                else:
-                    backend.vw_sgd_minibatch(
-                        idx[sl], val[sl], o, labels[s:e], w, g, lr, l2,
-                        power_t, loss,
-                        ex_w[s:e] if ex_w is not None else None, s_tbl,
-                        invariant=self.get("invariant"))
+                backend.vw_sgd_minibatch(
+                    idx[sl], val[sl], o, labels[s:e], w, g, lr, l2,
+                    power_t, loss,
+                    ex_w[s:e] if ex_w is not None else None, s_tbl,
+                    invariant=self.get("invariant"))
             learn_s += time.perf_counter() - t0
             # end-of-pass sync: RCCL all_reduce of weights + accumulators
             t0 = time.perf_counter()
             if comm.is_distributed:
-                if use_packed:
-                    comm.all_reduce(wg)  # one fused collective for w and G
-                    wg /= comm.world_size
-                else:
-                    comm.all_reduce(w)
-                    w /= comm.world_size
-                    comm.all_reduce(g)
-                    g /= comm.world_size
+                comm.all_reduce(w)
+                w /= comm.world_size
+                comm.all_reduce(g)
+                g /= comm.world_size
                 if s_tbl is not None:
                     comm.all_reduce(s_tbl, op="max")
             if l1 > 0:  # proximal truncation
-                wv = wg.view(-1, 2)[:, 0] if use_packed else w
-                wv.copy_(torch.sign(wv) * (wv.abs() - lr * l1).clamp_min(0))
+                w.copy_(torch.sign(w) * (w.abs() - lr * l1).clamp_min(0))
             multipass_s += time.perf_counter() - t0
 
-        if use_packed and not self.get("bfgs"):
-            pair = wg.view(-1, 2)
-            w = pair[:, 0].contiguous()
-            g = pair[:, 1].contiguous()
         model = self._model_class()(weights=w.cpu().numpy(),
                                     adaptive=g.cpu().numpy())
         for pname in ("labelCol", "featuresCol", "additionalFeatures",

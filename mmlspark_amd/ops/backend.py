@@ -147,17 +147,6 @@ def vw_sgd_minibatch(indices, values, offsets, labels, weights_tbl, adaptive_tbl
                                     invariant)
 
 
-def vw_sgd_minibatch_packed(indices, values, offsets, labels, wg_tbl,
-                            lr, l2, power_t, loss: str, ex_weight=None,
-                            normalize_tbl=None, invariant=False):
-    """GPU-only: hogwild SGD over an interleaved (w,G) table — both per-
-    feature atomics hit one cacheline (~1.6x on the line-RMW-bound kernel)."""
-    return _require_ext().vw_sgd_minibatch_packed(
-        indices, values, offsets, labels, wg_tbl, lr, l2, power_t,
-        {"squared": 0, "logistic": 1, "hinge": 2}[loss], ex_weight,
-        normalize_tbl, invariant)
-
-
 def vw_predict(indices, values, offsets, weights_tbl):
     if weights_tbl.is_cuda:
         return _require_ext().vw_predict(indices, values, offsets, weights_tbl)

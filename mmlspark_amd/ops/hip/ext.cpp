@@ -37,9 +37,6 @@ void launch_vw_sgd(const int*, const float*, const long*, const float*,
                    int, int, long, float*, hipStream_t);
 void launch_vw_predict(const int*, const float*, const long*, const float*,
                        long, float*, hipStream_t);
-void launch_vw_sgd_packed(const int*, const float*, const long*,
-                          const float*, const float*, float*, float*, float,
-                          float, float, int, int, long, float*, hipStream_t);
 void launch_tree_shap(const int*, const float*, const int*, const int*,
                       const float*, const float*, const long*, const int*,
                       const unsigned*, const float*, long, int, int, int, int,
@@ -246,28 +243,6 @@ torch::Tensor vw_sgd_minibatch(torch::Tensor idx, torch::Tensor val,
   return preds;
 }
 
-torch::Tensor vw_sgd_minibatch_packed(torch::Tensor idx, torch::Tensor val,
-                                      torch::Tensor off, torch::Tensor label,
-                                      torch::Tensor wg_tbl,
-                                      double lr, double l2, double power_t,
-                                      long loss,
-                                      c10::optional<torch::Tensor> ex_weight,
-                                      c10::optional<torch::Tensor> s_tbl,
-                                      bool invariant) {
-  CHECK_DEV(wg_tbl); CHECK_CONTIG(wg_tbl);
-  const long n_ex = off.numel() - 1;
-  auto preds = torch::zeros({n_ex}, wg_tbl.options());
-  const float* wptr = ex_weight.has_value() ? ex_weight->data_ptr<float>()
-                                            : nullptr;
-  float* sptr = s_tbl.has_value() ? s_tbl->data_ptr<float>() : nullptr;
-  launch_vw_sgd_packed(idx.data_ptr<int>(), val.data_ptr<float>(),
-                       off.data_ptr<long>(), label.data_ptr<float>(), wptr,
-                       wg_tbl.data_ptr<float>(), sptr, (float)lr, (float)l2,
-                       (float)power_t, (int)loss, invariant ? 1 : 0, n_ex,
-                       preds.data_ptr<float>(), cur_stream());
-  return preds;
-}
-
 torch::Tensor vw_predict(torch::Tensor idx, torch::Tensor val,
                          torch::Tensor off, torch::Tensor w_tbl) {
   CHECK_DEV(w_tbl); CHECK_CONTIG(w_tbl);
@@ -315,8 +290,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("partition_rows", &partition_rows,
         "stable ordered row partition, single sync");
   m.def("vw_sgd_minibatch", &vw_sgd_minibatch, "adaptive sparse SGD minibatch");
-  m.def("vw_sgd_minibatch_packed", &vw_sgd_minibatch_packed,
-        "sparse SGD over interleaved (w,G) table — same-cacheline atomics");
   m.def("vw_predict", &vw_predict, "sparse linear predict");
   m.def("tree_shap", &tree_shap, "path-dependent TreeSHAP contributions");
 }

@@ -146,112 +146,6 @@ __global__ void vw_sgd_k(const int* __restrict__ idx,
   }
 }
 
-// PACKED variant: (w, G) interleaved in one array so the two global
-// atomics of every feature update land on the SAME cacheline — the kernel
-// is line-RMW bound at the fabric, so this nearly halves its atomic
-// traffic.  s_tbl (--normalized) stays separate (read-mostly).
-__global__ void vw_sgd_packed_k(const int* __restrict__ idx,
-                         const float* __restrict__ val,
-                         const long* __restrict__ off,
-                         const float* __restrict__ label,
-                         const float* __restrict__ ex_weight,
-                         float* __restrict__ wg_tbl,
-                         float* __restrict__ s_tbl,
-                         float lr, float l2, float power_t, int loss,
-                         int invariant, long n_ex,
-                         float* __restrict__ preds_out) {
-  const long wid0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
-  const int lane = threadIdx.x & (WAVE - 1);
-  const long n_waves = ((long)gridDim.x * blockDim.x) / WAVE;
-  for (long ex = wid0; ex < n_ex; ex += n_waves) {
-    const long s = off[ex], e = off[ex + 1];
-    float dot = 0.0f;
-    for (long k = s + lane; k < e; k += WAVE) dot += wg_tbl[2 * (size_t)idx[k]] * val[k];
-    const float pred = wave_sum(dot);
-    if (preds_out && lane == 0) preds_out[ex] = pred;
-    const float y = label[ex];
-    const float h = ex_weight ? ex_weight[ex] : 1.0f;
-    float gl = dloss(loss, pred, y) * h;
-    if (invariant) {
-      // pass 1: sensitivity x_norm = sum x_i^2 * scale_i with the same
-      // per-coordinate scales the update will use (G + x^2 proxy)
-      float xn = 0.0f;
-      for (long k = s + lane; k < e; k += WAVE) {
-        const int i = idx[k];
-        const float x = val[k];
-        float sc;
-        const float Gp = wg_tbl[2 * (size_t)i + 1] + x * x;
-        if (power_t == 0.5f) sc = __frsqrt_rn(Gp + 1e-10f);
-        else sc = __powf(Gp + 1e-10f, -power_t);
-        if (s_tbl) {
-          const float ax = fabsf(x);
-          atomicMax((int*)&s_tbl[i], __float_as_int(ax));
-          const float sn = fmaxf(__int_as_float(((const int*)s_tbl)[i]), ax);
-          if (sn > 0.0f) sc /= sn;
-        }
-        xn += x * x * sc;
-      }
-      const float x_norm = wave_sum(xn);
-      if (x_norm <= 0.0f) continue;
-      const float dp = invariant_dp(loss, pred, y, h * lr * x_norm);
-      const float kk = dp / x_norm;
-      for (long k = s + lane; k < e; k += WAVE) {
-        const int i = idx[k];
-        const float x = val[k];
-        float sc;
-        const float Gp = wg_tbl[2 * (size_t)i + 1] + x * x;
-        if (power_t == 0.5f) sc = __frsqrt_rn(Gp + 1e-10f);
-        else sc = __powf(Gp + 1e-10f, -power_t);
-        if (s_tbl) {
-          const float sn = fmaxf(__int_as_float(((const int*)s_tbl)[i]),
-                                 fabsf(x));
-          if (sn > 0.0f) sc /= sn;
-        }
-        const float w = wg_tbl[2 * (size_t)i];
-        const float g = gl * x + l2 * w;
-        atomicAdd(&wg_tbl[2 * (size_t)i], (kk * x - lr * l2 * w) * sc);
-        atomicAdd(&wg_tbl[2 * (size_t)i + 1], g * g);
-      }
-      continue;
-    }
-    if (gl == 0.0f) continue;
-    for (long k = s + lane; k < e; k += WAVE) {
-      const int i = idx[k];
-      const float x = val[k];
-      float g = gl * x + l2 * wg_tbl[2 * (size_t)i];
-      const float Gold = atomicAdd(&wg_tbl[2 * (size_t)i + 1], g * g);
-      const float G = Gold + g * g;
-      // adaptive per-weight rate: lr * G^(-power_t); power_t=0.5 → rsqrt
-      float scale;
-      if (power_t == 0.5f) scale = __frsqrt_rn(G + 1e-10f);
-      else scale = __powf(G + 1e-10f, -power_t);
-      if (s_tbl) {  // --normalized: divide by running max|x| per weight
-        const float ax = fabsf(x);
-        atomicMax((int*)&s_tbl[i], __float_as_int(ax));
-        const float sn = fmaxf(__int_as_float(((const int*)s_tbl)[i]), ax);
-        if (sn > 0.0f) scale /= sn;
-      }
-      atomicAdd(&wg_tbl[2 * (size_t)i], -lr * g * scale);
-    }
-  }
-}
-
-
-extern "C" void launch_vw_sgd_packed(const int* idx, const float* val,
-                                     const long* off, const float* label,
-                                     const float* ex_weight, float* wg_tbl,
-                                     float* s_tbl, float lr, float l2,
-                                     float power_t, int loss, int invariant,
-                                     long n_ex, float* preds_out,
-                                     hipStream_t stream) {
-  if (n_ex == 0) return;
-  long blocks = (n_ex * WAVE + 255) / 256;
-  if (blocks > 4096) blocks = 4096;
-  hipLaunchKernelGGL(vw_sgd_packed_k, dim3((unsigned)blocks), dim3(256), 0,
-                     stream, idx, val, off, label, ex_weight, wg_tbl, s_tbl,
-                     lr, l2, power_t, loss, invariant, n_ex, preds_out);
-}
-
 extern "C" void launch_vw_sgd(const int* idx, const float* val,
                               const long* off, const float* label,
                               const float* ex_weight,

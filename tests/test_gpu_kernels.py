@@ -442,27 +442,3 @@ def test_vw_bfgs_gpu(binary_df):
     y = binary_df["label"].to_numpy()
     acc = (m.transform(binary_df)["prediction"].to_numpy() == y).mean()
     assert acc > 0.9, acc
-
-
-@requires_gpu
-def test_vw_packed_matches_separate_tables():
-    """Interleaved (w,G) kernel == separate-tables kernel on collision-free
-    data (same updates, same-cacheline atomics are a pure layout change)."""
-    from mmlspark_amd.ops import backend
-    g = torch.Generator().manual_seed(17)
-    tbl = 1 << 18
-    n, fp = 8000, 30
-    idx = (torch.arange(n * fp, dtype=torch.int32) % tbl).cuda()
-    val = torch.randn(n * fp, generator=g).cuda()
-    off = (torch.arange(0, n + 1, dtype=torch.int64) * fp).cuda()
-    y = torch.where(torch.rand(n, generator=g) > 0.5, 1.0, -1.0).cuda()
-    w = torch.zeros(tbl).cuda()
-    gacc = torch.zeros(tbl).cuda()
-    backend.vw_sgd_minibatch(idx, val, off, y, w, gacc, 0.5, 0.0, 0.5,
-                             "logistic")
-    wg = torch.zeros(2 * tbl).cuda()
-    backend.vw_sgd_minibatch_packed(idx, val, off, y, wg, 0.5, 0.0, 0.5,
-                                    "logistic")
-    pair = wg.view(-1, 2)
-    assert torch.allclose(pair[:, 0], w, atol=1e-6)
-    assert torch.allclose(pair[:, 1], gacc, atol=1e-6)
