@@ -153,3 +153,26 @@ def test_dictionary_examples_payload():
     row = pd.Series({"t": "hello", "tr": "hola"})
     p = d._payload(row)
     assert p == [{"Text": "hello", "Translation": "hola"}]
+
+
+def test_azure_search_writer_batches(mock_service):
+    """AzureSearchWriter (AzureSearchAPI.scala sink): docs batched into
+    @search.action-tagged POSTs; every row carries the batch response."""
+    from mmlspark_amd.io_http.cognitive import AzureSearchWriter
+    docs = [{"id": str(i), "text": f"d{i}"} for i in range(7)]
+    df = pd.DataFrame({"doc": docs})
+    w = AzureSearchWriter(url=mock_service, batchSize=3, actionType="upload",
+                          subscriptionKey="k", outputCol="result")
+    out = w.transform(df)
+    assert len(out) == 7
+    assert out["result"].notna().all()
+    # 7 docs at batchSize 3 → 3 distinct batch responses referenced
+    assert len({id(r) for r in out["result"]}) <= 3
+
+
+def test_speech_to_text_posts_audio_bytes():
+    from mmlspark_amd.io_http.cognitive import SpeechToTextSDK
+    s = SpeechToTextSDK(url="http://x/", audioBytesCol="audio")
+    row = pd.Series({"audio": b"RIFFxxxx"})
+    assert s._payload(row) is None           # body is raw audio, not JSON
+    assert s._headers(row)["Content-Type"] == "audio/wav"
