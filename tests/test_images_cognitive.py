@@ -166,8 +166,10 @@ def test_azure_search_writer_batches(mock_service):
     out = w.transform(df)
     assert len(out) == 7
     assert out["result"].notna().all()
-    # 7 docs at batchSize 3 → 3 distinct batch responses referenced
-    assert len({id(r) for r in out["result"]}) <= 3
+    # 7 docs at batchSize 3 → 3 distinct batch responses (dict per row)
+    uniq = {json.dumps(r, sort_keys=True, default=str)
+            for r in out["result"]}
+    assert len(uniq) <= 3
 
 
 def test_speech_to_text_posts_audio_bytes():
