@@ -216,11 +216,20 @@ class DeepVisionClassifier(Estimator):
 @register
 class DeepVisionModel(TorchModel):
     predictionCol = Param("predictionCol", "prediction column", "prediction")
+    probabilityCol = Param("probabilityCol", "softmax probability column",
+                           "probability")
 
     def _transform(self, df):
         out = super()._transform(df)
         logits = np.stack(out[self.get("outputCol")].to_numpy()) if len(out) \
             else np.zeros((0, 1))
+        if len(out):
+            z = logits - logits.max(axis=1, keepdims=True)
+            e = np.exp(z)
+            out[self.get("probabilityCol")] = matrix_to_vector_column(
+                e / e.sum(axis=1, keepdims=True))
+        else:
+            out[self.get("probabilityCol")] = []
         out[self.get("predictionCol")] = logits.argmax(axis=1).astype(np.float64) \
             if len(out) else []
         return out
