@@ -1,6 +1,8 @@
 """File IO codecs, PowerBI writer, codegen stub/doc generation."""
 import os
 
+import json
+
 import numpy as np
 import pandas as pd
 import pytest
@@ -176,3 +178,46 @@ def test_jpeg_codec_roundtrip():
     dgray = decode_image(encode_image(gray, "jpg"))
     assert dgray.shape == gray.shape
     assert np.abs(dgray.astype(int) - gray.astype(int)).max() <= 4
+
+
+def test_jpeg_huffman_tables_are_inverse():
+    from mmlspark_amd.io_http.jpeg_codec import (HT_AC_C, HT_AC_L, HT_DC_C,
+                                                 HT_DC_L, _build_codes,
+                                                 _build_decoder)
+    for bits, vals in (HT_DC_L, HT_DC_C, HT_AC_L, HT_AC_C):
+        enc = _build_codes(bits, vals)
+        dec = _build_decoder(bits, vals)
+        assert len(enc) == len(vals)
+        for v, (code, length) in enc.items():
+            assert dec[(length, code)] == v
+        # prefix-free: no code is a prefix of a longer one
+        codes = sorted(((l, c) for c, l in enc.values()))
+        for i, (l1, c1) in enumerate(codes):
+            for l2, c2 in codes[i + 1:]:
+                if l2 > l1:
+                    assert (c2 >> (l2 - l1)) != c1
+
+
+def test_http_transformer_concurrent_requests():
+    """HTTPTransformer with a thread pool (AsyncHTTPClient analog) keeps
+    row-to-response alignment under concurrency."""
+    from mmlspark_amd.io_http.client import HTTPTransformer
+    from mmlspark_amd.io_http.http_schema import HTTPRequestData
+    from mmlspark_amd.serving.server import ServingServer
+
+    srv = ServingServer(lambda ps: [{"echo": p["i"]} for p in ps],
+                        port=0, mode="continuous").start()
+    try:
+        n = 24
+        df = pd.DataFrame({"req": [
+            HTTPRequestData(url=f"http://127.0.0.1:{srv.port}/",
+                            method="POST",
+                            headers={"Content-Type": "application/json"},
+                            entity=json.dumps({"i": i}).encode())
+            for i in range(n)]})
+        out = HTTPTransformer(inputCol="req", outputCol="resp",
+                              concurrency=8).transform(df)
+        got = [r.json()["echo"] for r in out["resp"]]
+        assert got == list(range(n))  # order preserved despite concurrency
+    finally:
+        srv.stop()
