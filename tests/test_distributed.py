@@ -289,3 +289,49 @@ def test_distributed_bfgs_identical_weights():
         w1, acc1 = out[1]
     np.testing.assert_allclose(w0, w1, rtol=0, atol=0)  # bit-identical
     assert acc0 > 0.9 and acc1 > 0.9
+
+
+@pytest.mark.timeout(240)
+def test_distributed_gbdt_ws4_identical():
+    """world_size=4 (the scale bench's mid point): synchronized growth still
+    yields the identical booster on every rank."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [ctx.Process(target=_worker_gbdt_ws4, args=(r, 4, port, q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(4):
+        rank, s = q.get(timeout=200)
+        results[rank] = s
+    for p in procs:
+        p.join(timeout=30)
+    assert not any(str(s).startswith("ERROR") for s in results.values()), \
+        {k: str(v)[:200] for k, v in results.items()}
+    base = results[0]
+    for r in range(1, 4):
+        assert results[r] == base  # byte-identical model JSON
+
+
+def _worker_gbdt_ws4(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        torch.distributed.init_process_group(
+            "gloo", init_method=f"tcp://127.0.0.1:{port}",
+            rank=rank, world_size=world)
+        from mmlspark_amd.models.gbdt.objectives import make_objective
+        from mmlspark_amd.models.gbdt.trainer import TrainConfig, train_booster
+        from mmlspark_amd.parallel.comm import Comm
+
+        X, y = _make_data(0, n=4000)
+        sl = slice(rank * 1000, (rank + 1) * 1000)
+        booster, _ = train_booster(
+            torch.from_numpy(X[sl]), torch.from_numpy(y[sl]),
+            TrainConfig(num_iterations=6, num_leaves=15, seed=3),
+            make_objective("binary"), Comm())
+        q.put((rank, booster.save_to_string()))
+        torch.distributed.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"ERROR: {e!r}"))
