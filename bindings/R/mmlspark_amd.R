@@ -342,7 +342,7 @@ ml_deep_vision_classifier <- function(labelCol = NULL, imageCol = NULL, predicti
   stage
 }
 
-ml_deep_vision_model <- function(inputCol = NULL, outputCol = NULL, batchSize = NULL, moduleBytes = NULL, device = NULL, predictionCol = NULL) {
+ml_deep_vision_model <- function(inputCol = NULL, outputCol = NULL, batchSize = NULL, moduleBytes = NULL, device = NULL, predictionCol = NULL, probabilityCol = NULL) {
   stage <- mmlspark_amd$models$image_featurizer$DeepVisionModel()
   if (!is.null(inputCol)) stage$set("inputCol", inputCol)
   if (!is.null(outputCol)) stage$set("outputCol", outputCol)
@@ -350,6 +350,7 @@ ml_deep_vision_model <- function(inputCol = NULL, outputCol = NULL, batchSize = 
   if (!is.null(moduleBytes)) stage$set("moduleBytes", moduleBytes)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(predictionCol)) stage$set("predictionCol", predictionCol)
+  if (!is.null(probabilityCol)) stage$set("probabilityCol", probabilityCol)
   stage
 }
 
@@ -1392,6 +1393,15 @@ ml_repartition <- function(n = NULL, disable = NULL) {
   stage
 }
 
+ml_resize_image_transformer <- function(inputCol = NULL, outputCol = NULL, height = NULL, width = NULL) {
+  stage <- mmlspark_amd$models$images$ResizeImageTransformer()
+  if (!is.null(inputCol)) stage$set("inputCol", inputCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(height)) stage$set("height", height)
+  if (!is.null(width)) stage$set("width", width)
+  stage
+}
+
 ml_sar <- function(userCol = NULL, itemCol = NULL, ratingCol = NULL, timeCol = NULL, supportThreshold = NULL, similarityFunction = NULL, timeDecayCoeff = NULL) {
   stage <- mmlspark_amd$models$sar$SAR()
   if (!is.null(userCol)) stage$set("userCol", userCol)
@@ -1811,6 +1821,22 @@ ml_unicode_normalize <- function(inputCol = NULL, outputCol = NULL, form = NULL,
   if (!is.null(outputCol)) stage$set("outputCol", outputCol)
   if (!is.null(form)) stage$set("form", form)
   if (!is.null(lower)) stage$set("lower", lower)
+  stage
+}
+
+ml_unroll_binary_image <- function(inputCol = NULL, outputCol = NULL, height = NULL, width = NULL) {
+  stage <- mmlspark_amd$models$images$UnrollBinaryImage()
+  if (!is.null(inputCol)) stage$set("inputCol", inputCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(height)) stage$set("height", height)
+  if (!is.null(width)) stage$set("width", width)
+  stage
+}
+
+ml_unroll_image <- function(inputCol = NULL, outputCol = NULL) {
+  stage <- mmlspark_amd$models$images$UnrollImage()
+  if (!is.null(inputCol)) stage$set("inputCol", inputCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
   stage
 }
 

@@ -13,7 +13,7 @@ import numpy as np
 import pandas as pd
 import torch
 
-from ..core.param import Param, toList, toString
+from ..core.param import Param, toList, toString, toInt
 from ..core.pipeline import Transformer
 from ..core.registry import register
 
@@ -183,3 +183,67 @@ class ImageSetAugmenter(Transformer):
                                         for v in df[self.get("inputCol")]]
             frames.append(f)
         return pd.concat(frames, ignore_index=True)
+
+
+@register
+class ResizeImageTransformer(Transformer):
+    """Standalone resize (image/ResizeImageTransformer.scala): bilinear
+    resize to (height, width), preserving dtype."""
+    inputCol = Param("inputCol", "image column", "image")
+    outputCol = Param("outputCol", "output column", "image")
+    height = Param("height", "target height", 224, toInt)
+    width = Param("width", "target width", 224, toInt)
+
+    def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
+        h, w = self.get("height"), self.get("width")
+        out = df.copy()
+        out[self.get("outputCol")] = [
+            _to_array(_apply_stage(_to_tensor(np.asarray(v)),
+                                   {"op": "resize", "height": h, "width": w}),
+                      np.asarray(v).dtype == np.uint8)
+            for v in df[self.get("inputCol")]]
+        return out
+
+
+@register
+class UnrollImage(Transformer):
+    """Image → flat float vector (image/UnrollImage.scala:151): row-major
+    HWC unroll scaled to [0,1] floats — the shape CNTKModel-style consumers
+    expect."""
+    inputCol = Param("inputCol", "image column", "image")
+    outputCol = Param("outputCol", "unrolled vector column", "unrolled")
+
+    def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
+        out = df.copy()
+        vecs = []
+        for v in df[self.get("inputCol")]:
+            a = np.asarray(v)
+            a = a.astype(np.float32) / (255.0 if a.dtype == np.uint8 else 1.0)
+            vecs.append(a.reshape(-1))
+        out[self.get("outputCol")] = vecs
+        return out
+
+
+@register
+class UnrollBinaryImage(Transformer):
+    """Encoded image bytes → decoded (+optional resize) → flat vector
+    (image/UnrollImage.scala:186 UnrollBinaryImage)."""
+    inputCol = Param("inputCol", "image-bytes column", "data")
+    outputCol = Param("outputCol", "unrolled vector column", "unrolled")
+    height = Param("height", "optional resize height (0 = keep)", 0, toInt)
+    width = Param("width", "optional resize width (0 = keep)", 0, toInt)
+
+    def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
+        from ..io_http.files import decode_image
+        h, w = self.get("height"), self.get("width")
+        out = df.copy()
+        vecs = []
+        for b in df[self.get("inputCol")]:
+            img = decode_image(bytes(b))
+            if h and w:
+                img = _to_array(_apply_stage(
+                    _to_tensor(img), {"op": "resize", "height": h,
+                                      "width": w}), True)
+            vecs.append(img.astype(np.float32).reshape(-1) / 255.0)
+        out[self.get("outputCol")] = vecs
+        return out

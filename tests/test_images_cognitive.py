@@ -178,3 +178,22 @@ def test_speech_to_text_posts_audio_bytes():
     row = pd.Series({"audio": b"RIFFxxxx"})
     assert s._payload(row) is None           # body is raw audio, not JSON
     assert s._headers(row)["Content-Type"] == "audio/wav"
+
+
+def test_unroll_and_resize_stages():
+    """image/ subpackage parity (UnrollImage.scala:151,186 +
+    ResizeImageTransformer.scala): unroll to [0,1] floats, binary decode
+    path, standalone resize."""
+    from mmlspark_amd.io_http.files import encode_image
+    from mmlspark_amd.models.images import (ResizeImageTransformer,
+                                            UnrollBinaryImage, UnrollImage)
+    rng = np.random.default_rng(4)
+    img = rng.integers(0, 255, size=(20, 16, 3)).astype(np.uint8)
+    df = pd.DataFrame({"image": [img], "data": [encode_image(img, "png")]})
+    r = ResizeImageTransformer(height=8, width=8).transform(df)
+    assert np.asarray(r["image"].iloc[0]).shape == (8, 8, 3)
+    u = UnrollImage().transform(df)["unrolled"].iloc[0]
+    assert u.shape == (20 * 16 * 3,) and 0 <= u.min() and u.max() <= 1
+    np.testing.assert_allclose(u[:3], img[0, 0].astype(np.float32) / 255)
+    ub = UnrollBinaryImage(height=8, width=8).transform(df)["unrolled"].iloc[0]
+    assert ub.shape == (8 * 8 * 3,)
