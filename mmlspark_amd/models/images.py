@@ -207,9 +207,10 @@ class ResizeImageTransformer(Transformer):
 
 @register
 class UnrollImage(Transformer):
-    """Image → flat float vector (image/UnrollImage.scala:151): row-major
-    HWC unroll scaled to [0,1] floats — the shape CNTKModel-style consumers
-    expect."""
+    """Image → flat float vector in CHANNEL-MAJOR (CHW) order with raw
+    0-255 values — exactly the reference's rearrangement
+    (image/UnrollImage.scala:30-55: index = h*W*C + w*C + c, emitted
+    channel-by-channel)."""
     inputCol = Param("inputCol", "image column", "image")
     outputCol = Param("outputCol", "unrolled vector column", "unrolled")
 
@@ -218,8 +219,10 @@ class UnrollImage(Transformer):
         vecs = []
         for v in df[self.get("inputCol")]:
             a = np.asarray(v)
-            a = a.astype(np.float32) / (255.0 if a.dtype == np.uint8 else 1.0)
-            vecs.append(a.reshape(-1))
+            if a.ndim == 2:
+                a = a[:, :, None]
+            vecs.append(np.ascontiguousarray(
+                a.transpose(2, 0, 1)).reshape(-1).astype(np.float64))
         out[self.get("outputCol")] = vecs
         return out
 
@@ -244,6 +247,9 @@ class UnrollBinaryImage(Transformer):
                 img = _to_array(_apply_stage(
                     _to_tensor(img), {"op": "resize", "height": h,
                                       "width": w}), True)
-            vecs.append(img.astype(np.float32).reshape(-1) / 255.0)
+            if img.ndim == 2:
+                img = img[:, :, None]
+            vecs.append(np.ascontiguousarray(
+                img.transpose(2, 0, 1)).reshape(-1).astype(np.float64))
         out[self.get("outputCol")] = vecs
         return out

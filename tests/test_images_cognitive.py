@@ -193,7 +193,8 @@ def test_unroll_and_resize_stages():
     r = ResizeImageTransformer(height=8, width=8).transform(df)
     assert np.asarray(r["image"].iloc[0]).shape == (8, 8, 3)
     u = UnrollImage().transform(df)["unrolled"].iloc[0]
-    assert u.shape == (20 * 16 * 3,) and 0 <= u.min() and u.max() <= 1
-    np.testing.assert_allclose(u[:3], img[0, 0].astype(np.float32) / 255)
+    assert u.shape == (20 * 16 * 3,) and u.max() > 1  # raw 0-255, CHW
+    # channel-major: first W entries are channel 0 of row 0
+    np.testing.assert_allclose(u[:16], img[0, :, 0].astype(np.float64))
     ub = UnrollBinaryImage(height=8, width=8).transform(df)["unrolled"].iloc[0]
     assert ub.shape == (8 * 8 * 3,)
