@@ -905,7 +905,7 @@ ml_light_gbm_classification_model <- function(labelCol = NULL, featuresCol = NUL
   stage
 }
 
-ml_light_gbm_classifier <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, categoricalSlotNames = NULL, slotNames = NULL, modelString = NULL, lightGBMBooster = NULL, fobj = NULL, boostFromAverage = NULL, improvementTolerance = NULL, posBaggingFraction = NULL, negBaggingFraction = NULL, binSampleCount = NULL, maxBinByFeature = NULL, uniformDrop = NULL, xgboostDartMode = NULL, startIteration = NULL, chunkSize = NULL, defaultListenPort = NULL, driverListenPort = NULL, timeout = NULL, numTasks = NULL, numThreads = NULL, useSingleDatasetMode = NULL, matrixType = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL, isUnbalance = NULL) {
+ml_light_gbm_classifier <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, categoricalSlotNames = NULL, slotNames = NULL, modelString = NULL, lightGBMBooster = NULL, fobj = NULL, boostFromAverage = NULL, improvementTolerance = NULL, posBaggingFraction = NULL, negBaggingFraction = NULL, binSampleCount = NULL, maxBinByFeature = NULL, uniformDrop = NULL, xgboostDartMode = NULL, startIteration = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, chunkSize = NULL, defaultListenPort = NULL, driverListenPort = NULL, timeout = NULL, numTasks = NULL, numThreads = NULL, useSingleDatasetMode = NULL, matrixType = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL, isUnbalance = NULL) {
   stage <- mmlspark_amd$models$gbdt$estimators$LightGBMClassifier()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -960,6 +960,8 @@ ml_light_gbm_classifier <- function(labelCol = NULL, featuresCol = NULL, feature
   if (!is.null(uniformDrop)) stage$set("uniformDrop", uniformDrop)
   if (!is.null(xgboostDartMode)) stage$set("xgboostDartMode", xgboostDartMode)
   if (!is.null(startIteration)) stage$set("startIteration", startIteration)
+  if (!is.null(leafPredictionCol)) stage$set("leafPredictionCol", leafPredictionCol)
+  if (!is.null(featuresShapCol)) stage$set("featuresShapCol", featuresShapCol)
   if (!is.null(chunkSize)) stage$set("chunkSize", chunkSize)
   if (!is.null(defaultListenPort)) stage$set("defaultListenPort", defaultListenPort)
   if (!is.null(driverListenPort)) stage$set("driverListenPort", driverListenPort)
@@ -975,7 +977,7 @@ ml_light_gbm_classifier <- function(labelCol = NULL, featuresCol = NULL, feature
   stage
 }
 
-ml_light_gbm_ranker <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, categoricalSlotNames = NULL, slotNames = NULL, modelString = NULL, lightGBMBooster = NULL, fobj = NULL, boostFromAverage = NULL, improvementTolerance = NULL, posBaggingFraction = NULL, negBaggingFraction = NULL, binSampleCount = NULL, maxBinByFeature = NULL, uniformDrop = NULL, xgboostDartMode = NULL, startIteration = NULL, chunkSize = NULL, defaultListenPort = NULL, driverListenPort = NULL, timeout = NULL, numTasks = NULL, numThreads = NULL, useSingleDatasetMode = NULL, matrixType = NULL, device = NULL, groupCol = NULL, labelGain = NULL, maxPosition = NULL, evalAt = NULL) {
+ml_light_gbm_ranker <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, categoricalSlotNames = NULL, slotNames = NULL, modelString = NULL, lightGBMBooster = NULL, fobj = NULL, boostFromAverage = NULL, improvementTolerance = NULL, posBaggingFraction = NULL, negBaggingFraction = NULL, binSampleCount = NULL, maxBinByFeature = NULL, uniformDrop = NULL, xgboostDartMode = NULL, startIteration = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, chunkSize = NULL, defaultListenPort = NULL, driverListenPort = NULL, timeout = NULL, numTasks = NULL, numThreads = NULL, useSingleDatasetMode = NULL, matrixType = NULL, device = NULL, groupCol = NULL, labelGain = NULL, maxPosition = NULL, evalAt = NULL) {
   stage <- mmlspark_amd$models$gbdt$estimators$LightGBMRanker()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1030,6 +1032,8 @@ ml_light_gbm_ranker <- function(labelCol = NULL, featuresCol = NULL, featureCols
   if (!is.null(uniformDrop)) stage$set("uniformDrop", uniformDrop)
   if (!is.null(xgboostDartMode)) stage$set("xgboostDartMode", xgboostDartMode)
   if (!is.null(startIteration)) stage$set("startIteration", startIteration)
+  if (!is.null(leafPredictionCol)) stage$set("leafPredictionCol", leafPredictionCol)
+  if (!is.null(featuresShapCol)) stage$set("featuresShapCol", featuresShapCol)
   if (!is.null(chunkSize)) stage$set("chunkSize", chunkSize)
   if (!is.null(defaultListenPort)) stage$set("defaultListenPort", defaultListenPort)
   if (!is.null(driverListenPort)) stage$set("driverListenPort", driverListenPort)
@@ -1076,7 +1080,7 @@ ml_light_gbm_regression_model <- function(labelCol = NULL, featuresCol = NULL, f
   stage
 }
 
-ml_light_gbm_regressor <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, categoricalSlotNames = NULL, slotNames = NULL, modelString = NULL, lightGBMBooster = NULL, fobj = NULL, boostFromAverage = NULL, improvementTolerance = NULL, posBaggingFraction = NULL, negBaggingFraction = NULL, binSampleCount = NULL, maxBinByFeature = NULL, uniformDrop = NULL, xgboostDartMode = NULL, startIteration = NULL, chunkSize = NULL, defaultListenPort = NULL, driverListenPort = NULL, timeout = NULL, numTasks = NULL, numThreads = NULL, useSingleDatasetMode = NULL, matrixType = NULL, device = NULL, alpha = NULL, tweedieVariancePower = NULL) {
+ml_light_gbm_regressor <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, categoricalSlotNames = NULL, slotNames = NULL, modelString = NULL, lightGBMBooster = NULL, fobj = NULL, boostFromAverage = NULL, improvementTolerance = NULL, posBaggingFraction = NULL, negBaggingFraction = NULL, binSampleCount = NULL, maxBinByFeature = NULL, uniformDrop = NULL, xgboostDartMode = NULL, startIteration = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, chunkSize = NULL, defaultListenPort = NULL, driverListenPort = NULL, timeout = NULL, numTasks = NULL, numThreads = NULL, useSingleDatasetMode = NULL, matrixType = NULL, device = NULL, alpha = NULL, tweedieVariancePower = NULL) {
   stage <- mmlspark_amd$models$gbdt$estimators$LightGBMRegressor()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1131,6 +1135,8 @@ ml_light_gbm_regressor <- function(labelCol = NULL, featuresCol = NULL, featureC
   if (!is.null(uniformDrop)) stage$set("uniformDrop", uniformDrop)
   if (!is.null(xgboostDartMode)) stage$set("xgboostDartMode", xgboostDartMode)
   if (!is.null(startIteration)) stage$set("startIteration", startIteration)
+  if (!is.null(leafPredictionCol)) stage$set("leafPredictionCol", leafPredictionCol)
+  if (!is.null(featuresShapCol)) stage$set("featuresShapCol", featuresShapCol)
   if (!is.null(chunkSize)) stage$set("chunkSize", chunkSize)
   if (!is.null(defaultListenPort)) stage$set("defaultListenPort", defaultListenPort)
   if (!is.null(driverListenPort)) stage$set("driverListenPort", driverListenPort)

@@ -109,6 +109,12 @@ class _GBDTParams(Params):
                             "mode", False, toBool)
     startIteration = Param("startIteration", "first iteration used at "
                            "predict time", 0, toInt)
+    leafPredictionCol = Param("leafPredictionCol",
+                              "output column for per-tree leaf indices "
+                              "(propagated to the model)", None)
+    featuresShapCol = Param("featuresShapCol",
+                            "output column for SHAP contributions "
+                            "(propagated to the model)", None)
     # Spark/JVM topology params accepted for API compatibility; topology is
     # one-process-per-GPU with RCCL here, so they change nothing (documented
     # in PARITY.md — LightGBMParams.scala:54-100)
@@ -249,7 +255,8 @@ class _GBDTEstimatorBase(_GBDTParams, Estimator):
                 metrics_fn=default_metrics_fn(self.get("metric") or None),
                 init_score=init_score)
         model = self._model_class()(booster=booster)
-        for p in ("labelCol", "featuresCol", "featureCols", "predictionCol"):
+        for p in ("labelCol", "featuresCol", "featureCols", "predictionCol",
+                  "leafPredictionCol", "featuresShapCol"):
             model.set(p, self.get(p))
         model._training_stats = stats
         return model
