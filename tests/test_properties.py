@@ -78,3 +78,24 @@ def test_invariant_update_never_overshoots(h, lr):
     dp = _invariant_dp("squared", pred, y, torch.tensor([h * lr * x_norm]))
     new = float(pred + dp)
     assert 0.0 <= new <= 3.0 + 1e-5
+
+
+@given(st.integers(1, 30), st.integers(2, 8), st.integers(0, 10_000))
+@settings(max_examples=15, deadline=None)
+def test_lightgbm_text_roundtrip_random_boosters(n_trees, n_leaves, seed):
+    """Property: ANY trained booster survives to_lightgbm_text →
+    load_from_string with identical predictions."""
+    import pandas as pd
+    from mmlspark_amd.models.gbdt.booster import Booster
+    from mmlspark_amd.models.gbdt.estimators import LightGBMRegressor
+    rng = np.random.default_rng(seed)
+    X = rng.normal(size=(400, 4)).astype(np.float32)
+    y = (X[:, 0] * rng.normal() + np.sin(X[:, 1])).astype(np.float32)
+    df = pd.DataFrame({"features": list(X), "label": y})
+    m = LightGBMRegressor(numIterations=n_trees, numLeaves=n_leaves,
+                          minDataInLeaf=5, seed=seed).fit(df)
+    b2 = Booster.load_from_string(m.booster.to_lightgbm_text())
+    Xt = torch.from_numpy(X[:50])
+    np.testing.assert_allclose(b2.predict_raw(Xt).numpy(),
+                               m.booster.predict_raw(Xt).numpy(),
+                               rtol=1e-5, atol=1e-5)
