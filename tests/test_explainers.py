@@ -171,3 +171,21 @@ def test_image_shap():
     main_seg = int(vals[counts.argmax()])
     # the bright-corner superpixel carries the largest contribution
     assert exp[1 + main_seg] == exp[1:].max()
+
+
+def test_explainer_save_load_with_dataframe_param(tmp_path, model_and_data):
+    """ComplexParam DataFrame side-file (parquet) round trip through an
+    explainer — the ComplexParamsWriter path for backgroundData
+    (serialize.ComplexParam parity)."""
+    from mmlspark_amd.core.serialize import load_stage, save_stage
+    model, df, cols = model_and_data
+    shap = TabularSHAP(inputCols=cols, model=model, targetCol="probability",
+                       targetClasses=[1], numSamples=64,
+                       backgroundData=df.head(40))
+    path = str(tmp_path / "shap_stage")
+    save_stage(shap, path)
+    back = load_stage(path)
+    pd.testing.assert_frame_equal(back.get("backgroundData"),
+                                  shap.get("backgroundData"))
+    out = back.transform(df.head(2))
+    assert np.stack(out["explanation"].to_numpy()).shape == (2, 1, 6)
