@@ -7,7 +7,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import numpy as np
 import pandas as pd
 
-from mmlspark_amd.models.sar import SAR, RankingEvaluator, RecommendationIndexer
+from mmlspark_amd.models.sar import SAR, RecommendationIndexer
 
 rng = np.random.default_rng(0)
 rows = []

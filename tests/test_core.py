@@ -2,7 +2,6 @@
 import os
 
 import numpy as np
-import pandas as pd
 import pytest
 
 import mmlspark_amd as M

@@ -5,7 +5,6 @@ import json
 import numpy as np
 import pandas as pd
 import pytest
-import requests
 
 from mmlspark_amd.models.images import ImageSetAugmenter, ImageTransformer
 from mmlspark_amd.serving.server import ServingServer

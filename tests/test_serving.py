@@ -1,5 +1,4 @@
 """Serving server (continuous + micro-batch) and HTTP client stack."""
-import json
 import threading
 import time
 
