@@ -198,7 +198,6 @@ def test_categorical_features():
     # trees actually used categorical splits
     assert any((t.cat_offset >= 0).any() for t in m.booster.trees)
     # save/load preserves bitsets
-    import json as _json
     b2 = m.booster.load_from_string(m.booster.save_to_string())
     X_t = torch.from_numpy(X.astype(np.float32))
     assert torch.allclose(m.booster.predict_raw(X_t), b2.predict_raw(X_t))
