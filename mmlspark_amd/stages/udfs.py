@@ -43,3 +43,17 @@ def ml_transform(df: pd.DataFrame, *stages) -> pd.DataFrame:
 def ml_fit(df: pd.DataFrame, estimator):
     """Fluent fit (FluentAPI mlFit)."""
     return estimator.fit(df)
+
+
+def install_fluent_api():
+    """Opt-in monkey patch (core/spark/FluentAPI.py parity): gives every
+    pandas DataFrame `.mlTransform(*stages)` and `.mlFit(estimator)` so
+    reference-style fluent chains port verbatim."""
+    def _ml_transform(self, *stages):
+        return ml_transform(self, *stages)
+
+    def _ml_fit(self, estimator):
+        return ml_fit(self, estimator)
+
+    pd.DataFrame.mlTransform = _ml_transform
+    pd.DataFrame.mlFit = _ml_fit

@@ -290,3 +290,17 @@ def test_plot_helpers(tmp_path):
     assert cm.sum() == 500
     assert cm[1, 1] > cm[1, 0]  # classifier is informative
     plt.close("all")
+
+
+def test_fluent_api_monkeypatch(binary_df):
+    """install_fluent_api (FluentAPI.py parity): df.mlFit / df.mlTransform."""
+    from mmlspark_amd.stages.basic import SelectColumns
+    from mmlspark_amd.stages.udfs import install_fluent_api
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    install_fluent_api()
+    model = binary_df.mlFit(LightGBMClassifier(numIterations=5, numLeaves=7))
+    out = binary_df.mlTransform(
+        model, SelectColumns(cols=["prediction", "label"]))
+    assert list(out.columns) == ["prediction", "label"]
+    acc = (out["prediction"] == out["label"]).mean()
+    assert acc > 0.8
