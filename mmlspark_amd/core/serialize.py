@@ -152,7 +152,16 @@ def load_stage(path: str):
                 n = json.load(f)
             v = [load_stage(os.path.join(base, str(i))) for i in range(n)]
         elif kind == "tensors":
-            v = torch.load(base + ".pt", map_location="cpu", weights_only=False)
+            # state dicts / tensor payloads only — weights_only forbids
+            # arbitrary pickle execution when pointed at untrusted paths.
+            # (Set MMLSPARK_AMD_TRUSTED_LOAD=1 only for trusted legacy files
+            # that predate the weights-only format.)
+            if os.environ.get("MMLSPARK_AMD_TRUSTED_LOAD"):
+                v = torch.load(base + ".pt", map_location="cpu",
+                               weights_only=False)
+            else:
+                v = torch.load(base + ".pt", map_location="cpu",
+                               weights_only=True)
         elif kind == "bytes":
             with open(base + ".bin", "rb") as f:
                 v = f.read()

@@ -263,7 +263,8 @@ def _to_lightgbm_text(booster: "Booster") -> str:
     saveNativeModel gives its users (LightGBMClassifier.scala:185-205).
     Categorical bitset splits export with decision_type=1 (==) per LightGBM's
     categorical encoding via the cat_boundaries/cat_threshold arrays."""
-    obj = {"binary": "binary sigmoid:1", "multiclass": "multiclass",
+    sig = float(getattr(booster, "sigmoid", 1.0))
+    obj = {"binary": f"binary sigmoid:{sig:g}", "multiclass": "multiclass",
            "regression": "regression"}.get(booster.objective,
                                            booster.objective)
     lines = ["tree", "version=v3",

@@ -35,6 +35,27 @@ def auc(p: np.ndarray, y: np.ndarray) -> float:
     return float((sum_rank_pos - n_pos * (n_pos + 1) / 2) / (n_pos * n_neg))
 
 
+# Per-metric improvement direction (LightGBM's per-metric handling): the
+# direction must come from the metric actually compared, NOT from the
+# objective (an AUC-eval'd classifier is higher-better even though the
+# binary objective's loss is lower-better).
+_METRIC_HIGHER_BETTER = {
+    "auc": True, "ndcg": True, "map": True, "accuracy": True,
+    "average_precision": True, "auprc": True,
+    "binary_logloss": False, "multi_logloss": False, "cross_entropy": False,
+    "l1": False, "l2": False, "mae": False, "mse": False, "rmse": False,
+    "mape": False, "huber": False, "fair": False, "poisson": False,
+    "quantile": False, "tweedie": False, "gamma": False,
+}
+
+
+def metric_higher_is_better(name: str, default: bool = False) -> bool:
+    """Direction of a metric by name ('ndcg@5' → 'ndcg'); falls back to
+    `default` (the objective's direction) for unknown custom metrics."""
+    base = str(name).split("@")[0].strip().lower()
+    return _METRIC_HIGHER_BETTER.get(base, default)
+
+
 def default_metrics_fn(metric_name=None):
     def fn(booster, Xv, yv, wv, comm):
         out = {}

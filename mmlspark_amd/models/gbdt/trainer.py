@@ -802,18 +802,24 @@ def train_booster(X: torch.Tensor, y: torch.Tensor, cfg: TrainConfig,
             booster_now = session.booster()
             entry = {"iteration": it}
             score = None
+            score_name = None
             for vi, (Xv, yv, wv) in enumerate(valid_sets or []):
                 m = metrics_fn(booster_now, Xv, yv, wv, comm)
                 entry[f"valid_{vi}"] = m
                 if score is None and m:
-                    score = next(iter(m.values()))
+                    score_name, score = next(iter(m.items()))
             stats.evals.append(entry)
             stats.eval_s += time.perf_counter() - t0
             if score is not None and cfg.early_stopping_round > 0:
+                # direction keyed off the metric actually compared (per-metric
+                # table like LightGBM), not the objective's loss direction
+                from .metrics import metric_higher_is_better
+                higher_better = metric_higher_is_better(
+                    score_name, objective.higher_better_metric)
                 tol = cfg.improvement_tolerance  # improvementTolerance
                 better = (best_score is None
                           or (score > best_score + tol
-                              if objective.higher_better_metric
+                              if higher_better
                               else score < best_score - tol))
                 if better:
                     best_score, best_iter = score, it

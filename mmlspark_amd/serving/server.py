@@ -155,6 +155,13 @@ class ServingServer:
             self.epoch += 1
             try:
                 replies = self.handler([p.payload for p in batch])
+                if len(replies) != len(batch):
+                    # a short reply list would silently drop the unmatched
+                    # requests until reply_timeout; fail the epoch instead so
+                    # the at-least-once replay path below re-enqueues them
+                    raise RuntimeError(
+                        f"handler returned {len(replies)} replies for "
+                        f"{len(batch)} requests")
                 for pr, rep in zip(batch, replies):
                     pr.response = json.dumps(rep, default=_np_default).encode()
                     pr.event.set()
@@ -265,9 +272,14 @@ class LowLatencyGBDTScorer:
     def _raw(self):
         from ..ops import backend
         f = self.flat
+        # mirror Booster.predict_raw exactly: categorical bitset splits and
+        # the early-stopping tree range must survive the low-latency path
+        bi = getattr(self.booster, "best_iteration", -1)
         return backend.predict_forest(
             f["feature"], f["threshold"], f["left"], f["right"], f["value"],
-            f["offsets"], self.inp, self.booster.n_outputs, f["weights"])
+            f["offsets"], self.inp, self.booster.n_outputs, f["weights"],
+            num_iteration=(bi + 1 if bi >= 0 else -1),
+            cat_offset=f.get("cat_offset"), cat_words=f.get("cat_words"))
 
     def _capture(self):
         torch = self.torch

@@ -399,3 +399,44 @@ def test_warm_start_from_booster_object(binary_df):
     m2 = LightGBMClassifier(numIterations=5, numLeaves=7,
                             lightGBMBooster=m1.booster).fit(binary_df)
     assert m2.booster.num_trees == 10  # 5 warm + 5 new
+
+
+def test_metric_direction_table():
+    """direction keyed off the metric compared, not the objective."""
+    from mmlspark_amd.models.gbdt.metrics import metric_higher_is_better
+    assert metric_higher_is_better("auc")
+    assert metric_higher_is_better("ndcg@5")
+    assert metric_higher_is_better("map@10")
+    assert not metric_higher_is_better("binary_logloss")
+    assert not metric_higher_is_better("l2")
+    assert not metric_higher_is_better("rmse")
+    # unknown metric falls back to the provided default
+    assert metric_higher_is_better("my_custom", True)
+    assert not metric_higher_is_better("my_custom", False)
+
+
+def test_early_stopping_auc_tracks_best_not_worst(binary_df):
+    """With metric='auc' (higher-better) best_iteration must be the argmax
+    of AUC over the eval history — the round-1 bug tracked the argmin."""
+    df = binary_df.copy()
+    rng = np.random.default_rng(5)
+    df["isVal"] = rng.random(len(df)) < 0.3
+    m = LightGBMClassifier(numIterations=60, numLeaves=31, learningRate=0.4,
+                           metric="auc", validationIndicatorCol="isVal",
+                           earlyStoppingRound=5).fit(df)
+    evals = m._training_stats.evals
+    aucs = [e["valid_0"]["auc"] for e in evals]
+    bi = m.booster.best_iteration
+    assert bi >= 0
+    assert aucs[bi] == max(aucs), (bi, aucs)
+
+
+def test_lightgbm_text_export_preserves_sigmoid(binary_df):
+    m = LightGBMClassifier(numIterations=3, numLeaves=7).fit(binary_df)
+    b = m.booster
+    b.sigmoid = 2.5
+    txt = b.to_lightgbm_text()
+    assert "objective=binary sigmoid:2.5" in txt
+    # round trip through our own text importer keeps the sigmoid
+    b2 = Booster.load_from_string(txt)
+    assert abs(float(b2.sigmoid) - 2.5) < 1e-6

@@ -176,3 +176,45 @@ def test_distributed_serving_server():
             assert rr == {"worker": i, "y": i + 1}
     finally:
         srv.stop()
+
+
+def test_low_latency_scorer_categorical_and_best_iteration():
+    """the low-latency path must mirror Booster.predict_raw —
+    categorical bitset splits and the early-stopping tree range included."""
+    import torch
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    rng = np.random.default_rng(3)
+    n = 2000
+    cat = rng.integers(0, 10, size=n).astype(np.float32)
+    noise = rng.normal(size=(n, 3)).astype(np.float32)
+    y = np.isin(cat.astype(int), [1, 4, 7]).astype(np.float32)
+    X = np.column_stack([cat, noise]).astype(np.float32)
+    df = pd.DataFrame({"features": list(X), "label": y})
+    m = LightGBMClassifier(numIterations=8, numLeaves=15,
+                           categoricalSlotIndexes=[0],
+                           minDataInLeaf=5).fit(df)
+    b = m.booster
+    assert any((t.cat_offset >= 0).any() for t in b.trees)
+    b.best_iteration = 4  # force an early-stopping tree range
+    scorer = LowLatencyGBDTScorer(b, max_batch=8, use_graph=False)
+    got = scorer.score(X[:8])
+    ref = torch.sigmoid(b.sigmoid * b.predict_raw(
+        torch.from_numpy(X[:8]))).numpy()
+    np.testing.assert_allclose(got, ref, atol=1e-6)
+
+
+def test_micro_batch_reply_count_mismatch_fails_loudly():
+    """A handler returning too few replies must error the epoch (replay →
+    eventual 500), not silently drop requests into a 504."""
+    def short_handler(batch):
+        return [{"ok": 1}] * (len(batch) - 1) if len(batch) > 1 else []
+
+    srv = ServingServer(short_handler, port=0, mode="micro-batch",
+                        batch_wait_ms=5, reply_timeout=2.0).start()
+    try:
+        url = f"http://127.0.0.1:{srv.port}/"
+        r = requests.post(url, json={"x": 1}, timeout=10)
+        assert r.status_code == 500
+        assert "replies" in r.json().get("error", "")
+    finally:
+        srv.stop()
