@@ -42,6 +42,14 @@ class Comm:
     def is_distributed(self) -> bool:
         return self._initialized and self.world_size > 1
 
+    def native_group(self):
+        """The underlying c10d ProcessGroup object, for native (C++) code
+        that calls collectives GIL-free (e.g. the GBDT grower's per-split
+        histogram all_reduce).  None when not distributed."""
+        if not self.is_distributed:
+            return None
+        return self.group if self.group is not None else dist.group.WORLD
+
     # --- collectives ------------------------------------------------------
     def all_reduce(self, t: torch.Tensor, op: str = "sum") -> torch.Tensor:
         if self.is_distributed:

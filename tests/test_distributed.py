@@ -335,3 +335,47 @@ def _worker_gbdt_ws4(rank, world, port, q):
         torch.distributed.destroy_process_group()
     except Exception as e:  # pragma: no cover
         q.put((rank, f"ERROR: {e!r}"))
+
+
+def _worker_native_allreduce(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        torch.distributed.init_process_group(
+            "gloo", init_method=f"tcp://127.0.0.1:{port}",
+            rank=rank, world_size=world)
+        from mmlspark_amd.ops import _hip_grower
+        from mmlspark_amd.parallel.comm import Comm
+
+        comm = Comm()
+        pg = comm.native_group()
+        assert pg is not None
+        # int64 tensor — the histogram dtype the grower reduces
+        t = torch.arange(16, dtype=torch.int64) * (rank + 1)
+        _hip_grower.allreduce_native(pg, t)
+        q.put((rank, t.tolist()))
+        torch.distributed.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"ERROR: {e!r}"))
+
+
+@pytest.mark.timeout(120)
+def test_native_c10d_allreduce_binding():
+    """The C++ grower reduces histograms through the c10d ProcessGroup C++
+    API (no Python callback, no GIL).  Prove the exact binding it uses —
+    pg cast + allreduce + wait on an int64 tensor — at gloo world_size=2."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_native_allreduce,
+                         args=(r, 2, 29874, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, s = q.get(timeout=100)
+        results[rank] = s
+    for p in procs:
+        p.join(timeout=30)
+    assert not any(str(s).startswith("ERROR") for s in results.values()), results
+    expect = [int(i) * 3 for i in range(16)]  # (rank0 1x + rank1 2x)
+    assert results[0] == expect
+    assert results[1] == expect

@@ -292,9 +292,9 @@ def test_serving_gpu_hipgraph(binary_df):
 
 @requires_gpu
 def test_native_grower_distributed_codepath(binary_df):
-    """Exercise the multi-rank branches of the C++ grower on one GPU: the
-    histogram reduce callback (GIL hop) and the synced partition readback."""
-    import pandas as pd
+    """Exercise the multi-rank branches of the C++ grower on one GPU (pg=None
+    so the reduce is skipped but every distributed branch — device-side
+    partition counts, pinned nl readback, speculation — runs)."""
     from mmlspark_amd.models.gbdt.objectives import make_objective
     from mmlspark_amd.models.gbdt.trainer import (TrainConfig, TrainingSession)
     from mmlspark_amd.parallel.comm import Comm
@@ -307,18 +307,11 @@ def test_native_grower_distributed_codepath(binary_df):
     g, h = ses.objective.grad_hess(ses.preds, y, None)
     grower = ses.grower
     grower.set_scales(g[:, 0], h[:, 0])
-    calls = {"n": 0}
-
-    def fake_reduce(t):
-        calls["n"] += 1
-        return t
-
     d = _hip_grower.grow_tree_native(
         grower.binned, grower.binned_pair, ses.all_rows, g[:, 0].contiguous(),
         h[:, 0].contiguous(), cfg.max_bin, grower.nf, grower.scale_g,
         grower.scale_h, 0.0, 0.0, float(cfg.min_data_in_leaf), 1e-3, 0.0,
-        0.0, 15, -1, None, fake_reduce, True)
-    assert calls["n"] >= 15  # one reduce per histogram build
+        0.0, 15, -1, None, None, True)
     feature = d["feature"].numpy()
     assert (feature >= 0).sum() == 14  # 15 leaves → 14 internal nodes
     # leaf segments partition all rows exactly
