@@ -96,6 +96,10 @@ def infer_feature_names(df: pd.DataFrame, features_col: str,
                         feature_cols: Optional[Sequence[str]]) -> List[str]:
     if feature_cols:
         return list(feature_cols)
+    if len(df) and features_col in df.columns \
+            and isinstance(df[features_col].iloc[0], SparseVector):
+        d = df[features_col].iloc[0].size
+        return [f"{features_col}_{i}" for i in range(d)]
     if is_vector_column(df, features_col):
         d = len(np.asarray(df[features_col].iloc[0])) if len(df) else 0
         return [f"{features_col}_{i}" for i in range(d)]

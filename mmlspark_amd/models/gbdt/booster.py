@@ -71,6 +71,13 @@ class Booster:
         num_iteration overrides."""
         if num_iteration < 0 and getattr(self, "best_iteration", -1) >= 0:
             num_iteration = self.best_iteration + 1
+        if hasattr(X, "densify_chunks"):  # CsrMatrix: bounded-memory scoring
+            outs = [self.predict_raw(chunk, start_iteration, num_iteration)
+                    for _, chunk in X.densify_chunks()]
+            if not outs:
+                base = torch.from_numpy(self.base_score).to(X.device)
+                return base.expand(0, self.n_outputs).clone()
+            return torch.cat(outs, dim=0)
         X = X if isinstance(X, torch.Tensor) else torch.as_tensor(X, dtype=torch.float32)
         X = X.float()
         if not self.trees:
@@ -96,6 +103,9 @@ class Booster:
         return raw
 
     def predict_leaf(self, X: torch.Tensor) -> torch.Tensor:
+        if hasattr(X, "densify_chunks"):  # CsrMatrix
+            return torch.cat([self.predict_leaf(chunk)
+                              for _, chunk in X.densify_chunks()], dim=0)
         X = X.float()
         f = self._flat(X.device)
         return backend.predict_leaf(f["feature"], f["threshold"], f["left"],

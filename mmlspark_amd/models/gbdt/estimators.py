@@ -132,8 +132,9 @@ class _GBDTParams(Params):
     useSingleDatasetMode = Param("useSingleDatasetMode", "obsolete: one "
                                  "process per GPU owns its shard", False,
                                  toBool)
-    matrixType = Param("matrixType", "auto|dense|sparse (binned storage is "
-                       "always dense uint8)", "auto", toString)
+    matrixType = Param("matrixType", "auto|dense|sparse — sparse trains from "
+                       "binned CSR without densifying (auto samples the "
+                       "column like DatasetUtils.scala:49)", "auto", toString)
     device = Param("device", "cpu|cuda|auto", "auto", toString)
 
     def _train_config(self) -> TrainConfig:
@@ -189,15 +190,32 @@ class _GBDTEstimatorBase(_GBDTParams, Estimator):
         return make_objective(name, num_class=num_class,
                               fobj=self.get("fobj"))
 
+    def _use_sparse(self, df: pd.DataFrame) -> bool:
+        """matrixType='sparse' forces CSR; 'auto' samples the column like the
+        reference (DatasetUtils.scala:49 sampleRowsForArrayType)."""
+        from .sparse import looks_sparse
+        if self.get("featureCols"):
+            return False
+        fc = self.get("featuresCol")
+        mt = (self.get("matrixType") or "auto").lower()
+        if mt == "dense" or fc not in df.columns or not len(df):
+            return False
+        return mt == "sparse" or looks_sparse(df[fc])
+
     def _extract(self, df: pd.DataFrame, device):
-        X = features_matrix(df, self.get("featuresCol"), self.get("featureCols"))
         y = df[self.get("labelCol")].to_numpy(dtype=np.float32)
+        yt = torch.from_numpy(y).to(device)
         w = None
         if self.get("weightCol"):
             w = torch.from_numpy(
                 df[self.get("weightCol")].to_numpy(dtype=np.float32)).to(device)
+        if self._use_sparse(df):
+            from .sparse import CsrMatrix
+            Xs = CsrMatrix.from_sparse_vectors(
+                df[self.get("featuresCol")]).to(device)
+            return Xs, yt, w
+        X = features_matrix(df, self.get("featuresCol"), self.get("featureCols"))
         Xt = torch.from_numpy(np.ascontiguousarray(X)).to(device)
-        yt = torch.from_numpy(y).to(device)
         return Xt, yt, w
 
     def _fit(self, df: pd.DataFrame):
@@ -329,6 +347,11 @@ class _GBDTModelBase(Model):
         return self.booster.feature_importances(importance_type).tolist()
 
     def _X(self, df: pd.DataFrame, device):
+        from .sparse import CsrMatrix, looks_sparse
+        fc = self.get("featuresCol")
+        if (not self.get("featureCols") and fc in df.columns and len(df)
+                and looks_sparse(df[fc])):
+            return CsrMatrix.from_sparse_vectors(df[fc]).to(device)
         X = features_matrix(df, self.get("featuresCol"), self.get("featureCols"))
         return torch.from_numpy(np.ascontiguousarray(X)).to(device)
 

@@ -1,0 +1,261 @@
+"""Sparse CSR GBDT ingestion — train without densifying.
+
+The reference builds LightGBM datasets straight from CSR
+(`LGBM_DatasetCreateFromCSR`, lightgbm/.../dataset/DatasetAggregator.scala:442,
+sparse auto-detect DatasetUtils.scala:49).  MI355X-native equivalent:
+
+  * the shard lives in HBM as CSR of BINNED values — indptr int64,
+    col int32, bin uint8 (~5 B/nnz instead of nf bytes/row);
+  * per-leaf histograms accumulate only the stored entries (HIP kernel,
+    global fixed-point int64 atomics) and recover each feature's implicit
+    zeros by subtraction from exact integer leaf totals — LightGBM's
+    zero-bin optimization, in fixed point so the distributed all_reduce
+    stays bit-exact;
+  * leaf partition gathers the split feature's bin per row by binary
+    search in the row's column segment (missing → the feature's zero bin).
+
+Quantile boundaries come from a MIXTURE quantile over sampled rows: each
+feature's sorted nonzero sample values merged with its implied zeros, so a
+99%-sparse feature still gets sensible cuts around its nonzero mass.
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import numpy as np
+import torch
+
+from ...parallel.comm import Comm
+from .binning import BinMapper
+
+
+class CsrMatrix:
+    """Device CSR float matrix (rows compressed). The sparse analog of the
+    dense (n, nf) feature tensor accepted by TrainingSession."""
+
+    def __init__(self, indptr: torch.Tensor, col: torch.Tensor,
+                 val: torch.Tensor, shape):
+        assert indptr.dtype == torch.int64
+        self.indptr = indptr
+        self.col = col.to(torch.int32)
+        self.val = val.to(torch.float32)
+        self.shape = (int(shape[0]), int(shape[1]))
+
+    @property
+    def device(self):
+        return self.val.device
+
+    @property
+    def nnz(self) -> int:
+        return int(self.val.numel())
+
+    def to(self, device) -> "CsrMatrix":
+        return CsrMatrix(self.indptr.to(device), self.col.to(device),
+                         self.val.to(device), self.shape)
+
+    @staticmethod
+    def from_sparse_vectors(series, n_features: Optional[int] = None
+                            ) -> "CsrMatrix":
+        """Build from a pandas column of SparseVector (or scipy rows)."""
+        from ...core.schema import SparseVector
+        vals = series.to_numpy() if hasattr(series, "to_numpy") else series
+        n = len(vals)
+        nf = n_features
+        indices, values, lens = [], [], np.zeros(n, dtype=np.int64)
+        for i, v in enumerate(vals):
+            if isinstance(v, SparseVector):
+                idx, vv = v.indices, v.values
+                if nf is None:
+                    nf = v.size
+            elif hasattr(v, "indices") and hasattr(v, "data"):  # scipy row
+                idx, vv = v.indices, v.data
+                if nf is None:
+                    nf = v.shape[-1]
+            else:
+                arr = np.asarray(v, dtype=np.float32)
+                idx = np.nonzero(arr)[0].astype(np.int32)
+                vv = arr[idx]
+                if nf is None:
+                    nf = arr.shape[0]
+            idx = np.asarray(idx, dtype=np.int32)
+            vv = np.asarray(vv, dtype=np.float32)
+            if len(idx) > 1 and (np.diff(idx) < 0).any():
+                order = np.argsort(idx, kind="stable")  # kernels binary-search
+                idx, vv = idx[order], vv[order]
+            lens[i] = len(idx)
+            indices.append(idx)
+            values.append(vv)
+        indptr = np.zeros(n + 1, dtype=np.int64)
+        np.cumsum(lens, out=indptr[1:])
+        col = (np.concatenate(indices) if indices
+               else np.zeros(0, dtype=np.int32))
+        val = (np.concatenate(values) if values
+               else np.zeros(0, dtype=np.float32))
+        return CsrMatrix(torch.from_numpy(indptr), torch.from_numpy(col),
+                         torch.from_numpy(val), (n, nf or 0))
+
+    @staticmethod
+    def from_scipy(m) -> "CsrMatrix":
+        m = m.tocsr()
+        m.sort_indices()
+        return CsrMatrix(torch.from_numpy(m.indptr.astype(np.int64)),
+                         torch.from_numpy(m.indices.astype(np.int32)),
+                         torch.from_numpy(m.data.astype(np.float32)),
+                         m.shape)
+
+    def row_slice(self, start: int, end: int) -> "CsrMatrix":
+        a, b = int(self.indptr[start]), int(self.indptr[end])
+        return CsrMatrix(self.indptr[start:end + 1] - a, self.col[a:b],
+                         self.val[a:b], (end - start, self.shape[1]))
+
+    def densify(self, rows: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """Dense (m, nf) float32 for the selected rows (scoring fallback)."""
+        n, nf = self.shape
+        if rows is None:
+            rows = torch.arange(n, device=self.device)
+        rows = rows.long()
+        out = torch.zeros(rows.numel(), nf, dtype=torch.float32,
+                          device=self.device)
+        starts = self.indptr[rows]
+        counts = self.indptr[rows + 1] - starts
+        e = _expand_entries(starts, counts)
+        rpos = torch.repeat_interleave(
+            torch.arange(rows.numel(), device=self.device), counts)
+        out[rpos, self.col[e].long()] = self.val[e]
+        return out
+
+    def densify_chunks(self, chunk: int = 65536):
+        """Iterate (start, dense_chunk) — bounded-memory scoring."""
+        n = self.shape[0]
+        for s in range(0, max(n, 1), chunk):
+            e = min(n, s + chunk)
+            if s >= n:
+                break
+            idx = torch.arange(s, e, device=self.device)
+            yield s, self.densify(idx)
+
+
+def _expand_entries(starts: torch.Tensor, counts: torch.Tensor
+                    ) -> torch.Tensor:
+    """Entry indices for variable-length CSR segments: concat of
+    [starts[i], starts[i]+counts[i]) — the standard repeat_interleave
+    prefix trick, all on device."""
+    total = int(counts.sum())
+    if total == 0:
+        return torch.zeros(0, dtype=torch.int64, device=starts.device)
+    seg = torch.repeat_interleave(starts, counts)
+    off = torch.arange(total, device=starts.device)
+    bounds = torch.repeat_interleave(
+        torch.cumsum(counts, 0) - counts, counts)
+    return seg + (off - bounds)
+
+
+def fit_bin_mapper_csr(csr: CsrMatrix, n_bins: int = 255,
+                       sample_size: int = 200_000,
+                       comm: Optional[Comm] = None, seed: int = 0
+                       ) -> BinMapper:
+    """Quantile boundaries from a row sample WITHOUT densifying.
+
+    Per feature, the sampled column is [sorted negatives | zeros | sorted
+    positives]; a rank lookup into that virtual array gives exact mixture
+    quantiles.  Distributed: ranks exchange their sampled (col, val) sets
+    (all_gather_object — sample-sized, one-time) and compute identical
+    boundaries.
+    """
+    n, nf = csr.shape
+    gen = torch.Generator(device="cpu").manual_seed(seed)
+    k = min(n, max(1, sample_size // max(1, comm.world_size if comm else 1)))
+    rows = torch.randperm(n, generator=gen)[:k].to(csr.device).sort().values
+    starts = csr.indptr[rows.long()]
+    counts = csr.indptr[rows.long() + 1] - starts
+    e = _expand_entries(starts, counts)
+    cols = csr.col[e].long().cpu()
+    vals = torch.nan_to_num(csr.val[e], nan=0.0).cpu()
+    n_sample = int(k)
+    if comm is not None and comm.is_distributed:
+        parts = comm.all_gather_object(
+            (int(k), cols.numpy(), vals.numpy()))
+        n_sample = sum(p[0] for p in parts)
+        cols = torch.from_numpy(np.concatenate([p[1] for p in parts]))
+        vals = torch.from_numpy(np.concatenate([p[2] for p in parts]))
+    # sort by (feature, value): stable double argsort
+    order = torch.argsort(vals, stable=True)
+    order = order[torch.argsort(cols[order], stable=True)]
+    cols_s = cols[order]
+    vals_s = vals[order].double()
+    seg_count = torch.bincount(cols_s, minlength=nf)          # nnz per feature
+    seg_start = torch.cumsum(seg_count, 0) - seg_count
+    neg = torch.bincount(cols_s[vals_s < 0], minlength=nf)    # negatives per f
+    zeros = n_sample - seg_count                              # implied zeros
+    qs = torch.arange(1, n_bins, dtype=torch.float64) / n_bins
+    # virtual sorted column rank per (feature, quantile)
+    r = torch.clamp((qs.unsqueeze(0) * (n_sample - 1)).floor().long(),
+                    max=max(n_sample - 1, 0))                 # (nf, nb-1)
+    negf = neg.unsqueeze(1)
+    zf = zeros.unsqueeze(1)
+    in_neg = r < negf
+    in_zero = (~in_neg) & (r < negf + zf)
+    # index into vals_s for the non-zero regions
+    idx_neg = seg_start.unsqueeze(1) + r
+    idx_pos = seg_start.unsqueeze(1) + (r - zf)
+    idx = torch.where(in_neg, idx_neg, idx_pos)
+    idx = torch.clamp(idx, 0, max(int(vals_s.numel()) - 1, 0))
+    if vals_s.numel() == 0:
+        ub = torch.zeros(nf, n_bins - 1, dtype=torch.float64)
+    else:
+        ub = vals_s[idx]
+    ub = torch.where(in_zero, torch.zeros_like(ub), ub).float()
+    ub = torch.cummax(ub, dim=1).values
+    ub[:, -1] = float("inf")
+    return BinMapper(ub.contiguous().to(csr.device), n_bins)
+
+
+class SparseShard:
+    """Binned CSR shard: the sparse counterpart of the (ngroups, n, 4)
+    dense binned tensor."""
+
+    def __init__(self, indptr, col, binv, zero_bin, shape):
+        self.indptr = indptr
+        self.col = col
+        self.binv = binv          # uint8 (nnz,)
+        self.zero_bin = zero_bin  # int32 (nf,) — the bin holding value 0.0
+        self.shape = (int(shape[0]), int(shape[1]))
+
+    @property
+    def device(self):
+        return self.binv.device
+
+
+def bin_csr(csr: CsrMatrix, mapper: BinMapper, chunk: int = 4_000_000
+            ) -> SparseShard:
+    """Bin the stored entries; zero_bin[f] mirrors the dense rule
+    bin(x) = searchsorted(bounds_f, x) evaluated at x = 0."""
+    ub = mapper.upper_bounds.to(csr.device)
+    nb = mapper.n_bins
+    nnz = csr.nnz
+    binv = torch.empty(nnz, dtype=torch.uint8, device=csr.device)
+    for s in range(0, max(nnz, 1), chunk):
+        e = min(nnz, s + chunk)
+        if s >= nnz:
+            break
+        c = csr.col[s:e].long()
+        v = torch.nan_to_num(csr.val[s:e], nan=-float("inf"))
+        b = torch.searchsorted(ub[c].contiguous(), v.unsqueeze(1),
+                               right=False).squeeze(1)
+        binv[s:e] = b.clamp(max=nb - 1).to(torch.uint8)
+    zero = torch.searchsorted(
+        ub, torch.zeros(ub.shape[0], 1, device=csr.device),
+        right=False).squeeze(1).clamp(max=nb - 1).to(torch.int32)
+    return SparseShard(csr.indptr, csr.col, binv, zero, csr.shape)
+
+
+def looks_sparse(series, sample: int = 10) -> bool:
+    """Auto-detect: the reference samples 10 rows
+    (DatasetUtils.scala:49 sampleRowsForArrayType)."""
+    from ...core.schema import SparseVector
+    vals = series.to_numpy() if hasattr(series, "to_numpy") else series
+    for v in vals[:sample]:
+        if isinstance(v, SparseVector) or (hasattr(v, "indices")
+                                           and hasattr(v, "data")):
+            return True
+    return False

@@ -327,6 +327,28 @@ class TreeGrower:
     def _best_split(self, hist, feat_mask):
         return self._scan(hist.unsqueeze(0), feat_mask)[0]
 
+    def _partition(self, leaf) -> tuple:
+        """Split a leaf's row list by its committed split (stable order)."""
+        t0 = time.perf_counter()
+        if leaf.cats is not None:
+            gq, j = leaf.feat // 4, leaf.feat % 4
+            bins = self.binned[gq, leaf.rows.long(), j]
+            cats_t = torch.tensor(leaf.cats, dtype=bins.dtype,
+                                  device=bins.device)
+            mask = torch.isin(bins, cats_t)
+            rows_l, rows_r = leaf.rows[mask], leaf.rows[~mask]
+        else:
+            # counts above 2^24 lose exactness through the f32 scan
+            known = (int(leaf.CL) if (self.fixed
+                                      and not self.comm.is_distributed
+                                      and leaf.C < 1.6e7)
+                     else -1)
+            rows_l, rows_r = backend.partition_rows(
+                self.binned, leaf.rows, leaf.feat, leaf.bin,
+                known_left=known)
+        self.stats.partition_s += time.perf_counter() - t0
+        return rows_l, rows_r
+
     def _grow_native(self, rows_root, grad, hess, feat_mask):
         """Whole leaf-wise loop in the C++ driver (ops/hip/gbdt_grower.cpp)
         — one Python call per tree instead of ~6 per split."""
@@ -364,9 +386,12 @@ class TreeGrower:
             leaves.append(lf)
         return tree, leaves
 
+    native_ok = True  # sparse grower overrides (no arena path yet)
+
     def grow(self, rows_root: torch.Tensor, grad: torch.Tensor,
              hess: torch.Tensor, feat_mask) -> (Tree, List):
-        if (self.fixed and not self.cat_features and not self.voting
+        if (self.native_ok and self.fixed and not self.cat_features
+                and not self.voting
                 and not os.environ.get(
                     "MMLSPARK_AMD_NO_NATIVE_GROWER")):
             try:
@@ -424,24 +449,7 @@ class TreeGrower:
                 continue
             final_leaves.remove(leaf)
 
-            t0 = time.perf_counter()
-            if leaf.cats is not None:
-                gq, j = leaf.feat // 4, leaf.feat % 4
-                bins = self.binned[gq, leaf.rows.long(), j]
-                cats_t = torch.tensor(leaf.cats, dtype=bins.dtype,
-                                      device=bins.device)
-                mask = torch.isin(bins, cats_t)
-                rows_l, rows_r = leaf.rows[mask], leaf.rows[~mask]
-            else:
-                # counts above 2^24 lose exactness through the f32 scan
-                known = (int(leaf.CL) if (self.fixed
-                                          and not self.comm.is_distributed
-                                          and leaf.C < 1.6e7)
-                         else -1)
-                rows_l, rows_r = backend.partition_rows(
-                    self.binned, leaf.rows, leaf.feat, leaf.bin,
-                    known_left=known)
-            self.stats.partition_s += time.perf_counter() - t0
+            rows_l, rows_r = self._partition(leaf)
 
             GL, HL, CL = leaf.GL, leaf.HL, leaf.CL
             GR, HR, CR = leaf.G - GL, leaf.H - HL, leaf.C - CL
@@ -500,6 +508,87 @@ class TreeGrower:
                     gain_arr, leaf_idx, shrinkage=1.0,
                     cat_offset=cat_off, cat_words=cat_words)
         return tree, final_leaves
+
+
+class SparseTreeGrower(TreeGrower):
+    """Leaf-wise growth over a binned CSR shard (models/gbdt/sparse.py).
+
+    The histogram covers stored entries only; each feature's implicit zeros
+    are recovered by exact integer subtraction from leaf totals, so the
+    multi-rank all_reduce stays bit-exact.  Replaces the dense-only binned
+    matrix for high-dimensional sparse workloads (the reference's
+    LGBM_DatasetCreateFromCSR path, DatasetAggregator.scala:442)."""
+
+    native_ok = False  # python-loop grower; arena mode is dense-only
+
+    def __init__(self, shard, n_features: int, cfg: TrainConfig, comm: Comm,
+                 stats: TrainingStats, bin_mapper: BinMapper,
+                 n_global: Optional[int] = None):
+        self.shard = shard
+        self.binned = None
+        self.nf = n_features
+        self.nf_pad = n_features
+        self.cfg = cfg
+        self.comm = comm
+        self.stats = stats
+        self.bin_mapper = bin_mapper
+        self.device = shard.device
+        self.fixed = True  # int64 fixed point on CPU and GPU alike
+        self.n_global = n_global or shard.shape[0]
+        self.scale_g = 1.0
+        self.scale_h = 1.0
+        self.cat_features = []
+        self.cat_smooth = 10.0
+        self._gq = None
+        self._hq = None
+
+    def set_scales(self, grad: torch.Tensor, hess: torch.Tensor):
+        mx = torch.stack([grad.abs().max(), hess.max()])
+        self.comm.all_reduce(mx, op="max")
+        gmax = max(float(mx[0]), 1e-12)
+        hmax = max(float(mx[1]), 1e-12)
+        self.scale_g = (2.0 ** 61) / (max(self.n_global, 1) * gmax)
+        self.scale_h = (2.0 ** 24) / hmax
+        # quantize once per tree — every rank rounds identically
+        self._gq = torch.round(grad.double() * self.scale_g).to(torch.int64)
+        self._hq = torch.round(hess.double() * self.scale_h).to(torch.int64)
+
+    def _hist(self, rows, grad, hess, reduce=True):
+        t0 = time.perf_counter()
+        sh = self.shard
+        h = backend.csr_hist_fixed(sh.indptr, sh.col, sh.binv, self._gq,
+                                   self._hq, rows, self.nf, self.cfg.max_bin)
+        # implicit zeros: exact integer leaf totals − per-feature stored sums
+        r = rows.long()
+        tot = torch.stack([
+            self._gq[r].sum(), self._hq[r].sum(),
+            torch.tensor(int(rows.numel()), dtype=torch.int64,
+                         device=h.device)])
+        corr = tot.unsqueeze(0) - h.sum(dim=1)  # (nf, 3)
+        ar = torch.arange(self.nf, device=h.device)
+        h[ar, sh.zero_bin.long()] += corr
+        if self._sync_timers and self.device.type == "cuda":
+            torch.cuda.synchronize()
+        t1 = time.perf_counter()
+        self.stats.hist_s += t1 - t0
+        if reduce:
+            if self.voting:
+                self._voting_reduce(h)
+            else:
+                self.comm.all_reduce(h)
+            self.stats.comm_s += time.perf_counter() - t1
+        return h
+
+    def _partition(self, leaf) -> tuple:
+        t0 = time.perf_counter()
+        sh = self.shard
+        bins = backend.csr_gather_bins(
+            sh.indptr, sh.col, sh.binv, leaf.rows, leaf.feat,
+            int(sh.zero_bin[leaf.feat]))
+        mask = bins <= leaf.bin
+        rows_l, rows_r = leaf.rows[mask], leaf.rows[~mask]
+        self.stats.partition_s += time.perf_counter() - t0
+        return rows_l, rows_r
 
 
 def predict_tree_binned(tree: Tree, binned_i4: torch.Tensor,
@@ -581,8 +670,22 @@ class TrainingSession:
         if hasattr(objective, "group_sizes"):
             objective.group_sizes = group_sizes
 
+        from .sparse import CsrMatrix, bin_csr, fit_bin_mapper_csr
+        self.is_sparse = isinstance(X, CsrMatrix)
+        if self.is_sparse:
+            if cfg.categorical_features:
+                raise ValueError("categorical features need dense input "
+                                 "(set matrixType='dense')")
+            if cfg.boosting == "dart":
+                raise ValueError("dart boosting needs dense input "
+                                 "(set matrixType='dense')")
         if binned_cache is not None:
             self.bin_mapper, self.binned = binned_cache
+        elif self.is_sparse:
+            self.bin_mapper = fit_bin_mapper_csr(
+                X, n_bins=cfg.max_bin, comm=comm, seed=cfg.seed,
+                sample_size=cfg.bin_sample_count)
+            self.binned = bin_csr(X, self.bin_mapper)
         else:
             self.bin_mapper = BinMapper.fit(X, n_bins=cfg.max_bin, comm=comm,
                                             seed=cfg.seed,
@@ -613,7 +716,8 @@ class TrainingSession:
             t = torch.tensor([float(n)], device=device)
             comm.all_reduce(t)
             n_global = int(t[0])
-        self.grower = TreeGrower(self.binned, self.nf, cfg, comm, self.stats,
+        grower_cls = SparseTreeGrower if self.is_sparse else TreeGrower
+        self.grower = grower_cls(self.binned, self.nf, cfg, comm, self.stats,
                                  self.bin_mapper, n_global=n_global)
         self.gen = torch.Generator(device="cpu")
         self.all_rows = torch.arange(n, dtype=torch.int32, device=device)

@@ -152,3 +152,23 @@ def vw_predict(indices, values, offsets, weights_tbl):
         return _require_ext().vw_predict(indices, values, offsets, weights_tbl)
     from ..models.vw import sgd_ref
     return sgd_ref.vw_predict(indices, values, offsets, weights_tbl)
+
+
+# --------------------------------------------------------------- sparse CSR
+def csr_hist_fixed(indptr, col, binv, gq, hq, rows, nf, n_bins):
+    """Fixed-point histogram over STORED entries of `rows` from a binned CSR
+    shard; (nf, n_bins, 3) int64.  Implicit-zero correction is the caller's
+    (exact integer subtraction from leaf totals)."""
+    if binv.is_cuda:
+        return _require_ext().csr_hist_fixed(indptr, col, binv, gq, hq,
+                                             rows.contiguous(), nf, n_bins)
+    return cpu_ref.csr_hist_fixed(indptr, col, binv, gq, hq, rows, nf, n_bins)
+
+
+def csr_gather_bins(indptr, col, binv, rows, feature, zero_bin):
+    """Per-row bin of `feature` (missing → zero_bin) from a binned CSR shard."""
+    if binv.is_cuda:
+        return _require_ext().csr_gather_bins(indptr, col, binv,
+                                              rows.contiguous(), feature,
+                                              zero_bin).long()
+    return cpu_ref.csr_gather_bins(indptr, col, binv, rows, feature, zero_bin)

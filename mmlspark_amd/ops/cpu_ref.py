@@ -190,3 +190,47 @@ def split_scan(hists: torch.Tensor, n_bins: int, l1: float, l2: float,
         per_f[ar, bf], bf.float(), bb.float(),
         GL[ar, bf, bb], HL[ar, bf, bb], CL[ar, bf, bb]], dim=-1)
     return out
+
+
+# ------------------------------------------------------------------ sparse CSR
+def _expand_csr(indptr: torch.Tensor, rows: torch.Tensor):
+    """(entry_indices, row_position_per_entry) for the given rows."""
+    rows = rows.long()
+    starts = indptr[rows]
+    counts = indptr[rows + 1] - starts
+    total = int(counts.sum())
+    if total == 0:
+        z = torch.zeros(0, dtype=torch.int64, device=indptr.device)
+        return z, z
+    seg = torch.repeat_interleave(starts, counts)
+    off = torch.arange(total, device=indptr.device)
+    bounds = torch.repeat_interleave(torch.cumsum(counts, 0) - counts, counts)
+    rpos = torch.repeat_interleave(
+        torch.arange(rows.numel(), device=indptr.device), counts)
+    return seg + (off - bounds), rpos
+
+
+def csr_hist_fixed(indptr, col, binv, gq, hq, rows, nf, n_bins):
+    """Fixed-point histogram over stored CSR entries of `rows`
+    (implicit zeros are corrected by the caller from exact leaf totals).
+    Returns (nf, n_bins, 3) int64 — matches csr_hist_fixed_k."""
+    e, rpos = _expand_csr(indptr, rows)
+    hist = torch.zeros(nf * n_bins, 3, dtype=torch.int64, device=binv.device)
+    if e.numel():
+        r = rows.long()[rpos]
+        flat = col[e].long() * n_bins + binv[e].long()
+        src = torch.stack([gq[r], hq[r], torch.ones_like(gq[r])], dim=1)
+        hist.index_add_(0, flat, src)
+    return hist.view(nf, n_bins, 3)
+
+
+def csr_gather_bins(indptr, col, binv, rows, feature: int, zero_bin: int):
+    """Bin of `feature` for each row (missing → zero_bin); matches
+    csr_gather_bin_k."""
+    e, rpos = _expand_csr(indptr, rows)
+    out = torch.full((rows.numel(),), int(zero_bin), dtype=torch.int64,
+                     device=binv.device)
+    if e.numel():
+        m = col[e].long() == int(feature)
+        out[rpos[m]] = binv[e[m]].long()
+    return out

@@ -41,6 +41,12 @@ void launch_tree_shap(const int*, const float*, const int*, const int*,
                       const float*, const float*, const long*, const int*,
                       const unsigned*, const float*, long, int, int, int, int,
                       float*, hipStream_t);
+void launch_csr_hist_fixed(const long*, const int*,
+                           const unsigned char*, const long long*,
+                           const long long*, const int*, long, long long*,
+                           int, hipStream_t);
+void launch_csr_gather_bin(const long*, const int*, const unsigned char*,
+                           const int*, long, int, int, int*, hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -277,6 +283,50 @@ torch::Tensor tree_shap(torch::Tensor feat, torch::Tensor thr,
   return out;
 }
 
+// --------------------------------------------------------------- sparse CSR
+torch::Tensor csr_hist_fixed(torch::Tensor indptr, torch::Tensor col,
+                             torch::Tensor binv, torch::Tensor gq,
+                             torch::Tensor hq, torch::Tensor rows,
+                             long nf, long n_bins) {
+  CHECK_DEV(indptr); CHECK_CONTIG(indptr);
+  CHECK_DEV(col); CHECK_CONTIG(col);
+  CHECK_DEV(binv); CHECK_CONTIG(binv);
+  CHECK_DEV(gq); CHECK_CONTIG(gq);
+  CHECK_DEV(hq); CHECK_CONTIG(hq);
+  CHECK_DEV(rows); CHECK_CONTIG(rows);
+  TORCH_CHECK(indptr.dtype() == torch::kInt64, "indptr must be int64");
+  TORCH_CHECK(col.dtype() == torch::kInt32, "col must be int32");
+  TORCH_CHECK(binv.dtype() == torch::kUInt8, "binv must be uint8");
+  TORCH_CHECK(gq.dtype() == torch::kInt64 && hq.dtype() == torch::kInt64,
+              "gq/hq must be int64 fixed point");
+  auto hist = torch::zeros({nf, n_bins, 3},
+                           gq.options().dtype(torch::kInt64));
+  launch_csr_hist_fixed(indptr.data_ptr<long>(), col.data_ptr<int>(),
+                        binv.data_ptr<unsigned char>(),
+                        (const long long*)gq.data_ptr<int64_t>(),
+                        (const long long*)hq.data_ptr<int64_t>(),
+                        rows.data_ptr<int>(), rows.numel(),
+                        (long long*)hist.data_ptr<int64_t>(), (int)n_bins,
+                        cur_stream());
+  return hist;
+}
+
+torch::Tensor csr_gather_bins(torch::Tensor indptr, torch::Tensor col,
+                              torch::Tensor binv, torch::Tensor rows,
+                              long feature, long zero_bin) {
+  CHECK_DEV(indptr); CHECK_CONTIG(indptr);
+  CHECK_DEV(col); CHECK_CONTIG(col);
+  CHECK_DEV(binv); CHECK_CONTIG(binv);
+  CHECK_DEV(rows); CHECK_CONTIG(rows);
+  auto out = torch::empty({rows.numel()},
+                          rows.options().dtype(torch::kInt32));
+  launch_csr_gather_bin(indptr.data_ptr<long>(), col.data_ptr<int>(),
+                        binv.data_ptr<unsigned char>(), rows.data_ptr<int>(),
+                        rows.numel(), (int)feature, (int)zero_bin,
+                        out.data_ptr<int>(), cur_stream());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hist_build", &hist_build, "per-leaf (feature,bin) grad/hess/count histogram");
   m.def("hist_build_fixed_pair", &hist_build_fixed_pair,
@@ -292,4 +342,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("vw_sgd_minibatch", &vw_sgd_minibatch, "adaptive sparse SGD minibatch");
   m.def("vw_predict", &vw_predict, "sparse linear predict");
   m.def("tree_shap", &tree_shap, "path-dependent TreeSHAP contributions");
+  m.def("csr_hist_fixed", &csr_hist_fixed,
+        "fixed-point histogram over stored CSR entries");
+  m.def("csr_gather_bins", &csr_gather_bins,
+        "per-row bin of one feature from CSR (missing -> zero bin)");
 }

@@ -1372,3 +1372,84 @@ extern "C" void launch_tree_shap(const int* feat, const float* thr,
   else TSLAUNCH(32);
 #undef TSLAUNCH
 }
+
+// -------------------------------------------------------------- sparse CSR
+// Histogram over STORED entries only (the reference's CSR ingestion,
+// DatasetAggregator.scala:442); implicit zeros are corrected host-side by
+// subtraction from exact integer leaf totals (LightGBM zero-bin trick).
+// Fixed-point int64 global atomics keep the multi-rank all_reduce bit-exact.
+// Thread-per-row grid-stride; rows are sorted so indptr/grad reads coalesce.
+__global__ void csr_hist_fixed_k(const long* __restrict__ indptr,
+                                 const int* __restrict__ col,
+                                 const unsigned char* __restrict__ binv,
+                                 const long long* __restrict__ gq,
+                                 const long long* __restrict__ hq,
+                                 const int* __restrict__ rows, long m,
+                                 long long* __restrict__ hist, int n_bins) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < m;
+       i += (long)gridDim.x * blockDim.x) {
+    const int r = rows[i];
+    const unsigned long long g = (unsigned long long)gq[r];
+    const unsigned long long h = (unsigned long long)hq[r];
+    const long s = indptr[r], e = indptr[r + 1];
+    for (long j = s; j < e; ++j) {
+      unsigned long long* cell = (unsigned long long*)hist
+          + ((size_t)col[j] * n_bins + binv[j]) * 3;
+      atomicAdd(cell + 0, g);
+      atomicAdd(cell + 1, h);
+      atomicAdd(cell + 2, 1ull);
+    }
+  }
+}
+
+extern "C" void launch_csr_hist_fixed(const long* indptr, const int* col,
+                                      const unsigned char* binv,
+                                      const long long* gq,
+                                      const long long* hq, const int* rows,
+                                      long m, long long* hist, int n_bins,
+                                      hipStream_t stream) {
+  if (m == 0) return;
+  const int threads = 256;
+  const long want = (m + threads - 1) / threads;
+  const int blocks = (int)(want < 8192 ? want : 8192);  // ≫256 WGs, 8 XCDs
+  hipLaunchKernelGGL(csr_hist_fixed_k, dim3(blocks), dim3(threads), 0,
+                     stream, indptr, col, binv, gq, hq, rows, m, hist,
+                     n_bins);
+}
+
+// Bin of one feature per row: binary search the row's sorted column
+// segment; missing -> the feature's zero bin. Used for leaf partition.
+__global__ void csr_gather_bin_k(const long* __restrict__ indptr,
+                                 const int* __restrict__ col,
+                                 const unsigned char* __restrict__ binv,
+                                 const int* __restrict__ rows, long m,
+                                 int feature, int zero_bin,
+                                 int* __restrict__ out) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < m;
+       i += (long)gridDim.x * blockDim.x) {
+    const int r = rows[i];
+    long lo = indptr[r], hi = indptr[r + 1];
+    int b = zero_bin;
+    while (lo < hi) {
+      const long mid = (lo + hi) >> 1;
+      const int c = col[mid];
+      if (c == feature) { b = binv[mid]; break; }
+      if (c < feature) lo = mid + 1; else hi = mid;
+    }
+    out[i] = b;
+  }
+}
+
+extern "C" void launch_csr_gather_bin(const long* indptr, const int* col,
+                                      const unsigned char* binv,
+                                      const int* rows, long m, int feature,
+                                      int zero_bin, int* out,
+                                      hipStream_t stream) {
+  if (m == 0) return;
+  const int threads = 256;
+  const long want = (m + threads - 1) / threads;
+  const int blocks = (int)(want < 8192 ? want : 8192);
+  hipLaunchKernelGGL(csr_gather_bin_k, dim3(blocks), dim3(threads), 0,
+                     stream, indptr, col, binv, rows, m, feature, zero_bin,
+                     out);
+}
