@@ -282,37 +282,45 @@ class TreeGrower:
         return res
 
     def _cat_scan(self, histf, feat_mask):
-        """Sorted one-vs-rest categorical split over present categories."""
+        """Sorted one-vs-rest categorical split over present categories.
+
+        Deterministic: float32 numpy arithmetic with a lexsort tie-break by
+        bin id — the C++ arena grower (gbdt_grower.cpp cat_scan_host)
+        replicates this exactly so native and Python growers produce
+        bit-identical trees."""
         cfg = self.cfg
         best = None
         hcpu = histf.cpu()
         for f in self.cat_features:
             if feat_mask is not None and not bool(feat_mask[f]):
                 continue
-            h = hcpu[f]  # (nb, 3): g, h, count
-            present = torch.nonzero(h[:, 2] > 0).squeeze(-1)
-            if present.numel() < 2:
+            h = hcpu[f].numpy().astype(np.float32)  # (nb, 3): g, h, count
+            present = np.nonzero(h[:, 2] > 0)[0]
+            if len(present) < 2:
                 continue
             g = h[present, 0]
             hh = h[present, 1]
             c = h[present, 2]
-            order = torch.argsort(g / (hh + self.cat_smooth))
-            GL = g[order].cumsum(0)
-            HL = hh[order].cumsum(0)
-            CL = c[order].cumsum(0)
-            G, H, C = float(GL[-1]), float(HL[-1]), float(CL[-1])
+            ratio = (g / (hh + np.float32(self.cat_smooth))).astype(np.float32)
+            order = np.lexsort((present, ratio))  # stable: ratio, then bin id
+            GL = np.cumsum(g[order], dtype=np.float32)
+            HL = np.cumsum(hh[order], dtype=np.float32)
+            CL = np.cumsum(c[order], dtype=np.float32)
+            G, H, C = GL[-1], HL[-1], CL[-1]
 
             def sc(Gs, Hs):
-                Ga = (Gs.abs() - cfg.lambda_l1).clamp_min(0)
-                return Ga * Ga / (Hs + cfg.lambda_l2 + 1e-32)
+                Ga = np.maximum(np.abs(Gs) - np.float32(cfg.lambda_l1),
+                                np.float32(0))
+                return Ga * Ga / (Hs + np.float32(cfg.lambda_l2)
+                                  + np.float32(1e-32))
 
             gain = sc(GL, HL) + sc(G - GL, H - HL) - sc(
-                torch.tensor(G), torch.tensor(H))
+                np.float32(G), np.float32(H))
             valid = ((CL >= cfg.min_data_in_leaf)
                      & (C - CL >= cfg.min_data_in_leaf)
                      & (HL >= cfg.min_sum_hessian_in_leaf)
                      & (H - HL >= cfg.min_sum_hessian_in_leaf))
-            gain = torch.where(valid, gain, torch.full_like(gain, NEG_INF))
+            gain = np.where(valid, gain, np.float32(NEG_INF))
             gain[-1] = NEG_INF
             k = int(gain.argmax())
             bg = float(gain[k])
@@ -357,6 +365,19 @@ class TreeGrower:
         # the per-split histogram all_reduce runs INSIDE the C++ grower via
         # the c10d ProcessGroup C++ API — no GIL hop on the critical path
         pg = self.comm.native_group()
+        # categorical features: numeric scan masks them out; the grower's
+        # host-side cat scan handles them (filtered by this iteration's
+        # feature mask, mirroring _scan/_cat_scan)
+        cats_active = [f for f in self.cat_features
+                       if feat_mask is None or bool(feat_mask[f])]
+        num_mask = feat_mask
+        if self.cat_features:
+            num_mask = (feat_mask.clone() if feat_mask is not None else
+                        torch.ones(self.nf_pad, dtype=torch.bool,
+                                   device=self.device))
+            num_mask[self.cat_features] = False
+        cats_t = (torch.tensor(cats_active, dtype=torch.int64)
+                  if cats_active else None)
         t0 = time.perf_counter()
         d = _hip_grower.grow_tree_native(
             self.binned, self.binned_pair, rows_root.contiguous(), grad, hess,
@@ -364,19 +385,24 @@ class TreeGrower:
             self.nf, self.scale_g, self.scale_h, cfg.lambda_l1, cfg.lambda_l2,
             float(cfg.min_data_in_leaf), cfg.min_sum_hessian_in_leaf,
             cfg.min_gain_to_split, cfg.max_delta_step, cfg.num_leaves,
-            cfg.max_depth, feat_mask, pg,
+            cfg.max_depth, num_mask, cats_t, self.cat_smooth, pg,
             self.comm.is_distributed or bool(os.environ.get(
                 "MMLSPARK_AMD_FORCE_DIST_GROWER")))
         self.stats.hist_s += time.perf_counter() - t0
         feature = d["feature"].numpy()
         thr_bin = d["thr_bin"].numpy()
+        cat_offset = d["cat_offset"].numpy()
         threshold = np.array(
-            [self.bin_mapper.bin_upper_value(int(f), int(b)) if f >= 0 else 0.0
-             for f, b in zip(feature, thr_bin)], dtype=np.float32)
+            [float("nan") if cat_offset[i] >= 0 else
+             self.bin_mapper.bin_upper_value(int(f), int(b)) if f >= 0 else 0.0
+             for i, (f, b) in enumerate(zip(feature, thr_bin))],
+            dtype=np.float32)
         tree = Tree(feature, threshold, thr_bin, d["left"].numpy(),
                     d["right"].numpy(), d["value"].numpy(),
                     d["count"].numpy(), d["gain"].numpy(),
-                    d["leaf_index"].numpy(), shrinkage=1.0)
+                    d["leaf_index"].numpy(), shrinkage=1.0,
+                    cat_offset=cat_offset,
+                    cat_words=d["cat_words"].numpy().view(np.uint32))
         offs = d["leaf_offsets"].tolist()
         nodes = d["leaf_nodes"].tolist()
         leaves = []
@@ -390,8 +416,7 @@ class TreeGrower:
 
     def grow(self, rows_root: torch.Tensor, grad: torch.Tensor,
              hess: torch.Tensor, feat_mask) -> (Tree, List):
-        if (self.native_ok and self.fixed and not self.cat_features
-                and not self.voting
+        if (self.native_ok and self.fixed and not self.voting
                 and not os.environ.get(
                     "MMLSPARK_AMD_NO_NATIVE_GROWER")):
             try:

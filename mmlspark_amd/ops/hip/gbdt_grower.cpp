@@ -32,8 +32,10 @@
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 
+#include <algorithm>
 #include <cmath>
 #include <memory>
+#include <numeric>
 #include <queue>
 #include <vector>
 
@@ -43,7 +45,7 @@ void launch_arena_gather(const void*, long, const float*, const float*,
                          int*, hipStream_t);
 void launch_partition_arena(const void*, const void*, const int*, void*,
                             void*, int*, int*, long, int, long, long, int,
-                            int, int*, int*, hipStream_t);
+                            int, const unsigned*, int*, int*, hipStream_t);
 void launch_hist_pair_range(const void*, const void*, long, long, long,
                             long long*, int, int, int, const int*, int,
                             hipStream_t);
@@ -79,6 +81,12 @@ struct GrowCtx {
   c10::intrusive_ptr<c10d::ProcessGroup> pg;
   bool has_reduce;
   bool distributed;
+  // categorical features (already filtered by the iteration's feature mask);
+  // the host-side sorted one-vs-rest scan replicates trainer._cat_scan
+  // bit-for-bit (float32 arithmetic, lexsort tie-break by bin id)
+  std::vector<int> cat_feats;
+  double cat_smooth = 10.0;
+  torch::Tensor cat_idx_dev;  // i64 device indices for index_select
   torch::Tensor scratch;    // partition block counters (stream-ordered reuse)
   torch::Tensor total;      // partition left-count (stream-ordered reuse)
   torch::Tensor dst_idx;    // (n_arena,) i32 per-split dest ranks (reused)
@@ -88,6 +96,9 @@ struct SplitJob {
   torch::Tensor hist_l, hist_r;
   torch::Tensor scan_host;  // (2,6) f32 pinned
   torch::Tensor nl_host;    // (1,) i32 pinned — local left count readback
+  torch::Tensor cat_host;   // (nh, ncat, nb, 3) i64 pinned — cat hist rows
+  torch::Tensor bits_host;  // (8,) i32 pinned — partition bitset staging
+  torch::Tensor bits_dev;   // (8,) i32 device
   long nl_known = -1;       // host-known left count (single-rank fast path)
   hipEvent_t ev = nullptr;
 
@@ -107,6 +118,7 @@ struct LeafCand {
   double G = 0, H = 0, C = 0;
   double GL = 0, HL = 0, CL = 0;
   int feat = -1, bin = 0;
+  std::vector<int> cats;  // chosen category bins (empty = numeric split)
   std::shared_ptr<SplitJob> job;
 };
 
@@ -160,20 +172,140 @@ void launch_scan_async(GrowCtx& ctx, const torch::Tensor& hists_i64,
   (void)hipMemcpyAsync(job.scan_host.data_ptr<float>(), out.data_ptr<float>(),
                        nh * 6 * sizeof(float), hipMemcpyDeviceToHost,
                        grower_stream());
+  if (!ctx.cat_feats.empty()) {
+    // categorical rows of the reduced histogram ride back with the scan;
+    // the sorted one-vs-rest scan runs on host at commit (≤ ncat×256 bins)
+    const long ncat = (long)ctx.cat_feats.size();
+    auto cat_rows = hists_i64.index_select(1, ctx.cat_idx_dev).contiguous();
+    job.cat_host = torch::empty({nh, ncat, (long)ctx.n_bins, 3},
+                                torch::TensorOptions()
+                                    .dtype(torch::kInt64)
+                                    .pinned_memory(true));
+    (void)hipMemcpyAsync(job.cat_host.data_ptr<int64_t>(),
+                         cat_rows.data_ptr<int64_t>(),
+                         nh * ncat * ctx.n_bins * 3 * sizeof(int64_t),
+                         hipMemcpyDeviceToHost, grower_stream());
+  }
   (void)hipEventCreateWithFlags(&job.ev, hipEventDisableTiming);
   (void)hipEventRecord(job.ev, grower_stream());
+}
+
+struct CatBest {
+  float gain = -INFINITY;
+  int feat = -1;
+  float GL = 0, HL = 0, CL = 0;
+  std::vector<int> cats;
+};
+
+// Bit-for-bit replica of trainer._cat_scan: float32 arithmetic, sequential
+// float32 cumsums, lexsort tie-break by bin id, first-max argmax.
+CatBest cat_scan_host(const GrowCtx& ctx, const int64_t* hist) {
+  CatBest best;
+  const int nb = ctx.n_bins;
+  const double inv_g = 1.0 / ctx.scale_g, inv_h = 1.0 / ctx.scale_h;
+  for (size_t ci = 0; ci < ctx.cat_feats.size(); ++ci) {
+    const int64_t* h = hist + ci * (size_t)nb * 3;
+    std::vector<int> present;
+    for (int b = 0; b < nb; ++b)
+      if (h[b * 3 + 2] > 0) present.push_back(b);
+    const int m = (int)present.size();
+    if (m < 2) continue;
+    std::vector<float> g(m), hh(m), c(m), ratio(m);
+    for (int i = 0; i < m; ++i) {
+      const int b = present[i];
+      g[i] = (float)((double)h[b * 3 + 0] * inv_g);
+      hh[i] = (float)((double)h[b * 3 + 1] * inv_h);
+      c[i] = (float)h[b * 3 + 2];
+      ratio[i] = g[i] / (hh[i] + (float)ctx.cat_smooth);
+    }
+    std::vector<int> order(m);
+    std::iota(order.begin(), order.end(), 0);
+    std::stable_sort(order.begin(), order.end(), [&](int x, int y) {
+      if (ratio[x] != ratio[y]) return ratio[x] < ratio[y];
+      return present[x] < present[y];
+    });
+    std::vector<float> GL(m), HL(m), CL(m);
+    float ag = 0.f, ah = 0.f, ac = 0.f;
+    for (int i = 0; i < m; ++i) {
+      ag += g[order[i]]; ah += hh[order[i]]; ac += c[order[i]];
+      GL[i] = ag; HL[i] = ah; CL[i] = ac;
+    }
+    const float G = GL[m - 1], H = HL[m - 1], C = CL[m - 1];
+    auto sc = [&](float Gs, float Hs) {
+      const float Ga = std::max(std::fabs(Gs) - (float)ctx.l1, 0.0f);
+      return Ga * Ga / (Hs + (float)ctx.l2 + 1e-32f);
+    };
+    const float scGH = sc(G, H);
+    int bi = -1;
+    float bg = -INFINITY;
+    for (int i = 0; i < m - 1; ++i) {
+      const bool valid = CL[i] >= (float)ctx.min_data
+          && (C - CL[i]) >= (float)ctx.min_data
+          && HL[i] >= (float)ctx.min_hess
+          && (H - HL[i]) >= (float)ctx.min_hess;
+      const float gain = valid
+          ? sc(GL[i], HL[i]) + sc(G - GL[i], H - HL[i]) - scGH
+          : -INFINITY;
+      if (gain > bg) { bg = gain; bi = i; }
+    }
+    if (bi < 0 || !std::isfinite(bg)) continue;
+    if (best.feat < 0 || bg > best.gain) {
+      best.gain = bg;
+      best.feat = ctx.cat_feats[ci];
+      best.GL = GL[bi]; best.HL = HL[bi]; best.CL = CL[bi];
+      best.cats.assign(order.begin(), order.begin() + bi + 1);
+      for (auto& v : best.cats) v = present[v];
+    }
+  }
+  return best;
+}
+
+// Fill a candidate's split from the readbacks: numeric scan row `hi`, then
+// let the categorical best take over iff strictly better (trainer._scan).
+void resolve_best(const GrowCtx& ctx, SplitJob& job, int hi, LeafCand& c) {
+  auto a = job.scan_host.accessor<float, 2>();
+  c.gain = a[hi][0]; c.feat = (int)a[hi][1]; c.bin = (int)a[hi][2];
+  c.GL = a[hi][3]; c.HL = a[hi][4]; c.CL = a[hi][5];
+  c.cats.clear();
+  if (!ctx.cat_feats.empty()) {
+    const size_t stride =
+        ctx.cat_feats.size() * (size_t)ctx.n_bins * 3;
+    CatBest cb = cat_scan_host(
+        ctx, job.cat_host.data_ptr<int64_t>() + (size_t)hi * stride);
+    if (cb.feat >= 0 && (double)cb.gain > c.gain) {
+      c.gain = (double)cb.gain; c.feat = cb.feat; c.bin = 0;
+      c.GL = (double)cb.GL; c.HL = (double)cb.HL; c.CL = (double)cb.CL;
+      c.cats = std::move(cb.cats);
+    }
+  }
 }
 
 void launch_job(GrowCtx& ctx, LeafCand& leaf) {
   auto job = std::make_shared<SplitJob>();
   const long m = leaf.hi - leaf.lo;
   const int src = leaf.buf, dst = leaf.buf ^ 1;
+  const unsigned* cat_bits = nullptr;
+  if (!leaf.cats.empty()) {
+    // categorical split: stage the 256-bit category set and partition by
+    // bitset test instead of bin <= thr
+    job->bits_host = torch::zeros({8}, torch::TensorOptions()
+                                           .dtype(torch::kInt32)
+                                           .pinned_memory(true));
+    unsigned* wb = (unsigned*)job->bits_host.data_ptr<int>();
+    for (int b : leaf.cats) wb[b >> 5] |= (1u << (b & 31));
+    job->bits_dev = torch::empty({8}, ctx.rowid[0].options()
+                                          .dtype(torch::kInt32));
+    (void)hipMemcpyAsync(job->bits_dev.data_ptr<int>(), wb,
+                         8 * sizeof(int), hipMemcpyHostToDevice,
+                         grower_stream());
+    cat_bits = (const unsigned*)job->bits_dev.data_ptr<int>();
+  }
   launch_partition_arena(
       ctx.pair[src].data_ptr(), ctx.ghq[src].data_ptr(),
       ctx.rowid[src].data_ptr<int>(), ctx.pair[dst].data_ptr(),
       ctx.ghq[dst].data_ptr(), ctx.rowid[dst].data_ptr<int>(),
       ctx.dst_idx.data_ptr<int>() + leaf.lo, ctx.n_arena,
-      ctx.npairs, leaf.lo, m, leaf.feat, leaf.bin,
+      ctx.npairs, leaf.lo, m, leaf.feat, leaf.bin, cat_bits,
       ctx.scratch.data_ptr<int>(), ctx.total.data_ptr<int>(),
       grower_stream());
 
@@ -229,6 +361,8 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
                           double min_hess, double min_gain, double max_delta,
                           long num_leaves, long max_depth,
                           c10::optional<torch::Tensor> feat_mask,
+                          c10::optional<torch::Tensor> cat_feats,
+                          double cat_smooth,
                           py::object process_group, bool distributed) {
   GrowCtx ctx;
   const long n_full = binned_pair.size(1);
@@ -244,6 +378,14 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
   ctx.num_leaves = (int)num_leaves;
   ctx.max_depth = (int)max_depth;
   if (feat_mask.has_value()) ctx.feat_mask = *feat_mask;
+  if (cat_feats.has_value() && cat_feats->numel() > 0) {
+    auto cf = cat_feats->to(torch::kInt64).contiguous().cpu();
+    auto acc = cf.accessor<int64_t, 1>();
+    for (long i = 0; i < cf.numel(); ++i)
+      ctx.cat_feats.push_back((int)acc[i]);
+    ctx.cat_idx_dev = cf.to(binned.device());
+    ctx.cat_smooth = cat_smooth;
+  }
   if (!process_group.is_none()) {
     ctx.pg = process_group.cast<c10::intrusive_ptr<c10d::ProcessGroup>>();
   }
@@ -275,11 +417,12 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
                       grower_stream());
 
   std::vector<int> feature_, thr_bin_, left_, right_, leaf_idx_;
+  std::vector<int> cat_off_, cat_words_;
   std::vector<float> value_, count_, gain_;
   auto new_node = [&]() {
     feature_.push_back(-1); thr_bin_.push_back(0); left_.push_back(-1);
     right_.push_back(-1); value_.push_back(0.f); count_.push_back(0.f);
-    gain_.push_back(0.f); leaf_idx_.push_back(-1);
+    gain_.push_back(0.f); leaf_idx_.push_back(-1); cat_off_.push_back(-1);
     return (int)feature_.size() - 1;
   };
 
@@ -300,9 +443,7 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
     SplitJob j;
     launch_scan_async(ctx, root_hist.unsqueeze(0), j);
     (void)hipEventSynchronize(j.ev);
-    auto a = j.scan_host.accessor<float, 2>();
-    root->gain = a[0][0]; root->feat = (int)a[0][1]; root->bin = (int)a[0][2];
-    root->GL = a[0][3]; root->HL = a[0][4]; root->CL = a[0][5];
+    resolve_best(ctx, j, 0, *root);
   }
   count_[root->node_id] = (float)C0;
   value_[root->node_id] = (float)leaf_output(G0, H0, l1, l2, max_delta);
@@ -342,7 +483,6 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
 
     SplitJob& job = *leaf->job;
     (void)hipEventSynchronize(job.ev);
-    auto a = job.scan_host.accessor<float, 2>();
     const long m_parent = leaf->hi - leaf->lo;
     const long nl = job.nl_known >= 0
                         ? job.nl_known
@@ -354,6 +494,12 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
     feature_[nid] = leaf->feat;
     thr_bin_[nid] = leaf->bin;
     gain_[nid] = (float)leaf->gain;
+    if (!leaf->cats.empty()) {
+      unsigned words[8] = {0};
+      for (int b : leaf->cats) words[b >> 5] |= (1u << (b & 31));
+      cat_off_[nid] = (int)(cat_words_.size() / 8);
+      for (int w = 0; w < 8; ++w) cat_words_.push_back((int)words[w]);
+    }
     const int lid = new_node();
     const int rid = new_node();
     left_[nid] = lid;
@@ -373,10 +519,8 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
     lc->hist = job.hist_l; rc->hist = job.hist_r;
     lc->G = GL; lc->H = HL; lc->C = CL;
     rc->G = GR; rc->H = HR; rc->C = CR;
-    lc->gain = a[0][0]; lc->feat = (int)a[0][1]; lc->bin = (int)a[0][2];
-    lc->GL = a[0][3]; lc->HL = a[0][4]; lc->CL = a[0][5];
-    rc->gain = a[1][0]; rc->feat = (int)a[1][1]; rc->bin = (int)a[1][2];
-    rc->GL = a[1][3]; rc->HL = a[1][4]; rc->CL = a[1][5];
+    resolve_best(ctx, job, 0, *lc);
+    resolve_best(ctx, job, 1, *rc);
     lc->seq = seq++;
     rc->seq = seq++;
     leaf->job.reset();
@@ -423,6 +567,10 @@ py::dict grow_tree_native(torch::Tensor binned, torch::Tensor binned_pair,
   d["leaf_rows"] = leaf_rows;
   d["leaf_offsets"] = torch::tensor(offs, torch::kInt64);
   d["leaf_nodes"] = torch::tensor(seg_nodes, torch::kInt64);
+  d["cat_offset"] = torch::tensor(cat_off_, i32);
+  d["cat_words"] = cat_words_.empty()
+                       ? torch::zeros({0}, i32)
+                       : torch::tensor(cat_words_, i32);
   return d;
 }
 

@@ -612,14 +612,19 @@ extern "C" void launch_arena_gather(const void* pair_src, long n_src,
 
 __global__ void part_arena_count_k(const unsigned long long* __restrict__ pair,
                                    long n_arena, long lo, long m, int pf,
-                                   int jbyte, int thr, long chunk,
+                                   int jbyte, int thr,
+                                   const unsigned* __restrict__ cat_bits,
+                                   long chunk,
                                    int* __restrict__ block_counts) {
   const long start = (long)blockIdx.x * chunk;
   const long end = min(start + chunk, m);
   const unsigned long long* plane = pair + (size_t)pf * n_arena + lo;
   int cnt = 0;
-  for (long i = start + threadIdx.x; i < end; i += blockDim.x)
-    cnt += (int)(((plane[i] >> (8 * jbyte)) & 0xffull) <= (unsigned)thr);
+  for (long i = start + threadIdx.x; i < end; i += blockDim.x) {
+    const unsigned bv = (unsigned)((plane[i] >> (8 * jbyte)) & 0xffull);
+    cnt += (int)(cat_bits ? ((cat_bits[bv >> 5] >> (bv & 31)) & 1u)
+                          : (unsigned)(bv <= (unsigned)thr));
+  }
   __shared__ int sh[256];
   sh[threadIdx.x] = cnt;
   __syncthreads();
@@ -637,7 +642,8 @@ __global__ void part_arena_index_k(
     const longlong2* __restrict__ ghq_src, const int* __restrict__ rowid_src,
     longlong2* __restrict__ ghq_dst, int* __restrict__ rowid_dst,
     int* __restrict__ dst_idx, long n_arena, long lo, long m, int pf,
-    int jbyte, int thr, long chunk, const int* __restrict__ block_offsets,
+    int jbyte, int thr, const unsigned* __restrict__ cat_bits, long chunk,
+    const int* __restrict__ block_offsets,
     const int* __restrict__ total_left) {
   const long start = (long)blockIdx.x * chunk;
   const long end = min(start + chunk, m);
@@ -655,8 +661,11 @@ __global__ void part_arena_index_k(
   for (long i0 = start; i0 < end; i0 += blockDim.x) {
     const long i = i0 + threadIdx.x;
     bool valid = i < end, left = false;
-    if (valid)
-      left = ((split_plane[i] >> (8 * jbyte)) & 0xffull) <= (unsigned)thr;
+    if (valid) {
+      const unsigned bv = (unsigned)((split_plane[i] >> (8 * jbyte)) & 0xffull);
+      left = cat_bits ? (((cat_bits[bv >> 5] >> (bv & 31)) & 1u) != 0u)
+                      : (bv <= (unsigned)thr);
+    }
     const unsigned long long mask_l = __ballot(valid && left);
     const unsigned long long mask_r = __ballot(valid && !left);
     const unsigned long long lt = (1ull << lane) - 1ull;
@@ -710,7 +719,8 @@ extern "C" void launch_partition_arena(
     const void* pair_src, const void* ghq_src, const int* rowid_src,
     void* pair_dst, void* ghq_dst, int* rowid_dst, int* dst_idx,
     long n_arena, int npairs, long lo, long m, int feature, int thr,
-    int* scratch, int* total_left, hipStream_t stream) {
+    const unsigned* cat_bits, int* scratch, int* total_left,
+    hipStream_t stream) {
   if (m == 0) return;
   long chunk = 4096;
   long blocks = (m + chunk - 1) / chunk;
@@ -721,14 +731,14 @@ extern "C" void launch_partition_arena(
   const int pf = feature / 8, jbyte = feature % 8;
   hipLaunchKernelGGL(part_arena_count_k, dim3((unsigned)blocks), dim3(256),
                      0, stream, (const unsigned long long*)pair_src, n_arena,
-                     lo, m, pf, jbyte, thr, chunk, scratch);
+                     lo, m, pf, jbyte, thr, cat_bits, chunk, scratch);
   hipLaunchKernelGGL(part_scan_k, dim3(1), dim3(256), 0, stream, scratch,
                      (int)blocks, total_left);
   hipLaunchKernelGGL(part_arena_index_k, dim3((unsigned)blocks), dim3(256),
                      0, stream, (const unsigned long long*)pair_src,
                      (const longlong2*)ghq_src, rowid_src,
                      (longlong2*)ghq_dst, rowid_dst, dst_idx, n_arena, lo, m,
-                     pf, jbyte, thr, chunk, scratch, total_left);
+                     pf, jbyte, thr, cat_bits, chunk, scratch, total_left);
   long cchunk = 2048;
   long cblocks = (m + cchunk - 1) / cchunk;
   if (cblocks > 4096) {

@@ -311,7 +311,7 @@ def test_native_grower_distributed_codepath(binary_df):
         grower.binned, grower.binned_pair, ses.all_rows, g[:, 0].contiguous(),
         h[:, 0].contiguous(), cfg.max_bin, grower.nf, grower.scale_g,
         grower.scale_h, 0.0, 0.0, float(cfg.min_data_in_leaf), 1e-3, 0.0,
-        0.0, 15, -1, None, None, True)
+        0.0, 15, -1, None, None, 10.0, None, True)
     feature = d["feature"].numpy()
     assert (feature >= 0).sum() == 14  # 15 leaves → 14 internal nodes
     # leaf segments partition all rows exactly
@@ -437,3 +437,36 @@ def test_vw_bfgs_gpu(binary_df):
     y = binary_df["label"].to_numpy()
     acc = (m.transform(binary_df)["prediction"].to_numpy() == y).mean()
     assert acc > 0.9, acc
+
+
+@requires_gpu
+def test_categorical_native_grower_bit_identical():
+    """categorical one-vs-rest splits now run inside the
+    native arena grower (bitset partition + host cat scan) and must produce
+    BIT-IDENTICAL boosters to the Python grower on the same GPU data."""
+    import os
+    import pandas as pd
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    rng = np.random.default_rng(17)
+    n = 20_000
+    cat1 = rng.integers(0, 24, size=n).astype(np.float32)
+    cat2 = rng.integers(0, 6, size=n).astype(np.float32)
+    num = rng.normal(size=(n, 4)).astype(np.float32)
+    good = {2, 5, 9, 13, 20}
+    y = ((np.isin(cat1.astype(int), list(good))) ^ (num[:, 0] > 0.7)
+         ).astype(np.float32)
+    X = np.column_stack([cat1, num[:, :2], cat2, num[:, 2:]]).astype(np.float32)
+    df = pd.DataFrame({"features": list(X), "label": y})
+    kw = dict(numIterations=12, numLeaves=31, categoricalSlotIndexes=[0, 3],
+              minDataInLeaf=5, featureFraction=0.8, device="cuda")
+    m_native = LightGBMClassifier(**kw).fit(df)
+    os.environ["MMLSPARK_AMD_NO_NATIVE_GROWER"] = "1"
+    try:
+        m_py = LightGBMClassifier(**kw).fit(df)
+    finally:
+        del os.environ["MMLSPARK_AMD_NO_NATIVE_GROWER"]
+    s_native = m_native.booster.save_to_string()
+    s_py = m_py.booster.save_to_string()
+    assert any((t.cat_offset >= 0).any() for t in m_native.booster.trees), \
+        "native grower must actually take categorical splits"
+    assert s_native == s_py
