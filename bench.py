@@ -31,6 +31,12 @@ def main():
                     help="rows per GPU (weak scaling)")
     ap.add_argument("--features", type=int, default=100)
     ap.add_argument("--num-leaves", type=int, default=63)
+    ap.add_argument("--categorical", type=int, default=0,
+                    help="make the first K features categorical (ablation)")
+    ap.add_argument("--sparse", action="store_true",
+                    help="CSR ablation: train from sparse input")
+    ap.add_argument("--nnz", type=int, default=20,
+                    help="nonzeros per row in --sparse mode")
     args = ap.parse_args()
 
     sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
@@ -66,12 +72,33 @@ def main():
         noise = 0.5 * torch.randn(n, generator=g)
     logits = X @ w + 0.3 * torch.sin(3 * X[:, 0]) + 0.3 * X[:, 1] * X[:, 2]
     y = (logits + noise > 0).float()
+    cat_idx = None
+    if args.categorical > 0:
+        # ablation: first K features become 32-category ids whose effect on
+        # the label is non-monotone (forces one-vs-rest set splits)
+        k = min(args.categorical, nf)
+        cats = torch.randint(0, 32, (n, k), device=X.device).float()
+        parity = (cats.long() % 3 == 0).float().sum(dim=1)
+        y = ((logits + noise + 0.5 * parity) > 0.5).float()
+        X[:, :k] = cats
+        cat_idx = list(range(k))
     X = X.to(device)
     y = y.to(device)
+    if args.sparse:
+        # CSR ablation: keep --nnz random entries per row (same label task)
+        from mmlspark_amd.models.gbdt.sparse import CsrMatrix
+        keep = torch.argsort(torch.rand(n, nf, device=device),
+                             dim=1)[:, :args.nnz].sort(dim=1).values
+        val = torch.gather(X, 1, keep)
+        indptr = torch.arange(0, (n + 1) * args.nnz, args.nnz,
+                              dtype=torch.int64, device=device)
+        X = CsrMatrix(indptr, keep.to(torch.int32).reshape(-1),
+                      val.reshape(-1), (n, nf))
 
     cfg = TrainConfig(num_iterations=args.warmup + args.steps,
                       num_leaves=args.num_leaves, learning_rate=0.1,
-                      max_bin=255, min_data_in_leaf=20)
+                      max_bin=255, min_data_in_leaf=20,
+                      categorical_features=cat_idx)
     objective = make_objective("binary")
     session = TrainingSession(X, y, cfg, objective, comm)
 
@@ -124,6 +151,8 @@ def main():
                 "parallelism": f"dp{world}",
                 "sync": "RCCL histogram all_reduce over xGMI" if world > 1
                         else "single rank",
+                "categorical": args.categorical,
+                "matrix": (f"csr nnz={args.nnz}" if args.sparse else "dense"),
             },
         }), flush=True)
         stats = session.stats.as_dict()
