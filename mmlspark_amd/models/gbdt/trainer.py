@@ -238,9 +238,12 @@ class TreeGrower:
         """Stacked histograms → float32 real units for the split scan."""
         if not self.fixed:
             return hists
+        # float64 from the start: a float32 intermediate here would round
+        # 1/scale differently than the fixed-scan kernel's double inv_g and
+        # make native- vs Python-grown trees differ in final ulps
         inv = torch.tensor([1.0 / self.scale_g, 1.0 / self.scale_h, 1.0],
-                           device=hists.device)
-        return (hists.double() * inv.double()).float()
+                           dtype=torch.float64, device=hists.device)
+        return (hists.double() * inv).float()
 
     def _sums(self, hist: torch.Tensor):
         """(G, H, C) totals of one histogram (feature 0 owns every row)."""
