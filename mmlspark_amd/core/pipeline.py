@@ -44,6 +44,10 @@ class PipelineStage(Params):
 
 class Transformer(PipelineStage):
     def transform(self, df):
+        # pyarrow Tables / pyspark DataFrames accepted everywhere; the
+        # output comes back in the caller's kind (core/interop.py)
+        from .interop import coerce_input, restore_output
+        df, kind = coerce_input(df)
         t0 = time.time()
         try:
             out = self._transform(df)
@@ -51,7 +55,7 @@ class Transformer(PipelineStage):
             log_stage_event(self, "transform", error=repr(e))
             raise
         log_stage_event(self, "transform", ms=(time.time() - t0) * 1e3)
-        return out
+        return restore_output(out, kind)
 
     def _transform(self, df):
         raise NotImplementedError
@@ -62,6 +66,8 @@ class Transformer(PipelineStage):
 
 class Estimator(PipelineStage):
     def fit(self, df, params: Optional[dict] = None):
+        from .interop import coerce_input
+        df, _ = coerce_input(df)
         inst = self.copy(params) if params else self
         t0 = time.time()
         try:
