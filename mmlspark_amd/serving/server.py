@@ -62,6 +62,11 @@ class ServingServer:
         self.routing: Dict[str, _PendingRequest] = {}
         self.epoch = 0
         self.n_served = 0
+        # epoch-keyed in-flight requests (historyQueues analog,
+        # HTTPSourceV2.scala:488-517): filled when the batch loop takes an
+        # epoch, GC'd on commit, re-hydrated into a restarted worker
+        self.history: Dict[int, list] = {}
+        self.committed_epoch = -1
         self._stop = threading.Event()
         self._httpd: Optional[ThreadingHTTPServer] = None
         self._threads = []
@@ -153,6 +158,8 @@ class ServingServer:
                 except queue.Empty:
                     break
             self.epoch += 1
+            ep = self.epoch
+            self.history[ep] = batch  # registerPartition analog
             try:
                 replies = self.handler([p.payload for p in batch])
                 if len(replies) != len(batch):
@@ -165,6 +172,7 @@ class ServingServer:
                 for pr, rep in zip(batch, replies):
                     pr.response = json.dumps(rep, default=_np_default).encode()
                     pr.event.set()
+                self.commit(ep)  # fully replied → GC (HTTPSinkV2:129-136)
             except Exception as e:
                 # at-least-once: failed epoch re-enqueues unanswered requests
                 for pr in batch:
@@ -175,6 +183,44 @@ class ServingServer:
                             pr.code = 500
                             pr.response = json.dumps({"error": repr(e)}).encode()
                             pr.event.set()
+                self.commit(ep)  # re-enqueued requests join a later epoch
+
+    # --------------------------------------------------------- fault tolerance
+    def commit(self, epoch: int):
+        """Commit GC: drop history up to `epoch` (HTTPSourceV2.scala:557-575)."""
+        self.committed_epoch = max(self.committed_epoch, epoch)
+        for e in [e for e in self.history if e <= epoch]:
+            del self.history[e]
+
+    def kill(self):
+        """Simulate a worker crash: the listener and scoring loop die, but
+        queued/in-flight request state (queue, history, routing) survives
+        for re-hydration — the 'partition retries the same epoch' scenario."""
+        self._stop.set()
+        if self._httpd:
+            self._httpd.shutdown()
+            self._httpd.server_close()
+
+    def rehydrate_from(self, dead: "ServingServer") -> int:
+        """Adopt a crashed worker's unanswered requests: uncommitted epochs
+        first (they were mid-flight), then whatever still sat in its queue —
+        the registerPartition re-hydration path (HTTPSourceV2.scala:495-505)."""
+        moved = 0
+        for ep in sorted(dead.history):
+            for pr in dead.history[ep]:
+                if not pr.event.is_set():
+                    self.request_queue.put(pr)
+                    moved += 1
+        dead.history.clear()
+        while True:
+            try:
+                pr = dead.request_queue.get_nowait()
+            except queue.Empty:
+                break
+            if not pr.event.is_set():
+                self.request_queue.put(pr)
+                moved += 1
+        return moved
 
     # ------------------------------------------------------------- lifecycle
     def start(self):
@@ -323,35 +369,99 @@ class LowLatencyGBDTScorer:
 
 class DistributedServingServer:
     """Multi-worker serving (the HTTPSourceV2 distributed shape): one
-    ServingServer per worker plus a head endpoint that serves the aggregated
-    ServiceInfo list for load-balancer discovery (DriverServiceUtils /
+    ServingServer per worker plus a head endpoint (DriverServiceUtils /
     HTTPSourceStateHolder parity, HTTPSourceV2.scala:133-198,337).
+
+    Two head roles:
+      * discovery (always): GET /__service_info aggregates worker infos
+        for an external load balancer;
+      * proxy (proxy=True): the head forwards each request to a worker
+        round-robin and FAILS OVER to the next worker when one is down —
+        the load-balancer-with-retry pattern the reference assumes in
+        front of its WorkerServers.  Combined with kill_worker /
+        restart_worker epoch re-hydration, a worker crash mid-flight
+        drops no replies.
     Workers here are threads in one process (one per GPU rank in a real
-    deployment); requests go straight to workers, the head only discovers."""
+    deployment)."""
 
     def __init__(self, handler_factory, n_workers: int = 2,
                  host: str = "127.0.0.1", base_port: int = 0,
-                 mode: str = "continuous", name: str = "mmlspark-serving"):
+                 mode: str = "continuous", name: str = "mmlspark-serving",
+                 proxy: bool = False, reply_timeout: float = 30.0):
+        self.handler_factory = handler_factory
+        self.mode = mode
+        self.reply_timeout = reply_timeout
         self.workers = [
             ServingServer(handler_factory(i), host=host,
                           port=(base_port + i if base_port else 0),
-                          mode=mode, name=f"{name}-{i}")
+                          mode=mode, name=f"{name}-{i}",
+                          reply_timeout=reply_timeout)
             for i in range(n_workers)]
         self.head: Optional[ServingServer] = None
         self.name = name
         self.host = host
+        self.proxy = proxy
+        self._rr = 0
+        self._rr_lock = threading.Lock()
 
     def start(self):
         for w in self.workers:
             w.start()
 
-        def head_handler(payloads):
-            return [self.service_info() for _ in payloads]
+        if self.proxy:
+            import requests as _rq
+            session = _rq.Session()
+
+            def head_handler(payloads):
+                out = []
+                for p in payloads:
+                    with self._rr_lock:
+                        start = self._rr
+                        self._rr += 1
+                    last_err = None
+                    for k in range(len(self.workers) * 2):
+                        w = self.workers[(start + k) % len(self.workers)]
+                        if w._stop.is_set():
+                            continue  # known-dead: skip without a timeout
+                        try:
+                            r = session.post(
+                                f"http://{w.host}:{w.port}/", json=p,
+                                timeout=self.reply_timeout)
+                            if r.status_code == 200:
+                                out.append(r.json())
+                                break
+                            last_err = f"worker {w.name}: {r.status_code}"
+                        except Exception as e:  # connection refused/reset
+                            last_err = repr(e)
+                    else:
+                        raise RuntimeError(f"all workers failed: {last_err}")
+                return out
+        else:
+            def head_handler(payloads):
+                return [self.service_info() for _ in payloads]
 
         self.head = ServingServer(head_handler, host=self.host, port=0,
                                   mode="continuous",
                                   name=f"{self.name}-head").start()
         return self
+
+    # --------------------------------------------------------- fault injection
+    def kill_worker(self, i: int):
+        """Crash worker i (listener + scoring loop die; request state kept)."""
+        self.workers[i].kill()
+
+    def restart_worker(self, i: int) -> int:
+        """Start a fresh worker in slot i and re-hydrate the dead worker's
+        unanswered requests into it (registerPartition re-hydration,
+        HTTPSourceV2.scala:495-505).  Returns requests re-hydrated."""
+        dead = self.workers[i]
+        fresh = ServingServer(self.handler_factory(i), host=self.host, port=0,
+                              mode=self.mode, name=dead.name,
+                              reply_timeout=self.reply_timeout)
+        fresh.start()
+        moved = fresh.rehydrate_from(dead)
+        self.workers[i] = fresh
+        return moved
 
     def service_info(self):
         return {"name": self.name,

@@ -1,4 +1,5 @@
 """Serving server (continuous + micro-batch) and HTTP client stack."""
+import json
 import threading
 import time
 
@@ -7,7 +8,8 @@ import pandas as pd
 import pytest
 import requests
 
-from mmlspark_amd.serving.server import (LowLatencyGBDTScorer, ServingServer,
+from mmlspark_amd.serving.server import (DistributedServingServer,
+                                         LowLatencyGBDTScorer, ServingServer,
                                          TransformerHandler)
 
 
@@ -218,3 +220,157 @@ def test_micro_batch_reply_count_mismatch_fails_loudly():
         assert "replies" in r.json().get("error", "")
     finally:
         srv.stop()
+
+
+def test_commit_gc_bounds_history():
+    """Epoch history must be GC'd on commit (HTTPSourceV2.scala:557-575) —
+    a long-running server cannot accumulate replied epochs."""
+    srv = ServingServer(lambda batch: [{"ok": 1}] * len(batch), port=0,
+                        mode="micro-batch", batch_wait_ms=1).start()
+    try:
+        url = f"http://127.0.0.1:{srv.port}/"
+        s = requests.Session()
+        for i in range(40):
+            assert s.post(url, json={"i": i}, timeout=10).status_code == 200
+        assert srv.epoch >= 1
+        assert srv.committed_epoch >= 1
+        assert len(srv.history) == 0, srv.history
+    finally:
+        srv.stop()
+
+
+def test_worker_crash_rehydration_no_dropped_replies():
+    """Kill a micro-batch worker mid-flight; a restarted worker re-hydrates
+    its uncommitted epochs + queue and every client still gets its reply
+    (registerPartition re-hydration, HTTPSourceV2.scala:488-505)."""
+    import threading
+    hang = threading.Event()
+    entered = threading.Event()
+
+    def hanging_handler(batch):
+        entered.set()
+        hang.wait(timeout=60)  # simulates a crashed/stuck scoring process
+        raise RuntimeError("worker died")
+
+    dead = ServingServer(hanging_handler, port=0, mode="micro-batch",
+                         batch_wait_ms=1, reply_timeout=30.0).start()
+    url = f"http://127.0.0.1:{dead.port}/"
+    results = {}
+
+    def client(i):
+        try:
+            r = requests.post(url, json={"i": i}, timeout=30)
+            results[i] = (r.status_code, r.json())
+        except Exception as e:  # pragma: no cover
+            results[i] = ("ERR", repr(e))
+
+    threads = [threading.Thread(target=client, args=(i,)) for i in range(6)]
+    for t in threads:
+        t.start()
+    assert entered.wait(timeout=10)  # the batch loop took an epoch
+    time.sleep(0.2)                  # let the rest land in queue/history
+    dead.kill()                      # crash: listener + loop die, state kept
+
+    fresh = ServingServer(lambda batch: [{"ok": p["i"]} for p in batch],
+                          port=0, mode="micro-batch", batch_wait_ms=1).start()
+    moved = fresh.rehydrate_from(dead)
+    assert moved == 6, moved
+    for t in threads:
+        t.join(timeout=30)
+    hang.set()
+    fresh.stop()
+    assert len(results) == 6
+    for i, (code, body) in results.items():
+        assert code == 200, results
+        assert body == {"ok": i}
+
+
+def test_head_proxy_failover_no_dropped_replies():
+    """Proxy head fails over to live workers when one is killed mid-run —
+    the load-balancer pattern in front of WorkerServers; no request drops."""
+    import threading
+
+    def factory(i):
+        return lambda batch: [{"worker": i, "x": p["x"]} for p in batch]
+
+    dist = DistributedServingServer(factory, n_workers=2, mode="continuous",
+                                    proxy=True, reply_timeout=10).start()
+    try:
+        url = f"http://127.0.0.1:{dist.head.port}/"
+        s = requests.Session()
+        # both workers take traffic round-robin
+        seen = {s.post(url, json={"x": k}, timeout=10).json()["worker"]
+                for k in range(8)}
+        assert seen == {0, 1}
+
+        results = []
+        lock = threading.Lock()
+
+        def client(k):
+            r = requests.post(url, json={"x": k}, timeout=15)
+            with lock:
+                results.append((k, r.status_code, r.json()))
+
+        threads = [threading.Thread(target=client, args=(k,))
+                   for k in range(24)]
+        for j, t in enumerate(threads):
+            t.start()
+            if j == 8:
+                dist.kill_worker(0)  # mid-flight crash
+        for t in threads:
+            t.join(timeout=30)
+        assert len(results) == 24
+        for k, code, body in results:
+            assert code == 200, (k, code, body)
+            assert body["x"] == k
+        # after the kill everything lands on worker 1
+        assert s.post(url, json={"x": 99}, timeout=10).json()["worker"] == 1
+    finally:
+        dist.stop()
+
+
+def test_sixty_four_concurrent_clients_smoke(binary_df):
+    """64 concurrent keep-alive clients against micro-batch scoring (CPU
+    smoke; the measured req/s + p99 number comes from bench_serving.py
+    --clients 64 on the GPU box, committed under profiles/)."""
+    import http.client
+    import socket
+    import threading
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    model = LightGBMClassifier(numIterations=5, numLeaves=7).fit(binary_df)
+    scorer = LowLatencyGBDTScorer(model.booster, max_batch=64,
+                                  use_graph=False)
+
+    def handler(payloads):
+        X = np.asarray([p["features"] for p in payloads], dtype=np.float32)
+        return [{"score": s.tolist()} for s in scorer.score(X)]
+
+    srv = ServingServer(handler, port=0, mode="micro-batch", max_batch=64,
+                        batch_wait_ms=0.5).start()
+    errs = []
+    x = binary_df["features"].iloc[0].tolist()
+
+    def client(ci):
+        try:
+            conn = http.client.HTTPConnection("127.0.0.1", srv.port)
+            conn.connect()
+            conn.sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+            body = json.dumps({"features": x})
+            for _ in range(5):
+                conn.request("POST", "/", body=body,
+                             headers={"Content-Type": "application/json"})
+                r = conn.getresponse()
+                r.read()
+                if r.status != 200:
+                    errs.append((ci, r.status))
+            conn.close()
+        except Exception as e:  # pragma: no cover
+            errs.append((ci, repr(e)))
+
+    threads = [threading.Thread(target=client, args=(i,)) for i in range(64)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=60)
+    srv.stop()
+    assert not errs, errs[:5]
