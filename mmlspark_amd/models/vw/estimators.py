@@ -31,30 +31,36 @@ _LOSS_BY_NAME = {"squared": "squared", "logistic": "logistic", "hinge": "hinge"}
 
 
 def _extract_csr(df: pd.DataFrame, col: str, extra_cols=None, device="cpu"):
-    """SparseVector / dense-vector column(s) -> CSR tensors on device."""
+    """SparseVector / dense-vector column(s) -> CSR tensors on device.
+
+    Columns are combined VW-style: the primary column keeps its indices
+    (default namespace), every additional column's indices are offset by a
+    murmur3 seed of its name (namespace seeding,
+    VowpalWabbitMurmurWithPrefix semantics) so that columns of equal size
+    never alias onto the same table slots after the 2^b mask."""
+    from .murmur import hash_string
     cols = [col] + list(extra_cols or [])
     idx_parts, val_parts, counts = [], [], np.zeros(len(df), dtype=np.int64)
-    offset_base = 0
-    sizes = []
-    for c in cols:
+    total_dims = 0
+    for ci, c in enumerate(cols):
+        seed = 0 if ci == 0 else (hash_string(c) & 0x3FFFFFFF)
         vals = df[c].to_numpy()
         if len(vals) and isinstance(vals[0], SparseVector):
-            sizes.append(vals[0].size)
+            total_dims += vals[0].size
         else:
-            sizes.append(len(np.asarray(vals[0])) if len(vals) else 0)
-    for ci, c in enumerate(cols):
-        vals = df[c].to_numpy()
+            total_dims += len(np.asarray(vals[0])) if len(vals) else 0
         for i, v in enumerate(vals):
             if isinstance(v, SparseVector):
-                idx_parts.append((i, v.indices + offset_base, v.values))
+                vi = (v.indices.astype(np.int64) + seed) & 0x3FFFFFFF
+                idx_parts.append((i, vi.astype(np.int32), v.values))
                 counts[i] += len(v.indices)
             else:
                 dense = np.asarray(v, dtype=np.float32)
                 nz = np.nonzero(dense)[0]
-                idx_parts.append((i, nz.astype(np.int32) + offset_base,
-                                  dense[nz]))
+                vi = (nz.astype(np.int64) + seed) & 0x3FFFFFFF
+                idx_parts.append((i, vi.astype(np.int32), dense[nz]))
                 counts[i] += len(nz)
-        offset_base += sizes[ci]
+    offset_base = total_dims
     # assemble in row order
     idx_parts.sort(key=lambda t: t[0])
     if idx_parts:
@@ -69,6 +75,33 @@ def _extract_csr(df: pd.DataFrame, col: str, extra_cols=None, device="cpu"):
             torch.from_numpy(values.astype(np.float32)).to(device),
             torch.from_numpy(offsets).to(device),
             offset_base)
+
+
+def _filter_csr(idx, val, off, mask):
+    """Row-filter a CSR triple (device-side, no host loop)."""
+    counts = off[1:] - off[:-1]
+    keep = counts[mask]
+    new_off = torch.zeros(int(mask.sum()) + 1, dtype=off.dtype,
+                          device=off.device)
+    new_off[1:] = keep.cumsum(0)
+    total = int(new_off[-1])
+    if total == 0:
+        return idx[:0], val[:0], new_off
+    starts = off[:-1][mask]
+    seg = torch.repeat_interleave(starts, keep)
+    o = torch.arange(total, device=off.device)
+    e = seg + (o - torch.repeat_interleave(new_off[:-1], keep))
+    return idx[e], val[e], new_off
+
+
+def _avg_loss(pred: torch.Tensor, y: torch.Tensor, loss: str) -> float:
+    if loss == "logistic":
+        ll = torch.log1p(torch.exp(-y * pred))
+    elif loss == "hinge":
+        ll = (1 - y * pred).clamp_min(0)
+    else:  # squared
+        ll = (pred - y) ** 2
+    return float(ll.mean())
 
 
 class _VWParams(Params):
@@ -118,6 +151,15 @@ class _VWParams(Params):
                                     toBool)
     bfgs = Param("bfgs", "second-order full-batch L-BFGS optimization "
                  "(--bfgs) instead of online SGD", False, toBool)
+    holdoutOff = Param("holdoutOff", "--holdout_off: train multi-pass on "
+                       "every example (no holdout early termination)",
+                       False, toBool)
+    holdoutPeriod = Param("holdoutPeriod", "--holdout_period: every k-th "
+                          "example is held out when passes>1 (VW default 10)",
+                          10, toInt)
+    earlyTerminate = Param("earlyTerminate", "--early_terminate: stop after "
+                           "this many passes without holdout improvement",
+                           3, toInt)
     maxIterBfgs = Param("maxIterBfgs", "L-BFGS iteration budget", 100, toInt)
     device = Param("device", "cpu|cuda|auto", "auto", toString)
 
@@ -128,7 +170,8 @@ class _VWParams(Params):
         ig = self.get("ignoreNamespaces") or []
         if ig:
             extra = [c for c in extra if c[:1] not in ig]
-        specs = self.get("interactions") or []
+        specs = list(self.get("interactions") or [])
+        specs += self._resolve_quadratic(df)  # raw-input -q crossing
         from .featurizer import VowpalWabbitInteractions
         for k, spec in enumerate(specs):
             cols = (list(spec) if isinstance(spec, (list, tuple))
@@ -149,9 +192,13 @@ class _VWParams(Params):
         mapping = {"--l1": "l1", "--l2": "l2", "--learning_rate": "learningRate",
                    "--power_t": "powerT", "-b": "numBits", "--bit_precision":
                    "numBits", "--passes": "numPasses",
-                   "--loss_function": "lossFunction"}
+                   "--loss_function": "lossFunction",
+                   "--holdout_period": "holdoutPeriod",
+                   "--early_terminate": "earlyTerminate"}
         flags = {"--adaptive": "adaptive", "--normalized": "normalized",
-                 "--invariant": "invariant", "--bfgs": "bfgs"}
+                 "--invariant": "invariant", "--bfgs": "bfgs",
+                 "--holdout_off": "holdoutOff"}
+        self._quadratic_specs = []
         while i < len(s):
             if s[i] in mapping and i + 1 < len(s):
                 self.set(mapping[s[i]], s[i + 1])
@@ -159,8 +206,34 @@ class _VWParams(Params):
             elif s[i] in flags:
                 self.set(flags[s[i]], True)
                 i += 1
+            elif s[i] in ("-q", "--quadratic") and i + 1 < len(s):
+                # raw-input namespace crossing (VowpalWabbitBase arg
+                # surface): 'ab' crosses namespaces a×b; ':' is a wildcard
+                self._quadratic_specs.append(s[i + 1])
+                i += 2
             else:
                 i += 1
+
+    def _resolve_quadratic(self, df: pd.DataFrame) -> list:
+        """-q namespace pairs → concrete column pairs.  A namespace is a
+        column's first letter (VowpalWabbitFeaturizer semantics); ':' means
+        every namespace (so '-q ::' crosses all feature-column pairs)."""
+        specs = getattr(self, "_quadratic_specs", None) or []
+        if not specs:
+            return []
+        cols = [self.get("featuresCol")] + list(
+            self.get("additionalFeatures") or [])
+        cols = [c for c in cols if c in df.columns]
+        out = []
+        for spec in specs:
+            a, b = (spec + "::")[:2]
+            ca = cols if a == ":" else [c for c in cols if c[:1] == a]
+            cb = cols if b == ":" else [c for c in cols if c[:1] == b]
+            for x in ca:
+                for y in cb:
+                    if (x, y) not in out and (y, x) not in out:
+                        out.append((x, y))
+        return [list(p) for p in out]
 
 
 class _VWBase(_VWParams, Estimator):
@@ -215,12 +288,22 @@ class _VWBase(_VWParams, Estimator):
 
     def _fit(self, df: pd.DataFrame):
         self._parse_args()
+        # materialize -q namespace pairs into the interactions param so the
+        # SAME crossing applies at fit and at model transform time
+        q = self._resolve_quadratic(df)
+        if q:
+            self.set("interactions",
+                     list(self.get("interactions") or []) + q)
+            self._quadratic_specs = []
         comm = get_comm()
         device = default_device(self.get("device"))
         t_ingest = time.perf_counter()
         df, extra_cols = self._feature_frame(df)
         idx, val, off, _ = _extract_csr(df, self.get("featuresCol"),
                                         extra_cols, device)
+        # VW semantics: every feature index lands in the 2^b table (hash &
+        # mask); the concatenated column offsets can exceed it otherwise
+        idx = idx & ((1 << self.get("numBits")) - 1)
         y = df[self.get("labelCol")].to_numpy(dtype=np.float32)
         if self._binary_labels and getattr(self, "_convert_labels", True):
             y = np.where(y > 0, 1.0, -1.0).astype(np.float32)
@@ -262,19 +345,43 @@ class _VWBase(_VWParams, Estimator):
             w = self._fit_bfgs(idx, val, off, labels, ex_w, w, loss, l2,
                                comm)
             learn_s = time.perf_counter() - t0
-        for p in (range(0) if self.get("bfgs")
-                  else range(self.get("numPasses"))):
+        # multi-pass example cache replay (endPass/performRemainingPasses,
+        # VowpalWabbitBase.scala:363-368): the cache is the device-resident
+        # CSR itself.  VW holdout semantics: with passes>1 every k-th
+        # example is held out of training and scores each pass; training
+        # stops after earlyTerminate passes without holdout improvement
+        # (disable with --holdout_off).
+        n_passes = 0 if self.get("bfgs") else self.get("numPasses")
+        use_holdout = n_passes > 1 and not self.get("holdoutOff")
+        tr = (idx, val, off, labels, ex_w)
+        ho = None
+        if use_holdout:
+            period = max(2, self.get("holdoutPeriod"))
+            pos = torch.arange(n, device=off.device)
+            ho_mask = (pos % period) == (period - 1)
+            ti, tv, to = _filter_csr(idx, val, off, ~ho_mask)
+            tr = (ti, tv, to, labels[~ho_mask],
+                  ex_w[~ho_mask] if ex_w is not None else None)
+            hi, hv, hoff = _filter_csr(idx, val, off, ho_mask)
+            ho = (hi, hv, hoff, labels[ho_mask])
+        t_idx, t_val, t_off, t_y, t_w = tr
+        n_train = int(t_y.numel())
+        best_holdout = float("inf")
+        passes_run = 0
+        no_improve = 0
+        for p in range(n_passes):
             t0 = time.perf_counter()
-            for s in range(0, n, bs):
-                e = min(s + bs, n)
-                o = off[s:e + 1] - off[s]
-                sl = slice(int(off[s]), int(off[e]))
+            for s in range(0, n_train, bs):
+                e = min(s + bs, n_train)
+                o = t_off[s:e + 1] - t_off[s]
+                sl = slice(int(t_off[s]), int(t_off[e]))
                 backend.vw_sgd_minibatch(
-                    idx[sl], val[sl], o, labels[s:e], w, g, lr, l2,
+                    t_idx[sl], t_val[sl], o, t_y[s:e], w, g, lr, l2,
                     power_t, loss,
-                    ex_w[s:e] if ex_w is not None else None, s_tbl,
+                    t_w[s:e] if t_w is not None else None, s_tbl,
                     invariant=self.get("invariant"))
             learn_s += time.perf_counter() - t0
+            passes_run += 1
             # end-of-pass sync: RCCL all_reduce of weights + accumulators
             t0 = time.perf_counter()
             if comm.is_distributed:
@@ -287,6 +394,24 @@ class _VWBase(_VWParams, Estimator):
             if l1 > 0:  # proximal truncation
                 w.copy_(torch.sign(w) * (w.abs() - lr * l1).clamp_min(0))
             multipass_s += time.perf_counter() - t0
+            if ho is not None:
+                hloss = _avg_loss(backend.vw_predict(ho[0], ho[1], ho[2], w),
+                                  ho[3], loss)
+                if comm.is_distributed:
+                    t = torch.tensor([hloss, 1.0], device=w.device)
+                    comm.all_reduce(t)
+                    hloss = float(t[0] / t[1])
+                if hloss < best_holdout - 1e-9:
+                    best_holdout, no_improve = hloss, 0
+                    best_snap = (w.clone(), g.clone())
+                else:
+                    no_improve += 1
+                    if no_improve >= self.get("earlyTerminate"):
+                        break
+        if ho is not None and best_holdout < float("inf"):
+            # keep the best-holdout pass's weights (early termination would
+            # otherwise return a worse — possibly diverged — final pass)
+            w, g = best_snap
 
         model = self._model_class()(weights=w.cpu().numpy(),
                                     adaptive=g.cpu().numpy())
@@ -296,7 +421,9 @@ class _VWBase(_VWParams, Estimator):
             model.set(pname, self.get(pname))
         model.set("lossFunction", loss)
         model._stats = pd.DataFrame([{
-            "partitionId": comm.rank, "ipassCurrent": self.get("numPasses"),
+            "partitionId": comm.rank,
+            "ipassCurrent": (passes_run if n_passes else
+                             self.get("numPasses")),
             "numberOfExamplesPerPass": n,
             "totalNumberOfFeatures": int(off[-1]),
             "nativeIngestTimeNs": int(ingest_s * 1e9),
@@ -342,6 +469,7 @@ class _VWModelBase(_VWParams, Model):
         idx, val, off, _ = _extract_csr(df, self.get("featuresCol"),
                                         extra_cols, device)
         w = torch.from_numpy(self.weights).to(device)
+        idx = idx & (w.numel() - 1)  # table mask, matching _fit
         return backend.vw_predict(idx, val, off, w).cpu().numpy()
 
 

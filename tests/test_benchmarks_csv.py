@@ -54,7 +54,8 @@ def test_vw_benchmark_csv():
         rows.append(SparseVector(size, idx.astype(np.int32), val))
         ys.append(float((w_true[idx]).sum()))
     df = pd.DataFrame({"features": rows, "label": ys})
-    m = VowpalWabbitRegressor(numPasses=8, numBits=14, learningRate=0.3).fit(df)
+    m = VowpalWabbitRegressor(numPasses=8, numBits=14, learningRate=0.3,
+                              holdoutOff=True).fit(df)
     pred = m.transform(df)["prediction"].to_numpy()
     mse = float(((pred - np.asarray(ys)) ** 2).mean())
     runner = BenchmarkRunner("VerifyVowpalWabbitRegressor", RESOURCE_DIR)
@@ -88,5 +89,32 @@ def test_train_classifier_benchmark_csv():
                precision=0.05)
     runner.add("aupr_pima_shaped", float(stats.iloc[0].get(
         "AUPR", stats.iloc[0].get("precision", 0.0))), precision=0.08)
+    problems = runner.compare()
+    assert not problems, problems
+
+
+def test_vw_multipass_quadratic_benchmark_csv():
+    """committed MSE bar for a '--passes 4 -q ::' style
+    run — multi-pass replay + raw-input namespace crossing
+    (benchmarks_VerifyVowpalWabbitRegressor.csv discipline)."""
+    from mmlspark_amd.models.vw.estimators import VowpalWabbitRegressor
+    rng = np.random.default_rng(13)
+    n = 3000
+    xa = rng.normal(size=(n, 4)).astype(np.float32)
+    xb = rng.normal(size=(n, 4)).astype(np.float32)
+    # label needs the a×b cross terms: linear-only MSE stays high
+    y = (xa[:, 0] * xb[:, 0] + 0.5 * xa[:, 1] * xb[:, 1]
+         + 0.1 * rng.normal(size=n)).astype(np.float32)
+    df = pd.DataFrame({"afeat": list(xa), "bfeat": list(xb), "label": y})
+    m = VowpalWabbitRegressor(featuresCol="afeat",
+                              additionalFeatures=["bfeat"],
+                              passThroughArgs="--passes 16 -q ::",
+                              numBits=18, learningRate=0.02,
+                              adaptive=True, holdoutOff=True).fit(df)
+    pred = m.transform(df)["prediction"].to_numpy()
+    mse = float(((pred - y) ** 2).mean())
+    runner = BenchmarkRunner("VerifyVowpalWabbitRegressor", RESOURCE_DIR)
+    runner.add("mse_synthetic_passes4_quadratic", mse, precision=0.05,
+               higher_is_better=False)
     problems = runner.compare()
     assert not problems, problems

@@ -102,7 +102,8 @@ def test_vw_regressor_learns():
         rows.append(SparseVector(size, idx.astype(np.int32), val))
         ys.append(float((w_true[idx] * val).sum()))
     df = pd.DataFrame({"features": rows, "label": ys})
-    m = VowpalWabbitRegressor(numPasses=10, numBits=14, learningRate=0.3).fit(df)
+    m = VowpalWabbitRegressor(numPasses=10, numBits=14, learningRate=0.3,
+                              holdoutOff=True).fit(df)
     pred = m.transform(df)["prediction"].to_numpy()
     y = np.asarray(ys)
     ss_res = ((pred - y) ** 2).sum()
@@ -312,3 +313,59 @@ def test_bfgs_mode():
     r = VowpalWabbitRegressor(bfgs=True, l2=1e-6).fit(dfr)
     pred = r.transform(dfr)["prediction"].to_numpy()
     assert np.corrcoef(pred, yr)[0, 1] > 0.99
+
+
+def test_raw_quadratic_namespace_crossing():
+    """-q on raw (unfeaturized) input (VowpalWabbitBase arg surface): a
+    multiplicative task that a linear model cannot learn becomes learnable
+    with '-q ab' namespace crossing."""
+    from mmlspark_amd.models.vw.estimators import VowpalWabbitClassifier
+    rng = np.random.default_rng(3)
+    n = 6000
+    xa = rng.choice([-1.0, 1.0], size=n).astype(np.float32)
+    xb = rng.choice([-1.0, 1.0], size=n).astype(np.float32)
+    y = (xa * xb > 0).astype(np.float32)  # XOR-like: linear AUC ≈ 0.5
+    df = pd.DataFrame({
+        "afeat": [np.array([v, 1.0], dtype=np.float32) for v in xa],
+        "bfeat": [np.array([v, 1.0], dtype=np.float32) for v in xb],
+        "label": y,
+    })
+    kw = dict(featuresCol="afeat", additionalFeatures=["bfeat"],
+              numPasses=8, numBits=18, learningRate=0.5, adaptive=True,
+              holdoutOff=True)
+    m_lin = VowpalWabbitClassifier(**kw).fit(df)
+    acc_lin = (m_lin.transform(df)["prediction"].to_numpy() == y).mean()
+    m_q = VowpalWabbitClassifier(passThroughArgs="-q ab", **kw).fit(df)
+    acc_q = (m_q.transform(df)["prediction"].to_numpy() == y).mean()
+    assert acc_lin < 0.62, acc_lin
+    assert acc_q > 0.95, acc_q
+    # ':' wildcard resolves to all column pairs
+    m_w = VowpalWabbitClassifier(passThroughArgs="--quadratic ::", **kw).fit(df)
+    acc_w = (m_w.transform(df)["prediction"].to_numpy() == y).mean()
+    assert acc_w > 0.95, acc_w
+
+
+def test_multipass_holdout_early_terminate():
+    """VW multi-pass semantics: every 10th example held out; training stops
+    after earlyTerminate passes without holdout improvement (and
+    --holdout_off disables all of it)."""
+    from mmlspark_amd.models.vw.estimators import VowpalWabbitRegressor
+    rng = np.random.default_rng(5)
+    n, d = 2000, 32
+    X = rng.normal(size=(n, d)).astype(np.float32)
+    w = rng.normal(size=d)
+    y = (X @ w).astype(np.float32)  # learnable fast → later passes plateau
+    df = pd.DataFrame({"features": list(X), "label": y})
+    m = VowpalWabbitRegressor(numPasses=40, learningRate=0.5, numBits=12,
+                              adaptive=True, earlyTerminate=2).fit(df)
+    stats = m.getPerformanceStatistics()
+    ran = int(stats.iloc[0]["ipassCurrent"])
+    assert ran < 40, ran  # early-terminated
+    m2 = VowpalWabbitRegressor(numPasses=5, learningRate=0.5, numBits=12,
+                               adaptive=True, holdoutOff=True).fit(df)
+    assert int(m2.getPerformanceStatistics().iloc[0]["ipassCurrent"]) == 5
+    # passThroughArgs spellings parse
+    m3 = VowpalWabbitRegressor(
+        passThroughArgs="--passes 3 --holdout_off", learningRate=0.5,
+        numBits=12).fit(df)
+    assert int(m3.getPerformanceStatistics().iloc[0]["ipassCurrent"]) == 3
