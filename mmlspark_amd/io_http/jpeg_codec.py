@@ -179,6 +179,18 @@ def _extend(v: int, n: int) -> int:
 
 
 def decode_jpeg(data: bytes) -> np.ndarray:
+    """Decode a JPEG: the native C++ decoder (baseline + progressive SOF2,
+    jpeg_native.cpp) when built, else the pure-Python baseline path below."""
+    try:
+        import torch  # noqa: F401 — loads libc10 for the extension
+        from . import _jpeg_native
+        return _jpeg_native.decode_jpeg(bytes(data)).numpy()
+    except ImportError:
+        pass
+    return _decode_jpeg_py(data)
+
+
+def _decode_jpeg_py(data: bytes) -> np.ndarray:
     if data[:2] != b"\xff\xd8":
         raise ValueError("not a JPEG")
     pos = 2

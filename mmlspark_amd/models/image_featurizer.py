@@ -28,10 +28,18 @@ from .resnet import MODELS
 
 
 def _rows_to_batch(vals, size: int) -> torch.Tensor:
-    """Image cells (H,W,C uint8 | C,H,W float | flat) -> (N,3,size,size) f32."""
+    """Image cells (H,W,C uint8 | C,H,W float | flat | encoded bytes)
+    -> (N,3,size,size) f32.  Encoded bytes (JPEG/PNG/BMP/PPM) decode
+    through io_http.files (native C++ JPEG decoder when built) — the
+    bytes-to-features path of the reference's ImageFeaturizer
+    (core/.../core/image/ImageUtils.scala)."""
     outs = []
     for v in vals:
-        a = np.asarray(v)
+        if isinstance(v, (bytes, bytearray, memoryview)):
+            from ..io_http.files import decode_image
+            a = decode_image(bytes(v))
+        else:
+            a = np.asarray(v)
         if a.ndim == 1:  # flattened
             side = int(round((a.size / 3) ** 0.5))
             a = a.reshape(3, side, side) if a.size == 3 * side * side else a

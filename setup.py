@@ -66,6 +66,12 @@ setup(
                                 f"-I{ROCM}/include"],
             extra_link_args=[f"-L{ROCM}/lib", "-lamdhip64"],
         ),
+        cpp_extension.CppExtension(
+            name="mmlspark_amd.io_http._jpeg_native",
+            sources=[os.path.join(ROOT, "mmlspark_amd", "io_http",
+                                  "jpeg_native.cpp")],
+            extra_compile_args=["-O3"],
+        ),
     ],
     cmdclass={"build_ext": cpp_extension.BuildExtension.with_options(
         use_ninja=False)},
