@@ -1,0 +1,109 @@
+"""R binding verification (the reference generates AND
+tests R — Wrappable.scala:393, core/src/test/R/testthat).
+
+R itself is not in this image, so verification is (a) a structural lint of
+the generated reticulate code — balanced delimiters, one ml_* function per
+registered stage, every declared argument wired to a stage$set — and (b) a
+semantic round trip: execute each generated function body's Python
+equivalent (the import path + set() calls it emits) and check the stage
+comes back with exactly those params set."""
+import os
+import re
+
+import pytest
+
+import mmlspark_amd
+from mmlspark_amd.core.registry import all_stages
+
+R_PATH = os.path.join(os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))), "bindings", "R", "mmlspark_amd.R")
+
+
+@pytest.fixture(scope="module")
+def r_source():
+    if not os.path.exists(R_PATH):
+        from mmlspark_amd.core.codegen import generate_r_wrappers
+        generate_r_wrappers(R_PATH)
+    with open(R_PATH) as f:
+        return f.read()
+
+
+def _functions(src):
+    """Parse `name <- function(args) { body }` blocks."""
+    out = {}
+    for m in re.finditer(
+            r"^(ml_\w+) <- function\(([^)]*)\) \{\n(.*?)^\}", src,
+            re.M | re.S):
+        out[m.group(1)] = (m.group(2), m.group(3))
+    return out
+
+
+def test_r_file_structural_lint(r_source):
+    assert r_source.count("{") == r_source.count("}")
+    assert r_source.count("(") == r_source.count(")")
+    fns = _functions(r_source)
+    assert len(fns) >= 150, len(fns)  # one per registered stage
+    for name, (args, body) in fns.items():
+        # every declared argument must be wired to a stage$set call
+        argnames = [a.split("=")[0].strip() for a in args.split(",") if a.strip()]
+        for a in argnames:
+            assert f'stage$set("{a}", {a})' in body, (name, a)
+        # the constructor line references a real import path
+        m = re.search(r"stage <- mmlspark_amd((?:\$\w+)+)\(\)", body)
+        assert m, name
+        assert body.rstrip().endswith("stage")
+
+
+def test_r_functions_cover_registry(r_source):
+    fns = _functions(r_source)
+
+    def snake(n):
+        return re.sub(r"(?<!^)(?=[A-Z])", "_", n).lower()
+
+    missing = [n for n in all_stages()
+               if f"ml_{snake(n)}" not in fns]
+    assert not missing, missing[:10]
+
+
+def test_r_semantic_round_trip(r_source):
+    """Execute what the R body does (resolve the python path, construct,
+    set each param) for a sample of stages, incl. LightGBMClassifier."""
+    fns = _functions(r_source)
+    checked = 0
+    for name, (args, body) in sorted(fns.items()):
+        if checked >= 25 and name != "ml_light_gbm_classifier":
+            continue
+        m = re.search(r"stage <- mmlspark_amd((?:\$\w+)+)\(\)", body)
+        parts = m.group(1).replace("$", ".").lstrip(".").split(".")
+        import importlib
+        mod = importlib.import_module("mmlspark_amd." + ".".join(parts[:-1]))
+        stage = getattr(mod, parts[-1])()
+        argnames = [a.split("=")[0].strip() for a in args.split(",")
+                    if a.strip()]
+        params = stage.params()
+        for a in argnames:
+            assert a in params, (name, a)
+        # round-trip one simple param value through set/get (what the
+        # generated `stage$set(name, value)` line does)
+        for a in argnames:
+            p = params[a]
+            v = stage.get(a)
+            if getattr(p, "is_complex", False) or v is None:
+                continue
+            stage.set(a, v)
+            assert stage.get(a) == v, (name, a)
+            break
+        checked += 1
+    assert checked >= 25
+
+
+def test_r_lightgbm_params_match_python():
+    """The flagship estimator's R arg list == its Python param surface."""
+    with open(R_PATH) as f:
+        src = f.read()
+    fns = _functions(src)
+    args, _ = fns["ml_light_gbm_classifier"]
+    argnames = {a.split("=")[0].strip() for a in args.split(",") if a.strip()}
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    pnames = set(LightGBMClassifier().params().keys())
+    assert argnames == pnames, (argnames ^ pnames)
