@@ -33,9 +33,28 @@ def _rows_to_batch(vals, size: int) -> torch.Tensor:
     through io_http.files (native C++ JPEG decoder when built) — the
     bytes-to-features path of the reference's ImageFeaturizer
     (core/.../core/image/ImageUtils.scala)."""
+    decoded = None
+    if len(vals) and isinstance(vals[0], (bytes, bytearray, memoryview)):
+        # decode in a thread pool: the native JPEG decoder releases the GIL
+        from concurrent.futures import ThreadPoolExecutor
+        from ..io_http.files import decode_image
+        with ThreadPoolExecutor(max_workers=min(16, max(1, len(vals)))) as ex:
+            decoded = list(ex.map(lambda b: decode_image(bytes(b)), vals))
+        if all(d.ndim == 3 and d.shape == decoded[0].shape
+               and d.dtype == np.uint8 for d in decoded):
+            # uniform decoded batch: one stack + one vectorized normalize
+            batch = torch.from_numpy(np.stack(decoded)).permute(
+                0, 3, 1, 2).float().div_(255.0)
+            if batch.shape[-1] != size or batch.shape[-2] != size:
+                batch = torch.nn.functional.interpolate(
+                    batch, size=(size, size), mode="bilinear",
+                    align_corners=False)
+            return batch
     outs = []
-    for v in vals:
-        if isinstance(v, (bytes, bytearray, memoryview)):
+    for i, v in enumerate(vals):
+        if decoded is not None:
+            a = decoded[i]
+        elif isinstance(v, (bytes, bytearray, memoryview)):
             from ..io_http.files import decode_image
             a = decode_image(bytes(v))
         else:

@@ -229,6 +229,9 @@ class ServingServer:
             # adds ~40 ms per response on keep-alive connections)
             disable_nagle_algorithm = True
             daemon_threads = True
+            # 64+ concurrent clients connect simultaneously; the default
+            # listen backlog of 5 resets the burst
+            request_queue_size = 256
 
         self._httpd = _Server((self.host, self.port),
                               self._make_handler())

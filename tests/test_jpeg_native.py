@@ -307,3 +307,22 @@ def test_restart_markers_native():
     out = _jpeg_native.decode_jpeg(enc2).numpy()
     ref = _jpeg_native.decode_jpeg(bytes(enc)).numpy()
     np.testing.assert_array_equal(out, ref)
+
+
+def test_image_featurizer_accepts_jpeg_bytes():
+    """bytes → decode → features end to end (ImageUtils decode parity)."""
+    import pandas as pd
+    from mmlspark_amd.models.image_featurizer import ImageFeaturizer
+    img = _test_image(h=64, w=64, seed=2)
+    enc = jpeg_codec.encode_jpeg(img, quality=90)
+    feat = ImageFeaturizer(modelName="ResNet18", cutOutputLayers=1,
+                           imageSize=64)
+    out = feat.transform(pd.DataFrame({"image": [enc, enc, enc]}))
+    F = np.stack(out["features"].to_numpy())
+    assert F.shape[0] == 3 and F.shape[1] >= 128
+    assert np.isfinite(F).all()
+    # bytes path ≈ array path on the same pixels
+    arr = jpeg_codec.decode_jpeg(enc)
+    out2 = feat.transform(pd.DataFrame({"image": [arr, arr, arr]}))
+    F2 = np.stack(out2["features"].to_numpy())
+    np.testing.assert_allclose(F, F2, atol=1e-4)
