@@ -72,4 +72,55 @@ class ModelDownloader:
         raise KeyError(f"model {name!r} not in repo {self.local_path}")
 
     def load_state(self, name: str):
-        return torch.load(self.download_by_name(name).uri, map_location="cpu")
+        return torch.load(self.download_by_name(name).uri, map_location="cpu",
+                          weights_only=True)
+
+
+class RemoteRepo:
+    """HTTP model repository (HDFSRepo analog,
+    downloader/ModelDownloader.scala:42): a base URL serving models.json
+    plus the model payloads.  download_by_name fetches into a local
+    ModelDownloader cache with sha256 verification — the transport is
+    behind this interface so an offline deployment swaps in LocalRepo
+    (= ModelDownloader) unchanged."""
+
+    def __init__(self, base_url: str, cache: ModelDownloader,
+                 timeout: float = 60.0):
+        self.base_url = base_url.rstrip("/")
+        self.cache = cache
+        self.timeout = timeout
+
+    def _get(self, path: str) -> bytes:
+        import requests
+        r = requests.get(f"{self.base_url}/{path}", timeout=self.timeout)
+        r.raise_for_status()
+        return r.content
+
+    def list_models(self) -> List[ModelSchema]:
+        return [ModelSchema(**d) for d in json.loads(self._get("models.json"))]
+
+    def download_by_name(self, name: str) -> ModelSchema:
+        for m in self.list_models():
+            if m.name != name:
+                continue
+            local = os.path.join(self.cache.local_path, f"{name}.pt")
+            if not (os.path.exists(local)
+                    and ModelDownloader._sha256(local) == m.hash):
+                blob = self._get(os.path.basename(m.uri))
+                if hashlib.sha256(blob).hexdigest() != m.hash:
+                    raise IOError(f"hash mismatch downloading {name}")
+                with open(local, "wb") as f:
+                    f.write(blob)
+            m.uri = local
+            # register in the local cache index so later offline runs
+            # resolve it through ModelDownloader directly
+            models = [x for x in self.cache.list_models() if x.name != name]
+            models.append(m)
+            with open(self.cache._index_path(), "w") as f:
+                json.dump([asdict(x) for x in models], f, indent=1)
+            return m
+        raise KeyError(f"model {name!r} not in remote repo {self.base_url}")
+
+    def load_state(self, name: str):
+        return torch.load(self.download_by_name(name).uri, map_location="cpu",
+                          weights_only=True)
