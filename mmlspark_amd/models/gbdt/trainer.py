@@ -584,14 +584,11 @@ class SparseTreeGrower(TreeGrower):
     def _hist(self, rows, grad, hess, reduce=True):
         t0 = time.perf_counter()
         sh = self.shard
-        h = backend.csr_hist_fixed(sh.indptr, sh.col, sh.binv, self._gq,
-                                   self._hq, rows, self.nf, self.cfg.max_bin)
+        h, tot = backend.csr_hist_fixed_tot(sh.indptr, sh.col, sh.binv,
+                                            self._gq, self._hq, rows,
+                                            self.nf, self.cfg.max_bin)
         # implicit zeros: exact integer leaf totals − per-feature stored sums
-        r = rows.long()
-        tot = torch.stack([
-            self._gq[r].sum(), self._hq[r].sum(),
-            torch.tensor(int(rows.numel()), dtype=torch.int64,
-                         device=h.device)])
+        tot = tot.to(h.device)
         corr = tot.unsqueeze(0) - h.sum(dim=1)  # (nf, 3)
         ar = torch.arange(self.nf, device=h.device)
         h[ar, sh.zero_bin.long()] += corr

@@ -172,3 +172,18 @@ def csr_gather_bins(indptr, col, binv, rows, feature, zero_bin):
                                               rows.contiguous(), feature,
                                               zero_bin).long()
     return cpu_ref.csr_gather_bins(indptr, col, binv, rows, feature, zero_bin)
+
+
+def csr_hist_fixed_tot(indptr, col, binv, gq, hq, rows, nf, n_bins):
+    """Histogram over stored entries PLUS exact integer leaf totals
+    (sum gq, sum hq, count) in one launch — (hist, tot).  GPU: the
+    wave-cooperative v2 kernel; CPU: reference + torch sums."""
+    if binv.is_cuda:
+        return _require_ext().csr_hist_fixed_tot(indptr, col, binv, gq, hq,
+                                                 rows.contiguous(), nf,
+                                                 n_bins)
+    h = cpu_ref.csr_hist_fixed(indptr, col, binv, gq, hq, rows, nf, n_bins)
+    r = rows.long()
+    tot = torch.stack([gq[r].sum(), hq[r].sum(),
+                       torch.tensor(int(rows.numel()), dtype=torch.int64)])
+    return h, tot

@@ -47,6 +47,9 @@ void launch_csr_hist_fixed(const long*, const int*,
                            int, hipStream_t);
 void launch_csr_gather_bin(const long*, const int*, const unsigned char*,
                            const int*, long, int, int, int*, hipStream_t);
+void launch_csr_hist_fixed_v2(const long*, const int*, const unsigned char*,
+                              const long long*, const long long*, const int*,
+                              long, long long*, int, long long*, hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -311,6 +314,31 @@ torch::Tensor csr_hist_fixed(torch::Tensor indptr, torch::Tensor col,
   return hist;
 }
 
+// v2: wave-cooperative row chunks + in-kernel leaf totals
+std::tuple<torch::Tensor, torch::Tensor> csr_hist_fixed_tot(
+    torch::Tensor indptr, torch::Tensor col, torch::Tensor binv,
+    torch::Tensor gq, torch::Tensor hq, torch::Tensor rows, long nf,
+    long n_bins) {
+  CHECK_DEV(indptr); CHECK_CONTIG(indptr);
+  CHECK_DEV(col); CHECK_CONTIG(col);
+  CHECK_DEV(binv); CHECK_CONTIG(binv);
+  CHECK_DEV(gq); CHECK_CONTIG(gq);
+  CHECK_DEV(hq); CHECK_CONTIG(hq);
+  CHECK_DEV(rows); CHECK_CONTIG(rows);
+  auto hist = torch::zeros({nf, n_bins, 3},
+                           gq.options().dtype(torch::kInt64));
+  auto tot = torch::zeros({3}, gq.options().dtype(torch::kInt64));
+  launch_csr_hist_fixed_v2(indptr.data_ptr<long>(), col.data_ptr<int>(),
+                           binv.data_ptr<unsigned char>(),
+                           (const long long*)gq.data_ptr<int64_t>(),
+                           (const long long*)hq.data_ptr<int64_t>(),
+                           rows.data_ptr<int>(), rows.numel(),
+                           (long long*)hist.data_ptr<int64_t>(), (int)n_bins,
+                           (long long*)tot.data_ptr<int64_t>(),
+                           cur_stream());
+  return {hist, tot};
+}
+
 torch::Tensor csr_gather_bins(torch::Tensor indptr, torch::Tensor col,
                               torch::Tensor binv, torch::Tensor rows,
                               long feature, long zero_bin) {
@@ -346,4 +374,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fixed-point histogram over stored CSR entries");
   m.def("csr_gather_bins", &csr_gather_bins,
         "per-row bin of one feature from CSR (missing -> zero bin)");
+  m.def("csr_hist_fixed_tot", &csr_hist_fixed_tot,
+        "wave-cooperative CSR histogram + in-kernel leaf totals");
 }

@@ -1463,3 +1463,104 @@ extern "C" void launch_csr_gather_bin(const long* indptr, const int* col,
                      stream, indptr, col, binv, rows, m, feature, zero_bin,
                      out);
 }
+
+// v2 sparse histogram: one WAVE cooperates on 64 consecutive rows of the
+// (sorted) row list.  Row bounds + quantized (g,h) stage through LDS; an
+// in-wave prefix over row lengths maps entry slots to rows, so col/binv
+// reads are coalesced runs instead of 20-strided gathers, and gq/hq come
+// from LDS instead of per-entry global gathers.  Leaf totals (sum gq, sum
+// hq, count) accumulate in-kernel (one atomic per wave) so the host-side
+// correction needs no extra passes over the row list.
+__global__ void csr_hist_fixed_v2_k(const long* __restrict__ indptr,
+                                    const int* __restrict__ col,
+                                    const unsigned char* __restrict__ binv,
+                                    const long long* __restrict__ gq,
+                                    const long long* __restrict__ hq,
+                                    const int* __restrict__ rows, long m,
+                                    long long* __restrict__ hist, int n_bins,
+                                    long long* __restrict__ tot) {
+  __shared__ long s_start[4][64];
+  __shared__ int s_len[4][65];      // inclusive prefix at +1
+  __shared__ long long s_g[4][64], s_h[4][64];
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  unsigned long long tg = 0, th = 0, tc = 0;
+
+  // block-uniform outer loop (same trip count for all 4 waves — the
+  // barriers below must not diverge); each wave owns a 64-row sub-chunk
+  for (long bbase = (long)blockIdx.x * 256; bbase < m;
+       bbase += (long)gridDim.x * 256) {
+    const long base = bbase + (long)wid * 64;
+    const int nrows = (int)max(0l, min((long)64, m - base));
+    long my_len = 0;
+    if (lane < nrows) {
+      const int r = rows[base + lane];
+      const long s = indptr[r];
+      s_start[wid][lane] = s;
+      my_len = indptr[r + 1] - s;
+      const long long g = gq[r];
+      const long long h = hq[r];
+      s_g[wid][lane] = g;
+      s_h[wid][lane] = h;
+      tg += (unsigned long long)g;
+      th += (unsigned long long)h;
+      tc += 1;
+    }
+    // inclusive prefix of row lengths across the wave (LDS scan)
+    s_len[wid][0] = 0;
+    int v = (int)my_len;
+#pragma unroll
+    for (int d = 1; d < 64; d <<= 1) {
+      const int up = __shfl_up((int)v, d);
+      if (lane >= d) v += up;
+    }
+    s_len[wid][lane + 1] = v;
+    __syncthreads();
+    const int total = s_len[wid][nrows];
+    const int* cum = s_len[wid];
+    for (int e = lane; e < total; e += 64) {
+      // binary search: largest k with cum[k] <= e
+      int lo2 = 0, hi2 = nrows;
+      while (lo2 + 1 < hi2) {
+        const int mid = (lo2 + hi2) >> 1;
+        if (cum[mid] <= e) lo2 = mid; else hi2 = mid;
+      }
+      const long j = s_start[wid][lo2] + (e - cum[lo2]);
+      unsigned long long* cell = (unsigned long long*)hist
+          + ((size_t)col[j] * n_bins + binv[j]) * 3;
+      atomicAdd(cell + 0, (unsigned long long)s_g[wid][lo2]);
+      atomicAdd(cell + 1, (unsigned long long)s_h[wid][lo2]);
+      atomicAdd(cell + 2, 1ull);
+    }
+    __syncthreads();
+  }
+  // wave-reduce totals, one atomic triple per wave
+#pragma unroll
+  for (int d = 32; d > 0; d >>= 1) {
+    tg += (unsigned long long)__shfl_down((long long)tg, d);
+    th += (unsigned long long)__shfl_down((long long)th, d);
+    tc += (unsigned long long)__shfl_down((long long)tc, d);
+  }
+  if (lane == 0 && tot) {
+    atomicAdd((unsigned long long*)tot + 0, tg);
+    atomicAdd((unsigned long long*)tot + 1, th);
+    atomicAdd((unsigned long long*)tot + 2, tc);
+  }
+}
+
+extern "C" void launch_csr_hist_fixed_v2(const long* indptr, const int* col,
+                                         const unsigned char* binv,
+                                         const long long* gq,
+                                         const long long* hq,
+                                         const int* rows, long m,
+                                         long long* hist, int n_bins,
+                                         long long* tot,
+                                         hipStream_t stream) {
+  if (m == 0) return;
+  const int threads = 256;                  // 4 waves x 64 rows
+  const long want = (m + 255) / 256;
+  const int blocks = (int)(want < 4096 ? (want > 0 ? want : 1) : 4096);
+  hipLaunchKernelGGL(csr_hist_fixed_v2_k, dim3(blocks), dim3(threads), 0,
+                     stream, indptr, col, binv, gq, hq, rows, m, hist,
+                     n_bins, tot);
+}

@@ -132,3 +132,29 @@ def test_sparse_high_dim_1m_x_100k():
           f"hist {stats.hist_s:.2f}s total {stats.total_s:.2f}s")
     assert booster.num_trees == 3
     assert peak < 40 * 2**30  # far under the 100 GB+ a dense path would need
+
+
+@requires_gpu
+def test_csr_hist_v2_fused_totals_matches_cpu():
+    """Wave-cooperative v2 kernel (LDS row staging + in-kernel totals)
+    must be integer-exact vs the reference, including ragged tails."""
+    from mmlspark_amd.ops import backend, cpu_ref
+    gen = torch.Generator().manual_seed(7)
+    for n, nf, nb, nnz in ((50_000, 300, 31, 12), (4097, 64, 255, 3),
+                           (63, 10, 15, 5)):
+        indptr, col, binv = _rand_csr(gen, n, nf, nnz)
+        gq = torch.randint(-2**40, 2**40, (n,), generator=gen,
+                           dtype=torch.int64)
+        hq = torch.randint(0, 2**24, (n,), generator=gen, dtype=torch.int64)
+        rows = torch.randperm(n, generator=gen)[: max(1, n // 3)].to(
+            torch.int32).sort().values
+        ref = cpu_ref.csr_hist_fixed(indptr, col, binv, gq, hq, rows, nf, nb)
+        r = rows.long()
+        ref_tot = torch.stack([gq[r].sum(), hq[r].sum(),
+                               torch.tensor(int(rows.numel()),
+                                            dtype=torch.int64)])
+        out, tot = backend.csr_hist_fixed_tot(
+            indptr.cuda(), col.cuda(), binv.cuda(), gq.cuda(), hq.cuda(),
+            rows.cuda(), nf, nb)
+        assert torch.equal(out.cpu(), ref), (n, nf)
+        assert torch.equal(tot.cpu(), ref_tot), (n, tot.cpu(), ref_tot)
