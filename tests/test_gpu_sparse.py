@@ -143,6 +143,7 @@ def test_csr_hist_v2_fused_totals_matches_cpu():
     for n, nf, nb, nnz in ((50_000, 300, 31, 12), (4097, 64, 255, 3),
                            (63, 10, 15, 5)):
         indptr, col, binv = _rand_csr(gen, n, nf, nnz)
+        binv = (binv % nb).contiguous()  # bins must lie in [0, nb)
         gq = torch.randint(-2**40, 2**40, (n,), generator=gen,
                            dtype=torch.int64)
         hq = torch.randint(0, 2**24, (n,), generator=gen, dtype=torch.int64)
