@@ -50,6 +50,10 @@ void launch_csr_gather_bin(const long*, const int*, const unsigned char*,
 void launch_csr_hist_fixed_v2(const long*, const int*, const unsigned char*,
                               const long long*, const long long*, const int*,
                               long, long long*, int, long long*, hipStream_t);
+void launch_csr_hist_fixed_lds(const long*, const int*, const unsigned char*,
+                               const long long*, const long long*,
+                               const int*, long, long long*, int, int,
+                               long long*, hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -328,14 +332,29 @@ std::tuple<torch::Tensor, torch::Tensor> csr_hist_fixed_tot(
   auto hist = torch::zeros({nf, n_bins, 3},
                            gq.options().dtype(torch::kInt64));
   auto tot = torch::zeros({3}, gq.options().dtype(torch::kInt64));
-  launch_csr_hist_fixed_v2(indptr.data_ptr<long>(), col.data_ptr<int>(),
-                           binv.data_ptr<unsigned char>(),
-                           (const long long*)gq.data_ptr<int64_t>(),
-                           (const long long*)hq.data_ptr<int64_t>(),
-                           rows.data_ptr<int>(), rows.numel(),
-                           (long long*)hist.data_ptr<int64_t>(), (int)n_bins,
-                           (long long*)tot.data_ptr<int64_t>(),
-                           cur_stream());
+  if (nf <= 2048) {
+    // small/medium nf: LDS-privatized feature chunks (redundant coalesced
+    // reads, 9x-faster LDS atomics, one global flush per cell)
+    launch_csr_hist_fixed_lds(indptr.data_ptr<long>(), col.data_ptr<int>(),
+                              binv.data_ptr<unsigned char>(),
+                              (const long long*)gq.data_ptr<int64_t>(),
+                              (const long long*)hq.data_ptr<int64_t>(),
+                              rows.data_ptr<int>(), rows.numel(),
+                              (long long*)hist.data_ptr<int64_t>(),
+                              (int)n_bins, (int)nf,
+                              (long long*)tot.data_ptr<int64_t>(),
+                              cur_stream());
+  } else {
+    // wide-nf: atomics spread across nf*n_bins cells — contention is low
+    launch_csr_hist_fixed_v2(indptr.data_ptr<long>(), col.data_ptr<int>(),
+                             binv.data_ptr<unsigned char>(),
+                             (const long long*)gq.data_ptr<int64_t>(),
+                             (const long long*)hq.data_ptr<int64_t>(),
+                             rows.data_ptr<int>(), rows.numel(),
+                             (long long*)hist.data_ptr<int64_t>(), (int)n_bins,
+                             (long long*)tot.data_ptr<int64_t>(),
+                             cur_stream());
+  }
   return {hist, tot};
 }
 

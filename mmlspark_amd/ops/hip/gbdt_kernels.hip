@@ -1564,3 +1564,133 @@ extern "C" void launch_csr_hist_fixed_v2(const long* indptr, const int* col,
                      stream, indptr, col, binv, gq, hq, rows, m, hist,
                      n_bins, tot);
 }
+
+// LDS-privatized sparse histogram for small/medium nf: grid dim y carves
+// features into chunks of 8; each block stages an 8-feature × n_bins packed
+// (g, count|h) histogram in LDS (the 9× integer-LDS-atomic win of the dense
+// kernel) and streams its row chunk's entries COALESCED, filtering to its
+// feature chunk.  Global atomics drop from 3-per-entry to one flush per
+// LDS cell.  Row staging + entry→row mapping as in v2.  Row count per
+// block must stay ≤ 2^19 so the 44-bit packed h-sum cannot overflow
+// (launcher enforces via grid sizing).
+__global__ void csr_hist_fixed_lds_k(const long* __restrict__ indptr,
+                                     const int* __restrict__ col,
+                                     const unsigned char* __restrict__ binv,
+                                     const long long* __restrict__ gq,
+                                     const long long* __restrict__ hq,
+                                     const int* __restrict__ rows, long m,
+                                     long long* __restrict__ hist,
+                                     int n_bins, int nf,
+                                     long long* __restrict__ tot) {
+  extern __shared__ unsigned long long lds64[];  // [8][n_bins][2]
+  const int f0 = blockIdx.y * 8;
+  const int tid = threadIdx.x;
+  const int lds_elems = 8 * n_bins * 2;
+  for (int i = tid; i < lds_elems; i += blockDim.x) lds64[i] = 0ull;
+
+  __shared__ long s_start[4][64];
+  __shared__ int s_len[4][65];
+  __shared__ long long s_g[4][64], s_h[4][64];
+  const int wid = tid >> 6;
+  const int lane = tid & 63;
+  constexpr unsigned long long CNT_ONE = 1ull << 44;
+  unsigned long long tg = 0, th = 0, tc = 0;
+  __syncthreads();
+
+  for (long bbase = (long)blockIdx.x * 256; bbase < m;
+       bbase += (long)gridDim.x * 256) {
+    const long base = bbase + (long)wid * 64;
+    const int nrows = (int)max(0l, min((long)64, m - base));
+    long my_len = 0;
+    if (lane < nrows) {
+      const int r = rows[base + lane];
+      const long s = indptr[r];
+      s_start[wid][lane] = s;
+      my_len = indptr[r + 1] - s;
+      const long long g = gq[r];
+      const long long h = hq[r];
+      s_g[wid][lane] = g;
+      s_h[wid][lane] = h;
+      if (blockIdx.y == 0) {  // totals once, not per feature chunk
+        tg += (unsigned long long)g;
+        th += (unsigned long long)h;
+        tc += 1;
+      }
+    }
+    s_len[wid][0] = 0;
+    int v = (int)my_len;
+#pragma unroll
+    for (int d = 1; d < 64; d <<= 1) {
+      const int up = __shfl_up((int)v, d);
+      if (lane >= d) v += up;
+    }
+    s_len[wid][lane + 1] = v;
+    __syncthreads();
+    const int total = s_len[wid][nrows];
+    const int* cum = s_len[wid];
+    for (int e = lane; e < total; e += 64) {
+      int lo2 = 0, hi2 = nrows;
+      while (lo2 + 1 < hi2) {
+        const int mid = (lo2 + hi2) >> 1;
+        if (cum[mid] <= e) lo2 = mid; else hi2 = mid;
+      }
+      const long j = s_start[wid][lo2] + (e - cum[lo2]);
+      const int c = col[j];
+      if (c >= f0 && c < f0 + 8) {
+        unsigned long long* cell =
+            &lds64[(((c - f0) * n_bins) + binv[j]) * 2];
+        atomicAdd(cell + 0, (unsigned long long)s_g[wid][lo2]);
+        atomicAdd(cell + 1, CNT_ONE + (unsigned long long)s_h[wid][lo2]);
+      }
+    }
+    __syncthreads();
+  }
+
+  // flush LDS chunk to the global (nf, n_bins, 3) histogram
+  for (int i = tid; i < 8 * n_bins; i += blockDim.x) {
+    const int f = i / n_bins;
+    if (f0 + f >= nf) break;
+    const unsigned long long gsum = lds64[i * 2 + 0];
+    const unsigned long long hpacked = lds64[i * 2 + 1];
+    if (gsum == 0ull && hpacked == 0ull) continue;
+    long long* out = hist + ((size_t)(f0 + f) * n_bins + (i % n_bins)) * 3;
+    atomicAdd((unsigned long long*)(out + 0), gsum);
+    atomicAdd((unsigned long long*)(out + 1),
+              hpacked & ((1ull << 44) - 1ull));
+    atomicAdd((unsigned long long*)(out + 2), hpacked >> 44);
+  }
+  if (blockIdx.y == 0) {
+#pragma unroll
+    for (int d = 32; d > 0; d >>= 1) {
+      tg += (unsigned long long)__shfl_down((long long)tg, d);
+      th += (unsigned long long)__shfl_down((long long)th, d);
+      tc += (unsigned long long)__shfl_down((long long)tc, d);
+    }
+    if (lane == 0 && tot) {
+      atomicAdd((unsigned long long*)tot + 0, tg);
+      atomicAdd((unsigned long long*)tot + 1, th);
+      atomicAdd((unsigned long long*)tot + 2, tc);
+    }
+  }
+}
+
+extern "C" void launch_csr_hist_fixed_lds(const long* indptr, const int* col,
+                                          const unsigned char* binv,
+                                          const long long* gq,
+                                          const long long* hq,
+                                          const int* rows, long m,
+                                          long long* hist, int n_bins,
+                                          int nf, long long* tot,
+                                          hipStream_t stream) {
+  if (m == 0) return;
+  long bx = (m + 2047) / 2048;
+  if (bx > 4096) bx = 4096;
+  const long min_bx = (m + (1l << 19) - 1) >> 19;  // ≤2^19 rows per block
+  if (bx < min_bx) bx = min_bx;
+  if (bx < 1) bx = 1;
+  const int by = (nf + 7) / 8;
+  const size_t lds_bytes = (size_t)8 * n_bins * 2 * sizeof(long long);
+  hipLaunchKernelGGL(csr_hist_fixed_lds_k, dim3((unsigned)bx, (unsigned)by),
+                     dim3(256), lds_bytes, stream, indptr, col, binv, gq,
+                     hq, rows, m, hist, n_bins, nf, tot);
+}
