@@ -135,6 +135,11 @@ class _GBDTParams(Params):
     matrixType = Param("matrixType", "auto|dense|sparse — sparse trains from "
                        "binned CSR without densifying (auto samples the "
                        "column like DatasetUtils.scala:49)", "auto", toString)
+    checkpointDir = Param("checkpointDir", "directory for iteration-level "
+                          "checkpoints (elastic restart resumes from it)",
+                          None)
+    checkpointInterval = Param("checkpointInterval", "write a checkpoint "
+                               "every k iterations (0 = off)", 0, toInt)
     device = Param("device", "cpu|cuda|auto", "auto", toString)
 
     def _train_config(self) -> TrainConfig:
@@ -271,7 +276,9 @@ class _GBDTEstimatorBase(_GBDTParams, Estimator):
                 weight=w, group_sizes=group, feature_names=names,
                 valid_sets=valid_sets, init_booster=booster,
                 metrics_fn=default_metrics_fn(self.get("metric") or None),
-                init_score=init_score)
+                init_score=init_score,
+                checkpoint_dir=self.get("checkpointDir"),
+                checkpoint_every=self.get("checkpointInterval"))
         model = self._model_class()(booster=booster)
         for p in ("labelCol", "featuresCol", "featureCols", "predictionCol",
                   "leafPredictionCol", "featuresShapCol"):

@@ -901,6 +901,30 @@ class TrainingSession:
                        tree_weights=np.array(tw, dtype=np.float32))
 
 
+def load_checkpoint(checkpoint_dir: str):
+    """(booster, completed_iterations) from an iteration-level checkpoint,
+    or None.  The elastic-restart resume point (the reference's recovery
+    story is Spark barrier gang restart + modelString warm start,
+    LightGBMBase.scala:46-61; here a restarted gang resumes mid-training)."""
+    path = os.path.join(checkpoint_dir, "checkpoint.json")
+    if not os.path.exists(path):
+        return None
+    import json as _json
+    with open(path) as f:
+        d = _json.load(f)
+    return Booster.load_from_string(d["model"]), int(d["iteration"])
+
+
+def _save_checkpoint(checkpoint_dir: str, booster: Booster, iteration: int):
+    import json as _json
+    os.makedirs(checkpoint_dir, exist_ok=True)
+    tmp = os.path.join(checkpoint_dir, ".checkpoint.tmp")
+    with open(tmp, "w") as f:
+        _json.dump({"iteration": iteration,
+                    "model": booster.save_to_string()}, f)
+    os.replace(tmp, os.path.join(checkpoint_dir, "checkpoint.json"))
+
+
 def train_booster(X: torch.Tensor, y: torch.Tensor, cfg: TrainConfig,
                   objective: Objective, comm: Comm,
                   weight: Optional[torch.Tensor] = None,
@@ -909,9 +933,20 @@ def train_booster(X: torch.Tensor, y: torch.Tensor, cfg: TrainConfig,
                   valid_sets: Optional[List[tuple]] = None,
                   init_booster: Optional[Booster] = None,
                   metrics_fn=None,
-                  binned_cache=None, init_score=None) -> (Booster, TrainingStats):
-    """Full training loop with eval + early stopping over TrainingSession."""
+                  binned_cache=None, init_score=None,
+                  checkpoint_dir: Optional[str] = None,
+                  checkpoint_every: int = 0) -> (Booster, TrainingStats):
+    """Full training loop with eval + early stopping over TrainingSession.
+
+    With checkpoint_dir + checkpoint_every, rank 0 writes an atomic
+    iteration-level checkpoint every k iterations and a fresh call resumes
+    from it — the fault-tolerance piece a restarted gang needs."""
     t_start = time.perf_counter()
+    start_it = 0
+    if checkpoint_dir:
+        ck = load_checkpoint(checkpoint_dir)
+        if ck is not None:
+            init_booster, start_it = ck
     session = TrainingSession(X, y, cfg, objective, comm, weight=weight,
                               group_sizes=group_sizes,
                               feature_names=feature_names,
@@ -923,8 +958,11 @@ def train_booster(X: torch.Tensor, y: torch.Tensor, cfg: TrainConfig,
     best_iter = -1
     rounds_no_improve = 0
 
-    for it in range(cfg.num_iterations):
+    for it in range(start_it, cfg.num_iterations):
         session.step()
+        if (checkpoint_dir and checkpoint_every > 0
+                and (it + 1) % checkpoint_every == 0 and comm.rank == 0):
+            _save_checkpoint(checkpoint_dir, session.booster(), it + 1)
         if metrics_fn is not None and (valid_sets or cfg.is_provide_training_metric):
             t0 = time.perf_counter()
             booster_now = session.booster()
