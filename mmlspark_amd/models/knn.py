@@ -62,18 +62,35 @@ class KNNModel(_KNNParamsMixin, Model):
                 payload["labels"] = np.asarray(labels)
             self.set("indexData", {k: v for k, v in payload.items()})
 
-    def _search(self, Q: np.ndarray, cond_masks=None):
+    def _search(self, Q: np.ndarray, cond_masks=None, cond_bits=None,
+                label_ids=None):
+        """Tiled MFMA-GEMM distances + device top-k.
+
+        Conditioning: either host bool masks (small corpora), or — the
+        large-corpus path — per-query uint64 label bitmasks tested on
+        device against the index's label-id vector, so no (n_q, n_index)
+        host matrix ever exists."""
         data = self.get("indexData")
         device = default_device("auto")
         X = torch.from_numpy(data["index"]).to(device)
         xsq = (X * X).sum(dim=1)
         k = min(self.get("k"), X.shape[0])
-        bs = self.get("batchSize")
+        # memory-aware batch clamp: the (bs, n) distance tile (plus the
+        # transient mask tile) must stay bounded as the corpus grows
+        bs = min(self.get("batchSize"),
+                 max(128, int((1 << 31) / max(1, X.shape[0]))))
+        lid = (torch.from_numpy(label_ids).to(device)
+               if label_ids is not None else None)
         all_idx, all_dist = [], []
         for s in range(0, len(Q), bs):
             q = torch.from_numpy(Q[s:s + bs]).to(device)
             d2 = (q * q).sum(1, keepdim=True) - 2.0 * (q @ X.t()) + xsq
-            if cond_masks is not None:
+            if cond_bits is not None:
+                bits = torch.from_numpy(cond_bits[s:s + bs]).to(device)
+                allowed = ((bits.unsqueeze(1) >> lid.unsqueeze(0)) & 1).bool()
+                d2 = torch.where(allowed, d2,
+                                 torch.full_like(d2, float("inf")))
+            elif cond_masks is not None:
                 m = torch.from_numpy(cond_masks[s:s + bs]).to(device)
                 d2 = torch.where(m, d2, torch.full_like(d2, float("inf")))
             dist, idx = torch.topk(d2, k, dim=1, largest=False)
@@ -130,13 +147,32 @@ class ConditionalKNNModel(KNNModel):
         labels = data["labels"]
         vals = data.get("values")
         conds = df[self.get("conditionerCol")].to_numpy()
-        masks = np.zeros((len(df), len(labels)), dtype=bool)
-        for i, c in enumerate(conds):
-            allowed = set(c) if isinstance(c, (list, tuple, set, np.ndarray)) \
-                else {c}
-            masks[i] = np.isin(labels, list(allowed))
-        idx, dist = self._search(Q, masks) if len(df) else (np.zeros((0, 0), int),
-                                                            np.zeros((0, 0)))
+        uniq, label_ids = np.unique(labels, return_inverse=True)
+        if len(df) == 0:
+            idx, dist = np.zeros((0, 0), int), np.zeros((0, 0))
+        elif len(uniq) <= 64:
+            # large-corpus path: per-query uint64 allowed-label bitmask,
+            # tested on device — no (n_q, n_index) host matrix
+            pos = {v: i for i, v in enumerate(uniq)}
+            bits = np.zeros(len(df), dtype=np.int64)
+            for i, c in enumerate(conds):
+                allowed = (c if isinstance(c, (list, tuple, set, np.ndarray))
+                           else [c])
+                b = 0
+                for v in allowed:
+                    j = pos.get(v)
+                    if j is not None:
+                        b |= (1 << j)
+                bits[i] = b
+            idx, dist = self._search(Q, cond_bits=bits,
+                                     label_ids=label_ids.astype(np.int64))
+        else:
+            masks = np.zeros((len(df), len(labels)), dtype=bool)
+            for i, c in enumerate(conds):
+                allowed = set(c) if isinstance(
+                    c, (list, tuple, set, np.ndarray)) else {c}
+                masks[i] = np.isin(labels, list(allowed))
+            idx, dist = self._search(Q, masks)
         matches = []
         for i in range(len(df)):
             row = []

@@ -42,6 +42,30 @@ def bench_knn():
                       n_query / dt, "n_index": n_index, "dim": dim, "k": k}))
 
 
+def bench_conditional_knn():
+    """conditional (label-filtered) search at a corpus
+    size where the reference's broadcast ball tree would not fit a UDF —
+    2M x d128, 16 labels, device bitmask filtering."""
+    from mmlspark_amd.models.knn import ConditionalKNN
+    rng = np.random.default_rng(1)
+    n_index, n_query, dim, k = 2_000_000, 20_000, 128, 10
+    idx_df = pd.DataFrame({
+        "features": list(rng.normal(size=(n_index, dim)).astype(np.float32)),
+        "labels": rng.integers(0, 16, size=n_index),
+    })
+    q = pd.DataFrame({
+        "features": list(rng.normal(size=(n_query, dim)).astype(np.float32)),
+        "conditioner": [list(rng.choice(16, size=3, replace=False))
+                        for _ in range(n_query)],
+    })
+    model = ConditionalKNN(k=k, labelCol="labels").fit(idx_df)
+    dt = timeit(lambda: model.transform(q), warmup=1, iters=3)
+    print(json.dumps({"bench": "conditional_knn_bitmask",
+                      "queries_per_sec": n_query / dt,
+                      "n_index": n_index, "dim": dim, "k": k,
+                      "n_labels": 16}))
+
+
 def bench_iforest():
     from mmlspark_amd.models.iforest import IsolationForest
     rng = np.random.default_rng(1)
@@ -130,7 +154,9 @@ def bench_serving_concurrent():
 
 if __name__ == "__main__":
     which = sys.argv[1] if len(sys.argv) > 1 else "all"
-    for name, fn in [("knn", bench_knn), ("iforest", bench_iforest),
+    for name, fn in [("knn", bench_knn),
+                     ("cknn", bench_conditional_knn),
+                     ("iforest", bench_iforest),
                      ("sar", bench_sar),
                      ("serving", bench_serving_concurrent)]:
         if which in ("all", name):
