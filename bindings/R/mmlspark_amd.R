@@ -905,7 +905,7 @@ ml_light_gbm_classification_model <- function(labelCol = NULL, featuresCol = NUL
   stage
 }
 
-ml_light_gbm_classifier <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, categoricalSlotNames = NULL, slotNames = NULL, modelString = NULL, lightGBMBooster = NULL, fobj = NULL, boostFromAverage = NULL, improvementTolerance = NULL, posBaggingFraction = NULL, negBaggingFraction = NULL, binSampleCount = NULL, maxBinByFeature = NULL, uniformDrop = NULL, xgboostDartMode = NULL, startIteration = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, chunkSize = NULL, defaultListenPort = NULL, driverListenPort = NULL, timeout = NULL, numTasks = NULL, numThreads = NULL, useSingleDatasetMode = NULL, matrixType = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL, isUnbalance = NULL) {
+ml_light_gbm_classifier <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, categoricalSlotNames = NULL, slotNames = NULL, modelString = NULL, lightGBMBooster = NULL, fobj = NULL, boostFromAverage = NULL, improvementTolerance = NULL, posBaggingFraction = NULL, negBaggingFraction = NULL, binSampleCount = NULL, maxBinByFeature = NULL, uniformDrop = NULL, xgboostDartMode = NULL, startIteration = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, chunkSize = NULL, defaultListenPort = NULL, driverListenPort = NULL, timeout = NULL, numTasks = NULL, numThreads = NULL, useSingleDatasetMode = NULL, matrixType = NULL, checkpointDir = NULL, checkpointInterval = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL, isUnbalance = NULL) {
   stage <- mmlspark_amd$models$gbdt$estimators$LightGBMClassifier()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -970,6 +970,8 @@ ml_light_gbm_classifier <- function(labelCol = NULL, featuresCol = NULL, feature
   if (!is.null(numThreads)) stage$set("numThreads", numThreads)
   if (!is.null(useSingleDatasetMode)) stage$set("useSingleDatasetMode", useSingleDatasetMode)
   if (!is.null(matrixType)) stage$set("matrixType", matrixType)
+  if (!is.null(checkpointDir)) stage$set("checkpointDir", checkpointDir)
+  if (!is.null(checkpointInterval)) stage$set("checkpointInterval", checkpointInterval)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(rawPredictionCol)) stage$set("rawPredictionCol", rawPredictionCol)
   if (!is.null(probabilityCol)) stage$set("probabilityCol", probabilityCol)
@@ -977,7 +979,7 @@ ml_light_gbm_classifier <- function(labelCol = NULL, featuresCol = NULL, feature
   stage
 }
 
-ml_light_gbm_ranker <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, categoricalSlotNames = NULL, slotNames = NULL, modelString = NULL, lightGBMBooster = NULL, fobj = NULL, boostFromAverage = NULL, improvementTolerance = NULL, posBaggingFraction = NULL, negBaggingFraction = NULL, binSampleCount = NULL, maxBinByFeature = NULL, uniformDrop = NULL, xgboostDartMode = NULL, startIteration = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, chunkSize = NULL, defaultListenPort = NULL, driverListenPort = NULL, timeout = NULL, numTasks = NULL, numThreads = NULL, useSingleDatasetMode = NULL, matrixType = NULL, device = NULL, groupCol = NULL, labelGain = NULL, maxPosition = NULL, evalAt = NULL) {
+ml_light_gbm_ranker <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, categoricalSlotNames = NULL, slotNames = NULL, modelString = NULL, lightGBMBooster = NULL, fobj = NULL, boostFromAverage = NULL, improvementTolerance = NULL, posBaggingFraction = NULL, negBaggingFraction = NULL, binSampleCount = NULL, maxBinByFeature = NULL, uniformDrop = NULL, xgboostDartMode = NULL, startIteration = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, chunkSize = NULL, defaultListenPort = NULL, driverListenPort = NULL, timeout = NULL, numTasks = NULL, numThreads = NULL, useSingleDatasetMode = NULL, matrixType = NULL, checkpointDir = NULL, checkpointInterval = NULL, device = NULL, groupCol = NULL, labelGain = NULL, maxPosition = NULL, evalAt = NULL) {
   stage <- mmlspark_amd$models$gbdt$estimators$LightGBMRanker()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1042,6 +1044,8 @@ ml_light_gbm_ranker <- function(labelCol = NULL, featuresCol = NULL, featureCols
   if (!is.null(numThreads)) stage$set("numThreads", numThreads)
   if (!is.null(useSingleDatasetMode)) stage$set("useSingleDatasetMode", useSingleDatasetMode)
   if (!is.null(matrixType)) stage$set("matrixType", matrixType)
+  if (!is.null(checkpointDir)) stage$set("checkpointDir", checkpointDir)
+  if (!is.null(checkpointInterval)) stage$set("checkpointInterval", checkpointInterval)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(groupCol)) stage$set("groupCol", groupCol)
   if (!is.null(labelGain)) stage$set("labelGain", labelGain)
@@ -1080,7 +1084,7 @@ ml_light_gbm_regression_model <- function(labelCol = NULL, featuresCol = NULL, f
   stage
 }
 
-ml_light_gbm_regressor <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, categoricalSlotNames = NULL, slotNames = NULL, modelString = NULL, lightGBMBooster = NULL, fobj = NULL, boostFromAverage = NULL, improvementTolerance = NULL, posBaggingFraction = NULL, negBaggingFraction = NULL, binSampleCount = NULL, maxBinByFeature = NULL, uniformDrop = NULL, xgboostDartMode = NULL, startIteration = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, chunkSize = NULL, defaultListenPort = NULL, driverListenPort = NULL, timeout = NULL, numTasks = NULL, numThreads = NULL, useSingleDatasetMode = NULL, matrixType = NULL, device = NULL, alpha = NULL, tweedieVariancePower = NULL) {
+ml_light_gbm_regressor <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, categoricalSlotNames = NULL, slotNames = NULL, modelString = NULL, lightGBMBooster = NULL, fobj = NULL, boostFromAverage = NULL, improvementTolerance = NULL, posBaggingFraction = NULL, negBaggingFraction = NULL, binSampleCount = NULL, maxBinByFeature = NULL, uniformDrop = NULL, xgboostDartMode = NULL, startIteration = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, chunkSize = NULL, defaultListenPort = NULL, driverListenPort = NULL, timeout = NULL, numTasks = NULL, numThreads = NULL, useSingleDatasetMode = NULL, matrixType = NULL, checkpointDir = NULL, checkpointInterval = NULL, device = NULL, alpha = NULL, tweedieVariancePower = NULL) {
   stage <- mmlspark_amd$models$gbdt$estimators$LightGBMRegressor()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1145,6 +1149,8 @@ ml_light_gbm_regressor <- function(labelCol = NULL, featuresCol = NULL, featureC
   if (!is.null(numThreads)) stage$set("numThreads", numThreads)
   if (!is.null(useSingleDatasetMode)) stage$set("useSingleDatasetMode", useSingleDatasetMode)
   if (!is.null(matrixType)) stage$set("matrixType", matrixType)
+  if (!is.null(checkpointDir)) stage$set("checkpointDir", checkpointDir)
+  if (!is.null(checkpointInterval)) stage$set("checkpointInterval", checkpointInterval)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(alpha)) stage$set("alpha", alpha)
   if (!is.null(tweedieVariancePower)) stage$set("tweedieVariancePower", tweedieVariancePower)
@@ -1907,7 +1913,7 @@ ml_verify_faces <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyC
   stage
 }
 
-ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, maxIterBfgs = NULL, device = NULL, weightsArrays = NULL, rawPredictionCol = NULL, probabilityCol = NULL) {
+ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, holdoutOff = NULL, holdoutPeriod = NULL, earlyTerminate = NULL, maxIterBfgs = NULL, device = NULL, weightsArrays = NULL, rawPredictionCol = NULL, probabilityCol = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitClassificationModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1934,6 +1940,9 @@ ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol =
   if (!is.null(ignoreNamespaces)) stage$set("ignoreNamespaces", ignoreNamespaces)
   if (!is.null(useBarrierExecutionMode)) stage$set("useBarrierExecutionMode", useBarrierExecutionMode)
   if (!is.null(bfgs)) stage$set("bfgs", bfgs)
+  if (!is.null(holdoutOff)) stage$set("holdoutOff", holdoutOff)
+  if (!is.null(holdoutPeriod)) stage$set("holdoutPeriod", holdoutPeriod)
+  if (!is.null(earlyTerminate)) stage$set("earlyTerminate", earlyTerminate)
   if (!is.null(maxIterBfgs)) stage$set("maxIterBfgs", maxIterBfgs)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(weightsArrays)) stage$set("weightsArrays", weightsArrays)
@@ -1942,7 +1951,7 @@ ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol =
   stage
 }
 
-ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, maxIterBfgs = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL, labelConversion = NULL) {
+ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, holdoutOff = NULL, holdoutPeriod = NULL, earlyTerminate = NULL, maxIterBfgs = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL, labelConversion = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitClassifier()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1969,6 +1978,9 @@ ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, add
   if (!is.null(ignoreNamespaces)) stage$set("ignoreNamespaces", ignoreNamespaces)
   if (!is.null(useBarrierExecutionMode)) stage$set("useBarrierExecutionMode", useBarrierExecutionMode)
   if (!is.null(bfgs)) stage$set("bfgs", bfgs)
+  if (!is.null(holdoutOff)) stage$set("holdoutOff", holdoutOff)
+  if (!is.null(holdoutPeriod)) stage$set("holdoutPeriod", holdoutPeriod)
+  if (!is.null(earlyTerminate)) stage$set("earlyTerminate", earlyTerminate)
   if (!is.null(maxIterBfgs)) stage$set("maxIterBfgs", maxIterBfgs)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(rawPredictionCol)) stage$set("rawPredictionCol", rawPredictionCol)
@@ -1977,7 +1989,7 @@ ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, add
   stage
 }
 
-ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, maxIterBfgs = NULL, device = NULL, sharedCol = NULL, chosenActionCol = NULL, probabilityCol = NULL, epsilon = NULL) {
+ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, holdoutOff = NULL, holdoutPeriod = NULL, earlyTerminate = NULL, maxIterBfgs = NULL, device = NULL, sharedCol = NULL, chosenActionCol = NULL, probabilityCol = NULL, epsilon = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitContextualBandit()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -2004,6 +2016,9 @@ ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NU
   if (!is.null(ignoreNamespaces)) stage$set("ignoreNamespaces", ignoreNamespaces)
   if (!is.null(useBarrierExecutionMode)) stage$set("useBarrierExecutionMode", useBarrierExecutionMode)
   if (!is.null(bfgs)) stage$set("bfgs", bfgs)
+  if (!is.null(holdoutOff)) stage$set("holdoutOff", holdoutOff)
+  if (!is.null(holdoutPeriod)) stage$set("holdoutPeriod", holdoutPeriod)
+  if (!is.null(earlyTerminate)) stage$set("earlyTerminate", earlyTerminate)
   if (!is.null(maxIterBfgs)) stage$set("maxIterBfgs", maxIterBfgs)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(sharedCol)) stage$set("sharedCol", sharedCol)
@@ -2013,7 +2028,7 @@ ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NU
   stage
 }
 
-ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, maxIterBfgs = NULL, device = NULL, weightsArrays = NULL, sharedCol = NULL, epsilon = NULL) {
+ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, holdoutOff = NULL, holdoutPeriod = NULL, earlyTerminate = NULL, maxIterBfgs = NULL, device = NULL, weightsArrays = NULL, sharedCol = NULL, epsilon = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitContextualBanditModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -2040,6 +2055,9 @@ ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCo
   if (!is.null(ignoreNamespaces)) stage$set("ignoreNamespaces", ignoreNamespaces)
   if (!is.null(useBarrierExecutionMode)) stage$set("useBarrierExecutionMode", useBarrierExecutionMode)
   if (!is.null(bfgs)) stage$set("bfgs", bfgs)
+  if (!is.null(holdoutOff)) stage$set("holdoutOff", holdoutOff)
+  if (!is.null(holdoutPeriod)) stage$set("holdoutPeriod", holdoutPeriod)
+  if (!is.null(earlyTerminate)) stage$set("earlyTerminate", earlyTerminate)
   if (!is.null(maxIterBfgs)) stage$set("maxIterBfgs", maxIterBfgs)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(weightsArrays)) stage$set("weightsArrays", weightsArrays)
@@ -2070,7 +2088,7 @@ ml_vowpal_wabbit_interactions <- function(inputCols = NULL, outputCol = NULL, nu
   stage
 }
 
-ml_vowpal_wabbit_regressor <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, maxIterBfgs = NULL, device = NULL) {
+ml_vowpal_wabbit_regressor <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, holdoutOff = NULL, holdoutPeriod = NULL, earlyTerminate = NULL, maxIterBfgs = NULL, device = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitRegressor()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -2097,12 +2115,15 @@ ml_vowpal_wabbit_regressor <- function(labelCol = NULL, featuresCol = NULL, addi
   if (!is.null(ignoreNamespaces)) stage$set("ignoreNamespaces", ignoreNamespaces)
   if (!is.null(useBarrierExecutionMode)) stage$set("useBarrierExecutionMode", useBarrierExecutionMode)
   if (!is.null(bfgs)) stage$set("bfgs", bfgs)
+  if (!is.null(holdoutOff)) stage$set("holdoutOff", holdoutOff)
+  if (!is.null(holdoutPeriod)) stage$set("holdoutPeriod", holdoutPeriod)
+  if (!is.null(earlyTerminate)) stage$set("earlyTerminate", earlyTerminate)
   if (!is.null(maxIterBfgs)) stage$set("maxIterBfgs", maxIterBfgs)
   if (!is.null(device)) stage$set("device", device)
   stage
 }
 
-ml_vowpal_wabbit_regressor_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, maxIterBfgs = NULL, device = NULL, weightsArrays = NULL) {
+ml_vowpal_wabbit_regressor_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, holdoutOff = NULL, holdoutPeriod = NULL, earlyTerminate = NULL, maxIterBfgs = NULL, device = NULL, weightsArrays = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitRegressorModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -2129,6 +2150,9 @@ ml_vowpal_wabbit_regressor_model <- function(labelCol = NULL, featuresCol = NULL
   if (!is.null(ignoreNamespaces)) stage$set("ignoreNamespaces", ignoreNamespaces)
   if (!is.null(useBarrierExecutionMode)) stage$set("useBarrierExecutionMode", useBarrierExecutionMode)
   if (!is.null(bfgs)) stage$set("bfgs", bfgs)
+  if (!is.null(holdoutOff)) stage$set("holdoutOff", holdoutOff)
+  if (!is.null(holdoutPeriod)) stage$set("holdoutPeriod", holdoutPeriod)
+  if (!is.null(earlyTerminate)) stage$set("earlyTerminate", earlyTerminate)
   if (!is.null(maxIterBfgs)) stage$set("maxIterBfgs", maxIterBfgs)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(weightsArrays)) stage$set("weightsArrays", weightsArrays)

@@ -184,10 +184,12 @@ def decode_jpeg(data: bytes) -> np.ndarray:
     try:
         import torch  # noqa: F401 — loads libc10 for the extension
         from . import _jpeg_native
-        return _jpeg_native.decode_jpeg(bytes(data)).numpy()
     except ImportError:
-        pass
-    return _decode_jpeg_py(data)
+        return _decode_jpeg_py(data)
+    try:
+        return _jpeg_native.decode_jpeg(bytes(data)).numpy()
+    except RuntimeError as e:  # same error contract as the Python decoder
+        raise ValueError(str(e)) from e
 
 
 def _decode_jpeg_py(data: bytes) -> np.ndarray:
