@@ -55,13 +55,12 @@ def test_r_file_structural_lint(r_source):
 
 
 def test_r_functions_cover_registry(r_source):
+    import mmlspark_amd as pkg
+    pkg._register_all()  # full registry, independent of test import order
+    from mmlspark_amd.core.codegen import _snake
     fns = _functions(r_source)
-
-    def snake(n):
-        return re.sub(r"(?<!^)(?=[A-Z])", "_", n).lower()
-
     missing = [n for n in all_stages()
-               if f"ml_{snake(n)}" not in fns]
+               if f"ml_{_snake(n)}" not in fns]
     assert not missing, missing[:10]
 
 
