@@ -607,11 +607,13 @@ class SparseTreeGrower(TreeGrower):
     def _partition(self, leaf) -> tuple:
         t0 = time.perf_counter()
         sh = self.shard
-        bins = backend.csr_gather_bins(
+        # counts above 2^24 lose exactness through the f32 scan
+        known = (int(leaf.CL) if (not self.comm.is_distributed
+                                  and leaf.C < 1.6e7)
+                 else -1)
+        rows_l, rows_r = backend.csr_partition_rows(
             sh.indptr, sh.col, sh.binv, leaf.rows, leaf.feat,
-            int(sh.zero_bin[leaf.feat]))
-        mask = bins <= leaf.bin
-        rows_l, rows_r = leaf.rows[mask], leaf.rows[~mask]
+            int(sh.zero_bin[leaf.feat]), leaf.bin, known_left=known)
         self.stats.partition_s += time.perf_counter() - t0
         return rows_l, rows_r
 

@@ -187,3 +187,17 @@ def csr_hist_fixed_tot(indptr, col, binv, gq, hq, rows, nf, n_bins):
     tot = torch.stack([gq[r].sum(), hq[r].sum(),
                        torch.tensor(int(rows.numel()), dtype=torch.int64)])
     return h, tot
+
+
+def csr_partition_rows(indptr, col, binv, rows, feature, zero_bin,
+                       threshold_bin, known_left=-1):
+    """Stable ordered partition of a CSR shard's row list by one feature's
+    bin (missing → zero_bin): fused predicate + 3-kernel partition on GPU,
+    gather + boolean masks on CPU."""
+    if binv.is_cuda:
+        return _require_ext().csr_partition_rows(
+            indptr, col, binv, rows.contiguous(), feature, zero_bin,
+            threshold_bin, known_left)
+    bins = cpu_ref.csr_gather_bins(indptr, col, binv, rows, feature, zero_bin)
+    mask = bins <= threshold_bin
+    return rows[mask], rows[~mask]

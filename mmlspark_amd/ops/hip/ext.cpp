@@ -54,6 +54,9 @@ void launch_csr_hist_fixed_lds(const long*, const int*, const unsigned char*,
                                const long long*, const long long*,
                                const int*, long, long long*, int, int,
                                long long*, hipStream_t);
+void launch_csr_partition(const long*, const int*, const unsigned char*,
+                          const int*, long, int, int, int, const unsigned*,
+                          int*, int*, int*, hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -374,6 +377,29 @@ torch::Tensor csr_gather_bins(torch::Tensor indptr, torch::Tensor col,
   return out;
 }
 
+std::tuple<torch::Tensor, torch::Tensor> csr_partition_rows(
+    torch::Tensor indptr, torch::Tensor col, torch::Tensor binv,
+    torch::Tensor rows, long feature, long zero_bin, long thr,
+    long known_left) {
+  CHECK_DEV(indptr); CHECK_CONTIG(indptr);
+  CHECK_DEV(col); CHECK_CONTIG(col);
+  CHECK_DEV(binv); CHECK_CONTIG(binv);
+  CHECK_DEV(rows); CHECK_CONTIG(rows);
+  const long m = rows.numel();
+  auto out = torch::empty({m}, rows.options());
+  auto scratch = torch::empty({4096}, rows.options());
+  auto total = torch::zeros({1}, rows.options());
+  if (m > 0) {
+    launch_csr_partition(indptr.data_ptr<long>(), col.data_ptr<int>(),
+                         binv.data_ptr<unsigned char>(), rows.data_ptr<int>(),
+                         m, (int)feature, (int)zero_bin, (int)thr, nullptr,
+                         out.data_ptr<int>(), scratch.data_ptr<int>(),
+                         total.data_ptr<int>(), cur_stream());
+  }
+  const long nl = known_left >= 0 ? known_left : total.item<int>();
+  return {out.slice(0, 0, nl), out.slice(0, nl, m)};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hist_build", &hist_build, "per-leaf (feature,bin) grad/hess/count histogram");
   m.def("hist_build_fixed_pair", &hist_build_fixed_pair,
@@ -393,6 +419,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fixed-point histogram over stored CSR entries");
   m.def("csr_gather_bins", &csr_gather_bins,
         "per-row bin of one feature from CSR (missing -> zero bin)");
+  m.def("csr_partition_rows", &csr_partition_rows,
+        "fused CSR leaf partition (binary-search predicate, single sync)");
   m.def("csr_hist_fixed_tot", &csr_hist_fixed_tot,
         "wave-cooperative CSR histogram + in-kernel leaf totals");
 }

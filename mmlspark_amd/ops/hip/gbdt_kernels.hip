@@ -1694,3 +1694,139 @@ extern "C" void launch_csr_hist_fixed_lds(const long* indptr, const int* col,
                      dim3(256), lds_bytes, stream, indptr, col, binv, gq,
                      hq, rows, m, hist, n_bins, nf, tot);
 }
+
+// Fused CSR leaf partition: predicate = binary-searched bin of the split
+// feature (missing → zero bin) compared against thr / a category bitset —
+// replaces csr_gather + torch boolean masks (two nonzero passes + gathers)
+// with the same 3-kernel stable ordered partition the dense path uses.
+DEV_INLINE int csr_bin_of(const long* indptr, const int* col,
+                          const unsigned char* binv, int r, int feature,
+                          int zero_bin) {
+  long lo = indptr[r], hi = indptr[r + 1];
+  while (lo < hi) {
+    const long mid = (lo + hi) >> 1;
+    const int c = col[mid];
+    if (c == feature) return binv[mid];
+    if (c < feature) lo = mid + 1; else hi = mid;
+  }
+  return zero_bin;
+}
+
+DEV_INLINE bool csr_goes_left(const long* indptr, const int* col,
+                              const unsigned char* binv, int r, int feature,
+                              int zero_bin, int thr,
+                              const unsigned* cat_bits) {
+  const int b = csr_bin_of(indptr, col, binv, r, feature, zero_bin);
+  if (cat_bits) return ((cat_bits[b >> 5] >> (b & 31)) & 1u) != 0u;
+  return b <= thr;
+}
+
+__global__ void csr_part_count_k(const long* __restrict__ indptr,
+                                 const int* __restrict__ col,
+                                 const unsigned char* __restrict__ binv,
+                                 const int* __restrict__ rows, long m,
+                                 int feature, int zero_bin, int thr,
+                                 const unsigned* __restrict__ cat_bits,
+                                 long chunk, int* __restrict__ block_counts) {
+  const long start = (long)blockIdx.x * chunk;
+  const long end = min(start + chunk, m);
+  int cnt = 0;
+  for (long i = start + threadIdx.x; i < end; i += blockDim.x)
+    cnt += (int)csr_goes_left(indptr, col, binv, rows[i], feature, zero_bin,
+                              thr, cat_bits);
+  __shared__ int sh[256];
+  sh[threadIdx.x] = cnt;
+  __syncthreads();
+  for (int d = 128; d > 0; d >>= 1) {
+    if (threadIdx.x < d) sh[threadIdx.x] += sh[threadIdx.x + d];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) block_counts[blockIdx.x] = sh[0];
+}
+
+__global__ void csr_part_scatter_k(const long* __restrict__ indptr,
+                                   const int* __restrict__ col,
+                                   const unsigned char* __restrict__ binv,
+                                   const int* __restrict__ rows, long m,
+                                   int feature, int zero_bin, int thr,
+                                   const unsigned* __restrict__ cat_bits,
+                                   long chunk,
+                                   const int* __restrict__ block_offsets,
+                                   const int* __restrict__ total_left,
+                                   int* __restrict__ out) {
+  const long start = (long)blockIdx.x * chunk;
+  const long end = min(start + chunk, m);
+  const long nl_total = *total_left;
+  __shared__ long base_l, base_r;
+  __shared__ int wave_l[4], wave_r[4];
+  if (threadIdx.x == 0) {
+    base_l = block_offsets[blockIdx.x];
+    base_r = nl_total + (start - block_offsets[blockIdx.x]);
+  }
+  __syncthreads();
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  for (long i0 = start; i0 < end; i0 += blockDim.x) {
+    const long i = i0 + threadIdx.x;
+    int r = 0;
+    bool valid = i < end, left = false;
+    if (valid) {
+      r = rows[i];
+      left = csr_goes_left(indptr, col, binv, r, feature, zero_bin, thr,
+                           cat_bits);
+    }
+    const unsigned long long mask_l = __ballot(valid && left);
+    const unsigned long long mask_r = __ballot(valid && !left);
+    const unsigned long long lt = (1ull << lane) - 1ull;
+    if (lane == 0) {
+      wave_l[wid] = __popcll(mask_l);
+      wave_r[wid] = __popcll(mask_r);
+    }
+    __syncthreads();
+    long wl = base_l, wr = base_r;
+    for (int w = 0; w < wid; ++w) {
+      wl += wave_l[w];
+      wr += wave_r[w];
+    }
+    if (valid) {
+      if (left) out[wl + __popcll(mask_l & lt)] = r;
+      else out[wr + __popcll(mask_r & lt)] = r;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      long al = 0, ar = 0;
+      for (int w = 0; w < 4; ++w) {
+        al += wave_l[w];
+        ar += wave_r[w];
+      }
+      base_l += al;
+      base_r += ar;
+    }
+    __syncthreads();
+  }
+}
+
+extern "C" void launch_csr_partition(const long* indptr, const int* col,
+                                     const unsigned char* binv,
+                                     const int* rows, long m, int feature,
+                                     int zero_bin, int thr,
+                                     const unsigned* cat_bits, int* out,
+                                     int* scratch, int* total_left,
+                                     hipStream_t stream) {
+  if (m == 0) return;
+  long chunk = 4096;
+  long blocks = (m + chunk - 1) / chunk;
+  if (blocks > 4096) {
+    chunk = (m + 4095) / 4096;
+    blocks = (m + chunk - 1) / chunk;
+  }
+  hipLaunchKernelGGL(csr_part_count_k, dim3((unsigned)blocks), dim3(256), 0,
+                     stream, indptr, col, binv, rows, m, feature, zero_bin,
+                     thr, cat_bits, chunk, scratch);
+  hipLaunchKernelGGL(part_scan_k, dim3(1), dim3(256), 0, stream, scratch,
+                     (int)blocks, total_left);
+  hipLaunchKernelGGL(csr_part_scatter_k, dim3((unsigned)blocks), dim3(256),
+                     0, stream, indptr, col, binv, rows, m, feature,
+                     zero_bin, thr, cat_bits, chunk, scratch, total_left,
+                     out);
+}

@@ -159,3 +159,28 @@ def test_csr_hist_v2_fused_totals_matches_cpu():
             rows.cuda(), nf, nb)
         assert torch.equal(out.cpu(), ref), (n, nf)
         assert torch.equal(tot.cpu(), ref_tot), (n, tot.cpu(), ref_tot)
+
+
+@requires_gpu
+def test_csr_partition_rows_kernel_matches_cpu():
+    """Fused CSR partition (predicate + stable ordered split) == the CPU
+    gather+mask reference, including the known-left no-sync path."""
+    from mmlspark_amd.ops import backend
+    gen = torch.Generator().manual_seed(21)
+    n, nf, nb = 70_000, 150, 63
+    indptr, col, binv = _rand_csr(gen, n, nf, 9)
+    binv = (binv % nb).contiguous()
+    rows = torch.randperm(n, generator=gen)[: n // 2].to(
+        torch.int32).sort().values
+    for f, zb, thr in ((3, 5, 20), (149, 0, 0), (80, 31, 62)):
+        ref_l, ref_r = backend.csr_partition_rows(
+            indptr, col, binv, rows, f, zb, thr)
+        got_l, got_r = backend.csr_partition_rows(
+            indptr.cuda(), col.cuda(), binv.cuda(), rows.cuda(), f, zb, thr)
+        assert torch.equal(got_l.cpu(), ref_l), (f, thr)
+        assert torch.equal(got_r.cpu(), ref_r), (f, thr)
+        gl2, gr2 = backend.csr_partition_rows(
+            indptr.cuda(), col.cuda(), binv.cuda(), rows.cuda(), f, zb, thr,
+            known_left=ref_l.numel())
+        assert torch.equal(gl2.cpu(), ref_l)
+        assert torch.equal(gr2.cpu(), ref_r)
