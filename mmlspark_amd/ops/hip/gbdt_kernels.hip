@@ -16,6 +16,7 @@
 //    near-coalesced.
 #include <hip/hip_runtime.h>
 #include <cstdint>
+#include <cstdlib>
 
 #define WAVE 64
 #define DEV_INLINE __device__ __forceinline__
@@ -1573,6 +1574,7 @@ extern "C" void launch_csr_hist_fixed_v2(const long* indptr, const int* col,
 // LDS cell.  Row staging + entry→row mapping as in v2.  Row count per
 // block must stay ≤ 2^19 so the 44-bit packed h-sum cannot overflow
 // (launcher enforces via grid sizing).
+template <int FW>  // feature-chunk width (8: 4 blocks/CU at nb=255; 16: 2)
 __global__ void csr_hist_fixed_lds_k(const long* __restrict__ indptr,
                                      const int* __restrict__ col,
                                      const unsigned char* __restrict__ binv,
@@ -1582,10 +1584,10 @@ __global__ void csr_hist_fixed_lds_k(const long* __restrict__ indptr,
                                      long long* __restrict__ hist,
                                      int n_bins, int nf,
                                      long long* __restrict__ tot) {
-  extern __shared__ unsigned long long lds64[];  // [8][n_bins][2]
-  const int f0 = blockIdx.y * 8;
+  extern __shared__ unsigned long long lds64[];  // [FW][n_bins][2]
+  const int f0 = blockIdx.y * FW;
   const int tid = threadIdx.x;
-  const int lds_elems = 8 * n_bins * 2;
+  const int lds_elems = FW * n_bins * 2;
   for (int i = tid; i < lds_elems; i += blockDim.x) lds64[i] = 0ull;
 
   __shared__ long s_start[4][64];
@@ -1636,7 +1638,7 @@ __global__ void csr_hist_fixed_lds_k(const long* __restrict__ indptr,
       }
       const long j = s_start[wid][lo2] + (e - cum[lo2]);
       const int c = col[j];
-      if (c >= f0 && c < f0 + 8) {
+      if (c >= f0 && c < f0 + FW) {
         unsigned long long* cell =
             &lds64[(((c - f0) * n_bins) + binv[j]) * 2];
         atomicAdd(cell + 0, (unsigned long long)s_g[wid][lo2]);
@@ -1647,7 +1649,7 @@ __global__ void csr_hist_fixed_lds_k(const long* __restrict__ indptr,
   }
 
   // flush LDS chunk to the global (nf, n_bins, 3) histogram
-  for (int i = tid; i < 8 * n_bins; i += blockDim.x) {
+  for (int i = tid; i < FW * n_bins; i += blockDim.x) {
     const int f = i / n_bins;
     if (f0 + f >= nf) break;
     const unsigned long long gsum = lds64[i * 2 + 0];
@@ -1688,11 +1690,26 @@ extern "C" void launch_csr_hist_fixed_lds(const long* indptr, const int* col,
   const long min_bx = (m + (1l << 19) - 1) >> 19;  // ≤2^19 rows per block
   if (bx < min_bx) bx = min_bx;
   if (bx < 1) bx = 1;
-  const int by = (nf + 7) / 8;
-  const size_t lds_bytes = (size_t)8 * n_bins * 2 * sizeof(long long);
-  hipLaunchKernelGGL(csr_hist_fixed_lds_k, dim3((unsigned)bx, (unsigned)by),
-                     dim3(256), lds_bytes, stream, indptr, col, binv, gq,
-                     hq, rows, m, hist, n_bins, nf, tot);
+  // A/B-measured on MI355X: FW=16 (half the redundant entry passes, half
+  // the LDS occupancy) ties FW=8 at 10M×100 nnz=20 — the kernel is
+  // LDS-atomic-bound, not read-bound — so default to the safer FW=8
+  static const int fw_env = [] {
+    const char* e = getenv("MMLSPARK_AMD_SPARSE_FW");
+    return e ? atoi(e) : 8;
+  }();
+  const int FW = (fw_env == 16) ? 16 : 8;
+  const int by = (nf + FW - 1) / FW;
+  const size_t lds_bytes = (size_t)FW * n_bins * 2 * sizeof(long long);
+  if (FW == 8)
+    hipLaunchKernelGGL(csr_hist_fixed_lds_k<8>,
+                       dim3((unsigned)bx, (unsigned)by), dim3(256),
+                       lds_bytes, stream, indptr, col, binv, gq, hq, rows,
+                       m, hist, n_bins, nf, tot);
+  else
+    hipLaunchKernelGGL(csr_hist_fixed_lds_k<16>,
+                       dim3((unsigned)bx, (unsigned)by), dim3(256),
+                       lds_bytes, stream, indptr, col, binv, gq, hq, rows,
+                       m, hist, n_bins, nf, tot);
 }
 
 // Fused CSR leaf partition: predicate = binary-searched bin of the split
