@@ -16,7 +16,7 @@ TIMEOUT=${CI_TIMEOUT:-1200}       # seconds per split (ref: 20-min sbt cap)
 declare -A SPLITS=(
   [core]="tests/test_core.py tests/test_io_codegen.py tests/test_interop.py tests/test_r_bindings.py"
   [stages]="tests/test_stages.py tests/test_properties.py tests/test_fuzzing.py"
-  [gbdt]="tests/test_gbdt.py tests/test_gbdt_sparse.py tests/test_benchmarks_csv.py tests/test_external_anchor.py"
+  [gbdt]="tests/test_gbdt.py tests/test_gbdt_sparse.py tests/test_benchmarks_csv.py tests/test_external_anchor.py tests/test_elastic.py"
   [vw]="tests/test_vw.py"
   [distributed]="tests/test_distributed.py"
   [serving]="tests/test_serving.py"
