@@ -125,8 +125,11 @@ class CsrMatrix:
         return out
 
     def densify_chunks(self, chunk: int = 65536):
-        """Iterate (start, dense_chunk) — bounded-memory scoring."""
-        n = self.shape[0]
+        """Iterate (start, dense_chunk) — bounded-memory scoring.  The
+        chunk shrinks with feature count so the dense tile stays ≈≤1 GB
+        even at 100k features."""
+        n, nf = self.shape
+        chunk = max(256, min(chunk, int((1 << 28) / max(1, nf))))
         for s in range(0, max(n, 1), chunk):
             e = min(n, s + chunk)
             if s >= n:

@@ -1628,21 +1628,30 @@ __global__ void csr_hist_fixed_lds_k(const long* __restrict__ indptr,
     }
     s_len[wid][lane + 1] = v;
     __syncthreads();
+    // contiguous per-lane entry ranges: ONE binary search per lane, then
+    // an amortized forward row walk — the strided variant's per-entry
+    // binary search made the kernel latency-bound (PMC: 21x wait/busy)
     const int total = s_len[wid][nrows];
     const int* cum = s_len[wid];
-    for (int e = lane; e < total; e += 64) {
-      int lo2 = 0, hi2 = nrows;
-      while (lo2 + 1 < hi2) {
-        const int mid = (lo2 + hi2) >> 1;
-        if (cum[mid] <= e) lo2 = mid; else hi2 = mid;
+    const int per = (total + 63) >> 6;
+    const int e0 = lane * per;
+    const int e1 = min(total, e0 + per);
+    if (e0 < e1) {
+      int k = 0, hi2 = nrows;
+      while (k + 1 < hi2) {
+        const int mid = (k + hi2) >> 1;
+        if (cum[mid] <= e0) k = mid; else hi2 = mid;
       }
-      const long j = s_start[wid][lo2] + (e - cum[lo2]);
-      const int c = col[j];
-      if (c >= f0 && c < f0 + FW) {
-        unsigned long long* cell =
-            &lds64[(((c - f0) * n_bins) + binv[j]) * 2];
-        atomicAdd(cell + 0, (unsigned long long)s_g[wid][lo2]);
-        atomicAdd(cell + 1, CNT_ONE + (unsigned long long)s_h[wid][lo2]);
+      for (int e = e0; e < e1; ++e) {
+        while (cum[k + 1] <= e) ++k;
+        const long j = s_start[wid][k] + (e - cum[k]);
+        const int c = col[j];
+        if (c >= f0 && c < f0 + FW) {
+          unsigned long long* cell =
+              &lds64[(((c - f0) * n_bins) + binv[j]) * 2];
+          atomicAdd(cell + 0, (unsigned long long)s_g[wid][k]);
+          atomicAdd(cell + 1, CNT_ONE + (unsigned long long)s_h[wid][k]);
+        }
       }
     }
     __syncthreads();

@@ -413,7 +413,13 @@ class DistributedServingServer:
 
         if self.proxy:
             import requests as _rq
-            session = _rq.Session()
+            local = threading.local()  # Session is not thread-safe; the
+            # continuous head calls the handler from many HTTP threads
+
+            def _session():
+                if not hasattr(local, "s"):
+                    local.s = _rq.Session()
+                return local.s
 
             def head_handler(payloads):
                 out = []
@@ -427,7 +433,7 @@ class DistributedServingServer:
                         if w._stop.is_set():
                             continue  # known-dead: skip without a timeout
                         try:
-                            r = session.post(
+                            r = _session().post(
                                 f"http://{w.host}:{w.port}/", json=p,
                                 timeout=self.reply_timeout)
                             if r.status_code == 200:
