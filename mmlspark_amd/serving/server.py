@@ -481,3 +481,128 @@ class DistributedServingServer:
             w.stop()
         if self.head:
             self.head.stop()
+
+
+class _WorkerProc:
+    __slots__ = ("proc", "host", "port", "name", "dead")
+
+    def __init__(self, proc, host, port, name):
+        self.proc = proc
+        self.host, self.port, self.name = host, port, name
+        self.dead = False
+
+    def alive(self):
+        return not self.dead and self.proc.poll() is None
+
+
+class ProcessServingCluster:
+    """Multi-PROCESS serving: N `mmlspark_amd.serving.worker` subprocesses
+    (one per GPU rank in a real deployment — set HIP_VISIBLE_DEVICES per
+    worker via `worker_env`) behind a proxy head with round-robin failover.
+    The process analog of the reference's per-executor WorkerServer fleet
+    + driver discovery (HTTPSourceV2.scala:475,133-198)."""
+
+    def __init__(self, model_dir: str, n_workers: int = 2,
+                 output_cols: str = "prediction", scorer: bool = False,
+                 host: str = "127.0.0.1", mode: str = "micro-batch",
+                 name: str = "mmlspark-serving", reply_timeout: float = 30.0,
+                 worker_env=None):
+        self.model_dir = model_dir
+        self.n_workers = n_workers
+        self.output_cols = output_cols
+        self.scorer = scorer
+        self.host = host
+        self.mode = mode
+        self.name = name
+        self.reply_timeout = reply_timeout
+        self.worker_env = worker_env  # callable(i) -> env dict, or None
+        self.workers: list = []
+        self.head: Optional[ServingServer] = None
+        self._rr = 0
+        self._rr_lock = threading.Lock()
+
+    def _spawn(self, i: int) -> _WorkerProc:
+        import os
+        import subprocess
+        import sys
+        argv = [sys.executable, "-m", "mmlspark_amd.serving.worker",
+                "--model", self.model_dir, "--host", self.host,
+                "--port", "0", "--mode", self.mode,
+                "--output-cols", self.output_cols,
+                "--name", f"{self.name}-{i}"]
+        if self.scorer:
+            argv.append("--scorer")
+        env = dict(os.environ)
+        if self.worker_env:
+            env.update(self.worker_env(i))
+        proc = subprocess.Popen(argv, stdout=subprocess.PIPE, text=True,
+                                env=env)
+        line = proc.stdout.readline()  # {"ready": true, "port": ...}
+        info = json.loads(line)
+        return _WorkerProc(proc, self.host, int(info["port"]),
+                           f"{self.name}-{i}")
+
+    def start(self):
+        import requests as _rq
+        self.workers = [self._spawn(i) for i in range(self.n_workers)]
+        local = threading.local()
+
+        def _session():
+            if not hasattr(local, "s"):
+                local.s = _rq.Session()
+            return local.s
+
+        def head_handler(payloads):
+            out = []
+            for p in payloads:
+                with self._rr_lock:
+                    start = self._rr
+                    self._rr += 1
+                last_err = None
+                for k in range(len(self.workers) * 2):
+                    w = self.workers[(start + k) % len(self.workers)]
+                    if not w.alive():
+                        continue
+                    try:
+                        r = _session().post(f"http://{w.host}:{w.port}/",
+                                            json=p,
+                                            timeout=self.reply_timeout)
+                        if r.status_code == 200:
+                            out.append(r.json())
+                            break
+                        last_err = f"{w.name}: {r.status_code}"
+                    except Exception as e:
+                        last_err = repr(e)
+                else:
+                    raise RuntimeError(f"all workers failed: {last_err}")
+            return out
+
+        self.head = ServingServer(head_handler, host=self.host, port=0,
+                                  mode="continuous",
+                                  name=f"{self.name}-head").start()
+        return self
+
+    def kill_worker(self, i: int):
+        """Hard-kill the worker PROCESS (real crash, not a simulation)."""
+        self.workers[i].proc.kill()
+        self.workers[i].dead = True
+
+    def restart_worker(self, i: int):
+        self.workers[i] = self._spawn(i)
+
+    def service_info(self):
+        return {"name": self.name,
+                "workers": [{"name": w.name, "host": w.host, "port": w.port,
+                             "alive": w.alive()} for w in self.workers]}
+
+    def stop(self):
+        for w in self.workers:
+            if w.proc.poll() is None:
+                w.proc.terminate()
+        for w in self.workers:
+            try:
+                w.proc.wait(timeout=10)
+            except Exception:
+                w.proc.kill()
+        if self.head:
+            self.head.stop()

@@ -374,3 +374,38 @@ def test_sixty_four_concurrent_clients_smoke(binary_df):
         t.join(timeout=60)
     srv.stop()
     assert not errs, errs[:5]
+
+
+def test_process_serving_cluster_failover(tmp_path, binary_df):
+    """Multi-PROCESS workers (the reference's per-executor WorkerServer
+    shape): real subprocess workers behind a failover head; a hard-killed
+    worker process drops no replies, and a restarted one takes traffic."""
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    from mmlspark_amd.serving.server import ProcessServingCluster
+    model = LightGBMClassifier(numIterations=5, numLeaves=7).fit(binary_df)
+    mdir = str(tmp_path / "model")
+    model.save(mdir)
+    x = binary_df["features"].iloc[0].tolist()
+    cluster = ProcessServingCluster(mdir, n_workers=2,
+                                    output_cols="prediction",
+                                    mode="micro-batch",
+                                    reply_timeout=15).start()
+    try:
+        url = f"http://127.0.0.1:{cluster.head.port}/"
+        s = requests.Session()
+        for _ in range(6):
+            r = s.post(url, json={"features": x}, timeout=20)
+            assert r.status_code == 200
+            assert r.json()["prediction"] in (0.0, 1.0)
+        cluster.kill_worker(0)  # hard process kill
+        for _ in range(6):     # failover to worker 1, nothing dropped
+            r = s.post(url, json={"features": x}, timeout=20)
+            assert r.status_code == 200
+        cluster.restart_worker(0)
+        info = cluster.service_info()
+        assert all(w["alive"] for w in info["workers"])
+        for _ in range(4):
+            assert s.post(url, json={"features": x},
+                          timeout=20).status_code == 200
+    finally:
+        cluster.stop()
