@@ -198,3 +198,54 @@ def _sparse_dist_worker(rank, world, port, q):
         torch.distributed.destroy_process_group()
     except Exception as e:  # pragma: no cover
         q.put((rank, f"ERROR: {e!r}"))
+
+
+def test_sparse_property_random_shapes():
+    """Property sweep: random sparse shapes/params must train, score,
+    save/load without error (empty rows, single-feature, tiny data,
+    unsorted indices, duplicate-free)."""
+    rng = np.random.default_rng(23)
+    for trial in range(6):
+        n = int(rng.integers(30, 400))
+        nf = int(rng.integers(2, 200))
+        rows = []
+        for i in range(n):
+            k = int(rng.integers(0, min(nf, 12) + 1))  # may be 0 (empty row)
+            idx = rng.choice(nf, size=k, replace=False).astype(np.int32)
+            rng.shuffle(idx)  # unsorted on purpose — ingestion must sort
+            rows.append(SparseVector(nf, idx,
+                                     rng.normal(size=k).astype(np.float32)))
+        y = rng.integers(0, 2, size=n).astype(np.float32)
+        df = pd.DataFrame({"features": rows, "label": y})
+        m = LightGBMClassifier(numIterations=3,
+                               numLeaves=int(rng.integers(2, 15)),
+                               maxBin=int(rng.integers(4, 64)),
+                               minDataInLeaf=2).fit(df)
+        out = m.transform(df)
+        assert len(out) == n
+        p = np.stack(out["probability"].to_numpy())
+        assert np.isfinite(p).all(), trial
+        b2 = m.booster.load_from_string(m.booster.save_to_string())
+        assert b2.num_trees == m.booster.num_trees
+
+
+def test_sparse_weight_and_validation_cols():
+    """weightCol + validationIndicatorCol flow through the sparse path."""
+    rng = np.random.default_rng(29)
+    n, nf = 1500, 40
+    rows, y = [], np.zeros(n, dtype=np.float32)
+    w = rng.normal(size=nf)
+    for i in range(n):
+        idx = np.sort(rng.choice(nf, size=8, replace=False)).astype(np.int32)
+        val = rng.normal(size=8).astype(np.float32)
+        y[i] = 1.0 if (w[idx] * val).sum() > 0 else 0.0
+        rows.append(SparseVector(nf, idx, val))
+    df = pd.DataFrame({"features": rows, "label": y,
+                       "wt": rng.random(n).astype(np.float32) + 0.5,
+                       "isVal": rng.random(n) < 0.25})
+    m = LightGBMClassifier(numIterations=25, numLeaves=15, weightCol="wt",
+                           validationIndicatorCol="isVal", metric="auc",
+                           earlyStoppingRound=5).fit(df)
+    assert m.booster.num_trees >= 1
+    evals = m._training_stats.evals
+    assert evals and "auc" in evals[0]["valid_0"]
