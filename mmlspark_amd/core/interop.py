@@ -158,11 +158,20 @@ def coerce_input(df) -> Tuple[pd.DataFrame, str]:
 
 
 def restore_output(out, kind: str):
-    """transform() result back to the caller's DataFrame kind."""
+    """transform() result back to the caller's DataFrame kind.
+
+    Columns that have no Arrow/Spark representation (e.g. rich Python
+    objects some transformers emit) fall back to pandas with a warning
+    rather than failing the whole transform."""
     if kind == PANDAS or not isinstance(out, pd.DataFrame):
         return out
-    if kind == ARROW:
-        return pandas_to_arrow(out)
-    if kind == SPARK:
-        return pandas_to_spark(out)
+    try:
+        if kind == ARROW:
+            return pandas_to_arrow(out)
+        if kind == SPARK:
+            return pandas_to_spark(out)
+    except Exception as e:
+        import warnings
+        warnings.warn(f"could not convert transform output back to {kind} "
+                      f"({e!r}); returning pandas", RuntimeWarning)
     return out
