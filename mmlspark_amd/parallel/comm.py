@@ -109,7 +109,10 @@ def init_from_env(timeout_s: int = 600) -> Comm:
         return _GLOBAL_COMM
     if dist.is_available() and not dist.is_initialized() and "RANK" in os.environ \
             and "WORLD_SIZE" in os.environ and int(os.environ["WORLD_SIZE"]) > 1:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # MMLSPARK_AMD_BACKEND=gloo lets multi-rank runs share one GPU for
+        # rehearsal (RCCL refuses duplicate devices); default is RCCL on GPU
+        backend = os.environ.get("MMLSPARK_AMD_BACKEND") or (
+            "nccl" if torch.cuda.is_available() else "gloo")
         if torch.cuda.is_available():
             local_rank = int(os.environ.get("LOCAL_RANK", os.environ["RANK"]))
             torch.cuda.set_device(local_rank % torch.cuda.device_count())
