@@ -35,8 +35,10 @@ def main():
     use_gpu = torch.cuda.is_available()
     device = torch.device("cuda") if use_gpu else torch.device("cpu")
     if use_gpu and world > 1:
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
-        device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", rank)))
+        # modulo lets a one-GPU rehearsal run the multi-proc path (gloo)
+        li = int(os.environ.get("LOCAL_RANK", rank)) % torch.cuda.device_count()
+        torch.cuda.set_device(li)
+        device = torch.device("cuda", li)
     bs = args.batch if use_gpu else 16
 
     torch.manual_seed(7 + rank)

@@ -31,8 +31,10 @@ def main():
     use_gpu = torch.cuda.is_available()
     device = torch.device("cuda") if use_gpu else torch.device("cpu")
     if use_gpu and world > 1:
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
-        device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", rank)))
+        # modulo lets a one-GPU rehearsal run the multi-proc path (gloo)
+        li = int(os.environ.get("LOCAL_RANK", rank)) % torch.cuda.device_count()
+        torch.cuda.set_device(li)
+        device = torch.device("cuda", li)
 
     n = args.rows if use_gpu else min(args.rows, 100_000)
     tbl = 1 << args.bits
