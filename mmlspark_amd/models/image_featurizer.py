@@ -150,6 +150,9 @@ class ImageFeaturizer(TorchModel):
     """Transfer-learning image featurizer (ImageFeaturizer.scala:41)."""
 
     modelName = Param("modelName", "backbone: ResNet18/34/50/101", "ResNet50")
+    batchSize = Param("batchSize", "inference minibatch size (larger than "
+                      "the TorchModel default: decode threads + the conv "
+                      "stack both want deep batches)", 256, toInt)
     cutOutputLayers = Param("cutOutputLayers",
                             "how many output layers to cut (1 = pooled "
                             "features)", 1, toInt)
