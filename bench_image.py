@@ -119,8 +119,10 @@ def bytes_to_features(args):
     blobs = [imgs[i % 16] for i in range(n_imgs)]
     feat = ImageFeaturizer(modelName="ResNet50", cutOutputLayers=1,
                            imageSize=224)
-    df = pd.DataFrame({"image": blobs[:32]})
-    feat.transform(df)  # warmup (weights init + first kernels)
+    # warm with the SAME batch shape as the measured run — MIOpen tunes
+    # per conv shape and the first find is seconds, not milliseconds
+    nb = min(len(blobs), 2 * feat.get("batchSize"))
+    feat.transform(pd.DataFrame({"image": blobs[:nb]}))
     if use_gpu:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
