@@ -37,6 +37,10 @@ def main():
                     help="CSR ablation: train from sparse input")
     ap.add_argument("--nnz", type=int, default=20,
                     help="nonzeros per row in --sparse mode")
+    ap.add_argument("--parallelism", default="data_parallel",
+                    choices=["data_parallel", "voting_parallel"],
+                    help="histogram sync strategy (voting reduces only the "
+                         "globally-voted top-K features per split)")
     args = ap.parse_args()
 
     sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
@@ -100,7 +104,8 @@ def main():
     cfg = TrainConfig(num_iterations=args.warmup + args.steps,
                       num_leaves=args.num_leaves, learning_rate=0.1,
                       max_bin=255, min_data_in_leaf=20,
-                      categorical_features=cat_idx)
+                      categorical_features=cat_idx,
+                      parallelism=args.parallelism)
     objective = make_objective("binary")
     session = TrainingSession(X, y, cfg, objective, comm)
 
@@ -150,7 +155,9 @@ def main():
                 "max_bin": 255,
                 "objective": "binary",
                 "global_batch": n * world,
-                "parallelism": f"dp{world}",
+                "parallelism": (f"dp{world}" if args.parallelism ==
+                                "data_parallel" else
+                                f"voting{world} topK20"),
                 "sync": "RCCL histogram all_reduce over xGMI" if world > 1
                         else "single rank",
                 "categorical": args.categorical,
