@@ -63,6 +63,27 @@ def main():
     if not use_gpu:  # CPU smoke: shrink so the default invocation finishes fast
         n = min(n, 200_000)
 
+    if args.sparse:
+        # CSR generated DIRECTLY (a dense intermediate would need n×nf
+        # floats — 1.5 TB at 4M×100k); label from a sparse linear model
+        from mmlspark_amd.models.gbdt.sparse import CsrMatrix
+        gen_dev = device if use_gpu else torch.device("cpu")
+        if use_gpu:
+            torch.cuda.manual_seed(1234 + rank)
+        keep = torch.randint(0, nf, (n, args.nnz), device=gen_dev,
+                             dtype=torch.int64).sort(dim=1).values
+        val = torch.randn(n, args.nnz, device=gen_dev)
+        w = torch.randn(nf, device=gen_dev) / (args.nnz ** 0.5)
+        logits = (w[keep.reshape(-1)].view(n, args.nnz) * val).sum(dim=1)
+        y = (logits + 0.5 * torch.randn(n, device=gen_dev) > 0).float()
+        indptr = torch.arange(0, (n + 1) * args.nnz, args.nnz,
+                              dtype=torch.int64, device=gen_dev)
+        X = CsrMatrix(indptr, keep.to(torch.int32).reshape(-1),
+                      val.reshape(-1), (n, nf))
+        del keep, val, w, logits
+        return run_session(args, comm, rank, world, use_gpu, device, X, y,
+                           n, nf, cat_idx=None)
+
     # synthetic Higgs-like binary data, random-init (no network/datasets
     # here); generated directly on the device so an 8-rank launch does not
     # stage 8×4 GB through host RAM
@@ -90,17 +111,14 @@ def main():
         cat_idx = list(range(k))
     X = X.to(device)
     y = y.to(device)
-    if args.sparse:
-        # CSR ablation: keep --nnz random entries per row (same label task)
-        from mmlspark_amd.models.gbdt.sparse import CsrMatrix
-        keep = torch.argsort(torch.rand(n, nf, device=device),
-                             dim=1)[:, :args.nnz].sort(dim=1).values
-        val = torch.gather(X, 1, keep)
-        indptr = torch.arange(0, (n + 1) * args.nnz, args.nnz,
-                              dtype=torch.int64, device=device)
-        X = CsrMatrix(indptr, keep.to(torch.int32).reshape(-1),
-                      val.reshape(-1), (n, nf))
+    return run_session(args, comm, rank, world, use_gpu, device, X, y, n,
+                       nf, cat_idx)
 
+
+def run_session(args, comm, rank, world, use_gpu, device, X, y, n, nf,
+                cat_idx):
+    from mmlspark_amd.models.gbdt.objectives import make_objective
+    from mmlspark_amd.models.gbdt.trainer import TrainConfig, TrainingSession
     cfg = TrainConfig(num_iterations=args.warmup + args.steps,
                       num_leaves=args.num_leaves, learning_rate=0.1,
                       max_bin=255, min_data_in_leaf=20,
