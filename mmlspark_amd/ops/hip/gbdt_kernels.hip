@@ -1115,6 +1115,18 @@ __global__ void split_scan_k(const void* __restrict__ hist_v, int n_bins,
     sc[tid] = in ? H[tid * 3 + 2] : 0.0f;
   }
   __syncthreads();
+  // degenerate feature: every row in ONE bin (e.g. a sparse feature absent
+  // from this leaf — all mass in its zero bin) can never split validly;
+  // skip the 3 prefix scans + gain pass.  Dominant on wide-sparse shapes
+  // where most of 100k features are untouched per leaf.
+  const int nz_bins = __syncthreads_count(in && sc[tid] > 0.0f);
+  if (nz_bins <= 1) {
+    if (tid == 0) {
+      out[0] = -INFINITY; out[1] = (float)f; out[2] = 0;
+      out[3] = 0; out[4] = 0; out[5] = 0;
+    }
+    return;
+  }
   // Hillis-Steele inclusive scan over 256 slots
 #pragma unroll
   for (int d = 1; d < 256; d <<= 1) {
