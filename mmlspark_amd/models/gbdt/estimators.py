@@ -261,8 +261,15 @@ class _GBDTEstimatorBase(_GBDTParams, Estimator):
         from .metrics import default_metrics_fn
         booster = init
         stats = None
-        for b in batches:
+        for bi, b in enumerate(batches):
             part = df if n_batches <= 1 else df.iloc[b]
+            # each sequential batch gets its own checkpoint namespace —
+            # otherwise batch 2+ would "resume" from batch 1's completed
+            # checkpoint and train nothing
+            ckdir = self.get("checkpointDir")
+            if ckdir and n_batches > 1:
+                import os as _os
+                ckdir = _os.path.join(ckdir, f"batch{bi}")
             Xt, yt, w = self._extract(part, device)
             w = self._adjust_weights(yt, w)
             group = self._group_sizes(part, device)
@@ -277,7 +284,7 @@ class _GBDTEstimatorBase(_GBDTParams, Estimator):
                 valid_sets=valid_sets, init_booster=booster,
                 metrics_fn=default_metrics_fn(self.get("metric") or None),
                 init_score=init_score,
-                checkpoint_dir=self.get("checkpointDir"),
+                checkpoint_dir=ckdir,
                 checkpoint_every=self.get("checkpointInterval"))
         model = self._model_class()(booster=booster)
         for p in ("labelCol", "featuresCol", "featureCols", "predictionCol",

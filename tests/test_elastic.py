@@ -120,3 +120,18 @@ def test_estimator_checkpoint_params(tmp_path):
     with open(os.path.join(ckdir, "checkpoint.json")) as f:
         d = json.load(f)
     assert d["iteration"] == 6
+
+
+def test_checkpoints_with_num_batches(tmp_path):
+    """numBatches + checkpointDir: each batch checkpoints in its own
+    namespace so later batches never 'resume' a finished earlier batch."""
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    X, y = _data(seed=9)
+    df = pd.DataFrame({"features": list(X.numpy()), "label": y.numpy()})
+    ckdir = str(tmp_path / "ck")
+    m = LightGBMClassifier(numIterations=4, numLeaves=7, numBatches=2,
+                           checkpointDir=ckdir,
+                           checkpointInterval=2).fit(df)
+    assert m.booster.num_trees == 8  # 4 per batch — batch 2 actually ran
+    assert os.path.exists(os.path.join(ckdir, "batch0", "checkpoint.json"))
+    assert os.path.exists(os.path.join(ckdir, "batch1", "checkpoint.json"))
