@@ -290,10 +290,14 @@ class _VWBase(_VWParams, Estimator):
         self._parse_args()
         # materialize -q namespace pairs into the interactions param so the
         # SAME crossing applies at fit and at model transform time
+        # (idempotent: refitting must not duplicate pairs)
         q = self._resolve_quadratic(df)
         if q:
-            self.set("interactions",
-                     list(self.get("interactions") or []) + q)
+            existing = [list(p) if isinstance(p, (list, tuple)) else [p]
+                        for p in (self.get("interactions") or [])]
+            q = [p for p in q if p not in existing]
+            if q:
+                self.set("interactions", existing + q)
             self._quadratic_specs = []
         comm = get_comm()
         device = default_device(self.get("device"))
