@@ -421,7 +421,16 @@ class LightGBMClassificationModel(_GBDTModelBase):
         b = self.booster
         si, ni = self.get("startIteration"), self.get("numIterations")
         raw = b.predict_raw(X, si, ni)
-        prob = b.predict_prob(X, si, ni)
+        # probability from the raw margins already computed — traversing
+        # the forest once, not twice (this path is the explainer fan-out's
+        # inner loop)
+        if b.objective == "binary":
+            p1 = torch.sigmoid(b.sigmoid * raw)
+            prob = torch.cat([1 - p1, p1], dim=-1)
+        elif b.objective in ("multiclass", "softmax"):
+            prob = torch.softmax(raw, dim=-1)
+        else:
+            prob = raw
         out = df.copy()
         if b.objective == "binary":
             raw2 = torch.cat([-raw, raw], dim=-1)
