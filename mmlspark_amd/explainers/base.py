@@ -38,3 +38,22 @@ class LocalExplainer(Transformer):
             mat = np.stack([np.asarray(v) for v in col])
             return mat[:, classes]
         return np.asarray(col, dtype=np.float64)[:, None]
+
+    def _score_matrix(self, X: np.ndarray, cols=None) -> np.ndarray:
+        """Score a raw feature MATRIX.  Models exposing score_matrix (the
+        GBDT models) skip the whole DataFrame round trip — the fan-out
+        builds millions of object cells otherwise; generic transformers
+        fall back to _score_samples."""
+        model = self.get("model")
+        if hasattr(model, "score_matrix"):
+            v = model.score_matrix(X, self.get("targetCol"),
+                                   self.get("targetClasses"))
+            if v is not None:
+                return v
+        if cols is not None:
+            return self._score_samples(pd.DataFrame(X, columns=list(cols)))
+        from ..core.schema import matrix_to_vector_column
+        fcol = model.get("featuresCol") if "featuresCol" in model.params() \
+            else "features"
+        return self._score_samples(pd.DataFrame({
+            fcol: matrix_to_vector_column(X.astype(np.float32))}))

@@ -63,7 +63,7 @@ class TabularLIME(LIMEBase):
         for x in rows:
             states = lime_sample_states(m, n_samp, rng)
             pert = VectorSampler(bg, rng).apply(x, states)
-            scores = self._score_samples(pd.DataFrame(pert, columns=cols))
+            scores = self._score_matrix(pert, cols)
             explanations.append(self._fit_states(states, scores))
         out = df.copy()
         out[self.get("outputCol")] = explanations
@@ -88,8 +88,7 @@ class VectorLIME(LIMEBase):
         for x in rows:
             states = lime_sample_states(m, n_samp, rng)
             pert = VectorSampler(bg, rng).apply(x, states).astype(np.float32)
-            scores = self._score_samples(
-                pd.DataFrame({fcol: matrix_to_vector_column(pert)}))
+            scores = self._score_matrix(pert)
             explanations.append(self._fit_states(states, scores))
         out = df.copy()
         out[self.get("outputCol")] = explanations

@@ -63,11 +63,9 @@ class TabularSHAP(KernelSHAPBase):
                 pert = sampler.apply(x, Z)
                 frames.append(pert)
             all_pert = np.concatenate(frames) if frames else np.zeros((0, m))
-            samples_df = pd.DataFrame(all_pert, columns=cols)
             # v_null from background mean prediction, v_full from the row
-            extra = pd.DataFrame(np.concatenate([bg, chunk]), columns=cols)
-            scores = self._score_samples(pd.concat([samples_df, extra],
-                                                   ignore_index=True))
+            scores = self._score_matrix(
+                np.concatenate([all_pert, bg, chunk]), cols)
             nZ = Z.shape[0]
             v_null = scores[len(all_pert):len(all_pert) + len(bg)].mean(axis=0)
             for i, x in enumerate(chunk):
@@ -101,10 +99,8 @@ class VectorSHAP(KernelSHAPBase):
             pert_frames = [VectorSampler(bg, rng).apply(x, Z) for x in chunk]
             all_pert = np.concatenate(pert_frames)
             full = np.concatenate([bg, chunk])
-            samples_df = pd.DataFrame({
-                fcol: matrix_to_vector_column(
-                    np.concatenate([all_pert, full]).astype(np.float32))})
-            scores = self._score_samples(samples_df)
+            scores = self._score_matrix(
+                np.concatenate([all_pert, full]).astype(np.float32))
             nZ = Z.shape[0]
             v_null = scores[len(all_pert):len(all_pert) + len(bg)].mean(axis=0)
             for i in range(len(chunk)):

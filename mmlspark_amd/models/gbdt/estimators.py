@@ -360,6 +360,46 @@ class _GBDTModelBase(Model):
     def getFeatureImportances(self, importance_type: str = "split"):
         return self.booster.feature_importances(importance_type).tolist()
 
+    def score_matrix(self, X: np.ndarray, target_col: str, classes):
+        """Explainer fast path (LocalExplainer._score_matrix): raw feature
+        matrix → target-column values with ONE forest pass and no DataFrame
+        construction — the perturbation fan-out scores millions of rows."""
+        device = default_device(self.get("device"))
+        Xt = torch.from_numpy(
+            np.ascontiguousarray(X, dtype=np.float32)).to(device)
+        b = self.booster
+        raw = b.predict_raw(Xt, self.get("startIteration"),
+                            self.get("numIterations"))
+        params = self.params()
+        prob_col = self.get("probabilityCol") if "probabilityCol" in params \
+            else None
+        raw_col = self.get("rawPredictionCol") if "rawPredictionCol" in params \
+            else None
+        if target_col == prob_col:
+            if b.objective == "binary":
+                p1 = torch.sigmoid(b.sigmoid * raw)
+                vals = torch.cat([1 - p1, p1], dim=-1)
+            elif b.objective in ("multiclass", "softmax"):
+                vals = torch.softmax(raw, dim=-1)
+            else:
+                vals = raw
+        elif target_col == raw_col:
+            vals = torch.cat([-raw, raw], dim=-1) if b.objective == "binary" \
+                else raw
+        elif target_col == self.get("predictionCol"):
+            if b.objective == "binary":
+                vals = (torch.sigmoid(b.sigmoid * raw) > 0.5).float()
+            elif b.objective in ("multiclass", "softmax"):
+                vals = raw.argmax(dim=-1, keepdim=True).float()
+            else:
+                vals = raw
+        else:  # unknown target column: the caller falls back to transform()
+            return None
+        v = vals.cpu().numpy().astype(np.float64)
+        if v.shape[1] == 1:  # scalar target (regression/prediction)
+            return v
+        return v[:, list(classes)]
+
     def _X(self, df: pd.DataFrame, device):
         from .sparse import CsrMatrix, looks_sparse
         fc = self.get("featuresCol")
