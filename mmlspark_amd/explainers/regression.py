@@ -104,3 +104,31 @@ def constrained_kernel_shap_solve(Z: torch.Tensor, v: torch.Tensor,
     phi_rest = res.coefficients
     phi_last = (v_full - v_null) - phi_rest.sum()
     return torch.cat([phi_rest, phi_last.reshape(1)])
+
+
+def batched_kernel_shap_solve(Z: torch.Tensor, V: torch.Tensor,
+                              weights: torch.Tensor, v_null: torch.Tensor,
+                              v_full: torch.Tensor) -> torch.Tensor:
+    """Solve the constrained KernelSHAP WLS for MANY rows sharing one
+    coalition design: factor A = Zr'WZr once, apply to every row/class.
+
+    Z: (nZ, m); V: (R, nZ) model outputs per row; v_null: scalar or (R,);
+    v_full: (R,).  Returns (R, m) phis."""
+    m = Z.shape[1]
+    R = V.shape[0]
+    if m == 1:
+        return (v_full - v_null).reshape(R, 1)
+    zl = Z[:, -1]                      # (nZ,)
+    Zr = Z[:, :-1] - zl.unsqueeze(1)   # (nZ, m-1)
+    Zw = Zr * weights.unsqueeze(1)
+    A = Zw.t() @ Zr                    # (m-1, m-1) — shared
+    vn = v_null if torch.is_tensor(v_null) else torch.tensor(
+        v_null, dtype=V.dtype)
+    vn = vn.reshape(-1) if vn.dim() else vn.reshape(1)
+    span = (v_full - vn)               # (R,)
+    Yr = V - vn.reshape(-1, 1) - span.reshape(-1, 1) * zl.unsqueeze(0)
+    B = Zw.t() @ Yr.t()                # (m-1, R)
+    eye = torch.eye(m - 1, dtype=A.dtype)
+    phi_rest = torch.linalg.solve(A + 1e-10 * eye, B).t()  # (R, m-1)
+    phi_last = span - phi_rest.sum(dim=1)
+    return torch.cat([phi_rest, phi_last.reshape(R, 1)], dim=1)

@@ -17,7 +17,8 @@ from ..core.param import Param, toInt, toList
 from ..core.registry import register
 from ..core.schema import matrix_to_vector_column, vector_column_to_matrix
 from .base import LocalExplainer
-from .regression import constrained_kernel_shap_solve
+from .regression import (batched_kernel_shap_solve,
+                         constrained_kernel_shap_solve)
 from .sampler import (ImageSampler, TextSampler, VectorSampler,
                       sample_coalitions, slic_superpixels)
 
@@ -25,6 +26,26 @@ from .sampler import (ImageSampler, TextSampler, VectorSampler,
 class KernelSHAPBase(LocalExplainer):
     def _default_samples(self, m):
         return 2 * m + 2048
+
+    def _solve_batch(self, Z, scores, nZ, n_pert, n_bg, w, v_null,
+                     n_rows):
+        """All rows of a chunk share Z/w: one factorization, every
+        row × class solved in a single matrix solve."""
+        C = scores.shape[1]
+        Zt = torch.from_numpy(Z.astype(np.float64))
+        wt = torch.from_numpy(w)
+        per_class = []
+        for k in range(C):
+            V = torch.from_numpy(
+                scores[:n_pert, k].reshape(n_rows, nZ).astype(np.float64))
+            v_fulls = torch.from_numpy(
+                scores[n_pert + n_bg:, k].astype(np.float64))
+            per_class.append(batched_kernel_shap_solve(
+                Zt, V, wt, float(v_null[k]), v_fulls).numpy())
+        return [np.stack([np.concatenate([[float(v_null[k])],
+                                          per_class[k][i]])
+                          for k in range(C)])
+                for i in range(n_rows)]
 
     def _solve(self, Z, v, w, v_null, v_full):
         phis = []
@@ -68,10 +89,9 @@ class TabularSHAP(KernelSHAPBase):
                 np.concatenate([all_pert, bg, chunk]), cols)
             nZ = Z.shape[0]
             v_null = scores[len(all_pert):len(all_pert) + len(bg)].mean(axis=0)
-            for i, x in enumerate(chunk):
-                v = scores[i * nZ:(i + 1) * nZ]
-                v_full = scores[len(all_pert) + len(bg) + i]
-                explanations.append(self._solve(Z, v, w, v_null, v_full))
+            explanations.extend(self._solve_batch(
+                Z, scores, nZ, len(all_pert), len(bg), w, v_null,
+                len(chunk)))
         out = df.copy()
         out[self.get("outputCol")] = explanations
         return out
@@ -103,10 +123,9 @@ class VectorSHAP(KernelSHAPBase):
                 np.concatenate([all_pert, full]).astype(np.float32))
             nZ = Z.shape[0]
             v_null = scores[len(all_pert):len(all_pert) + len(bg)].mean(axis=0)
-            for i in range(len(chunk)):
-                v = scores[i * nZ:(i + 1) * nZ]
-                v_full = scores[len(all_pert) + len(bg) + i]
-                explanations.append(self._solve(Z, v, w, v_null, v_full))
+            explanations.extend(self._solve_batch(
+                Z, scores, nZ, len(all_pert), len(bg), w, v_null,
+                len(chunk)))
         out = df.copy()
         out[self.get("outputCol")] = explanations
         return out
