@@ -60,11 +60,19 @@ class TabularLIME(LIMEBase):
         n_samp = self.get("numSamples") or self._default_samples(m)
         explanations = []
         rows = df[cols].to_numpy(dtype=np.float64)
-        for x in rows:
-            states = lime_sample_states(m, n_samp, rng)
-            pert = VectorSampler(bg, rng).apply(x, states)
-            scores = self._score_matrix(pert, cols)
-            explanations.append(self._fit_states(states, scores))
+        batch = self.get("rowBatch")
+        for s0 in range(0, len(rows), batch):
+            chunk = rows[s0:s0 + batch]
+            states_l = [lime_sample_states(m, n_samp, rng) for _ in chunk]
+            perts = [VectorSampler(bg, rng).apply(x, st)
+                     for x, st in zip(chunk, states_l)]
+            # one scoring fan-out for the whole chunk
+            scores = self._score_matrix(np.concatenate(perts), cols)                 if perts else np.zeros((0, 1))
+            o = 0
+            for st, p_ in zip(states_l, perts):
+                explanations.append(
+                    self._fit_states(st, scores[o:o + len(p_)]))
+                o += len(p_)
         out = df.copy()
         out[self.get("outputCol")] = explanations
         return out
@@ -85,11 +93,18 @@ class VectorLIME(LIMEBase):
         n_samp = self.get("numSamples") or self._default_samples(m)
         rows = vector_column_to_matrix(df, fcol).astype(np.float64)
         explanations = []
-        for x in rows:
-            states = lime_sample_states(m, n_samp, rng)
-            pert = VectorSampler(bg, rng).apply(x, states).astype(np.float32)
-            scores = self._score_matrix(pert)
-            explanations.append(self._fit_states(states, scores))
+        batch = self.get("rowBatch")
+        for s0 in range(0, len(rows), batch):
+            chunk = rows[s0:s0 + batch]
+            states_l = [lime_sample_states(m, n_samp, rng) for _ in chunk]
+            perts = [VectorSampler(bg, rng).apply(x, st).astype(np.float32)
+                     for x, st in zip(chunk, states_l)]
+            scores = self._score_matrix(np.concatenate(perts))                 if perts else np.zeros((0, 1))
+            o = 0
+            for st, p_ in zip(states_l, perts):
+                explanations.append(
+                    self._fit_states(st, scores[o:o + len(p_)]))
+                o += len(p_)
         out = df.copy()
         out[self.get("outputCol")] = explanations
         return out
