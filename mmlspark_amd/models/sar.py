@@ -152,13 +152,17 @@ class SARModel(Model):
         if remove_seen:
             scores = torch.where(A > 0, torch.full_like(scores, -1e30), scores)
         vals, idx = torch.topk(scores, min(k, scores.shape[1]), dim=1)
+        # ONE device→host transfer (a per-user .cpu() was a sync per row)
+        vals_np = vals.cpu().numpy()
+        idx_np = idx.cpu().numpy()
+        ucol, icol = self.get("userCol"), self.get("itemCol")
         rows = []
-        for uidx in range(scores.shape[0]):
+        for uidx in range(vals_np.shape[0]):
             rows.append({
-                self.get("userCol"): uidx,
+                ucol: uidx,
                 "recommendations": [
-                    {self.get("itemCol"): int(j), "rating": float(v)}
-                    for j, v in zip(idx[uidx].cpu(), vals[uidx].cpu())
+                    {icol: int(j), "rating": float(v)}
+                    for j, v in zip(idx_np[uidx], vals_np[uidx])
                     if v > -1e29]})
         return pd.DataFrame(rows)
 
