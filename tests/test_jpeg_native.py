@@ -326,3 +326,156 @@ def test_image_featurizer_accepts_jpeg_bytes():
     out2 = feat.transform(pd.DataFrame({"image": [arr, arr, arr]}))
     F2 = np.stack(out2["features"].to_numpy())
     np.testing.assert_allclose(F, F2, atol=1e-4)
+
+
+def encode_progressive_ac_sa(img, quality=90):
+    """Full successive-approximation progressive stream: DC (Al=1 +
+    refine), AC first scans at Al=1, AC refinement scans at Ah=1,Al=0 —
+    the encoder-side mirror of ITU-T.81 G.2.2/G.2.3 so the decoder's AC
+    refinement path (EOBRUN + correction bits) is exercised end to end."""
+    h, w = img.shape[:2]
+    planes = _quantized_planes(img, quality, (1, 1))
+    out = _headers(h, w, quality, (1, 1), progressive=True)
+    dc_codes = [_build_codes(*HT_DC_L), _build_codes(*HT_DC_C)]
+    ac_codes = [_build_codes(*HT_AC_L), _build_codes(*HT_AC_C)]
+    bh, bw_ = planes[0].shape[:2]
+
+    # DC scans (same as encode_progressive)
+    out += _sos([(1, 0, 0), (2, 1, 1), (3, 1, 1)], 0, 0, 0, 1)
+    bw = _BitWriter()
+    pred = [0, 0, 0]
+    for by in range(bh):
+        for bx in range(bw_):
+            for ci in range(3):
+                v = int(planes[ci][by, bx, 0]) >> 1
+                _write_dc(bw, dc_codes[0 if ci == 0 else 1], v - pred[ci])
+                pred[ci] = v
+    bw.flush()
+    out += bytes(bw.out)
+    out += _sos([(1, 0, 0), (2, 1, 1), (3, 1, 1)], 0, 0, 1, 0)
+    bw = _BitWriter()
+    for by in range(bh):
+        for bx in range(bw_):
+            for ci in range(3):
+                bw.write(int(planes[ci][by, bx, 0]) & 1, 1)
+    bw.flush()
+    out += bytes(bw.out)
+
+    def ac_first(ci, Al):
+        """AC first scan of v>>Al.  Annex-K AC tables carry only EOB0, so
+        every block that ends early emits its own EOB0 (the decoder's
+        eobrun = (1<<0)-1 = 0 ends just that block)."""
+        bw = _BitWriter()
+        codes = ac_codes[0 if ci == 0 else 1]
+        for by in range(bh):
+            for bx in range(bw_):
+                zz = planes[ci][by, bx]
+                sh = [(abs(int(zz[k])) >> Al) *
+                      (1 if int(zz[k]) >= 0 else -1) for k in range(1, 64)]
+                last = 0
+                for k in range(63):
+                    if sh[k]:
+                        last = k + 1
+                run = 0
+                for k in range(last):
+                    v = sh[k]
+                    if v == 0:
+                        run += 1
+                        continue
+                    while run > 15:
+                        c, ln = codes[0xF0]
+                        bw.write(c, ln)
+                        run -= 16
+                    n = _mag(v)
+                    c, ln = codes[(run << 4) | n]
+                    bw.write(c, ln)
+                    bw.write(v if v >= 0 else v + (1 << n) - 1, n)
+                    run = 0
+                if last < 63:
+                    c, ln = codes[0x00]
+                    bw.write(c, ln)
+        bw.flush()
+        return bytes(bw.out)
+
+    def ac_refine(ci, Ah, Al):
+        """AC refinement (G.2.3 / jpeg6b encode_mcu_AC_refine), one EOB0
+        per early-ending block: the EOB symbol is followed by the block's
+        remaining correction bits (decoder: eobrun=1 → append bits for the
+        rest of THIS block)."""
+        bw = _BitWriter()
+        codes = ac_codes[0 if ci == 0 else 1]
+        for by in range(bh):
+            for bx in range(bw_):
+                zz = planes[ci][by, bx]
+                absv = [abs(int(zz[k])) >> Al for k in range(1, 64)]
+                signs = [1 if int(zz[k]) >= 0 else -1 for k in range(1, 64)]
+                EOB = 0
+                for k in range(63):
+                    if absv[k] == 1:  # newly significant at this precision
+                        EOB = k + 1
+                run = 0
+                block_bits = []
+                k = 0
+                while k < EOB:
+                    t = absv[k]
+                    if t == 0:
+                        run += 1
+                        k += 1
+                        continue
+                    if t > 1:  # already significant: buffer correction bit
+                        block_bits.append(t & 1)
+                        k += 1
+                        continue
+                    while run > 15:
+                        c, ln = codes[0xF0]
+                        bw.write(c, ln)
+                        for b in block_bits:
+                            bw.write(b, 1)
+                        block_bits.clear()
+                        run -= 16
+                    c, ln = codes[(run << 4) | 1]
+                    bw.write(c, ln)
+                    bw.write(1 if signs[k] > 0 else 0, 1)
+                    for b in block_bits:
+                        bw.write(b, 1)
+                    block_bits.clear()
+                    run = 0
+                    k += 1
+                if EOB < 63:
+                    # EOB0 + correction bits for every already-significant
+                    # coefficient in [EOB, 63)
+                    c, ln = codes[0x00]
+                    bw.write(c, ln)
+                    for b in block_bits:
+                        bw.write(b, 1)
+                    for kk in range(EOB, 63):
+                        if absv[kk] > 1:
+                            bw.write(absv[kk] & 1, 1)
+                else:
+                    for b in block_bits:
+                        bw.write(b, 1)
+        bw.flush()
+        return bytes(bw.out)
+
+    for ci in range(3):
+        td = 0 if ci == 0 else 1
+        out += _sos([(ci + 1, td, td)], 1, 63, 0, 1)
+        out += ac_first(ci, 1)
+    for ci in range(3):
+        td = 0 if ci == 0 else 1
+        out += _sos([(ci + 1, td, td)], 1, 63, 1, 0)
+        out += ac_refine(ci, 1, 0)
+    return out + b"\xff\xd9"
+
+
+def test_progressive_ac_successive_approximation():
+    """AC successive approximation (first Al=1 + refinement Ah=1) must
+    reconstruct EXACTLY what a baseline stream of the same coefficients
+    gives — exercising the decoder's G.2.3 path (EOBRUN + correction
+    bits) that plain spectral-selection streams never reach."""
+    img = _test_image(h=80, w=96, seed=11)
+    prog = encode_progressive_ac_sa(img, quality=85)
+    base = jpeg_codec.encode_jpeg(img, quality=85)
+    out_p = _jpeg_native.decode_jpeg(prog).numpy()
+    out_b = _jpeg_native.decode_jpeg(base).numpy()
+    np.testing.assert_array_equal(out_p, out_b)
