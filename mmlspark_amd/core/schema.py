@@ -73,8 +73,9 @@ def vector_column_to_matrix(df: pd.DataFrame, col: str,
 
 
 def matrix_to_vector_column(mat: np.ndarray) -> list:
-    """Dense (n, d) matrix -> list of 1-D arrays for an object column."""
-    return [np.asarray(row) for row in mat]
+    """Dense (n, d) matrix -> list of 1-D arrays for an object column.
+    list() iterates at C level and row views need no copy."""
+    return list(np.asarray(mat))
 
 
 def features_matrix(df: pd.DataFrame, features_col: str = "features",

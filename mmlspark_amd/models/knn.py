@@ -104,13 +104,13 @@ class KNNModel(_KNNParamsMixin, Model):
         vals = data.get("values")
         idx, dist = self._search(Q) if len(df) else (np.zeros((0, 0), int),
                                                      np.zeros((0, 0)))
-        matches = []
-        for i in range(len(df)):
-            matches.append([
-                {"value": (vals[j].item() if hasattr(vals[j], "item") else vals[j])
-                          if len(vals) else int(j),
-                 "distance": float(dist[i, c])}
-                for c, j in enumerate(idx[i])])
+        has_vals = len(vals) > 0
+        matches = [
+            [{"value": (vals[j].item() if hasattr(vals[j], "item")
+                        else vals[j]) if has_vals else j,
+              "distance": d}
+             for j, d in zip(ir, dr)]
+            for ir, dr in zip(idx.tolist(), dist.tolist())]
         out = df.copy()
         out[self.get("outputCol")] = matches
         return out
