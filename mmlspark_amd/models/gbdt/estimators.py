@@ -257,6 +257,7 @@ class _GBDTEstimatorBase(_GBDTParams, Estimator):
         if valid_df is not None and len(valid_df):
             Xv, yv, wv = self._extract(valid_df, device)
             valid_sets = [(Xv, yv, wv)]
+            valid_groups = self._group_sizes(valid_df, device)
 
         from .metrics import default_metrics_fn
         booster = init
@@ -282,7 +283,8 @@ class _GBDTEstimatorBase(_GBDTParams, Estimator):
                 Xt, yt, self._train_config(), self._make_objective(yt), comm,
                 weight=w, group_sizes=group, feature_names=names,
                 valid_sets=valid_sets, init_booster=booster,
-                metrics_fn=default_metrics_fn(self.get("metric") or None),
+                metrics_fn=self._metrics_fn(
+                    valid_groups if valid_sets else None),
                 init_score=init_score,
                 checkpoint_dir=ckdir,
                 checkpoint_every=self.get("checkpointInterval"))
@@ -295,6 +297,10 @@ class _GBDTEstimatorBase(_GBDTParams, Estimator):
 
     def _group_sizes(self, df, device):
         return None
+
+    def _metrics_fn(self, valid_groups=None):
+        from .metrics import default_metrics_fn
+        return default_metrics_fn(self.get("metric") or None)
 
     def _adjust_weights(self, yt, w):
         return w
@@ -547,6 +553,45 @@ class LightGBMRanker(_GBDTEstimatorBase):
         # grouping column, LightGBMParams.scala:76) — sort locally by group
         df = df.sort_values(self.get("groupCol"), kind="stable").reset_index(drop=True)
         return super()._fit(df)
+
+    def _metrics_fn(self, valid_groups=None):
+        """Group-aware NDCG@k on the validation split (evalAt /
+        maxPosition, labelGain semantics — LightGBMRanker.scala:36-52);
+        higher-is-better early stopping keys off the 'ndcg@k' name."""
+        eval_at = list(self.get("evalAt") or [self.get("maxPosition")])
+        gains_tbl = self.get("labelGain")
+        sizes = (valid_groups.tolist() if valid_groups is not None else None)
+
+        def fn(booster, Xv, yv, wv, comm):
+            raw = booster.predict_raw(Xv).squeeze(-1)
+            scores = raw.cpu().numpy()
+            y = yv.cpu().numpy()
+            group_sizes = sizes if sizes else [len(y)]
+            out = {}
+            for k in eval_at:
+                vals = []
+                start = 0
+                for sz in group_sizes:
+                    sl = slice(start, start + int(sz))
+                    start += int(sz)
+                    ys, ss = y[sl], scores[sl]
+                    if len(ys) == 0:
+                        continue
+                    order = np.argsort(-ss, kind="stable")
+                    g = (np.asarray(gains_tbl)[ys.astype(int)]
+                         if gains_tbl is not None else 2.0 ** ys - 1.0)
+                    disc = 1.0 / np.log2(np.arange(2, len(ys) + 2))
+                    kk = min(int(k), len(ys))
+                    dcg = float((g[order][:kk] * disc[:kk]).sum())
+                    idcg = float((np.sort(g)[::-1][:kk] * disc[:kk]).sum())
+                    vals.append(dcg / idcg if idcg > 0 else 0.0)
+                m = float(np.mean(vals)) if vals else 0.0
+                t = torch.tensor([m, 1.0], device=Xv.device)
+                comm.all_reduce(t)
+                out[f"ndcg@{int(k)}"] = float(t[0] / t[1])
+            return out
+
+        return fn
 
     def _model_class(self):
         return LightGBMRankerModel

@@ -440,3 +440,34 @@ def test_lightgbm_text_export_preserves_sigmoid(binary_df):
     # round trip through our own text importer keeps the sigmoid
     b2 = Booster.load_from_string(txt)
     assert abs(float(b2.sigmoid) - 2.5) < 1e-6
+
+
+def test_ranker_validation_ndcg_and_early_stopping():
+    """Ranker validation now reports group-aware NDCG@k (evalAt) and early
+    stopping treats it as higher-is-better (the round-1 ADVICE inversion)."""
+    rng = np.random.default_rng(21)
+    n_groups, per = 120, 20
+    rows = []
+    for gid in range(n_groups):
+        q = rng.normal(size=4)
+        for _ in range(per):
+            x = rng.normal(size=4)
+            rel = float(np.clip(round(2 + 1.5 * (q @ x) / 4
+                                      + rng.normal() * 0.3), 0, 4))
+            rows.append({"features": np.concatenate([q, x]).astype(np.float32),
+                         "label": rel, "group": gid})
+    df = pd.DataFrame(rows)
+    df["isVal"] = (df["group"] % 5 == 0)
+    m = LightGBMRanker(numIterations=60, numLeaves=15, learningRate=0.1,
+                       groupCol="group", evalAt=[3, 5],
+                       validationIndicatorCol="isVal",
+                       earlyStoppingRound=8).fit(df)
+    evals = m._training_stats.evals
+    assert evals, "validation metrics must be recorded"
+    keys = set(evals[0]["valid_0"].keys())
+    assert keys == {"ndcg@3", "ndcg@5"}, keys
+    ndcgs = [e["valid_0"]["ndcg@3"] for e in evals]
+    bi = m.booster.best_iteration
+    assert bi >= 0
+    assert ndcgs[bi] == max(ndcgs), (bi, ndcgs)  # higher-is-better tracked
+    assert max(ndcgs) > 0.6
