@@ -101,6 +101,68 @@ def _apply_stage(t: torch.Tensor, stage: dict) -> torch.Tensor:
     raise ValueError(f"unknown image op {op!r}")
 
 
+def _apply_stage_batched(t: torch.Tensor, stage: dict) -> torch.Tensor:
+    """Batched (N, H, W, C) mirror of _apply_stage — uniform-shape batches
+    run every op as ONE device launch instead of a per-image Python loop."""
+    op = stage["op"]
+    if op == "resize":
+        h, w = int(stage["height"]), int(stage["width"])
+        x = t.permute(0, 3, 1, 2)
+        x = torch.nn.functional.interpolate(x, size=(h, w), mode="bilinear",
+                                            align_corners=False)
+        return x.permute(0, 2, 3, 1)
+    if op == "crop":
+        x, y = int(stage["x"]), int(stage["y"])
+        h, w = int(stage["height"]), int(stage["width"])
+        return t[:, y:y + h, x:x + w]
+    if op == "flip":
+        code = int(stage.get("flipCode", 1))
+        if code >= 1:
+            return torch.flip(t, dims=[2])
+        if code == 0:
+            return torch.flip(t, dims=[1])
+        return torch.flip(t, dims=[1, 2])
+    if op == "colorFormat":
+        fmt = stage.get("format", "gray")
+        if fmt in ("gray", "grayscale"):
+            if t.shape[-1] >= 3:
+                gray = (0.299 * t[..., 2] + 0.587 * t[..., 1]
+                        + 0.114 * t[..., 0])
+                return gray.unsqueeze(-1)
+            return t
+        if fmt == "bgr2rgb" and t.shape[-1] >= 3:
+            return t.flip(-1)
+        return t
+    if op in ("blur", "gaussianKernel"):
+        if op == "blur":
+            kh, kw = int(stage["height"]), int(stage["width"])
+            k = torch.ones(1, 1, kh, kw, dtype=t.dtype,
+                           device=t.device) / (kh * kw)
+        else:
+            size = int(stage.get("apertureSize", 3))
+            sigma = float(stage.get("sigma", 1.0))
+            ax = (torch.arange(size, dtype=t.dtype, device=t.device)
+                  - (size - 1) / 2.0)
+            g1 = torch.exp(-(ax ** 2) / (2 * sigma * sigma))
+            k = (g1[:, None] * g1[None, :])
+            k = (k / k.sum()).reshape(1, 1, size, size)
+            kh = kw = size
+        N, H, W, C = t.shape
+        x = t.permute(0, 3, 1, 2).reshape(N * C, 1, H, W)
+        x = torch.nn.functional.conv2d(x, k, padding=(kh // 2, kw // 2))
+        x = x[:, :, :H, :W].reshape(N, C, H, W).permute(0, 2, 3, 1)
+        return x
+    if op == "threshold":
+        return _apply_stage(t, stage)   # purely elementwise
+    if op == "normalize":
+        mean = torch.tensor(stage.get("mean", [0.0]), dtype=t.dtype,
+                            device=t.device)
+        std = torch.tensor(stage.get("std", [1.0]), dtype=t.dtype,
+                           device=t.device)
+        return (t / 255.0 - mean) / std
+    raise ValueError(f"unknown image op {op!r}")
+
+
 @register
 class ImageTransformer(Transformer):
     """Stage-list image pipeline. Stages added with fluent helpers
@@ -144,15 +206,35 @@ class ImageTransformer(Transformer):
         return self._add(op="normalize", mean=mean, std=std)
 
     def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
+        from ..utils.devices import default_device
         stages = self.get("stages") or []
-        as_uint8 = self.get("outputType") == "uint8"
+        keep_u8 = (self.get("outputType") == "uint8"
+                   and not any(s["op"] == "normalize" for s in stages))
+        vals = df[self.get("inputCol")].to_numpy()
         outs = []
-        for img in df[self.get("inputCol")]:
-            t = _to_tensor(img)
-            for st in stages:
-                t = _apply_stage(t, st)
-            outs.append(_to_array(t, as_uint8 and not any(
-                s["op"] == "normalize" for s in stages)))
+        # uniform-shape batches run each op as one device launch
+        uniform = (len(vals) > 1 and all(
+            isinstance(v, np.ndarray) and v.shape == vals[0].shape
+            for v in vals))
+        if uniform:
+            device = default_device("auto")
+            chunk = max(1, int((1 << 28) / max(1, vals[0].size * 4)))
+            for s0 in range(0, len(vals), chunk):
+                t = torch.from_numpy(
+                    np.stack(vals[s0:s0 + chunk])).float().to(device)
+                if t.ndim == 3:
+                    t = t.unsqueeze(-1)
+                for st in stages:
+                    t = _apply_stage_batched(t, st)
+                a = t.clamp(0, 255).cpu().numpy()
+                a = a.astype(np.uint8) if keep_u8 else a
+                outs.extend(list(a))
+        else:
+            for img in vals:
+                t = _to_tensor(img)
+                for st in stages:
+                    t = _apply_stage(t, st)
+                outs.append(_to_array(t, keep_u8))
         out = df.copy()
         out[self.get("outputCol")] = outs
         return out

@@ -197,3 +197,31 @@ def test_unroll_and_resize_stages():
     np.testing.assert_allclose(u[:16], img[0, :, 0].astype(np.float64))
     ub = UnrollBinaryImage(height=8, width=8).transform(df)["unrolled"].iloc[0]
     assert ub.shape == (8 * 8 * 3,)
+
+
+def test_image_transformer_batched_matches_per_image():
+    """The uniform-shape batched path must match the per-image path for
+    every op (resize/crop/flip/gray/blur/gaussian/threshold/normalize)."""
+    from mmlspark_amd.models.images import ImageTransformer
+    rng = np.random.default_rng(7)
+    imgs = [rng.integers(0, 255, size=(48, 40, 3)).astype(np.uint8)
+            for _ in range(6)]
+    mixed = imgs[:5] + [rng.integers(0, 255, size=(32, 32, 3)).astype(np.uint8)]
+
+    pipelines = [
+        lambda t: t.resize(24, 28).crop(2, 3, 16, 18).normalize(
+            mean=[0.4, 0.5, 0.6], std=[0.2, 0.3, 0.4]),
+        lambda t: t.flip(1).blur(3, 3).threshold(90.0, 255.0),
+        lambda t: t.colorFormat("gray").gaussianKernel(5, 1.2),
+        lambda t: t.colorFormat("bgr2rgb").flip(0).flip(-1),
+    ]
+    for make in pipelines:
+        t = make(ImageTransformer(inputCol="image", outputCol="o"))
+        batched = t.transform(pd.DataFrame({"image": imgs}))["o"]
+        # mixed shapes force the per-image fallback on the SAME transformer
+        per_img = t.transform(pd.DataFrame({"image": mixed}))["o"]
+        for i in range(5):
+            np.testing.assert_allclose(
+                np.asarray(batched.iloc[i], dtype=np.float64),
+                np.asarray(per_img.iloc[i], dtype=np.float64),
+                atol=1e-3)

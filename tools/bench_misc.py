@@ -98,6 +98,23 @@ def bench_sar():
                       "interactions": n_inter}))
 
 
+def bench_images():
+    """ImageTransformer stage-list pipeline (resize/crop/normalize) — the
+    opencv-module analog — images/s on device tensors."""
+    from mmlspark_amd.models.images import ImageTransformer
+    rng = np.random.default_rng(5)
+    n = 4096
+    imgs = [rng.integers(0, 255, size=(256, 256, 3)).astype(np.uint8)
+            for _ in range(n)]
+    df = pd.DataFrame({"image": imgs})
+    t = (ImageTransformer(inputCol="image", outputCol="out")
+         .resize(224, 224).crop(12, 12, 200, 200)
+         .normalize(mean=[0.485, 0.456, 0.406], std=[0.229, 0.224, 0.225]))
+    dt = timeit(lambda: t.transform(df), warmup=1, iters=3)
+    print(json.dumps({"bench": "image_transformer", "images_per_sec": n / dt,
+                      "pipeline": "resize224+crop200+normalize"}))
+
+
 def bench_serving_concurrent():
     import http.client
     import socket
@@ -157,6 +174,7 @@ if __name__ == "__main__":
     for name, fn in [("knn", bench_knn),
                      ("cknn", bench_conditional_knn),
                      ("iforest", bench_iforest),
+                     ("images", bench_images),
                      ("sar", bench_sar),
                      ("serving", bench_serving_concurrent)]:
         if which in ("all", name):
