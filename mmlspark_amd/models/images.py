@@ -27,8 +27,11 @@ def _to_tensor(img) -> torch.Tensor:
 
 
 def _to_array(t: torch.Tensor, as_uint8: bool) -> np.ndarray:
-    a = t.clamp(0, 255).cpu().numpy()
-    return a.astype(np.uint8) if as_uint8 else a
+    # clamp only on the uint8 path: float outputs (e.g. normalize, whose
+    # values are signed) must pass through untouched
+    if as_uint8:
+        return t.clamp(0, 255).cpu().numpy().astype(np.uint8)
+    return t.cpu().numpy()
 
 
 def _apply_stage(t: torch.Tensor, stage: dict) -> torch.Tensor:
@@ -226,8 +229,10 @@ class ImageTransformer(Transformer):
                     t = t.unsqueeze(-1)
                 for st in stages:
                     t = _apply_stage_batched(t, st)
-                a = t.clamp(0, 255).cpu().numpy()
-                a = a.astype(np.uint8) if keep_u8 else a
+                if keep_u8:
+                    a = t.clamp(0, 255).cpu().numpy().astype(np.uint8)
+                else:
+                    a = t.cpu().numpy()
                 outs.extend(list(a))
         else:
             for img in vals:

@@ -225,3 +225,15 @@ def test_image_transformer_batched_matches_per_image():
                 np.asarray(batched.iloc[i], dtype=np.float64),
                 np.asarray(per_img.iloc[i], dtype=np.float64),
                 atol=1e-3)
+
+
+def test_image_normalize_keeps_negative_values():
+    """normalize outputs are signed; they must not be clamped at 0."""
+    from mmlspark_amd.models.images import ImageTransformer
+    img = np.zeros((8, 8, 3), dtype=np.uint8)  # → (0/255 - mean)/std < 0
+    t = ImageTransformer(inputCol="image", outputCol="o").normalize(
+        mean=[0.485, 0.456, 0.406], std=[0.229, 0.224, 0.225])
+    for frame in (pd.DataFrame({"image": [img]}),              # per-image
+                  pd.DataFrame({"image": [img, img]})):        # batched
+        out = np.asarray(t.transform(frame)["o"].iloc[0])
+        assert out.min() < -1.0, out.min()
