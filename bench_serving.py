@@ -140,7 +140,8 @@ def concurrent_phase(model, X, args):
                 errs[ci] += 1
         conn.close()
 
-    threads = [__import__("threading").Thread(target=client, args=(i,))
+    import threading as _th
+    threads = [_th.Thread(target=client, args=(i,))
                for i in range(n_clients)]
     for t in threads:
         t.start()
