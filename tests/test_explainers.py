@@ -189,3 +189,31 @@ def test_explainer_save_load_with_dataframe_param(tmp_path, model_and_data):
                                   shap.get("backgroundData"))
     out = back.transform(df.head(2))
     assert np.stack(out["explanation"].to_numpy()).shape == (2, 1, 6)
+
+
+def test_tabular_shap_generic_model_fallback():
+    """A model WITHOUT score_matrix (any plain Transformer) must still be
+    explainable — the DataFrame fallback of _score_matrix."""
+    from mmlspark_amd.core.pipeline import Model
+
+    class LinModel(Model):
+        def _transform(self, df):
+            out = df.copy()
+            out["probability"] = [
+                np.array([0.0, float(r["f0"] * 2 + r["f1"])])
+                for _, r in df.iterrows()]
+            return out
+
+    rng = np.random.default_rng(0)
+    df = pd.DataFrame({"f0": rng.normal(size=8), "f1": rng.normal(size=8),
+                       "f2": rng.normal(size=8)})
+    shap = TabularSHAP(inputCols=["f0", "f1", "f2"], model=LinModel(),
+                       targetCol="probability", targetClasses=[1],
+                       backgroundData=df, numSamples=64, rowBatch=4)
+    out = shap.transform(df)
+    exp = np.stack(out[shap.get("outputCol")].to_numpy())
+    assert exp.shape == (8, 1, 4)
+    # linear model: phi_f0 ≈ 2*(x0 - E[x0]), phi_f2 ≈ 0
+    x0 = df["f0"].to_numpy()
+    np.testing.assert_allclose(exp[:, 0, 1], 2 * (x0 - x0.mean()), atol=0.2)
+    np.testing.assert_allclose(exp[:, 0, 3], 0, atol=0.15)
