@@ -84,9 +84,19 @@ class VectorSampler:
         self.background = background  # (b, d)
         self.rng = rng
 
-    def apply(self, instance: np.ndarray, states: np.ndarray) -> np.ndarray:
+    def apply(self, instance: np.ndarray, states: np.ndarray,
+              draws_per_state: int = 1) -> np.ndarray:
+        """With draws_per_state == len(background) (the mean-aggregation
+        path using the FULL background), cycle every background row once
+        per coalition — exact marginalization instead of sampling noise."""
         n = states.shape[0]
-        bidx = self.rng.integers(0, len(self.background), size=n)
+        if (draws_per_state > 1
+                and draws_per_state == len(self.background)
+                and n % draws_per_state == 0):
+            bidx = np.tile(np.arange(len(self.background)),
+                           n // draws_per_state)
+        else:
+            bidx = self.rng.integers(0, len(self.background), size=n)
         bg = self.background[bidx]
         return states * instance[None, :] + (1 - states) * bg
 

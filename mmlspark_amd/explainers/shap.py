@@ -23,6 +23,18 @@ from .sampler import (ImageSampler, TextSampler, VectorSampler,
                       sample_coalitions, slic_superpixels)
 
 
+def _mean_over_draws(scores, n_rows, nZ, B):
+    """Collapse B background draws per coalition to their mean: the head
+    (n_rows*nZ*B) perturbation scores reduce to n_rows*nZ coalition values;
+    the tail (background + full rows) passes through untouched."""
+    if B == 1:
+        return scores
+    head = scores[: n_rows * nZ * B]
+    tail = scores[n_rows * nZ * B:]
+    head = head.reshape(n_rows * nZ, B, -1).mean(axis=1)
+    return np.concatenate([head, tail])
+
+
 class KernelSHAPBase(LocalExplainer):
     def _default_samples(self, m):
         return 2 * m + 2048
@@ -73,6 +85,12 @@ class TabularSHAP(KernelSHAPBase):
         rng = np.random.default_rng(self.get("seed"))
         n_samp = self.get("numSamples") or self._default_samples(m)
         Z, w = sample_coalitions(m, n_samp, rng)
+        nZ = Z.shape[0]
+        # reference semantics: v(S) is the MEAN model output over background
+        # replacements (KernelSHAPBase.scala:69-93 aggregates per coalition);
+        # spend the leftover sample budget on background draws per coalition
+        B = max(1, min(len(bg), n_samp // max(1, nZ)))
+        Zrep = np.repeat(Z, B, axis=0)
         explanations = []
         batch = self.get("rowBatch")
         rows = df[cols].to_numpy(dtype=np.float64)
@@ -81,16 +99,17 @@ class TabularSHAP(KernelSHAPBase):
             frames = []
             for x in chunk:
                 sampler = VectorSampler(bg, rng)
-                pert = sampler.apply(x, Z)
+                pert = sampler.apply(x, Zrep, draws_per_state=B)
                 frames.append(pert)
             all_pert = np.concatenate(frames) if frames else np.zeros((0, m))
             # v_null from background mean prediction, v_full from the row
-            scores = self._score_matrix(
+            raw_scores = self._score_matrix(
                 np.concatenate([all_pert, bg, chunk]), cols)
-            nZ = Z.shape[0]
-            v_null = scores[len(all_pert):len(all_pert) + len(bg)].mean(axis=0)
+            scores = _mean_over_draws(raw_scores, len(chunk), nZ, B)
+            v_null = scores[len(chunk) * nZ:
+                            len(chunk) * nZ + len(bg)].mean(axis=0)
             explanations.extend(self._solve_batch(
-                Z, scores, nZ, len(all_pert), len(bg), w, v_null,
+                Z, scores, nZ, len(chunk) * nZ, len(bg), w, v_null,
                 len(chunk)))
         out = df.copy()
         out[self.get("outputCol")] = explanations
@@ -111,20 +130,26 @@ class VectorSHAP(KernelSHAPBase):
         rng = np.random.default_rng(self.get("seed"))
         n_samp = self.get("numSamples") or self._default_samples(m)
         Z, w = sample_coalitions(m, n_samp, rng)
+        nZ = Z.shape[0]
+        B = max(1, min(len(bg), n_samp // max(1, nZ)))
+        Zrep = np.repeat(Z, B, axis=0)
         rows = vector_column_to_matrix(df, fcol).astype(np.float64)
         explanations = []
         batch = self.get("rowBatch")
         for s in range(0, len(rows), batch):
             chunk = rows[s:s + batch]
-            pert_frames = [VectorSampler(bg, rng).apply(x, Z) for x in chunk]
+            pert_frames = [VectorSampler(bg, rng).apply(x, Zrep,
+                                                        draws_per_state=B)
+                           for x in chunk]
             all_pert = np.concatenate(pert_frames)
             full = np.concatenate([bg, chunk])
-            scores = self._score_matrix(
+            raw_scores = self._score_matrix(
                 np.concatenate([all_pert, full]).astype(np.float32))
-            nZ = Z.shape[0]
-            v_null = scores[len(all_pert):len(all_pert) + len(bg)].mean(axis=0)
+            scores = _mean_over_draws(raw_scores, len(chunk), nZ, B)
+            v_null = scores[len(chunk) * nZ:
+                            len(chunk) * nZ + len(bg)].mean(axis=0)
             explanations.extend(self._solve_batch(
-                Z, scores, nZ, len(all_pert), len(bg), w, v_null,
+                Z, scores, nZ, len(chunk) * nZ, len(bg), w, v_null,
                 len(chunk)))
         out = df.copy()
         out[self.get("outputCol")] = explanations
