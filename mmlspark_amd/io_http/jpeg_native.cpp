@@ -481,7 +481,6 @@ torch::Tensor decode_jpeg_native(py::bytes data_b) {
         px[2] = (uint8_t)std::min(255.f, std::max(0.f, b + 0.5f));
       }
     }
-    (void)pwr;
   }
   return out;
 }
