@@ -538,6 +538,11 @@ class ProcessServingCluster:
         proc = subprocess.Popen(argv, stdout=subprocess.PIPE, text=True,
                                 env=env)
         line = proc.stdout.readline()  # {"ready": true, "port": ...}
+        if not line:
+            rc = proc.poll()
+            raise RuntimeError(
+                f"serving worker {i} exited before reporting ready "
+                f"(rc={rc}); see its stderr above")
         info = json.loads(line)
         return _WorkerProc(proc, self.host, int(info["port"]),
                            f"{self.name}-{i}")
