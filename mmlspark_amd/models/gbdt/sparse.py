@@ -229,13 +229,17 @@ class SparseShard:
         return self.binv.device
 
 
-def bin_csr(csr: CsrMatrix, mapper: BinMapper, chunk: int = 4_000_000
-            ) -> SparseShard:
+def bin_csr(csr: CsrMatrix, mapper: BinMapper,
+            chunk: Optional[int] = None) -> SparseShard:
     """Bin the stored entries; zero_bin[f] mirrors the dense rule
     bin(x) = searchsorted(bounds_f, x) evaluated at x = 0."""
     ub = mapper.upper_bounds.to(csr.device)
     nb = mapper.n_bins
     nnz = csr.nnz
+    if chunk is None:
+        # the per-chunk gathered boundary matrix is (chunk, nb-1) floats —
+        # keep the transient around 256 MB
+        chunk = max(65536, (1 << 26) // max(1, nb - 1))
     binv = torch.empty(nnz, dtype=torch.uint8, device=csr.device)
     for s in range(0, max(nnz, 1), chunk):
         e = min(nnz, s + chunk)
