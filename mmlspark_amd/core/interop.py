@@ -63,6 +63,20 @@ def arrow_to_pandas(table) -> pd.DataFrame:
         typ = col.type
         if pa.types.is_fixed_size_list(typ) or pa.types.is_list(typ) \
                 or pa.types.is_large_list(typ):
+            arr = col.combine_chunks()
+            if arr.null_count == 0 and pa.types.is_floating(typ.value_type) \
+                    or (arr.null_count == 0
+                        and pa.types.is_integer(typ.value_type)):
+                # zero-copy-ish: one flat buffer + per-row views
+                flat = np.asarray(arr.values).astype(np.float32, copy=False)
+                if pa.types.is_fixed_size_list(typ):
+                    rows = list(flat.reshape(-1, typ.list_size))
+                else:
+                    offs = np.asarray(arr.offsets)
+                    rows = [flat[offs[i]:offs[i + 1]]
+                            for i in range(len(arr))]
+                out[name] = pd.Series(rows, dtype=object)
+                continue
             vals = col.to_pylist()
             out[name] = pd.Series(
                 [None if v is None else np.asarray(v, dtype=np.float32)
