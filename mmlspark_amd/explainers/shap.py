@@ -3,9 +3,14 @@ TabularSHAP:16, VectorSHAP, ImageSHAP, TextSHAP).
 
 Per row: sample coalitions (exact small-|z| enumeration, default budget
 2*m+2048 — KernelSHAPBase.scala:135), build perturbed samples, score ALL of
-them through the model in one batched transform (GPU forest kernel), then
-solve the constrained weighted least squares per target class.  Output per
-row: (n_classes, m+1) — [base_value, phi_1..phi_m] per class."""
+them through the model in one batched pass (GBDT models skip the DataFrame
+entirely via score_matrix), MEAN-aggregate each coalition over background
+replacements (the reference's per-coalition aggregation,
+KernelSHAPBase.scala:69-93 — leftover sample budget becomes background
+draws, cycling the full background exactly when it fits), then solve the
+constrained weighted least squares for every row and class at once (the
+coalition design is shared, so one factorization serves the whole chunk).
+Output per row: (n_classes, m+1) — [base_value, phi_1..phi_m] per class."""
 from __future__ import annotations
 
 
