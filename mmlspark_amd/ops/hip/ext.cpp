@@ -56,7 +56,7 @@ void launch_csr_hist_fixed_lds(const long*, const int*, const unsigned char*,
                                long long*, hipStream_t);
 void launch_csr_partition(const long*, const int*, const unsigned char*,
                           const int*, long, int, int, int, const unsigned*,
-                          int*, int*, int*, hipStream_t);
+                          unsigned char*, int*, int*, int*, hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -387,12 +387,14 @@ std::tuple<torch::Tensor, torch::Tensor> csr_partition_rows(
   CHECK_DEV(rows); CHECK_CONTIG(rows);
   const long m = rows.numel();
   auto out = torch::empty({m}, rows.options());
+  auto pred = torch::empty({m}, rows.options().dtype(torch::kUInt8));
   auto scratch = torch::empty({4096}, rows.options());
   auto total = torch::zeros({1}, rows.options());
   if (m > 0) {
     launch_csr_partition(indptr.data_ptr<long>(), col.data_ptr<int>(),
                          binv.data_ptr<unsigned char>(), rows.data_ptr<int>(),
                          m, (int)feature, (int)zero_bin, (int)thr, nullptr,
+                         pred.data_ptr<unsigned char>(),
                          out.data_ptr<int>(), scratch.data_ptr<int>(),
                          total.data_ptr<int>(), cur_stream());
   }

@@ -1765,13 +1765,17 @@ __global__ void csr_part_count_k(const long* __restrict__ indptr,
                                  const int* __restrict__ rows, long m,
                                  int feature, int zero_bin, int thr,
                                  const unsigned* __restrict__ cat_bits,
-                                 long chunk, int* __restrict__ block_counts) {
+                                 long chunk, int* __restrict__ block_counts,
+                                 unsigned char* __restrict__ pred) {
   const long start = (long)blockIdx.x * chunk;
   const long end = min(start + chunk, m);
   int cnt = 0;
-  for (long i = start + threadIdx.x; i < end; i += blockDim.x)
-    cnt += (int)csr_goes_left(indptr, col, binv, rows[i], feature, zero_bin,
-                              thr, cat_bits);
+  for (long i = start + threadIdx.x; i < end; i += blockDim.x) {
+    const bool left = csr_goes_left(indptr, col, binv, rows[i], feature,
+                                    zero_bin, thr, cat_bits);
+    pred[i] = (unsigned char)left;  // scatter pass reuses the predicate
+    cnt += (int)left;
+  }
   __shared__ int sh[256];
   sh[threadIdx.x] = cnt;
   __syncthreads();
@@ -1782,12 +1786,8 @@ __global__ void csr_part_count_k(const long* __restrict__ indptr,
   if (threadIdx.x == 0) block_counts[blockIdx.x] = sh[0];
 }
 
-__global__ void csr_part_scatter_k(const long* __restrict__ indptr,
-                                   const int* __restrict__ col,
-                                   const unsigned char* __restrict__ binv,
+__global__ void csr_part_scatter_k(const unsigned char* __restrict__ pred,
                                    const int* __restrict__ rows, long m,
-                                   int feature, int zero_bin, int thr,
-                                   const unsigned* __restrict__ cat_bits,
                                    long chunk,
                                    const int* __restrict__ block_offsets,
                                    const int* __restrict__ total_left,
@@ -1810,8 +1810,7 @@ __global__ void csr_part_scatter_k(const long* __restrict__ indptr,
     bool valid = i < end, left = false;
     if (valid) {
       r = rows[i];
-      left = csr_goes_left(indptr, col, binv, r, feature, zero_bin, thr,
-                           cat_bits);
+      left = pred[i] != 0;
     }
     const unsigned long long mask_l = __ballot(valid && left);
     const unsigned long long mask_r = __ballot(valid && !left);
@@ -1848,7 +1847,8 @@ extern "C" void launch_csr_partition(const long* indptr, const int* col,
                                      const unsigned char* binv,
                                      const int* rows, long m, int feature,
                                      int zero_bin, int thr,
-                                     const unsigned* cat_bits, int* out,
+                                     const unsigned* cat_bits,
+                                     unsigned char* pred, int* out,
                                      int* scratch, int* total_left,
                                      hipStream_t stream) {
   if (m == 0) return;
@@ -1860,11 +1860,10 @@ extern "C" void launch_csr_partition(const long* indptr, const int* col,
   }
   hipLaunchKernelGGL(csr_part_count_k, dim3((unsigned)blocks), dim3(256), 0,
                      stream, indptr, col, binv, rows, m, feature, zero_bin,
-                     thr, cat_bits, chunk, scratch);
+                     thr, cat_bits, chunk, scratch, pred);
   hipLaunchKernelGGL(part_scan_k, dim3(1), dim3(256), 0, stream, scratch,
                      (int)blocks, total_left);
   hipLaunchKernelGGL(csr_part_scatter_k, dim3((unsigned)blocks), dim3(256),
-                     0, stream, indptr, col, binv, rows, m, feature,
-                     zero_bin, thr, cat_bits, chunk, scratch, total_left,
+                     0, stream, pred, rows, m, chunk, scratch, total_left,
                      out);
 }
