@@ -1711,17 +1711,23 @@ extern "C" void launch_csr_hist_fixed_lds(const long* indptr, const int* col,
   const long min_bx = (m + (1l << 19) - 1) >> 19;  // ≤2^19 rows per block
   if (bx < min_bx) bx = min_bx;
   if (bx < 1) bx = 1;
-  // A/B-measured on MI355X: FW=16 (half the redundant entry passes, half
-  // the LDS occupancy) ties FW=8 at 10M×100 nnz=20 — the kernel is
-  // LDS-atomic-bound, not read-bound — so default to the safer FW=8
+  // A/B-measured on MI355X at 10M×100 nnz=20: FW=16 (half the passes,
+  // half the occupancy) ties FW=8; FW=4 (more occupancy, double passes)
+  // is ~1.7× SLOWER — reads win over latency hiding below FW=8, atomics
+  // cap above it.  FW=8 is the sweet spot and the default
   static const int fw_env = [] {
     const char* e = getenv("MMLSPARK_AMD_SPARSE_FW");
     return e ? atoi(e) : 8;
   }();
-  const int FW = (fw_env == 16) ? 16 : 8;
+  const int FW = (fw_env == 16) ? 16 : (fw_env == 4 ? 4 : 8);
   const int by = (nf + FW - 1) / FW;
   const size_t lds_bytes = (size_t)FW * n_bins * 2 * sizeof(long long);
-  if (FW == 8)
+  if (FW == 4)
+    hipLaunchKernelGGL(csr_hist_fixed_lds_k<4>,
+                       dim3((unsigned)bx, (unsigned)by), dim3(256),
+                       lds_bytes, stream, indptr, col, binv, gq, hq, rows,
+                       m, hist, n_bins, nf, tot);
+  else if (FW == 8)
     hipLaunchKernelGGL(csr_hist_fixed_lds_k<8>,
                        dim3((unsigned)bx, (unsigned)by), dim3(256),
                        lds_bytes, stream, indptr, col, binv, gq, hq, rows,
