@@ -123,3 +123,27 @@ def test_fit_transform_accepts_spark_dataframe():  # pragma: no cover
     out = m_spark.transform(sdf)
     assert out.__class__.__module__.startswith("pyspark.sql")
     assert "probability" in out.columns
+
+
+def test_arrow_nulls_and_chunked_tables():
+    """Null list entries fall back to the slow path (None preserved);
+    multi-chunk tables combine correctly through the flat-buffer path."""
+    rows = [np.arange(4, dtype=np.float32), None,
+            np.ones(4, dtype=np.float32)]
+    t = pa.table({"features": pa.array(
+        [None if r is None else r.tolist() for r in rows],
+        type=pa.list_(pa.float32()))})
+    back = arrow_to_pandas(t)
+    assert back["features"].iloc[1] is None
+    np.testing.assert_allclose(back["features"].iloc[0], rows[0])
+
+    # chunked: two record batches concatenated
+    df = pd.DataFrame({"features": [np.full(3, i, dtype=np.float32)
+                                    for i in range(10)]})
+    t1 = pandas_to_arrow(df.iloc[:6])
+    t2 = pandas_to_arrow(df.iloc[6:])
+    both = pa.concat_tables([t1, t2])
+    assert both.column(0).num_chunks == 2
+    back = arrow_to_pandas(both)
+    assert len(back) == 10
+    np.testing.assert_allclose(back["features"].iloc[9], np.full(3, 9.0))

@@ -409,3 +409,29 @@ def test_process_serving_cluster_failover(tmp_path, binary_df):
                           timeout=20).status_code == 200
     finally:
         cluster.stop()
+
+
+def test_worker_cli_scorer_mode(tmp_path, binary_df):
+    """`python -m mmlspark_amd.serving.worker --scorer`: the low-latency
+    scorer path served by a standalone worker process."""
+    import subprocess
+    import sys
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    model = LightGBMClassifier(numIterations=5, numLeaves=7).fit(binary_df)
+    mdir = str(tmp_path / "m")
+    model.save(mdir)
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "mmlspark_amd.serving.worker",
+         "--model", mdir, "--port", "0", "--mode", "continuous", "--scorer"],
+        stdout=subprocess.PIPE, text=True)
+    try:
+        info = json.loads(proc.stdout.readline())
+        assert info["ready"]
+        x = binary_df["features"].iloc[0].tolist()
+        r = requests.post(f"http://127.0.0.1:{info['port']}/",
+                          json={"features": x}, timeout=20)
+        assert r.status_code == 200
+        assert "score" in r.json()
+    finally:
+        proc.terminate()
+        proc.wait(timeout=10)
