@@ -135,3 +135,71 @@ def test_checkpoints_with_num_batches(tmp_path):
     assert m.booster.num_trees == 8  # 4 per batch — batch 2 actually ran
     assert os.path.exists(os.path.join(ckdir, "batch0", "checkpoint.json"))
     assert os.path.exists(os.path.join(ckdir, "batch1", "checkpoint.json"))
+
+
+def _dist_ck_worker(rank, world, port, ckdir, n_iters, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        torch.distributed.init_process_group(
+            "gloo", init_method=f"tcp://127.0.0.1:{port}",
+            rank=rank, world_size=world)
+        from mmlspark_amd.models.gbdt.objectives import make_objective
+        from mmlspark_amd.models.gbdt.trainer import TrainConfig, train_booster
+        from mmlspark_amd.parallel.comm import Comm
+        rng = np.random.default_rng(0)
+        X = rng.normal(size=(4000, 10)).astype(np.float32)
+        w = rng.normal(size=10)
+        y = ((X @ w) > 0).astype(np.float32)
+        sl = slice(rank * 2000, (rank + 1) * 2000)
+        cfg = TrainConfig(num_iterations=n_iters, num_leaves=15, seed=2)
+        b, _ = train_booster(torch.from_numpy(X[sl]), torch.from_numpy(y[sl]),
+                             cfg, make_objective("binary"), Comm(),
+                             checkpoint_dir=ckdir, checkpoint_every=2)
+        q.put((rank, b.save_to_string()))
+        torch.distributed.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"ERROR: {e!r}"))
+
+
+@pytest.mark.timeout(240)
+def test_distributed_checkpoint_resume_ws2(tmp_path):
+    """Gang restart at world_size=2: run to iteration 4 (checkpointing),
+    relaunch the gang with num_iterations=8 — every rank resumes from the
+    shared checkpoint, stays bit-identical across ranks, and ends with 8
+    trees."""
+    import torch.multiprocessing as mp
+    ckdir = str(tmp_path / "ck")
+
+    def run(n_iters, port):
+        ctx = mp.get_context("spawn")
+        q = ctx.Queue()
+        ps = [ctx.Process(target=_dist_ck_worker,
+                          args=(r, 2, port, ckdir, n_iters, q))
+              for r in range(2)]
+        for p in ps:
+            p.start()
+        res = {}
+        for _ in range(2):
+            r, s = q.get(timeout=200)
+            res[r] = s
+        for p in ps:
+            p.join(timeout=30)
+        assert not any(str(s).startswith("ERROR") for s in res.values()), res
+        return res
+
+    res4 = run(4, 29961)      # "crashes" after 4 iterations (checkpointed)
+    assert res4[0] == res4[1]
+    from mmlspark_amd.models.gbdt.trainer import load_checkpoint
+    b_ck, it = load_checkpoint(ckdir)
+    assert it == 4 and b_ck.num_trees == 4
+
+    res8 = run(8, 29962)      # gang restart resumes 5..8
+    assert res8[0] == res8[1]
+    from mmlspark_amd.models.gbdt.booster import Booster
+    b = Booster.load_from_string(res8[0])
+    assert b.num_trees == 8
+    # the first 4 trees are exactly the pre-crash trees
+    import json as _json
+    t_resumed = _json.loads(res8[0])["trees"][:4]
+    t_before = _json.loads(res4[0])["trees"]
+    assert t_resumed == t_before
