@@ -969,6 +969,11 @@ def train_booster(X: torch.Tensor, y: torch.Tensor, cfg: TrainConfig,
             t0 = time.perf_counter()
             booster_now = session.booster()
             entry = {"iteration": it}
+            if cfg.is_provide_training_metric:
+                # per-iteration TRAIN metrics (isProvideTrainingMetric,
+                # TrainUtils.scala:117-128 logs train eval each iteration)
+                entry["training"] = metrics_fn(booster_now, X, y, weight,
+                                               comm)
             score = None
             score_name = None
             for vi, (Xv, yv, wv) in enumerate(valid_sets or []):

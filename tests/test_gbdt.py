@@ -471,3 +471,16 @@ def test_ranker_validation_ndcg_and_early_stopping():
     assert bi >= 0
     assert ndcgs[bi] == max(ndcgs), (bi, ndcgs)  # higher-is-better tracked
     assert max(ndcgs) > 0.6
+
+
+def test_is_provide_training_metric(binary_df):
+    """isProvideTrainingMetric logs per-iteration TRAIN metrics
+    (TrainUtils.scala:117-128 parity) even without a validation split."""
+    m = LightGBMClassifier(numIterations=5, numLeaves=7,
+                           isProvideTrainingMetric=True).fit(binary_df)
+    evals = m._training_stats.evals
+    assert len(evals) == 5
+    assert "training" in evals[0]
+    assert "binary_logloss" in evals[0]["training"]
+    losses = [e["training"]["binary_logloss"] for e in evals]
+    assert losses[-1] < losses[0]  # training loss decreases
