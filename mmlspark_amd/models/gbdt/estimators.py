@@ -157,6 +157,7 @@ class _GBDTParams(Params):
             feature_fraction=self.get("featureFraction"),
             bagging_fraction=self.get("baggingFraction"),
             bagging_freq=self.get("baggingFreq"),
+            bagging_seed=self.get("baggingSeed"),
             boosting=self.get("boostingType"),
             top_rate=self.get("topRate"),
             other_rate=self.get("otherRate"),

@@ -58,6 +58,7 @@ class TrainConfig:
     feature_fraction: float = 1.0
     bagging_fraction: float = 1.0
     bagging_freq: int = 0
+    bagging_seed: int = 3        # baggingSeed (LightGBM default 3)
     boosting: str = "gbdt"  # gbdt | rf | dart | goss
     top_rate: float = 0.2       # goss
     other_rate: float = 0.1     # goss
@@ -807,8 +808,8 @@ class TrainingSession:
         grad, hess = self.objective.grad_hess(preds_used, self.y, self.weight)
         self.stats.grad_s += time.perf_counter() - t0
 
-        # ---- row sampling ---------------------------------------------------
-        gen.manual_seed(cfg.seed * 104729 + it * 31 + self.comm.rank)
+        # ---- row sampling (baggingSeed drives bagging/GOSS draws) -----------
+        gen.manual_seed(cfg.bagging_seed * 104729 + it * 31 + self.comm.rank)
         if goss_mode and it >= 1:
             rows_root = _goss_sample(grad, hess, cfg, gen)
         elif (cfg.pos_bagging_fraction < 1.0 or cfg.neg_bagging_fraction < 1.0) \

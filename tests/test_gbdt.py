@@ -484,3 +484,14 @@ def test_is_provide_training_metric(binary_df):
     assert "binary_logloss" in evals[0]["training"]
     losses = [e["training"]["binary_logloss"] for e in evals]
     assert losses[-1] < losses[0]  # training loss decreases
+
+
+def test_bagging_seed_controls_sampling(binary_df):
+    """baggingSeed must change bagging draws (same seed → same model)."""
+    kw = dict(numIterations=8, numLeaves=15, baggingFraction=0.6,
+              baggingFreq=1, seed=0)
+    a = LightGBMClassifier(baggingSeed=1, **kw).fit(binary_df)
+    b = LightGBMClassifier(baggingSeed=1, **kw).fit(binary_df)
+    c = LightGBMClassifier(baggingSeed=2, **kw).fit(binary_df)
+    assert a.booster.save_to_string() == b.booster.save_to_string()
+    assert a.booster.save_to_string() != c.booster.save_to_string()
