@@ -33,16 +33,24 @@ class LIMEBase(LocalExplainer):
         w = np.exp(-(d / self.get("kernelWidth")) ** 2)
         return np.sqrt(w)
 
-    def _fit_states(self, states, scores):
+    def _fit_states(self, states, scores, r2_out=None):
+        """Weighted lasso per class; optionally records the weighted R² of
+        the local surrogate (the reference's metricsCol output,
+        LIMEBase.scala r2 column)."""
         w = self._weights(states)
         coefs = []
+        r2s = []
+        Xs = torch.from_numpy(states.astype(np.float64))
+        wt = torch.from_numpy(w)
         for k in range(scores.shape[1]):
-            res = lasso_regression(
-                torch.from_numpy(states.astype(np.float64)),
-                torch.from_numpy(scores[:, k].astype(np.float64)),
-                alpha=self.get("regularization"),
-                sample_weight=torch.from_numpy(w))
+            yv = torch.from_numpy(scores[:, k].astype(np.float64))
+            res = lasso_regression(Xs, yv, alpha=self.get("regularization"),
+                                   sample_weight=wt)
             coefs.append(res.coefficients.numpy())
+            if r2_out is not None:
+                r2s.append(float(res.r_squared))
+        if r2_out is not None:
+            r2_out.append(np.asarray(r2s))
         return np.stack(coefs)  # (n_classes, m)
 
 
@@ -59,6 +67,7 @@ class TabularLIME(LIMEBase):
         rng = np.random.default_rng(self.get("seed"))
         n_samp = self.get("numSamples") or self._default_samples(m)
         explanations = []
+        r2s = []
         rows = df[cols].to_numpy(dtype=np.float64)
         batch = self.get("rowBatch")
         for s0 in range(0, len(rows), batch):
@@ -71,10 +80,13 @@ class TabularLIME(LIMEBase):
             o = 0
             for st, p_ in zip(states_l, perts):
                 explanations.append(
-                    self._fit_states(st, scores[o:o + len(p_)]))
+                    self._fit_states(st, scores[o:o + len(p_)],
+                                     r2_out=r2s))
                 o += len(p_)
         out = df.copy()
         out[self.get("outputCol")] = explanations
+        if self.get("metricsCol"):
+            out[self.get("metricsCol")] = r2s
         return out
 
 
@@ -93,6 +105,7 @@ class VectorLIME(LIMEBase):
         n_samp = self.get("numSamples") or self._default_samples(m)
         rows = vector_column_to_matrix(df, fcol).astype(np.float64)
         explanations = []
+        r2s = []
         batch = self.get("rowBatch")
         for s0 in range(0, len(rows), batch):
             chunk = rows[s0:s0 + batch]
@@ -103,10 +116,13 @@ class VectorLIME(LIMEBase):
             o = 0
             for st, p_ in zip(states_l, perts):
                 explanations.append(
-                    self._fit_states(st, scores[o:o + len(p_)]))
+                    self._fit_states(st, scores[o:o + len(p_)],
+                                     r2_out=r2s))
                 o += len(p_)
         out = df.copy()
         out[self.get("outputCol")] = explanations
+        if self.get("metricsCol"):
+            out[self.get("metricsCol")] = r2s
         return out
 
 
@@ -117,7 +133,7 @@ class TextLIME(LIMEBase):
 
     def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
         rng = np.random.default_rng(self.get("seed"))
-        explanations, tokens_col = [], []
+        explanations, tokens_col, r2s = [], [], []
         for _, row in df.iterrows():
             tokens = str(row[self.get("inputCol")]).split()
             tokens_col.append(tokens)
@@ -127,9 +143,12 @@ class TextLIME(LIMEBase):
             texts = TextSampler(tokens).apply(states)
             scores = self._score_samples(
                 pd.DataFrame({self.get("inputCol"): texts}))
-            explanations.append(self._fit_states(states, scores))
+            explanations.append(self._fit_states(states, scores,
+                                                 r2_out=r2s))
         out = df.copy()
         out[self.get("outputCol")] = explanations
+        if self.get("metricsCol"):
+            out[self.get("metricsCol")] = r2s
         out[self.get("tokensCol")] = tokens_col
         return out
 
@@ -143,7 +162,7 @@ class ImageLIME(LIMEBase):
 
     def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
         rng = np.random.default_rng(self.get("seed"))
-        explanations, segs_col = [], []
+        explanations, segs_col, r2s = [], [], []
         for _, row in df.iterrows():
             img = np.asarray(row[self.get("inputCol")])
             segments = slic_superpixels(img, self.get("cellSize"),
@@ -155,9 +174,12 @@ class ImageLIME(LIMEBase):
             imgs = ImageSampler(img, segments).apply(states)
             scores = self._score_samples(
                 pd.DataFrame({self.get("inputCol"): imgs}))
-            explanations.append(self._fit_states(states, scores))
+            explanations.append(self._fit_states(states, scores,
+                                                 r2_out=r2s))
         out = df.copy()
         out[self.get("outputCol")] = explanations
+        if self.get("metricsCol"):
+            out[self.get("metricsCol")] = r2s
         out[self.get("superpixelCol")] = segs_col
         return out
 

@@ -218,3 +218,23 @@ def test_tabular_shap_generic_model_fallback():
     np.testing.assert_allclose(exp[:, 0, 1], 2 * (x0 - x0.mean()),
                                atol=0.1)
     np.testing.assert_allclose(exp[:, 0, 3], 0, atol=0.1)
+
+
+def test_lime_metrics_col_r2():
+    """metricsCol emits the local surrogate's per-class R²
+    (LIMEBase's r2 metrics output)."""
+    from mmlspark_amd.models.gbdt.estimators import LightGBMClassifier
+    rng = np.random.default_rng(2)
+    X = rng.normal(size=(800, 5)).astype(np.float32)
+    cols = [f"f{i}" for i in range(5)]
+    df = pd.DataFrame(X, columns=cols)
+    df["label"] = (X[:, 0] > 0).astype(np.float32)
+    model = LightGBMClassifier(featureCols=cols, numIterations=20,
+                               numLeaves=7).fit(df)
+    lime = TabularLIME(inputCols=cols, model=model, targetCol="probability",
+                       targetClasses=[1], backgroundData=df.head(100),
+                       numSamples=400, metricsCol="r2")
+    out = lime.transform(df.head(6))
+    r2 = np.stack(out["r2"].to_numpy())
+    assert r2.shape == (6, 1)
+    assert (r2 > 0.2).all() and (r2 <= 1.0 + 1e-9).all(), r2
