@@ -120,7 +120,9 @@ class _VWParams(Params):
     lossFunction = Param("lossFunction", "squared|logistic|hinge", None)
     batchSize = Param("batchSize", "SGD minibatch size (GPU hogwild window)",
                       4096, toInt)
-    hashSeed = Param("hashSeed", "murmur seed", 0, toInt)
+    hashSeed = Param("hashSeed", "murmur seed (accepted for API compat; hashing "
+                     "happens in VowpalWabbitFeaturizer via its own seed param — "
+                     "the estimator consumes pre-hashed vectors)", 0, toInt)
     adaptive = Param("adaptive", "per-weight adaptive (AdaGrad) rates "
                      "(--adaptive)", True, toBool)
     normalized = Param("normalized", "scale updates by running max|x| per "
@@ -660,4 +662,14 @@ class VowpalWabbitContextualBanditModel(_VWModelBase):
         out = df.copy()
         out["predictedCosts"] = scores_col
         out[self.get("predictionCol")] = chosen_col
+        # epsilon-greedy action distribution (VW --epsilon pmf: the greedy
+        # action gets 1-eps+eps/K, every action eps/K)
+        eps = float(self.get("epsilon"))
+        probs_col = []
+        for scores, chosen in zip(scores_col, chosen_col):
+            K = max(len(scores), 1)
+            p_ = np.full(K, eps / K, dtype=np.float32)
+            p_[chosen - 1] += 1.0 - eps
+            probs_col.append(p_)
+        out["probabilities"] = probs_col
         return out

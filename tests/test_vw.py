@@ -369,3 +369,31 @@ def test_multipass_holdout_early_terminate():
         passThroughArgs="--passes 3 --holdout_off", learningRate=0.5,
         numBits=12).fit(df)
     assert int(m3.getPerformanceStatistics().iloc[0]["ipassCurrent"]) == 3
+
+
+def test_contextual_bandit_epsilon_pmf():
+    """The CB model emits the epsilon-greedy action distribution
+    (VW --epsilon pmf): greedy gets 1-eps+eps/K, the rest eps/K."""
+    from mmlspark_amd.models.vw.estimators import VowpalWabbitContextualBandit
+    rng = np.random.default_rng(0)
+    size = 1 << 14
+    rows = []
+    for _ in range(300):
+        shared = SparseVector(size, [100], [1.0])
+        actions = [SparseVector(size, [2000 + a], [1.0]) for a in range(3)]
+        chosen = int(rng.integers(1, 4))
+        cost = float(chosen != 2)  # action 2 is best
+        rows.append({"shared": shared, "features": actions,
+                     "chosenAction": chosen, "cost": cost,
+                     "probability": 1 / 3})
+    df = pd.DataFrame(rows)
+    m = VowpalWabbitContextualBandit(numPasses=4, numBits=14,
+                                     learningRate=0.5, epsilon=0.3).fit(df)
+    out = m.transform(df.head(10))
+    assert "probabilities" in out.columns
+    for _, r in out.iterrows():
+        p = np.asarray(r["probabilities"])
+        assert p.shape == (3,)
+        np.testing.assert_allclose(p.sum(), 1.0, atol=1e-6)
+        assert abs(p.max() - (0.7 + 0.1)) < 1e-6
+        assert int(np.argmax(p)) + 1 == r["prediction"]
