@@ -304,3 +304,62 @@ def test_fluent_api_monkeypatch(binary_df):
     assert list(out.columns) == ["prediction", "label"]
     acc = (out["prediction"] == out["label"]).mean()
     assert acc > 0.8
+
+
+def test_fast_vector_assembler():
+    from mmlspark_amd.stages.featurize import FastVectorAssembler
+    from mmlspark_amd.core.schema import SparseVector
+    df = pd.DataFrame({
+        "a": [1.0, 2.0, 3.0],
+        "v": [np.array([1, 2], dtype=np.float32)] * 3,
+        "s": [SparseVector(4, [1], [5.0]), SparseVector(4, [0, 3], [1.0, 2.0]),
+              SparseVector(4, [], [])],
+    })
+    out = FastVectorAssembler(inputCols=["a", "v", "s"], outputCol="f") \
+        .transform(df)
+    f = np.stack(out["f"].to_numpy())
+    assert f.shape == (3, 7)
+    np.testing.assert_allclose(f[0], [1, 1, 2, 0, 5, 0, 0])
+    np.testing.assert_allclose(f[1], [2, 1, 2, 1, 0, 0, 2])
+    np.testing.assert_allclose(f[2], [3, 1, 2, 0, 0, 0, 0])
+
+
+def test_shared_variable_and_singleton():
+    from mmlspark_amd.utils.shared import (SharedSingleton, SharedVariable,
+                                           clear_pool)
+    clear_pool()
+    calls = []
+
+    def make():
+        calls.append(1)
+        return object()
+
+    sv = SharedVariable(make)
+    assert sv.get() is sv.get()
+    assert len(calls) == 1
+    # singleton: two instances from the same ctor share one value
+    s1, s2 = SharedSingleton(make), SharedSingleton(make)
+    assert s1.get() is s2.get()
+    # pickling carries the key, not the value (module-level ctor so the
+    # thunk itself pickles)
+    import pickle
+    svp = SharedVariable(dict)
+    first = svp.get()
+    sv2 = pickle.loads(pickle.dumps(svp))
+    assert sv2.get() is first
+
+
+def test_cluster_topology(monkeypatch):
+    from mmlspark_amd.parallel.cluster import (get_driver_host,
+                                               get_num_executors,
+                                               get_topology)
+    monkeypatch.setenv("WORLD_SIZE", "8")
+    monkeypatch.setenv("RANK", "3")
+    monkeypatch.setenv("LOCAL_RANK", "3")
+    monkeypatch.setenv("LOCAL_WORLD_SIZE", "4")
+    monkeypatch.setenv("MASTER_ADDR", "10.0.0.1")
+    t = get_topology()
+    assert (t.world_size, t.rank, t.local_rank) == (8, 3, 3)
+    assert t.n_nodes == 2
+    assert get_num_executors() == 8
+    assert get_driver_host() == "10.0.0.1"
