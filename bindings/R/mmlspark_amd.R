@@ -536,6 +536,13 @@ ml_explode <- function(inputCol = NULL, outputCol = NULL) {
   stage
 }
 
+ml_fast_vector_assembler <- function(inputCols = NULL, outputCol = NULL) {
+  stage <- mmlspark_amd$stages$featurize$FastVectorAssembler()
+  if (!is.null(inputCols)) stage$set("inputCols", inputCols)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  stage
+}
+
 ml_featurize <- function(inputCols = NULL, outputCol = NULL, oneHotEncodeCategoricals = NULL, numFeatures = NULL) {
   stage <- mmlspark_amd$stages$featurize$Featurize()
   if (!is.null(inputCols)) stage$set("inputCols", inputCols)
