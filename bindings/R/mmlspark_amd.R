@@ -187,6 +187,19 @@ ml_break_sentence <- function(url = NULL, subscriptionKey = NULL, subscriptionKe
   stage
 }
 
+ml_cntk_model <- function(inputCol = NULL, outputCol = NULL, batchSize = NULL, moduleBytes = NULL, device = NULL, feedDict = NULL, fetchDict = NULL, convertOutputToDenseVector = NULL) {
+  stage <- mmlspark_amd$models$image_featurizer$CNTKModel()
+  if (!is.null(inputCol)) stage$set("inputCol", inputCol)
+  if (!is.null(outputCol)) stage$set("outputCol", outputCol)
+  if (!is.null(batchSize)) stage$set("batchSize", batchSize)
+  if (!is.null(moduleBytes)) stage$set("moduleBytes", moduleBytes)
+  if (!is.null(device)) stage$set("device", device)
+  if (!is.null(feedDict)) stage$set("feedDict", feedDict)
+  if (!is.null(fetchDict)) stage$set("fetchDict", fetchDict)
+  if (!is.null(convertOutputToDenseVector)) stage$set("convertOutputToDenseVector", convertOutputToDenseVector)
+  stage
+}
+
 ml_cacher <- function(disable = NULL) {
   stage <- mmlspark_amd$stages$basic$Cacher()
   if (!is.null(disable)) stage$set("disable", disable)

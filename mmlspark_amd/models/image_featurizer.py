@@ -263,3 +263,30 @@ class DeepVisionModel(TorchModel):
         out[self.get("predictionCol")] = logits.argmax(axis=1).astype(np.float64) \
             if len(out) else []
         return out
+
+
+@register
+class CNTKModel(TorchModel):
+    """Name-compatible alias of TorchModel for reference users
+    (cntk/CNTKModel.scala:520-562): batched DNN inference with feed/fetch
+    dict naming.  The MI355X runtime holds torch modules, not CNTK graphs,
+    so `feedDict`/`fetchDict` map the single model input/output to DataFrame
+    columns (CNTKModel.scala:212-226 param surface); multi-head graphs are
+    expressed as torch modules returning one tensor per registered output.
+    """
+
+    feedDict = Param("feedDict", "model-input-name -> input column "
+                     "({name: col}; single entry sets inputCol)", None)
+    fetchDict = Param("fetchDict", "output column -> model-output-name "
+                      "({col: name}; single entry sets outputCol)", None)
+    convertOutputToDenseVector = Param(
+        "convertOutputToDenseVector", "emit output as a dense vector column "
+        "(always true here — outputs are vector columns)", True, toBool)
+
+    def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
+        fd, fc = self.get("feedDict"), self.get("fetchDict")
+        if fd:
+            self.set("inputCol", next(iter(fd.values())))
+        if fc:
+            self.set("outputCol", next(iter(fc.keys())))
+        return super()._transform(df)
