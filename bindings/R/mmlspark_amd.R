@@ -999,7 +999,7 @@ ml_light_gbm_classifier <- function(labelCol = NULL, featuresCol = NULL, feature
   stage
 }
 
-ml_light_gbm_ranker <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, categoricalSlotNames = NULL, slotNames = NULL, modelString = NULL, lightGBMBooster = NULL, fobj = NULL, boostFromAverage = NULL, improvementTolerance = NULL, posBaggingFraction = NULL, negBaggingFraction = NULL, binSampleCount = NULL, maxBinByFeature = NULL, uniformDrop = NULL, xgboostDartMode = NULL, startIteration = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, chunkSize = NULL, defaultListenPort = NULL, driverListenPort = NULL, timeout = NULL, numTasks = NULL, numThreads = NULL, useSingleDatasetMode = NULL, matrixType = NULL, checkpointDir = NULL, checkpointInterval = NULL, device = NULL, groupCol = NULL, labelGain = NULL, maxPosition = NULL, evalAt = NULL) {
+ml_light_gbm_ranker <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, weightCol = NULL, validationIndicatorCol = NULL, initScoreCol = NULL, predictionCol = NULL, numIterations = NULL, learningRate = NULL, numLeaves = NULL, maxDepth = NULL, maxBin = NULL, lambdaL1 = NULL, lambdaL2 = NULL, minDataInLeaf = NULL, minSumHessianInLeaf = NULL, minGainToSplit = NULL, featureFraction = NULL, baggingFraction = NULL, baggingFreq = NULL, baggingSeed = NULL, boostingType = NULL, topRate = NULL, otherRate = NULL, dropRate = NULL, skipDrop = NULL, maxDrop = NULL, maxDeltaStep = NULL, earlyStoppingRound = NULL, objective = NULL, metric = NULL, seed = NULL, numBatches = NULL, verbosity = NULL, isProvideTrainingMetric = NULL, useBarrierExecutionMode = NULL, parallelism = NULL, topK = NULL, categoricalSlotIndexes = NULL, categoricalSlotNames = NULL, slotNames = NULL, modelString = NULL, lightGBMBooster = NULL, fobj = NULL, boostFromAverage = NULL, improvementTolerance = NULL, posBaggingFraction = NULL, negBaggingFraction = NULL, binSampleCount = NULL, maxBinByFeature = NULL, uniformDrop = NULL, xgboostDartMode = NULL, startIteration = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, chunkSize = NULL, defaultListenPort = NULL, driverListenPort = NULL, timeout = NULL, numTasks = NULL, numThreads = NULL, useSingleDatasetMode = NULL, matrixType = NULL, checkpointDir = NULL, checkpointInterval = NULL, device = NULL, groupCol = NULL, labelGain = NULL, maxPosition = NULL, evalAt = NULL, repartitionByGroupingColumn = NULL) {
   stage <- mmlspark_amd$models$gbdt$estimators$LightGBMRanker()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1071,6 +1071,7 @@ ml_light_gbm_ranker <- function(labelCol = NULL, featuresCol = NULL, featureCols
   if (!is.null(labelGain)) stage$set("labelGain", labelGain)
   if (!is.null(maxPosition)) stage$set("maxPosition", maxPosition)
   if (!is.null(evalAt)) stage$set("evalAt", evalAt)
+  if (!is.null(repartitionByGroupingColumn)) stage$set("repartitionByGroupingColumn", repartitionByGroupingColumn)
   stage
 }
 

@@ -540,6 +540,10 @@ class LightGBMRanker(_GBDTEstimatorBase):
     labelGain = Param("labelGain", "per-label gain table", None)
     maxPosition = Param("maxPosition", "NDCG truncation", 10, toInt)
     evalAt = Param("evalAt", "NDCG eval positions", None)
+    repartitionByGroupingColumn = Param(
+        "repartitionByGroupingColumn", "make each query group contiguous "
+        "before training (LightGBMParams.scala:76); False trusts the input "
+        "ordering", True, toBool)
 
     def _make_objective(self, y):
         return make_objective("lambdarank", label_gain=self.get("labelGain"))
@@ -552,7 +556,9 @@ class LightGBMRanker(_GBDTEstimatorBase):
     def _fit(self, df):
         # rows of one query group must be contiguous (reference repartitions by
         # grouping column, LightGBMParams.scala:76) — sort locally by group
-        df = df.sort_values(self.get("groupCol"), kind="stable").reset_index(drop=True)
+        if self.get("repartitionByGroupingColumn"):
+            df = df.sort_values(self.get("groupCol"),
+                                kind="stable").reset_index(drop=True)
         return super()._fit(df)
 
     def _metrics_fn(self, valid_groups=None):
