@@ -2010,7 +2010,7 @@ ml_vowpal_wabbit_classifier <- function(labelCol = NULL, featuresCol = NULL, add
   stage
 }
 
-ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, holdoutOff = NULL, holdoutPeriod = NULL, earlyTerminate = NULL, maxIterBfgs = NULL, device = NULL, sharedCol = NULL, chosenActionCol = NULL, probabilityCol = NULL, epsilon = NULL) {
+ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, holdoutOff = NULL, holdoutPeriod = NULL, earlyTerminate = NULL, maxIterBfgs = NULL, device = NULL, sharedCol = NULL, additionalSharedFeatures = NULL, chosenActionCol = NULL, probabilityCol = NULL, epsilon = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitContextualBandit()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -2043,13 +2043,14 @@ ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NU
   if (!is.null(maxIterBfgs)) stage$set("maxIterBfgs", maxIterBfgs)
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(sharedCol)) stage$set("sharedCol", sharedCol)
+  if (!is.null(additionalSharedFeatures)) stage$set("additionalSharedFeatures", additionalSharedFeatures)
   if (!is.null(chosenActionCol)) stage$set("chosenActionCol", chosenActionCol)
   if (!is.null(probabilityCol)) stage$set("probabilityCol", probabilityCol)
   if (!is.null(epsilon)) stage$set("epsilon", epsilon)
   stage
 }
 
-ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, holdoutOff = NULL, holdoutPeriod = NULL, earlyTerminate = NULL, maxIterBfgs = NULL, device = NULL, weightsArrays = NULL, sharedCol = NULL, epsilon = NULL) {
+ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, holdoutOff = NULL, holdoutPeriod = NULL, earlyTerminate = NULL, maxIterBfgs = NULL, device = NULL, weightsArrays = NULL, sharedCol = NULL, additionalSharedFeatures = NULL, epsilon = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitContextualBanditModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -2083,6 +2084,7 @@ ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCo
   if (!is.null(device)) stage$set("device", device)
   if (!is.null(weightsArrays)) stage$set("weightsArrays", weightsArrays)
   if (!is.null(sharedCol)) stage$set("sharedCol", sharedCol)
+  if (!is.null(additionalSharedFeatures)) stage$set("additionalSharedFeatures", additionalSharedFeatures)
   if (!is.null(epsilon)) stage$set("epsilon", epsilon)
   stage
 }

@@ -567,6 +567,10 @@ class VowpalWabbitContextualBandit(_VWBase):
     importance weight 1/p_logged (reference --cb_type ips semantics)."""
     _default_loss = "squared"
     sharedCol = Param("sharedCol", "shared-context sparse column", "shared")
+    additionalSharedFeatures = Param(
+        "additionalSharedFeatures", "extra shared-context columns, merged "
+        "into the shared namespace with per-column murmur seeds "
+        "(VowpalWabbitContextualBandit.scala:95)", None)
     featuresCol = Param("featuresCol", "per-action features (list of "
                         "SparseVector per row)", "features")
     chosenActionCol = Param("chosenActionCol", "1-based chosen action index",
@@ -576,13 +580,32 @@ class VowpalWabbitContextualBandit(_VWBase):
     labelCol = Param("labelCol", "observed cost", "cost")
     epsilon = Param("epsilon", "exploration for predicted policy", 0.05, toFloat)
 
+    def _merged_shared(self, row) -> SparseVector:
+        """sharedCol ⊕ additionalSharedFeatures, each extra column's indices
+        offset by a murmur seed of its name (same namespace-seeding rule as
+        _extract_csr)."""
+        from .murmur import hash_string
+        base = row[self.get("sharedCol")]
+        extras = self.get("additionalSharedFeatures") or []
+        if not extras:
+            return base
+        idx = [base.indices.astype(np.int64)]
+        val = [base.values]
+        for c in extras:
+            v = row[c]
+            seed = hash_string(c) & 0x3FFFFFFF
+            idx.append((v.indices.astype(np.int64) + seed) & 0x3FFFFFFF)
+            val.append(v.values)
+        return SparseVector(1 << 30, np.concatenate(idx).astype(np.int32),
+                            np.concatenate(val))
+
     def _combine(self, shared: SparseVector, action: SparseVector, mask: int):
         from .featurizer import FNV_PRIME
         si, sv = shared.indices.astype(np.int64), shared.values
         ai, av = action.indices.astype(np.int64), action.values
         cross_i = ((si[:, None] * FNV_PRIME) ^ ai[None, :]).reshape(-1)
         cross_v = (sv[:, None] * av[None, :]).reshape(-1)
-        idx = np.concatenate([si, ai, cross_i & mask])
+        idx = np.concatenate([si & mask, ai & mask, cross_i & mask])
         val = np.concatenate([sv, av, cross_v])
         return idx.astype(np.int32), val.astype(np.float32)
 
@@ -598,7 +621,7 @@ class VowpalWabbitContextualBandit(_VWBase):
 
         idx_parts, val_parts, counts, labels, ws = [], [], [], [], []
         for _, row in df.iterrows():
-            shared = row[self.get("sharedCol")]
+            shared = self._merged_shared(row)
             actions = row[self.get("featuresCol")]
             chosen = int(row[self.get("chosenActionCol")]) - 1
             cost = float(row[self.get("labelCol")])
@@ -632,8 +655,8 @@ class VowpalWabbitContextualBandit(_VWBase):
                                          invariant=self.get("invariant"))
         model = VowpalWabbitContextualBanditModel(weights=w.cpu().numpy(),
                                                   adaptive=g.cpu().numpy())
-        for p in ("sharedCol", "featuresCol", "predictionCol", "numBits",
-                  "epsilon"):
+        for p in ("sharedCol", "additionalSharedFeatures", "featuresCol",
+                  "predictionCol", "numBits", "epsilon"):
             model.set(p, self.get(p))
         return model
 
@@ -641,17 +664,22 @@ class VowpalWabbitContextualBandit(_VWBase):
 @register
 class VowpalWabbitContextualBanditModel(_VWModelBase):
     sharedCol = Param("sharedCol", "shared-context sparse column", "shared")
+    additionalSharedFeatures = Param(
+        "additionalSharedFeatures", "extra shared-context columns", None)
     epsilon = Param("epsilon", "exploration rate", 0.05, toFloat)
 
     def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
         est = VowpalWabbitContextualBandit()
         est.set("numBits", self.get("numBits"))
+        est.set("sharedCol", self.get("sharedCol"))
+        est.set("additionalSharedFeatures",
+                self.get("additionalSharedFeatures"))
         mask = (1 << self.get("numBits")) - 1
         w = self.weights
         scores_col = []
         chosen_col = []
         for _, row in df.iterrows():
-            shared = row[self.get("sharedCol")]
+            shared = est._merged_shared(row)
             actions = row[self.get("featuresCol")]
             scores = []
             for a in actions:

@@ -397,3 +397,35 @@ def test_contextual_bandit_epsilon_pmf():
         np.testing.assert_allclose(p.sum(), 1.0, atol=1e-6)
         assert abs(p.max() - (0.7 + 0.1)) < 1e-6
         assert int(np.argmax(p)) + 1 == r["prediction"]
+
+
+def test_contextual_bandit_additional_shared_features():
+    """Extra shared columns change (and help) the policy; they are merged
+    into the shared namespace with per-column murmur seeds."""
+    rng = np.random.default_rng(4)
+    size = 1 << 14
+    rows = []
+    for _ in range(1500):
+        ctx = int(rng.integers(0, 3))
+        # the discriminative context lives ONLY in the extra shared column
+        extra = SparseVector(size, [300 + ctx], [1.0])
+        shared = SparseVector(size, [100], [1.0])
+        actions = [SparseVector(size, [2000 + a], [1.0]) for a in range(3)]
+        logged = int(rng.integers(0, 3))
+        cost = 0.0 if logged == ctx else 1.0
+        rows.append({"shared": shared, "extraCtx": extra, "features": actions,
+                     "chosenAction": logged + 1, "cost": cost,
+                     "probability": 1 / 3, "ctx": ctx})
+    df = pd.DataFrame(rows)
+    cb = VowpalWabbitContextualBandit(
+        numPasses=5, numBits=14, learningRate=0.5,
+        additionalSharedFeatures=["extraCtx"]).fit(df)
+    out = cb.transform(df)
+    acc = ((out["prediction"].to_numpy() - 1) == df["ctx"].to_numpy()).mean()
+    assert acc > 0.9
+    # without the extra column the context is invisible → near-chance policy
+    cb0 = VowpalWabbitContextualBandit(numPasses=5, numBits=14,
+                                       learningRate=0.5).fit(df)
+    out0 = cb0.transform(df)
+    acc0 = ((out0["prediction"].to_numpy() - 1) == df["ctx"].to_numpy()).mean()
+    assert acc0 < 0.6
