@@ -435,6 +435,11 @@ class LightGBMClassifier(_GBDTEstimatorBase):
     isUnbalance = Param("isUnbalance", "re-weight positives by n_neg/n_pos "
                         "(binary only)", False, toBool)
 
+    def getActualNumClasses(self) -> int:
+        """Number of classes inferred at fit time (LightGBMClassifier
+        actualNumClasses); 2 until fit."""
+        return getattr(self, "_actual_num_classes", 2)
+
     def _adjust_weights(self, yt, w):
         if not self.get("isUnbalance"):
             return w
@@ -451,7 +456,7 @@ class LightGBMClassifier(_GBDTEstimatorBase):
 
     def _fit(self, df):
         model = super()._fit(df)
-        for p in ("rawPredictionCol", "probabilityCol"):
+        for p in ("rawPredictionCol", "probabilityCol", "isUnbalance"):
             model.set(p, self.get(p))
         return model
 
@@ -461,6 +466,14 @@ class LightGBMClassificationModel(_GBDTModelBase):
     rawPredictionCol = Param("rawPredictionCol", "raw margin column", "rawPrediction")
     probabilityCol = Param("probabilityCol", "probability column", "probability")
     thresholds = Param("thresholds", "per-class prediction thresholds", None)
+    isUnbalance = Param("isUnbalance", "was the estimator fit with unbalance "
+                        "re-weighting (informational)", False, toBool)
+
+    def getActualNumClasses(self) -> int:
+        """LightGBMClassificationModel.getActualNumClasses: classes in the
+        trained booster (1 output plane = binary)."""
+        n = self.booster.num_planes if hasattr(self.booster, "num_planes")             else 1
+        return 2 if n <= 1 else n
 
     def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
         device = default_device(self.get("device"))
@@ -508,6 +521,12 @@ class LightGBMRegressor(_GBDTEstimatorBase):
     alpha = Param("alpha", "huber/quantile alpha", 0.9, toFloat)
     tweedieVariancePower = Param("tweedieVariancePower", "tweedie rho", 1.5, toFloat)
 
+    def _fit(self, df):
+        model = super()._fit(df)
+        for p in ("alpha", "tweedieVariancePower"):
+            model.set(p, self.get(p))
+        return model
+
     def _make_objective(self, y):
         name = self.get("objective") or "regression"
         return make_objective(name, alpha=self.get("alpha"),
@@ -519,6 +538,10 @@ class LightGBMRegressor(_GBDTEstimatorBase):
 
 @register
 class LightGBMRegressionModel(_GBDTModelBase):
+    alpha = Param("alpha", "huber/quantile alpha used at fit", 0.9, toFloat)
+    tweedieVariancePower = Param("tweedieVariancePower",
+                                 "tweedie rho used at fit", 1.5, toFloat)
+
     def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
         device = default_device(self.get("device"))
         X = self._X(df, device)
@@ -559,7 +582,10 @@ class LightGBMRanker(_GBDTEstimatorBase):
         if self.get("repartitionByGroupingColumn"):
             df = df.sort_values(self.get("groupCol"),
                                 kind="stable").reset_index(drop=True)
-        return super()._fit(df)
+        model = super()._fit(df)
+        for p in ("labelGain", "maxPosition", "evalAt"):
+            model.set(p, self.get(p))
+        return model
 
     def _metrics_fn(self, valid_groups=None):
         """Group-aware NDCG@k on the validation split (evalAt /
@@ -606,6 +632,11 @@ class LightGBMRanker(_GBDTEstimatorBase):
 
 @register
 class LightGBMRankerModel(_GBDTModelBase):
+    labelGain = Param("labelGain", "per-label gain table used at fit", None)
+    maxPosition = Param("maxPosition", "NDCG truncation used at fit", 10,
+                        toInt)
+    evalAt = Param("evalAt", "NDCG eval positions used at fit", None)
+
     def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
         device = default_device(self.get("device"))
         X = self._X(df, device)

@@ -74,6 +74,14 @@ class SAR(Estimator):
                                "jaccard|lift|cooccurrence", "jaccard", toString)
     timeDecayCoeff = Param("timeDecayCoeff", "affinity half-life (days)", 30,
                            toInt)
+    startTime = Param("startTime", "custom 'now' reference time for the "
+                      "decay when scoring historical data (SAR.scala:242)",
+                      None)
+    startTimeFormat = Param("startTimeFormat", "strftime format of startTime "
+                            "(SAR.scala:50)", "EEE MMM dd HH:mm:ss Z yyyy")
+    activityTimeFormat = Param("activityTimeFormat", "strftime format of a "
+                               "string timeCol (SAR.scala:53)",
+                               "yyyy/MM/dd'T'h:mm:ss")
 
     def _fit(self, df: pd.DataFrame):
         device = default_device("auto")
@@ -86,8 +94,17 @@ class SAR(Estimator):
 
         # time-decayed affinity (SAR.scala:80-130): rating * 2^(-(t_ref - t)/T)
         if self.get("timeCol") and self.get("timeCol") in df.columns:
-            t = torch.from_numpy(df[self.get("timeCol")].to_numpy(np.float64))
-            t_ref = float(t.max())
+            tv = df[self.get("timeCol")]
+            if tv.dtype == object or str(tv.dtype).startswith("datetime"):
+                # string/datetime timestamps (activityTimeFormat semantics —
+                # pandas infers the format; epoch seconds)
+                tv = pd.to_datetime(tv).astype("int64") / 1e9
+                t = torch.from_numpy(tv.to_numpy(np.float64))
+            else:
+                t = torch.from_numpy(tv.to_numpy(np.float64))
+            st = self.get("startTime")
+            t_ref = (float(pd.Timestamp(st).timestamp()) if st
+                     else float(t.max()))
             half_life = self.get("timeDecayCoeff") * 86400.0
             decay = torch.pow(2.0, -(t_ref - t) / half_life).float()
             r = r * decay
@@ -125,6 +142,19 @@ class SARModel(Model):
     predictionCol = Param("predictionCol", "score column", "prediction")
     sarArrays = Param("sarArrays", "affinity + similarity matrices", None,
                       is_complex=True)
+
+    def getItemDataFrame(self) -> pd.DataFrame:
+        """Item-item similarity as a DataFrame (SARModel itemDataFrame):
+        one row per item, `similarity` = that item's row vector."""
+        S = self.get("sarArrays")["similarity"]
+        return pd.DataFrame({"itemID": np.arange(len(S)),
+                             "similarity": list(S.astype(np.float64))})
+
+    def getUserDataFrame(self) -> pd.DataFrame:
+        """User-affinity as a DataFrame (SARModel userDataFrame)."""
+        A = self.get("sarArrays")["affinity"]
+        return pd.DataFrame({"userID": np.arange(len(A)),
+                             "affinity": list(A.astype(np.float64))})
 
     def _mats(self):
         d = self.get("sarArrays")

@@ -511,7 +511,7 @@ class VowpalWabbitClassifier(_VWBase):
 
     def _fit(self, df):
         model = super()._fit(df)
-        for p in ("rawPredictionCol", "probabilityCol"):
+        for p in ("rawPredictionCol", "probabilityCol", "labelConversion"):
             model.set(p, self.get(p))
         return model
 
@@ -523,6 +523,8 @@ class VowpalWabbitClassifier(_VWBase):
 class VowpalWabbitClassificationModel(_VWModelBase):
     rawPredictionCol = Param("rawPredictionCol", "margin column", "rawPrediction")
     probabilityCol = Param("probabilityCol", "probability column", "probability")
+    labelConversion = Param("labelConversion", "labels were converted {0,1}→"
+                            "{-1,+1} at fit (informational)", True, toBool)
 
     def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
         raw = self._raw(df)
@@ -666,6 +668,10 @@ class VowpalWabbitContextualBanditModel(_VWModelBase):
     sharedCol = Param("sharedCol", "shared-context sparse column", "shared")
     additionalSharedFeatures = Param(
         "additionalSharedFeatures", "extra shared-context columns", None)
+    chosenActionCol = Param("chosenActionCol", "1-based chosen action index "
+                            "column (for offline evaluation)", "chosenAction")
+    probabilityCol = Param("probabilityCol", "logged action probability "
+                           "column (for offline evaluation)", "probability")
     epsilon = Param("epsilon", "exploration rate", 0.05, toFloat)
 
     def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
