@@ -198,9 +198,27 @@ class FindBestModel(Estimator):
             scores.append(s)
             rows.append({"model": type(m).__name__, "uid": m.uid, metric: s})
         best_i = int(np.argmax(scores) if higher_better else np.argmin(scores))
-        out = BestModel(best=self.get("models")[best_i])
+        best = self.get("models")[best_i]
+        out = BestModel(best=best)
         out.set("allModelMetrics", pd.DataFrame(rows))
         out.set("bestModelMetrics", float(scores[best_i]))
+        # BestModel.scoredDataset / rocCurve (FindBestModel.scala:134 params)
+        scored = best.transform(df)
+        out.set("scoredDataset", scored)
+        prob_col = next((c for c in ("probability", "rawPrediction")
+                         if c in scored.columns), None)
+        y = df[self.get("labelCol")].to_numpy()
+        if prob_col is not None and set(np.unique(y)) <= {0, 1, 0.0, 1.0}:
+            p1 = np.asarray([np.asarray(v).ravel()[-1]
+                             for v in scored[prob_col]])
+            order = np.argsort(-p1)
+            ys = y[order]
+            tp = np.cumsum(ys)
+            fp = np.cumsum(1 - ys)
+            P, N = max(tp[-1], 1), max(fp[-1], 1)
+            out.set("rocCurve", pd.DataFrame(
+                {"falsePositiveRate": np.concatenate([[0.0], fp / N]),
+                 "truePositiveRate": np.concatenate([[0.0], tp / P])}))
         return out
 
 
@@ -210,6 +228,16 @@ class BestModel(Model):
     allModelMetrics = Param("allModelMetrics", "evaluation table", None,
                             is_complex=True)
     bestModelMetrics = Param("bestModelMetrics", "winning metric", None)
+    scoredDataset = Param("scoredDataset", "best model's scores on the "
+                          "evaluation dataset", None, is_complex=True)
+    rocCurve = Param("rocCurve", "ROC points of the best model (binary)",
+                     None, is_complex=True)
+
+    def getScoredDataset(self):
+        return self.get("scoredDataset")
+
+    def getRocCurve(self):
+        return self.get("rocCurve")
 
     def __init__(self, best=None, **kwargs):
         super().__init__(**kwargs)
