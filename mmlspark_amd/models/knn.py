@@ -28,6 +28,9 @@ class _KNNParamsMixin:
     outputCol = Param("outputCol", "matches output column", "output")
     k = Param("k", "number of neighbors", 5, toInt)
     batchSize = Param("batchSize", "query rows per device batch", 4096, toInt)
+    leafSize = Param("leafSize", "ball-tree leaf size (used only by the CPU "
+                     "BallTree view, getBallTree(); the GPU path is exact "
+                     "brute-force — KNN.scala leafSize)", 50, toInt)
 
 
 @register
@@ -39,7 +42,7 @@ class KNN(_KNNParamsMixin, Estimator):
         values = df[self.get("valuesCol")].tolist() \
             if self.get("valuesCol") in df.columns else list(range(len(df)))
         model = KNNModel(index=X, values=values)
-        for p in ("featuresCol", "outputCol", "k", "batchSize"):
+        for p in ("featuresCol", "outputCol", "k", "batchSize", "leafSize"):
             model.set(p, self.get(p))
         return model
 
@@ -48,6 +51,18 @@ class KNN(_KNNParamsMixin, Estimator):
 class KNNModel(_KNNParamsMixin, Model):
     indexData = Param("indexData", "indexed matrix + payload", None,
                       is_complex=True)
+
+    def getBallTree(self):
+        """The reference's broadcast ball tree (BallTree.scala:109) over the
+        same index; built lazily on the CPU.  Scoring here uses the exact
+        brute-force MFMA path, so this exists for API/introspection parity
+        and CPU-side traversal."""
+        from .balltree import BallTree
+        if getattr(self, "_ball_tree", None) is None:
+            d = self.get("indexData")
+            self._ball_tree = BallTree(d["index"], list(d["values"]),
+                                       leaf_size=self.get("leafSize"))
+        return self._ball_tree
 
     def __init__(self, index: Optional[np.ndarray] = None,
                  values: Optional[List] = None, labels=None, **kwargs):
