@@ -42,12 +42,43 @@ class TextFeaturizer(Estimator):
     numFeatures = Param("numFeatures", "hash dimension", 1 << 18, toInt)
     useIDF = Param("useIDF", "apply inverse document frequency", True, toBool)
     minDocFreq = Param("minDocFreq", "min document frequency", 1, toInt)
+    useStopWordsRemover = Param("useStopWordsRemover", "drop stop words "
+                                "(TextFeaturizer.scala useStopWordsRemover)",
+                                False, toBool)
+    stopWords = Param("stopWords", "stop words (comma-separated string or "
+                      "list); default English list when unset", None)
+    caseSensitiveStopWords = Param("caseSensitiveStopWords",
+                                   "match stop words case-sensitively",
+                                   False, toBool)
+    binary = Param("binary", "binary term frequency (presence instead of "
+                   "counts)", False, toBool)
+
+    _DEFAULT_STOP_WORDS = (
+        "a,an,the,and,or,but,if,then,is,are,was,were,be,been,of,to,in,on,"
+        "for,with,as,at,by,it,its,this,that,these,those,from,not,no,so,too,"
+        "very,can,will,just,do,does,did,have,has,had,i,you,he,she,we,they")
+
+    def _stop_set(self):
+        sw = self.get("stopWords")
+        if sw is None:
+            words = self._DEFAULT_STOP_WORDS.split(",")
+        elif isinstance(sw, str):
+            words = sw.split(",")
+        else:
+            words = list(sw)
+        if not self.get("caseSensitiveStopWords"):
+            words = [w.lower() for w in words]
+        return set(words)
 
     def _terms(self, s):
         toks = _tokenize(s, self.get("tokenizerPattern"),
                          self.get("toLowercase"),
                          self.get("minTokenLength")) \
             if self.get("useTokenizer") else list(s)
+        if self.get("useStopWordsRemover"):
+            stop = self._stop_set()
+            cs = self.get("caseSensitiveStopWords")
+            toks = [t for t in toks if (t if cs else t.lower()) not in stop]
         if self.get("useNGram"):
             toks = toks + _ngrams(toks, self.get("nGramLength"))
         return toks
@@ -68,7 +99,8 @@ class TextFeaturizer(Estimator):
         m.set("idf", idf if self.get("useIDF") else np.ones(dim, np.float32))
         for p in ("inputCol", "outputCol", "useTokenizer", "tokenizerPattern",
                   "toLowercase", "minTokenLength", "useNGram", "nGramLength",
-                  "numFeatures"):
+                  "numFeatures", "useStopWordsRemover", "stopWords",
+                  "caseSensitiveStopWords", "binary"):
             m.set(p, self.get(p))
         return m
 
@@ -84,6 +116,12 @@ class TextFeaturizerModel(Model):
     useNGram = Param("useNGram", "add ngrams", False, toBool)
     nGramLength = Param("nGramLength", "ngram order", 2, toInt)
     numFeatures = Param("numFeatures", "hash dimension", 1 << 18, toInt)
+    useStopWordsRemover = Param("useStopWordsRemover", "drop stop words",
+                                False, toBool)
+    stopWords = Param("stopWords", "stop words", None)
+    caseSensitiveStopWords = Param("caseSensitiveStopWords",
+                                   "case-sensitive match", False, toBool)
+    binary = Param("binary", "binary term frequency", False, toBool)
     idf = Param("idf", "idf weights", None, is_complex=True)
 
     def _transform(self, df):
@@ -91,7 +129,9 @@ class TextFeaturizerModel(Model):
         idf = np.asarray(self.get("idf"))
         fe = TextFeaturizer()
         for p in ("useTokenizer", "tokenizerPattern", "toLowercase",
-                  "minTokenLength", "useNGram", "nGramLength"):
+                  "minTokenLength", "useNGram", "nGramLength",
+                  "useStopWordsRemover", "stopWords",
+                  "caseSensitiveStopWords"):
             fe.set(p, self.get(p))
         vecs = []
         for s in df[self.get("inputCol")]:
@@ -101,6 +141,8 @@ class TextFeaturizerModel(Model):
                 counts[h] = counts.get(h, 0) + 1
             idx = np.array(sorted(counts), dtype=np.int32)
             val = np.array([counts[i] for i in idx], dtype=np.float32)
+            if self.get("binary"):
+                val = np.ones_like(val)
             val = val * idf[idx]
             vecs.append(SparseVector(dim, idx, val))
         out = df.copy()

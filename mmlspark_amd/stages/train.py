@@ -8,7 +8,7 @@ from __future__ import annotations
 import numpy as np
 import pandas as pd
 
-from ..core.param import Param, toInt, toList, toString
+from ..core.param import Param, toBool, toInt, toList, toString
 from ..core.pipeline import Estimator, Model, Transformer
 from ..core.registry import register
 from .featurize import Featurize
@@ -23,6 +23,9 @@ class TrainClassifier(Estimator):
     featuresCol = Param("featuresCol", "generated features column",
                         "TrainClassifier_features")
     numFeatures = Param("numFeatures", "hash dim for high-card strings", 0, toInt)
+    reindexLabel = Param("reindexLabel", "re-index labels to 0..K-1 "
+                         "(TrainClassifier.scala reindexLabel); False trusts "
+                         "numeric labels as-is", True, toBool)
 
     def _fit(self, df: pd.DataFrame):
         label = self.get("labelCol")
@@ -30,8 +33,12 @@ class TrainClassifier(Estimator):
         if inner is None:
             from ..models.gbdt.estimators import LightGBMClassifier
             inner = LightGBMClassifier()
-        levels = sorted((str(v) for v in pd.unique(df[label].dropna())))
-        lut = {v: i for i, v in enumerate(levels)}
+        if not self.get("reindexLabel"):
+            levels = sorted({str(v) for v in pd.unique(df[label].dropna())})
+            lut = {str(v): float(v) for v in pd.unique(df[label].dropna())}
+        else:
+            levels = sorted((str(v) for v in pd.unique(df[label].dropna())))
+            lut = {v: i for i, v in enumerate(levels)}
         feat_cols = [c for c in df.columns if c != label]
         featurizer = Featurize(inputCols=feat_cols,
                                outputCol=self.get("featuresCol")).fit(df)
@@ -43,7 +50,8 @@ class TrainClassifier(Estimator):
         fitted = inner.fit(dff)
         m = TrainedClassifierModel(featurizer=featurizer, inner=fitted)
         m.set("labelCol", label)
-        m.set("levels", levels)
+        # without reindexing there is no index→level mapping to invert
+        m.set("levels", levels if self.get("reindexLabel") else None)
         m.set("featuresCol", self.get("featuresCol"))
         return m
 
