@@ -51,7 +51,8 @@ class TrainClassifier(Estimator):
         m = TrainedClassifierModel(featurizer=featurizer, inner=fitted)
         m.set("labelCol", label)
         # without reindexing there is no index→level mapping to invert
-        m.set("levels", levels if self.get("reindexLabel") else None)
+        if self.get("reindexLabel"):
+            m.set("levels", levels)
         m.set("featuresCol", self.get("featuresCol"))
         return m
 
