@@ -144,11 +144,13 @@ ml_azure_search_writer <- function(url = NULL, subscriptionKey = NULL, subscript
   stage
 }
 
-ml_best_model <- function(bestModel = NULL, allModelMetrics = NULL, bestModelMetrics = NULL) {
+ml_best_model <- function(bestModel = NULL, allModelMetrics = NULL, bestModelMetrics = NULL, scoredDataset = NULL, rocCurve = NULL) {
   stage <- mmlspark_amd$stages$automl$BestModel()
   if (!is.null(bestModel)) stage$set("bestModel", bestModel)
   if (!is.null(allModelMetrics)) stage$set("allModelMetrics", allModelMetrics)
   if (!is.null(bestModelMetrics)) stage$set("bestModelMetrics", bestModelMetrics)
+  if (!is.null(scoredDataset)) stage$set("scoredDataset", scoredDataset)
+  if (!is.null(rocCurve)) stage$set("rocCurve", rocCurve)
   stage
 }
 
@@ -266,25 +268,27 @@ ml_compute_per_instance_statistics <- function(labelCol = NULL, scoresCol = NULL
   stage
 }
 
-ml_conditional_knn <- function(featuresCol = NULL, valuesCol = NULL, outputCol = NULL, k = NULL, batchSize = NULL, labelCol = NULL, conditionerCol = NULL) {
+ml_conditional_knn <- function(featuresCol = NULL, valuesCol = NULL, outputCol = NULL, k = NULL, batchSize = NULL, leafSize = NULL, labelCol = NULL, conditionerCol = NULL) {
   stage <- mmlspark_amd$models$knn$ConditionalKNN()
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
   if (!is.null(valuesCol)) stage$set("valuesCol", valuesCol)
   if (!is.null(outputCol)) stage$set("outputCol", outputCol)
   if (!is.null(k)) stage$set("k", k)
   if (!is.null(batchSize)) stage$set("batchSize", batchSize)
+  if (!is.null(leafSize)) stage$set("leafSize", leafSize)
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(conditionerCol)) stage$set("conditionerCol", conditionerCol)
   stage
 }
 
-ml_conditional_knn_model <- function(featuresCol = NULL, valuesCol = NULL, outputCol = NULL, k = NULL, batchSize = NULL, indexData = NULL, conditionerCol = NULL) {
+ml_conditional_knn_model <- function(featuresCol = NULL, valuesCol = NULL, outputCol = NULL, k = NULL, batchSize = NULL, leafSize = NULL, indexData = NULL, conditionerCol = NULL) {
   stage <- mmlspark_amd$models$knn$ConditionalKNNModel()
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
   if (!is.null(valuesCol)) stage$set("valuesCol", valuesCol)
   if (!is.null(outputCol)) stage$set("outputCol", outputCol)
   if (!is.null(k)) stage$set("k", k)
   if (!is.null(batchSize)) stage$set("batchSize", batchSize)
+  if (!is.null(leafSize)) stage$set("leafSize", leafSize)
   if (!is.null(indexData)) stage$set("indexData", indexData)
   if (!is.null(conditionerCol)) stage$set("conditionerCol", conditionerCol)
   stage
@@ -820,23 +824,25 @@ ml_json_output_parser <- function(inputCol = NULL, outputCol = NULL) {
   stage
 }
 
-ml_knn <- function(featuresCol = NULL, valuesCol = NULL, outputCol = NULL, k = NULL, batchSize = NULL) {
+ml_knn <- function(featuresCol = NULL, valuesCol = NULL, outputCol = NULL, k = NULL, batchSize = NULL, leafSize = NULL) {
   stage <- mmlspark_amd$models$knn$KNN()
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
   if (!is.null(valuesCol)) stage$set("valuesCol", valuesCol)
   if (!is.null(outputCol)) stage$set("outputCol", outputCol)
   if (!is.null(k)) stage$set("k", k)
   if (!is.null(batchSize)) stage$set("batchSize", batchSize)
+  if (!is.null(leafSize)) stage$set("leafSize", leafSize)
   stage
 }
 
-ml_knn_model <- function(featuresCol = NULL, valuesCol = NULL, outputCol = NULL, k = NULL, batchSize = NULL, indexData = NULL) {
+ml_knn_model <- function(featuresCol = NULL, valuesCol = NULL, outputCol = NULL, k = NULL, batchSize = NULL, leafSize = NULL, indexData = NULL) {
   stage <- mmlspark_amd$models$knn$KNNModel()
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
   if (!is.null(valuesCol)) stage$set("valuesCol", valuesCol)
   if (!is.null(outputCol)) stage$set("outputCol", outputCol)
   if (!is.null(k)) stage$set("k", k)
   if (!is.null(batchSize)) stage$set("batchSize", batchSize)
+  if (!is.null(leafSize)) stage$set("leafSize", leafSize)
   if (!is.null(indexData)) stage$set("indexData", indexData)
   stage
 }
@@ -907,7 +913,7 @@ ml_language_detector_v2 <- function(url = NULL, subscriptionKey = NULL, subscrip
   stage
 }
 
-ml_light_gbm_classification_model <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, predictionCol = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, boosterModelStr = NULL, startIteration = NULL, numIterations = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL, thresholds = NULL) {
+ml_light_gbm_classification_model <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, predictionCol = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, boosterModelStr = NULL, startIteration = NULL, numIterations = NULL, device = NULL, rawPredictionCol = NULL, probabilityCol = NULL, thresholds = NULL, isUnbalance = NULL) {
   stage <- mmlspark_amd$models$gbdt$estimators$LightGBMClassificationModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -922,6 +928,7 @@ ml_light_gbm_classification_model <- function(labelCol = NULL, featuresCol = NUL
   if (!is.null(rawPredictionCol)) stage$set("rawPredictionCol", rawPredictionCol)
   if (!is.null(probabilityCol)) stage$set("probabilityCol", probabilityCol)
   if (!is.null(thresholds)) stage$set("thresholds", thresholds)
+  if (!is.null(isUnbalance)) stage$set("isUnbalance", isUnbalance)
   stage
 }
 
@@ -1075,7 +1082,7 @@ ml_light_gbm_ranker <- function(labelCol = NULL, featuresCol = NULL, featureCols
   stage
 }
 
-ml_light_gbm_ranker_model <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, predictionCol = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, boosterModelStr = NULL, startIteration = NULL, numIterations = NULL, device = NULL) {
+ml_light_gbm_ranker_model <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, predictionCol = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, boosterModelStr = NULL, startIteration = NULL, numIterations = NULL, device = NULL, labelGain = NULL, maxPosition = NULL, evalAt = NULL) {
   stage <- mmlspark_amd$models$gbdt$estimators$LightGBMRankerModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1087,10 +1094,13 @@ ml_light_gbm_ranker_model <- function(labelCol = NULL, featuresCol = NULL, featu
   if (!is.null(startIteration)) stage$set("startIteration", startIteration)
   if (!is.null(numIterations)) stage$set("numIterations", numIterations)
   if (!is.null(device)) stage$set("device", device)
+  if (!is.null(labelGain)) stage$set("labelGain", labelGain)
+  if (!is.null(maxPosition)) stage$set("maxPosition", maxPosition)
+  if (!is.null(evalAt)) stage$set("evalAt", evalAt)
   stage
 }
 
-ml_light_gbm_regression_model <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, predictionCol = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, boosterModelStr = NULL, startIteration = NULL, numIterations = NULL, device = NULL) {
+ml_light_gbm_regression_model <- function(labelCol = NULL, featuresCol = NULL, featureCols = NULL, predictionCol = NULL, leafPredictionCol = NULL, featuresShapCol = NULL, boosterModelStr = NULL, startIteration = NULL, numIterations = NULL, device = NULL, alpha = NULL, tweedieVariancePower = NULL) {
   stage <- mmlspark_amd$models$gbdt$estimators$LightGBMRegressionModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1102,6 +1112,8 @@ ml_light_gbm_regression_model <- function(labelCol = NULL, featuresCol = NULL, f
   if (!is.null(startIteration)) stage$set("startIteration", startIteration)
   if (!is.null(numIterations)) stage$set("numIterations", numIterations)
   if (!is.null(device)) stage$set("device", device)
+  if (!is.null(alpha)) stage$set("alpha", alpha)
+  if (!is.null(tweedieVariancePower)) stage$set("tweedieVariancePower", tweedieVariancePower)
   stage
 }
 
@@ -1435,7 +1447,7 @@ ml_resize_image_transformer <- function(inputCol = NULL, outputCol = NULL, heigh
   stage
 }
 
-ml_sar <- function(userCol = NULL, itemCol = NULL, ratingCol = NULL, timeCol = NULL, supportThreshold = NULL, similarityFunction = NULL, timeDecayCoeff = NULL) {
+ml_sar <- function(userCol = NULL, itemCol = NULL, ratingCol = NULL, timeCol = NULL, supportThreshold = NULL, similarityFunction = NULL, timeDecayCoeff = NULL, startTime = NULL, startTimeFormat = NULL, activityTimeFormat = NULL) {
   stage <- mmlspark_amd$models$sar$SAR()
   if (!is.null(userCol)) stage$set("userCol", userCol)
   if (!is.null(itemCol)) stage$set("itemCol", itemCol)
@@ -1444,6 +1456,9 @@ ml_sar <- function(userCol = NULL, itemCol = NULL, ratingCol = NULL, timeCol = N
   if (!is.null(supportThreshold)) stage$set("supportThreshold", supportThreshold)
   if (!is.null(similarityFunction)) stage$set("similarityFunction", similarityFunction)
   if (!is.null(timeDecayCoeff)) stage$set("timeDecayCoeff", timeDecayCoeff)
+  if (!is.null(startTime)) stage$set("startTime", startTime)
+  if (!is.null(startTimeFormat)) stage$set("startTimeFormat", startTimeFormat)
+  if (!is.null(activityTimeFormat)) stage$set("activityTimeFormat", activityTimeFormat)
   stage
 }
 
@@ -1627,7 +1642,7 @@ ml_tag_image <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyCol 
   stage
 }
 
-ml_text_featurizer <- function(inputCol = NULL, outputCol = NULL, useTokenizer = NULL, tokenizerPattern = NULL, toLowercase = NULL, minTokenLength = NULL, useNGram = NULL, nGramLength = NULL, numFeatures = NULL, useIDF = NULL, minDocFreq = NULL) {
+ml_text_featurizer <- function(inputCol = NULL, outputCol = NULL, useTokenizer = NULL, tokenizerPattern = NULL, toLowercase = NULL, minTokenLength = NULL, useNGram = NULL, nGramLength = NULL, numFeatures = NULL, useIDF = NULL, minDocFreq = NULL, useStopWordsRemover = NULL, stopWords = NULL, caseSensitiveStopWords = NULL, binary = NULL) {
   stage <- mmlspark_amd$stages$text$TextFeaturizer()
   if (!is.null(inputCol)) stage$set("inputCol", inputCol)
   if (!is.null(outputCol)) stage$set("outputCol", outputCol)
@@ -1640,10 +1655,14 @@ ml_text_featurizer <- function(inputCol = NULL, outputCol = NULL, useTokenizer =
   if (!is.null(numFeatures)) stage$set("numFeatures", numFeatures)
   if (!is.null(useIDF)) stage$set("useIDF", useIDF)
   if (!is.null(minDocFreq)) stage$set("minDocFreq", minDocFreq)
+  if (!is.null(useStopWordsRemover)) stage$set("useStopWordsRemover", useStopWordsRemover)
+  if (!is.null(stopWords)) stage$set("stopWords", stopWords)
+  if (!is.null(caseSensitiveStopWords)) stage$set("caseSensitiveStopWords", caseSensitiveStopWords)
+  if (!is.null(binary)) stage$set("binary", binary)
   stage
 }
 
-ml_text_featurizer_model <- function(inputCol = NULL, outputCol = NULL, useTokenizer = NULL, tokenizerPattern = NULL, toLowercase = NULL, minTokenLength = NULL, useNGram = NULL, nGramLength = NULL, numFeatures = NULL, idf = NULL) {
+ml_text_featurizer_model <- function(inputCol = NULL, outputCol = NULL, useTokenizer = NULL, tokenizerPattern = NULL, toLowercase = NULL, minTokenLength = NULL, useNGram = NULL, nGramLength = NULL, numFeatures = NULL, useStopWordsRemover = NULL, stopWords = NULL, caseSensitiveStopWords = NULL, binary = NULL, idf = NULL) {
   stage <- mmlspark_amd$stages$text$TextFeaturizerModel()
   if (!is.null(inputCol)) stage$set("inputCol", inputCol)
   if (!is.null(outputCol)) stage$set("outputCol", outputCol)
@@ -1654,6 +1673,10 @@ ml_text_featurizer_model <- function(inputCol = NULL, outputCol = NULL, useToken
   if (!is.null(useNGram)) stage$set("useNGram", useNGram)
   if (!is.null(nGramLength)) stage$set("nGramLength", nGramLength)
   if (!is.null(numFeatures)) stage$set("numFeatures", numFeatures)
+  if (!is.null(useStopWordsRemover)) stage$set("useStopWordsRemover", useStopWordsRemover)
+  if (!is.null(stopWords)) stage$set("stopWords", stopWords)
+  if (!is.null(caseSensitiveStopWords)) stage$set("caseSensitiveStopWords", caseSensitiveStopWords)
+  if (!is.null(binary)) stage$set("binary", binary)
   if (!is.null(idf)) stage$set("idf", idf)
   stage
 }
@@ -1753,12 +1776,13 @@ ml_torch_model <- function(inputCol = NULL, outputCol = NULL, batchSize = NULL, 
   stage
 }
 
-ml_train_classifier <- function(model = NULL, labelCol = NULL, featuresCol = NULL, numFeatures = NULL) {
+ml_train_classifier <- function(model = NULL, labelCol = NULL, featuresCol = NULL, numFeatures = NULL, reindexLabel = NULL) {
   stage <- mmlspark_amd$stages$train$TrainClassifier()
   if (!is.null(model)) stage$set("model", model)
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
   if (!is.null(numFeatures)) stage$set("numFeatures", numFeatures)
+  if (!is.null(reindexLabel)) stage$set("reindexLabel", reindexLabel)
   stage
 }
 
@@ -1934,7 +1958,7 @@ ml_verify_faces <- function(url = NULL, subscriptionKey = NULL, subscriptionKeyC
   stage
 }
 
-ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, holdoutOff = NULL, holdoutPeriod = NULL, earlyTerminate = NULL, maxIterBfgs = NULL, device = NULL, weightsArrays = NULL, rawPredictionCol = NULL, probabilityCol = NULL) {
+ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, holdoutOff = NULL, holdoutPeriod = NULL, earlyTerminate = NULL, maxIterBfgs = NULL, device = NULL, weightsArrays = NULL, rawPredictionCol = NULL, probabilityCol = NULL, labelConversion = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitClassificationModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -1969,6 +1993,7 @@ ml_vowpal_wabbit_classification_model <- function(labelCol = NULL, featuresCol =
   if (!is.null(weightsArrays)) stage$set("weightsArrays", weightsArrays)
   if (!is.null(rawPredictionCol)) stage$set("rawPredictionCol", rawPredictionCol)
   if (!is.null(probabilityCol)) stage$set("probabilityCol", probabilityCol)
+  if (!is.null(labelConversion)) stage$set("labelConversion", labelConversion)
   stage
 }
 
@@ -2050,7 +2075,7 @@ ml_vowpal_wabbit_contextual_bandit <- function(labelCol = NULL, featuresCol = NU
   stage
 }
 
-ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, holdoutOff = NULL, holdoutPeriod = NULL, earlyTerminate = NULL, maxIterBfgs = NULL, device = NULL, weightsArrays = NULL, sharedCol = NULL, additionalSharedFeatures = NULL, epsilon = NULL) {
+ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCol = NULL, additionalFeatures = NULL, weightCol = NULL, predictionCol = NULL, numPasses = NULL, learningRate = NULL, powerT = NULL, l1 = NULL, l2 = NULL, numBits = NULL, lossFunction = NULL, batchSize = NULL, hashSeed = NULL, adaptive = NULL, normalized = NULL, invariant = NULL, initialModel = NULL, passThroughArgs = NULL, args = NULL, testArgs = NULL, interactions = NULL, ignoreNamespaces = NULL, useBarrierExecutionMode = NULL, bfgs = NULL, holdoutOff = NULL, holdoutPeriod = NULL, earlyTerminate = NULL, maxIterBfgs = NULL, device = NULL, weightsArrays = NULL, sharedCol = NULL, additionalSharedFeatures = NULL, chosenActionCol = NULL, probabilityCol = NULL, epsilon = NULL) {
   stage <- mmlspark_amd$models$vw$estimators$VowpalWabbitContextualBanditModel()
   if (!is.null(labelCol)) stage$set("labelCol", labelCol)
   if (!is.null(featuresCol)) stage$set("featuresCol", featuresCol)
@@ -2085,6 +2110,8 @@ ml_vowpal_wabbit_contextual_bandit_model <- function(labelCol = NULL, featuresCo
   if (!is.null(weightsArrays)) stage$set("weightsArrays", weightsArrays)
   if (!is.null(sharedCol)) stage$set("sharedCol", sharedCol)
   if (!is.null(additionalSharedFeatures)) stage$set("additionalSharedFeatures", additionalSharedFeatures)
+  if (!is.null(chosenActionCol)) stage$set("chosenActionCol", chosenActionCol)
+  if (!is.null(probabilityCol)) stage$set("probabilityCol", probabilityCol)
   if (!is.null(epsilon)) stage$set("epsilon", epsilon)
   stage
 }
