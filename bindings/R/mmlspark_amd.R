@@ -189,7 +189,7 @@ ml_break_sentence <- function(url = NULL, subscriptionKey = NULL, subscriptionKe
   stage
 }
 
-ml_cntk_model <- function(inputCol = NULL, outputCol = NULL, batchSize = NULL, moduleBytes = NULL, device = NULL, feedDict = NULL, fetchDict = NULL, convertOutputToDenseVector = NULL) {
+ml_cntk_model <- function(inputCol = NULL, outputCol = NULL, batchSize = NULL, moduleBytes = NULL, device = NULL, feedDict = NULL, fetchDict = NULL, convertOutputToDenseVector = NULL, batchInput = NULL, shapeOutput = NULL) {
   stage <- mmlspark_amd$models$image_featurizer$CNTKModel()
   if (!is.null(inputCol)) stage$set("inputCol", inputCol)
   if (!is.null(outputCol)) stage$set("outputCol", outputCol)
@@ -199,6 +199,8 @@ ml_cntk_model <- function(inputCol = NULL, outputCol = NULL, batchSize = NULL, m
   if (!is.null(feedDict)) stage$set("feedDict", feedDict)
   if (!is.null(fetchDict)) stage$set("fetchDict", fetchDict)
   if (!is.null(convertOutputToDenseVector)) stage$set("convertOutputToDenseVector", convertOutputToDenseVector)
+  if (!is.null(batchInput)) stage$set("batchInput", batchInput)
+  if (!is.null(shapeOutput)) stage$set("shapeOutput", shapeOutput)
   stage
 }
 
@@ -1759,10 +1761,11 @@ ml_time_interval_mini_batch_transformer <- function(millisToWait = NULL, maxBatc
   stage
 }
 
-ml_timer <- function(stage = NULL, logToScala = NULL) {
+ml_timer <- function(stage = NULL, logToScala = NULL, disableMaterialization = NULL) {
   stage <- mmlspark_amd$stages$basic$Timer()
   if (!is.null(stage)) stage$set("stage", stage)
   if (!is.null(logToScala)) stage$set("logToScala", logToScala)
+  if (!is.null(disableMaterialization)) stage$set("disableMaterialization", disableMaterialization)
   stage
 }
 

@@ -282,6 +282,10 @@ class CNTKModel(TorchModel):
     convertOutputToDenseVector = Param(
         "convertOutputToDenseVector", "emit output as a dense vector column "
         "(always true here — outputs are vector columns)", True, toBool)
+    batchInput = Param("batchInput", "minibatch rows before inference "
+                       "(always on — the transform is batched)", True, toBool)
+    shapeOutput = Param("shapeOutput", "reshape output to the model's "
+                        "declared shape (flat vector here)", False, toBool)
 
     def _transform(self, df: pd.DataFrame) -> pd.DataFrame:
         fd, fc = self.get("feedDict"), self.get("fetchDict")

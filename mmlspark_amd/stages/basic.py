@@ -119,6 +119,10 @@ class Timer(Transformer):
     """Times an inner stage (Timer.scala); logs to the telemetry ring."""
     stage = Param("stage", "inner stage", None, is_complex=True)
     logToScala = Param("logToScala", "log to driver", True, toBool)
+    disableMaterialization = Param("disableMaterialization", "skip forcing "
+                                   "the inner result (no-op: pandas frames "
+                                   "are eager — Timer.scala parity)", True,
+                                   toBool)
 
     def _transform(self, df):
         from ..core.telemetry import log_stage_event
