@@ -15,13 +15,13 @@ TIMEOUT=${CI_TIMEOUT:-1200}       # seconds per split (ref: 20-min sbt cap)
 
 declare -A SPLITS=(
   [core]="tests/test_core.py tests/test_io_codegen.py tests/test_interop.py tests/test_r_bindings.py"
-  [stages]="tests/test_stages.py tests/test_properties.py tests/test_fuzzing.py"
+  [stages]="tests/test_stages.py tests/test_properties.py tests/test_fuzzing.py tests/test_text.py tests/test_automl.py tests/test_coverage_extras.py"
   [gbdt]="tests/test_gbdt.py tests/test_gbdt_sparse.py tests/test_benchmarks_csv.py tests/test_external_anchor.py tests/test_elastic.py"
   [vw]="tests/test_vw.py"
   [distributed]="tests/test_distributed.py"
   [serving]="tests/test_serving.py"
   [cognitive]="tests/test_images_cognitive.py tests/test_jpeg_native.py"
-  [models]="tests/test_models_misc.py tests/test_cyber.py tests/test_explainers.py tests/test_deep_learning.py"
+  [models]="tests/test_models_misc.py tests/test_cyber.py tests/test_explainers.py tests/test_deep_learning.py tests/test_knn.py tests/test_sar.py tests/test_image_featurizer.py"
   [e2e]="tests/test_integration_e2e.py tests/test_examples_e2e.py"
   [gpu]="tests -m gpu"
 )
