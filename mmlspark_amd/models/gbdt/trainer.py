@@ -126,7 +126,12 @@ def _leaf_output(G, H, cfg: TrainConfig) -> float:
     g = abs(G) - cfg.lambda_l1
     if g <= 0:
         return 0.0
-    w = -math.copysign(g, G) / (H + cfg.lambda_l2)
+    denom = H + cfg.lambda_l2
+    if denom <= 0:
+        # a leaf whose (quantized) hessians all rounded to zero with no L2 —
+        # LightGBM outputs 0 below min_sum_hessian rather than diverging
+        return 0.0
+    w = -math.copysign(g, G) / denom
     if cfg.max_delta_step > 0:
         w = max(-cfg.max_delta_step, min(cfg.max_delta_step, w))
     return w

@@ -338,7 +338,9 @@ double leaf_output(double G, double H, double l1, double l2,
                    double max_delta) {
   double g = std::abs(G) - l1;
   if (g <= 0) return 0.0;
-  double w = -std::copysign(g, G) / (H + l2);
+  double denom = H + l2;
+  if (denom <= 0) return 0.0;  // all-zero quantized hessians, no L2
+  double w = -std::copysign(g, G) / denom;
   if (max_delta > 0) w = std::clamp(w, -max_delta, max_delta);
   return w;
 }

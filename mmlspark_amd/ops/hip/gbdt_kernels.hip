@@ -1586,7 +1586,7 @@ extern "C" void launch_csr_hist_fixed_v2(const long* indptr, const int* col,
 // LDS cell.  Row staging + entry→row mapping as in v2.  Row count per
 // block must stay ≤ 2^19 so the 44-bit packed h-sum cannot overflow
 // (launcher enforces via grid sizing).
-template <int FW>  // feature-chunk width (8: 4 blocks/CU at nb=255; 16: 2)
+template <int FW, bool ILP = false>  // FW: feature-chunk width; ILP: 4-entry load batching
 __global__ void csr_hist_fixed_lds_k(const long* __restrict__ indptr,
                                      const int* __restrict__ col,
                                      const unsigned char* __restrict__ binv,
@@ -1654,15 +1654,44 @@ __global__ void csr_hist_fixed_lds_k(const long* __restrict__ indptr,
         const int mid = (k + hi2) >> 1;
         if (cum[mid] <= e0) k = mid; else hi2 = mid;
       }
-      for (int e = e0; e < e1; ++e) {
-        while (cum[k + 1] <= e) ++k;
-        const long j = s_start[wid][k] + (e - cum[k]);
-        const int c = col[j];
-        if (c >= f0 && c < f0 + FW) {
+      if (ILP) {
+        // batch 4 entries: resolve rows + issue all col/bin loads before
+        // any LDS atomic, so the 4 global loads overlap instead of
+        // serializing behind each atomic's LDS dependency chain
+        for (int e = e0; e < e1;) {
+          int ks[4]; long js[4]; int cnt = 0;
+          for (; cnt < 4 && e + cnt < e1; ++cnt) {
+            while (cum[k + 1] <= e + cnt) ++k;
+            ks[cnt] = k;
+            js[cnt] = s_start[wid][k] + (e + cnt - cum[k]);
+          }
+          int cs[4]; unsigned char bs[4];
+          for (int t = 0; t < cnt; ++t) {
+            cs[t] = col[js[t]];
+            bs[t] = binv[js[t]];
+          }
+          for (int t = 0; t < cnt; ++t) {
+            if (cs[t] >= f0 && cs[t] < f0 + FW) {
+              unsigned long long* cell =
+                  &lds64[(((cs[t] - f0) * n_bins) + bs[t]) * 2];
+              atomicAdd(cell + 0, (unsigned long long)s_g[wid][ks[t]]);
+              atomicAdd(cell + 1,
+                        CNT_ONE + (unsigned long long)s_h[wid][ks[t]]);
+            }
+          }
+          e += cnt;
+        }
+      } else {
+        for (int e = e0; e < e1; ++e) {
+          while (cum[k + 1] <= e) ++k;
+          const long j = s_start[wid][k] + (e - cum[k]);
+          const int c = col[j];
+          if (c >= f0 && c < f0 + FW) {
           unsigned long long* cell =
               &lds64[(((c - f0) * n_bins) + binv[j]) * 2];
           atomicAdd(cell + 0, (unsigned long long)s_g[wid][k]);
           atomicAdd(cell + 1, CNT_ONE + (unsigned long long)s_h[wid][k]);
+          }
         }
       }
     }
@@ -1719,9 +1748,20 @@ extern "C" void launch_csr_hist_fixed_lds(const long* indptr, const int* col,
     const char* e = getenv("MMLSPARK_AMD_SPARSE_FW");
     return e ? atoi(e) : 8;
   }();
+  static const bool ilp_env = [] {
+    const char* e = getenv("MMLSPARK_AMD_SPARSE_ILP");
+    return e && atoi(e) != 0;
+  }();
   const int FW = (fw_env == 16) ? 16 : (fw_env == 4 ? 4 : 8);
   const int by = (nf + FW - 1) / FW;
   const size_t lds_bytes = (size_t)FW * n_bins * 2 * sizeof(long long);
+  if (FW == 8 && ilp_env) {
+    hipLaunchKernelGGL((csr_hist_fixed_lds_k<8, true>),
+                       dim3((unsigned)bx, (unsigned)by), dim3(256),
+                       lds_bytes, stream, indptr, col, binv, gq, hq, rows,
+                       m, hist, n_bins, nf, tot);
+    return;
+  }
   if (FW == 4)
     hipLaunchKernelGGL(csr_hist_fixed_lds_k<4>,
                        dim3((unsigned)bx, (unsigned)by), dim3(256),

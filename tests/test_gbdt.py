@@ -495,3 +495,13 @@ def test_bagging_seed_controls_sampling(binary_df):
     c = LightGBMClassifier(baggingSeed=2, **kw).fit(binary_df)
     assert a.booster.save_to_string() == b.booster.save_to_string()
     assert a.booster.save_to_string() != c.booster.save_to_string()
+
+
+def test_leaf_output_zero_hessian_guard():
+    """H + lambda_l2 == 0 (quantized hessians rounded to zero in a tiny
+    leaf) must yield output 0.0, not ZeroDivisionError."""
+    from mmlspark_amd.models.gbdt.trainer import TrainConfig, _leaf_output
+    cfg = TrainConfig(lambda_l1=0.0, lambda_l2=0.0)
+    assert _leaf_output(1.5, 0.0, cfg) == 0.0
+    assert _leaf_output(-1.5, 0.0, cfg) == 0.0
+    assert _leaf_output(1.5, 2.0, cfg) == -0.75
