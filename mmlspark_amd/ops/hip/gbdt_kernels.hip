@@ -1656,8 +1656,10 @@ __global__ void csr_hist_fixed_lds_k(const long* __restrict__ indptr,
       }
       if (ILP) {
         // batch 4 entries: resolve rows + issue all col/bin loads before
-        // any LDS atomic, so the 4 global loads overlap instead of
-        // serializing behind each atomic's LDS dependency chain
+        // any LDS atomic.  A/B-measured on MI355X @10M×100 nnz=20: 131M vs
+        // 211M rows/s — 1.6× SLOWER (the ks/js/cs/bs staging spills and the
+        // LDS-atomic chain, not load latency, is the bound).  Kept gated
+        // OFF behind MMLSPARK_AMD_SPARSE_ILP as the recorded negative result
         for (int e = e0; e < e1;) {
           int ks[4]; long js[4]; int cnt = 0;
           for (; cnt < 4 && e + cnt < e1; ++cnt) {
