@@ -63,7 +63,7 @@ def roc(df, y_col: str, y_hat_col: str, thresh: float = .5):
     P = max(int(tp[-1]), 1)
     N = max(int(fp[-1]), 1)
     tpr, fpr = tp / P, fp / N
-    auc = float(np.trapz(tpr, fpr))
+    auc = float(np.trapezoid(tpr, fpr))
     plt.plot(fpr, tpr)
     plt.xlabel("False Positive Rate", fontsize=20)
     plt.ylabel("True Positive Rate", fontsize=20)
