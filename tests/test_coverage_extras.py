@@ -103,3 +103,14 @@ def test_superpixel_transformer_stage():
     segs = out["superpixels"].iloc[0]
     assert segs.shape == (32, 32)
     assert len(np.unique(segs)) >= 4  # multiple segments found
+
+
+def test_backend_fails_loudly_without_extension(monkeypatch):
+    """On a GPU tensor a missing HIP extension must raise, never silently
+    fall back to eager torch (round-end native-load check contract)."""
+    import mmlspark_amd.ops.backend as B
+    monkeypatch.setattr(B, "_EXT", False)
+    monkeypatch.setattr(B, "_EXT_ERR", "simulated")
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError, match="HIP extension is required"):
+        B._require_ext()
