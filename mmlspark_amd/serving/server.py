@@ -495,6 +495,22 @@ class _WorkerProc:
         return not self.dead and self.proc.poll() is None
 
 
+class _RemoteWorker:
+    """A worker on ANOTHER host/process that self-registered with the head
+    (DriverServiceUtils rendezvous: workers POST their ServiceInfo to the
+    driver's HttpServer, HTTPSourceV2.scala:133-198).  The head cannot
+    poll() it; liveness is learned from request failures (failover skips
+    it after a kill_remote / repeated errors)."""
+    __slots__ = ("host", "port", "name", "dead")
+
+    def __init__(self, host, port, name):
+        self.host, self.port, self.name = host, int(port), name
+        self.dead = False
+
+    def alive(self):
+        return not self.dead
+
+
 class ProcessServingCluster:
     """Multi-PROCESS serving: N `mmlspark_amd.serving.worker` subprocesses
     (one per GPU rank in a real deployment — set HIP_VISIBLE_DEVICES per
@@ -560,6 +576,14 @@ class ProcessServingCluster:
         def head_handler(payloads):
             out = []
             for p in payloads:
+                # cross-host rendezvous: a worker anywhere POSTs
+                # {"__register__": {host, port, name}} to join the rotation
+                if isinstance(p, dict) and "__register__" in p:
+                    info = p["__register__"]
+                    out.append(self.register_remote(
+                        info.get("host", "127.0.0.1"), info["port"],
+                        info.get("name", f"remote-{info['port']}")))
+                    continue
                 with self._rr_lock:
                     start = self._rr
                     self._rr += 1
@@ -587,10 +611,26 @@ class ProcessServingCluster:
                                   name=f"{self.name}-head").start()
         return self
 
+    def register_remote(self, host, port, name=None) -> dict:
+        """Add a worker running on another host/process to the head's
+        round-robin rotation (driver-rendezvous parity).  Idempotent on
+        (host, port)."""
+        name = name or f"remote-{port}"
+        for w in self.workers:
+            if w.host == host and w.port == int(port):
+                w.dead = False
+                return {"registered": True, "name": w.name, "known": True}
+        self.workers.append(_RemoteWorker(host, port, name))
+        return {"registered": True, "name": name, "known": False}
+
     def kill_worker(self, i: int):
         """Hard-kill the worker PROCESS (real crash, not a simulation)."""
-        self.workers[i].proc.kill()
-        self.workers[i].dead = True
+        w = self.workers[i]
+        if isinstance(w, _RemoteWorker):
+            w.dead = True  # remote: mark out of rotation (cannot signal it)
+        else:
+            w.proc.kill()
+            w.dead = True
 
     def restart_worker(self, i: int):
         self.workers[i] = self._spawn(i)
@@ -601,10 +641,11 @@ class ProcessServingCluster:
                              "alive": w.alive()} for w in self.workers]}
 
     def stop(self):
-        for w in self.workers:
+        local = [w for w in self.workers if isinstance(w, _WorkerProc)]
+        for w in local:
             if w.proc.poll() is None:
                 w.proc.terminate()
-        for w in self.workers:
+        for w in local:
             try:
                 w.proc.wait(timeout=10)
             except Exception:

@@ -50,7 +50,9 @@ def main(argv=None):
     if args.report_to:
         import requests
         try:
-            requests.post(args.report_to, json=srv.service_info(), timeout=10)
+            requests.post(args.report_to,
+                          json={"__register__": srv.service_info()},
+                          timeout=10)
         except Exception as e:  # discovery is best-effort
             print(json.dumps({"report_error": repr(e)}), flush=True)
 
