@@ -479,3 +479,117 @@ def test_progressive_ac_successive_approximation():
     out_p = _jpeg_native.decode_jpeg(prog).numpy()
     out_b = _jpeg_native.decode_jpeg(base).numpy()
     np.testing.assert_array_equal(out_p, out_b)
+
+
+def encode_progressive_eobn(img, quality=30):
+    """Progressive stream whose AC first scans use a CUSTOM Huffman table
+    carrying EOBn symbols (r=0..5) so multi-block EOB runs (ITU-T.81
+    G.1.2.2 EOBRUN; decoder jpeg_native.cpp eobrun path) are exercised —
+    Annex-K tables carry only EOB0, so the in-tree streams never produced
+    runs >1 block before this encoder.  All custom codes are 8 bits
+    (167 symbols ≤ 256, valid canonical table)."""
+    h, w = img.shape[:2]
+    planes = _quantized_planes(img, quality, (1, 1))
+    out = _headers(h, w, quality, (1, 1), progressive=True)
+
+    # custom AC table id (tc=1, th=2): every (run,size) pair we can emit,
+    # ZRL, and EOBn for runs up to 2^6-1 blocks
+    vals = [(r << 4) | s for r in range(16) for s in range(1, 11)]
+    vals += [0xF0] + [(r << 4) for r in range(6)]
+    bits = [0] * 16
+    bits[7] = len(vals)  # all codes 8 bits long
+    out += (b"\xff\xc4" + struct.pack(">H", 2 + 1 + 16 + len(vals))
+            + bytes([(1 << 4) | 2]) + bytes(bits) + bytes(vals))
+    ac = _build_codes(bits, vals)
+    dc_codes = [_build_codes(*HT_DC_L), _build_codes(*HT_DC_C)]
+    bh, bw_ = planes[0].shape[:2]
+
+    # interleaved DC first scan, Ah=Al=0 (no refinement scan needed)
+    out += _sos([(1, 0, 0), (2, 1, 1), (3, 1, 1)], 0, 0, 0, 0)
+    bw = _BitWriter()
+    pred = [0, 0, 0]
+    for by in range(bh):
+        for bx in range(bw_):
+            for ci in range(3):
+                v = int(planes[ci][by, bx, 0])
+                _write_dc(bw, dc_codes[0 if ci == 0 else 1], v - pred[ci])
+                pred[ci] = v
+    bw.flush()
+    out += bytes(bw.out)
+
+    def ac_first(ci):
+        bw = _BitWriter()
+        state = {"eobrun": 0}
+
+        def flush():
+            n = state["eobrun"]
+            if n == 0:
+                return
+            r = n.bit_length() - 1
+            c, ln = ac[(r << 4)]
+            bw.write(c, ln)
+            if r:
+                bw.write(n - (1 << r), r)
+            state["eobrun"] = 0
+
+        for by in range(bh):
+            for bx in range(bw_):
+                zz = planes[ci][by, bx]
+                last = 0
+                for k in range(1, 64):
+                    if zz[k]:
+                        last = k
+                if last == 0:  # all-zero AC: extend the EOB run
+                    state["eobrun"] += 1
+                    if state["eobrun"] == 63:
+                        flush()
+                    continue
+                flush()
+                run = 0
+                for k in range(1, last + 1):
+                    v = int(zz[k])
+                    if v == 0:
+                        run += 1
+                        continue
+                    while run > 15:
+                        c, ln = ac[0xF0]
+                        bw.write(c, ln)
+                        run -= 16
+                    n = _mag(v)
+                    assert n <= 10, "test image too sharp for custom table"
+                    c, ln = ac[(run << 4) | n]
+                    bw.write(c, ln)
+                    bw.write(v if v >= 0 else v + (1 << n) - 1, n)
+                    run = 0
+                if last < 63:  # early end joins the next run
+                    state["eobrun"] += 1
+                    if state["eobrun"] == 63:
+                        flush()
+        flush()
+        bw.flush()
+        return bytes(bw.out)
+
+    for ci in range(3):
+        out += _sos([(ci + 1, 0 if ci == 0 else 1, 2)], 1, 63, 0, 0)
+        out += ac_first(ci)
+    return out + b"\xff\xd9"
+
+
+def test_progressive_eobn_runs_decode():
+    """Multi-block EOBn runs (n>0 → eobrun spans whole blocks) decode to
+    exactly the baseline reconstruction of the same coefficients."""
+    # smooth image + coarse quantization → long all-zero-AC runs
+    yy, xx = np.mgrid[0:96, 0:128]
+    img = np.clip(128 + 40 * np.sin(xx / 60.0) + 30 * np.cos(yy / 70.0),
+                  0, 255).astype(np.uint8)
+    img = np.stack([img, img, img], -1)
+    prog = encode_progressive_eobn(img, quality=25)
+    base = jpeg_codec.encode_jpeg(img, quality=25)
+    out_p = _jpeg_native.decode_jpeg(prog).numpy()
+    out_b = _jpeg_native.decode_jpeg(base).numpy()
+    np.testing.assert_array_equal(out_p, out_b)
+    # the stream must actually contain EOBn symbols with r>0: re-encode a
+    # fully-flat image and assert the chroma AC scan is just a few bytes
+    flat = np.full((64, 64, 3), 128, np.uint8)
+    tiny = encode_progressive_eobn(flat, quality=25)
+    assert len(tiny) < len(jpeg_codec.encode_jpeg(flat, quality=25)) + 4096
