@@ -99,3 +99,61 @@ def test_lightgbm_text_roundtrip_random_boosters(n_trees, n_leaves, seed):
     np.testing.assert_allclose(b2.predict_raw(Xt).numpy(),
                                m.booster.predict_raw(Xt).numpy(),
                                rtol=1e-5, atol=1e-5)
+
+
+@given(st.lists(st.lists(st.floats(-1e6, 1e6, width=32), min_size=0,
+                         max_size=8), min_size=1, max_size=20))
+@settings(max_examples=50, deadline=None)
+def test_arrow_round_trip_preserves_vectors(rows):
+    import pandas as pd
+    import pytest
+    pa = pytest.importorskip("pyarrow")
+    from mmlspark_amd.core.interop import arrow_to_pandas, pandas_to_arrow
+    df = pd.DataFrame({"v": [np.asarray(r, dtype=np.float32) for r in rows],
+                       "s": list(range(len(rows)))})
+    back = arrow_to_pandas(pandas_to_arrow(df))
+    assert len(back) == len(df)
+    for a, b in zip(back["v"], df["v"]):
+        np.testing.assert_array_equal(np.asarray(a), b)
+    assert back["s"].tolist() == df["s"].tolist()
+
+
+@given(st.integers(1, 30), st.integers(1, 6), st.integers(0, 10**6))
+@settings(max_examples=50, deadline=None)
+def test_fast_vector_assembler_matches_concat(n, nv, seed):
+    import pandas as pd
+    from mmlspark_amd.stages.featurize import FastVectorAssembler
+    rng = np.random.default_rng(seed)
+    df = pd.DataFrame({
+        "a": rng.normal(size=n),
+        "v": list(rng.normal(size=(n, nv)).astype(np.float32)),
+    })
+    out = FastVectorAssembler(inputCols=["a", "v"], outputCol="f") \
+        .transform(df)
+    got = np.stack(out["f"].to_numpy())
+    want = np.concatenate([df["a"].to_numpy(np.float32)[:, None],
+                           np.stack(df["v"].to_numpy())], axis=1)
+    np.testing.assert_allclose(got, want, rtol=1e-6)
+
+
+@given(st.integers(0, 10**6))
+@settings(max_examples=20, deadline=None)
+def test_booster_lightgbm_text_round_trip(seed):
+    """to_lightgbm_text → load gives numerically identical predictions for
+    random small forests (save/load parity with stock LightGBM v3 format)."""
+    import pandas as pd
+    from mmlspark_amd.models.gbdt.booster import (Booster,
+                                                  _from_lightgbm_text,
+                                                  _to_lightgbm_text)
+    from mmlspark_amd.models.gbdt.estimators import LightGBMRegressor
+    rng = np.random.default_rng(seed)
+    X = rng.normal(size=(300, 5)).astype(np.float32)
+    y = (X[:, 0] * 2 + np.sin(X[:, 1])).astype(np.float64)
+    df = pd.DataFrame({"features": list(X), "label": y})
+    m = LightGBMRegressor(numIterations=3, numLeaves=7).fit(df)
+    b = m.booster
+    b2 = _from_lightgbm_text(_to_lightgbm_text(b))
+    Xt = torch.from_numpy(X)
+    np.testing.assert_allclose(b.predict_raw(Xt).numpy(),
+                               b2.predict_raw(Xt).numpy(), rtol=1e-5,
+                               atol=1e-6)
