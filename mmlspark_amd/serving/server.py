@@ -633,6 +633,9 @@ class ProcessServingCluster:
             w.dead = True
 
     def restart_worker(self, i: int):
+        if isinstance(self.workers[i], _RemoteWorker):
+            raise ValueError("worker %d is remote — restart it on its own "
+                             "host; it will re-register" % i)
         self.workers[i] = self._spawn(i)
 
     def service_info(self):

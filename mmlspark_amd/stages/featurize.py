@@ -257,12 +257,15 @@ class FastVectorAssembler(Transformer):
                 mat[:, off] = s.to_numpy(np.float32)
             else:
                 vals = s.to_numpy()
-                if len(vals) and isinstance(vals[0], SparseVector):
+                first = next((v for v in vals if v is not None), None)
+                if isinstance(first, SparseVector):
                     for r, v in enumerate(vals):
-                        mat[r, off + v.indices] = v.values
-                elif len(vals):
+                        if v is not None:
+                            mat[r, off + v.indices] = v.values
+                elif first is not None:
                     mat[:, off:off + w] = np.stack(
-                        [np.asarray(v, dtype=np.float32) for v in vals])
+                        [np.full(w, np.nan, np.float32) if v is None
+                         else np.asarray(v, dtype=np.float32) for v in vals])
             off += w
         out = df.copy()
         out[self.get("outputCol")] = matrix_to_vector_column(mat)

@@ -59,6 +59,9 @@ class TextFeaturizer(Estimator):
         "very,can,will,just,do,does,did,have,has,had,i,you,he,she,we,they")
 
     def _stop_set(self):
+        cached = getattr(self, "_stop_cache", None)
+        if cached is not None:
+            return cached
         sw = self.get("stopWords")
         if sw is None:
             words = self._DEFAULT_STOP_WORDS.split(",")
@@ -68,7 +71,8 @@ class TextFeaturizer(Estimator):
             words = list(sw)
         if not self.get("caseSensitiveStopWords"):
             words = [w.lower() for w in words]
-        return set(words)
+        self._stop_cache = set(words)
+        return self._stop_cache
 
     def _terms(self, s):
         toks = _tokenize(s, self.get("tokenizerPattern"),
