@@ -19,8 +19,8 @@ declare -A SPLITS=(
   [gbdt]="tests/test_gbdt.py tests/test_gbdt_sparse.py tests/test_benchmarks_csv.py tests/test_external_anchor.py tests/test_elastic.py"
   [vw]="tests/test_vw.py"
   [distributed]="tests/test_distributed.py"
-  [serving]="tests/test_serving.py tests/test_serving_cluster_remote.py tests/test_jpeg_native.py"
-  [cognitive]="tests/test_images_cognitive.py"
+  [serving]="tests/test_serving.py tests/test_serving_cluster_remote.py"
+  [cognitive]="tests/test_images_cognitive.py tests/test_jpeg_native.py"
   [models]="tests/test_models_misc.py tests/test_cyber.py tests/test_explainers.py tests/test_deep_learning.py tests/test_knn.py tests/test_sar.py tests/test_image_featurizer.py"
   [e2e]="tests/test_integration_e2e.py tests/test_examples_e2e.py"
   [gpu]="tests -m gpu"
