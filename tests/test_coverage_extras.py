@@ -114,3 +114,12 @@ def test_backend_fails_loudly_without_extension(monkeypatch):
     import pytest as _pytest
     with _pytest.raises(RuntimeError, match="HIP extension is required"):
         B._require_ext()
+
+
+def test_default_device_selection():
+    import torch
+    from mmlspark_amd.utils.devices import default_device
+    assert default_device("cpu").type == "cpu"
+    auto = default_device("auto")
+    assert auto.type == ("cuda" if torch.cuda.is_available() else "cpu")
+    assert default_device(None).type == auto.type
