@@ -221,3 +221,23 @@ def test_http_transformer_concurrent_requests():
         assert got == list(range(n))  # order preserved despite concurrency
     finally:
         srv.stop()
+
+
+def test_committed_bindings_are_fresh(tmp_path):
+    """The committed R wrappers and Python stubs must match what codegen
+    produces from the current registry (staleness guard — the analog of
+    the reference's CI assertion that generated wrappers are current)."""
+    import filecmp
+    import os
+    import mmlspark_amd
+    mmlspark_amd._register_all()
+    from mmlspark_amd.core.codegen import generate_r_wrappers, generate_stubs
+    generate_r_wrappers(str(tmp_path / "gen.R"))
+    committed = open("bindings/R/mmlspark_amd.R").read()
+    assert open(tmp_path / "gen.R").read() == committed, \
+        "bindings/R/mmlspark_amd.R is stale — rerun generate_r_wrappers"
+    generate_stubs(str(tmp_path / "stubs"))
+    for f in sorted(os.listdir(tmp_path / "stubs")):
+        assert filecmp.cmp(tmp_path / "stubs" / f,
+                           os.path.join("bindings/python_stubs", f),
+                           shallow=False), f"stale stub: {f}"
