@@ -241,3 +241,9 @@ def test_committed_bindings_are_fresh(tmp_path):
         assert filecmp.cmp(tmp_path / "stubs" / f,
                            os.path.join("bindings/python_stubs", f),
                            shallow=False), f"stale stub: {f}"
+    from mmlspark_amd.core.codegen import generate_docs
+    generate_docs(str(tmp_path / "docs"))
+    for f in sorted(os.listdir(tmp_path / "docs")):
+        assert filecmp.cmp(tmp_path / "docs" / f,
+                           os.path.join("docs/api", f),
+                           shallow=False), f"stale API doc: {f}"
