@@ -432,3 +432,63 @@ def test_vw_multipass_allreduce_ws4():
     costs = [results[r]["multipass_ns"] for r in range(4)]
     assert all(c > 0 for c in costs)
     print("[vw ws4] per-pass allreduce cost ns:", costs)
+
+
+def _worker_gbdt_cat(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        torch.distributed.init_process_group(
+            "gloo", init_method=f"tcp://127.0.0.1:{port}",
+            rank=rank, world_size=world)
+        from mmlspark_amd.models.gbdt.objectives import make_objective
+        from mmlspark_amd.models.gbdt.trainer import TrainConfig, train_booster
+        from mmlspark_amd.parallel.comm import Comm
+
+        rng = np.random.default_rng(0)
+        n = 4000
+        Xnum = rng.normal(size=(n, 4)).astype(np.float32)
+        cat = rng.integers(0, 12, size=n)
+        # non-monotone categorical effect forces one-vs-rest set splits
+        eff = np.array([1.5, -2, 0.3, 2, -1, 0, 1, -1.5, 0.7, -0.4, 2.2, -2.5])
+        y = ((Xnum[:, 0] + eff[cat] + rng.normal(size=n) * 0.4) > 0) \
+            .astype(np.float32)
+        X = np.concatenate([cat[:, None].astype(np.float32), Xnum], axis=1)
+        sl = slice(rank * 2000, (rank + 1) * 2000)
+        cfg = TrainConfig(num_iterations=8, num_leaves=15, seed=7,
+                          categorical_features=[0])
+        booster, _ = train_booster(torch.from_numpy(X[sl]),
+                                   torch.from_numpy(y[sl]), cfg,
+                                   make_objective("binary"), Comm())
+        q.put((rank, booster.save_to_string()))
+        torch.distributed.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"ERROR: {e!r}"))
+
+
+@pytest.mark.timeout(180)
+def test_distributed_categorical_identical_models():
+    """Categorical one-vs-rest splits under ws=2 histogram sync: every rank
+    must grow the identical tree sequence (cat scan is deterministic over
+    the REDUCED histogram, so bitsets agree bit-for-bit)."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29911
+    procs = [ctx.Process(target=_worker_gbdt_cat, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, s = q.get(timeout=150)
+        results[rank] = s
+    for p in procs:
+        p.join(timeout=30)
+    assert not any(str(s).startswith("ERROR") for s in results.values()), \
+        results
+    assert results[0] == results[1]  # byte-identical boosters
+    t = json.loads(results[0])["trees"]
+    assert len(t) == 8
+    # at least one categorical (set) split must actually appear:
+    # cat_offset[node] >= 0 marks a 256-bit bitset split
+    assert any(any(o >= 0 for o in tree.get("cat_offset", []))
+               for tree in t), "no categorical split in any tree"
