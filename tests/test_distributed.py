@@ -492,3 +492,52 @@ def test_distributed_categorical_identical_models():
     # cat_offset[node] >= 0 marks a 256-bit bitset split
     assert any(any(o >= 0 for o in tree.get("cat_offset", []))
                for tree in t), "no categorical split in any tree"
+
+
+def _worker_gbdt_boosting(rank, world, port, q, boosting):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        torch.distributed.init_process_group(
+            "gloo", init_method=f"tcp://127.0.0.1:{port}",
+            rank=rank, world_size=world)
+        from mmlspark_amd.models.gbdt.objectives import make_objective
+        from mmlspark_amd.models.gbdt.trainer import TrainConfig, train_booster
+        from mmlspark_amd.parallel.comm import Comm
+        X, y = _make_data(0, n=4000)
+        sl = slice(rank * 2000, (rank + 1) * 2000)
+        cfg = TrainConfig(num_iterations=8, num_leaves=15, seed=7,
+                          boosting=boosting,
+                          bagging_fraction=0.7 if boosting == "rf" else 1.0,
+                          bagging_freq=1 if boosting == "rf" else 0)
+        booster, _ = train_booster(torch.from_numpy(X[sl]),
+                                   torch.from_numpy(y[sl]), cfg,
+                                   make_objective("binary"), Comm())
+        q.put((rank, booster.save_to_string()))
+        torch.distributed.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"ERROR: {e!r}"))
+
+
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("boosting", ["goss", "dart", "rf"])
+def test_distributed_boosting_variants_identical(boosting):
+    """goss/dart/rf under ws=2 histogram sync: per-rank row sampling uses
+    rank-seeded RNGs (data-parallel), yet reduced histograms must drive
+    every rank to the byte-identical booster."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29920 + {"goss": 0, "dart": 1, "rf": 2}[boosting]
+    procs = [ctx.Process(target=_worker_gbdt_boosting,
+                         args=(r, 2, port, q, boosting)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, s = q.get(timeout=250)
+        results[rank] = s
+    for p in procs:
+        p.join(timeout=30)
+    assert not any(str(s).startswith("ERROR") for s in results.values()), \
+        results
+    assert results[0] == results[1]
+    assert len(json.loads(results[0])["trees"]) >= 8
