@@ -541,3 +541,53 @@ def test_distributed_boosting_variants_identical(boosting):
         results
     assert results[0] == results[1]
     assert len(json.loads(results[0])["trees"]) >= 8
+
+
+def _worker_ranker(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        torch.distributed.init_process_group(
+            "gloo", init_method=f"tcp://127.0.0.1:{port}",
+            rank=rank, world_size=world)
+        import pandas as pd
+        from mmlspark_amd.models.gbdt.estimators import LightGBMRanker
+        rng = np.random.default_rng(0)
+        nq, per_q = 200, 10
+        feats = rng.normal(size=(nq * per_q, 6)).astype(np.float32)
+        rel = feats[:, 0] + 0.3 * rng.normal(size=nq * per_q)
+        labels = np.digitize(rel, np.quantile(rel, [0.5, 0.8, 0.95])) \
+            .astype(np.float64)
+        groups = np.repeat(np.arange(nq), per_q)
+        # whole query groups per rank (repartitionByGroupingColumn shape)
+        mask = (groups % world) == rank
+        df = pd.DataFrame({"group": groups[mask],
+                           "features": list(feats[mask]),
+                           "label": labels[mask]})
+        m = LightGBMRanker(numIterations=6, numLeaves=15,
+                           groupCol="group").fit(df)
+        q.put((rank, m.booster.save_to_string()))
+        torch.distributed.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"ERROR: {e!r}"))
+
+
+@pytest.mark.timeout(300)
+def test_distributed_ranker_identical_models():
+    """LambdaRank under ws=2: per-rank NDCG gradients over local query
+    groups, reduced histograms → byte-identical boosters on both ranks."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29931
+    procs = [ctx.Process(target=_worker_ranker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, s = q.get(timeout=250)
+        results[rank] = s
+    for p in procs:
+        p.join(timeout=30)
+    assert not any(str(s).startswith("ERROR") for s in results.values()), \
+        results
+    assert results[0] == results[1]
