@@ -591,3 +591,54 @@ def test_distributed_ranker_identical_models():
     assert not any(str(s).startswith("ERROR") for s in results.values()), \
         results
     assert results[0] == results[1]
+
+
+def _worker_earlystop(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        torch.distributed.init_process_group(
+            "gloo", init_method=f"tcp://127.0.0.1:{port}",
+            rank=rank, world_size=world)
+        from mmlspark_amd.models.gbdt.objectives import make_objective
+        from mmlspark_amd.models.gbdt.trainer import TrainConfig, train_booster
+        from mmlspark_amd.parallel.comm import Comm
+        X, y = _make_data(0, n=4000)
+        Xv, yv = _make_data(99, n=800)  # same valid set on every rank
+        sl = slice(rank * 2000, (rank + 1) * 2000)
+        cfg = TrainConfig(num_iterations=60, num_leaves=31, seed=7,
+                          early_stopping_round=5)
+        booster, _ = train_booster(
+            torch.from_numpy(X[sl]), torch.from_numpy(y[sl]), cfg,
+            make_objective("binary"), Comm(),
+            valid_sets=[(torch.from_numpy(Xv), torch.from_numpy(yv), None)])
+        q.put((rank, json.dumps({"best": booster.best_iteration,
+                                 "model": booster.save_to_string()})))
+        torch.distributed.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"ERROR: {e!r}"))
+
+
+@pytest.mark.timeout(300)
+def test_distributed_early_stopping_agrees():
+    """Early stopping under ws=2: the shared validation set gives every
+    rank the same metric stream, so best_iteration and the final booster
+    must agree exactly (a rank divergence here would deadlock real runs)."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29941
+    procs = [ctx.Process(target=_worker_earlystop, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, s = q.get(timeout=250)
+        results[rank] = s
+    for p in procs:
+        p.join(timeout=30)
+    assert not any(str(s).startswith("ERROR") for s in results.values()), \
+        results
+    a, b = json.loads(results[0]), json.loads(results[1])
+    assert a["best"] == b["best"]
+    assert a["model"] == b["model"]
+    assert a["best"] is not None and a["best"] < 60  # it actually stopped
