@@ -36,15 +36,26 @@ class BinMapper:
             max_bin_by_feature: Optional[list] = None) -> "BinMapper":
         n, nf = X.shape
         gen = torch.Generator(device="cpu").manual_seed(seed)
-        k = min(n, max(1, sample_size // max(1, comm.world_size if comm else 1)))
+        quota = max(1, sample_size // max(1, comm.world_size if comm else 1))
+        k = min(n, quota)
         idx = torch.randperm(n, generator=gen)[:k].to(X.device)
         sample = X[idx]
         if comm is not None and comm.is_distributed:
-            # equal-size shards so all_gather works; ranks with fewer rows pad
-            # by repeating (slight weight skew only in tiny-data tests)
-            if sample.shape[0] < k:
-                reps = (k + sample.shape[0] - 1) // max(1, sample.shape[0])
-                sample = sample.repeat(reps, 1)[:k]
+            # all_gather needs EQUAL shapes on every rank.  Agree on a common
+            # per-rank sample size first (one tiny collective): the largest
+            # shard's k, capped at the quota.  Smaller shards pad by
+            # repeating their sample (slight weight skew toward small shards
+            # only — exact repeats do not move quantiles of a shard's own
+            # distribution); an empty shard ships zeros
+            ks = comm.all_gather(torch.tensor([k], dtype=torch.int64,
+                                              device=X.device))
+            common = int(min(quota, max(int(t[0]) for t in ks)))
+            common = max(1, common)
+            if sample.shape[0] == 0:
+                sample = X.new_zeros((common, nf))
+            elif sample.shape[0] < common:
+                reps = (common + sample.shape[0] - 1) // sample.shape[0]
+                sample = sample.repeat(reps, 1)[:common]
             sample = torch.cat(comm.all_gather(sample.contiguous()), dim=0)
         qs = torch.linspace(0, 1, n_bins, device=sample.device)[1:]  # n_bins-1 cuts
         sample = torch.nan_to_num(sample, nan=float("inf"))

@@ -642,3 +642,49 @@ def test_distributed_early_stopping_agrees():
     assert a["best"] == b["best"]
     assert a["model"] == b["model"]
     assert a["best"] is not None and a["best"] < 60  # it actually stopped
+
+
+def _worker_uneven(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        torch.distributed.init_process_group(
+            "gloo", init_method=f"tcp://127.0.0.1:{port}",
+            rank=rank, world_size=world)
+        from mmlspark_amd.models.gbdt.objectives import make_objective
+        from mmlspark_amd.models.gbdt.trainer import TrainConfig, train_booster
+        from mmlspark_amd.parallel.comm import Comm
+        X, y = _make_data(0, n=4000)
+        bounds = [0, 1800, 3300, 4000]  # deliberately unequal shards
+        sl = slice(bounds[rank], bounds[rank + 1])
+        cfg = TrainConfig(num_iterations=6, num_leaves=15, seed=7)
+        booster, _ = train_booster(torch.from_numpy(X[sl]),
+                                   torch.from_numpy(y[sl]), cfg,
+                                   make_objective("binary"), Comm())
+        q.put((rank, booster.save_to_string()))
+        torch.distributed.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"ERROR: {e!r}"))
+
+
+@pytest.mark.timeout(300)
+def test_distributed_uneven_shards_ws3():
+    """World size 3 with unequal shard sizes (1800/1500/700): reduced
+    histograms weight rows correctly regardless of shard balance and all
+    ranks grow the identical booster — the empty/uneven-partition
+    robustness of LightGBMBase.scala:346-354 in distributed form."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29951
+    procs = [ctx.Process(target=_worker_uneven, args=(r, 3, port, q))
+             for r in range(3)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(3):
+        rank, s = q.get(timeout=250)
+        results[rank] = s
+    for p in procs:
+        p.join(timeout=30)
+    assert not any(str(s).startswith("ERROR") for s in results.values()), \
+        results
+    assert results[0] == results[1] == results[2]
