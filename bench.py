@@ -180,7 +180,10 @@ def run_session(args, comm, rank, world, use_gpu, device, X, y, n, nf,
                 "parallelism": (f"dp{world}" if args.parallelism ==
                                 "data_parallel" else
                                 f"voting{world} topK20"),
-                "sync": "RCCL histogram all_reduce over xGMI" if world > 1
+                "sync": (("RCCL" if use_gpu else
+                          torch.distributed.get_backend())
+                         + " histogram all_reduce"
+                         + (" over xGMI" if use_gpu else "")) if world > 1
                         else "single rank",
                 "categorical": args.categorical,
                 "matrix": (f"csr nnz={args.nnz}" if args.sparse else "dense"),
