@@ -952,9 +952,21 @@ def train_booster(X: torch.Tensor, y: torch.Tensor, cfg: TrainConfig,
     t_start = time.perf_counter()
     start_it = 0
     if checkpoint_dir:
-        ck = load_checkpoint(checkpoint_dir)
-        if ck is not None:
-            init_booster, start_it = ck
+        # resume state must be IDENTICAL on every rank or collective counts
+        # diverge and the gang deadlocks — rank 0's checkpoint view wins
+        # (checkpoint_dir may be node-local; rank>0 may see nothing)
+        if comm.is_distributed:
+            ck0 = load_checkpoint(checkpoint_dir) if comm.rank == 0 else None
+            obj = comm.all_gather_object(
+                None if ck0 is None
+                else (ck0[0].save_to_string(), ck0[1]))[0]
+            if obj is not None:
+                init_booster = Booster.load_from_string(obj[0])
+                start_it = obj[1]
+        else:
+            ck = load_checkpoint(checkpoint_dir)
+            if ck is not None:
+                init_booster, start_it = ck
     session = TrainingSession(X, y, cfg, objective, comm, weight=weight,
                               group_sizes=group_sizes,
                               feature_names=feature_names,

@@ -203,3 +203,71 @@ def test_distributed_checkpoint_resume_ws2(tmp_path):
     t_resumed = _json.loads(res8[0])["trees"][:4]
     t_before = _json.loads(res4[0])["trees"]
     assert t_resumed == t_before
+
+
+def _worker_resume_rank0_only(rank, world, port, q, ckdir):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        torch.distributed.init_process_group(
+            "gloo", init_method=f"tcp://127.0.0.1:{port}",
+            rank=rank, world_size=world)
+        from mmlspark_amd.models.gbdt.objectives import make_objective
+        from mmlspark_amd.models.gbdt.trainer import TrainConfig, train_booster
+        from mmlspark_amd.parallel.comm import Comm
+        rng = np.random.default_rng(0)
+        X = rng.normal(size=(2000, 8)).astype(np.float32)
+        y = (X[:, 0] > 0).astype(np.float32)
+        sl = slice(rank * 1000, (rank + 1) * 1000)
+        cfg = TrainConfig(num_iterations=6, num_leaves=7, seed=3)
+        # only rank 0 can see the checkpoint dir (node-local disk shape)
+        ck = ckdir if rank == 0 else os.path.join(ckdir, "nonexistent")
+        booster, _ = train_booster(torch.from_numpy(X[sl]),
+                                   torch.from_numpy(y[sl]), cfg,
+                                   make_objective("binary"), Comm(),
+                                   checkpoint_dir=ck, checkpoint_every=2)
+        q.put((rank, booster.save_to_string()))
+        torch.distributed.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"ERROR: {e!r}"))
+
+
+@pytest.mark.timeout(240)
+def test_resume_with_rank0_only_checkpoint(tmp_path):
+    """Rank 0 holds a 4-iteration checkpoint, rank 1 sees nothing (node-
+    local disks): rank 0's resume state must be broadcast so both ranks
+    start at the same iteration instead of deadlocking on mismatched
+    collective counts, and end with identical boosters."""
+    import multiprocessing as mp
+    from mmlspark_amd.models.gbdt.objectives import make_objective
+    from mmlspark_amd.models.gbdt.trainer import (TrainConfig, _save_checkpoint,
+                                                  train_booster)
+    from mmlspark_amd.parallel.comm import Comm
+    ckdir = str(tmp_path / "ck")
+    # seed the checkpoint: a 4-iteration single-rank run on the SAME data
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(2000, 8)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.float32)
+    pre, _ = train_booster(torch.from_numpy(X[:1000]),
+                           torch.from_numpy(y[:1000]),
+                           TrainConfig(num_iterations=4, num_leaves=7, seed=3),
+                           make_objective("binary"), Comm())
+    _save_checkpoint(ckdir, pre, 4)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29961
+    procs = [ctx.Process(target=_worker_resume_rank0_only,
+                         args=(r, 2, port, q, ckdir)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, s = q.get(timeout=180)
+        results[rank] = s
+    for p in procs:
+        p.join(timeout=30)
+    assert not any(str(s).startswith("ERROR") for s in results.values()), \
+        results
+    assert results[0] == results[1]
+    import json as _json
+    assert len(_json.loads(results[0])["trees"]) == 6  # 4 resumed + 2 new
