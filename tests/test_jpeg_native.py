@@ -593,3 +593,29 @@ def test_progressive_eobn_runs_decode():
     flat = np.full((64, 64, 3), 128, np.uint8)
     tiny = encode_progressive_eobn(flat, quality=25)
     assert len(tiny) < len(jpeg_codec.encode_jpeg(flat, quality=25)) + 4096
+
+
+def test_malformed_streams_never_crash():
+    """Truncated / bit-flipped / garbage streams must either decode or
+    raise cleanly (RuntimeError→ValueError contract) — never segfault the
+    serving process.  Seeded 150-case fuzz."""
+    img = (np.random.default_rng(0).integers(0, 255, (48, 48, 3))
+           ).astype(np.uint8)
+    enc = jpeg_codec.encode_jpeg(img, quality=80)
+    rng = np.random.default_rng(1)
+    for trial in range(150):
+        b = bytearray(enc)
+        mode = trial % 3
+        if mode == 0:
+            b = b[:rng.integers(2, len(b))]
+        elif mode == 1:
+            for _ in range(rng.integers(1, 8)):
+                b[rng.integers(0, len(b))] = rng.integers(0, 256)
+        else:
+            b = bytes([0xFF, 0xD8]) + bytes(
+                rng.integers(0, 256, rng.integers(10, 300)).astype(np.uint8))
+        try:
+            out = _jpeg_native.decode_jpeg(bytes(b))
+            assert out.numel() >= 0
+        except RuntimeError:
+            pass  # clean rejection
