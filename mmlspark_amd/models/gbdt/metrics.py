@@ -67,6 +67,10 @@ def default_metrics_fn(metric_name=None):
             t = torch.tensor([s, n], device=Xv.device)
             comm.all_reduce(t)
             out["binary_logloss"] = float(t[0] / t[1])
+            # distributed note: this is the MEAN of per-rank AUCs over each
+            # rank's validation shard — deterministic and identical on every
+            # rank (what early stopping needs), but not the pooled global
+            # AUC; pass a replicated validation set for an exact global AUC
             a = auc(p.cpu().numpy(), y.cpu().numpy())
             t = torch.tensor([a, 1.0], device=Xv.device)
             comm.all_reduce(t)
