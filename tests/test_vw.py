@@ -429,3 +429,22 @@ def test_contextual_bandit_additional_shared_features():
     out0 = cb0.transform(df)
     acc0 = ((out0["prediction"].to_numpy() - 1) == df["ctx"].to_numpy()).mean()
     assert acc0 < 0.6
+
+
+def test_contextual_bandit_metrics_ips_snips():
+    """IPS/SNIPS off-policy estimators: with uniform logging (p=1/K) and a
+    deterministic target policy, IPS must be unbiased for the target
+    policy's expected cost and SNIPS must match it up to normalization."""
+    from mmlspark_amd.models.vw.estimators import ContextualBanditMetrics
+    rng = np.random.default_rng(0)
+    K = 4
+    # true per-action costs; target policy always picks action 0 (cost 0.2)
+    costs = np.array([0.2, 0.8, 0.5, 1.0])
+    m = ContextualBanditMetrics()
+    for _ in range(20000):
+        logged = int(rng.integers(0, K))
+        m.add(prob_logged=1.0 / K, cost=float(costs[logged]),
+              prob_pred_matches=1.0 if logged == 0 else 0.0)
+    assert abs(m.ips_estimate - costs[0]) < 0.03
+    assert abs(m.snips_estimate - costs[0]) < 0.03
+    assert m.total == 20000
