@@ -505,3 +505,32 @@ def test_leaf_output_zero_hessian_guard():
     assert _leaf_output(1.5, 0.0, cfg) == 0.0
     assert _leaf_output(-1.5, 0.0, cfg) == 0.0
     assert _leaf_output(1.5, 2.0, cfg) == -0.75
+
+
+def test_booster_merge_appends_trees():
+    """merge (mergeBooster:252) appends trees: the merged model's raw score
+    equals the sum of both models' raws (additive forests, same objective)."""
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(400, 5)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.float64)
+    df = pd.DataFrame({"features": list(X), "label": y})
+    m1 = LightGBMClassifier(numIterations=3, numLeaves=7, seed=1).fit(df)
+    m2 = LightGBMClassifier(numIterations=2, numLeaves=7, seed=2).fit(df)
+    b1, b2 = m1.booster, m2.booster
+    Xt = torch.from_numpy(X)
+    r1 = b1.predict_raw(Xt).clone()
+    r2 = b2.predict_raw(Xt).clone()
+    import copy
+    merged = copy.deepcopy(b1).merge(b2)
+    assert merged.num_trees == b1.num_trees + b2.num_trees
+    # merged keeps SELF's base_score and appends the other's trees, so the
+    # expected raw is r1 + (r2 - base2) — the numBatches continuation
+    # semantics (batch-2 boosters carry their lift in trees, not base)
+    np.testing.assert_allclose(
+        merged.predict_raw(Xt).numpy(),
+        (r1 + r2).numpy() - b2.base_score, rtol=1e-5, atol=1e-5)
+    # merged model round-trips through the JSON format
+    from mmlspark_amd.models.gbdt.booster import Booster
+    back = Booster.load_from_string(merged.save_to_string())
+    np.testing.assert_allclose(back.predict_raw(Xt).numpy(),
+                               merged.predict_raw(Xt).numpy())
