@@ -763,15 +763,28 @@ class TrainingSession:
         if not self.cfg.boost_from_average:  # boostFromAverage=False
             return torch.zeros(1, device=device)
         if comm.is_distributed:
+            # the stored base_score must be IDENTICAL on every rank — a
+            # shard-local init would make per-rank models differ even though
+            # the reduced-histogram trees agree
             w = weight if weight is not None else torch.ones_like(
                 y, dtype=torch.float32)
             sums = torch.stack([(y.float() * w).sum(), w.sum()]).to(device)
             comm.all_reduce(sums)
+            mean = sums[0] / sums[1].clamp_min(1e-12)
             if obj.name == "binary":
-                mean = (sums[0] / sums[1]).clamp(1e-6, 1 - 1e-6)
-                return torch.log(mean / (1 - mean)).reshape(1)
+                p = mean.clamp(1e-6, 1 - 1e-6)
+                sig = getattr(obj, "sigmoid", 1.0)
+                return (torch.log(p / (1 - p)) / sig).reshape(1)
             if obj.name in ("regression", "regression_l2"):
-                return (sums[0] / sums[1]).reshape(1)
+                return mean.reshape(1)
+            if obj.name in ("poisson", "tweedie"):
+                return torch.log(mean.clamp_min(1e-8)).reshape(1)
+            # every other objective (l1 median, quantile, custom fobj):
+            # average the shard-local inits — approximate but identical on
+            # all ranks, and the first boosting round absorbs the error
+            s0 = obj.init_score(y, weight).to(device).float()
+            comm.all_reduce(s0)
+            return s0 / comm.world_size
         return obj.init_score(y, weight).to(device)
 
     # ------------------------------------------------------------------ step

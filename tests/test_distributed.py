@@ -688,3 +688,52 @@ def test_distributed_uneven_shards_ws3():
     assert not any(str(s).startswith("ERROR") for s in results.values()), \
         results
     assert results[0] == results[1] == results[2]
+
+
+def _worker_objective(rank, world, port, q, objective):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        torch.distributed.init_process_group(
+            "gloo", init_method=f"tcp://127.0.0.1:{port}",
+            rank=rank, world_size=world)
+        from mmlspark_amd.models.gbdt.objectives import make_objective
+        from mmlspark_amd.models.gbdt.trainer import TrainConfig, train_booster
+        from mmlspark_amd.parallel.comm import Comm
+        rng = np.random.default_rng(0)
+        X = rng.normal(size=(3000, 6)).astype(np.float32)
+        y = np.exp(X[:, 0] * 0.5 + rng.normal(size=3000) * 0.2) \
+            .astype(np.float32)  # positive labels for poisson/tweedie
+        # deliberately skewed shards so shard-local label stats DIFFER
+        sl = slice(0, 2000) if rank == 0 else slice(2000, 3000)
+        cfg = TrainConfig(num_iterations=4, num_leaves=15, seed=7)
+        booster, _ = train_booster(torch.from_numpy(X[sl]),
+                                   torch.from_numpy(y[sl]), cfg,
+                                   make_objective(objective), Comm())
+        q.put((rank, booster.save_to_string()))
+        torch.distributed.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"ERROR: {e!r}"))
+
+
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("objective", ["poisson", "tweedie", "regression_l1"])
+def test_distributed_init_score_consistency(objective):
+    """Objectives whose init score depends on label statistics (log-mean,
+    median): skewed shards give different LOCAL stats, yet the stored
+    base_score — and therefore the whole serialized model — must agree."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29971 + {"poisson": 0, "tweedie": 1, "regression_l1": 2}[objective]
+    procs = [ctx.Process(target=_worker_objective,
+                         args=(r, 2, port, q, objective)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, s = q.get(timeout=250)
+        results[rank] = s
+    for p in procs:
+        p.join(timeout=30)
+    assert not any(str(s).startswith("ERROR") for s in results.values()), \
+        results
+    assert results[0] == results[1]
