@@ -534,3 +534,29 @@ def test_booster_merge_appends_trees():
     back = Booster.load_from_string(merged.save_to_string())
     np.testing.assert_allclose(back.predict_raw(Xt).numpy(),
                                merged.predict_raw(Xt).numpy())
+
+
+def test_nan_features_train_and_predict():
+    """NaN feature values must train (binned to the top bin) and score
+    (routed like the top bin: raw <= thr is false) without poisoning
+    outputs — the reference's useMissing/NaN tolerance."""
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(2000, 6)).astype(np.float32)
+    y = (X[:, 0] + X[:, 1] > 0).astype(np.float64)
+    Xn = X.copy()
+    mask = rng.random(X.shape) < 0.1
+    mask[:, 0] = False  # keep the signal feature mostly intact
+    Xn[mask] = np.nan
+    df = pd.DataFrame({"features": list(Xn), "label": y})
+    m = LightGBMClassifier(numIterations=20, numLeaves=15).fit(df)
+    out = m.transform(df)
+    prob = np.stack(out["probability"].to_numpy())
+    assert np.isfinite(prob).all()
+    acc = (out["prediction"].to_numpy() == y).mean()
+    assert acc > 0.85, acc
+    # NaN routes exactly like +inf at predict time (both fail raw <= thr)
+    row_nan = Xn[:1].copy(); row_nan[0, 2] = np.nan
+    row_inf = Xn[:1].copy(); row_inf[0, 2] = np.inf
+    p1 = m.booster.predict_raw(torch.from_numpy(np.nan_to_num(row_nan, nan=np.nan)))
+    p2 = m.booster.predict_raw(torch.from_numpy(row_inf))
+    np.testing.assert_allclose(p1.numpy(), p2.numpy())
