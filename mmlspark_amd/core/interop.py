@@ -41,6 +41,22 @@ def _is_spark(obj) -> bool:
     return m.startswith("pyspark.sql")
 
 
+_IMAGE_FIELDS = {"origin", "height", "width", "nChannels", "mode", "data"}
+
+
+def _image_struct_to_obj(v):
+    """Spark ImageSchema struct → HWC uint8 ndarray (the layout every
+    image stage here consumes; ImageSchemaUtils / ImageUtils parity —
+    Spark stores BGR row-major bytes)."""
+    if v is None:
+        return None
+    h, w, c = int(v["height"]), int(v["width"]), int(v["nChannels"])
+    buf = v["data"]
+    arr = np.frombuffer(bytes(buf), dtype=np.uint8)[: h * w * c]
+    a = arr.reshape(h, w, c)
+    return a[:, :, 0] if c == 1 else a
+
+
 def _vector_struct_to_obj(v):
     """SparkML VectorUDT struct (dict after Arrow/pandas conversion)."""
     if v is None:
@@ -85,6 +101,11 @@ def arrow_to_pandas(table) -> pd.DataFrame:
                 f.name for f in typ}:
             out[name] = pd.Series(
                 [_vector_struct_to_obj(v) for v in col.to_pylist()],
+                dtype=object)
+        elif pa.types.is_struct(typ) and _IMAGE_FIELDS <= {
+                f.name for f in typ}:
+            out[name] = pd.Series(
+                [_image_struct_to_obj(v) for v in col.to_pylist()],
                 dtype=object)
         else:
             out[name] = col.to_pandas()

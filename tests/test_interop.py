@@ -147,3 +147,25 @@ def test_arrow_nulls_and_chunked_tables():
     back = arrow_to_pandas(both)
     assert len(back) == 10
     np.testing.assert_allclose(back["features"].iloc[9], np.full(3, 9.0))
+
+
+def test_arrow_image_schema_struct_decodes():
+    """Spark ImageSchema structs (origin/height/width/nChannels/mode/data)
+    arriving through Arrow become HWC uint8 arrays the image stages use
+    (ImageSchemaUtils parity)."""
+    rng = np.random.default_rng(0)
+    imgs = [rng.integers(0, 255, (8, 6, 3)).astype(np.uint8)
+            for _ in range(3)]
+    rows = [{"origin": f"mem://{i}", "height": 8, "width": 6,
+             "nChannels": 3, "mode": 16,
+             "data": im.tobytes()} for i, im in enumerate(imgs)]
+    t = pa.table({"image": pa.array(rows)})
+    back = arrow_to_pandas(t)
+    got = back["image"].iloc[1]
+    assert got.shape == (8, 6, 3) and got.dtype == np.uint8
+    np.testing.assert_array_equal(got, imgs[1])
+    # flows straight into ImageTransformer
+    from mmlspark_amd.models.images import ImageTransformer
+    out = ImageTransformer(inputCol="image", outputCol="resized") \
+        .resize(4, 4).transform(back)
+    assert out["resized"].iloc[0].shape[:2] == (4, 4)
