@@ -121,6 +121,20 @@ def pandas_to_arrow(df: pd.DataFrame):
                 s.dropna().iloc[0] if s.notna().any() else None,
                 (np.ndarray, list, SparseVector)):
             first = s.dropna().iloc[0]
+            if isinstance(first, np.ndarray) and first.ndim >= 2 \
+                    and first.dtype == np.uint8:
+                # image arrays → ImageSchema structs (round-trip symmetry)
+                def _to_img(v):
+                    if v is None:
+                        return None
+                    a = v if v.ndim == 3 else v[:, :, None]
+                    return {"origin": "", "height": a.shape[0],
+                            "width": a.shape[1], "nChannels": a.shape[2],
+                            "mode": 16 if a.shape[2] == 3 else 0,
+                            "data": a.tobytes()}
+                arrays.append(pa.array([_to_img(v) for v in s]))
+                names.append(str(name))
+                continue
             if isinstance(first, SparseVector):
                 arrays.append(pa.array(
                     [None if v is None else

@@ -169,3 +169,17 @@ def test_arrow_image_schema_struct_decodes():
     out = ImageTransformer(inputCol="image", outputCol="resized") \
         .resize(4, 4).transform(back)
     assert out["resized"].iloc[0].shape[:2] == (4, 4)
+
+
+def test_arrow_image_round_trip():
+    """uint8 HWC image columns survive pandas → Arrow → pandas unchanged
+    (encoded as ImageSchema structs on the way out)."""
+    rng = np.random.default_rng(1)
+    imgs = [rng.integers(0, 255, (5, 7, 3)).astype(np.uint8)
+            for _ in range(4)]
+    df = pd.DataFrame({"image": imgs, "k": list(range(4))})
+    t = pandas_to_arrow(df)
+    assert pa.types.is_struct(t.schema.field("image").type)
+    back = arrow_to_pandas(t)
+    for a, b in zip(back["image"], imgs):
+        np.testing.assert_array_equal(a, b)
