@@ -162,7 +162,12 @@ def spark_to_pandas(sdf) -> pd.DataFrame:
         if pdf[c].dtype == object and len(pdf):
             v0 = pdf[c].dropna()
             v0 = v0.iloc[0] if len(v0) else None
-            if isinstance(v0, DenseVector):
+            if hasattr(v0, "asDict") and _IMAGE_FIELDS <= set(
+                    v0.asDict().keys()):  # ImageSchema Row
+                pdf[c] = pdf[c].map(
+                    lambda v: None if v is None
+                    else _image_struct_to_obj(v.asDict()))
+            elif isinstance(v0, DenseVector):
                 pdf[c] = pdf[c].map(
                     lambda v: None if v is None
                     else np.asarray(v.toArray(), dtype=np.float32))
