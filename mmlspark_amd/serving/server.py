@@ -616,11 +616,13 @@ class ProcessServingCluster:
         round-robin rotation (driver-rendezvous parity).  Idempotent on
         (host, port)."""
         name = name or f"remote-{port}"
-        for w in self.workers:
-            if w.host == host and w.port == int(port):
-                w.dead = False
-                return {"registered": True, "name": w.name, "known": True}
-        self.workers.append(_RemoteWorker(host, port, name))
+        with self._rr_lock:  # registrations can race from HTTP threads
+            for w in self.workers:
+                if w.host == host and w.port == int(port):
+                    w.dead = False
+                    return {"registered": True, "name": w.name,
+                            "known": True}
+            self.workers.append(_RemoteWorker(host, port, name))
         return {"registered": True, "name": name, "known": False}
 
     def kill_worker(self, i: int):
